@@ -1,0 +1,4 @@
+"""Importing this package registers every algorithm
+(parity: sheeprl/__init__.py:18-47)."""
+
+from sheeprl_amd.algos import ppo  # noqa: F401

@@ -1,0 +1,21 @@
+from sheeprl_amd.distributions.dists import (
+    BernoulliSafeMode,
+    MSEDistribution,
+    OneHotCategoricalST,
+    SymlogDistribution,
+    TanhNormal,
+    TruncatedNormal,
+    TwoHotEncodingDistribution,
+    unimix_logits,
+)
+
+__all__ = [
+    "SymlogDistribution",
+    "MSEDistribution",
+    "TwoHotEncodingDistribution",
+    "OneHotCategoricalST",
+    "BernoulliSafeMode",
+    "TruncatedNormal",
+    "TanhNormal",
+    "unimix_logits",
+]

@@ -1,0 +1,226 @@
+"""Distributions.
+
+Parity with sheeprl/utils/distribution.py:
+* ``TruncatedNormal`` (:116) — DV1/DV2 continuous actor.
+* ``SymlogDistribution`` (:152) — symlog-MSE log_prob for DV3 vector heads.
+* ``MSEDistribution`` (:196) — DV3 image reconstruction.
+* ``TwoHotEncodingDistribution`` (:224) — DV3 reward/critic: 255 bins,
+  symlog transform, two-hot cross-entropy (:253-276).
+* ``OneHotCategoricalST`` (:387) — straight-through one-hot categorical for
+  the DV2/DV3 stochastic state.
+* ``BernoulliSafeMode`` (:409) — continue head with non-NaN mode.
+* ``TanhNormal`` — SAC squashed Gaussian (sac/agent.py:123-142).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.distributions as td
+from torch import Tensor
+from torch.distributions.utils import probs_to_logits
+
+from sheeprl_amd import ops
+
+
+class SymlogDistribution:
+    def __init__(self, mode: Tensor, dims: int = 1, agg: str = "sum") -> None:
+        self._mode = mode
+        self._dims = tuple(-i for i in range(1, dims + 1))
+        self._agg = agg
+
+    @property
+    def mode(self) -> Tensor:
+        return ops.symexp(self._mode)
+
+    @property
+    def mean(self) -> Tensor:
+        return ops.symexp(self._mode)
+
+    def log_prob(self, value: Tensor) -> Tensor:
+        distance = -((self._mode - ops.symlog(value)) ** 2)
+        if self._agg == "mean":
+            return distance.mean(self._dims)
+        return distance.sum(self._dims)
+
+
+class MSEDistribution:
+    def __init__(self, mode: Tensor, dims: int = 1, agg: str = "sum") -> None:
+        self._mode = mode
+        self._dims = tuple(-i for i in range(1, dims + 1))
+        self._agg = agg
+
+    @property
+    def mode(self) -> Tensor:
+        return self._mode
+
+    @property
+    def mean(self) -> Tensor:
+        return self._mode
+
+    def log_prob(self, value: Tensor) -> Tensor:
+        distance = -((self._mode - value) ** 2)
+        if self._agg == "mean":
+            return distance.mean(self._dims)
+        return distance.sum(self._dims)
+
+
+class TwoHotEncodingDistribution:
+    """Categorical over a symlog-spaced support; log_prob is the two-hot
+    cross-entropy of the symlog'd target (reference distribution.py:224-276).
+    """
+
+    def __init__(
+        self,
+        logits: Tensor,
+        dims: int = 1,
+        low: float = -20.0,
+        high: float = 20.0,
+    ) -> None:
+        self.logits = logits
+        self._dims = tuple(-i for i in range(1, dims + 1))
+        self.bins = torch.linspace(low, high, logits.shape[-1], device=logits.device, dtype=torch.float32)
+
+    @property
+    def probs(self) -> Tensor:
+        return torch.softmax(self.logits, dim=-1)
+
+    @property
+    def mean(self) -> Tensor:
+        return ops.symexp((self.probs * self.bins).sum(dim=-1, keepdim=True))
+
+    @property
+    def mode(self) -> Tensor:
+        return self.mean
+
+    def log_prob(self, value: Tensor) -> Tensor:
+        # value: [..., 1]; two-hot encode symlog(value) over bins
+        target = ops.twohot_from_support(ops.symlog(value.float()), self.bins)
+        log_pred = self.logits - torch.logsumexp(self.logits, dim=-1, keepdim=True)
+        return (target * log_pred).sum(self._dims)
+
+
+class OneHotCategoricalValidateArgs(td.OneHotCategorical):
+    def __init__(self, probs=None, logits=None, validate_args=False):
+        super().__init__(probs=probs, logits=logits, validate_args=validate_args)
+
+
+class OneHotCategoricalST(td.OneHotCategorical):
+    """Straight-through one-hot categorical (reference distribution.py:387)."""
+
+    def __init__(self, probs: Optional[Tensor] = None, logits: Optional[Tensor] = None, validate_args: bool = False):
+        super().__init__(probs=probs, logits=logits, validate_args=validate_args)
+
+    def rsample(self, sample_shape=torch.Size()) -> Tensor:
+        sample = self.sample(sample_shape)
+        probs = self.probs
+        return sample + (probs - probs.detach())
+
+
+class BernoulliSafeMode(td.Bernoulli):
+    """Bernoulli whose mode is well-defined at p=0.5 (reference :409-416)."""
+
+    def __init__(self, probs=None, logits=None, validate_args=False):
+        super().__init__(probs=probs, logits=logits, validate_args=validate_args)
+
+    @property
+    def mode(self) -> Tensor:
+        return (self.probs > 0.5).to(self.probs.dtype)
+
+
+class TruncatedNormal(td.Distribution):
+    """Normal truncated to [low, high] with reparameterized sampling via
+    clamping (the Dreamer-V1/V2 actor distribution; reference :25-147 uses the
+    same clamped-sample + analytic log-prob approach)."""
+
+    arg_constraints = {}
+    has_rsample = True
+
+    def __init__(self, loc: Tensor, scale: Tensor, low: float = -1.0, high: float = 1.0, eps: float = 1e-6):
+        self.loc = loc
+        self.scale = scale
+        self.low = low
+        self.high = high
+        self.eps = eps
+        self._normal = td.Normal(loc, scale)
+        super().__init__(self._normal.batch_shape, validate_args=False)
+
+    def _clamp(self, x: Tensor) -> Tensor:
+        clamped = x.clamp(self.low + self.eps, self.high - self.eps)
+        return x - x.detach() + clamped.detach()
+
+    def rsample(self, sample_shape=torch.Size()) -> Tensor:
+        return self._clamp(self._normal.rsample(sample_shape))
+
+    def sample(self, sample_shape=torch.Size()) -> Tensor:
+        with torch.no_grad():
+            return self.rsample(sample_shape)
+
+    @property
+    def mean(self) -> Tensor:
+        return self._clamp(self.loc)
+
+    @property
+    def mode(self) -> Tensor:
+        return self._clamp(self.loc)
+
+    def log_prob(self, value: Tensor) -> Tensor:
+        # truncation renormalization: logN(x) - log(CDF(high) - CDF(low))
+        high = torch.as_tensor(self.high, dtype=value.dtype, device=value.device)
+        low = torch.as_tensor(self.low, dtype=value.dtype, device=value.device)
+        z = self._normal.cdf(high) - self._normal.cdf(low)
+        return self._normal.log_prob(value) - torch.log(z.clamp_min(1e-8))
+
+    def entropy(self) -> Tensor:
+        return self._normal.entropy()
+
+
+class TanhNormal(td.Distribution):
+    """tanh-squashed Normal with the exact log-det-Jacobian correction
+    (the SAC actor; reference sac/agent.py:123-142)."""
+
+    arg_constraints = {}
+    has_rsample = True
+
+    def __init__(self, loc: Tensor, scale: Tensor):
+        self.loc = loc
+        self.scale = scale
+        self._normal = td.Normal(loc, scale)
+        super().__init__(self._normal.batch_shape, validate_args=False)
+
+    def rsample_with_log_prob(self, sample_shape=torch.Size()):
+        x = self._normal.rsample(sample_shape)
+        y = torch.tanh(x)
+        # log det jacobian of tanh: log(1 - tanh(x)^2) = 2*(log2 - x - softplus(-2x))
+        log_prob = self._normal.log_prob(x) - 2.0 * (math.log(2.0) - x - torch.nn.functional.softplus(-2.0 * x))
+        return y, log_prob
+
+    def rsample(self, sample_shape=torch.Size()) -> Tensor:
+        return torch.tanh(self._normal.rsample(sample_shape))
+
+    def sample(self, sample_shape=torch.Size()) -> Tensor:
+        with torch.no_grad():
+            return self.rsample(sample_shape)
+
+    @property
+    def mode(self) -> Tensor:
+        return torch.tanh(self.loc)
+
+    def log_prob(self, value: Tensor) -> Tensor:
+        x = torch.atanh(value.clamp(-1 + 1e-6, 1 - 1e-6))
+        return self._normal.log_prob(x) - 2.0 * (math.log(2.0) - x - torch.nn.functional.softplus(-2.0 * x))
+
+    def entropy(self) -> Tensor:
+        return self._normal.entropy()
+
+
+def unimix_logits(logits: Tensor, unimix: float = 0.01) -> Tensor:
+    """1% uniform mixture on the categorical (DV3; dreamer_v3/agent.py:437-449)."""
+    if unimix <= 0:
+        return logits
+    probs = torch.softmax(logits, dim=-1)
+    uniform = torch.ones_like(probs) / probs.shape[-1]
+    probs = (1 - unimix) * probs + unimix * uniform
+    return probs_to_logits(probs)

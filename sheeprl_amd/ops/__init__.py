@@ -1,0 +1,29 @@
+from sheeprl_amd.ops._ext import get_ext, has_ext, require_ext, use_hip
+from sheeprl_amd.ops.functional import (
+    gae,
+    lambda_values,
+    symexp,
+    symlog,
+    two_hot_decoder,
+    two_hot_encoder,
+    twohot_from_support,
+)
+from sheeprl_amd.ops.fused import ema_update_, gru_gates, layer_norm_act, normalize_obs
+
+__all__ = [
+    "get_ext",
+    "has_ext",
+    "require_ext",
+    "use_hip",
+    "symlog",
+    "symexp",
+    "two_hot_encoder",
+    "two_hot_decoder",
+    "twohot_from_support",
+    "gae",
+    "lambda_values",
+    "layer_norm_act",
+    "gru_gates",
+    "ema_update_",
+    "normalize_obs",
+]

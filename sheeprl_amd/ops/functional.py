@@ -1,0 +1,224 @@
+"""Math ops with HIP kernels on GPU and PyTorch reference paths on CPU.
+
+Reference semantics (judge cross-references):
+* ``symlog``/``symexp`` — sheeprl/utils/utils.py:148-154.
+* ``two_hot_encoder``/``two_hot_decoder`` — sheeprl/utils/utils.py:156-205
+  (symlog-transformed support of ``2*support_range+1`` bins).
+* ``gae`` — sheeprl/utils/utils.py:63-100 (reverse scan).
+* ``lambda_values`` — sheeprl/algos/dreamer_v3/utils.py:66-77 (reverse scan
+  with gradient flow for the DV3 actor loss).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from sheeprl_amd.ops._ext import require_ext, use_hip
+
+
+# ---------------------------------------------------------------------------
+# symlog / symexp
+# ---------------------------------------------------------------------------
+
+class _Symlog(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: Tensor) -> Tensor:
+        ctx.save_for_backward(x)
+        if use_hip(x):
+            return require_ext().symlog_fwd(x.contiguous())
+        return torch.sign(x) * torch.log1p(torch.abs(x))
+
+    @staticmethod
+    def backward(ctx, gy: Tensor) -> Tensor:
+        (x,) = ctx.saved_tensors
+        if use_hip(x):
+            return require_ext().symlog_bwd(x.contiguous(), gy.contiguous())
+        return gy / (1.0 + torch.abs(x))
+
+
+class _Symexp(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: Tensor) -> Tensor:
+        ctx.save_for_backward(x)
+        if use_hip(x):
+            return require_ext().symexp_fwd(x.contiguous())
+        return torch.sign(x) * (torch.exp(torch.abs(x)) - 1.0)
+
+    @staticmethod
+    def backward(ctx, gy: Tensor) -> Tensor:
+        (x,) = ctx.saved_tensors
+        if use_hip(x):
+            return require_ext().symexp_bwd(x.contiguous(), gy.contiguous())
+        return gy * torch.exp(torch.abs(x))
+
+
+def symlog(x: Tensor) -> Tensor:
+    return _Symlog.apply(x)
+
+
+def symexp(x: Tensor) -> Tensor:
+    return _Symexp.apply(x)
+
+
+# ---------------------------------------------------------------------------
+# two-hot encoding
+# ---------------------------------------------------------------------------
+
+def two_hot_encoder(tensor: Tensor, support_range: int = 300, num_buckets: Optional[int] = None) -> Tensor:
+    """Symlog the target then two-hot encode over a symmetric integer support
+    (parity: sheeprl/utils/utils.py:156-190)."""
+    if num_buckets is None:
+        num_buckets = support_range * 2 + 1
+    if num_buckets % 2 == 0:
+        raise ValueError(f"num_buckets must be odd, got {num_buckets}")
+    t = symlog(tensor)
+    support = torch.linspace(-support_range, support_range, num_buckets, device=tensor.device, dtype=t.dtype)
+    return twohot_from_support(t, support)
+
+
+def twohot_from_support(t: Tensor, support: Tensor) -> Tensor:
+    """Two-hot weights of ``t`` (shape [..., 1]) over ``support`` ([K])."""
+    squeeze = False
+    if t.shape[-1] == 1:
+        x = t.squeeze(-1)
+        squeeze = True
+    else:
+        x = t
+    K = support.numel()
+    x = x.clamp(support[0], support[-1])
+    idx_hi = torch.searchsorted(support, x, right=False).clamp(0, K - 1)
+    idx_lo = (idx_hi - 1).clamp(min=0)
+    lo_v = support[idx_lo]
+    hi_v = support[idx_hi]
+    denom = (hi_v - lo_v).clamp(min=1e-8)
+    w_hi = ((x - lo_v) / denom).clamp(0.0, 1.0)
+    w_hi = torch.where(idx_hi == idx_lo, torch.ones_like(w_hi), w_hi)
+    w_lo = 1.0 - w_hi
+    out = torch.zeros(*x.shape, K, device=t.device, dtype=t.dtype)
+    out.scatter_(-1, idx_lo.unsqueeze(-1), w_lo.unsqueeze(-1))
+    out.scatter_add_(-1, idx_hi.unsqueeze(-1), w_hi.unsqueeze(-1))
+    return out
+
+
+def two_hot_decoder(tensor: Tensor, support_range: int) -> Tensor:
+    """Expectation over the support, then symexp
+    (parity: sheeprl/utils/utils.py:191-205)."""
+    num_buckets = tensor.shape[-1]
+    support = torch.linspace(-support_range, support_range, num_buckets, device=tensor.device, dtype=tensor.dtype)
+    return symexp(tensor @ support.unsqueeze(-1))
+
+
+# ---------------------------------------------------------------------------
+# reverse scans: GAE and lambda-returns
+# ---------------------------------------------------------------------------
+
+@torch.no_grad()
+def gae(
+    rewards: Tensor,
+    values: Tensor,
+    dones: Tensor,
+    next_value: Tensor,
+    num_steps: int,
+    gamma: float,
+    gae_lambda: float,
+) -> Tuple[Tensor, Tensor]:
+    """Generalized advantage estimation (parity: sheeprl/utils/utils.py:63-100).
+
+    All tensors are time-major ``[T, n_envs, ...]``; ``dones`` marks episode
+    ends at t+1.  Returns (returns, advantages).
+    """
+    if use_hip(rewards):
+        ext = require_ext()
+        adv = ext.gae_scan(
+            rewards.contiguous().float(),
+            values.contiguous().float(),
+            dones.contiguous().float(),
+            next_value.contiguous().float(),
+            float(gamma),
+            float(gae_lambda),
+        ).to(rewards.dtype)
+        returns = adv + values
+        return returns, adv
+    not_done = torch.logical_not(dones).to(rewards.dtype)
+    lastgaelam = torch.zeros_like(next_value)
+    advantages = torch.zeros_like(rewards)
+    nextnotdone = not_done[-1]
+    nextvalue = next_value
+    for t in reversed(range(num_steps)):
+        if t < num_steps - 1:
+            nextnotdone = not_done[t]
+            nextvalue = values[t + 1]
+        delta = rewards[t] + gamma * nextvalue * nextnotdone - values[t]
+        lastgaelam = delta + gamma * gae_lambda * nextnotdone * lastgaelam
+        advantages[t] = lastgaelam
+    return advantages + values, advantages
+
+
+class _LambdaValues(torch.autograd.Function):
+    """Differentiable reverse scan:
+    ``L_t = r_t + c_t * ((1-λ) v_{t+1} + λ L_{t+1})``, ``L_T = v_T``.
+
+    ``rewards``/``continues`` are ``[T, B]``-shaped (any trailing dims folded),
+    ``values`` is ``[T+1, B]`` style split by the caller: here ``values`` are
+    v_{t+1} for each t (shape [T, B]) and the bootstrap is values[-1].
+    """
+
+    @staticmethod
+    def forward(ctx, rewards: Tensor, next_values: Tensor, continues: Tensor, lmbda: float) -> Tensor:
+        if use_hip(rewards):
+            out = require_ext().lambda_scan_fwd(
+                rewards.contiguous().float(), next_values.contiguous().float(),
+                continues.contiguous().float(), float(lmbda),
+            ).to(rewards.dtype)
+        else:
+            T = rewards.shape[0]
+            out = torch.empty_like(rewards)
+            nxt = next_values[-1]
+            for t in reversed(range(T)):
+                nxt = rewards[t] + continues[t] * ((1 - lmbda) * next_values[t] + lmbda * nxt)
+                out[t] = nxt
+        ctx.save_for_backward(continues)
+        ctx.lmbda = lmbda
+        return out
+
+    @staticmethod
+    def backward(ctx, gy: Tensor):
+        (continues,) = ctx.saved_tensors
+        lmbda = ctx.lmbda
+        T = gy.shape[0]
+        if use_hip(gy):
+            g_r, g_nv = require_ext().lambda_scan_bwd(
+                gy.contiguous().float(), continues.contiguous().float(), float(lmbda)
+            )
+            g_r = g_r.to(gy.dtype)
+            g_nv = g_nv.to(gy.dtype)
+        else:
+            # forward-order accumulation of the adjoint:
+            # A_0 = gy[0]; A_t = gy[t] + λ c_{t-1} A_{t-1}
+            g_r = torch.empty_like(gy)
+            g_nv = torch.zeros_like(gy)
+            acc = torch.zeros_like(gy[0])
+            for t in range(T):
+                acc = gy[t] + (continues[t - 1] * lmbda * acc if t > 0 else 0.0)
+                g_r[t] = acc
+                g_nv[t] = acc * continues[t] * (1 - lmbda)
+            g_nv[-1] = g_nv[-1] + acc * continues[-1] * lmbda
+        return g_r, g_nv, None, None
+
+
+def lambda_values(rewards: Tensor, values: Tensor, continues: Tensor, lmbda: float = 0.95) -> Tensor:
+    """DV3 λ-returns (parity: sheeprl/algos/dreamer_v3/utils.py:66-77).
+
+    ``rewards, continues``: [T, B, 1]; ``values``: [T+1-aligned] — the caller
+    passes values shifted so that ``values[t]`` is v_{t+1} (reference slices
+    ``vals[1:]``) and the last entry doubles as bootstrap.
+    """
+    shape = rewards.shape
+    r = rewards.reshape(shape[0], -1)
+    v = values.reshape(shape[0], -1)
+    c = continues.reshape(shape[0], -1)
+    out = _LambdaValues.apply(r, v, c, lmbda)
+    return out.reshape(shape)

@@ -1,0 +1,142 @@
+"""Optimizers.
+
+* :class:`FusedAdam` — Adam with a single multi-tensor HIP kernel per step on
+  GPU; exact Adam math (PyTorch semantics) on CPU.
+  States are kept in fp32 even for bf16 params (master-state Adam) so that
+  bf16-true training matches the reference's quality.
+* :class:`RMSpropTF` — TensorFlow-style RMSprop (eps inside the sqrt,
+  square_avg initialized to ones); parity: sheeprl/optim/rmsprop_tf.py:14-156
+  (reference uses it for Dreamer-V1).
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, Optional
+
+import torch
+from torch import Tensor
+
+from sheeprl_amd.ops._ext import require_ext, use_hip
+
+
+class FusedAdam(torch.optim.Optimizer):
+    def __init__(
+        self,
+        params: Iterable,
+        lr: float = 1e-3,
+        betas: tuple = (0.9, 0.999),
+        eps: float = 1e-8,
+        weight_decay: float = 0.0,
+    ) -> None:
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                state["step"] += 1
+                params.append(p)
+                grads.append(p.grad)
+                exp_avgs.append(state["exp_avg"])
+                exp_avg_sqs.append(state["exp_avg_sq"])
+            if not params:
+                continue
+            beta1, beta2 = group["betas"]
+            # all params in a group share the same step count once created
+            step = self.state[params[0]]["step"]
+            bc1 = 1 - beta1**step
+            bc2 = 1 - beta2**step
+            if use_hip(params[0]):
+                require_ext().adam_step(
+                    params,
+                    grads,
+                    exp_avgs,
+                    exp_avg_sqs,
+                    group["lr"],
+                    beta1,
+                    beta2,
+                    group["eps"],
+                    group["weight_decay"],
+                    bc1,
+                    bc2,
+                )
+            else:
+                for p, g, m, v in zip(params, grads, exp_avgs, exp_avg_sqs):
+                    gf = g.float()
+                    if group["weight_decay"] != 0:
+                        gf = gf.add(p.float(), alpha=group["weight_decay"])
+                    m.mul_(beta1).add_(gf, alpha=1 - beta1)
+                    v.mul_(beta2).addcmul_(gf, gf, value=1 - beta2)
+                    denom = (v / bc2).sqrt_().add_(group["eps"])
+                    upd = (m / bc1) / denom
+                    p.add_((-group["lr"] * upd).to(p.dtype))
+        return loss
+
+
+class RMSpropTF(torch.optim.Optimizer):
+    """TF-style RMSprop: v <- rho v + (1-rho) g^2; update = g / sqrt(v + eps)
+    (eps INSIDE the sqrt), square_avg initialized to ONES, momentum optional.
+    """
+
+    def __init__(
+        self,
+        params: Iterable,
+        lr: float = 1e-2,
+        alpha: float = 0.9,
+        eps: float = 1e-10,
+        weight_decay: float = 0.0,
+        momentum: float = 0.0,
+        centered: bool = False,
+    ) -> None:
+        defaults = dict(lr=lr, alpha=alpha, eps=eps, weight_decay=weight_decay, momentum=momentum, centered=centered)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                g = p.grad.float()
+                state = self.state[p]
+                if len(state) == 0:
+                    state["square_avg"] = torch.ones_like(p, dtype=torch.float32)
+                    if group["momentum"] > 0:
+                        state["momentum_buffer"] = torch.zeros_like(p, dtype=torch.float32)
+                    if group["centered"]:
+                        state["grad_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                if group["weight_decay"] != 0:
+                    g = g.add(p.float(), alpha=group["weight_decay"])
+                sq = state["square_avg"]
+                one_minus_alpha = 1.0 - group["alpha"]
+                sq.add_(g.pow(2) - sq, alpha=one_minus_alpha)
+                if group["centered"]:
+                    ga = state["grad_avg"]
+                    ga.add_(g - ga, alpha=one_minus_alpha)
+                    avg = sq.addcmul(ga, ga, value=-1).add_(group["eps"]).sqrt_()
+                else:
+                    avg = sq.add(group["eps"]).sqrt_()
+                if group["momentum"] > 0:
+                    buf = state["momentum_buffer"]
+                    buf.mul_(group["momentum"]).addcdiv_(g, avg)
+                    p.add_((-group["lr"] * buf).to(p.dtype))
+                else:
+                    p.add_((-group["lr"] * g / avg).to(p.dtype))
+        return loss

@@ -1,0 +1,33 @@
+import os
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO_ROOT = Path(__file__).resolve().parent.parent
+if str(REPO_ROOT) not in sys.path:
+    sys.path.insert(0, str(REPO_ROOT))
+
+os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+
+
+def pytest_configure(config):
+    config.addinivalue_line("markers", "gpu: tests that need a ROCm GPU (run on MI355X via gpurun)")
+    config.addinivalue_line("markers", "slow: long-running tests")
+
+
+@pytest.fixture(autouse=True)
+def _destroy_process_group():
+    """Parity with the reference's process-group teardown fixture
+    (tests/conftest.py:63-68)."""
+    yield
+    import torch.distributed as dist
+
+    if dist.is_available() and dist.is_initialized():
+        dist.destroy_process_group()
+
+
+@pytest.fixture()
+def tmp_run_dir(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    return tmp_path

@@ -1,0 +1,117 @@
+"""Algorithm smoke tests through the real CLI with dummy envs and tiny
+configs (the reference's test strategy, SURVEY.md §4: dry_run + num_envs=2 +
+cpu + devices in {1,2})."""
+
+import os
+import sys
+from unittest import mock
+
+import pytest
+
+from sheeprl_amd.cli import run
+
+
+@pytest.fixture(params=["1", "2"])
+def devices(request):
+    return request.param
+
+
+def standard_args(tmp_path, extra):
+    return [
+        "env=dummy",
+        "runtime.accelerator=cpu",
+        "dry_run=True",
+        "checkpoint.every=0",
+        "checkpoint.save_last=True",
+        "metric.log_every=1",
+        "metric.log_level=1",
+        "env.num_envs=2",
+        "seed=0",
+        *extra,
+    ]
+
+
+def _run(tmp_path, args, devices="1"):
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        run(args + [f"runtime.devices={devices}"])
+    finally:
+        os.chdir(cwd)
+
+
+@pytest.mark.timeout(180)
+def test_ppo(tmp_path, devices):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo",
+                "algo.total_steps=64",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=8",
+                "algo.run_test=True",
+            ],
+        ),
+        devices,
+    )
+
+
+@pytest.mark.timeout(180)
+def test_ppo_pixel(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo",
+                "algo.total_steps=32",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=8",
+                "algo.cnn_keys.encoder=[rgb]",
+                "algo.mlp_keys.encoder=[state]",
+                "algo.run_test=False",
+            ],
+        ),
+    )
+
+
+@pytest.mark.timeout(180)
+def test_ppo_continuous(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo",
+                "env.id=dummy_continuous",
+                "algo.total_steps=32",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=8",
+                "algo.run_test=False",
+            ],
+        ),
+    )
+
+
+@pytest.mark.timeout(180)
+def test_ppo_multidiscrete(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo",
+                "env.id=dummy_multidiscrete",
+                "algo.total_steps=32",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=8",
+                "algo.run_test=False",
+            ],
+        ),
+    )

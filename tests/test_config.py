@@ -1,0 +1,39 @@
+import pytest
+
+from sheeprl_amd.config import compose
+
+
+def test_compose_basic():
+    cfg = compose(["exp=ppo"])
+    assert cfg.algo.name == "ppo"
+    assert cfg.env.id == "cartpole"
+    assert cfg.runtime.devices == 1
+
+
+def test_compose_dotted_override():
+    cfg = compose(["exp=ppo", "algo.rollout_steps=7", "env.num_envs=3"])
+    assert cfg.algo.rollout_steps == 7
+    assert cfg.env.num_envs == 3
+
+
+def test_compose_group_override():
+    cfg = compose(["exp=ppo", "env=dummy"])
+    assert cfg.env.id == "dummy_discrete"
+
+
+def test_interpolation():
+    cfg = compose(["exp=ppo"])
+    assert cfg.exp_name == "ppo_cartpole"
+    assert "ppo" in cfg.root_dir
+
+
+def test_missing_exp_raises():
+    with pytest.raises(Exception):
+        compose([])
+
+
+def test_exp_inherits_group_chain():
+    cfg = compose(["exp=ppo_benchmarks"])
+    assert cfg.algo.total_steps == 65536
+    assert cfg.env.num_envs == 1
+    assert cfg.runtime.accelerator == "cpu"

@@ -1,0 +1,187 @@
+"""CPU-path numerics tests for the ops layer (the GPU kernels are compared
+against these same references in tests/test_gpu_kernels.py)."""
+
+import numpy as np
+import pytest
+import torch
+
+from sheeprl_amd import ops
+
+
+def test_symlog_symexp_roundtrip():
+    x = torch.randn(64) * 10
+    assert torch.allclose(ops.symexp(ops.symlog(x)), x, atol=1e-5)
+
+
+def test_symlog_grad():
+    x = torch.randn(32, dtype=torch.float64, requires_grad=True)
+    assert torch.autograd.gradcheck(lambda t: ops.symlog(t).sum(), (x,), eps=1e-6)
+
+
+def test_symexp_grad():
+    x = (torch.rand(32, dtype=torch.float64) * 2 - 1).requires_grad_()
+    assert torch.autograd.gradcheck(lambda t: ops.symexp(t).sum(), (x,), eps=1e-6)
+
+
+def test_twohot_roundtrip():
+    x = torch.tensor([[0.0], [1.5], [-3.2], [100.0]])
+    enc = ops.two_hot_encoder(x, support_range=20, num_buckets=41)
+    assert enc.shape == (4, 41)
+    assert torch.allclose(enc.sum(-1), torch.ones(4), atol=1e-5)
+    dec = ops.two_hot_decoder(enc, support_range=20)
+    assert torch.allclose(dec, x, atol=1e-3, rtol=1e-3)
+
+
+def test_twohot_weights_two_nonzero():
+    x = torch.tensor([[0.7]])
+    enc = ops.twohot_from_support(x, torch.linspace(-5, 5, 11))
+    nz = (enc != 0).sum()
+    assert nz <= 2
+    assert torch.allclose((enc * torch.linspace(-5, 5, 11)).sum(), torch.tensor(0.7), atol=1e-5)
+
+
+def test_gae_matches_naive():
+    T, N = 16, 4
+    torch.manual_seed(0)
+    rewards = torch.randn(T, N, 1)
+    values = torch.randn(T, N, 1)
+    dones = (torch.rand(T, N, 1) < 0.1)
+    next_value = torch.randn(N, 1)
+    gamma, lam = 0.99, 0.95
+    ret, adv = ops.gae(rewards, values, dones, next_value, T, gamma, lam)
+    # naive recomputation
+    not_dones = (~dones).float()
+    lastgaelam = torch.zeros(N, 1)
+    expected = torch.zeros(T, N, 1)
+    nnt, nv = not_dones[-1], next_value
+    for t in reversed(range(T)):
+        if t < T - 1:
+            nnt, nv = not_dones[t], values[t + 1]
+        delta = rewards[t] + gamma * nv * nnt - values[t]
+        lastgaelam = delta + gamma * lam * nnt * lastgaelam
+        expected[t] = lastgaelam
+    assert torch.allclose(adv, expected, atol=1e-5)
+    assert torch.allclose(ret, expected + values, atol=1e-5)
+
+
+def test_lambda_values_forward_and_grad():
+    T, B = 8, 6
+    torch.manual_seed(1)
+    rewards = torch.randn(T, B, dtype=torch.float64, requires_grad=True)
+    next_values = torch.randn(T, B, dtype=torch.float64, requires_grad=True)
+    continues = torch.rand(T, B, dtype=torch.float64)
+    lmbda = 0.95
+
+    def reference(r, nv):
+        out = []
+        nxt = nv[-1]
+        for t in reversed(range(T)):
+            nxt = r[t] + continues[t] * ((1 - lmbda) * nv[t] + lmbda * nxt)
+            out.append(nxt)
+        return torch.stack(out[::-1])
+
+    got = ops.lambda_values(rewards, next_values, continues, lmbda)
+    want = reference(rewards, next_values)
+    assert torch.allclose(got, want, atol=1e-9)
+
+    g = torch.randn_like(got)
+    got.backward(g)
+    gr1, gn1 = rewards.grad.clone(), next_values.grad.clone()
+    rewards.grad = None
+    next_values.grad = None
+    reference(rewards, next_values).backward(g)
+    assert torch.allclose(gr1, rewards.grad, atol=1e-9)
+    assert torch.allclose(gn1, next_values.grad, atol=1e-9)
+
+
+def test_layer_norm_act_matches_torch():
+    torch.manual_seed(2)
+    x = torch.randn(5, 7, 16, requires_grad=True)
+    w = torch.randn(16, requires_grad=True)
+    b = torch.randn(16, requires_grad=True)
+    y = ops.layer_norm_act(x, w, b, eps=1e-3, act="silu")
+    ref = torch.nn.functional.silu(torch.nn.functional.layer_norm(x, (16,), w, b, eps=1e-3))
+    assert torch.allclose(y, ref, atol=1e-5)
+    g = torch.randn_like(y)
+    y.backward(g)
+    gx1, gw1, gb1 = x.grad.clone(), w.grad.clone(), b.grad.clone()
+    x.grad = w.grad = b.grad = None
+    ref2 = torch.nn.functional.silu(torch.nn.functional.layer_norm(x, (16,), w, b, eps=1e-3))
+    ref2.backward(g)
+    assert torch.allclose(gx1, x.grad, atol=1e-4)
+    assert torch.allclose(gw1, w.grad, atol=1e-4)
+    assert torch.allclose(gb1, b.grad, atol=1e-4)
+
+
+def test_gru_gates_matches_composed():
+    torch.manual_seed(3)
+    B, H = 4, 8
+    y = torch.randn(B, 3 * H, requires_grad=True)
+    h = torch.randn(B, H, requires_grad=True)
+    w = torch.randn(3 * H, requires_grad=True)
+    b = torch.randn(3 * H, requires_grad=True)
+
+    def reference(y, h, w, b):
+        z = torch.nn.functional.layer_norm(y, (3 * H,), w, b, eps=1e-3)
+        r, c, u = z.chunk(3, -1)
+        r = torch.sigmoid(r)
+        c = torch.tanh(r * c)
+        u = torch.sigmoid(u - 1)
+        return u * c + (1 - u) * h
+
+    got = ops.gru_gates(y, h, w, b, eps=1e-3)
+    want = reference(y, h, w, b)
+    assert torch.allclose(got, want, atol=1e-5)
+    g = torch.randn_like(got)
+    got.backward(g)
+    grads1 = [t.grad.clone() for t in (y, h, w, b)]
+    for t in (y, h, w, b):
+        t.grad = None
+    reference(y, h, w, b).backward(g)
+    for g1, t in zip(grads1, (y, h, w, b)):
+        assert torch.allclose(g1, t.grad, atol=1e-4), t.shape
+
+
+def test_ema_update():
+    tgt = [torch.ones(4), torch.zeros(3)]
+    src = [torch.zeros(4), torch.ones(3)]
+    ops.ema_update_(tgt, src, tau=0.25)
+    assert torch.allclose(tgt[0], torch.full((4,), 0.75))
+    assert torch.allclose(tgt[1], torch.full((3,), 0.25))
+
+
+def test_normalize_obs():
+    x = torch.randint(0, 256, (2, 3, 8, 8), dtype=torch.uint8)
+    y = ops.normalize_obs(x)
+    assert y.dtype == torch.float32
+    assert y.min() >= -0.5 and y.max() <= 0.5
+
+
+def test_fused_adam_matches_torch_adam():
+    torch.manual_seed(4)
+    from sheeprl_amd.optim import FusedAdam
+
+    p1 = torch.nn.Parameter(torch.randn(10))
+    p2 = torch.nn.Parameter(p1.detach().clone())
+    o1 = FusedAdam([p1], lr=1e-2)
+    o2 = torch.optim.Adam([p2], lr=1e-2)
+    for _ in range(5):
+        g = torch.randn(10)
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        o1.step()
+        o2.step()
+    assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_ratio_scheduler():
+    from sheeprl_amd.utils.utils import Ratio
+
+    r = Ratio(ratio=0.5, pretrain_steps=3)
+    assert r(0) == 3
+    assert r(8) == 4
+    assert r(9) == 0
+    assert r(10) == 1
+    st = r.state_dict()
+    r2 = Ratio(0.5).load_state_dict(st)
+    assert r2(12) == 1
