@@ -1,0 +1,287 @@
+"""Benchmark entrypoint (driver contract).
+
+Measures the flagship training configuration — Dreamer-V3 Atari-100K
+(S model: dense 512, GRU 512, CNN mult 32, batch 16 x seq 64, replay_ratio 1)
+per BASELINE.json — on synthetic Atari-shaped data (no ALE in the image) with
+random-init weights.  One "step" is one training iteration of the real
+algorithm loop: one vectorized env interaction (policy forward + env step +
+buffer add) plus the replay-ratio-driven gradient step (world model + actor +
+critic update).  The reported metric is whole-job env-frames/sec
+(policy steps x action_repeat x world_size / wall), the reference's headline
+accounting for "Atari MsPacman 100K frames in 14h on 1x RTX 3080"
+ = 400k env frames / 50400 s = 7.94 env-frames/s.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+For N>1 the driver launches this under torch.distributed.run (one rank/GPU,
+RCCL over xGMI); each rank runs an identical replica (weak scaling, DP).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+import numpy as np
+import torch
+
+BASELINE_ENV_FRAMES_PER_SEC = 400_000 / (14 * 3600)  # RTX 3080, BASELINE.md row 2
+
+
+def _build(cfg_overrides, device_type: str):
+    from sheeprl_amd.config import compose
+
+    overrides = [
+        "exp=dreamer_v3_100k_ms_pacman",
+        f"runtime.accelerator={device_type}",
+        "runtime.precision=" + ("bf16" if device_type == "cuda" else "fp32"),
+        "metric.log_level=0",
+        "metric.disable_timer=True",
+        "checkpoint.every=0",
+        "checkpoint.save_last=False",
+        "buffer.size=4096",
+        "algo.run_test=False",
+        "env.sync_env=True",
+    ] + list(cfg_overrides)
+    return compose(overrides)
+
+
+def _setup(cfg, rank: int, world_size: int):
+    """Build env, agent, buffer, optimizers — the same components the real
+    training loop uses (sheeprl_amd/algos/dreamer_v3/dreamer_v3.py)."""
+    import torch.nn.functional as F
+
+    from sheeprl_amd.algos.dreamer_v3.agent import build_agent
+    from sheeprl_amd.algos.dreamer_v3.utils import Moments, prepare_obs
+    from sheeprl_amd.data import EnvIndependentReplayBuffer, SequentialReplayBuffer
+    from sheeprl_amd.envs import spaces, vectorize_env
+    from sheeprl_amd.optim import FusedAdam
+    from sheeprl_amd.parallel import Runtime
+    from sheeprl_amd.utils.utils import Ratio, seed_everything
+
+    seed_everything(cfg.seed + rank)
+    runtime = Runtime(
+        devices=world_size,
+        accelerator=cfg.runtime.accelerator,
+        precision=cfg.runtime.precision,
+    )
+    runtime.world_size = world_size
+    runtime.global_rank = rank
+    runtime.local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world_size > 1:
+        runtime._init_process_group(init_method="env://")
+    else:
+        runtime._setup_device()
+
+    envs = vectorize_env(cfg, cfg.seed, rank)
+    obs_space = envs.single_observation_space
+    action_space = envs.single_action_space
+    actions_dim = [action_space.n]
+
+    world_model, actor, critic, target_critic, player = build_agent(
+        runtime, actions_dim, False, cfg, obs_space
+    )
+    world_optimizer = FusedAdam(world_model.parameters(), lr=cfg.algo.world_model.optimizer.lr,
+                                eps=cfg.algo.world_model.optimizer.eps)
+    actor_optimizer = FusedAdam(actor.parameters(), lr=cfg.algo.actor.optimizer.lr, eps=cfg.algo.actor.optimizer.eps)
+    critic_optimizer = FusedAdam(critic.parameters(), lr=cfg.algo.critic.optimizer.lr,
+                                 eps=cfg.algo.critic.optimizer.eps)
+    moments = Moments(
+        cfg.algo.actor.moments.decay,
+        cfg.algo.actor.moments.max,
+        cfg.algo.actor.moments.percentile.low,
+        cfg.algo.actor.moments.percentile.high,
+    ).to(runtime.device)
+
+    rb = EnvIndependentReplayBuffer(
+        int(cfg.buffer.size),
+        n_envs=cfg.env.num_envs,
+        obs_keys=list(cfg.algo.cnn_keys.encoder),
+        buffer_cls=SequentialReplayBuffer,
+    )
+    return runtime, envs, (world_model, actor, critic, target_critic, player), (
+        world_optimizer, actor_optimizer, critic_optimizer), moments, rb
+
+
+def _prefill(cfg, envs, rb, n_steps: int):
+    """Fill the replay buffer with random-policy synthetic data (untimed)."""
+    num_envs = cfg.env.num_envs
+    n_act = envs.single_action_space.n
+    obs, _ = envs.reset(seed=cfg.seed)
+    step_data = {}
+    cnn_keys = list(cfg.algo.cnn_keys.encoder)
+    for k in cnn_keys:
+        step_data[k] = np.asarray(obs[k])[None]
+    step_data["rewards"] = np.zeros((1, num_envs, 1), np.float32)
+    step_data["terminated"] = np.zeros((1, num_envs, 1), np.float32)
+    step_data["truncated"] = np.zeros((1, num_envs, 1), np.float32)
+    step_data["is_first"] = np.ones_like(step_data["terminated"])
+    rng = np.random.default_rng(0)
+    for _ in range(n_steps):
+        a = rng.integers(0, n_act, size=num_envs)
+        onehot = np.eye(n_act, dtype=np.float32)[a]
+        step_data["actions"] = onehot[None]
+        rb.add(step_data)
+        obs, rewards, term, trunc, _ = envs.step(a)
+        for k in cnn_keys:
+            step_data[k] = np.asarray(obs[k])[None]
+        step_data["rewards"] = np.asarray(rewards, np.float32).reshape(1, num_envs, 1)
+        step_data["terminated"] = np.asarray(term, np.float32).reshape(1, num_envs, 1)
+        step_data["truncated"] = np.asarray(trunc, np.float32).reshape(1, num_envs, 1)
+        step_data["is_first"] = np.zeros_like(step_data["terminated"])
+    return obs, step_data
+
+
+def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
+    import torch.distributed as dist
+
+    from sheeprl_amd.algos.dreamer_v3.dreamer_v3 import train
+    from sheeprl_amd.algos.dreamer_v3.utils import prepare_obs
+    from sheeprl_amd.utils.metric import MetricAggregator
+
+    rank = int(os.environ.get("RANK", 0))
+    world_size = int(os.environ.get("WORLD_SIZE", n_gpus))
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    cfg = _build(overrides, device_type)
+    if device_type == "cuda":
+        from sheeprl_amd.ops import has_ext
+
+        if not has_ext():
+            raise RuntimeError("HIP extension _sheep_hip is not built — refusing to bench the eager fallback")
+
+    runtime, envs, models, optims, moments, rb = _setup(cfg, rank, world_size)
+    world_model, actor, critic, target_critic, player = models
+    world_optimizer, actor_optimizer, critic_optimizer = optims
+    device = runtime.device
+    num_envs = cfg.env.num_envs
+    actions_dim = [envs.single_action_space.n]
+    n_act = actions_dim[0]
+    seq_len = cfg.algo.per_rank_sequence_length
+
+    obs, step_data = _prefill(cfg, envs, rb, n_steps=max(seq_len + 1, 80))
+    player.init_states()
+
+    aggregator = MetricAggregator({})
+    MetricAggregator.disabled = True
+
+    def one_iter(obs, step_data):
+        # --- env interaction (the real policy forward + env step + buffer add)
+        with torch.inference_mode():
+            torch_obs = prepare_obs(runtime, obs, cnn_keys=list(cfg.algo.cnn_keys.encoder), num_envs=num_envs)
+            acts = player.get_actions(torch_obs)
+            actions = torch.cat(acts, -1).view(num_envs, -1).float().cpu().numpy()
+            real_actions = torch.stack([a.argmax(dim=-1) for a in acts], dim=-1).view(num_envs).cpu().numpy()
+        step_data["actions"] = actions[None]
+        rb.add(step_data)
+        obs, rewards, term, trunc, infos = envs.step(real_actions)
+        for k in cfg.algo.cnn_keys.encoder:
+            step_data[k] = np.asarray(obs[k])[None]
+        step_data["rewards"] = np.asarray(rewards, np.float32).reshape(1, num_envs, 1)
+        step_data["terminated"] = np.asarray(term, np.float32).reshape(1, num_envs, 1)
+        step_data["truncated"] = np.asarray(trunc, np.float32).reshape(1, num_envs, 1)
+        step_data["is_first"] = np.zeros_like(step_data["terminated"])
+        dones = np.logical_or(term, trunc)
+        if dones.any():
+            player.init_states(np.nonzero(dones)[0].tolist())
+        # --- one gradient step (replay_ratio=1 at num_envs=1)
+        sample = rb.sample_tensors(
+            cfg.algo.per_rank_batch_size,
+            sequence_length=seq_len,
+            n_samples=1,
+            device=device,
+        )
+        batch = {k: v[0] for k, v in sample.items()}
+        train(
+            runtime, world_model, actor, critic, target_critic,
+            world_optimizer, actor_optimizer, critic_optimizer,
+            batch, aggregator, cfg, False, actions_dim, moments,
+        )
+        return obs
+
+    # warmup
+    for _ in range(warmup):
+        obs = one_iter(obs, step_data)
+
+    if runtime.is_distributed:
+        dist.barrier()
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        obs = one_iter(obs, step_data)
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    if runtime.is_distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    if runtime.is_distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if runtime.backend == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    policy_steps = steps * num_envs * world_size
+    env_frames_per_sec = policy_steps * cfg.env.action_repeat / elapsed
+    result = {
+        "metric": "env_frames_per_sec",
+        "value": round(env_frames_per_sec, 3),
+        "unit": "frames/s",
+        "n_gpus": world_size,
+        "steps": steps,
+        "warmup": warmup,
+        "ms_per_step": round(elapsed / steps * 1000, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": round(env_frames_per_sec / (BASELINE_ENV_FRAMES_PER_SEC * 1), 3),
+        "dtype": "bf16" if device_type == "cuda" else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "dreamer_v3_S",
+            "global_batch": cfg.algo.per_rank_batch_size * world_size,
+            "seq_len": seq_len,
+            "parallelism": f"dp{world_size}",
+            "benchmark": "DreamerV3 Atari-100K (MsPacman shape, synthetic env)",
+            "grad_steps_per_policy_step": cfg.algo.replay_ratio,
+            "action_repeat": cfg.env.action_repeat,
+        },
+    }
+    envs.close()
+    return result
+
+
+def smoke_step() -> None:
+    """One tiny forward+backward of the flagship model on cuda:0 (driver
+    contract: __graft_entry__.smoke)."""
+    res = run_bench(
+        1,
+        steps=2,
+        warmup=0,
+        overrides=[
+            "algo=dreamer_v3_XS",
+            "algo.per_rank_batch_size=4",
+            "algo.per_rank_sequence_length=8",
+            "buffer.size=256",
+        ],
+    )
+    print("smoke:", json.dumps(res))
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--override", action="append", default=[])
+    args = p.parse_args()
+    rank = int(os.environ.get("RANK", 0))
+    result = run_bench(args.gpus, args.steps, args.warmup, args.override)
+    if rank == 0:
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

@@ -2,3 +2,4 @@
 (parity: sheeprl/__init__.py:18-47)."""
 
 from sheeprl_amd.algos import ppo  # noqa: F401
+from sheeprl_amd.algos import dreamer_v3  # noqa: F401

@@ -1,0 +1,704 @@
+"""Dreamer-V3 agent: encoder/decoder, RSSM, actor, critic, player.
+
+Parity surface: sheeprl/algos/dreamer_v3/agent.py — CNNEncoder :42,
+MLPEncoder :102, CNNDecoder :154, MLPDecoder :230, RecurrentModel :281,
+RSSM :344 (dynamic :398, _representation :458, _transition :474,
+imagination :487), PlayerDV3 :596, Actor :694, build_agent :935 (Hafner init
+application :1168-1180, player weight tying :1229-1235).
+
+MI355X design: every Linear+LN+SiLU triple is the fused DenseBlock; the GRU
+cell's post-GEMM math is the fused ``ops.gru_gates`` kernel; the stochastic
+state sampling stays in fp32 for categorical stability under bf16-true.
+"""
+
+from __future__ import annotations
+
+import copy
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+import torch.distributions as td
+import torch.nn.functional as F
+from torch import Tensor, nn
+
+from sheeprl_amd import ops
+from sheeprl_amd.distributions import OneHotCategoricalST, unimix_logits
+from sheeprl_amd.envs import spaces
+from sheeprl_amd.models import (
+    CNN,
+    DeCNN,
+    DenseBlock,
+    LayerNormGRUCell,
+    MLP,
+    MultiDecoder,
+    MultiEncoder,
+    cnn_forward,
+)
+from sheeprl_amd.parallel import Runtime
+
+
+# ---------------------------------------------------------------------------
+# init (Hafner) — parity: dreamer_v3/utils.py:143-187
+# ---------------------------------------------------------------------------
+
+def init_weights(m: nn.Module) -> None:
+    if isinstance(m, nn.Linear):
+        denoms = (m.in_features + m.out_features) / 2.0
+        std = np.sqrt(1.0 / denoms) / 0.87962566103423978
+        nn.init.trunc_normal_(m.weight.data, mean=0.0, std=std, a=-2.0 * std, b=2.0 * std)
+        if m.bias is not None:
+            m.bias.data.fill_(0.0)
+    elif isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
+        space = m.kernel_size[0] * m.kernel_size[1]
+        in_num = space * m.in_channels
+        out_num = space * m.out_channels
+        denoms = (in_num + out_num) / 2.0
+        std = np.sqrt(1.0 / denoms) / 0.87962566103423978
+        nn.init.trunc_normal_(m.weight.data, mean=0.0, std=std, a=-2.0 * std, b=2.0 * std)
+        if m.bias is not None:
+            m.bias.data.fill_(0.0)
+
+
+def uniform_init_weights(given_scale: float):
+    def f(m: nn.Module) -> None:
+        if isinstance(m, nn.Linear):
+            denoms = (m.in_features + m.out_features) / 2.0
+            limit = np.sqrt(3 * given_scale / denoms)
+            nn.init.uniform_(m.weight.data, a=-limit, b=limit)
+            if m.bias is not None:
+                m.bias.data.fill_(0.0)
+        elif isinstance(m, (nn.Conv2d, nn.ConvTranspose2d)):
+            space = m.kernel_size[0] * m.kernel_size[1]
+            denoms = (space * m.in_channels + space * m.out_channels) / 2.0
+            limit = np.sqrt(3 * given_scale / denoms)
+            nn.init.uniform_(m.weight.data, a=-limit, b=limit)
+            if m.bias is not None:
+                m.bias.data.fill_(0.0)
+
+    return f
+
+
+def compute_stochastic_state(logits: Tensor, discrete: int = 32, sample: bool = True) -> Tensor:
+    """One-hot straight-through sample of the [*, stoch, discrete] categorical
+    (parity: dreamer_v2/utils.py:44)."""
+    logits = logits.view(*logits.shape[:-1], -1, discrete)
+    dist = td.Independent(OneHotCategoricalST(logits=logits.float()), 1)
+    out = dist.rsample() if sample else dist.mode
+    return out.to(logits.dtype) if out.dtype != logits.dtype else out
+
+
+# ---------------------------------------------------------------------------
+# encoders / decoders
+# ---------------------------------------------------------------------------
+
+class CNNEncoder(nn.Module):
+    def __init__(
+        self,
+        keys: Sequence[str],
+        input_channels: Sequence[int],
+        image_size: Tuple[int, int],
+        channels_multiplier: int,
+        layer_norm_eps: float = 1e-3,
+        stages: int = 4,
+    ) -> None:
+        super().__init__()
+        self.keys = list(keys)
+        self.input_dim = (sum(input_channels), *image_size)
+        self.model = nn.Sequential(
+            CNN(
+                in_channels=self.input_dim[0],
+                hidden_channels=[(2**i) * channels_multiplier for i in range(stages)],
+                kernel_sizes=[4] * stages,
+                strides=[2] * stages,
+                paddings=[1] * stages,
+                activation="silu",
+                layer_norm=True,
+                layer_norm_eps=layer_norm_eps,
+            ),
+            nn.Flatten(-3, -1),
+        )
+        with torch.no_grad():
+            self.output_dim = self.model(torch.zeros(1, *self.input_dim)).shape[-1]
+
+    def forward(self, obs: Dict[str, Tensor]) -> Tensor:
+        x = torch.cat([obs[k] for k in self.keys], dim=-3)
+        return cnn_forward(self.model, x, x.shape[-3:], flatten=False)
+
+
+class MLPEncoder(nn.Module):
+    def __init__(
+        self,
+        keys: Sequence[str],
+        input_dims: Sequence[int],
+        mlp_layers: int = 4,
+        dense_units: int = 512,
+        layer_norm_eps: float = 1e-3,
+        symlog_inputs: bool = True,
+    ) -> None:
+        super().__init__()
+        self.keys = list(keys)
+        self.input_dim = sum(input_dims)
+        self.model = MLP(
+            self.input_dim,
+            None,
+            [dense_units] * mlp_layers,
+            activation="silu",
+            layer_norm=True,
+            layer_norm_eps=layer_norm_eps,
+        )
+        self.output_dim = dense_units
+        self.symlog_inputs = symlog_inputs
+
+    def forward(self, obs: Dict[str, Tensor]) -> Tensor:
+        x = torch.cat([ops.symlog(obs[k]) if self.symlog_inputs else obs[k] for k in self.keys], -1)
+        return self.model(x)
+
+
+class CNNDecoder(nn.Module):
+    def __init__(
+        self,
+        keys: Sequence[str],
+        output_channels: Sequence[int],
+        channels_multiplier: int,
+        latent_state_size: int,
+        cnn_encoder_output_dim: int,
+        image_size: Tuple[int, int],
+        layer_norm_eps: float = 1e-3,
+        stages: int = 4,
+    ) -> None:
+        super().__init__()
+        self.keys = list(keys)
+        self.output_channels = list(output_channels)
+        self.output_dim = (sum(output_channels), *image_size)
+        self.model = nn.Sequential(
+            nn.Linear(latent_state_size, cnn_encoder_output_dim),
+            nn.Unflatten(1, (-1, 4, 4)),
+            DeCNN(
+                in_channels=(2 ** (stages - 1)) * channels_multiplier,
+                hidden_channels=[(2**i) * channels_multiplier for i in reversed(range(stages - 1))]
+                + [self.output_dim[0]],
+                kernel_sizes=[4] * stages,
+                strides=[2] * stages,
+                paddings=[1] * stages,
+                activation="silu",
+                layer_norm=True,
+                layer_norm_eps=layer_norm_eps,
+                last_layer_plain=True,
+            ),
+        )
+
+    def forward(self, latent_states: Tensor) -> Dict[str, Tensor]:
+        out = cnn_forward(self.model, latent_states, (latent_states.shape[-1],), flatten=False)
+        return {k: rec for k, rec in zip(self.keys, torch.split(out, self.output_channels, -3))}
+
+
+class MLPDecoder(nn.Module):
+    def __init__(
+        self,
+        keys: Sequence[str],
+        output_dims: Sequence[int],
+        latent_state_size: int,
+        mlp_layers: int = 4,
+        dense_units: int = 512,
+        layer_norm_eps: float = 1e-3,
+    ) -> None:
+        super().__init__()
+        self.keys = list(keys)
+        self.model = MLP(
+            latent_state_size,
+            None,
+            [dense_units] * mlp_layers,
+            activation="silu",
+            layer_norm=True,
+            layer_norm_eps=layer_norm_eps,
+        )
+        self.heads = nn.ModuleList([nn.Linear(dense_units, dim) for dim in output_dims])
+
+    def forward(self, latent_states: Tensor) -> Dict[str, Tensor]:
+        x = self.model(latent_states)
+        return {k: h(x) for k, h in zip(self.keys, self.heads)}
+
+
+# ---------------------------------------------------------------------------
+# RSSM
+# ---------------------------------------------------------------------------
+
+class RecurrentModel(nn.Module):
+    def __init__(
+        self,
+        input_size: int,
+        recurrent_state_size: int,
+        dense_units: int,
+        layer_norm_eps: float = 1e-3,
+    ) -> None:
+        super().__init__()
+        self.mlp = DenseBlock(
+            input_size, dense_units, bias=False, layer_norm=True, layer_norm_eps=layer_norm_eps, activation="silu"
+        )
+        self.rnn = LayerNormGRUCell(dense_units, recurrent_state_size, bias=False, layer_norm=True,
+                                    layer_norm_eps=layer_norm_eps)
+        self.recurrent_state_size = recurrent_state_size
+
+    def forward(self, input: Tensor, recurrent_state: Tensor) -> Tensor:
+        return self.rnn(self.mlp(input), recurrent_state)
+
+
+class RSSM(nn.Module):
+    def __init__(
+        self,
+        recurrent_model: RecurrentModel,
+        representation_model: MLP,
+        transition_model: MLP,
+        discrete: int = 32,
+        unimix: float = 0.01,
+        learnable_initial_recurrent_state: bool = True,
+    ) -> None:
+        super().__init__()
+        self.recurrent_model = recurrent_model
+        self.representation_model = representation_model
+        self.transition_model = transition_model
+        self.discrete = discrete
+        self.unimix = unimix
+        init = torch.zeros(recurrent_model.recurrent_state_size, dtype=torch.float32)
+        if learnable_initial_recurrent_state:
+            self.initial_recurrent_state = nn.Parameter(init)
+        else:
+            self.register_buffer("initial_recurrent_state", init)
+
+    def get_initial_states(self, batch_shape: Sequence[int]) -> Tuple[Tensor, Tensor]:
+        initial_recurrent_state = torch.tanh(self.initial_recurrent_state).expand(*batch_shape, -1)
+        initial_posterior = self._transition(initial_recurrent_state, sample_state=False)[1]
+        return initial_recurrent_state, initial_posterior
+
+    def dynamic(
+        self, posterior: Tensor, recurrent_state: Tensor, action: Tensor, embedded_obs: Tensor, is_first: Tensor
+    ) -> Tuple[Tensor, Tensor, Tensor, Tensor, Tensor]:
+        action = (1 - is_first) * action
+        initial_recurrent_state, initial_posterior = self.get_initial_states(recurrent_state.shape[:2])
+        recurrent_state = (1 - is_first) * recurrent_state + is_first * initial_recurrent_state
+        posterior = posterior.view(*posterior.shape[:-2], -1)
+        posterior = (1 - is_first) * posterior + is_first * initial_posterior.view_as(posterior)
+        recurrent_state = self.recurrent_model(torch.cat((posterior, action), -1), recurrent_state)
+        prior_logits, prior = self._transition(recurrent_state)
+        posterior_logits, posterior = self._representation(recurrent_state, embedded_obs)
+        return recurrent_state, posterior, prior, posterior_logits, prior_logits
+
+    def _uniform_mix(self, logits: Tensor) -> Tensor:
+        dim = logits.dim()
+        if dim == 3:
+            logits = logits.view(*logits.shape[:-1], -1, self.discrete)
+        elif dim != 4:
+            raise RuntimeError(f"expected 3D or 4D logits, got {dim}D")
+        logits = unimix_logits(logits.float(), self.unimix)
+        return logits.view(*logits.shape[:-2], -1)
+
+    def _representation(self, recurrent_state: Tensor, embedded_obs: Tensor) -> Tuple[Tensor, Tensor]:
+        logits = self.representation_model(torch.cat((recurrent_state, embedded_obs), -1))
+        logits = self._uniform_mix(logits)
+        return logits, compute_stochastic_state(logits, discrete=self.discrete)
+
+    def _transition(self, recurrent_out: Tensor, sample_state: bool = True) -> Tuple[Tensor, Tensor]:
+        logits = self.transition_model(recurrent_out)
+        logits = self._uniform_mix(logits)
+        return logits, compute_stochastic_state(logits, discrete=self.discrete, sample=sample_state)
+
+    def imagination(self, prior: Tensor, recurrent_state: Tensor, actions: Tensor) -> Tuple[Tensor, Tensor]:
+        recurrent_state = self.recurrent_model(torch.cat((prior, actions), -1), recurrent_state)
+        _, imagined_prior = self._transition(recurrent_state)
+        return imagined_prior, recurrent_state
+
+
+class DecoupledRSSM(RSSM):
+    """Posterior computed for all timesteps in parallel from embeddings only
+    (parity: dreamer_v3/agent.py:501-593) — the sequence-parallel seam: only
+    the cheap gate recurrence stays sequential."""
+
+    def dynamic(  # type: ignore[override]
+        self, posterior: Tensor, recurrent_state: Tensor, action: Tensor, is_first: Tensor
+    ) -> Tuple[Tensor, Tensor, Tensor]:
+        action = (1 - is_first) * action
+        initial_recurrent_state, initial_posterior = self.get_initial_states(recurrent_state.shape[:2])
+        recurrent_state = (1 - is_first) * recurrent_state + is_first * initial_recurrent_state
+        posterior = posterior.view(*posterior.shape[:-2], -1)
+        posterior = (1 - is_first) * posterior + is_first * initial_posterior.view_as(posterior)
+        recurrent_state = self.recurrent_model(torch.cat((posterior, action), -1), recurrent_state)
+        prior_logits, _ = self._transition(recurrent_state)
+        return recurrent_state, None, prior_logits
+
+    def _representation(self, embedded_obs: Tensor) -> Tuple[Tensor, Tensor]:  # type: ignore[override]
+        logits = self.representation_model(embedded_obs)
+        logits = self._uniform_mix(logits)
+        return logits, compute_stochastic_state(logits, discrete=self.discrete)
+
+
+class WorldModel(nn.Module):
+    def __init__(
+        self,
+        encoder: MultiEncoder,
+        rssm: RSSM,
+        observation_model: MultiDecoder,
+        reward_model: MLP,
+        continue_model: MLP,
+    ) -> None:
+        super().__init__()
+        self.encoder = encoder
+        self.rssm = rssm
+        self.observation_model = observation_model
+        self.reward_model = reward_model
+        self.continue_model = continue_model
+
+
+# ---------------------------------------------------------------------------
+# actor
+# ---------------------------------------------------------------------------
+
+class Actor(nn.Module):
+    def __init__(
+        self,
+        latent_state_size: int,
+        actions_dim: Sequence[int],
+        is_continuous: bool,
+        distribution: str = "auto",
+        init_std: float = 2.0,
+        min_std: float = 0.1,
+        max_std: float = 1.0,
+        dense_units: int = 1024,
+        mlp_layers: int = 5,
+        layer_norm_eps: float = 1e-3,
+        unimix: float = 0.01,
+        action_clip: float = 1.0,
+    ) -> None:
+        super().__init__()
+        self.distribution = distribution.lower()
+        if self.distribution not in ("auto", "normal", "tanh_normal", "discrete", "scaled_normal"):
+            raise ValueError(f"unknown actor distribution '{distribution}'")
+        if self.distribution == "auto":
+            self.distribution = "scaled_normal" if is_continuous else "discrete"
+        self.model = MLP(
+            latent_state_size,
+            None,
+            [dense_units] * mlp_layers,
+            activation="silu",
+            layer_norm=True,
+            layer_norm_eps=layer_norm_eps,
+        )
+        if is_continuous:
+            self.mlp_heads = nn.ModuleList([nn.Linear(dense_units, int(sum(actions_dim)) * 2)])
+        else:
+            self.mlp_heads = nn.ModuleList([nn.Linear(dense_units, d) for d in actions_dim])
+        self.actions_dim = list(actions_dim)
+        self.is_continuous = is_continuous
+        self.init_std = init_std
+        self.min_std = min_std
+        self.max_std = max_std
+        self._unimix = unimix
+        self._action_clip = action_clip
+
+    def forward(
+        self, state: Tensor, greedy: bool = False, mask: Optional[Dict[str, Tensor]] = None
+    ) -> Tuple[Tuple[Tensor, ...], Tuple[td.Distribution, ...]]:
+        out = self.model(state)
+        pre_dist = [head(out) for head in self.mlp_heads]
+        if self.is_continuous:
+            mean, std = torch.chunk(pre_dist[0].float(), 2, -1)
+            if self.distribution == "tanh_normal":
+                mean = 5 * torch.tanh(mean / 5)
+                std = F.softplus(std + self.init_std) + self.min_std
+                dist = td.Independent(
+                    td.TransformedDistribution(td.Normal(mean, std), td.TanhTransform()), 1
+                )
+            elif self.distribution == "normal":
+                dist = td.Independent(td.Normal(mean, std), 1)
+            else:  # scaled_normal
+                std = (self.max_std - self.min_std) * torch.sigmoid(std + self.init_std) + self.min_std
+                dist = td.Independent(td.Normal(torch.tanh(mean), std), 1)
+            if not greedy:
+                actions = dist.rsample()
+            else:
+                sample = dist.sample((100,))
+                log_prob = dist.log_prob(sample)
+                actions = sample[log_prob.argmax(0)].view(1, 1, -1)
+            if self._action_clip > 0.0:
+                clip = torch.full_like(actions, self._action_clip)
+                actions = actions * (clip / torch.maximum(clip, torch.abs(actions))).detach()
+            return (actions,), (dist,)
+        actions_list: List[Tensor] = []
+        dists: List[td.Distribution] = []
+        for logits in pre_dist:
+            dist = OneHotCategoricalST(logits=unimix_logits(logits.float(), self._unimix))
+            dists.append(dist)
+            actions_list.append(dist.rsample() if not greedy else dist.mode)
+        return tuple(actions_list), tuple(dists)
+
+
+# ---------------------------------------------------------------------------
+# player
+# ---------------------------------------------------------------------------
+
+class PlayerDV3(nn.Module):
+    def __init__(
+        self,
+        encoder: MultiEncoder,
+        rssm: RSSM,
+        actor: Actor,
+        actions_dim: Sequence[int],
+        num_envs: int,
+        stochastic_size: int,
+        recurrent_state_size: int,
+        device: torch.device,
+        discrete_size: int = 32,
+        actor_type: Optional[str] = None,
+    ) -> None:
+        super().__init__()
+        self.encoder = encoder
+        self.rssm = rssm
+        self.actor = actor
+        self.actions_dim = list(actions_dim)
+        self.num_envs = num_envs
+        self.stochastic_size = stochastic_size
+        self.recurrent_state_size = recurrent_state_size
+        self.device = device
+        self.discrete_size = discrete_size
+        self.actor_type = actor_type
+        self.decoupled_rssm = isinstance(rssm, DecoupledRSSM)
+
+    @torch.no_grad()
+    def init_states(self, reset_envs: Optional[Sequence[int]] = None) -> None:
+        dtype = next(self.rssm.parameters()).dtype
+        if reset_envs is None or len(reset_envs) == 0:
+            self.actions = torch.zeros(1, self.num_envs, int(np.sum(self.actions_dim)), device=self.device, dtype=dtype)
+            self.recurrent_state, stochastic_state = self.rssm.get_initial_states((1, self.num_envs))
+            self.recurrent_state = self.recurrent_state.contiguous().to(dtype)
+            self.stochastic_state = stochastic_state.reshape(1, self.num_envs, -1).to(dtype)
+        else:
+            self.actions[:, reset_envs] = 0.0
+            rec, stoch = self.rssm.get_initial_states((1, len(reset_envs)))
+            self.recurrent_state[:, reset_envs] = rec.to(self.recurrent_state.dtype)
+            self.stochastic_state[:, reset_envs] = stoch.reshape(1, len(reset_envs), -1).to(self.stochastic_state.dtype)
+
+    @torch.no_grad()
+    def get_actions(
+        self, obs: Dict[str, Tensor], greedy: bool = False, mask: Optional[Dict[str, Tensor]] = None
+    ) -> Sequence[Tensor]:
+        embedded_obs = self.encoder(obs)
+        self.recurrent_state = self.rssm.recurrent_model(
+            torch.cat((self.stochastic_state, self.actions), -1), self.recurrent_state
+        )
+        if self.decoupled_rssm:
+            _, stoch = self.rssm._representation(embedded_obs)
+        else:
+            _, stoch = self.rssm._representation(self.recurrent_state, embedded_obs)
+        self.stochastic_state = stoch.view(*stoch.shape[:-2], self.stochastic_size * self.discrete_size)
+        actions, _ = self.actor(torch.cat((self.stochastic_state, self.recurrent_state), -1), greedy, mask)
+        self.actions = torch.cat(actions, -1)
+        return actions
+
+
+# ---------------------------------------------------------------------------
+# build
+# ---------------------------------------------------------------------------
+
+def build_agent(
+    runtime: Runtime,
+    actions_dim: Sequence[int],
+    is_continuous: bool,
+    cfg: Any,
+    obs_space: spaces.Dict,
+    world_model_state: Optional[Dict[str, Tensor]] = None,
+    actor_state: Optional[Dict[str, Tensor]] = None,
+    critic_state: Optional[Dict[str, Tensor]] = None,
+    target_critic_state: Optional[Dict[str, Tensor]] = None,
+) -> Tuple[WorldModel, Actor, MLP, nn.Module, PlayerDV3]:
+    wm_cfg = cfg.algo.world_model
+    actor_cfg = cfg.algo.actor
+    critic_cfg = cfg.algo.critic
+    eps = 1e-3
+
+    recurrent_state_size = wm_cfg.recurrent_model.recurrent_state_size
+    stochastic_size = wm_cfg.stochastic_size * wm_cfg.discrete_size
+    latent_state_size = stochastic_size + recurrent_state_size
+    cnn_keys = list(cfg.algo.cnn_keys.encoder or [])
+    mlp_keys = list(cfg.algo.mlp_keys.encoder or [])
+    stages = int(np.log2(cfg.env.screen_size) - np.log2(4))
+
+    cnn_encoder = (
+        CNNEncoder(
+            keys=cnn_keys,
+            input_channels=[int(np.prod(obs_space[k].shape[:-2])) for k in cnn_keys],
+            image_size=tuple(obs_space[cnn_keys[0]].shape[-2:]),
+            channels_multiplier=wm_cfg.encoder.cnn_channels_multiplier,
+            layer_norm_eps=eps,
+            stages=stages,
+        )
+        if cnn_keys
+        else None
+    )
+    mlp_encoder = (
+        MLPEncoder(
+            keys=mlp_keys,
+            input_dims=[int(obs_space[k].shape[0]) for k in mlp_keys],
+            mlp_layers=wm_cfg.encoder.mlp_layers,
+            dense_units=wm_cfg.encoder.dense_units,
+            layer_norm_eps=eps,
+        )
+        if mlp_keys
+        else None
+    )
+    encoder = MultiEncoder(cnn_encoder, mlp_encoder)
+
+    recurrent_model = RecurrentModel(
+        input_size=int(sum(actions_dim) + stochastic_size),
+        recurrent_state_size=recurrent_state_size,
+        dense_units=wm_cfg.recurrent_model.dense_units,
+        layer_norm_eps=eps,
+    )
+    repr_in = encoder.output_dim + (0 if wm_cfg.decoupled_rssm else recurrent_state_size)
+    representation_model = MLP(
+        repr_in,
+        stochastic_size,
+        [wm_cfg.representation_model.hidden_size],
+        activation="silu",
+        layer_norm=True,
+        layer_norm_eps=eps,
+    )
+    transition_model = MLP(
+        recurrent_state_size,
+        stochastic_size,
+        [wm_cfg.transition_model.hidden_size],
+        activation="silu",
+        layer_norm=True,
+        layer_norm_eps=eps,
+    )
+    rssm_cls = DecoupledRSSM if wm_cfg.decoupled_rssm else RSSM
+    recurrent_model.apply(init_weights)
+    representation_model.apply(init_weights)
+    transition_model.apply(init_weights)
+    rssm = rssm_cls(
+        recurrent_model=recurrent_model,
+        representation_model=representation_model,
+        transition_model=transition_model,
+        discrete=wm_cfg.discrete_size,
+        unimix=cfg.algo.unimix,
+        learnable_initial_recurrent_state=wm_cfg.learnable_initial_recurrent_state,
+    )
+
+    cnn_decoder = (
+        CNNDecoder(
+            keys=list(cfg.algo.cnn_keys.decoder or cnn_keys),
+            output_channels=[int(np.prod(obs_space[k].shape[:-2])) for k in cnn_keys],
+            channels_multiplier=wm_cfg.observation_model.cnn_channels_multiplier,
+            latent_state_size=latent_state_size,
+            cnn_encoder_output_dim=cnn_encoder.output_dim,
+            image_size=tuple(obs_space[cnn_keys[0]].shape[-2:]),
+            layer_norm_eps=eps,
+            stages=stages,
+        )
+        if cnn_keys
+        else None
+    )
+    mlp_decoder = (
+        MLPDecoder(
+            keys=list(cfg.algo.mlp_keys.decoder or mlp_keys),
+            output_dims=[int(obs_space[k].shape[0]) for k in mlp_keys],
+            latent_state_size=latent_state_size,
+            mlp_layers=wm_cfg.observation_model.mlp_layers,
+            dense_units=wm_cfg.observation_model.dense_units,
+            layer_norm_eps=eps,
+        )
+        if mlp_keys
+        else None
+    )
+    observation_model = MultiDecoder(cnn_decoder, mlp_decoder)
+
+    reward_model = MLP(
+        latent_state_size,
+        wm_cfg.reward_model.bins,
+        [wm_cfg.reward_model.dense_units] * wm_cfg.reward_model.mlp_layers,
+        activation="silu",
+        layer_norm=True,
+        layer_norm_eps=eps,
+    )
+    continue_model = MLP(
+        latent_state_size,
+        1,
+        [wm_cfg.discount_model.dense_units] * wm_cfg.discount_model.mlp_layers,
+        activation="silu",
+        layer_norm=True,
+        layer_norm_eps=eps,
+    )
+    world_model = WorldModel(encoder, rssm, observation_model, reward_model, continue_model)
+
+    actor = Actor(
+        latent_state_size=latent_state_size,
+        actions_dim=actions_dim,
+        is_continuous=is_continuous,
+        distribution=cfg.distribution.get("type", "auto"),
+        init_std=actor_cfg.init_std,
+        min_std=actor_cfg.min_std,
+        max_std=actor_cfg.max_std,
+        dense_units=actor_cfg.dense_units,
+        mlp_layers=actor_cfg.mlp_layers,
+        layer_norm_eps=eps,
+        unimix=cfg.algo.unimix,
+        action_clip=actor_cfg.action_clip,
+    )
+    critic = MLP(
+        latent_state_size,
+        critic_cfg.bins,
+        [critic_cfg.dense_units] * critic_cfg.mlp_layers,
+        activation="silu",
+        layer_norm=True,
+        layer_norm_eps=eps,
+    )
+
+    encoder.apply(init_weights)
+    observation_model.apply(init_weights)
+    reward_model.apply(init_weights)
+    continue_model.apply(init_weights)
+    actor.apply(init_weights)
+    critic.apply(init_weights)
+
+    if cfg.algo.hafner_initialization:
+        actor.mlp_heads.apply(uniform_init_weights(1.0))
+        critic.model[-1].apply(uniform_init_weights(0.0))
+        rssm.transition_model.model[-1].apply(uniform_init_weights(1.0))
+        rssm.representation_model.model[-1].apply(uniform_init_weights(1.0))
+        world_model.reward_model.model[-1].apply(uniform_init_weights(0.0))
+        world_model.continue_model.model[-1].apply(uniform_init_weights(1.0))
+        if mlp_decoder is not None:
+            mlp_decoder.heads.apply(uniform_init_weights(1.0))
+        if cnn_decoder is not None:
+            cnn_decoder.model[-1].model[-1].apply(uniform_init_weights(1.0))
+
+    if world_model_state:
+        world_model.load_state_dict(world_model_state)
+    if actor_state:
+        actor.load_state_dict(actor_state)
+    if critic_state:
+        critic.load_state_dict(critic_state)
+
+    # one GradSync per training module: few, large buckets for the xGMI links
+    world_model = runtime.setup_module(world_model)
+    actor = runtime.setup_module(actor)
+    critic = runtime.setup_module(critic)
+
+    target_critic = copy.deepcopy(critic)
+    if target_critic_state:
+        target_critic.load_state_dict(target_critic_state)
+    target_critic = runtime.setup_module(target_critic, sync=False)
+    for p in target_critic.parameters():
+        p.requires_grad_(False)
+
+    player = PlayerDV3(
+        world_model.encoder,
+        world_model.rssm,
+        actor,
+        actions_dim,
+        cfg.env.num_envs,
+        wm_cfg.stochastic_size,
+        recurrent_state_size,
+        runtime.device,
+        discrete_size=wm_cfg.discrete_size,
+    )
+    return world_model, actor, critic, target_critic, player

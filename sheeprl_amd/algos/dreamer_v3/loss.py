@@ -1,0 +1,51 @@
+"""Dreamer-V3 world-model loss (parity: sheeprl/algos/dreamer_v3/loss.py:9 —
+two-sided KL balancing with free nats :64-75)."""
+
+from __future__ import annotations
+
+from typing import Dict, Optional, Tuple
+
+import torch
+from torch import Tensor
+
+
+def categorical_kl(p_logits: Tensor, q_logits: Tensor) -> Tensor:
+    """KL(P || Q) for [*, stoch, discrete] categorical logits, summed over the
+    stoch dimension; computed in fp32 from log-softmaxes (stable under
+    bf16-true, SURVEY.md §7 hard-part 2)."""
+    p_log = torch.log_softmax(p_logits.float(), dim=-1)
+    q_log = torch.log_softmax(q_logits.float(), dim=-1)
+    p = p_log.exp()
+    return (p * (p_log - q_log)).sum(dim=(-2, -1))
+
+
+def reconstruction_loss(
+    po: Dict[str, object],
+    observations: Dict[str, Tensor],
+    pr: object,
+    rewards: Tensor,
+    priors_logits: Tensor,
+    posteriors_logits: Tensor,
+    kl_dynamic: float = 0.5,
+    kl_representation: float = 0.1,
+    kl_free_nats: float = 1.0,
+    kl_regularizer: float = 1.0,
+    pc: Optional[object] = None,
+    continue_targets: Optional[Tensor] = None,
+    continue_scale_factor: float = 1.0,
+) -> Tuple[Tensor, Tensor, Tensor, Tensor, Tensor, Tensor]:
+    observation_loss = -sum(po[k].log_prob(observations[k]) for k in po.keys())
+    reward_loss = -pr.log_prob(rewards)
+    # KL balancing: dynamic = KL(sg(post) || prior), representation = KL(post || sg(prior))
+    kl = dyn_loss = categorical_kl(posteriors_logits.detach(), priors_logits)
+    free_nats = torch.full_like(dyn_loss, kl_free_nats)
+    dyn_loss = kl_dynamic * torch.maximum(dyn_loss, free_nats)
+    repr_loss = categorical_kl(posteriors_logits, priors_logits.detach())
+    repr_loss = kl_representation * torch.maximum(repr_loss, free_nats)
+    kl_loss = dyn_loss + repr_loss
+    if pc is not None and continue_targets is not None:
+        continue_loss = continue_scale_factor * -pc.log_prob(continue_targets)
+    else:
+        continue_loss = torch.zeros_like(reward_loss)
+    rec_loss = (kl_regularizer * kl_loss + observation_loss + reward_loss + continue_loss).mean()
+    return rec_loss, kl, kl_loss.mean(), reward_loss.mean(), observation_loss.mean(), continue_loss.mean()

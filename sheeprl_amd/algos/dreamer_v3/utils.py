@@ -1,0 +1,117 @@
+"""Dreamer-V3 helpers (parity: sheeprl/algos/dreamer_v3/utils.py —
+Moments :40, compute_lambda_values :66, prepare_obs :80, test :94)."""
+
+from __future__ import annotations
+
+from typing import Any, Dict, Optional, Sequence
+
+import numpy as np
+import torch
+from torch import Tensor, nn
+
+from sheeprl_amd import ops
+from sheeprl_amd.parallel import Runtime
+
+AGGREGATOR_KEYS = {
+    "Rewards/rew_avg",
+    "Game/ep_len_avg",
+    "Loss/world_model_loss",
+    "Loss/value_loss",
+    "Loss/policy_loss",
+    "Loss/observation_loss",
+    "Loss/reward_loss",
+    "Loss/state_loss",
+    "Loss/continue_loss",
+    "State/kl",
+    "State/post_entropy",
+    "State/prior_entropy",
+    "Grads/world_model",
+    "Grads/actor",
+    "Grads/critic",
+}
+MODELS_TO_REGISTER = {"world_model", "actor", "critic", "target_critic", "moments"}
+
+
+class Moments(nn.Module):
+    """Distributed percentile EMA used to normalize λ-returns
+    (parity: dreamer_v3/utils.py:40-63; the all-gather is the RCCL hop)."""
+
+    def __init__(
+        self,
+        decay: float = 0.99,
+        max_: float = 1e8,
+        percentile_low: float = 0.05,
+        percentile_high: float = 0.95,
+    ) -> None:
+        super().__init__()
+        self._decay = decay
+        self._max = torch.tensor(max_)
+        self._percentile_low = percentile_low
+        self._percentile_high = percentile_high
+        self.register_buffer("low", torch.zeros((), dtype=torch.float32))
+        self.register_buffer("high", torch.zeros((), dtype=torch.float32))
+
+    def forward(self, x: Tensor, runtime: Runtime):
+        gathered = runtime.all_gather(x.detach()).float()
+        low = torch.quantile(gathered, self._percentile_low)
+        high = torch.quantile(gathered, self._percentile_high)
+        self.low = self._decay * self.low + (1 - self._decay) * low.to(self.low.device)
+        self.high = self._decay * self.high + (1 - self._decay) * high.to(self.high.device)
+        invscale = torch.max(1 / self._max.to(self.low.device), self.high - self.low)
+        return self.low.detach(), invscale.detach()
+
+
+def compute_lambda_values(rewards: Tensor, values: Tensor, continues: Tensor, lmbda: float = 0.95) -> Tensor:
+    """λ-returns with gradient flow, via the fused reverse-scan kernel:
+    L_t = r_t + c_t ((1-λ) v_t + λ L_{t+1}), bootstrap L_T = v_{T-1}
+    (exact formulation of dreamer_v3/utils.py:66-77)."""
+    return ops.lambda_values(rewards, values, continues, lmbda)
+
+
+def prepare_obs(
+    runtime: Runtime, obs: Dict[str, np.ndarray], cnn_keys: Sequence[str] = (), num_envs: int = 1
+) -> Dict[str, Tensor]:
+    torch_obs = {}
+    with torch.no_grad():
+        for k, v in obs.items():
+            t = torch.as_tensor(np.ascontiguousarray(v), device=runtime.device)
+            if k in cnn_keys:
+                t = t.view(1, num_envs, *v.shape[-3:])
+                t = ops.normalize_obs(t)
+            else:
+                t = t.float().view(1, num_envs, -1)
+            torch_obs[k] = t.to(runtime.param_dtype) if t.dtype.is_floating_point else t
+    return torch_obs
+
+
+@torch.no_grad()
+def test(
+    player: Any,
+    runtime: Runtime,
+    env_fn: Any,
+    cfg: Any,
+    log_dir: str = ".",
+    greedy: bool = True,
+) -> float:
+    """One greedy episode (parity: dreamer_v3/utils.py:94-139)."""
+    env = env_fn()
+    obs, _ = env.reset(seed=cfg.seed)
+    player.num_envs = 1
+    player.init_states()
+    done = False
+    cum_reward = 0.0
+    cnn_keys = list(cfg.algo.cnn_keys.encoder or [])
+    while not done:
+        batched = {k: np.expand_dims(np.asarray(v), 0) for k, v in obs.items()}
+        torch_obs = prepare_obs(runtime, batched, cnn_keys=cnn_keys, num_envs=1)
+        actions = player.get_actions(torch_obs, greedy=greedy)
+        if player.actor.is_continuous:
+            real_actions = torch.cat(actions, -1).cpu().numpy().reshape(-1)
+        else:
+            real_actions = np.array([a.argmax(dim=-1).cpu().numpy().item() for a in actions])
+            real_actions = real_actions[0] if len(real_actions) == 1 else real_actions
+        obs, reward, term, trunc, _ = env.step(real_actions)
+        cum_reward += float(reward)
+        done = bool(term or trunc)
+    env.close()
+    return cum_reward

@@ -1,0 +1,584 @@
+// CDNA4 (gfx950 / MI355X) kernels for sheeprl-amd.
+//
+// Replaces the PyTorch-op subgraphs listed in SURVEY.md §2.8 with fused HIP
+// kernels: symlog/symexp (utils.py:148-154), fused LayerNorm+SiLU (the MLP/CNN
+// epilogue, models.py:16/122), the LayerNormGRUCell post-GEMM gate math
+// (models.py:396-403), the GAE and λ-return reverse scans (utils.py:63-100,
+// dreamer_v3/utils.py:66-77), multi-tensor Adam/EMA (optim) and uint8 obs
+// normalization (dreamer_v3/utils.py:80-91).
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  * wave = 64 lanes; blocks are multiples of 64 (256 default).
+//  * reductions: __shfl_xor over 64-lane waves, then LDS across waves.
+//  * elementwise kernels are grid-stride, vectorized 16B/lane where the
+//    tensor is contiguous and size-aligned (G13).
+//  * all accumulation in fp32 regardless of storage dtype (bf16-true safe).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <vector>
+
+#define CHECK_IN(x) TORCH_CHECK(x.is_cuda() && x.is_contiguous(), #x " must be contiguous CUDA tensor")
+
+namespace {
+
+constexpr int kBlock = 256;
+
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// block-wide sum over kBlock threads (4 waves)
+__device__ __forceinline__ float block_sum(float v, float* lds) {
+  v = wave_sum(v);
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  if (lane == 0) lds[wid] = v;
+  __syncthreads();
+  float out = 0.f;
+  if (threadIdx.x < (blockDim.x >> 6)) out = lds[threadIdx.x];
+  out = wave_sum(out);  // lanes beyond nwaves hold 0
+  // broadcast via lds
+  if (threadIdx.x == 0) lds[8] = out;
+  __syncthreads();
+  return lds[8];
+}
+
+template <typename T>
+__device__ __forceinline__ float ld(const T* p, long i) {
+  return static_cast<float>(p[i]);
+}
+template <>
+__device__ __forceinline__ float ld<__hip_bfloat16>(const __hip_bfloat16* p, long i) {
+  return __bfloat162float(p[i]);
+}
+
+template <typename T>
+__device__ __forceinline__ void st(T* p, long i, float v) {
+  p[i] = static_cast<T>(v);
+}
+template <>
+__device__ __forceinline__ void st<__hip_bfloat16>(__hip_bfloat16* p, long i, float v) {
+  p[i] = __float2bfloat16(v);
+}
+
+// ---------------------------------------------------------------------------
+// symlog / symexp
+// ---------------------------------------------------------------------------
+
+template <typename T, int OP>
+__global__ void symmath_kernel(const T* __restrict__ x, const T* __restrict__ gy, T* __restrict__ y, long n) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
+    float v = ld(x, i);
+    float a = fabsf(v);
+    float s = v >= 0.f ? 1.f : -1.f;
+    float out;
+    if (OP == 0) out = s * log1pf(a);                         // symlog fwd
+    else if (OP == 1) out = ld(gy, i) / (1.f + a);            // symlog bwd
+    else if (OP == 2) out = s * (expf(a) - 1.f);              // symexp fwd
+    else out = ld(gy, i) * expf(a);                           // symexp bwd
+    st(y, i, out);
+  }
+}
+
+template <int OP>
+torch::Tensor symmath(const torch::Tensor& x, const c10::optional<torch::Tensor>& gy) {
+  CHECK_IN(x);
+  auto y = torch::empty_like(x);
+  long n = x.numel();
+  if (n == 0) return y;
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "symmath", [&] {
+    using T = scalar_t;
+    const T* gp = gy.has_value() ? (const T*)gy->data_ptr() : nullptr;
+    hipLaunchKernelGGL((symmath_kernel<T, OP>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)x.data_ptr(), gp, (T*)y.data_ptr(), n);
+  });
+  return y;
+}
+
+// ---------------------------------------------------------------------------
+// fused LayerNorm (+SiLU) forward/backward; rows of length D
+// ---------------------------------------------------------------------------
+
+template <typename T, bool SILU>
+__global__ void ln_act_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
+                                  const float* __restrict__ b, T* __restrict__ y, float* __restrict__ mean_out,
+                                  float* __restrict__ rstd_out, int D, float eps) {
+  __shared__ float lds[9];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)D;
+  T* yr = y + row * (long)D;
+  float s = 0.f;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) s += ld(xr, j);
+  float mean = block_sum(s, lds) / D;
+  __syncthreads();
+  float s2 = 0.f;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float d = ld(xr, j) - mean;
+    s2 += d * d;
+  }
+  float var = block_sum(s2, lds) / D;
+  float rstd = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float xhat = (ld(xr, j) - mean) * rstd;
+    float z = xhat * w[j] + b[j];
+    if (SILU) z = z / (1.f + expf(-z)) ;
+    st(yr, j, z);
+  }
+}
+
+std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& b,
+                                      double eps, bool silu) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dim() == 2, "ln_act_fwd expects [N, D]");
+  long N = x.size(0);
+  int D = (int)x.size(1);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({N}, x.options().dtype(at::kFloat));
+  auto rstd = torch::empty({N}, x.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  auto bf = b.to(at::kFloat).contiguous();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_fwd", [&] {
+    using T = scalar_t;
+    if (silu)
+      hipLaunchKernelGGL((ln_act_fwd_kernel<T, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                         (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(), (T*)y.data_ptr(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+    else
+      hipLaunchKernelGGL((ln_act_fwd_kernel<T, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                         (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(), (T*)y.data_ptr(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+  });
+  return {y, mean, rstd};
+}
+
+template <typename T, bool SILU>
+__global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict__ x, const float* __restrict__ w,
+                                  const float* __restrict__ b, const float* __restrict__ mean,
+                                  const float* __restrict__ rstd, T* __restrict__ gx, float* __restrict__ gw,
+                                  float* __restrict__ gb, int D) {
+  __shared__ float lds[9];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)D;
+  const T* gr = gy + row * (long)D;
+  T* gxr = gx + row * (long)D;
+  const float m = mean[row], r = rstd[row];
+  // first pass: compute gz and the two row sums
+  float s1 = 0.f, s2 = 0.f;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float xhat = (ld(xr, j) - m) * r;
+    float gz = ld(gr, j);
+    if (SILU) {
+      float z = xhat * w[j] + b[j];
+      float sig = 1.f / (1.f + expf(-z));
+      gz *= sig * (1.f + z * (1.f - sig));
+    }
+    float gxhat = gz * w[j];
+    s1 += gxhat;
+    s2 += gxhat * xhat;
+  }
+  float S1 = block_sum(s1, lds) / D;
+  __syncthreads();
+  float S2 = block_sum(s2, lds) / D;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float xhat = (ld(xr, j) - m) * r;
+    float gz = ld(gr, j);
+    if (SILU) {
+      float z = xhat * w[j] + b[j];
+      float sig = 1.f / (1.f + expf(-z));
+      gz *= sig * (1.f + z * (1.f - sig));
+    }
+    atomicAdd(&gw[j], gz * xhat);
+    atomicAdd(&gb[j], gz);
+    float gxhat = gz * w[j];
+    st(gxr, j, (gxhat - S1 - xhat * S2) * r);
+  }
+}
+
+std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tensor& x, const torch::Tensor& w,
+                                      const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd,
+                                      bool silu) {
+  CHECK_IN(gy);
+  CHECK_IN(x);
+  long N = x.size(0);
+  int D = (int)x.size(1);
+  auto gx = torch::empty_like(x);
+  auto gw = torch::zeros({D}, x.options().dtype(at::kFloat));
+  auto gb = torch::zeros({D}, x.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  auto bf = b.to(at::kFloat).contiguous();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_bwd", [&] {
+    using T = scalar_t;
+    if (silu)
+      hipLaunchKernelGGL((ln_act_bwd_kernel<T, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                         (const T*)gy.data_ptr(), (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
+                         gb.data_ptr<float>(), D);
+    else
+      hipLaunchKernelGGL((ln_act_bwd_kernel<T, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                         (const T*)gy.data_ptr(), (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
+                         gb.data_ptr<float>(), D);
+  });
+  return {gx, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
+}
+
+// ---------------------------------------------------------------------------
+// GRU gates: z = LN(y)*w+b; r,c,u = chunk(z,3); h' = σ(u-1)*tanh(σ(r)*c) + (1-σ(u-1))*h
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ h, const float* __restrict__ w,
+                                     const float* __restrict__ b, T* __restrict__ hout, float* __restrict__ mean_out,
+                                     float* __restrict__ rstd_out, int H, float eps) {
+  __shared__ float lds[9];
+  const long row = blockIdx.x;
+  const int D = 3 * H;
+  const T* yr = y + row * (long)D;
+  const T* hr = h + row * (long)H;
+  T* outr = hout + row * (long)H;
+  float s = 0.f;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) s += ld(yr, j);
+  float mean = block_sum(s, lds) / D;
+  __syncthreads();
+  float s2 = 0.f;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float d = ld(yr, j) - mean;
+    s2 += d * d;
+  }
+  float var = block_sum(s2, lds) / D;
+  float rstd = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  for (int j = threadIdx.x; j < H; j += blockDim.x) {
+    float zr = ((ld(yr, j) - mean) * rstd) * w[j] + b[j];
+    float zc = ((ld(yr, H + j) - mean) * rstd) * w[H + j] + b[H + j];
+    float zu = ((ld(yr, 2 * H + j) - mean) * rstd) * w[2 * H + j] + b[2 * H + j];
+    float r = 1.f / (1.f + expf(-zr));
+    float c = tanhf(r * zc);
+    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    st(outr, j, u * c + (1.f - u) * ld(hr, j));
+  }
+}
+
+std::vector<torch::Tensor> gru_gates_fwd(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
+                                         const torch::Tensor& b, double eps) {
+  CHECK_IN(y);
+  CHECK_IN(h);
+  TORCH_CHECK(y.dim() == 2 && h.dim() == 2 && y.size(1) == 3 * h.size(1), "gru_gates_fwd shapes");
+  long N = y.size(0);
+  int H = (int)h.size(1);
+  auto hout = torch::empty_like(h);
+  auto mean = torch::empty({N}, y.options().dtype(at::kFloat));
+  auto rstd = torch::empty({N}, y.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  auto bf = b.to(at::kFloat).contiguous();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_fwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((gru_gates_fwd_kernel<T>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                       (const T*)y.data_ptr(), (const T*)h.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
+                       (T*)hout.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), H, (float)eps);
+  });
+  return {hout, mean, rstd};
+}
+
+template <typename T>
+__global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restrict__ y, const T* __restrict__ h,
+                                     const float* __restrict__ w, const float* __restrict__ b,
+                                     const float* __restrict__ mean, const float* __restrict__ rstd,
+                                     T* __restrict__ gy, T* __restrict__ ghprev, float* __restrict__ gw,
+                                     float* __restrict__ gb, int H) {
+  __shared__ float lds[9];
+  const long row = blockIdx.x;
+  const int D = 3 * H;
+  const T* yr = y + row * (long)D;
+  const T* hr = h + row * (long)H;
+  const T* ghr = gh + row * (long)H;
+  T* gyr = gy + row * (long)D;
+  T* ghp = ghprev + row * (long)H;
+  const float m = mean[row], rs = rstd[row];
+  // recompute gates and form gz for this thread's columns; accumulate LN sums
+  float s1 = 0.f, s2 = 0.f;
+  for (int j = threadIdx.x; j < H; j += blockDim.x) {
+    float xh_r = (ld(yr, j) - m) * rs;
+    float xh_c = (ld(yr, H + j) - m) * rs;
+    float xh_u = (ld(yr, 2 * H + j) - m) * rs;
+    float zr = xh_r * w[j] + b[j];
+    float zc = xh_c * w[H + j] + b[H + j];
+    float zu = xh_u * w[2 * H + j] + b[2 * H + j];
+    float r = 1.f / (1.f + expf(-zr));
+    float rc = r * zc;
+    float c = tanhf(rc);
+    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    float g = ld(ghr, j);
+    float gu = g * (c - ld(hr, j));
+    float gc = g * u;
+    float gzu = gu * u * (1.f - u);
+    float grc = gc * (1.f - c * c);
+    float gzc = grc * r;
+    float gr = grc * zc;
+    float gzr = gr * r * (1.f - r);
+    st(ghp, j, g * (1.f - u));
+    // store gz temporarily in gy (pre-LN-backward)
+    st(gyr, j, gzr);
+    st(gyr, H + j, gzc);
+    st(gyr, 2 * H + j, gzu);
+    atomicAdd(&gw[j], gzr * xh_r);
+    atomicAdd(&gb[j], gzr);
+    atomicAdd(&gw[H + j], gzc * xh_c);
+    atomicAdd(&gb[H + j], gzc);
+    atomicAdd(&gw[2 * H + j], gzu * xh_u);
+    atomicAdd(&gb[2 * H + j], gzu);
+    float gxh_r = gzr * w[j];
+    float gxh_c = gzc * w[H + j];
+    float gxh_u = gzu * w[2 * H + j];
+    s1 += gxh_r + gxh_c + gxh_u;
+    s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
+  }
+  float S1 = block_sum(s1, lds) / D;
+  __syncthreads();
+  float S2 = block_sum(s2, lds) / D;
+  __syncthreads();
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float xhat = (ld(yr, j) - m) * rs;
+    float gz = ld(gyr, j);
+    float gxhat = gz * w[j];
+    st(gyr, j, (gxhat - S1 - xhat * S2) * rs);
+  }
+}
+
+std::vector<torch::Tensor> gru_gates_bwd(const torch::Tensor& gh, const torch::Tensor& y, const torch::Tensor& h,
+                                         const torch::Tensor& w, const torch::Tensor& b, const torch::Tensor& mean,
+                                         const torch::Tensor& rstd) {
+  CHECK_IN(gh);
+  CHECK_IN(y);
+  CHECK_IN(h);
+  long N = y.size(0);
+  int H = (int)h.size(1);
+  auto gy = torch::empty_like(y);
+  auto ghprev = torch::empty_like(h);
+  auto gw = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
+  auto gb = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
+  auto wf = w.to(at::kFloat).contiguous();
+  auto bf = b.to(at::kFloat).contiguous();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_bwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((gru_gates_bwd_kernel<T>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                       (const T*)gh.data_ptr(), (const T*)y.data_ptr(), (const T*)h.data_ptr(), wf.data_ptr<float>(),
+                       bf.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gy.data_ptr(),
+                       (T*)ghprev.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), H);
+  });
+  return {gy, ghprev, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
+}
+
+// ---------------------------------------------------------------------------
+// reverse scans (fp32, [T, B] layout, lane-per-column)
+// ---------------------------------------------------------------------------
+
+__global__ void gae_scan_kernel(const float* __restrict__ rewards, const float* __restrict__ values,
+                                const float* __restrict__ dones, const float* __restrict__ next_value,
+                                float* __restrict__ adv, int T, long B, float gamma, float lam) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < B; i += (long)gridDim.x * blockDim.x) {
+    float lastgaelam = 0.f;
+    float nnt = 1.f - dones[(long)(T - 1) * B + i];
+    float nv = next_value[i];
+    for (int t = T - 1; t >= 0; --t) {
+      if (t < T - 1) {
+        nnt = 1.f - dones[(long)t * B + i];
+        nv = values[(long)(t + 1) * B + i];
+      }
+      float delta = rewards[(long)t * B + i] + gamma * nv * nnt - values[(long)t * B + i];
+      lastgaelam = delta + gamma * lam * nnt * lastgaelam;
+      adv[(long)t * B + i] = lastgaelam;
+    }
+  }
+}
+
+torch::Tensor gae_scan(const torch::Tensor& rewards, const torch::Tensor& values, const torch::Tensor& dones,
+                       const torch::Tensor& next_value, double gamma, double lam) {
+  CHECK_IN(rewards);
+  int T = (int)rewards.size(0);
+  long B = rewards.numel() / T;
+  auto adv = torch::empty_like(rewards);
+  int blocks = (int)std::min((B + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(gae_scan_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(), rewards.data_ptr<float>(),
+                     values.data_ptr<float>(), dones.data_ptr<float>(), next_value.data_ptr<float>(),
+                     adv.data_ptr<float>(), T, B, (float)gamma, (float)lam);
+  return adv;
+}
+
+__global__ void lambda_fwd_kernel(const float* __restrict__ r, const float* __restrict__ nv,
+                                  const float* __restrict__ c, float* __restrict__ out, int T, long B, float lam) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < B; i += (long)gridDim.x * blockDim.x) {
+    float nxt = nv[(long)(T - 1) * B + i];
+    for (int t = T - 1; t >= 0; --t) {
+      nxt = r[(long)t * B + i] + c[(long)t * B + i] * ((1.f - lam) * nv[(long)t * B + i] + lam * nxt);
+      out[(long)t * B + i] = nxt;
+    }
+  }
+}
+
+torch::Tensor lambda_scan_fwd(const torch::Tensor& r, const torch::Tensor& nv, const torch::Tensor& c, double lam) {
+  CHECK_IN(r);
+  int T = (int)r.size(0);
+  long B = r.numel() / T;
+  auto out = torch::empty_like(r);
+  int blocks = (int)std::min((B + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(lambda_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(), r.data_ptr<float>(),
+                     nv.data_ptr<float>(), c.data_ptr<float>(), out.data_ptr<float>(), T, B, (float)lam);
+  return out;
+}
+
+__global__ void lambda_bwd_kernel(const float* __restrict__ gy, const float* __restrict__ c, float* __restrict__ gr,
+                                  float* __restrict__ gnv, int T, long B, float lam) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < B; i += (long)gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int t = 0; t < T; ++t) {
+      float prev_c = (t > 0) ? c[(long)(t - 1) * B + i] : 0.f;
+      acc = gy[(long)t * B + i] + ((t > 0) ? prev_c * lam * acc : 0.f);
+      gr[(long)t * B + i] = acc;
+      gnv[(long)t * B + i] = acc * c[(long)t * B + i] * (1.f - lam);
+    }
+    long last = (long)(T - 1) * B + i;
+    gnv[last] += acc * c[last] * lam;
+  }
+}
+
+std::vector<torch::Tensor> lambda_scan_bwd(const torch::Tensor& gy, const torch::Tensor& c, double lam) {
+  CHECK_IN(gy);
+  int T = (int)gy.size(0);
+  long B = gy.numel() / T;
+  auto gr = torch::empty_like(gy);
+  auto gnv = torch::empty_like(gy);
+  int blocks = (int)std::min((B + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(lambda_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(), gy.data_ptr<float>(),
+                     c.data_ptr<float>(), gr.data_ptr<float>(), gnv.data_ptr<float>(), T, B, (float)lam);
+  return {gr, gnv};
+}
+
+// ---------------------------------------------------------------------------
+// multi-tensor Adam / EMA
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void adam_kernel(T* __restrict__ p, const T* __restrict__ g, float* __restrict__ m, float* __restrict__ v,
+                            long n, float lr, float b1, float b2, float eps, float wd, float bc1, float bc2) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
+    float gf = ld(g, i);
+    float pf = ld(p, i);
+    if (wd != 0.f) gf += wd * pf;
+    float mi = b1 * m[i] + (1.f - b1) * gf;
+    float vi = b2 * v[i] + (1.f - b2) * gf * gf;
+    m[i] = mi;
+    v[i] = vi;
+    float upd = (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    st(p, i, pf - lr * upd);
+  }
+}
+
+void adam_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads, std::vector<torch::Tensor> ms,
+               std::vector<torch::Tensor> vs, double lr, double b1, double b2, double eps, double wd, double bc1,
+               double bc2) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  for (size_t k = 0; k < params.size(); ++k) {
+    auto& p = params[k];
+    long n = p.numel();
+    if (n == 0) continue;
+    int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+    auto gc = grads[k].contiguous();
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, p.scalar_type(), "adam_step", [&] {
+      using T = scalar_t;
+      hipLaunchKernelGGL((adam_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(), (T*)p.data_ptr(),
+                         (const T*)gc.data_ptr(), ms[k].data_ptr<float>(), vs[k].data_ptr<float>(), n, (float)lr,
+                         (float)b1, (float)b2, (float)eps, (float)wd, (float)bc1, (float)bc2);
+    });
+  }
+}
+
+template <typename T>
+__global__ void ema_kernel(T* __restrict__ t, const T* __restrict__ s, long n, float tau) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
+    st(t, i, (1.f - tau) * ld(t, i) + tau * ld(s, i));
+  }
+}
+
+void ema_update(std::vector<torch::Tensor> tgts, std::vector<torch::Tensor> srcs, double tau) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  for (size_t k = 0; k < tgts.size(); ++k) {
+    long n = tgts[k].numel();
+    if (n == 0) continue;
+    int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, tgts[k].scalar_type(), "ema_update", [&] {
+      using T = scalar_t;
+      hipLaunchKernelGGL((ema_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(), (T*)tgts[k].data_ptr(),
+                         (const T*)srcs[k].data_ptr(), n, (float)tau);
+    });
+  }
+}
+
+// ---------------------------------------------------------------------------
+// uint8 obs -> float in [-0.5, 0.5]
+// ---------------------------------------------------------------------------
+
+__global__ void obs_norm_kernel(const unsigned char* __restrict__ x, float* __restrict__ y, long n) {
+  const long n4 = n / 4;
+  const uchar4* x4 = reinterpret_cast<const uchar4*>(x);
+  float4* y4 = reinterpret_cast<float4*>(y);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n4; i += (long)gridDim.x * blockDim.x) {
+    uchar4 v = x4[i];
+    y4[i] = make_float4(v.x / 255.f - 0.5f, v.y / 255.f - 0.5f, v.z / 255.f - 0.5f, v.w / 255.f - 0.5f);
+  }
+  for (long i = n4 * 4 + blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x)
+    y[i] = x[i] / 255.f - 0.5f;
+}
+
+torch::Tensor obs_norm(const torch::Tensor& x) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.scalar_type() == at::kByte, "obs_norm expects uint8");
+  auto y = torch::empty(x.sizes(), x.options().dtype(at::kFloat));
+  long n = x.numel();
+  int blocks = (int)std::min((n / 4 + kBlock - 1) / kBlock + 1, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(obs_norm_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     (const unsigned char*)x.data_ptr(), y.data_ptr<float>(), n);
+  return y;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("symlog_fwd", [](const torch::Tensor& x) { return symmath<0>(x, c10::nullopt); });
+  m.def("symlog_bwd", [](const torch::Tensor& x, const torch::Tensor& g) { return symmath<1>(x, g); });
+  m.def("symexp_fwd", [](const torch::Tensor& x) { return symmath<2>(x, c10::nullopt); });
+  m.def("symexp_bwd", [](const torch::Tensor& x, const torch::Tensor& g) { return symmath<3>(x, g); });
+  m.def("ln_act_fwd", &ln_act_fwd);
+  m.def("ln_act_bwd", &ln_act_bwd);
+  m.def("gru_gates_fwd", &gru_gates_fwd);
+  m.def("gru_gates_bwd", &gru_gates_bwd);
+  m.def("gae_scan", &gae_scan);
+  m.def("lambda_scan_fwd", &lambda_scan_fwd);
+  m.def("lambda_scan_bwd", &lambda_scan_bwd);
+  m.def("adam_step", &adam_step);
+  m.def("ema_update", &ema_update);
+  m.def("obs_norm", &obs_norm);
+}

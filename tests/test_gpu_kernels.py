@@ -1,0 +1,196 @@
+"""GPU kernel numerics: every HIP kernel vs the plain PyTorch fp32 reference
+(the CPU path of the same op).  Run on MI355X via gpurun: pytest -m gpu."""
+
+import pytest
+import torch
+
+requires_gpu = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from sheeprl_amd import ops
+    from sheeprl_amd.ops import has_ext
+
+
+@pytest.fixture(scope="module", autouse=True)
+def _check_env():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    # on a GPU box the extension MUST be present: fail loudly, never fall back
+    from sheeprl_amd.ops import has_ext
+
+    assert has_ext(), "HIP extension _sheep_hip not built — GPU path would silently degrade"
+
+
+def _cpu_gpu(x):
+    return x, x.cuda()
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_symlog_symexp(dtype):
+    x = (torch.randn(1000) * 5).to(dtype)
+    tol = 1e-6 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(ops.symlog(x.cuda()).cpu().float(), ops.symlog(x.float()), atol=tol, rtol=tol)
+    y = torch.randn(1000).to(dtype) * 2
+    assert torch.allclose(ops.symexp(y.cuda()).cpu().float(), ops.symexp(y.float()), atol=tol, rtol=2e-2)
+
+
+@requires_gpu
+def test_symlog_grad_gpu():
+    x = torch.randn(257, requires_grad=True)
+    xg = x.detach().clone().cuda().requires_grad_()
+    g = torch.randn(257)
+    ops.symlog(x).backward(g)
+    ops.symlog(xg).backward(g.cuda())
+    assert torch.allclose(x.grad, xg.grad.cpu(), atol=1e-6)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("D", [255, 512, 1536, 4096])
+def test_ln_act(dtype, D):
+    torch.manual_seed(0)
+    N = 64
+    x = torch.randn(N, D)
+    w = torch.randn(D)
+    b = torch.randn(D)
+    ref = ops.layer_norm_act(x.requires_grad_(), w.requires_grad_(), b.requires_grad_(), 1e-3, "silu")
+    xg = x.detach().to(dtype).cuda().requires_grad_()
+    wg = w.detach().cuda().requires_grad_()
+    bg = b.detach().cuda().requires_grad_()
+    got = ops.layer_norm_act(xg, wg, bg, 1e-3, "silu")
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(got.cpu().float(), ref.detach(), atol=tol, rtol=tol)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.to(dtype).cuda())
+    btol = 1e-4 if dtype == torch.float32 else 8e-2
+    assert torch.allclose(xg.grad.cpu().float(), x.grad, atol=btol, rtol=btol)
+    assert torch.allclose(wg.grad.cpu().float(), w.grad, atol=1e-2, rtol=1e-2)
+    assert torch.allclose(bg.grad.cpu().float(), b.grad, atol=1e-2, rtol=1e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("H", [8, 512, 1024])
+def test_gru_gates(dtype, H):
+    torch.manual_seed(1)
+    B = 48
+    y = torch.randn(B, 3 * H)
+    h = torch.randn(B, H)
+    w = torch.rand(3 * H) + 0.5
+    b = torch.randn(3 * H) * 0.1
+    ref = ops.gru_gates(y.requires_grad_(), h.requires_grad_(), w.requires_grad_(), b.requires_grad_(), 1e-3)
+    yg = y.detach().to(dtype).cuda().requires_grad_()
+    hg = h.detach().to(dtype).cuda().requires_grad_()
+    wg = w.detach().cuda().requires_grad_()
+    bg = b.detach().cuda().requires_grad_()
+    got = ops.gru_gates(yg, hg, wg, bg, 1e-3)
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(got.cpu().float(), ref.detach(), atol=tol, rtol=tol)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.to(dtype).cuda())
+    btol = 1e-4 if dtype == torch.float32 else 1e-1
+    assert torch.allclose(yg.grad.cpu().float(), y.grad, atol=btol, rtol=btol)
+    assert torch.allclose(hg.grad.cpu().float(), h.grad, atol=btol, rtol=btol)
+    assert torch.allclose(wg.grad.cpu().float(), w.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(bg.grad.cpu().float(), b.grad, atol=5e-2, rtol=5e-2)
+
+
+@requires_gpu
+def test_gae_gpu():
+    T, N = 128, 16
+    torch.manual_seed(2)
+    rewards = torch.randn(T, N, 1)
+    values = torch.randn(T, N, 1)
+    dones = torch.rand(T, N, 1) < 0.05
+    next_value = torch.randn(N, 1)
+    ret_c, adv_c = ops.gae(rewards, values, dones, next_value, T, 0.99, 0.95)
+    ret_g, adv_g = ops.gae(rewards.cuda(), values.cuda(), dones.cuda(), next_value.cuda(), T, 0.99, 0.95)
+    assert torch.allclose(adv_g.cpu(), adv_c, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(ret_g.cpu(), ret_c, atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+def test_lambda_values_gpu():
+    T, B = 15, 1024
+    torch.manual_seed(3)
+    r = torch.randn(T, B, requires_grad=True)
+    nv = torch.randn(T, B, requires_grad=True)
+    c = (torch.rand(T, B) * 0.99)
+    ref = ops.lambda_values(r, nv, c, 0.95)
+    rg = r.detach().cuda().requires_grad_()
+    nvg = nv.detach().cuda().requires_grad_()
+    got = ops.lambda_values(rg, nvg, c.cuda(), 0.95)
+    assert torch.allclose(got.cpu(), ref.detach(), atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.cuda())
+    assert torch.allclose(rg.grad.cpu(), r.grad, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(nvg.grad.cpu(), nv.grad, atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_adam_gpu(dtype):
+    from sheeprl_amd.optim import FusedAdam
+
+    torch.manual_seed(4)
+    p_ref = torch.nn.Parameter(torch.randn(1000))
+    p_gpu = torch.nn.Parameter(p_ref.detach().to(dtype).cuda())
+    o_ref = FusedAdam([p_ref], lr=1e-2)
+    o_gpu = FusedAdam([p_gpu], lr=1e-2)
+    for _ in range(10):
+        g = torch.randn(1000)
+        p_ref.grad = g.clone()
+        p_gpu.grad = g.to(dtype).cuda()
+        o_ref.step()
+        o_gpu.step()
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(p_gpu.detach().cpu().float(), p_ref.detach(), atol=tol, rtol=tol)
+
+
+@requires_gpu
+def test_ema_gpu():
+    t = [torch.randn(100).cuda(), torch.randn(7, 3).cuda()]
+    s = [torch.randn(100).cuda(), torch.randn(7, 3).cuda()]
+    tc = [x.cpu().clone() for x in t]
+    sc = [x.cpu().clone() for x in s]
+    ops.ema_update_(t, s, 0.02)
+    ops.ema_update_(tc, sc, 0.02)
+    for a, b in zip(t, tc):
+        assert torch.allclose(a.cpu(), b, atol=1e-6)
+
+
+@requires_gpu
+def test_obs_norm_gpu():
+    x = torch.randint(0, 256, (4, 3, 64, 64), dtype=torch.uint8)
+    got = ops.normalize_obs(x.cuda()).cpu()
+    ref = ops.normalize_obs(x)
+    assert torch.allclose(got, ref, atol=1e-6)
+
+
+@requires_gpu
+def test_twohot_gpu():
+    x = torch.randn(512, 1).cuda() * 30
+    enc = ops.two_hot_encoder(x, support_range=20, num_buckets=255)
+    dec = ops.two_hot_decoder(enc, support_range=20)
+    assert torch.allclose(dec.cpu(), x.cpu().clamp(-4.8e8, 4.8e8), rtol=2e-2, atol=2e-2)
+
+
+@requires_gpu
+def test_gru_cell_module_gpu():
+    """LayerNormGRUCell end-to-end vs CPU fp32 (GEMM via hipBLASLt + fused gates)."""
+    from sheeprl_amd.models import LayerNormGRUCell
+
+    torch.manual_seed(5)
+    cell = LayerNormGRUCell(32, 64, layer_norm=True)
+    x = torch.randn(16, 32)
+    h = torch.randn(16, 64)
+    ref = cell(x, h)
+    cell_g = LayerNormGRUCell(32, 64, layer_norm=True)
+    cell_g.load_state_dict(cell.state_dict())
+    cell_g = cell_g.cuda()
+    got = cell_g(x.cuda(), h.cuda())
+    assert torch.allclose(got.cpu(), ref, atol=1e-4, rtol=1e-4)
