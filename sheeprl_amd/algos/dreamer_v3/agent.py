@@ -266,6 +266,10 @@ class RSSM(nn.Module):
         else:
             self.register_buffer("initial_recurrent_state", init)
 
+    @property
+    def _dtype(self) -> torch.dtype:
+        return next(self.recurrent_model.parameters()).dtype
+
     def get_initial_states(self, batch_shape: Sequence[int]) -> Tuple[Tensor, Tensor]:
         initial_recurrent_state = torch.tanh(self.initial_recurrent_state).expand(*batch_shape, -1)
         initial_posterior = self._transition(initial_recurrent_state, sample_state=False)[1]
@@ -274,11 +278,13 @@ class RSSM(nn.Module):
     def dynamic(
         self, posterior: Tensor, recurrent_state: Tensor, action: Tensor, embedded_obs: Tensor, is_first: Tensor
     ) -> Tuple[Tensor, Tensor, Tensor, Tensor, Tensor]:
-        action = (1 - is_first) * action
+        dt = self._dtype
+        is_first = is_first.to(dt)
+        action = (1 - is_first) * action.to(dt)
         initial_recurrent_state, initial_posterior = self.get_initial_states(recurrent_state.shape[:2])
-        recurrent_state = (1 - is_first) * recurrent_state + is_first * initial_recurrent_state
-        posterior = posterior.view(*posterior.shape[:-2], -1)
-        posterior = (1 - is_first) * posterior + is_first * initial_posterior.view_as(posterior)
+        recurrent_state = (1 - is_first) * recurrent_state.to(dt) + is_first * initial_recurrent_state.to(dt)
+        posterior = posterior.view(*posterior.shape[:-2], -1).to(dt)
+        posterior = (1 - is_first) * posterior + is_first * initial_posterior.view_as(posterior).to(dt)
         recurrent_state = self.recurrent_model(torch.cat((posterior, action), -1), recurrent_state)
         prior_logits, prior = self._transition(recurrent_state)
         posterior_logits, posterior = self._representation(recurrent_state, embedded_obs)
@@ -294,17 +300,19 @@ class RSSM(nn.Module):
         return logits.view(*logits.shape[:-2], -1)
 
     def _representation(self, recurrent_state: Tensor, embedded_obs: Tensor) -> Tuple[Tensor, Tensor]:
-        logits = self.representation_model(torch.cat((recurrent_state, embedded_obs), -1))
+        dt = self._dtype
+        logits = self.representation_model(torch.cat((recurrent_state.to(dt), embedded_obs.to(dt)), -1))
         logits = self._uniform_mix(logits)
-        return logits, compute_stochastic_state(logits, discrete=self.discrete)
+        return logits, compute_stochastic_state(logits, discrete=self.discrete).to(dt)
 
     def _transition(self, recurrent_out: Tensor, sample_state: bool = True) -> Tuple[Tensor, Tensor]:
-        logits = self.transition_model(recurrent_out)
+        logits = self.transition_model(recurrent_out.to(self._dtype))
         logits = self._uniform_mix(logits)
-        return logits, compute_stochastic_state(logits, discrete=self.discrete, sample=sample_state)
+        return logits, compute_stochastic_state(logits, discrete=self.discrete, sample=sample_state).to(self._dtype)
 
     def imagination(self, prior: Tensor, recurrent_state: Tensor, actions: Tensor) -> Tuple[Tensor, Tensor]:
-        recurrent_state = self.recurrent_model(torch.cat((prior, actions), -1), recurrent_state)
+        dt = self._dtype
+        recurrent_state = self.recurrent_model(torch.cat((prior.to(dt), actions.to(dt)), -1), recurrent_state.to(dt))
         _, imagined_prior = self._transition(recurrent_state)
         return imagined_prior, recurrent_state
 
@@ -317,19 +325,22 @@ class DecoupledRSSM(RSSM):
     def dynamic(  # type: ignore[override]
         self, posterior: Tensor, recurrent_state: Tensor, action: Tensor, is_first: Tensor
     ) -> Tuple[Tensor, Tensor, Tensor]:
-        action = (1 - is_first) * action
+        dt = self._dtype
+        is_first = is_first.to(dt)
+        action = (1 - is_first) * action.to(dt)
         initial_recurrent_state, initial_posterior = self.get_initial_states(recurrent_state.shape[:2])
-        recurrent_state = (1 - is_first) * recurrent_state + is_first * initial_recurrent_state
-        posterior = posterior.view(*posterior.shape[:-2], -1)
-        posterior = (1 - is_first) * posterior + is_first * initial_posterior.view_as(posterior)
+        recurrent_state = (1 - is_first) * recurrent_state.to(dt) + is_first * initial_recurrent_state.to(dt)
+        posterior = posterior.view(*posterior.shape[:-2], -1).to(dt)
+        posterior = (1 - is_first) * posterior + is_first * initial_posterior.view_as(posterior).to(dt)
         recurrent_state = self.recurrent_model(torch.cat((posterior, action), -1), recurrent_state)
         prior_logits, _ = self._transition(recurrent_state)
         return recurrent_state, None, prior_logits
 
     def _representation(self, embedded_obs: Tensor) -> Tuple[Tensor, Tensor]:  # type: ignore[override]
-        logits = self.representation_model(embedded_obs)
+        dt = self._dtype
+        logits = self.representation_model(embedded_obs.to(dt))
         logits = self._uniform_mix(logits)
-        return logits, compute_stochastic_state(logits, discrete=self.discrete)
+        return logits, compute_stochastic_state(logits, discrete=self.discrete).to(dt)
 
 
 class WorldModel(nn.Module):
@@ -491,7 +502,7 @@ class PlayerDV3(nn.Module):
             _, stoch = self.rssm._representation(self.recurrent_state, embedded_obs)
         self.stochastic_state = stoch.view(*stoch.shape[:-2], self.stochastic_size * self.discrete_size)
         actions, _ = self.actor(torch.cat((self.stochastic_state, self.recurrent_state), -1), greedy, mask)
-        self.actions = torch.cat(actions, -1)
+        self.actions = torch.cat(actions, -1).to(self.stochastic_state.dtype)
         return actions
 
 
