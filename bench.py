@@ -65,6 +65,8 @@ def _setup(cfg, rank: int, world_size: int):
     from sheeprl_amd.utils.utils import Ratio, seed_everything
 
     seed_everything(cfg.seed + rank)
+    torch.backends.cudnn.benchmark = True  # MIOpen find mode: avoid naive-conv fallback
+    torch.set_float32_matmul_precision("high")
     runtime = Runtime(
         devices=world_size,
         accelerator=cfg.runtime.accelerator,

@@ -196,11 +196,18 @@ class ConvBlock(nn.Module):
             self.ln_eps = layer_norm_eps
 
     def forward(self, x: Tensor) -> Tensor:
+        if x.is_cuda and x.dim() == 4:
+            # NHWC path: MIOpen picks CK/igemm kernels and the channel
+            # LayerNorm becomes a zero-copy row norm (permute of a
+            # channels_last tensor is contiguous NHWC)
+            x = x.contiguous(memory_format=torch.channels_last)
         y = self.conv(x)
         if self.layer_norm:
             z = y.permute(0, 2, 3, 1)
             z = ops.layer_norm_act(z, self.ln_weight, self.ln_bias, self.ln_eps, self._act_name)
-            y = z.permute(0, 3, 1, 2).contiguous()
+            y = z.permute(0, 3, 1, 2)
+            if not y.is_cuda:
+                y = y.contiguous()
             if self.act is not None and not isinstance(self.act, nn.Identity):
                 y = self.act(y)
             return y

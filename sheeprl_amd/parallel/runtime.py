@@ -175,6 +175,9 @@ class Runtime:
         module = module.to(self._device)
         if self.param_dtype != torch.float32:
             module = module.to(self.param_dtype)
+        if self._device.type == "cuda":
+            # NHWC conv weights (MIOpen CK/igemm path instead of naive NCHW)
+            module = module.to(memory_format=torch.channels_last)
         if sync and self.is_distributed:
             gs = GradSync(
                 module,
