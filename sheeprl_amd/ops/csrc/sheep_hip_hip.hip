@@ -107,9 +107,12 @@ torch::Tensor symmath(const torch::Tensor& x, const c10::optional<torch::Tensor>
 // fused LayerNorm (+SiLU) forward/backward; rows of length D
 // ---------------------------------------------------------------------------
 
-template <typename T, bool SILU>
-__global__ void ln_act_fwd_kernel(const T* __restrict__ x, const float* __restrict__ w,
-                                  const float* __restrict__ b, T* __restrict__ y, float* __restrict__ mean_out,
+// W can be a different float dtype from X (bf16-true modules keep bf16 affine
+// params; the GPU tests pass fp32) — templated on both to avoid per-call
+// conversion launches.
+template <typename T, typename TW, bool SILU>
+__global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict__ w,
+                                  const TW* __restrict__ b, T* __restrict__ y, float* __restrict__ mean_out,
                                   float* __restrict__ rstd_out, int D, float eps) {
   __shared__ float lds[9];
   const long row = blockIdx.x;
@@ -132,8 +135,8 @@ __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const float* __restri
   }
   for (int j = threadIdx.x; j < D; j += blockDim.x) {
     float xhat = (ld(xr, j) - mean) * rstd;
-    float z = xhat * w[j] + b[j];
-    if (SILU) z = z / (1.f + expf(-z)) ;
+    float z = xhat * ld(w, j) + ld(b, j);
+    if (SILU) z = z / (1.f + expf(-z));
     st(yr, j, z);
   }
 }
@@ -147,63 +150,83 @@ std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tenso
   auto y = torch::empty_like(x);
   auto mean = torch::empty({N}, x.options().dtype(at::kFloat));
   auto rstd = torch::empty({N}, x.options().dtype(at::kFloat));
-  auto wf = w.to(at::kFloat).contiguous();
-  auto bf = b.to(at::kFloat).contiguous();
+  auto wc = w.contiguous();
+  auto bc = b.contiguous();
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_fwd", [&] {
     using T = scalar_t;
-    if (silu)
-      hipLaunchKernelGGL((ln_act_fwd_kernel<T, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                         (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(), (T*)y.data_ptr(),
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
-    else
-      hipLaunchKernelGGL((ln_act_fwd_kernel<T, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                         (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(), (T*)y.data_ptr(),
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_fwd_w", [&] {
+      using TW = scalar_t;
+      if (silu)
+        hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                           (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
+                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+      else
+        hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                           (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
+                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+    });
   });
   return {y, mean, rstd};
 }
 
-template <typename T, bool SILU>
-__global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict__ x, const float* __restrict__ w,
-                                  const float* __restrict__ b, const float* __restrict__ mean,
+// Two-stage weight-grad reduction: a bounded grid of blocks strides the rows,
+// each accumulating gw/gb in an LDS fp32 image of the row, with ONE atomicAdd
+// per column per block at the end (instead of one per column per ROW, which
+// serialized on atomics contention — measured 334 us/call on the DV3 bench).
+template <typename T, typename TW, bool SILU>
+__global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict__ x, const TW* __restrict__ w,
+                                  const TW* __restrict__ b, const float* __restrict__ mean,
                                   const float* __restrict__ rstd, T* __restrict__ gx, float* __restrict__ gw,
-                                  float* __restrict__ gb, int D) {
-  __shared__ float lds[9];
-  const long row = blockIdx.x;
-  const T* xr = x + row * (long)D;
-  const T* gr = gy + row * (long)D;
-  T* gxr = gx + row * (long)D;
-  const float m = mean[row], r = rstd[row];
-  // first pass: compute gz and the two row sums
-  float s1 = 0.f, s2 = 0.f;
+                                  float* __restrict__ gb, long N, int D) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* lds = smem;          // 9 floats for block_sum
+  float* gw_acc = smem + 16;  // [D]
+  float* gb_acc = gw_acc + D; // [D]
   for (int j = threadIdx.x; j < D; j += blockDim.x) {
-    float xhat = (ld(xr, j) - m) * r;
-    float gz = ld(gr, j);
-    if (SILU) {
-      float z = xhat * w[j] + b[j];
-      float sig = 1.f / (1.f + expf(-z));
-      gz *= sig * (1.f + z * (1.f - sig));
-    }
-    float gxhat = gz * w[j];
-    s1 += gxhat;
-    s2 += gxhat * xhat;
+    gw_acc[j] = 0.f;
+    gb_acc[j] = 0.f;
   }
-  float S1 = block_sum(s1, lds) / D;
   __syncthreads();
-  float S2 = block_sum(s2, lds) / D;
-  for (int j = threadIdx.x; j < D; j += blockDim.x) {
-    float xhat = (ld(xr, j) - m) * r;
-    float gz = ld(gr, j);
-    if (SILU) {
-      float z = xhat * w[j] + b[j];
-      float sig = 1.f / (1.f + expf(-z));
-      gz *= sig * (1.f + z * (1.f - sig));
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + row * (long)D;
+    const T* gr = gy + row * (long)D;
+    T* gxr = gx + row * (long)D;
+    const float m = mean[row], r = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+    for (int j = threadIdx.x; j < D; j += blockDim.x) {
+      float xhat = (ld(xr, j) - m) * r;
+      float gz = ld(gr, j);
+      if (SILU) {
+        float z = xhat * ld(w, j) + ld(b, j);
+        float sig = 1.f / (1.f + expf(-z));
+        gz *= sig * (1.f + z * (1.f - sig));
+      }
+      float gxhat = gz * ld(w, j);
+      s1 += gxhat;
+      s2 += gxhat * xhat;
     }
-    atomicAdd(&gw[j], gz * xhat);
-    atomicAdd(&gb[j], gz);
-    float gxhat = gz * w[j];
-    st(gxr, j, (gxhat - S1 - xhat * S2) * r);
+    float S1 = block_sum(s1, lds) / D;
+    __syncthreads();
+    float S2 = block_sum(s2, lds) / D;
+    for (int j = threadIdx.x; j < D; j += blockDim.x) {
+      float xhat = (ld(xr, j) - m) * r;
+      float gz = ld(gr, j);
+      if (SILU) {
+        float z = xhat * ld(w, j) + ld(b, j);
+        float sig = 1.f / (1.f + expf(-z));
+        gz *= sig * (1.f + z * (1.f - sig));
+      }
+      gw_acc[j] += gz * xhat;
+      gb_acc[j] += gz;
+      float gxhat = gz * ld(w, j);
+      st(gxr, j, (gxhat - S1 - xhat * S2) * r);
+    }
+    __syncthreads();
+  }
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    atomicAdd(&gw[j], gw_acc[j]);
+    atomicAdd(&gb[j], gb_acc[j]);
   }
 }
 
@@ -217,21 +240,27 @@ std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tens
   auto gx = torch::empty_like(x);
   auto gw = torch::zeros({D}, x.options().dtype(at::kFloat));
   auto gb = torch::zeros({D}, x.options().dtype(at::kFloat));
-  auto wf = w.to(at::kFloat).contiguous();
-  auto bf = b.to(at::kFloat).contiguous();
+  auto wc = w.contiguous();
+  auto bc = b.contiguous();
+  size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
+  TORCH_CHECK(shmem <= 160 * 1024, "ln_act_bwd: D too large for LDS accumulation");
+  int blocks = (int)std::min(N, (long)512);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_bwd", [&] {
     using T = scalar_t;
-    if (silu)
-      hipLaunchKernelGGL((ln_act_bwd_kernel<T, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                         (const T*)gy.data_ptr(), (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
-                         gb.data_ptr<float>(), D);
-    else
-      hipLaunchKernelGGL((ln_act_bwd_kernel<T, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                         (const T*)gy.data_ptr(), (const T*)x.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
-                         gb.data_ptr<float>(), D);
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_bwd_w", [&] {
+      using TW = scalar_t;
+      if (silu)
+        hipLaunchKernelGGL((ln_act_bwd_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
+                           (const T*)gy.data_ptr(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                           (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D);
+      else
+        hipLaunchKernelGGL((ln_act_bwd_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
+                           (const T*)gy.data_ptr(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                           (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D);
+    });
   });
   return {gx, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
 }
@@ -240,9 +269,9 @@ std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tens
 // GRU gates: z = LN(y)*w+b; r,c,u = chunk(z,3); h' = σ(u-1)*tanh(σ(r)*c) + (1-σ(u-1))*h
 // ---------------------------------------------------------------------------
 
-template <typename T>
-__global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ h, const float* __restrict__ w,
-                                     const float* __restrict__ b, T* __restrict__ hout, float* __restrict__ mean_out,
+template <typename T, typename TW>
+__global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ h, const TW* __restrict__ w,
+                                     const TW* __restrict__ b, T* __restrict__ hout, float* __restrict__ mean_out,
                                      float* __restrict__ rstd_out, int H, float eps) {
   __shared__ float lds[9];
   const long row = blockIdx.x;
@@ -266,9 +295,9 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
     rstd_out[row] = rstd;
   }
   for (int j = threadIdx.x; j < H; j += blockDim.x) {
-    float zr = ((ld(yr, j) - mean) * rstd) * w[j] + b[j];
-    float zc = ((ld(yr, H + j) - mean) * rstd) * w[H + j] + b[H + j];
-    float zu = ((ld(yr, 2 * H + j) - mean) * rstd) * w[2 * H + j] + b[2 * H + j];
+    float zr = ((ld(yr, j) - mean) * rstd) * ld(w, j) + ld(b, j);
+    float zc = ((ld(yr, H + j) - mean) * rstd) * ld(w, H + j) + ld(b, H + j);
+    float zu = ((ld(yr, 2 * H + j) - mean) * rstd) * ld(w, 2 * H + j) + ld(b, 2 * H + j);
     float r = 1.f / (1.f + expf(-zr));
     float c = tanhf(r * zc);
     float u = 1.f / (1.f + expf(-(zu - 1.f)));
@@ -286,80 +315,97 @@ std::vector<torch::Tensor> gru_gates_fwd(const torch::Tensor& y, const torch::Te
   auto hout = torch::empty_like(h);
   auto mean = torch::empty({N}, y.options().dtype(at::kFloat));
   auto rstd = torch::empty({N}, y.options().dtype(at::kFloat));
-  auto wf = w.to(at::kFloat).contiguous();
-  auto bf = b.to(at::kFloat).contiguous();
+  auto wc = w.contiguous();
+  auto bc = b.contiguous();
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_fwd", [&] {
     using T = scalar_t;
-    hipLaunchKernelGGL((gru_gates_fwd_kernel<T>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                       (const T*)y.data_ptr(), (const T*)h.data_ptr(), wf.data_ptr<float>(), bf.data_ptr<float>(),
-                       (T*)hout.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), H, (float)eps);
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "gru_gates_fwd_w", [&] {
+      using TW = scalar_t;
+      hipLaunchKernelGGL((gru_gates_fwd_kernel<T, TW>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                         (const T*)y.data_ptr(), (const T*)h.data_ptr(), (const TW*)wc.data_ptr(),
+                         (const TW*)bc.data_ptr(), (T*)hout.data_ptr(), mean.data_ptr<float>(),
+                         rstd.data_ptr<float>(), H, (float)eps);
+    });
   });
   return {hout, mean, rstd};
 }
 
-template <typename T>
+template <typename T, typename TW>
 __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restrict__ y, const T* __restrict__ h,
-                                     const float* __restrict__ w, const float* __restrict__ b,
+                                     const TW* __restrict__ w, const TW* __restrict__ b,
                                      const float* __restrict__ mean, const float* __restrict__ rstd,
                                      T* __restrict__ gy, T* __restrict__ ghprev, float* __restrict__ gw,
-                                     float* __restrict__ gb, int H) {
-  __shared__ float lds[9];
-  const long row = blockIdx.x;
+                                     float* __restrict__ gb, long N, int H) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* lds = smem;
+  float* gw_acc = smem + 16;       // [3H]
+  float* gb_acc = gw_acc + 3 * H;  // [3H]
   const int D = 3 * H;
-  const T* yr = y + row * (long)D;
-  const T* hr = h + row * (long)H;
-  const T* ghr = gh + row * (long)H;
-  T* gyr = gy + row * (long)D;
-  T* ghp = ghprev + row * (long)H;
-  const float m = mean[row], rs = rstd[row];
-  // recompute gates and form gz for this thread's columns; accumulate LN sums
-  float s1 = 0.f, s2 = 0.f;
-  for (int j = threadIdx.x; j < H; j += blockDim.x) {
-    float xh_r = (ld(yr, j) - m) * rs;
-    float xh_c = (ld(yr, H + j) - m) * rs;
-    float xh_u = (ld(yr, 2 * H + j) - m) * rs;
-    float zr = xh_r * w[j] + b[j];
-    float zc = xh_c * w[H + j] + b[H + j];
-    float zu = xh_u * w[2 * H + j] + b[2 * H + j];
-    float r = 1.f / (1.f + expf(-zr));
-    float rc = r * zc;
-    float c = tanhf(rc);
-    float u = 1.f / (1.f + expf(-(zu - 1.f)));
-    float g = ld(ghr, j);
-    float gu = g * (c - ld(hr, j));
-    float gc = g * u;
-    float gzu = gu * u * (1.f - u);
-    float grc = gc * (1.f - c * c);
-    float gzc = grc * r;
-    float gr = grc * zc;
-    float gzr = gr * r * (1.f - r);
-    st(ghp, j, g * (1.f - u));
-    // store gz temporarily in gy (pre-LN-backward)
-    st(gyr, j, gzr);
-    st(gyr, H + j, gzc);
-    st(gyr, 2 * H + j, gzu);
-    atomicAdd(&gw[j], gzr * xh_r);
-    atomicAdd(&gb[j], gzr);
-    atomicAdd(&gw[H + j], gzc * xh_c);
-    atomicAdd(&gb[H + j], gzc);
-    atomicAdd(&gw[2 * H + j], gzu * xh_u);
-    atomicAdd(&gb[2 * H + j], gzu);
-    float gxh_r = gzr * w[j];
-    float gxh_c = gzc * w[H + j];
-    float gxh_u = gzu * w[2 * H + j];
-    s1 += gxh_r + gxh_c + gxh_u;
-    s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
-  }
-  float S1 = block_sum(s1, lds) / D;
-  __syncthreads();
-  float S2 = block_sum(s2, lds) / D;
-  __syncthreads();
   for (int j = threadIdx.x; j < D; j += blockDim.x) {
-    float xhat = (ld(yr, j) - m) * rs;
-    float gz = ld(gyr, j);
-    float gxhat = gz * w[j];
-    st(gyr, j, (gxhat - S1 - xhat * S2) * rs);
+    gw_acc[j] = 0.f;
+    gb_acc[j] = 0.f;
+  }
+  __syncthreads();
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* yr = y + row * (long)D;
+    const T* hr = h + row * (long)H;
+    const T* ghr = gh + row * (long)H;
+    T* gyr = gy + row * (long)D;
+    T* ghp = ghprev + row * (long)H;
+    const float m = mean[row], rs = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+    for (int j = threadIdx.x; j < H; j += blockDim.x) {
+      float xh_r = (ld(yr, j) - m) * rs;
+      float xh_c = (ld(yr, H + j) - m) * rs;
+      float xh_u = (ld(yr, 2 * H + j) - m) * rs;
+      float zr = xh_r * ld(w, j) + ld(b, j);
+      float zc = xh_c * ld(w, H + j) + ld(b, H + j);
+      float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
+      float r = 1.f / (1.f + expf(-zr));
+      float rc = r * zc;
+      float c = tanhf(rc);
+      float u = 1.f / (1.f + expf(-(zu - 1.f)));
+      float g = ld(ghr, j);
+      float gu = g * (c - ld(hr, j));
+      float gc = g * u;
+      float gzu = gu * u * (1.f - u);
+      float grc = gc * (1.f - c * c);
+      float gzc = grc * r;
+      float gr = grc * zc;
+      float gzr = gr * r * (1.f - r);
+      st(ghp, j, g * (1.f - u));
+      // store gz temporarily in gy (pre-LN-backward)
+      st(gyr, j, gzr);
+      st(gyr, H + j, gzc);
+      st(gyr, 2 * H + j, gzu);
+      gw_acc[j] += gzr * xh_r;
+      gb_acc[j] += gzr;
+      gw_acc[H + j] += gzc * xh_c;
+      gb_acc[H + j] += gzc;
+      gw_acc[2 * H + j] += gzu * xh_u;
+      gb_acc[2 * H + j] += gzu;
+      float gxh_r = gzr * ld(w, j);
+      float gxh_c = gzc * ld(w, H + j);
+      float gxh_u = gzu * ld(w, 2 * H + j);
+      s1 += gxh_r + gxh_c + gxh_u;
+      s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
+    }
+    float S1 = block_sum(s1, lds) / D;
+    __syncthreads();
+    float S2 = block_sum(s2, lds) / D;
+    __syncthreads();
+    for (int j = threadIdx.x; j < D; j += blockDim.x) {
+      float xhat = (ld(yr, j) - m) * rs;
+      float gz = ld(gyr, j);
+      float gxhat = gz * ld(w, j);
+      st(gyr, j, (gxhat - S1 - xhat * S2) * rs);
+    }
+    __syncthreads();
+  }
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    atomicAdd(&gw[j], gw_acc[j]);
+    atomicAdd(&gb[j], gb_acc[j]);
   }
 }
 
@@ -375,15 +421,22 @@ std::vector<torch::Tensor> gru_gates_bwd(const torch::Tensor& gh, const torch::T
   auto ghprev = torch::empty_like(h);
   auto gw = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
   auto gb = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
-  auto wf = w.to(at::kFloat).contiguous();
-  auto bf = b.to(at::kFloat).contiguous();
+  auto wc = w.contiguous();
+  auto bc = b.contiguous();
+  size_t shmem = (16 + 6 * (size_t)H) * sizeof(float);
+  TORCH_CHECK(shmem <= 160 * 1024, "gru_gates_bwd: H too large for LDS accumulation");
+  int blocks = (int)std::min(N, (long)512);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_bwd", [&] {
     using T = scalar_t;
-    hipLaunchKernelGGL((gru_gates_bwd_kernel<T>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                       (const T*)gh.data_ptr(), (const T*)y.data_ptr(), (const T*)h.data_ptr(), wf.data_ptr<float>(),
-                       bf.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gy.data_ptr(),
-                       (T*)ghprev.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), H);
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "gru_gates_bwd_w", [&] {
+      using TW = scalar_t;
+      hipLaunchKernelGGL((gru_gates_bwd_kernel<T, TW>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
+                         (const T*)gh.data_ptr(), (const T*)y.data_ptr(), (const T*)h.data_ptr(),
+                         (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
+                         rstd.data_ptr<float>(), (T*)gy.data_ptr(), (T*)ghprev.data_ptr(), gw.data_ptr<float>(),
+                         gb.data_ptr<float>(), N, H);
+    });
   });
   return {gy, ghprev, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
 }
