@@ -209,6 +209,18 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     for _ in range(warmup):
         obs = one_iter(obs, step_data)
 
+    if os.environ.get("SHEEPRL_AMD_TORCH_PROFILE"):
+        from torch.profiler import ProfilerActivity, profile
+
+        with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+            for _ in range(2):
+                obs = one_iter(obs, step_data)
+        out = os.environ["SHEEPRL_AMD_TORCH_PROFILE"]
+        with open(out, "w") as f:
+            f.write(prof.key_averages().table(sort_by="self_cuda_time_total", row_limit=40))
+            f.write("\n\n")
+            f.write(prof.key_averages().table(sort_by="self_cpu_time_total", row_limit=40))
+
     if runtime.is_distributed:
         dist.barrier()
     if device_type == "cuda":
