@@ -171,6 +171,33 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     aggregator = MetricAggregator({})
     MetricAggregator.disabled = True
 
+    def train_fn(batch):
+        train(
+            runtime, world_model, actor, critic, target_critic,
+            world_optimizer, actor_optimizer, critic_optimizer,
+            batch, aggregator, cfg, False, actions_dim, moments,
+        )
+
+    # hipGraph-capture the whole gradient step (fwd+bwd+optimizers): the
+    # T=64 RSSM scan + imagination is ~20k tiny kernels, host-bound in eager.
+    graphed = None
+    if device_type == "cuda" and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1":
+        from sheeprl_amd.parallel.graphs import CUDAGraphStep
+
+        example = {
+            k: v[0] for k, v in rb.sample_tensors(
+                cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device=device
+            ).items()
+        }
+        try:
+            graphed = CUDAGraphStep(train_fn, example, warmup=3)
+            if rank == 0:
+                print("[bench] train step captured in a hipGraph", file=sys.stderr)
+        except Exception as e:  # noqa: BLE001
+            graphed = None
+            if rank == 0:
+                print(f"[bench] hipGraph capture failed ({e}); running eager", file=sys.stderr)
+
     def one_iter(obs, step_data):
         # --- env interaction (the real policy forward + env step + buffer add)
         with torch.inference_mode():
@@ -191,6 +218,9 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         if dones.any():
             player.init_states(np.nonzero(dones)[0].tolist())
         # --- one gradient step (replay_ratio=1 at num_envs=1)
+        from sheeprl_amd import ops as _ops
+
+        _ops.ema_update_(list(target_critic.parameters()), list(critic.parameters()), cfg.algo.critic.tau)
         sample = rb.sample_tensors(
             cfg.algo.per_rank_batch_size,
             sequence_length=seq_len,
@@ -198,11 +228,10 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             device=device,
         )
         batch = {k: v[0] for k, v in sample.items()}
-        train(
-            runtime, world_model, actor, critic, target_critic,
-            world_optimizer, actor_optimizer, critic_optimizer,
-            batch, aggregator, cfg, False, actions_dim, moments,
-        )
+        if graphed is not None:
+            graphed(batch)
+        else:
+            train_fn(batch)
         return obs
 
     # warmup

@@ -55,10 +55,11 @@ class Moments(nn.Module):
         gathered = runtime.all_gather(x.detach()).float()
         low = torch.quantile(gathered, self._percentile_low)
         high = torch.quantile(gathered, self._percentile_high)
-        self.low = self._decay * self.low + (1 - self._decay) * low.to(self.low.device)
-        self.high = self._decay * self.high + (1 - self._decay) * high.to(self.high.device)
+        # in-place EMA so the buffers keep fixed storage (hipGraph-replayable)
+        self.low.mul_(self._decay).add_(low.to(self.low.device), alpha=1 - self._decay)
+        self.high.mul_(self._decay).add_(high.to(self.high.device), alpha=1 - self._decay)
         invscale = torch.max(1 / self._max.to(self.low.device), self.high - self.low)
-        return self.low.detach(), invscale.detach()
+        return self.low.detach().clone(), invscale.detach()
 
 
 def compute_lambda_values(rewards: Tensor, values: Tensor, continues: Tensor, lmbda: float = 0.95) -> Tensor:

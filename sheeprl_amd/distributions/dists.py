@@ -107,11 +107,30 @@ class OneHotCategoricalValidateArgs(td.OneHotCategorical):
         super().__init__(probs=probs, logits=logits, validate_args=validate_args)
 
 
+def gumbel_onehot_sample(logits: Tensor) -> Tensor:
+    """Gumbel-max categorical sample as a one-hot tensor.
+
+    Same distribution as ``torch.multinomial`` sampling but built only from
+    philox RNG ops (``torch.rand_like``), so it is hipGraph-capture-safe and
+    avoids multinomial's sort.
+    """
+    u = torch.rand_like(logits)
+    g = -torch.log(-torch.log(u.clamp_min(1e-20)).clamp_min(1e-20))
+    idx = (logits + g).argmax(-1)
+    return torch.nn.functional.one_hot(idx, logits.shape[-1]).to(logits.dtype)
+
+
 class OneHotCategoricalST(td.OneHotCategorical):
     """Straight-through one-hot categorical (reference distribution.py:387)."""
 
     def __init__(self, probs: Optional[Tensor] = None, logits: Optional[Tensor] = None, validate_args: bool = False):
         super().__init__(probs=probs, logits=logits, validate_args=validate_args)
+
+    def sample(self, sample_shape=torch.Size()) -> Tensor:
+        if sample_shape != torch.Size():
+            return super().sample(sample_shape)
+        with torch.no_grad():
+            return gumbel_onehot_sample(self.logits)
 
     def rsample(self, sample_shape=torch.Size()) -> Tensor:
         sample = self.sample(sample_shape)

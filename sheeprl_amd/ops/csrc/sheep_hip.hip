@@ -568,6 +568,52 @@ void adam_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> gra
   }
 }
 
+__global__ void step_inc_kernel(float* step_t) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) step_t[0] += 1.0f;
+}
+
+// Device-side bias correction (bc = 1 - beta^step read from a device scalar),
+// so the whole optimizer step is hipGraph-replayable with correct step counts.
+template <typename T>
+__global__ void adam_dev_kernel(T* __restrict__ p, const T* __restrict__ g, float* __restrict__ m,
+                                float* __restrict__ v, const float* __restrict__ step_t, long n, float lr, float b1,
+                                float b2, float eps, float wd) {
+  const float step = step_t[0];
+  const float bc1 = 1.f - powf(b1, step);
+  const float bc2 = 1.f - powf(b2, step);
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
+    float gf = ld(g, i);
+    float pf = ld(p, i);
+    if (wd != 0.f) gf += wd * pf;
+    float mi = b1 * m[i] + (1.f - b1) * gf;
+    float vi = b2 * v[i] + (1.f - b2) * gf * gf;
+    m[i] = mi;
+    v[i] = vi;
+    float upd = (mi / bc1) / (sqrtf(vi / bc2) + eps);
+    st(p, i, pf - lr * upd);
+  }
+}
+
+void adam_step_dev(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+                   std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs, torch::Tensor step_t, double lr,
+                   double b1, double b2, double eps, double wd) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(step_inc_kernel, dim3(1), dim3(64), 0, stream.stream(), step_t.data_ptr<float>());
+  for (size_t k = 0; k < params.size(); ++k) {
+    auto& p = params[k];
+    long n = p.numel();
+    if (n == 0) continue;
+    int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+    auto gc = grads[k].contiguous();
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, p.scalar_type(), "adam_step_dev", [&] {
+      using T = scalar_t;
+      hipLaunchKernelGGL((adam_dev_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(), (T*)p.data_ptr(),
+                         (const T*)gc.data_ptr(), ms[k].data_ptr<float>(), vs[k].data_ptr<float>(),
+                         step_t.data_ptr<float>(), n, (float)lr, (float)b1, (float)b2, (float)eps, (float)wd);
+    });
+  }
+}
+
 template <typename T>
 __global__ void ema_kernel(T* __restrict__ t, const T* __restrict__ s, long n, float tau) {
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
@@ -632,6 +678,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lambda_scan_fwd", &lambda_scan_fwd);
   m.def("lambda_scan_bwd", &lambda_scan_bwd);
   m.def("adam_step", &adam_step);
+  m.def("adam_step_dev", &adam_step_dev);
   m.def("ema_update", &ema_update);
   m.def("obs_norm", &obs_norm);
 }

@@ -60,18 +60,23 @@ class FusedAdam(torch.optim.Optimizer):
             bc1 = 1 - beta1**step
             bc2 = 1 - beta2**step
             if use_hip(params[0]):
-                require_ext().adam_step(
+                # device-side step counter: bias correction is computed in the
+                # kernel from a device scalar so the step is hipGraph-capturable
+                st0 = self.state[group["params"][0]]
+                if "step_t" not in st0 or st0["step_t"].device != params[0].device:
+                    st0["step_t"] = torch.zeros(1, dtype=torch.float32, device=params[0].device)
+                    st0["step_t"].fill_(float(step - 1))
+                require_ext().adam_step_dev(
                     params,
                     grads,
                     exp_avgs,
                     exp_avg_sqs,
+                    st0["step_t"],
                     group["lr"],
                     beta1,
                     beta2,
                     group["eps"],
                     group["weight_decay"],
-                    bc1,
-                    bc2,
                 )
             else:
                 for p, g, m, v in zip(params, grads, exp_avgs, exp_avg_sqs):
