@@ -9,4 +9,12 @@ hot ops, and an RCCL-over-xGMI runtime instead of Lightning Fabric.
 
 __version__ = "0.1.0"
 
+import torch.distributions as _td
+
+# Framework-wide default (parity: cfg.distribution.validate_args=False in the
+# reference).  Validation constructs do a host-side `_is_all_true` per
+# distribution — 232 stream syncs per DV3 train step, and they forbid hipGraph
+# capture.  Re-enable per-run with distribution.validate_args=True.
+_td.Distribution.set_default_validate_args(False)
+
 from sheeprl_amd.utils.dotdict import DotDict  # noqa: F401
