@@ -45,7 +45,7 @@ class Moments(nn.Module):
     ) -> None:
         super().__init__()
         self._decay = decay
-        self._max = torch.tensor(max_)
+        self._max = float(max_)
         self._percentile_low = percentile_low
         self._percentile_high = percentile_high
         self.register_buffer("low", torch.zeros((), dtype=torch.float32))
@@ -58,7 +58,7 @@ class Moments(nn.Module):
         # in-place EMA so the buffers keep fixed storage (hipGraph-replayable)
         self.low.mul_(self._decay).add_(low.to(self.low.device), alpha=1 - self._decay)
         self.high.mul_(self._decay).add_(high.to(self.high.device), alpha=1 - self._decay)
-        invscale = torch.max(1 / self._max.to(self.low.device), self.high - self.low)
+        invscale = torch.clamp(self.high - self.low, min=1.0 / self._max)
         return self.low.detach().clone(), invscale.detach()
 
 
