@@ -238,6 +238,36 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     for _ in range(warmup):
         obs = one_iter(obs, step_data)
 
+    if os.environ.get("SHEEPRL_AMD_PHASE_TIMING"):
+        # coarse phase split over a few synchronized steps
+        import numpy as _np
+
+        phases = {"env": 0.0, "sample": 0.0, "train": 0.0}
+        for _ in range(5):
+            torch.cuda.synchronize()
+            t = time.perf_counter()
+            with torch.inference_mode():
+                torch_obs = prepare_obs(runtime, obs, cnn_keys=list(cfg.algo.cnn_keys.encoder), num_envs=num_envs)
+                acts = player.get_actions(torch_obs)
+                real_actions = torch.stack([a.argmax(dim=-1) for a in acts], dim=-1).view(num_envs).cpu().numpy()
+            obs2, rewards, term, trunc, infos = envs.step(real_actions)
+            torch.cuda.synchronize()
+            phases["env"] += time.perf_counter() - t
+            t = time.perf_counter()
+            sample = rb.sample_tensors(cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device=device)
+            batch = {k: v[0] for k, v in sample.items()}
+            torch.cuda.synchronize()
+            phases["sample"] += time.perf_counter() - t
+            t = time.perf_counter()
+            if graphed is not None:
+                graphed(batch)
+            else:
+                train_fn(batch)
+            torch.cuda.synchronize()
+            phases["train"] += time.perf_counter() - t
+            obs = obs2
+        print({k: round(v / 5 * 1000, 2) for k, v in phases.items()}, file=sys.stderr)
+
     if os.environ.get("SHEEPRL_AMD_TORCH_PROFILE"):
         from torch.profiler import ProfilerActivity, profile
 
