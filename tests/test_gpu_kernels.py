@@ -245,8 +245,7 @@ def test_dv3_graphed_train_step():
     torch.cuda.synchronize()
     for n, p in world_model.named_parameters():
         assert torch.isfinite(p).all(), f"non-finite param {n} after graphed steps"
-    # step counter advanced on device across replays
+    # device step counter must advance on every replay (12 = 2 warmup + capture + ... + 10 replays)
     st = wo.state[wo.param_groups[0]["params"][0]]
-    assert float(st["step_t"].item()) == float(st["step"]) + 12 - st["step"], "device step did not advance"
-    assert float(st["step_t"].item()) >= 12
+    assert float(st["step_t"].item()) >= 12, "device Adam step did not advance across graph replays"
     envs.close()
