@@ -290,6 +290,15 @@ class RSSM(nn.Module):
         posterior_logits, posterior = self._representation(recurrent_state, embedded_obs)
         return recurrent_state, posterior, prior, posterior_logits, prior_logits
 
+    def _stoch_head(self, raw_logits: Tensor, sample: bool) -> Tuple[Tensor, Tensor]:
+        """Fused unimix + log-probs + one-hot-ST sampling (ops.categorical_st).
+
+        Returns (mixed log-prob logits [*, S*K] fp32, sample [*, S, K])."""
+        shape = raw_logits.shape
+        raw4 = raw_logits.view(*shape[:-1], -1, self.discrete)
+        m, onehot = ops.categorical_st(raw4, self.unimix, sample=sample)
+        return m.reshape(*shape), onehot.to(self._dtype)
+
     def _uniform_mix(self, logits: Tensor) -> Tensor:
         dim = logits.dim()
         if dim == 3:
@@ -301,14 +310,12 @@ class RSSM(nn.Module):
 
     def _representation(self, recurrent_state: Tensor, embedded_obs: Tensor) -> Tuple[Tensor, Tensor]:
         dt = self._dtype
-        logits = self.representation_model(torch.cat((recurrent_state.to(dt), embedded_obs.to(dt)), -1))
-        logits = self._uniform_mix(logits)
-        return logits, compute_stochastic_state(logits, discrete=self.discrete).to(dt)
+        raw = self.representation_model(torch.cat((recurrent_state.to(dt), embedded_obs.to(dt)), -1))
+        return self._stoch_head(raw, sample=True)
 
     def _transition(self, recurrent_out: Tensor, sample_state: bool = True) -> Tuple[Tensor, Tensor]:
-        logits = self.transition_model(recurrent_out.to(self._dtype))
-        logits = self._uniform_mix(logits)
-        return logits, compute_stochastic_state(logits, discrete=self.discrete, sample=sample_state).to(self._dtype)
+        raw = self.transition_model(recurrent_out.to(self._dtype))
+        return self._stoch_head(raw, sample=sample_state)
 
     def imagination(self, prior: Tensor, recurrent_state: Tensor, actions: Tensor) -> Tuple[Tensor, Tensor]:
         dt = self._dtype
@@ -337,10 +344,8 @@ class DecoupledRSSM(RSSM):
         return recurrent_state, None, prior_logits
 
     def _representation(self, embedded_obs: Tensor) -> Tuple[Tensor, Tensor]:  # type: ignore[override]
-        dt = self._dtype
-        logits = self.representation_model(embedded_obs.to(dt))
-        logits = self._uniform_mix(logits)
-        return logits, compute_stochastic_state(logits, discrete=self.discrete).to(dt)
+        raw = self.representation_model(embedded_obs.to(self._dtype))
+        return self._stoch_head(raw, sample=True)
 
 
 class WorldModel(nn.Module):
@@ -435,11 +440,15 @@ class Actor(nn.Module):
                 actions = actions * (clip / torch.maximum(clip, torch.abs(actions))).detach()
             return (actions,), (dist,)
         actions_list: List[Tensor] = []
-        dists: List[td.Distribution] = []
+        dists: List[Any] = []
         for logits in pre_dist:
-            dist = OneHotCategoricalST(logits=unimix_logits(logits.float(), self._unimix))
-            dists.append(dist)
-            actions_list.append(dist.rsample() if not greedy else dist.mode)
+            # fused unimix + sample + ST (ops.categorical_st); the dist wraps
+            # the normalized log-probs for log_prob/entropy in the actor loss
+            m, onehot = ops.categorical_st(logits, self._unimix, sample=not greedy)
+            from sheeprl_amd.distributions import LogProbCategorical
+
+            dists.append(LogProbCategorical(m))
+            actions_list.append(onehot)
         return tuple(actions_list), tuple(dists)
 
 

@@ -1,4 +1,5 @@
 from sheeprl_amd.distributions.dists import (
+    LogProbCategorical,
     BernoulliSafeMode,
     MSEDistribution,
     OneHotCategoricalST,
@@ -18,4 +19,5 @@ __all__ = [
     "TruncatedNormal",
     "TanhNormal",
     "unimix_logits",
+    "LogProbCategorical",
 ]

@@ -67,6 +67,16 @@ class MSEDistribution:
         return distance.sum(self._dims)
 
 
+_SUPPORT_CACHE: dict = {}
+
+
+def _support(low: float, high: float, k: int, device: torch.device) -> Tensor:
+    key = (low, high, k, str(device))
+    if key not in _SUPPORT_CACHE:
+        _SUPPORT_CACHE[key] = torch.linspace(low, high, k, device=device, dtype=torch.float32)
+    return _SUPPORT_CACHE[key]
+
+
 class TwoHotEncodingDistribution:
     """Categorical over a symlog-spaced support; log_prob is the two-hot
     cross-entropy of the symlog'd target (reference distribution.py:224-276).
@@ -81,7 +91,7 @@ class TwoHotEncodingDistribution:
     ) -> None:
         self.logits = logits
         self._dims = tuple(-i for i in range(1, dims + 1))
-        self.bins = torch.linspace(low, high, logits.shape[-1], device=logits.device, dtype=torch.float32)
+        self.bins = _support(low, high, logits.shape[-1], logits.device)
 
     @property
     def probs(self) -> Tensor:
@@ -136,6 +146,30 @@ class OneHotCategoricalST(td.OneHotCategorical):
         sample = self.sample(sample_shape)
         probs = self.probs
         return sample + (probs - probs.detach())
+
+
+class LogProbCategorical:
+    """Lightweight one-hot categorical over ALREADY-NORMALIZED log-probs
+    (the fused ``categorical_st`` head's output): log_prob/entropy without
+    re-normalization kernels."""
+
+    def __init__(self, log_probs: Tensor) -> None:
+        self.logits = log_probs
+
+    def log_prob(self, onehot: Tensor) -> Tensor:
+        return (self.logits * onehot.to(self.logits.dtype)).sum(-1)
+
+    def entropy(self) -> Tensor:
+        return -(self.logits.exp() * self.logits).sum(-1)
+
+    @property
+    def probs(self) -> Tensor:
+        return self.logits.exp()
+
+    @property
+    def mode(self) -> Tensor:
+        idx = self.logits.argmax(-1)
+        return torch.nn.functional.one_hot(idx, self.logits.shape[-1]).to(self.logits.dtype)
 
 
 class BernoulliSafeMode(td.Bernoulli):

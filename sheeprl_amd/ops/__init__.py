@@ -8,6 +8,7 @@ from sheeprl_amd.ops.functional import (
     two_hot_encoder,
     twohot_from_support,
 )
+from sheeprl_amd.ops.categorical import categorical_st
 from sheeprl_amd.ops.fused import ema_update_, gru_gates, layer_norm_act, normalize_obs
 
 __all__ = [
@@ -23,6 +24,7 @@ __all__ = [
     "gae",
     "lambda_values",
     "layer_norm_act",
+    "categorical_st",
     "gru_gates",
     "ema_update_",
     "normalize_obs",

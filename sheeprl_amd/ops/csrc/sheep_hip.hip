@@ -568,6 +568,143 @@ void adam_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> gra
   }
 }
 
+// ---------------------------------------------------------------------------
+// fused unimix-categorical straight-through head
+// rows of K logits: p = (1-u)*softmax(L) + u/K ; m = log p ;
+// sample = onehot(argmax(m + gumbel)) (or argmax(m) when !SAMPLE).
+// One wave per row (K <= 64 lanes, looped above that).
+// ---------------------------------------------------------------------------
+
+template <typename T, bool SAMPLE>
+__global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __restrict__ urand,
+                                  float* __restrict__ m_out, T* __restrict__ onehot, float* __restrict__ s_out,
+                                  long nrows, int K, float unimix) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= nrows) return;
+  const T* L = raw + row * (long)K;
+  float* mr = m_out + row * (long)K;
+  float* sr = s_out + row * (long)K;
+  T* oh = onehot + row * (long)K;
+  // row max
+  float lmax = -1e30f;
+  for (int j = lane; j < K; j += 64) lmax = fmaxf(lmax, ld(L, j));
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) lmax = fmaxf(lmax, __shfl_xor(lmax, off, 64));
+  // exp + sum
+  float lsum = 0.f;
+  for (int j = lane; j < K; j += 64) lsum += expf(ld(L, j) - lmax);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, 64);
+  const float inv = 1.f / lsum;
+  // p, m, gumbel-argmax
+  float best = -1e30f;
+  int best_j = 0;
+  for (int j = lane; j < K; j += 64) {
+    float s = expf(ld(L, j) - lmax) * inv;
+    float p = (1.f - unimix) * s + unimix / K;
+    float m = logf(p);
+    sr[j] = s;
+    mr[j] = m;
+    float score = m;
+    if (SAMPLE) {
+      float u = urand[row * (long)K + j];
+      float t = fmaxf(-logf(fmaxf(u, 1e-20f)), 1e-20f);
+      score += -logf(t);  // Gumbel(0,1) noise
+    }
+    if (score > best) {
+      best = score;
+      best_j = j;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float ob = __shfl_xor(best, off, 64);
+    int oj = __shfl_xor(best_j, off, 64);
+    if (ob > best || (ob == best && oj < best_j)) {
+      best = ob;
+      best_j = oj;
+    }
+  }
+  for (int j = lane; j < K; j += 64) st(oh, j, j == best_j ? 1.f : 0.f);
+}
+
+// backward: given gm (grad wrt m=log p) and gon (grad wrt the ST sample whose
+// gradient path is p), produce grad wrt raw logits.
+// g_L_j = (1-unimix) * s_j * [ (gm_j/p_j + gon_j) - sum_i (gm_i/p_i + gon_i) * s_i ]
+template <typename T>
+__global__ void cat_st_bwd_kernel(const float* __restrict__ gm, const T* __restrict__ gon,
+                                  const float* __restrict__ s_saved, T* __restrict__ graw, long nrows, int K,
+                                  float unimix) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= nrows) return;
+  const float* gmr = gm + row * (long)K;
+  const T* gor = gon + row * (long)K;
+  const float* sr = s_saved + row * (long)K;
+  T* gr = graw + row * (long)K;
+  float acc = 0.f;
+  for (int j = lane; j < K; j += 64) {
+    float sj = sr[j];
+    float pj = (1.f - unimix) * sj + unimix / K;
+    float t = gmr[j] / pj + ld(gor, j);
+    acc += t * sj;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_xor(acc, off, 64);
+  for (int j = lane; j < K; j += 64) {
+    float sj = sr[j];
+    float pj = (1.f - unimix) * sj + unimix / K;
+    float t = gmr[j] / pj + ld(gor, j);
+    st(gr, j, (1.f - unimix) * sj * (t - acc));
+  }
+}
+
+std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optional<torch::Tensor>& urand,
+                                      double unimix, bool sample) {
+  CHECK_IN(raw);
+  int K = (int)raw.size(-1);
+  long nrows = raw.numel() / K;
+  auto m = torch::empty(raw.sizes(), raw.options().dtype(at::kFloat));
+  auto s = torch::empty(raw.sizes(), raw.options().dtype(at::kFloat));
+  auto onehot = torch::empty_like(raw);
+  const int rows_per_block = kBlock / 64;
+  int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, raw.scalar_type(), "cat_st_fwd", [&] {
+    using T = scalar_t;
+    const float* up = urand.has_value() ? urand->data_ptr<float>() : nullptr;
+    if (sample)
+      hipLaunchKernelGGL((cat_st_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)raw.data_ptr(), up, m.data_ptr<float>(), (T*)onehot.data_ptr(),
+                         s.data_ptr<float>(), nrows, K, (float)unimix);
+    else
+      hipLaunchKernelGGL((cat_st_fwd_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)raw.data_ptr(), up, m.data_ptr<float>(), (T*)onehot.data_ptr(),
+                         s.data_ptr<float>(), nrows, K, (float)unimix);
+  });
+  return {m, onehot, s};
+}
+
+torch::Tensor cat_st_bwd(const torch::Tensor& gm, const torch::Tensor& gon, const torch::Tensor& s,
+                         double unimix) {
+  CHECK_IN(gm);
+  int K = (int)gm.size(-1);
+  long nrows = gm.numel() / K;
+  auto graw = torch::empty(gm.sizes(), gm.options().dtype(gon.scalar_type()));
+  const int rows_per_block = kBlock / 64;
+  int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto gonc = gon.contiguous();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, graw.scalar_type(), "cat_st_bwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((cat_st_bwd_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       gm.data_ptr<float>(), (const T*)gonc.data_ptr(), s.data_ptr<float>(), (T*)graw.data_ptr(),
+                       nrows, K, (float)unimix);
+  });
+  return graw;
+}
+
 __global__ void step_inc_kernel(float* step_t) {
   if (threadIdx.x == 0 && blockIdx.x == 0) step_t[0] += 1.0f;
 }
@@ -679,6 +816,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lambda_scan_bwd", &lambda_scan_bwd);
   m.def("adam_step", &adam_step);
   m.def("adam_step_dev", &adam_step_dev);
+  m.def("cat_st_fwd", &cat_st_fwd);
+  m.def("cat_st_bwd", &cat_st_bwd);
   m.def("ema_update", &ema_update);
   m.def("obs_norm", &obs_norm);
 }
