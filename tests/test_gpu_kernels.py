@@ -249,3 +249,36 @@ def test_dv3_graphed_train_step():
     st = wo.state[wo.param_groups[0]["params"][0]]
     assert float(st["step_t"].item()) >= 12, "device Adam step did not advance across graph replays"
     envs.close()
+
+
+@requires_gpu
+@pytest.mark.parametrize("K", [4, 9, 32, 255])
+def test_categorical_st_gpu(K):
+    torch.manual_seed(7)
+    raw = torch.randn(64, 8, K)
+    raw_c = raw.clone().requires_grad_()
+    raw_g = raw.cuda().requires_grad_()
+    m_c, _ = ops.categorical_st(raw_c, unimix=0.01, sample=False)
+    m_g, oh_g = ops.categorical_st(raw_g, unimix=0.01, sample=False)
+    assert torch.allclose(m_g.cpu(), m_c.detach(), atol=1e-4, rtol=1e-4)
+    # mode must match argmax of mixed log probs
+    assert (oh_g.argmax(-1).cpu() == m_c.argmax(-1)).all()
+    # backward: same upstream grads through m and the ST sample
+    gm = torch.randn_like(m_c)
+    gon = torch.randn(64, 8, K)
+    (m_c * gm + (torch.zeros_like(m_c) + 1) * 0).sum().backward(retain_graph=True)
+    raw_c.grad = None
+    torch.autograd.backward([m_c, _], [gm, gon])
+    torch.autograd.backward([m_g, oh_g], [gm.cuda(), gon.cuda()])
+    assert torch.allclose(raw_g.grad.cpu(), raw_c.grad, atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+def test_categorical_st_sampling_distribution():
+    """Gumbel-max sampling on GPU must follow the mixed categorical probs."""
+    torch.manual_seed(8)
+    logits = torch.tensor([[2.0, 0.0, -1.0, 0.5]]).repeat(20000, 1).cuda()
+    _, oh = ops.categorical_st(logits, unimix=0.01, sample=True)
+    freq = oh.float().mean(0).cpu()
+    p = (0.99 * torch.softmax(torch.tensor([2.0, 0.0, -1.0, 0.5]), -1) + 0.01 / 4)
+    assert torch.allclose(freq, p, atol=0.02), (freq, p)
