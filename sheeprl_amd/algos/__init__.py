@@ -3,3 +3,6 @@
 
 from sheeprl_amd.algos import ppo  # noqa: F401
 from sheeprl_amd.algos import dreamer_v3  # noqa: F401
+from sheeprl_amd.algos import sac  # noqa: F401
+from sheeprl_amd.algos import a2c  # noqa: F401
+from sheeprl_amd.algos import droq  # noqa: F401

@@ -185,3 +185,67 @@ def test_dreamer_v3_checkpoint_resume(tmp_path):
     ckpts = glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True)
     assert ckpts, "no checkpoint written"
     _run(tmp_path, args + [f"checkpoint.resume_from={ckpts[-1]}"])
+
+
+@pytest.mark.timeout(180)
+def test_sac(tmp_path, devices):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=sac",
+                "env=dummy",
+                "env.id=dummy_continuous",
+                "algo.total_steps=32",
+                "algo.learning_starts=8",
+                "algo.per_rank_batch_size=16",
+                "algo.replay_ratio=0.5",
+                "buffer.size=256",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+        devices,
+    )
+
+
+@pytest.mark.timeout(180)
+def test_a2c(tmp_path, devices):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=a2c",
+                "env=dummy",
+                "algo.total_steps=40",
+                "algo.rollout_steps=5",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+        devices,
+    )
+
+
+@pytest.mark.timeout(180)
+def test_droq(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=droq",
+                "env=dummy",
+                "env.id=dummy_continuous",
+                "algo.total_steps=32",
+                "algo.learning_starts=8",
+                "algo.per_rank_batch_size=16",
+                "algo.replay_ratio=1.0",
+                "buffer.size=256",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+    )
