@@ -1,1 +1,1 @@
-from sheeprl_amd.algos.ppo import evaluate, ppo  # noqa: F401  (registers ppo + eval)
+from sheeprl_amd.algos.ppo import evaluate, ppo, ppo_decoupled  # noqa: F401

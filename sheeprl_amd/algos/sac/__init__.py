@@ -1,1 +1,1 @@
-from sheeprl_amd.algos.sac import evaluate, sac  # noqa: F401
+from sheeprl_amd.algos.sac import evaluate, sac, sac_decoupled  # noqa: F401

@@ -235,6 +235,8 @@ class Runtime:
     def all_reduce(self, value: torch.Tensor, op: str = "mean", group: Any = None) -> torch.Tensor:
         if not self.is_distributed:
             return value
+        if group is None:
+            group = getattr(self, "_default_group", None)
         t = value.detach().clone() if isinstance(value, torch.Tensor) else torch.tensor(value, device=self._device)
         dist.all_reduce(t, op=dist.ReduceOp.SUM, group=group)
         if op == "mean":
@@ -245,6 +247,8 @@ class Runtime:
         """All-gather tensors (or dicts of tensors); returns stacked [world, ...]."""
         if not self.is_distributed:
             return data
+        if group is None:
+            group = getattr(self, "_default_group", None)
         if isinstance(data, dict):
             return {k: self.all_gather(v, group) for k, v in data.items()}
         t = data if isinstance(data, torch.Tensor) else torch.as_tensor(data, device=self._device)

@@ -249,3 +249,56 @@ def test_droq(tmp_path):
             ],
         ),
     )
+
+
+@pytest.mark.timeout(300)
+def test_ppo_decoupled(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo_decoupled",
+                "env=dummy",
+                "algo.total_steps=64",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=8",
+                "algo.run_test=False",
+                "dry_run=False",
+            ],
+        ),
+        devices="2",
+    )
+
+
+def test_decoupled_requires_multi_device(tmp_path):
+    with pytest.raises(Exception):
+        _run(
+            tmp_path,
+            standard_args(tmp_path, ["exp=ppo_decoupled", "env=dummy", "algo.run_test=False"]),
+            devices="1",
+        )
+
+
+@pytest.mark.timeout(300)
+def test_sac_decoupled(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=sac_decoupled",
+                "env=dummy",
+                "env.id=dummy_continuous",
+                "algo.total_steps=32",
+                "algo.learning_starts=8",
+                "algo.per_rank_batch_size=8",
+                "algo.replay_ratio=0.5",
+                "buffer.size=256",
+                "algo.run_test=False",
+                "dry_run=False",
+            ],
+        ),
+        devices="2",
+    )
