@@ -6,3 +6,5 @@ from sheeprl_amd.algos import dreamer_v3  # noqa: F401
 from sheeprl_amd.algos import sac  # noqa: F401
 from sheeprl_amd.algos import a2c  # noqa: F401
 from sheeprl_amd.algos import droq  # noqa: F401
+from sheeprl_amd.algos import ppo_recurrent  # noqa: F401
+from sheeprl_amd.algos import sac_ae  # noqa: F401

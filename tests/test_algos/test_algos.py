@@ -302,3 +302,47 @@ def test_sac_decoupled(tmp_path):
         ),
         devices="2",
     )
+
+
+@pytest.mark.timeout(300)
+def test_ppo_recurrent(tmp_path, devices):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo_recurrent",
+                "env=dummy",
+                "algo.total_steps=64",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=2",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+        devices,
+    )
+
+
+@pytest.mark.timeout(300)
+def test_sac_ae(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=sac_ae",
+                "env.id=dummy_continuous",
+                "algo.total_steps=24",
+                "algo.learning_starts=8",
+                "algo.per_rank_batch_size=8",
+                "algo.replay_ratio=0.5",
+                "algo.cnn_keys.encoder=[rgb]",
+                "algo.mlp_keys.encoder=[state]",
+                "buffer.size=128",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+    )
