@@ -8,3 +8,5 @@ from sheeprl_amd.algos import a2c  # noqa: F401
 from sheeprl_amd.algos import droq  # noqa: F401
 from sheeprl_amd.algos import ppo_recurrent  # noqa: F401
 from sheeprl_amd.algos import sac_ae  # noqa: F401
+from sheeprl_amd.algos import dreamer_v2  # noqa: F401
+from sheeprl_amd.algos import dreamer_v1  # noqa: F401

@@ -101,6 +101,8 @@ class CNNEncoder(nn.Module):
         channels_multiplier: int,
         layer_norm_eps: float = 1e-3,
         stages: int = 4,
+        activation: str = "silu",
+        layer_norm: bool = True,
     ) -> None:
         super().__init__()
         self.keys = list(keys)
@@ -112,8 +114,8 @@ class CNNEncoder(nn.Module):
                 kernel_sizes=[4] * stages,
                 strides=[2] * stages,
                 paddings=[1] * stages,
-                activation="silu",
-                layer_norm=True,
+                activation=activation,
+                layer_norm=layer_norm,
                 layer_norm_eps=layer_norm_eps,
             ),
             nn.Flatten(-3, -1),
@@ -135,6 +137,8 @@ class MLPEncoder(nn.Module):
         dense_units: int = 512,
         layer_norm_eps: float = 1e-3,
         symlog_inputs: bool = True,
+        activation: str = "silu",
+        layer_norm: bool = True,
     ) -> None:
         super().__init__()
         self.keys = list(keys)
@@ -143,8 +147,8 @@ class MLPEncoder(nn.Module):
             self.input_dim,
             None,
             [dense_units] * mlp_layers,
-            activation="silu",
-            layer_norm=True,
+            activation=activation,
+            layer_norm=layer_norm,
             layer_norm_eps=layer_norm_eps,
         )
         self.output_dim = dense_units
@@ -166,6 +170,8 @@ class CNNDecoder(nn.Module):
         image_size: Tuple[int, int],
         layer_norm_eps: float = 1e-3,
         stages: int = 4,
+        activation: str = "silu",
+        layer_norm: bool = True,
     ) -> None:
         super().__init__()
         self.keys = list(keys)
@@ -181,8 +187,8 @@ class CNNDecoder(nn.Module):
                 kernel_sizes=[4] * stages,
                 strides=[2] * stages,
                 paddings=[1] * stages,
-                activation="silu",
-                layer_norm=True,
+                activation=activation,
+                layer_norm=layer_norm,
                 layer_norm_eps=layer_norm_eps,
                 last_layer_plain=True,
             ),
@@ -202,6 +208,8 @@ class MLPDecoder(nn.Module):
         mlp_layers: int = 4,
         dense_units: int = 512,
         layer_norm_eps: float = 1e-3,
+        activation: str = "silu",
+        layer_norm: bool = True,
     ) -> None:
         super().__init__()
         self.keys = list(keys)
@@ -209,8 +217,8 @@ class MLPDecoder(nn.Module):
             latent_state_size,
             None,
             [dense_units] * mlp_layers,
-            activation="silu",
-            layer_norm=True,
+            activation=activation,
+            layer_norm=layer_norm,
             layer_norm_eps=layer_norm_eps,
         )
         self.heads = nn.ModuleList([nn.Linear(dense_units, dim) for dim in output_dims])
@@ -231,10 +239,13 @@ class RecurrentModel(nn.Module):
         recurrent_state_size: int,
         dense_units: int,
         layer_norm_eps: float = 1e-3,
+        activation: str = "silu",
+        layer_norm: bool = True,
     ) -> None:
         super().__init__()
         self.mlp = DenseBlock(
-            input_size, dense_units, bias=False, layer_norm=True, layer_norm_eps=layer_norm_eps, activation="silu"
+            input_size, dense_units, bias=not layer_norm, layer_norm=layer_norm,
+            layer_norm_eps=layer_norm_eps, activation=activation
         )
         self.rnn = LayerNormGRUCell(dense_units, recurrent_state_size, bias=False, layer_norm=True,
                                     layer_norm_eps=layer_norm_eps)
@@ -384,6 +395,8 @@ class Actor(nn.Module):
         layer_norm_eps: float = 1e-3,
         unimix: float = 0.01,
         action_clip: float = 1.0,
+        activation: str = "silu",
+        layer_norm: bool = True,
     ) -> None:
         super().__init__()
         self.distribution = distribution.lower()
@@ -395,8 +408,8 @@ class Actor(nn.Module):
             latent_state_size,
             None,
             [dense_units] * mlp_layers,
-            activation="silu",
-            layer_norm=True,
+            activation=activation,
+            layer_norm=layer_norm,
             layer_norm_eps=layer_norm_eps,
         )
         if is_continuous:

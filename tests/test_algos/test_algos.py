@@ -346,3 +346,76 @@ def test_sac_ae(tmp_path):
             ],
         ),
     )
+
+
+@pytest.mark.timeout(300)
+def test_dreamer_v2(tmp_path, devices):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=dreamer_v2",
+                "env=dummy",
+                "algo.dense_units=8",
+                "algo.mlp_layers=1",
+                "algo.world_model.encoder.cnn_channels_multiplier=2",
+                "algo.world_model.recurrent_model.recurrent_state_size=8",
+                "algo.world_model.transition_model.hidden_size=8",
+                "algo.world_model.representation_model.hidden_size=8",
+                "algo.world_model.discrete_size=4",
+                "algo.world_model.stochastic_size=4",
+                "algo.per_rank_batch_size=2",
+                "algo.per_rank_sequence_length=4",
+                "algo.per_rank_pretrain_steps=1",
+                "algo.horizon=3",
+                "algo.mlp_keys.encoder=[state]",
+                "algo.total_steps=16",
+                "algo.learning_starts=4",
+                "algo.replay_ratio=0.5",
+                "buffer.size=64",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+        devices,
+    )
+
+
+@pytest.mark.timeout(300)
+def test_dreamer_v1(tmp_path, devices):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=dreamer_v1",
+                "env=dummy",
+                "algo.world_model.stochastic_size=4",
+                "algo.world_model.encoder.cnn_channels_multiplier=2",
+                "algo.world_model.encoder.dense_units=8",
+                "algo.world_model.recurrent_model.recurrent_state_size=8",
+                "algo.world_model.recurrent_model.dense_units=8",
+                "algo.world_model.transition_model.hidden_size=8",
+                "algo.world_model.representation_model.hidden_size=8",
+                "algo.world_model.observation_model.dense_units=8",
+                "algo.world_model.reward_model.dense_units=8",
+                "algo.actor.dense_units=8",
+                "algo.actor.mlp_layers=1",
+                "algo.critic.dense_units=8",
+                "algo.critic.mlp_layers=1",
+                "algo.per_rank_batch_size=2",
+                "algo.per_rank_sequence_length=4",
+                "algo.per_rank_pretrain_steps=1",
+                "algo.horizon=3",
+                "algo.mlp_keys.encoder=[state]",
+                "algo.total_steps=16",
+                "algo.learning_starts=4",
+                "algo.replay_ratio=0.5",
+                "buffer.size=64",
+                "algo.run_test=True",
+                "dry_run=False",
+            ],
+        ),
+        devices,
+    )
