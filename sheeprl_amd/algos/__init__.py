@@ -10,3 +10,6 @@ from sheeprl_amd.algos import ppo_recurrent  # noqa: F401
 from sheeprl_amd.algos import sac_ae  # noqa: F401
 from sheeprl_amd.algos import dreamer_v2  # noqa: F401
 from sheeprl_amd.algos import dreamer_v1  # noqa: F401
+from sheeprl_amd.algos import p2e_dv3  # noqa: F401
+from sheeprl_amd.algos import p2e_dv2  # noqa: F401
+from sheeprl_amd.algos import p2e_dv1  # noqa: F401

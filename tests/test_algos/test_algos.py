@@ -419,3 +419,125 @@ def test_dreamer_v1(tmp_path, devices):
         ),
         devices,
     )
+
+
+P2E_TINY = [
+    "algo.dense_units=8",
+    "algo.mlp_layers=1",
+    "algo.world_model.encoder.cnn_channels_multiplier=2",
+    "algo.world_model.recurrent_model.recurrent_state_size=8",
+    "algo.world_model.transition_model.hidden_size=8",
+    "algo.world_model.representation_model.hidden_size=8",
+    "algo.world_model.discrete_size=4",
+    "algo.world_model.stochastic_size=4",
+    "algo.ensembles.n=2",
+    "algo.per_rank_batch_size=2",
+    "algo.per_rank_sequence_length=4",
+    "algo.horizon=3",
+    "algo.mlp_keys.encoder=[state]",
+    "algo.total_steps=16",
+    "algo.learning_starts=4",
+    "algo.replay_ratio=0.5",
+    "buffer.size=64",
+    "dry_run=False",
+]
+
+
+@pytest.mark.timeout(300)
+def test_p2e_dv3_exploration_then_finetuning(tmp_path):
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            ["exp=p2e_dv3_exploration", "env=dummy", "algo.run_test=True",
+             "checkpoint.save_last=True", *P2E_TINY],
+        ),
+    )
+    import glob
+
+    ckpts = sorted(glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True))
+    assert ckpts, "exploration produced no checkpoint"
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            ["exp=p2e_dv3_finetuning", "env=dummy", "algo.run_test=False",
+             f"checkpoint.exploration_ckpt_path={ckpts[-1]}", *P2E_TINY],
+        ),
+    )
+
+
+@pytest.mark.timeout(300)
+def test_p2e_dv2_exploration_then_finetuning(tmp_path):
+    args = [
+        "env=dummy",
+        "algo.dense_units=8",
+        "algo.mlp_layers=1",
+        "algo.world_model.encoder.cnn_channels_multiplier=2",
+        "algo.world_model.recurrent_model.recurrent_state_size=8",
+        "algo.world_model.transition_model.hidden_size=8",
+        "algo.world_model.representation_model.hidden_size=8",
+        "algo.world_model.discrete_size=4",
+        "algo.world_model.stochastic_size=4",
+        "algo.ensembles.n=2",
+        "algo.per_rank_batch_size=2",
+        "algo.per_rank_sequence_length=4",
+        "algo.per_rank_pretrain_steps=1",
+        "algo.horizon=3",
+        "algo.mlp_keys.encoder=[state]",
+        "algo.total_steps=16",
+        "algo.learning_starts=4",
+        "algo.replay_ratio=0.5",
+        "buffer.size=64",
+        "dry_run=False",
+    ]
+    _run(tmp_path, standard_args(tmp_path, ["exp=p2e_dv2_exploration", "algo.run_test=False",
+                                            "checkpoint.save_last=True", *args]))
+    import glob
+
+    ckpts = sorted(glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True))
+    assert ckpts
+    _run(tmp_path, standard_args(tmp_path, ["exp=p2e_dv2_finetuning", "algo.run_test=False",
+                                            f"checkpoint.exploration_ckpt_path={ckpts[-1]}", *args]))
+
+
+@pytest.mark.timeout(300)
+def test_p2e_dv1_exploration_then_finetuning(tmp_path):
+    args = [
+        "env=dummy",
+        "env.id=dummy_continuous",
+        "algo.world_model.stochastic_size=4",
+        "algo.world_model.encoder.cnn_channels_multiplier=2",
+        "algo.world_model.encoder.dense_units=8",
+        "algo.world_model.recurrent_model.recurrent_state_size=8",
+        "algo.world_model.recurrent_model.dense_units=8",
+        "algo.world_model.transition_model.hidden_size=8",
+        "algo.world_model.representation_model.hidden_size=8",
+        "algo.world_model.observation_model.dense_units=8",
+        "algo.world_model.reward_model.dense_units=8",
+        "algo.actor.dense_units=8",
+        "algo.actor.mlp_layers=1",
+        "algo.critic.dense_units=8",
+        "algo.critic.mlp_layers=1",
+        "algo.ensembles.n=2",
+        "algo.ensembles.dense_units=8",
+        "algo.ensembles.mlp_layers=1",
+        "algo.per_rank_batch_size=2",
+        "algo.per_rank_sequence_length=4",
+        "algo.per_rank_pretrain_steps=1",
+        "algo.horizon=3",
+        "algo.mlp_keys.encoder=[state]",
+        "algo.total_steps=16",
+        "algo.learning_starts=4",
+        "algo.replay_ratio=0.5",
+        "buffer.size=64",
+        "dry_run=False",
+    ]
+    _run(tmp_path, standard_args(tmp_path, ["exp=p2e_dv1_exploration", "algo.run_test=False",
+                                            "checkpoint.save_last=True", *args]))
+    import glob
+
+    ckpts = sorted(glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True))
+    assert ckpts
+    _run(tmp_path, standard_args(tmp_path, ["exp=p2e_dv1_finetuning", "algo.run_test=False",
+                                            f"checkpoint.exploration_ckpt_path={ckpts[-1]}", *args]))
