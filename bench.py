@@ -313,7 +313,9 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         "dtype": "bf16" if device_type == "cuda" else "fp32",
         "data": "synthetic",
         "config": {
-            "model": "dreamer_v3_S",
+            "model": {256: "dreamer_v3_XS", 512: "dreamer_v3_S", 1024: "dreamer_v3_M",
+                      2048: "dreamer_v3_L", 4096: "dreamer_v3_XL"}.get(
+                cfg.algo.world_model.recurrent_model.recurrent_state_size, "dreamer_v3_custom"),
             "global_batch": cfg.algo.per_rank_batch_size * world_size,
             "seq_len": seq_len,
             "parallelism": f"dp{world_size}",

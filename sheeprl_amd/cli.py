@@ -185,6 +185,30 @@ def evaluation(args: Optional[List[str]] = None) -> None:
     eval_algorithm(cfg)
 
 
+def registration(args: Optional[List[str]] = None) -> None:
+    """Register a checkpoint's models into the local model registry
+    (parity: cli.py:408-450; local backend — MLflow is not in this image)."""
+    argv = list(sys.argv[1:] if args is None else args)
+    kv = dict(a.split("=", 1) for a in argv if "=" in a)
+    ckpt = kv.pop("checkpoint_path", None)
+    if ckpt is None:
+        raise ValueError("registration requires checkpoint_path=<path to .ckpt>")
+    ckpt_path = pathlib.Path(ckpt)
+    run_cfg_path = ckpt_path.parent.parent / "config.yaml"
+    if not run_cfg_path.exists():
+        raise RuntimeError(f"no config.yaml next to checkpoint: {run_cfg_path}")
+    import yaml
+
+    from sheeprl_amd.utils.model_manager import register_models_from_checkpoint
+
+    with open(run_cfg_path) as f:
+        cfg = DotDict(yaml.safe_load(f))
+    registry_dir = kv.get("registry_dir", "models_registry")
+    versions = register_models_from_checkpoint(cfg, str(ckpt_path), registry_dir)
+    for name, v in versions.items():
+        print(f"registered {name} -> v{v}")
+
+
 def available_agents() -> None:
     """Print the registered algorithms (parity: available_agents.py:7)."""
     _import_algorithms()
@@ -209,6 +233,8 @@ def main() -> None:
     argv = sys.argv[1:]
     if argv and argv[0] == "eval":
         evaluation(argv[1:])
+    elif argv and argv[0] == "register":
+        registration(argv[1:])
     elif argv and argv[0] == "agents":
         available_agents()
     else:

@@ -541,3 +541,40 @@ def test_p2e_dv1_exploration_then_finetuning(tmp_path):
     assert ckpts
     _run(tmp_path, standard_args(tmp_path, ["exp=p2e_dv1_finetuning", "algo.run_test=False",
                                             f"checkpoint.exploration_ckpt_path={ckpts[-1]}", *args]))
+
+
+@pytest.mark.timeout(300)
+def test_eval_and_registration_cli(tmp_path):
+    """Train tiny PPO, then exercise `eval` and `register` CLI paths
+    (parity: tests/test_algos/test_cli.py eval smoke + registration)."""
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo",
+                "algo.total_steps=32",
+                "algo.rollout_steps=8",
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=8",
+                "algo.run_test=False",
+                "checkpoint.save_last=True",
+                "dry_run=False",
+            ],
+        ),
+    )
+    import glob
+    import os
+
+    from sheeprl_amd.cli import evaluation, registration
+
+    ckpts = sorted(glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True))
+    assert ckpts
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        evaluation([f"checkpoint_path={ckpts[-1]}", "runtime.accelerator=cpu"])
+        registration([f"checkpoint_path={ckpts[-1]}", f"registry_dir={tmp_path}/registry"])
+        assert (tmp_path / "registry").exists()
+    finally:
+        os.chdir(cwd)
