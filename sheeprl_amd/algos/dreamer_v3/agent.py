@@ -301,6 +301,46 @@ class RSSM(nn.Module):
         posterior_logits, posterior = self._representation(recurrent_state, embedded_obs)
         return recurrent_state, posterior, prior, posterior_logits, prior_logits
 
+    def dynamic_posterior(
+        self,
+        posterior: Tensor,
+        recurrent_state: Tensor,
+        action: Tensor,
+        embedded_obs: Tensor,
+        is_first: Tensor,
+        initial_states: Optional[Tuple[Tensor, Tensor]] = None,
+    ) -> Tuple[Tensor, Tensor, Tensor]:
+        """Scan step WITHOUT the transition head.
+
+        In training the prior is only consumed by the KL loss (the recurrence
+        feeds back the posterior), so the transition MLP runs once batched
+        over all T afterwards (``transition_logits``) instead of per step.
+        ``initial_states`` lets the caller hoist the (parameter-dependent but
+        step-invariant) initial-state computation out of the scan — together
+        this removes two MLP+categorical-head evaluations per scan step.
+        """
+        dt = self._dtype
+        is_first = is_first.to(dt)
+        action = (1 - is_first) * action.to(dt)
+        if initial_states is None:
+            initial_recurrent_state, initial_posterior = self.get_initial_states(recurrent_state.shape[:2])
+        else:
+            initial_recurrent_state, initial_posterior = initial_states
+        recurrent_state = (1 - is_first) * recurrent_state.to(dt) + is_first * initial_recurrent_state.to(dt)
+        posterior = posterior.view(*posterior.shape[:-2], -1).to(dt)
+        posterior = (1 - is_first) * posterior + is_first * initial_posterior.reshape(
+            *posterior.shape[:-1], -1
+        ).to(dt).expand_as(posterior)
+        recurrent_state = self.recurrent_model(torch.cat((posterior, action), -1), recurrent_state)
+        posterior_logits, posterior = self._representation(recurrent_state, embedded_obs)
+        return recurrent_state, posterior, posterior_logits
+
+    def transition_logits(self, recurrent_states: Tensor) -> Tensor:
+        """Batched prior logits over a whole [T, B, H] stack."""
+        raw = self.transition_model(recurrent_states.to(self._dtype))
+        m, _ = self._stoch_head(raw, sample=False)
+        return m
+
     def _stoch_head(self, raw_logits: Tensor, sample: bool) -> Tuple[Tensor, Tensor]:
         """Fused unimix + log-probs + one-hot-ST sampling (ops.categorical_st).
 

@@ -108,18 +108,22 @@ def train(
             sequence_length, batch_size, stochastic_size, discrete_size, device=device, dtype=dtype
         )
         posteriors_logits = torch.empty(sequence_length, batch_size, stoch_state_size, device=device)
+        # initial states are step-invariant: hoist them out of the scan; the
+        # prior (transition) head batches over all T afterwards
+        initial_states = world_model.rssm.get_initial_states((1, batch_size))
         for i in range(sequence_length):
-            recurrent_state, posterior, _, posterior_logits, prior_logits = world_model.rssm.dynamic(
+            recurrent_state, posterior, posterior_logits = world_model.rssm.dynamic_posterior(
                 posterior,
                 recurrent_state,
                 batch_actions[i : i + 1],
                 embedded_obs[i : i + 1],
                 data["is_first"][i : i + 1],
+                initial_states=initial_states,
             )
             recurrent_states[i] = recurrent_state
-            priors_logits[i] = prior_logits
             posteriors[i] = posterior
             posteriors_logits[i] = posterior_logits
+        priors_logits = world_model.rssm.transition_logits(recurrent_states)
     latent_states = torch.cat((posteriors.view(*posteriors.shape[:-2], -1), recurrent_states), -1)
 
     reconstructed_obs = world_model.observation_model(latent_states)
