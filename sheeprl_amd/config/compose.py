@@ -142,8 +142,11 @@ def compose(overrides: Sequence[str] = (), root: str = "config.yaml") -> DotDict
         else:
             dotted.append((k, _value_from_str(v)))
 
-    # Experiment selection can override other groups.
+    # Experiment selection can override other groups.  Walk the exp
+    # inheritance chain child -> parent, then apply group selections
+    # PARENT-FIRST so a child's `override /algo: ...` beats its base exp's.
     exp_chain: List[dict] = []
+    exp_defaults_chain: List[list] = []
     exp_name = group_sel.get("exp")
     if exp_name not in (None, "???"):
         seen: set = set()
@@ -154,10 +157,18 @@ def compose(overrides: Sequence[str] = (), root: str = "config.yaml") -> DotDict
             seen.add(name)
             raw = load_yaml(_find(f"exp/{name}.yaml"))
             defaults = raw.pop("defaults", [])
-            base = _parse_defaults(defaults, group_sel, None, "exp")
+            exp_defaults_chain.append(defaults)
             exp_chain.append(raw)
+            # find the same-group base without touching group_sel yet
+            base = None
+            for entry in defaults:
+                if isinstance(entry, str) and entry != "_self_":
+                    base = entry
             name = base
         exp_chain.reverse()
+        exp_defaults_chain.reverse()
+        for defaults in exp_defaults_chain:  # base first, child last (wins)
+            _parse_defaults(defaults, group_sel, None, "exp")
 
     # Re-apply CLI group selections (they beat the exp's defaults).
     for ov in overrides:

@@ -37,3 +37,13 @@ def test_exp_inherits_group_chain():
     assert cfg.algo.total_steps == 65536
     assert cfg.env.num_envs == 1
     assert cfg.runtime.accelerator == "cpu"
+
+
+def test_exp_child_group_override_beats_base():
+    """A child exp's `override /algo:` must beat its base exp's selection
+    (regression: the bench silently ran the XL model instead of S)."""
+    from sheeprl_amd.config import compose
+
+    cfg = compose(["exp=dreamer_v3_100k_ms_pacman"])
+    assert cfg.algo.world_model.recurrent_model.recurrent_state_size == 512
+    assert cfg.algo.dense_units == 512
