@@ -280,6 +280,37 @@ def train(
     world_optimizer.zero_grad(set_to_none=True)
 
 
+def _capture_train_step(
+    runtime, world_model, actor, critic, target_critic,
+    world_optimizer, actor_optimizer, critic_optimizer,
+    example_batch, cfg, is_continuous, actions_dim, moments,
+):
+    """hipGraph-capture the gradient step; returns a replayable callable or
+    None when capture fails (falls back to eager)."""
+    from sheeprl_amd.parallel.graphs import CUDAGraphStep
+    from sheeprl_amd.utils.metric import MetricAggregator
+
+    def train_fn(batch):
+        was_disabled = MetricAggregator.disabled
+        MetricAggregator.disabled = True
+        try:
+            train(
+                runtime, world_model, actor, critic, target_critic,
+                world_optimizer, actor_optimizer, critic_optimizer,
+                batch, None, cfg, is_continuous, actions_dim, moments,
+            )
+        finally:
+            MetricAggregator.disabled = was_disabled
+
+    try:
+        step = CUDAGraphStep(train_fn, example_batch, warmup=2)
+        runtime.print("[dreamer_v3] gradient step captured in a hipGraph")
+        return step
+    except Exception as e:  # noqa: BLE001
+        runtime.print(f"[dreamer_v3] hipGraph capture failed ({e}); eager training")
+        return None
+
+
 @register_algorithm(name="dreamer_v3")
 def main(runtime: Runtime, cfg: Any) -> None:
     device = runtime.device
@@ -396,6 +427,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
 
     clip_rewards_fn = (lambda r: np.tanh(r)) if cfg.env.clip_rewards else (lambda r: r)
 
+    want_graphs = (
+        bool(cfg.algo.get("hip_graphs", True))
+        and device.type == "cuda"
+        and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
+    )
+    graphed_step = None
+
     # initial step data
     step_data: Dict[str, np.ndarray] = {}
     obs, _ = envs.reset(seed=cfg.seed + runtime.global_rank * num_envs)
@@ -492,7 +530,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
                 step_data["is_first"][:, dones_idxes] = 1.0
                 player.init_states(dones_idxes)
 
-        # training phase, replay-ratio driven
+        # training phase, replay-ratio driven.  On CUDA the whole gradient
+        # step is hipGraph-captured after warmup (algo.hip_graphs, default
+        # on): ~20k kernel launches collapse into one replay.  Metrics can't
+        # cross the capture (host syncs), so every 16th gradient step runs
+        # eager to feed the aggregator.
         if isinstance(rb, EnvIndependentReplayBuffer):
             rb_ready = any(len(b) >= cfg.algo.per_rank_sequence_length for b in rb.buffer)
         else:
@@ -524,23 +566,40 @@ def main(runtime: Runtime, cfg: Any) -> None:
                             from_numpy=cfg.buffer.from_numpy,
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
-                        train(
-                            runtime,
-                            world_model,
-                            actor,
-                            critic,
-                            target_critic,
-                            world_optimizer,
-                            actor_optimizer,
-                            critic_optimizer,
-                            batch,
-                            aggregator,
-                            cfg,
-                            is_continuous,
-                            actions_dim,
-                            moments,
-                        )
+                        use_eager = graphed_step is None or (cumulative_per_rank_gradient_steps % 16 == 0)
+                        if use_eager:
+                            train(
+                                runtime,
+                                world_model,
+                                actor,
+                                critic,
+                                target_critic,
+                                world_optimizer,
+                                actor_optimizer,
+                                critic_optimizer,
+                                batch,
+                                aggregator,
+                                cfg,
+                                is_continuous,
+                                actions_dim,
+                                moments,
+                            )
+                        else:
+                            graphed_step(batch)
                         cumulative_per_rank_gradient_steps += 1
+                        # capture once the shapes/allocator have settled
+                        if (
+                            graphed_step is None
+                            and want_graphs
+                            and cumulative_per_rank_gradient_steps >= 3
+                        ):
+                            graphed_step = _capture_train_step(
+                                runtime, world_model, actor, critic, target_critic,
+                                world_optimizer, actor_optimizer, critic_optimizer,
+                                batch, cfg, is_continuous, actions_dim, moments,
+                            )
+                            if graphed_step is None:
+                                want_graphs = False
 
         # logging
         if policy_step - last_log >= cfg.metric.log_every or iter_num == total_iters or cfg.dry_run:
