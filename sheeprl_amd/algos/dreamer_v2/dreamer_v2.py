@@ -122,7 +122,8 @@ def train(
     posteriors = torch.empty(sequence_length, batch_size, stochastic_size, discrete_size, device=device, dtype=dtype)
     posteriors_logits = torch.empty(sequence_length, batch_size, stoch_state_size, device=device)
     embedded_obs = world_model.encoder(batch_obs)
-    initial_states = world_model.rssm.get_initial_states((1, batch_size))
+    _ir, _ip = world_model.rssm.get_initial_states((1, batch_size))
+    initial_states = (_ir.contiguous(), _ip.contiguous())
     for i in range(sequence_length):
         recurrent_state, posterior, posterior_logits = world_model.rssm.dynamic_posterior(
             posterior, recurrent_state, batch_actions[i : i + 1], embedded_obs[i : i + 1],

@@ -321,16 +321,17 @@ class RSSM(nn.Module):
         """
         dt = self._dtype
         is_first = is_first.to(dt)
-        action = (1 - is_first) * action.to(dt)
         if initial_states is None:
             initial_recurrent_state, initial_posterior = self.get_initial_states(recurrent_state.shape[:2])
         else:
             initial_recurrent_state, initial_posterior = initial_states
-        recurrent_state = (1 - is_first) * recurrent_state.to(dt) + is_first * initial_recurrent_state.to(dt)
+        # fused per-row reset masking (one kernel per tensor instead of ~4)
+        action = ops.masked_lerp(action.to(dt), None, is_first)
+        recurrent_state = ops.masked_lerp(recurrent_state.to(dt), initial_recurrent_state.to(dt), is_first)
         posterior = posterior.view(*posterior.shape[:-2], -1).to(dt)
-        posterior = (1 - is_first) * posterior + is_first * initial_posterior.reshape(
-            *posterior.shape[:-1], -1
-        ).to(dt).expand_as(posterior)
+        posterior = ops.masked_lerp(
+            posterior, initial_posterior.reshape(*posterior.shape[:-1], -1).to(dt), is_first
+        )
         recurrent_state = self.recurrent_model(torch.cat((posterior, action), -1), recurrent_state)
         posterior_logits, posterior = self._representation(recurrent_state, embedded_obs)
         return recurrent_state, posterior, posterior_logits

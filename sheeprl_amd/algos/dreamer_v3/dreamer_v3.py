@@ -110,7 +110,8 @@ def train(
         posteriors_logits = torch.empty(sequence_length, batch_size, stoch_state_size, device=device)
         # initial states are step-invariant: hoist them out of the scan; the
         # prior (transition) head batches over all T afterwards
-        initial_states = world_model.rssm.get_initial_states((1, batch_size))
+        _ir, _ip = world_model.rssm.get_initial_states((1, batch_size))
+        initial_states = (_ir.contiguous(), _ip.contiguous())
         for i in range(sequence_length):
             recurrent_state, posterior, posterior_logits = world_model.rssm.dynamic_posterior(
                 posterior,

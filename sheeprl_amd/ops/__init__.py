@@ -9,7 +9,7 @@ from sheeprl_amd.ops.functional import (
     twohot_from_support,
 )
 from sheeprl_amd.ops.categorical import categorical_st
-from sheeprl_amd.ops.fused import ema_update_, gru_gates, layer_norm_act, normalize_obs
+from sheeprl_amd.ops.fused import ema_update_, gru_gates, layer_norm_act, masked_lerp, normalize_obs
 
 __all__ = [
     "get_ext",
@@ -28,4 +28,5 @@ __all__ = [
     "gru_gates",
     "ema_update_",
     "normalize_obs",
+    "masked_lerp",
 ]

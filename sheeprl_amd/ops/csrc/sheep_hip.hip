@@ -705,6 +705,80 @@ torch::Tensor cat_st_bwd(const torch::Tensor& gm, const torch::Tensor& gon, cons
   return graw;
 }
 
+// ---------------------------------------------------------------------------
+// masked lerp (episode-reset masking in the RSSM scan):
+//   y[b, j] = (1 - f[b]) * x[b, j] + f[b] * init[b, j]      (init optional)
+// backward: gx = (1-f) * g ; ginit = f * g
+// ---------------------------------------------------------------------------
+
+template <typename T, bool HAS_INIT>
+__global__ void masked_lerp_fwd_kernel(const T* __restrict__ x, const T* __restrict__ init,
+                                       const T* __restrict__ f, T* __restrict__ y, long rows, int cols) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < rows * (long)cols;
+       i += (long)gridDim.x * blockDim.x) {
+    const long b = i / cols;
+    float fb = ld(f, b);
+    float v = (1.f - fb) * ld(x, i);
+    if (HAS_INIT) v += fb * ld(init, i);
+    st(y, i, v);
+  }
+}
+
+template <typename T, bool HAS_INIT>
+__global__ void masked_lerp_bwd_kernel(const T* __restrict__ g, const T* __restrict__ f, T* __restrict__ gx,
+                                       T* __restrict__ ginit, long rows, int cols) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < rows * (long)cols;
+       i += (long)gridDim.x * blockDim.x) {
+    const long b = i / cols;
+    float fb = ld(f, b);
+    float gv = ld(g, i);
+    st(gx, i, (1.f - fb) * gv);
+    if (HAS_INIT) st(ginit, i, fb * gv);
+  }
+}
+
+torch::Tensor masked_lerp_fwd(const torch::Tensor& x, const c10::optional<torch::Tensor>& init,
+                              const torch::Tensor& f) {
+  CHECK_IN(x);
+  auto y = torch::empty_like(x);
+  int cols = (int)x.size(-1);
+  long rows = x.numel() / cols;
+  int blocks = (int)std::min((x.numel() + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "masked_lerp_fwd", [&] {
+    using T = scalar_t;
+    if (init.has_value())
+      hipLaunchKernelGGL((masked_lerp_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)x.data_ptr(), (const T*)init->data_ptr(), (const T*)f.data_ptr(),
+                         (T*)y.data_ptr(), rows, cols);
+    else
+      hipLaunchKernelGGL((masked_lerp_fwd_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)x.data_ptr(), nullptr, (const T*)f.data_ptr(), (T*)y.data_ptr(), rows, cols);
+  });
+  return y;
+}
+
+std::vector<torch::Tensor> masked_lerp_bwd(const torch::Tensor& g, const torch::Tensor& f, bool has_init) {
+  CHECK_IN(g);
+  auto gx = torch::empty_like(g);
+  auto ginit = has_init ? torch::empty_like(g) : torch::empty({0}, g.options());
+  int cols = (int)g.size(-1);
+  long rows = g.numel() / cols;
+  int blocks = (int)std::min((g.numel() + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, g.scalar_type(), "masked_lerp_bwd", [&] {
+    using T = scalar_t;
+    if (has_init)
+      hipLaunchKernelGGL((masked_lerp_bwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)g.data_ptr(), (const T*)f.data_ptr(), (T*)gx.data_ptr(), (T*)ginit.data_ptr(),
+                         rows, cols);
+    else
+      hipLaunchKernelGGL((masked_lerp_bwd_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)g.data_ptr(), (const T*)f.data_ptr(), (T*)gx.data_ptr(), nullptr, rows, cols);
+  });
+  return {gx, ginit};
+}
+
 __global__ void step_inc_kernel(float* step_t) {
   if (threadIdx.x == 0 && blockIdx.x == 0) step_t[0] += 1.0f;
 }
@@ -818,6 +892,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step_dev", &adam_step_dev);
   m.def("cat_st_fwd", &cat_st_fwd);
   m.def("cat_st_bwd", &cat_st_bwd);
+  m.def("masked_lerp_fwd", &masked_lerp_fwd);
+  m.def("masked_lerp_bwd", &masked_lerp_bwd);
   m.def("ema_update", &ema_update);
   m.def("obs_norm", &obs_norm);
 }

@@ -159,3 +159,37 @@ def normalize_obs(obs: Tensor) -> Tensor:
     if obs.dtype == torch.uint8 and use_hip(obs):
         return require_ext().obs_norm(obs.contiguous())
     return obs.float() / 255.0 - 0.5
+
+
+class _MaskedLerp(torch.autograd.Function):
+    """y = (1-f)*x + f*init with a per-row mask f (episode-reset masking in
+    the RSSM scan; one kernel instead of ~4 eager broadcast ops)."""
+
+    @staticmethod
+    def forward(ctx, x: Tensor, init: Optional[Tensor], f: Tensor) -> Tensor:
+        fc = f.reshape(-1).to(x.dtype).contiguous()  # one mask entry per row
+        if use_hip(x):
+            initc = init.contiguous().to(x.dtype) if init is not None else None
+            y = require_ext().masked_lerp_fwd(x.contiguous(), initc, fc)
+        else:
+            fb = fc.view(*x.shape[:-1], 1)
+            y = (1 - fb) * x + (fb * init if init is not None else 0.0)
+        ctx.save_for_backward(fc)
+        ctx.has_init = init is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, g: Tensor):
+        (fc,) = ctx.saved_tensors
+        if use_hip(g):
+            gx, ginit = require_ext().masked_lerp_bwd(g.contiguous(), fc.to(g.dtype), ctx.has_init)
+            return gx, (ginit if ctx.has_init else None), None
+        fb = fc.view(*g.shape[:-1], 1).to(g.dtype)
+        gx = (1 - fb) * g
+        ginit = fb * g if ctx.has_init else None
+        return gx, ginit, None
+
+
+def masked_lerp(x: Tensor, init: Optional[Tensor], f: Tensor) -> Tensor:
+    """Rows of ``x``/``init`` are lerped by the per-row mask ``f`` ([*, 1])."""
+    return _MaskedLerp.apply(x, init, f)
