@@ -141,6 +141,46 @@ __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict_
   }
 }
 
+// Wave-per-row variant for short rows (conv-channel LayerNorms: D <= 256 but
+// N up to ~1M rows).  4 waves per block, grid-stride over rows.
+template <typename T, typename TW, bool SILU>
+__global__ void ln_act_fwd_small_kernel(const T* __restrict__ x, const TW* __restrict__ w,
+                                        const TW* __restrict__ b, T* __restrict__ y,
+                                        float* __restrict__ mean_out, float* __restrict__ rstd_out, long N, int D,
+                                        float eps) {
+  const int lane = threadIdx.x & 63;
+  const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
+  for (long row = wave; row < N; row += nwaves) {
+    const T* xr = x + row * (long)D;
+    T* yr = y + row * (long)D;
+    float s = 0.f, s2 = 0.f;
+    for (int j = lane; j < D; j += 64) {
+      float v = ld(xr, j);
+      s += v;
+      s2 += v * v;
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s += __shfl_xor(s, off, 64);
+      s2 += __shfl_xor(s2, off, 64);
+    }
+    float mean = s / D;
+    float var = s2 / D - mean * mean;
+    float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (lane == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+    for (int j = lane; j < D; j += 64) {
+      float xhat = (ld(xr, j) - mean) * rstd;
+      float z = xhat * ld(w, j) + ld(b, j);
+      if (SILU) z = z / (1.f + expf(-z));
+      st(yr, j, z);
+    }
+  }
+}
+
 std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& b,
                                       double eps, bool silu) {
   CHECK_IN(x);
@@ -153,11 +193,24 @@ std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tenso
   auto wc = w.contiguous();
   auto bc = b.contiguous();
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  const bool small = D <= 256;
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_fwd", [&] {
     using T = scalar_t;
     AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_fwd_w", [&] {
       using TW = scalar_t;
-      if (silu)
+      if (small) {
+        int blocks = (int)std::min((N + 3) / 4, (long)2048);
+        if (silu)
+          hipLaunchKernelGGL((ln_act_fwd_small_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), 0,
+                             stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                             (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), N, D, (float)eps);
+        else
+          hipLaunchKernelGGL((ln_act_fwd_small_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), 0,
+                             stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                             (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), N, D, (float)eps);
+      } else if (silu)
         hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                            (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
                            (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
@@ -230,6 +283,77 @@ __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict_
   }
 }
 
+// Wave-per-row backward for short rows; gw/gb accumulate in per-wave LDS
+// slices, reduced block-wide and flushed with one atomicAdd per column per
+// block.
+template <typename T, typename TW, bool SILU>
+__global__ void ln_act_bwd_small_kernel(const T* __restrict__ gy, const T* __restrict__ x,
+                                        const TW* __restrict__ w, const TW* __restrict__ b,
+                                        const float* __restrict__ mean, const float* __restrict__ rstd,
+                                        T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb,
+                                        long N, int D) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  const int nw = blockDim.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  float* gw_acc = smem + (size_t)wid * 2 * D;      // per-wave [D]
+  float* gb_acc = gw_acc + D;
+  for (int j = lane; j < D; j += 64) {
+    gw_acc[j] = 0.f;
+    gb_acc[j] = 0.f;
+  }
+  const long wave = (long)blockIdx.x * nw + wid;
+  const long nwaves = (long)gridDim.x * nw;
+  for (long row = wave; row < N; row += nwaves) {
+    const T* xr = x + row * (long)D;
+    const T* gr = gy + row * (long)D;
+    T* gxr = gx + row * (long)D;
+    const float m = mean[row], r = rstd[row];
+    float s1 = 0.f, s2 = 0.f;
+    for (int j = lane; j < D; j += 64) {
+      float xhat = (ld(xr, j) - m) * r;
+      float gz = ld(gr, j);
+      if (SILU) {
+        float z = xhat * ld(w, j) + ld(b, j);
+        float sig = 1.f / (1.f + expf(-z));
+        gz *= sig * (1.f + z * (1.f - sig));
+      }
+      float gxhat = gz * ld(w, j);
+      s1 += gxhat;
+      s2 += gxhat * xhat;
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s1 += __shfl_xor(s1, off, 64);
+      s2 += __shfl_xor(s2, off, 64);
+    }
+    float S1 = s1 / D, S2 = s2 / D;
+    for (int j = lane; j < D; j += 64) {
+      float xhat = (ld(xr, j) - m) * r;
+      float gz = ld(gr, j);
+      if (SILU) {
+        float z = xhat * ld(w, j) + ld(b, j);
+        float sig = 1.f / (1.f + expf(-z));
+        gz *= sig * (1.f + z * (1.f - sig));
+      }
+      gw_acc[j] += gz * xhat;
+      gb_acc[j] += gz;
+      float gxhat = gz * ld(w, j);
+      st(gxr, j, (gxhat - S1 - xhat * S2) * r);
+    }
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float sw = 0.f, sb = 0.f;
+    for (int k = 0; k < nw; ++k) {
+      sw += smem[(size_t)k * 2 * D + j];
+      sb += smem[(size_t)k * 2 * D + D + j];
+    }
+    atomicAdd(&gw[j], sw);
+    atomicAdd(&gb[j], sb);
+  }
+}
+
 std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tensor& x, const torch::Tensor& w,
                                       const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd,
                                       bool silu) {
@@ -242,6 +366,30 @@ std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tens
   auto gb = torch::zeros({D}, x.options().dtype(at::kFloat));
   auto wc = w.contiguous();
   auto bc = b.contiguous();
+  if (D <= 256) {
+    size_t shmem = (size_t)(kBlock >> 6) * 2 * D * sizeof(float);
+    int blocks = (int)std::min((N + 3) / 4, (long)2048);
+    auto stream2 = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_bwd_s", [&] {
+      using T = scalar_t;
+      AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_bwd_s_w", [&] {
+        using TW = scalar_t;
+        if (silu)
+          hipLaunchKernelGGL((ln_act_bwd_small_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), shmem,
+                             stream2.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                             (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
+                             gb.data_ptr<float>(), N, D);
+        else
+          hipLaunchKernelGGL((ln_act_bwd_small_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), shmem,
+                             stream2.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                             (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
+                             gb.data_ptr<float>(), N, D);
+      });
+    });
+    return {gx, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
+  }
   size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "ln_act_bwd: D too large for LDS accumulation");
   int blocks = (int)std::min(N, (long)512);

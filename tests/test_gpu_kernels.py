@@ -47,7 +47,7 @@ def test_symlog_grad_gpu():
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
-@pytest.mark.parametrize("D", [255, 512, 1536, 4096])
+@pytest.mark.parametrize("D", [32, 64, 255, 512, 1536, 4096])
 def test_ln_act(dtype, D):
     torch.manual_seed(0)
     N = 64
