@@ -171,6 +171,18 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     aggregator = MetricAggregator({})
     MetricAggregator.disabled = True
 
+    # pinned-host prefetch: the next minibatch is gathered + staged on a side
+    # stream while the GPU executes the current gradient step
+    from sheeprl_amd.data.prefetch import DevicePrefetcher
+
+    def _sample_host():
+        s = rb.sample_tensors(
+            cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device="cpu"
+        )
+        return {k: v[0] for k, v in s.items()}
+
+    prefetcher = DevicePrefetcher(_sample_host, device, depth=1)
+
     def train_fn(batch):
         train(
             runtime, world_model, actor, critic, target_critic,
@@ -206,7 +218,8 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             actions = torch.cat(acts, -1).view(num_envs, -1).float().cpu().numpy()
             real_actions = torch.stack([a.argmax(dim=-1) for a in acts], dim=-1).view(num_envs).cpu().numpy()
         step_data["actions"] = actions[None]
-        rb.add(step_data)
+        with prefetcher.lock:
+            rb.add(step_data)
         obs, rewards, term, trunc, infos = envs.step(real_actions)
         for k in cfg.algo.cnn_keys.encoder:
             step_data[k] = np.asarray(obs[k])[None]
@@ -221,13 +234,7 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         from sheeprl_amd import ops as _ops
 
         _ops.ema_update_(list(target_critic.parameters()), list(critic.parameters()), cfg.algo.critic.tau)
-        sample = rb.sample_tensors(
-            cfg.algo.per_rank_batch_size,
-            sequence_length=seq_len,
-            n_samples=1,
-            device=device,
-        )
-        batch = {k: v[0] for k, v in sample.items()}
+        batch = prefetcher.next()
         if graphed is not None:
             graphed(batch)
         else:
@@ -254,8 +261,7 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             torch.cuda.synchronize()
             phases["env"] += time.perf_counter() - t
             t = time.perf_counter()
-            sample = rb.sample_tensors(cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device=device)
-            batch = {k: v[0] for k, v in sample.items()}
+            batch = prefetcher.next()
             torch.cuda.synchronize()
             phases["sample"] += time.perf_counter() - t
             t = time.perf_counter()
@@ -297,6 +303,7 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    prefetcher.close()
     policy_steps = steps * num_envs * world_size
     env_frames_per_sec = policy_steps * cfg.env.action_repeat / elapsed
     result = {

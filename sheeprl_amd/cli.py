@@ -123,6 +123,9 @@ class _ReproducibleEntry:
         fmm = cfg.get("float32_matmul_precision", "high")
         if fmm:
             torch.set_float32_matmul_precision(fmm)
+        import torch.distributions as _td
+
+        _td.Distribution.set_default_validate_args(bool(cfg.get("distribution", {}).get("validate_args", False)))
         return fn(runtime, cfg)
 
 
