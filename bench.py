@@ -181,7 +181,8 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         )
         return {k: v[0] for k, v in s.items()}
 
-    prefetcher = DevicePrefetcher(_sample_host, device, depth=1)
+    use_prefetch = os.environ.get("SHEEPRL_AMD_NO_PREFETCH", "0") != "1"
+    prefetcher = DevicePrefetcher(_sample_host, device, depth=1) if use_prefetch else None
 
     def train_fn(batch):
         train(
@@ -218,7 +219,10 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             actions = torch.cat(acts, -1).view(num_envs, -1).float().cpu().numpy()
             real_actions = torch.stack([a.argmax(dim=-1) for a in acts], dim=-1).view(num_envs).cpu().numpy()
         step_data["actions"] = actions[None]
-        with prefetcher.lock:
+        if prefetcher is not None:
+            with prefetcher.lock:
+                rb.add(step_data)
+        else:
             rb.add(step_data)
         obs, rewards, term, trunc, infos = envs.step(real_actions)
         for k in cfg.algo.cnn_keys.encoder:
@@ -234,7 +238,11 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         from sheeprl_amd import ops as _ops
 
         _ops.ema_update_(list(target_critic.parameters()), list(critic.parameters()), cfg.algo.critic.tau)
-        batch = prefetcher.next()
+        if prefetcher is not None:
+            batch = prefetcher.next()
+        else:
+            s = rb.sample_tensors(cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device=device)
+            batch = {k: v[0] for k, v in s.items()}
         if graphed is not None:
             graphed(batch)
         else:
@@ -261,7 +269,11 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             torch.cuda.synchronize()
             phases["env"] += time.perf_counter() - t
             t = time.perf_counter()
-            batch = prefetcher.next()
+            if prefetcher is not None:
+                batch = prefetcher.next()
+            else:
+                _s = rb.sample_tensors(cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device=device)
+                batch = {k: v[0] for k, v in _s.items()}
             torch.cuda.synchronize()
             phases["sample"] += time.perf_counter() - t
             t = time.perf_counter()
@@ -303,7 +315,8 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
-    prefetcher.close()
+    if prefetcher is not None:
+        prefetcher.close()
     policy_steps = steps * num_envs * world_size
     env_frames_per_sec = policy_steps * cfg.env.action_repeat / elapsed
     result = {
