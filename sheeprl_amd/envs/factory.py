@@ -78,6 +78,8 @@ def make_env(
             env = wrappers.ClipReward(env)
         if env_cfg.get("max_episode_steps") and env_cfg.max_episode_steps > 0:
             env = wrappers.TimeLimit(env, env_cfg.max_episode_steps)
+        if env_cfg.get("capture_video", False) and run_name and rank == 0 and vector_env_idx == 0 and cnn_keys:
+            env = wrappers.RecordVideo(env, f"logs/runs/{run_name}/videos", key=cnn_keys[0])
         env = wrappers.RecordEpisodeStatistics(env)
 
         env.observation_space.seed(seed + rank * 1024 + vector_env_idx)

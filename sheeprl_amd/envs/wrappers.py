@@ -390,3 +390,47 @@ class ResizeObservation(ObservationWrapper):
         for k in self.keys:
             out[k] = resize_area(obs[k], self.size)
         return out
+
+
+class RecordVideo(Wrapper):
+    """Record episode frames (parity role: gym's RecordVideoV0 used via
+    cfg.env.capture_video).  No video codecs ship in this image, so episodes
+    are saved as compressed ``.npz`` frame stacks under ``video_dir`` —
+    loadable with ``np.load(...)["frames"]`` ([T,C,H,W] uint8)."""
+
+    def __init__(self, env: Env, video_dir: str, key: str = "rgb", every_n_episodes: int = 1) -> None:
+        super().__init__(env)
+        import os
+
+        self.video_dir = video_dir
+        os.makedirs(video_dir, exist_ok=True)
+        self.key = key
+        self.every = max(1, every_n_episodes)
+        self._frames: list = []
+        self._episode = 0
+
+    def _grab(self, obs: Any) -> None:
+        if self._episode % self.every != 0:
+            return
+        frame = obs.get(self.key) if isinstance(obs, dict) else obs
+        if frame is not None:
+            self._frames.append(np.asarray(frame))
+
+    def reset(self, *, seed=None, options=None):
+        obs, info = self.env.reset(seed=seed, options=options)
+        self._frames = []
+        self._grab(obs)
+        return obs, info
+
+    def step(self, action):
+        obs, r, term, trunc, info = self.env.step(action)
+        self._grab(obs)
+        if term or trunc:
+            if self._frames and self._episode % self.every == 0:
+                import os
+
+                path = os.path.join(self.video_dir, f"episode_{self._episode}.npz")
+                np.savez_compressed(path, frames=np.stack(self._frames))
+            self._episode += 1
+            self._frames = []
+        return obs, r, term, trunc, info
