@@ -282,3 +282,77 @@ def test_categorical_st_sampling_distribution():
     freq = oh.float().mean(0).cpu()
     p = (0.99 * torch.softmax(torch.tensor([2.0, 0.0, -1.0, 0.5]), -1) + 0.01 / 4)
     assert torch.allclose(freq, p, atol=0.02), (freq, p)
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_fused_rssm_scan_matches_module_loop():
+    """ops.scan.rssm_scan (hand-written backward) vs the module-based
+    dynamic_posterior loop: identical outputs and gradients given the same
+    philox stream (fp32)."""
+    from sheeprl_amd.algos.dreamer_v3.agent import RSSM, RecurrentModel
+    from sheeprl_amd.models import MLP
+    from sheeprl_amd.ops.scan import rssm_scan, scan_applicable
+
+    T, B, E, A, H, S, K, DU, P = 5, 3, 24, 6, 16, 4, 4, 16, 20
+    SK = S * K
+    torch.manual_seed(0)
+    rssm = RSSM(
+        RecurrentModel(SK + A, H, DU),
+        MLP(E + H, SK, [P], activation="silu", layer_norm=True, layer_norm_eps=1e-3),
+        MLP(H, SK, [P], activation="silu", layer_norm=True, layer_norm_eps=1e-3),
+        discrete=K,
+        unimix=0.01,
+    ).cuda()
+    assert scan_applicable(rssm)
+
+    embed = torch.randn(T, B, E, device="cuda")
+    actions = torch.randn(T, B, A, device="cuda")
+    is_first = (torch.rand(T, B, 1, device="cuda") < 0.2).float()
+    is_first[0] = 1.0
+
+    def module_loop():
+        _ir, _ip = rssm.get_initial_states((1, B))
+        initial_states = (_ir.contiguous(), _ip.contiguous())
+        posterior = torch.zeros(1, B, S, K, device="cuda")
+        hs, zs, ms = [], [], []
+        recurrent_state = torch.zeros(1, B, H, device="cuda")
+        for i in range(T):
+            recurrent_state, posterior, plogits = rssm.dynamic_posterior(
+                posterior, recurrent_state, actions[i : i + 1], embed[i : i + 1],
+                is_first[i : i + 1], initial_states=initial_states,
+            )
+            hs.append(recurrent_state[0])
+            zs.append(posterior.view(B, SK))
+            ms.append(plogits[0])
+        return torch.stack(hs), torch.stack(zs), torch.stack(ms)
+
+    def fused():
+        _ir, _ip = rssm.get_initial_states((1, B))
+        return rssm_scan(rssm, embed, actions, is_first, (_ir.contiguous(), _ip.contiguous()))
+
+    gh = torch.randn(T, B, H, device="cuda")
+    gz = torch.randn(T, B, SK, device="cuda")
+    gm = torch.randn(T, B, SK, device="cuda")
+
+    torch.manual_seed(42)
+    h1, z1, m1 = module_loop()
+    torch.autograd.backward([h1, z1, m1], [gh, gz, gm])
+    grads1 = {n: p.grad.clone() for n, p in rssm.named_parameters() if p.grad is not None}
+    for p in rssm.parameters():
+        p.grad = None
+
+    torch.manual_seed(42)
+    h2, z2, m2 = fused()
+    assert torch.allclose(h2, h1.detach(), atol=1e-4, rtol=1e-4)
+    assert torch.equal(z2, z1.detach())
+    assert torch.allclose(m2, m1.detach(), atol=1e-4, rtol=1e-4)
+    torch.autograd.backward([h2, z2, m2], [gh, gz, gm])
+    for n, g1 in grads1.items():
+        p = dict(rssm.named_parameters())[n]
+        if p.grad is None:
+            assert torch.allclose(g1, torch.zeros_like(g1), atol=1e-5), f"{n} lost grad"
+            continue
+        assert torch.allclose(p.grad, g1, atol=2e-3, rtol=2e-3), (
+            n, (p.grad - g1).abs().max().item()
+        )
