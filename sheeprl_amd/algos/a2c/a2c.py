@@ -26,6 +26,7 @@ from sheeprl_amd.utils.registry import register_algorithm, register_evaluation
 from sheeprl_amd.utils.timer import timer
 
 AGGREGATOR_KEYS = {"Rewards/rew_avg", "Game/ep_len_avg", "Loss/value_loss", "Loss/policy_loss"}
+MODELS_TO_REGISTER = {"agent"}
 
 
 def train(

@@ -43,6 +43,7 @@ AGGREGATOR_KEYS = {
     "Loss/ensemble_loss",
     "Rewards/intrinsic",
 }
+MODELS_TO_REGISTER = {"world_model", "ensembles", "actor_task", "critic_task", "actor_exploration", "critic_exploration"}
 
 
 def _dv1_behaviour(runtime, cfg, world_model, actor, critic, actor_opt, critic_opt,

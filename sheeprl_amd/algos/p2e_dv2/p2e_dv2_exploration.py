@@ -48,6 +48,7 @@ AGGREGATOR_KEYS = {
     "Rewards/intrinsic",
     "State/kl",
 }
+MODELS_TO_REGISTER = {"world_model", "ensembles", "actor_task", "critic_task", "actor_exploration", "critic_exploration"}
 
 
 def _behaviour_update(

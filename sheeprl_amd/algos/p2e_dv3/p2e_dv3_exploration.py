@@ -45,6 +45,7 @@ AGGREGATOR_KEYS = {
     "State/kl",
     "Rewards/intrinsic_intrinsic",
 }
+MODELS_TO_REGISTER = {"world_model", "ensembles", "actor_task", "critic_task", "target_critic_task", "actor_exploration"}
 
 
 def train(

@@ -11,7 +11,7 @@ from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.registry import register_evaluation
 
 
-@register_evaluation(algorithms=["sac"])
+@register_evaluation(algorithms=["sac", "sac_decoupled"])
 def evaluate(runtime: Runtime, cfg: Any, state: Dict[str, Any]) -> float:
     env_fn = make_env(cfg, cfg.seed, 0)
     env = env_fn()

@@ -38,6 +38,7 @@ AGGREGATOR_KEYS = {
     "Loss/alpha_loss",
     "Loss/reconstruction_loss",
 }
+MODELS_TO_REGISTER = {"agent"}
 
 
 def train(
