@@ -43,6 +43,8 @@ class DROQCritic(nn.Module):
 
     def forward(self, obs: Tensor, action: Tensor) -> Tensor:
         return self.model(torch.cat([obs, action], dim=-1))
+
+
 MODELS_TO_REGISTER = {"agent"}
 
 
