@@ -16,7 +16,7 @@ from sheeprl_amd.algos.ppo.agent import build_agent
 from sheeprl_amd.algos.ppo.utils import prepare_obs, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
-from sheeprl_amd.envs import make_env, spaces, vectorize_env
+from sheeprl_amd.envs import make_env, vectorize_env
 from sheeprl_amd.ops import gae as compute_gae
 from sheeprl_amd.optim import FusedAdam
 from sheeprl_amd.parallel import Runtime

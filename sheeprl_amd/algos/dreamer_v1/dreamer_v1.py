@@ -12,9 +12,8 @@ ELU/no-LN configuration; the stochastic state is a 30-dim diagonal Normal.
 
 from __future__ import annotations
 
-import copy
 import os
-from typing import Any, Dict, Optional, Sequence, Tuple
+from typing import Any, Dict, Optional, Sequence
 
 import numpy as np
 import torch

@@ -23,7 +23,6 @@ from sheeprl_amd.algos.dreamer_v3.loss import categorical_kl
 from sheeprl_amd.algos.dreamer_v3.utils import prepare_obs, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import EnvIndependentReplayBuffer, EpisodeBuffer, SequentialReplayBuffer
-from sheeprl_amd.distributions import BernoulliSafeMode
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
 from sheeprl_amd.optim import FusedAdam
 from sheeprl_amd.parallel import Runtime

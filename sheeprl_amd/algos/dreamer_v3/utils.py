@@ -3,7 +3,7 @@ Moments :40, compute_lambda_values :66, prepare_obs :80, test :94)."""
 
 from __future__ import annotations
 
-from typing import Any, Dict, Optional, Sequence
+from typing import Any, Dict, Sequence
 
 import numpy as np
 import torch

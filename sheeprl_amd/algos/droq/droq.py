@@ -14,7 +14,6 @@ import numpy as np
 import torch
 from torch import Tensor, nn
 
-from sheeprl_amd.algos import sac as sac_pkg
 from sheeprl_amd.algos.sac.agent import SACActor, SACAgent, SACPlayer
 from sheeprl_amd.algos.sac.sac import main as sac_main
 from sheeprl_amd.algos.sac.utils import test

@@ -8,9 +8,8 @@ exploration actor/critic trained by dynamics backprop on the intrinsic
 
 from __future__ import annotations
 
-import copy
 import os
-from typing import Any, Dict, Sequence
+from typing import Any, Dict
 
 import numpy as np
 import torch
@@ -18,7 +17,7 @@ import torch.distributions as td
 import torch.nn.functional as F
 
 from sheeprl_amd import ops
-from sheeprl_amd.algos.dreamer_v1.dreamer_v1 import build_agent as dv1_build_agent, train as dv1_world_train
+from sheeprl_amd.algos.dreamer_v1.dreamer_v1 import build_agent as dv1_build_agent
 from sheeprl_amd.algos.dreamer_v3.agent import Actor, init_weights
 from sheeprl_amd.algos.dreamer_v3.utils import prepare_obs, test
 from sheeprl_amd.config import save_config

@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import copy
 import os
-from typing import Any, Dict, Sequence
+from typing import Any, Dict
 
 import numpy as np
 import torch

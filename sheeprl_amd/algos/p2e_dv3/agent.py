@@ -10,11 +10,11 @@ Moments normalizer.
 from __future__ import annotations
 
 import copy
-from typing import Any, Dict, Optional, Sequence, Tuple
+from typing import Any, Dict, Optional, Sequence
 
 import numpy as np
 import torch
-from torch import Tensor, nn
+from torch import nn
 
 from sheeprl_amd.algos.dreamer_v3.agent import (
     Actor,

@@ -8,7 +8,6 @@ continuous action spaces.
 
 from __future__ import annotations
 
-import copy
 from typing import Any, Dict, List, Optional, Sequence, Tuple
 
 import torch

@@ -8,7 +8,6 @@ backward), identical replicas.
 
 from __future__ import annotations
 
-import copy
 import os
 import time
 from typing import Any, Dict

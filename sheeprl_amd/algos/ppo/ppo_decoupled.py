@@ -23,7 +23,7 @@ import numpy as np
 import torch
 from torch.nn.utils import parameters_to_vector, vector_to_parameters
 
-from sheeprl_amd.algos.ppo.agent import PPOAgent, PPOPlayer, build_agent
+from sheeprl_amd.algos.ppo.agent import PPOAgent, PPOPlayer
 from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, value_loss
 from sheeprl_amd.algos.ppo.utils import AGGREGATOR_KEYS, prepare_obs, test
 from sheeprl_amd.config import save_config

@@ -8,8 +8,6 @@ import numpy as np
 import torch
 from torch import Tensor
 
-from sheeprl_amd import ops
-from sheeprl_amd.utils.metric import MetricAggregator
 
 AGGREGATOR_KEYS = {
     "Rewards/rew_avg",

@@ -14,7 +14,7 @@ from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, value_loss
 from sheeprl_amd.algos.ppo.utils import AGGREGATOR_KEYS, prepare_obs
 from sheeprl_amd.algos.ppo_recurrent.agent import build_agent
 from sheeprl_amd.config import save_config
-from sheeprl_amd.envs import make_env, spaces, vectorize_env
+from sheeprl_amd.envs import make_env, vectorize_env
 from sheeprl_amd.ops import gae as compute_gae
 from sheeprl_amd.optim import FusedAdam
 from sheeprl_amd.parallel import Runtime

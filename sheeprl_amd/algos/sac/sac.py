@@ -18,7 +18,7 @@ import torch
 
 from sheeprl_amd.algos.sac.agent import build_agent
 from sheeprl_amd.algos.sac.loss import critic_loss, entropy_loss, policy_loss
-from sheeprl_amd.algos.sac.utils import AGGREGATOR_KEYS, prepare_obs, test
+from sheeprl_amd.algos.sac.utils import AGGREGATOR_KEYS, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env

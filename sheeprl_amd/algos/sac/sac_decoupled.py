@@ -9,13 +9,13 @@ the player, sampled chunks scattered per training round :240-257), trainer
 from __future__ import annotations
 
 import os
-from typing import Any, Dict, List
+from typing import Any, List
 
 import numpy as np
 import torch
 from torch.nn.utils import parameters_to_vector, vector_to_parameters
 
-from sheeprl_amd.algos.sac.agent import SACAgent, SACPlayer, build_agent
+from sheeprl_amd.algos.sac.agent import SACAgent, SACPlayer
 from sheeprl_amd.algos.sac.sac import train as sac_train
 from sheeprl_amd.algos.sac.utils import AGGREGATOR_KEYS, test
 from sheeprl_amd.config import save_config

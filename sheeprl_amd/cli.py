@@ -16,16 +16,15 @@ Usage::
 from __future__ import annotations
 
 import copy
-import importlib
 import os
 import pathlib
 import sys
 import warnings
-from typing import Any, List, Optional
+from typing import List, Optional
 
 import torch
 
-from sheeprl_amd.config import compose, save_config
+from sheeprl_amd.config import compose
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.callback import CheckpointCallback
 from sheeprl_amd.utils.dotdict import DotDict

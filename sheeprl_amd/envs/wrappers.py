@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import time
 from collections import deque
-from typing import Any, Callable, Dict, Optional, Tuple
+from typing import Any, Callable, Dict, Tuple
 
 import numpy as np
 

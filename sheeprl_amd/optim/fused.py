@@ -11,7 +11,7 @@
 
 from __future__ import annotations
 
-from typing import Iterable, Optional
+from typing import Iterable
 
 import torch
 from torch import Tensor

@@ -11,7 +11,7 @@ Parity surface with the reference (sheeprl/utils/metric.py):
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Any, Dict, List, Optional
 
 import torch
 

@@ -14,7 +14,7 @@ from __future__ import annotations
 
 import os
 import random
-from typing import Any, Dict, Optional, Sequence
+from typing import Any, Dict, Optional
 
 import numpy as np
 import torch
