@@ -16,7 +16,6 @@ MI355X notes:
 
 from __future__ import annotations
 
-import copy
 import os
 from typing import Any, Dict, Sequence
 
@@ -40,7 +39,6 @@ from sheeprl_amd.data import EnvIndependentReplayBuffer, EpisodeBuffer, Sequenti
 from sheeprl_amd.distributions import (
     BernoulliSafeMode,
     MSEDistribution,
-    OneHotCategoricalST,
     SymlogDistribution,
     TwoHotEncodingDistribution,
 )
