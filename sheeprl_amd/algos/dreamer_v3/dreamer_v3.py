@@ -311,6 +311,8 @@ def _capture_train_step(
             MetricAggregator.disabled = was_disabled
 
     try:
+        # NOTE: warmup (2) + capture (1) each execute a real gradient step;
+        # these three updates are not counted by the cumulative counters.
         step = CUDAGraphStep(train_fn, example_batch, warmup=2)
         runtime.print("[dreamer_v3] gradient step captured in a hipGraph")
         return step

@@ -124,7 +124,12 @@ class GradSync:
             bucket.launch(self.group, self.world_size)
 
     def finalize(self) -> None:
-        """Launch stragglers, wait for all in-flight reduces, write back."""
+        """Launch stragglers, wait for all in-flight reduces, write back.
+
+        Collective matching relies on data-INDEPENDENT control flow: every
+        rank runs the same module graph per step (true for all shipped
+        algorithms), so the same buckets fire in the same order everywhere.
+        """
         if not self.enabled or not self._any_ready:
             return
         for b in self.buckets:
