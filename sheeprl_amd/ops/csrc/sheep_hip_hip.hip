@@ -113,11 +113,11 @@ torch::Tensor symmath(const torch::Tensor& x, const c10::optional<torch::Tensor>
 template <typename T, typename TW, bool SILU>
 __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict__ w,
                                   const TW* __restrict__ b, T* __restrict__ y, float* __restrict__ mean_out,
-                                  float* __restrict__ rstd_out, int D, float eps) {
+                                  float* __restrict__ rstd_out, int D, float eps, long ys) {
   __shared__ float lds[9];
   const long row = blockIdx.x;
   const T* xr = x + row * (long)D;
-  T* yr = y + row * (long)D;
+  T* yr = y + row * ys;
   float s = 0.f;
   for (int j = threadIdx.x; j < D; j += blockDim.x) s += ld(xr, j);
   float mean = block_sum(s, lds) / D;
@@ -147,13 +147,13 @@ template <typename T, typename TW, bool SILU>
 __global__ void ln_act_fwd_small_kernel(const T* __restrict__ x, const TW* __restrict__ w,
                                         const TW* __restrict__ b, T* __restrict__ y,
                                         float* __restrict__ mean_out, float* __restrict__ rstd_out, long N, int D,
-                                        float eps) {
+                                        float eps, long ys) {
   const int lane = threadIdx.x & 63;
   const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
   for (long row = wave; row < N; row += nwaves) {
     const T* xr = x + row * (long)D;
-    T* yr = y + row * (long)D;
+    T* yr = y + row * ys;
     float s = 0.f, s2 = 0.f;
     for (int j = lane; j < D; j += 64) {
       float v = ld(xr, j);
@@ -181,15 +181,10 @@ __global__ void ln_act_fwd_small_kernel(const T* __restrict__ x, const TW* __res
   }
 }
 
-std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& b,
-                                      double eps, bool silu) {
-  CHECK_IN(x);
-  TORCH_CHECK(x.dim() == 2, "ln_act_fwd expects [N, D]");
+void ln_act_fwd_core(const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& b, double eps,
+                     bool silu, torch::Tensor& y, torch::Tensor& mean, torch::Tensor& rstd, long ys) {
   long N = x.size(0);
   int D = (int)x.size(1);
-  auto y = torch::empty_like(x);
-  auto mean = torch::empty({N}, x.options().dtype(at::kFloat));
-  auto rstd = torch::empty({N}, x.options().dtype(at::kFloat));
   auto wc = w.contiguous();
   auto bc = b.contiguous();
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -204,23 +199,46 @@ std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tenso
           hipLaunchKernelGGL((ln_act_fwd_small_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), 0,
                              stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                              (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
-                             rstd.data_ptr<float>(), N, D, (float)eps);
+                             rstd.data_ptr<float>(), N, D, (float)eps, ys);
         else
           hipLaunchKernelGGL((ln_act_fwd_small_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), 0,
                              stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                              (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
-                             rstd.data_ptr<float>(), N, D, (float)eps);
+                             rstd.data_ptr<float>(), N, D, (float)eps, ys);
       } else if (silu)
         hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                            (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
-                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps, ys);
       else
         hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                            (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
-                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps, ys);
     });
   });
+}
+
+std::vector<torch::Tensor> ln_act_fwd(const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& b,
+                                      double eps, bool silu) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dim() == 2, "ln_act_fwd expects [N, D]");
+  long N = x.size(0);
+  int D = (int)x.size(1);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({N}, x.options().dtype(at::kFloat));
+  auto rstd = torch::empty({N}, x.options().dtype(at::kFloat));
+  ln_act_fwd_core(x, w, b, eps, silu, y, mean, rstd, (long)D);
   return {y, mean, rstd};
+}
+
+// Out-variant for the fused scan: y may be a row-strided 2-D view (a slice of
+// a stacked [T, B, *] buffer); mean/rstd are caller-provided fp32 [N] slices.
+torch::Tensor ln_act_fwd_o(const torch::Tensor& x, const torch::Tensor& w, const torch::Tensor& b, double eps,
+                           bool silu, torch::Tensor y, torch::Tensor mean, torch::Tensor rstd) {
+  CHECK_IN(x);
+  TORCH_CHECK(x.dim() == 2 && y.dim() == 2 && y.stride(1) == 1, "ln_act_fwd_o shapes");
+  TORCH_CHECK(y.scalar_type() == x.scalar_type() && mean.is_contiguous() && rstd.is_contiguous());
+  ln_act_fwd_core(x, w, b, eps, silu, y, mean, rstd, y.stride(0));
+  return y;
 }
 
 // Two-stage weight-grad reduction: a bounded grid of blocks strides the rows,
@@ -231,7 +249,7 @@ template <typename T, typename TW, bool SILU>
 __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict__ x, const TW* __restrict__ w,
                                   const TW* __restrict__ b, const float* __restrict__ mean,
                                   const float* __restrict__ rstd, T* __restrict__ gx, float* __restrict__ gw,
-                                  float* __restrict__ gb, long N, int D) {
+                                  float* __restrict__ gb, long N, int D, long gys) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* lds = smem;          // 9 floats for block_sum
   float* gw_acc = smem + 16;  // [D]
@@ -243,7 +261,7 @@ __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict_
   __syncthreads();
   for (long row = blockIdx.x; row < N; row += gridDim.x) {
     const T* xr = x + row * (long)D;
-    const T* gr = gy + row * (long)D;
+    const T* gr = gy + row * gys;
     T* gxr = gx + row * (long)D;
     const float m = mean[row], r = rstd[row];
     float s1 = 0.f, s2 = 0.f;
@@ -291,7 +309,7 @@ __global__ void ln_act_bwd_small_kernel(const T* __restrict__ gy, const T* __res
                                         const TW* __restrict__ w, const TW* __restrict__ b,
                                         const float* __restrict__ mean, const float* __restrict__ rstd,
                                         T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb,
-                                        long N, int D) {
+                                        long N, int D, long gys) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   const int nw = blockDim.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -306,7 +324,7 @@ __global__ void ln_act_bwd_small_kernel(const T* __restrict__ gy, const T* __res
   const long nwaves = (long)gridDim.x * nw;
   for (long row = wave; row < N; row += nwaves) {
     const T* xr = x + row * (long)D;
-    const T* gr = gy + row * (long)D;
+    const T* gr = gy + row * gys;
     T* gxr = gx + row * (long)D;
     const float m = mean[row], r = rstd[row];
     float s1 = 0.f, s2 = 0.f;
@@ -354,16 +372,12 @@ __global__ void ln_act_bwd_small_kernel(const T* __restrict__ gy, const T* __res
   }
 }
 
-std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tensor& x, const torch::Tensor& w,
-                                      const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd,
-                                      bool silu) {
-  CHECK_IN(gy);
+void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torch::Tensor& w,
+                     const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd, bool silu,
+                     torch::Tensor& gx, torch::Tensor& gw, torch::Tensor& gb, long gys) {
   CHECK_IN(x);
   long N = x.size(0);
   int D = (int)x.size(1);
-  auto gx = torch::empty_like(x);
-  auto gw = torch::zeros({D}, x.options().dtype(at::kFloat));
-  auto gb = torch::zeros({D}, x.options().dtype(at::kFloat));
   auto wc = w.contiguous();
   auto bc = b.contiguous();
   if (D <= 256) {
@@ -379,16 +393,16 @@ std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tens
                              stream2.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                              (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
                              rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
-                             gb.data_ptr<float>(), N, D);
+                             gb.data_ptr<float>(), N, D, gys);
         else
           hipLaunchKernelGGL((ln_act_bwd_small_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), shmem,
                              stream2.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                              (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
                              rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
-                             gb.data_ptr<float>(), N, D);
+                             gb.data_ptr<float>(), N, D, gys);
       });
     });
-    return {gx, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
+    return;
   }
   size_t shmem = (16 + 2 * (size_t)D) * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "ln_act_bwd: D too large for LDS accumulation");
@@ -402,15 +416,39 @@ std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tens
         hipLaunchKernelGGL((ln_act_bwd_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
                            (const T*)gy.data_ptr(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                            (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D);
+                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
       else
         hipLaunchKernelGGL((ln_act_bwd_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
                            (const T*)gy.data_ptr(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                            (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D);
+                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
     });
   });
+}
+
+std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tensor& x, const torch::Tensor& w,
+                                      const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd,
+                                      bool silu) {
+  CHECK_IN(gy);
+  CHECK_IN(x);
+  int D = (int)x.size(1);
+  auto gx = torch::empty_like(x);
+  auto gw = torch::zeros({D}, x.options().dtype(at::kFloat));
+  auto gb = torch::zeros({D}, x.options().dtype(at::kFloat));
+  ln_act_bwd_core(gy, x, w, b, mean, rstd, silu, gx, gw, gb, (long)D);
   return {gx, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
+}
+
+// Accumulate-variant for the fused scan backward: gy may be row-strided;
+// gx is a caller-provided contiguous output; gw/gb are fp32 accumulators
+// (zeroed once per scan by the caller) — no per-step zero-fill or cast.
+torch::Tensor ln_act_bwd_acc(const torch::Tensor& gy, const torch::Tensor& x, const torch::Tensor& w,
+                             const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd,
+                             bool silu, torch::Tensor gx, torch::Tensor gw, torch::Tensor gb) {
+  TORCH_CHECK(gy.dim() == 2 && gy.stride(1) == 1 && gx.is_contiguous(), "ln_act_bwd_acc shapes");
+  TORCH_CHECK(gw.scalar_type() == at::kFloat && gb.scalar_type() == at::kFloat, "acc buffers must be fp32");
+  ln_act_bwd_core(gy, x, w, b, mean, rstd, silu, gx, gw, gb, gy.stride(0));
+  return gx;
 }
 
 // ---------------------------------------------------------------------------
@@ -420,13 +458,15 @@ std::vector<torch::Tensor> ln_act_bwd(const torch::Tensor& gy, const torch::Tens
 template <typename T, typename TW>
 __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ h, const TW* __restrict__ w,
                                      const TW* __restrict__ b, T* __restrict__ hout, float* __restrict__ mean_out,
-                                     float* __restrict__ rstd_out, int H, float eps) {
+                                     float* __restrict__ rstd_out, int H, float eps, long hs,
+                                     T* __restrict__ hout2, long h2s) {
   __shared__ float lds[9];
   const long row = blockIdx.x;
   const int D = 3 * H;
   const T* yr = y + row * (long)D;
-  const T* hr = h + row * (long)H;
+  const T* hr = h + row * hs;
   T* outr = hout + row * (long)H;
+  T* outr2 = hout2 ? hout2 + row * h2s : nullptr;
   float s = 0.f;
   for (int j = threadIdx.x; j < D; j += blockDim.x) s += ld(yr, j);
   float mean = block_sum(s, lds) / D;
@@ -449,20 +489,19 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
     float r = 1.f / (1.f + expf(-zr));
     float c = tanhf(r * zc);
     float u = 1.f / (1.f + expf(-(zu - 1.f)));
-    st(outr, j, u * c + (1.f - u) * ld(hr, j));
+    float hv = u * c + (1.f - u) * ld(hr, j);
+    st(outr, j, hv);
+    if (outr2) st(outr2, j, hv);
   }
 }
 
-std::vector<torch::Tensor> gru_gates_fwd(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
-                                         const torch::Tensor& b, double eps) {
+void gru_gates_fwd_core(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
+                        const torch::Tensor& b, double eps, torch::Tensor& hout, torch::Tensor& mean,
+                        torch::Tensor& rstd, long hs, void* hout2, long h2s) {
   CHECK_IN(y);
-  CHECK_IN(h);
   TORCH_CHECK(y.dim() == 2 && h.dim() == 2 && y.size(1) == 3 * h.size(1), "gru_gates_fwd shapes");
   long N = y.size(0);
   int H = (int)h.size(1);
-  auto hout = torch::empty_like(h);
-  auto mean = torch::empty({N}, y.options().dtype(at::kFloat));
-  auto rstd = torch::empty({N}, y.options().dtype(at::kFloat));
   auto wc = w.contiguous();
   auto bc = b.contiguous();
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -473,10 +512,31 @@ std::vector<torch::Tensor> gru_gates_fwd(const torch::Tensor& y, const torch::Te
       hipLaunchKernelGGL((gru_gates_fwd_kernel<T, TW>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                          (const T*)y.data_ptr(), (const T*)h.data_ptr(), (const TW*)wc.data_ptr(),
                          (const TW*)bc.data_ptr(), (T*)hout.data_ptr(), mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), H, (float)eps);
+                         rstd.data_ptr<float>(), H, (float)eps, hs, (T*)hout2, h2s);
     });
   });
+}
+
+std::vector<torch::Tensor> gru_gates_fwd(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
+                                         const torch::Tensor& b, double eps) {
+  CHECK_IN(h);
+  auto hout = torch::empty_like(h);
+  auto mean = torch::empty({y.size(0)}, y.options().dtype(at::kFloat));
+  auto rstd = torch::empty({y.size(0)}, y.options().dtype(at::kFloat));
+  gru_gates_fwd_core(y, h, w, b, eps, hout, mean, rstd, h.size(1), nullptr, 0);
   return {hout, mean, rstd};
+}
+
+// Out-variant for the fused scan: hprev may be row-strided (a [:, :H] slice of
+// the stacked GRU-input buffer); h is written both to the contiguous hout
+// (next-step input / output sequence) and to a strided hout2 slice of the
+// stacked representation-input buffer.  mean/rstd are caller-provided slices.
+void gru_gates_fwd_o(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
+                     const torch::Tensor& b, double eps, torch::Tensor hout, torch::Tensor hout2,
+                     torch::Tensor mean, torch::Tensor rstd) {
+  TORCH_CHECK(h.dim() == 2 && h.stride(1) == 1 && hout.is_contiguous(), "gru_gates_fwd_o shapes");
+  TORCH_CHECK(hout2.dim() == 2 && hout2.stride(1) == 1 && hout2.scalar_type() == hout.scalar_type());
+  gru_gates_fwd_core(y, h, w, b, eps, hout, mean, rstd, h.stride(0), hout2.data_ptr(), hout2.stride(0));
 }
 
 template <typename T, typename TW>
@@ -484,7 +544,7 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
                                      const TW* __restrict__ w, const TW* __restrict__ b,
                                      const float* __restrict__ mean, const float* __restrict__ rstd,
                                      T* __restrict__ gy, T* __restrict__ ghprev, float* __restrict__ gw,
-                                     float* __restrict__ gb, long N, int H) {
+                                     float* __restrict__ gb, long N, int H, long hs) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* lds = smem;
   float* gw_acc = smem + 16;       // [3H]
@@ -497,7 +557,7 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
   __syncthreads();
   for (long row = blockIdx.x; row < N; row += gridDim.x) {
     const T* yr = y + row * (long)D;
-    const T* hr = h + row * (long)H;
+    const T* hr = h + row * hs;
     const T* ghr = gh + row * (long)H;
     T* gyr = gy + row * (long)D;
     T* ghp = ghprev + row * (long)H;
@@ -557,18 +617,14 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
   }
 }
 
-std::vector<torch::Tensor> gru_gates_bwd(const torch::Tensor& gh, const torch::Tensor& y, const torch::Tensor& h,
-                                         const torch::Tensor& w, const torch::Tensor& b, const torch::Tensor& mean,
-                                         const torch::Tensor& rstd) {
+void gru_gates_bwd_core(const torch::Tensor& gh, const torch::Tensor& y, const torch::Tensor& h,
+                        const torch::Tensor& w, const torch::Tensor& b, const torch::Tensor& mean,
+                        const torch::Tensor& rstd, torch::Tensor& gy, torch::Tensor& ghprev, torch::Tensor& gw,
+                        torch::Tensor& gb, long hs) {
   CHECK_IN(gh);
   CHECK_IN(y);
-  CHECK_IN(h);
   long N = y.size(0);
-  int H = (int)h.size(1);
-  auto gy = torch::empty_like(y);
-  auto ghprev = torch::empty_like(h);
-  auto gw = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
-  auto gb = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
+  int H = (int)(y.size(1) / 3);
   auto wc = w.contiguous();
   auto bc = b.contiguous();
   size_t shmem = (16 + 6 * (size_t)H) * sizeof(float);
@@ -583,10 +639,34 @@ std::vector<torch::Tensor> gru_gates_bwd(const torch::Tensor& gh, const torch::T
                          (const T*)gh.data_ptr(), (const T*)y.data_ptr(), (const T*)h.data_ptr(),
                          (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
                          rstd.data_ptr<float>(), (T*)gy.data_ptr(), (T*)ghprev.data_ptr(), gw.data_ptr<float>(),
-                         gb.data_ptr<float>(), N, H);
+                         gb.data_ptr<float>(), N, H, hs);
     });
   });
+}
+
+std::vector<torch::Tensor> gru_gates_bwd(const torch::Tensor& gh, const torch::Tensor& y, const torch::Tensor& h,
+                                         const torch::Tensor& w, const torch::Tensor& b, const torch::Tensor& mean,
+                                         const torch::Tensor& rstd) {
+  CHECK_IN(h);
+  int H = (int)h.size(1);
+  auto gy = torch::empty_like(y);
+  auto ghprev = torch::empty_like(h);
+  auto gw = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
+  auto gb = torch::zeros({3 * H}, y.options().dtype(at::kFloat));
+  gru_gates_bwd_core(gh, y, h, w, b, mean, rstd, gy, ghprev, gw, gb, (long)H);
   return {gy, ghprev, gw.to(w.scalar_type()), gb.to(b.scalar_type())};
+}
+
+// Accumulate-variant for the fused scan backward: hprev may be row-strided,
+// gy/ghprev are caller-provided outputs, gw/gb fp32 accumulators.
+void gru_gates_bwd_acc(const torch::Tensor& gh, const torch::Tensor& y, const torch::Tensor& h,
+                       const torch::Tensor& w, const torch::Tensor& b, const torch::Tensor& mean,
+                       const torch::Tensor& rstd, torch::Tensor gy, torch::Tensor ghprev, torch::Tensor gw,
+                       torch::Tensor gb) {
+  TORCH_CHECK(h.dim() == 2 && h.stride(1) == 1 && gy.is_contiguous() && ghprev.is_contiguous(),
+              "gru_gates_bwd_acc shapes");
+  TORCH_CHECK(gw.scalar_type() == at::kFloat && gb.scalar_type() == at::kFloat, "acc buffers must be fp32");
+  gru_gates_bwd_core(gh, y, h, w, b, mean, rstd, gy, ghprev, gw, gb, h.stride(0));
 }
 
 // ---------------------------------------------------------------------------
@@ -834,6 +914,27 @@ std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optio
   return {m, onehot, s};
 }
 
+// Out-variant for the fused scan forward: m/onehot/s are caller-provided
+// slices of stacked [T, ...] buffers (all contiguous); onehot is written in
+// the compute dtype directly (no separate cast kernel).
+void cat_st_fwd_o(const torch::Tensor& raw, const torch::Tensor& urand, double unimix, torch::Tensor m,
+                  torch::Tensor onehot, torch::Tensor s) {
+  CHECK_IN(raw);
+  TORCH_CHECK(m.is_contiguous() && onehot.is_contiguous() && s.is_contiguous(), "cat_st_fwd_o outputs");
+  TORCH_CHECK(onehot.scalar_type() == raw.scalar_type(), "onehot dtype must match raw");
+  int K = (int)raw.size(-1);
+  long nrows = raw.numel() / K;
+  const int rows_per_block = kBlock / 64;
+  int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, raw.scalar_type(), "cat_st_fwd_o", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((cat_st_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)raw.data_ptr(), urand.data_ptr<float>(), m.data_ptr<float>(),
+                       (T*)onehot.data_ptr(), s.data_ptr<float>(), nrows, K, (float)unimix);
+  });
+}
+
 torch::Tensor cat_st_bwd(const torch::Tensor& gm, const torch::Tensor& gon, const torch::Tensor& s,
                          double unimix) {
   CHECK_IN(gm);
@@ -851,6 +952,25 @@ torch::Tensor cat_st_bwd(const torch::Tensor& gm, const torch::Tensor& gon, cons
                        nrows, K, (float)unimix);
   });
   return graw;
+}
+
+// Out-variant for the fused scan backward: graw is a caller-provided slice.
+void cat_st_bwd_o(const torch::Tensor& gm, const torch::Tensor& gon, const torch::Tensor& s, double unimix,
+                  torch::Tensor graw) {
+  CHECK_IN(gm);
+  CHECK_IN(gon);
+  TORCH_CHECK(graw.is_contiguous() && graw.scalar_type() == gon.scalar_type(), "cat_st_bwd_o output");
+  int K = (int)gm.size(-1);
+  long nrows = gm.numel() / K;
+  const int rows_per_block = kBlock / 64;
+  int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, graw.scalar_type(), "cat_st_bwd_o", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((cat_st_bwd_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       gm.data_ptr<float>(), (const T*)gon.data_ptr(), s.data_ptr<float>(), (T*)graw.data_ptr(),
+                       nrows, K, (float)unimix);
+  });
 }
 
 // ---------------------------------------------------------------------------
@@ -885,6 +1005,44 @@ __global__ void masked_lerp_bwd_kernel(const T* __restrict__ g, const T* __restr
   }
 }
 
+// Strided variant for the fused scan: x and y may be row-strided 2-D views
+// (slices of stacked [T, B, *] buffers); init (when present) is contiguous.
+template <typename T, bool HAS_INIT>
+__global__ void masked_lerp_fwd_s_kernel(const T* __restrict__ x, const T* __restrict__ init,
+                                         const T* __restrict__ f, T* __restrict__ y, long rows, int cols,
+                                         long xs, long ys) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < rows * (long)cols;
+       i += (long)gridDim.x * blockDim.x) {
+    const long b = i / cols;
+    const long c = i - b * cols;
+    float fb = ld(f, b);
+    float v = (1.f - fb) * ld(x, b * xs + c);
+    if (HAS_INIT) v += fb * ld(init, i);
+    st(y, b * ys + c, v);
+  }
+}
+
+void masked_lerp_fwd_o(const torch::Tensor& x, const c10::optional<torch::Tensor>& init,
+                       const torch::Tensor& f, torch::Tensor y) {
+  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1 && y.dim() == 2 && y.stride(1) == 1, "masked_lerp_fwd_o shapes");
+  long rows = x.size(0);
+  int cols = (int)x.size(1);
+  long n = rows * (long)cols;
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "masked_lerp_fwd_o", [&] {
+    using T = scalar_t;
+    if (init.has_value())
+      hipLaunchKernelGGL((masked_lerp_fwd_s_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)x.data_ptr(), (const T*)init->data_ptr(), (const T*)f.data_ptr(),
+                         (T*)y.data_ptr(), rows, cols, x.stride(0), y.stride(0));
+    else
+      hipLaunchKernelGGL((masked_lerp_fwd_s_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)x.data_ptr(), nullptr, (const T*)f.data_ptr(), (T*)y.data_ptr(), rows, cols,
+                         x.stride(0), y.stride(0));
+  });
+}
+
 torch::Tensor masked_lerp_fwd(const torch::Tensor& x, const c10::optional<torch::Tensor>& init,
                               const torch::Tensor& f) {
   CHECK_IN(x);
@@ -904,6 +1062,47 @@ torch::Tensor masked_lerp_fwd(const torch::Tensor& x, const c10::optional<torch:
                          (const T*)x.data_ptr(), nullptr, (const T*)f.data_ptr(), (T*)y.data_ptr(), rows, cols);
   });
   return y;
+}
+
+// Accumulate-variant for the fused scan backward: g may be row-strided; gx is
+// a caller-provided contiguous output; gacc (optional) is a fp32 accumulator
+// of the same 2-D shape receiving the init-side gradient (f * g), summed
+// across scan steps without atomics (steps are stream-ordered).
+template <typename T, bool HAS_ACC>
+__global__ void masked_lerp_bwd_s_kernel(const T* __restrict__ g, const T* __restrict__ f, T* __restrict__ gx,
+                                         float* __restrict__ gacc, long rows, int cols, long gs) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < rows * (long)cols;
+       i += (long)gridDim.x * blockDim.x) {
+    const long b = i / cols;
+    const long c = i - b * cols;
+    float fb = ld(f, b);
+    float gv = ld(g, b * gs + c);
+    st(gx, i, (1.f - fb) * gv);
+    if (HAS_ACC) gacc[i] += fb * gv;
+  }
+}
+
+void masked_lerp_bwd_acc(const torch::Tensor& g, const torch::Tensor& f, torch::Tensor gx,
+                         const c10::optional<torch::Tensor>& gacc) {
+  TORCH_CHECK(g.dim() == 2 && g.stride(1) == 1 && gx.is_contiguous(), "masked_lerp_bwd_acc shapes");
+  long rows = g.size(0);
+  int cols = (int)g.size(1);
+  long n = rows * (long)cols;
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, g.scalar_type(), "masked_lerp_bwd_acc", [&] {
+    using T = scalar_t;
+    if (gacc.has_value()) {
+      TORCH_CHECK(gacc->scalar_type() == at::kFloat && gacc->is_contiguous(), "gacc must be contiguous fp32");
+      hipLaunchKernelGGL((masked_lerp_bwd_s_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)g.data_ptr(), (const T*)f.data_ptr(), (T*)gx.data_ptr(),
+                         gacc->data_ptr<float>(), rows, cols, g.stride(0));
+    } else {
+      hipLaunchKernelGGL((masked_lerp_bwd_s_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)g.data_ptr(), (const T*)f.data_ptr(), (T*)gx.data_ptr(), nullptr, rows, cols,
+                         g.stride(0));
+    }
+  });
 }
 
 std::vector<torch::Tensor> masked_lerp_bwd(const torch::Tensor& g, const torch::Tensor& f, bool has_init) {
@@ -1042,6 +1241,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cat_st_bwd", &cat_st_bwd);
   m.def("masked_lerp_fwd", &masked_lerp_fwd);
   m.def("masked_lerp_bwd", &masked_lerp_bwd);
+  // fused-scan variants (strided views / caller-provided outputs+accumulators)
+  m.def("ln_act_fwd_o", &ln_act_fwd_o);
+  m.def("ln_act_bwd_acc", &ln_act_bwd_acc);
+  m.def("gru_gates_fwd_o", &gru_gates_fwd_o);
+  m.def("gru_gates_bwd_acc", &gru_gates_bwd_acc);
+  m.def("masked_lerp_fwd_o", &masked_lerp_fwd_o);
+  m.def("masked_lerp_bwd_acc", &masked_lerp_bwd_acc);
+  m.def("cat_st_fwd_o", &cat_st_fwd_o);
+  m.def("cat_st_bwd_o", &cat_st_bwd_o);
   m.def("ema_update", &ema_update);
   m.def("obs_norm", &obs_norm);
 }

@@ -3,8 +3,20 @@
 The DV3 dynamic-learning scan is T=64 sequential steps; under autograd each
 step leaves ~40 bookkeeping kernels in the backward (per-use gradient
 accumulation adds on every weight, cat-backward copies, zero fills).  This
-Function runs the whole scan with explicit kernels and accumulates weight
-gradients in-place with `addmm_` — one GEMM per weight per step, nothing else.
+Function runs the whole scan with explicit kernels and a launch-lean layout:
+
+* every per-step intermediate is written DIRECTLY into its slice of a stacked
+  [T, B, *] buffer (strided-out kernel variants, ``mm(..., out=)``) — no
+  per-step stacking copies, no ``torch.cat`` (the concatenated GEMM inputs
+  are assembled in place: resets write into ``x_s``/``hu_s`` column blocks,
+  the GRU writes ``h`` into both ``h_seq`` and the representation input
+  ``r_s`` whose embed columns are filled once, batched, before the loop);
+* the backward reverse loop stores per-step output gradients into stacked
+  buffers and accumulates LN affine / init-state gradients into fp32 buffers
+  INSIDE the kernels (``*_acc`` variants) — zero per-step fills/casts/adds;
+* all weight gradients are then computed AFTER the loop as four big
+  [out, T*B] x [T*B, in] GEMMs (MFMA-sized) instead of 4x64 latency-bound
+  rank-B ``addmm_`` updates.
 
 Per step (math identical to RSSM.dynamic_posterior + RecurrentModel +
 LayerNormGRUCell + representation MLP + the fused categorical-ST head):
@@ -66,152 +78,140 @@ class _RSSMScan(torch.autograd.Function):
         h_seq = torch.empty(T, B, H, device=dev, dtype=dt)
         z_seq = torch.empty(T, B, SK, device=dev, dtype=dt)
         m_seq = torch.empty(T, B, SK, device=dev, dtype=torch.float32)
-        # saved intermediates
-        x_s = torch.empty(T, B, SK + A, device=dev, dtype=dt)
+        # stacked intermediates, written in place by the step kernels
+        x_s = torch.empty(T, B, SK + A, device=dev, dtype=dt)     # [z', a']
         g1_s = torch.empty(T, B, D, device=dev, dtype=dt)
-        m1_s = torch.empty(T, B, device=dev, dtype=torch.float32)
-        r1_s = torch.empty(T, B, device=dev, dtype=torch.float32)
-        hin_s = torch.empty(T, B, H, device=dev, dtype=dt)
-        hu_s = torch.empty(T, B, H + D, device=dev, dtype=dt)
+        mr1_s = torch.empty(2, T, B, device=dev, dtype=torch.float32)
+        hu_s = torch.empty(T, B, H + D, device=dev, dtype=dt)     # [h', u]
         y_s = torch.empty(T, B, 3 * H, device=dev, dtype=dt)
-        mg_s = torch.empty(T, B, device=dev, dtype=torch.float32)
-        rg_s = torch.empty(T, B, device=dev, dtype=torch.float32)
+        mrg_s = torch.empty(2, T, B, device=dev, dtype=torch.float32)
+        r_s = torch.empty(T, B, H + E, device=dev, dtype=dt)      # [h, embed]
+        mr3_s = torch.empty(2, T, B, device=dev, dtype=torch.float32)
         g3_s = torch.empty(T, B, P, device=dev, dtype=dt)
-        m3_s = torch.empty(T, B, device=dev, dtype=torch.float32)
-        r3_s = torch.empty(T, B, device=dev, dtype=torch.float32)
         p_s = torch.empty(T, B, P, device=dev, dtype=dt)
         s_s = torch.empty(T, B, S, discrete, device=dev, dtype=torch.float32)
+        raw = torch.empty(B, SK, device=dev, dtype=dt)
 
+        r_s[:, :, H:] = embed                 # batched, once
+        f_all = is_first.to(dt).reshape(T, B)
         ih = init_h[0]
         iz = init_z[0]
-        h = torch.zeros(B, H, device=dev, dtype=dt)
-        z = torch.zeros(B, SK, device=dev, dtype=dt)
+        h0 = torch.zeros(B, H, device=dev, dtype=dt)
+        z0 = torch.zeros(B, SK, device=dev, dtype=dt)
+        w1t, w2t, w3t, w4t = w1.t(), w2.t(), w3.t(), w4.t()
         for t in range(T):
-            f = is_first[t].to(dt)
-            a_eff = ext.masked_lerp_fwd(actions[t].contiguous().to(dt), None, f.reshape(-1))
-            h_in = ext.masked_lerp_fwd(h, ih, f.reshape(-1))
-            z_in = ext.masked_lerp_fwd(z, iz, f.reshape(-1))
-            x = torch.cat((z_in, a_eff), dim=-1)
-            g1 = x @ w1.t()
-            u, m1, r1 = ext.ln_act_fwd(g1, lnw1, lnb1, eps, True)
-            hu = torch.cat((h_in, u), dim=-1)
-            y = hu @ w2.t()
-            h, mg, rg = ext.gru_gates_fwd(y, h_in, lnwg, lnbg, eps)
-            r = torch.cat((h, embed[t]), dim=-1)
-            g3 = r @ w3.t()
-            p, m3, r3 = ext.ln_act_fwd(g3, lnw3, lnb3, eps, True)
-            raw = torch.addmm(b4, p, w4.t())
+            f = f_all[t]
+            h_prev = h_seq[t - 1] if t > 0 else h0
+            z_prev = z_seq[t - 1] if t > 0 else z0
+            ext.masked_lerp_fwd_o(actions[t], None, f, x_s[t, :, SK:])
+            ext.masked_lerp_fwd_o(z_prev, iz, f, x_s[t, :, :SK])
+            ext.masked_lerp_fwd_o(h_prev, ih, f, hu_s[t, :, :H])
+            torch.mm(x_s[t], w1t, out=g1_s[t])
+            ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
+            torch.mm(hu_s[t], w2t, out=y_s[t])
+            ext.gru_gates_fwd_o(y_s[t], hu_s[t, :, :H], lnwg, lnbg, eps,
+                                h_seq[t], r_s[t, :, :H], mrg_s[0, t], mrg_s[1, t])
+            torch.mm(r_s[t], w3t, out=g3_s[t])
+            ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
+            torch.addmm(b4, p_s[t], w4t, out=raw)
+            # per-step rand keeps the philox stream identical to the module
+            # loop (one rand of B*S*K values per step)
             urand = torch.rand(B, S, discrete, device=dev, dtype=torch.float32)
-            m, onehot, s = ext.cat_st_fwd(raw.view(B, S, discrete), urand, unimix, True)
-            z = onehot.view(B, SK).to(dt)
-
-            h_seq[t] = h
-            z_seq[t] = z
-            m_seq[t] = m.view(B, SK)
-            x_s[t] = x
-            g1_s[t] = g1
-            m1_s[t], r1_s[t] = m1, r1
-            hin_s[t] = h_in
-            hu_s[t] = hu
-            y_s[t] = y
-            mg_s[t], rg_s[t] = mg, rg
-            g3_s[t] = g3
-            m3_s[t], r3_s[t] = m3, r3
-            p_s[t] = p
-            s_s[t] = s
+            ext.cat_st_fwd_o(raw.view(B, S, discrete), urand, unimix,
+                             m_seq[t].view(B, S, discrete), z_seq[t].view(B, S, discrete), s_s[t])
 
         ctx.save_for_backward(
-            embed, actions, is_first, init_h, init_z,
-            w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4, b4,
-            h_seq, z_seq, x_s, g1_s, m1_s, r1_s, hin_s, hu_s, y_s, mg_s, rg_s,
-            g3_s, m3_s, r3_s, p_s, s_s,
+            f_all, w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4,
+            x_s, g1_s, mr1_s, hu_s, y_s, mrg_s, r_s, g3_s, mr3_s, p_s, s_s,
         )
         ctx.dims = (T, B, E, A, H, SK, D, P, S, discrete)
         ctx.unimix = unimix
-        ctx.eps = eps
         return h_seq, z_seq, m_seq
 
     @staticmethod
     def backward(ctx, g_h_seq: Tensor, g_z_seq: Tensor, g_m_seq: Tensor):
         ext = require_ext()
         (
-            embed, actions, is_first, init_h, init_z,
-            w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4, b4,
-            h_seq, z_seq, x_s, g1_s, m1_s, r1_s, hin_s, hu_s, y_s, mg_s, rg_s,
-            g3_s, m3_s, r3_s, p_s, s_s,
+            f_all, w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4,
+            x_s, g1_s, mr1_s, hu_s, y_s, mrg_s, r_s, g3_s, mr3_s, p_s, s_s,
         ) = ctx.saved_tensors
         (T, B, E, A, H, SK, D, P, S, discrete) = ctx.dims
-        unimix, eps = ctx.unimix, ctx.eps
-        dt = embed.dtype
+        unimix = ctx.unimix
+        dev = x_s.device
+        dt = x_s.dtype
+        g_h_seq = g_h_seq.contiguous()
+        g_z_seq = g_z_seq.contiguous()
+        g_m_seq = g_m_seq.contiguous()
 
-        gW1 = torch.zeros_like(w1)
-        glnw1 = torch.zeros_like(lnw1)
-        glnb1 = torch.zeros_like(lnb1)
-        gW2 = torch.zeros_like(w2)
-        glnwg = torch.zeros_like(lnwg)
-        glnbg = torch.zeros_like(lnbg)
-        gW3 = torch.zeros_like(w3)
-        glnw3 = torch.zeros_like(lnw3)
-        glnb3 = torch.zeros_like(lnb3)
-        gW4 = torch.zeros_like(w4)
-        gb4 = torch.zeros_like(b4)
-        g_embed = torch.empty_like(embed)
-        g_actions = torch.empty_like(actions)
-        g_init_h = torch.zeros_like(init_h)
-        g_init_z = torch.zeros_like(init_z)
+        # stacked per-step output grads (feed the batched weight-grad GEMMs)
+        graw_s = torch.empty(T, B, SK, device=dev, dtype=dt)
+        gg3_s = torch.empty(T, B, P, device=dev, dtype=dt)
+        gy_s = torch.empty(T, B, 3 * H, device=dev, dtype=dt)
+        gg1_s = torch.empty(T, B, D, device=dev, dtype=dt)
+        gr_s = torch.empty(T, B, H + E, device=dev, dtype=dt)
+        g_actions = torch.empty(T, B, A, device=dev, dtype=dt)
+        # fp32 in-kernel accumulators, one zero-fill for all of them
+        accs = torch.zeros(2 * D + 6 * H + 2 * P + B * (H + SK), device=dev, dtype=torch.float32)
+        off = 0
+        def _take(n):
+            nonlocal off
+            out = accs.narrow(0, off, n)
+            off += n
+            return out
+        glnw1, glnb1 = _take(D), _take(D)
+        glnwg, glnbg = _take(3 * H), _take(3 * H)
+        glnw3, glnb3 = _take(P), _take(P)
+        gih_acc = _take(B * H).view(B, H)
+        giz_acc = _take(B * SK).view(B, SK)
+        # reused per-step work buffers
+        gz_buf = torch.empty(B, SK, device=dev, dtype=dt)
+        gp = torch.empty(B, P, device=dev, dtype=dt)
+        gh_total = torch.empty(B, H, device=dev, dtype=dt)
+        ghp = torch.empty(B, H, device=dev, dtype=dt)
+        ghu = torch.empty(B, H + D, device=dev, dtype=dt)
+        gh_in = torch.empty(B, H, device=dev, dtype=dt)
+        gx = torch.empty(B, SK + A, device=dev, dtype=dt)
+        gh_carry = torch.zeros(B, H, device=dev, dtype=dt)
+        gz_carry = torch.zeros(B, SK, device=dev, dtype=dt)
 
-        gh_carry = torch.zeros(B, H, device=embed.device, dtype=dt)
-        gz_carry = torch.zeros(B, SK, device=embed.device, dtype=dt)
         for t in range(T - 1, -1, -1):
-            f = is_first[t].to(dt).reshape(-1)
-            gm_t = g_m_seq[t].contiguous().view(B, S, discrete)
-            gz_t = (g_z_seq[t] + gz_carry).view(B, S, discrete)
-            graw = ext.cat_st_bwd(gm_t, gz_t.to(dt), s_s[t], unimix).view(B, SK)
-            # W4 / b4
-            gp = graw @ w4
-            gW4.addmm_(graw.t(), p_s[t])
-            gb4.add_(graw.sum(0))
-            # ln3
-            gg3, gw3_, gb3_ = ext.ln_act_bwd(gp.contiguous(), g3_s[t], lnw3, lnb3, m3_s[t], r3_s[t], True)
-            glnw3.add_(gw3_)
-            glnb3.add_(gb3_)
-            # W3 over r = [h_t, embed_t]
-            gr = gg3 @ w3
-            r = torch.cat((h_seq[t], embed[t]), dim=-1)
-            gW3.addmm_(gg3.t(), r)
-            gh_total = g_h_seq[t] + gh_carry + gr[:, :H]
-            g_embed[t] = gr[:, H:]
-            # GRU gates
-            gy, gh_in1, gwg_, gbg_ = ext.gru_gates_bwd(
-                gh_total.contiguous(), y_s[t], hin_s[t], lnwg, lnbg, mg_s[t], rg_s[t]
-            )
-            glnwg.add_(gwg_)
-            glnbg.add_(gbg_)
-            # W2 over hu = [h', u]
-            ghu = gy @ w2
-            gW2.addmm_(gy.t(), hu_s[t])
-            gh_in = ghu[:, :H] + gh_in1
-            gu = ghu[:, H:]
-            # ln1
-            gg1, gw1_, gb1_ = ext.ln_act_bwd(gu.contiguous(), g1_s[t], lnw1, lnb1, m1_s[t], r1_s[t], True)
-            glnw1.add_(gw1_)
-            glnb1.add_(gb1_)
-            # W1 over x = [z', a']
-            gx = gg1 @ w1
-            gW1.addmm_(gg1.t(), x_s[t])
-            gz_in = gx[:, :SK]
-            ga_eff = gx[:, SK:]
-            # masked reset backward (carries flow to step t-1)
-            gh_carry, gih = ext.masked_lerp_bwd(gh_in.contiguous(), f, True)
-            g_init_h[0].add_(gih)
-            gz_carry, giz = ext.masked_lerp_bwd(gz_in.contiguous(), f, True)
-            g_init_z[0].add_(giz)
-            ga, _ = ext.masked_lerp_bwd(ga_eff.contiguous(), f, False)
-            g_actions[t] = ga
+            f = f_all[t]
+            torch.add(g_z_seq[t], gz_carry, out=gz_buf)
+            ext.cat_st_bwd_o(g_m_seq[t].view(B, S, discrete), gz_buf.view(B, S, discrete), s_s[t],
+                             unimix, graw_s[t].view(B, S, discrete))
+            torch.mm(graw_s[t], w4, out=gp)
+            ext.ln_act_bwd_acc(gp, g3_s[t], lnw3, lnb3, mr3_s[0, t], mr3_s[1, t], True,
+                               gg3_s[t], glnw3, glnb3)
+            torch.mm(gg3_s[t], w3, out=gr_s[t])
+            torch.add(g_h_seq[t], gh_carry, out=gh_total)
+            gh_total.add_(gr_s[t, :, :H])
+            ext.gru_gates_bwd_acc(gh_total, y_s[t], hu_s[t, :, :H], lnwg, lnbg,
+                                  mrg_s[0, t], mrg_s[1, t], gy_s[t], ghp, glnwg, glnbg)
+            torch.mm(gy_s[t], w2, out=ghu)
+            torch.add(ghu[:, :H], ghp, out=gh_in)
+            ext.ln_act_bwd_acc(ghu[:, H:], g1_s[t], lnw1, lnb1, mr1_s[0, t], mr1_s[1, t], True,
+                               gg1_s[t], glnw1, glnb1)
+            torch.mm(gg1_s[t], w1, out=gx)
+            ext.masked_lerp_bwd_acc(gh_in, f, gh_carry, gih_acc)
+            ext.masked_lerp_bwd_acc(gx[:, :SK], f, gz_carry, giz_acc)
+            ext.masked_lerp_bwd_acc(gx[:, SK:], f, g_actions[t], None)
 
+        # batched weight grads: one MFMA GEMM per weight over all T*B rows
+        TB = T * B
+        gW4 = torch.mm(graw_s.view(TB, SK).t(), p_s.view(TB, P))
+        gb4 = graw_s.view(TB, SK).sum(0)
+        gW3 = torch.mm(gg3_s.view(TB, P).t(), r_s.view(TB, H + E))
+        gW2 = torch.mm(gy_s.view(TB, 3 * H).t(), hu_s.view(TB, H + D))
+        gW1 = torch.mm(gg1_s.view(TB, D).t(), x_s.view(TB, SK + A))
         return (
-            g_embed, g_actions, None, g_init_h, g_init_z,
-            gW1, glnw1, glnb1, gW2, glnwg, glnbg, gW3, glnw3, glnb3, gW4, gb4,
+            gr_s[:, :, H:],                       # g_embed (strided view is fine)
+            g_actions, None,
+            gih_acc.to(dt).view(1, B, H),
+            giz_acc.to(dt).view(1, B, SK),
+            gW1, glnw1.to(lnw1.dtype), glnb1.to(lnb1.dtype),
+            gW2, glnwg.to(lnwg.dtype), glnbg.to(lnbg.dtype),
+            gW3, glnw3.to(lnw3.dtype), glnb3.to(lnb3.dtype),
+            gW4, gb4,
             None, None, None,
         )
 
@@ -236,8 +236,8 @@ def rssm_scan(
     T, B = embed.shape[:2]
     dt = next(rssm.recurrent_model.parameters()).dtype
     return _RSSMScan.apply(
-        embed.to(dt),
-        actions.to(dt),
+        embed.to(dt).contiguous(),
+        actions.to(dt).contiguous(),
         is_first,
         init_h.reshape(1, B, -1).to(dt).contiguous(),
         init_z.reshape(1, B, -1).to(dt).contiguous(),
