@@ -26,6 +26,7 @@ from sheeprl_amd import ops
 from sheeprl_amd.distributions import OneHotCategoricalST, unimix_logits
 from sheeprl_amd.envs import spaces
 from sheeprl_amd.models import (
+    FastLinear,
     CNN,
     DeCNN,
     DenseBlock,
@@ -178,7 +179,7 @@ class CNNDecoder(nn.Module):
         self.output_channels = list(output_channels)
         self.output_dim = (sum(output_channels), *image_size)
         self.model = nn.Sequential(
-            nn.Linear(latent_state_size, cnn_encoder_output_dim),
+            FastLinear(latent_state_size, cnn_encoder_output_dim),
             nn.Unflatten(1, (-1, 4, 4)),
             DeCNN(
                 in_channels=(2 ** (stages - 1)) * channels_multiplier,
@@ -221,7 +222,7 @@ class MLPDecoder(nn.Module):
             layer_norm=layer_norm,
             layer_norm_eps=layer_norm_eps,
         )
-        self.heads = nn.ModuleList([nn.Linear(dense_units, dim) for dim in output_dims])
+        self.heads = nn.ModuleList([FastLinear(dense_units, dim) for dim in output_dims])
 
     def forward(self, latent_states: Tensor) -> Dict[str, Tensor]:
         x = self.model(latent_states)
@@ -454,9 +455,9 @@ class Actor(nn.Module):
             layer_norm_eps=layer_norm_eps,
         )
         if is_continuous:
-            self.mlp_heads = nn.ModuleList([nn.Linear(dense_units, int(sum(actions_dim)) * 2)])
+            self.mlp_heads = nn.ModuleList([FastLinear(dense_units, int(sum(actions_dim)) * 2)])
         else:
-            self.mlp_heads = nn.ModuleList([nn.Linear(dense_units, d) for d in actions_dim])
+            self.mlp_heads = nn.ModuleList([FastLinear(dense_units, d) for d in actions_dim])
         self.actions_dim = list(actions_dim)
         self.is_continuous = is_continuous
         self.init_std = init_std

@@ -60,6 +60,7 @@ class _RSSMScan(torch.autograd.Function):
         lnw3: Tensor, lnb3: Tensor,
         w4: Tensor,          # [SK, P]
         b4: Tensor,          # [SK]
+        urand_all,           # [T, B, S, K] fp32 or None (test parity hook)
         unimix: float,
         eps: float,
         discrete: int,
@@ -94,6 +95,10 @@ class _RSSMScan(torch.autograd.Function):
 
         r_s[:, :, H:] = embed                 # batched, once
         f_all = is_first.to(dt).reshape(T, B)
+        if urand_all is None:
+            # one batched philox launch; tests pass explicit per-step-stacked
+            # urand to match the module loop's rand stream exactly
+            urand_all = torch.rand(T, B, S, discrete, device=dev, dtype=torch.float32)
         ih = init_h[0]
         iz = init_z[0]
         h0 = torch.zeros(B, H, device=dev, dtype=dt)
@@ -114,10 +119,7 @@ class _RSSMScan(torch.autograd.Function):
             torch.mm(r_s[t], w3t, out=g3_s[t])
             ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
             torch.addmm(b4, p_s[t], w4t, out=raw)
-            # per-step rand keeps the philox stream identical to the module
-            # loop (one rand of B*S*K values per step)
-            urand = torch.rand(B, S, discrete, device=dev, dtype=torch.float32)
-            ext.cat_st_fwd_o(raw.view(B, S, discrete), urand, unimix,
+            ext.cat_st_fwd_o(raw.view(B, S, discrete), urand_all[t], unimix,
                              m_seq[t].view(B, S, discrete), z_seq[t].view(B, S, discrete), s_s[t])
 
         ctx.save_for_backward(
@@ -163,43 +165,42 @@ class _RSSMScan(torch.autograd.Function):
         glnw3, glnb3 = _take(P), _take(P)
         gih_acc = _take(B * H).view(B, H)
         giz_acc = _take(B * SK).view(B, SK)
-        # reused per-step work buffers
-        gz_buf = torch.empty(B, SK, device=dev, dtype=dt)
+        # reused per-step work buffers (incoming-gradient sums happen inside
+        # the kernels via their optional extra-source arguments)
         gp = torch.empty(B, P, device=dev, dtype=dt)
-        gh_total = torch.empty(B, H, device=dev, dtype=dt)
         ghp = torch.empty(B, H, device=dev, dtype=dt)
         ghu = torch.empty(B, H + D, device=dev, dtype=dt)
-        gh_in = torch.empty(B, H, device=dev, dtype=dt)
         gx = torch.empty(B, SK + A, device=dev, dtype=dt)
-        gh_carry = torch.zeros(B, H, device=dev, dtype=dt)
-        gz_carry = torch.zeros(B, SK, device=dev, dtype=dt)
+        gh_carry = torch.empty(B, H, device=dev, dtype=dt)   # written at t before read at t-1
+        gz_carry = torch.empty(B, SK, device=dev, dtype=dt)
 
         for t in range(T - 1, -1, -1):
             f = f_all[t]
-            torch.add(g_z_seq[t], gz_carry, out=gz_buf)
-            ext.cat_st_bwd_o(g_m_seq[t].view(B, S, discrete), gz_buf.view(B, S, discrete), s_s[t],
-                             unimix, graw_s[t].view(B, S, discrete))
+            zc = gz_carry.view(B, S, discrete) if t < T - 1 else None
+            hc = gh_carry if t < T - 1 else None
+            ext.cat_st_bwd_o(g_m_seq[t].view(B, S, discrete), g_z_seq[t].view(B, S, discrete), zc,
+                             s_s[t], unimix, graw_s[t].view(B, S, discrete))
             torch.mm(graw_s[t], w4, out=gp)
             ext.ln_act_bwd_acc(gp, g3_s[t], lnw3, lnb3, mr3_s[0, t], mr3_s[1, t], True,
                                gg3_s[t], glnw3, glnb3)
             torch.mm(gg3_s[t], w3, out=gr_s[t])
-            torch.add(g_h_seq[t], gh_carry, out=gh_total)
-            gh_total.add_(gr_s[t, :, :H])
-            ext.gru_gates_bwd_acc(gh_total, y_s[t], hu_s[t, :, :H], lnwg, lnbg,
+            ext.gru_gates_bwd_acc(g_h_seq[t], hc, gr_s[t, :, :H], y_s[t], hu_s[t, :, :H], lnwg, lnbg,
                                   mrg_s[0, t], mrg_s[1, t], gy_s[t], ghp, glnwg, glnbg)
             torch.mm(gy_s[t], w2, out=ghu)
-            torch.add(ghu[:, :H], ghp, out=gh_in)
             ext.ln_act_bwd_acc(ghu[:, H:], g1_s[t], lnw1, lnb1, mr1_s[0, t], mr1_s[1, t], True,
                                gg1_s[t], glnw1, glnb1)
             torch.mm(gg1_s[t], w1, out=gx)
-            ext.masked_lerp_bwd_acc(gh_in, f, gh_carry, gih_acc)
-            ext.masked_lerp_bwd_acc(gx[:, :SK], f, gz_carry, giz_acc)
-            ext.masked_lerp_bwd_acc(gx[:, SK:], f, g_actions[t], None)
+            ext.masked_lerp_bwd_acc(ghu[:, :H], ghp, f, gh_carry, gih_acc)
+            ext.masked_lerp_bwd_acc(gx[:, :SK], None, f, gz_carry, giz_acc)
+            ext.masked_lerp_bwd_acc(gx[:, SK:], None, f, g_actions[t], None)
 
         # batched weight grads: one MFMA GEMM per weight over all T*B rows
         TB = T * B
+        ones_row = torch.ones(1, TB, device=dev, dtype=dt)
         gW4 = torch.mm(graw_s.view(TB, SK).t(), p_s.view(TB, P))
-        gb4 = graw_s.view(TB, SK).sum(0)
+        # bias grad as a GEMM against ones: torch's bf16 column-reduce sum(0)
+        # was measured at ~300 us for this shape; the GEMV path is ~5 us.
+        gb4 = torch.mm(ones_row, graw_s.view(TB, SK)).view(SK)
         gW3 = torch.mm(gg3_s.view(TB, P).t(), r_s.view(TB, H + E))
         gW2 = torch.mm(gy_s.view(TB, 3 * H).t(), hu_s.view(TB, H + D))
         gW1 = torch.mm(gg1_s.view(TB, D).t(), x_s.view(TB, SK + A))
@@ -212,7 +213,7 @@ class _RSSMScan(torch.autograd.Function):
             gW2, glnwg.to(lnwg.dtype), glnbg.to(lnbg.dtype),
             gW3, glnw3.to(lnw3.dtype), glnb3.to(lnb3.dtype),
             gW4, gb4,
-            None, None, None,
+            None, None, None, None,
         )
 
 
@@ -222,6 +223,7 @@ def rssm_scan(
     actions: Tensor,
     is_first: Tensor,
     initial_states: Tuple[Tensor, Tensor],
+    urand: Tensor = None,
 ) -> Tuple[Tensor, Tensor, Tensor]:
     """Run the coupled RSSM posterior scan with the fused Function.
 
@@ -252,6 +254,7 @@ def rssm_scan(
         rep[0].ln_bias,
         rep[1].linear.weight,
         rep[1].linear.bias,
+        urand,
         float(rssm.unimix),
         1e-3,
         int(rssm.discrete),

@@ -329,7 +329,9 @@ def test_fused_rssm_scan_matches_module_loop():
 
     def fused():
         _ir, _ip = rssm.get_initial_states((1, B))
-        return rssm_scan(rssm, embed, actions, is_first, (_ir.contiguous(), _ip.contiguous()))
+        # replicate the module loop's philox stream: one rand(B*S*K) per step
+        urand = torch.stack([torch.rand(B, S, K, device="cuda") for _ in range(T)])
+        return rssm_scan(rssm, embed, actions, is_first, (_ir.contiguous(), _ip.contiguous()), urand=urand)
 
     gh = torch.randn(T, B, H, device="cuda")
     gz = torch.randn(T, B, SK, device="cuda")
