@@ -1066,14 +1066,16 @@ void adam_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> gra
 template <typename T, bool SAMPLE>
 __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __restrict__ urand,
                                   float* __restrict__ m_out, T* __restrict__ onehot, float* __restrict__ s_out,
-                                  long nrows, int K, float unimix) {
+                                  long nrows, int K, float unimix, long ohs, long spb) {
   const int lane = threadIdx.x & 63;
   const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (row >= nrows) return;
   const T* L = raw + row * (long)K;
   float* mr = m_out + row * (long)K;
   float* sr = s_out + row * (long)K;
-  T* oh = onehot + row * (long)K;
+  // onehot may be a [B, S*K] row-strided slice: row = b*spb + s maps to
+  // b*ohs + s*K (ohs = S*K, spb = S reproduces the contiguous layout)
+  T* oh = onehot + (row / spb) * ohs + (row % spb) * (long)K;
   // row max
   float lmax = -1e30f;
   for (int j = lane; j < K; j += 64) lmax = fmaxf(lmax, ld(L, j));
@@ -1166,11 +1168,11 @@ std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optio
     if (sample)
       hipLaunchKernelGGL((cat_st_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
                          (const T*)raw.data_ptr(), up, m.data_ptr<float>(), (T*)onehot.data_ptr(),
-                         s.data_ptr<float>(), nrows, K, (float)unimix);
+                         s.data_ptr<float>(), nrows, K, (float)unimix, (long)K, 1);
     else
       hipLaunchKernelGGL((cat_st_fwd_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
                          (const T*)raw.data_ptr(), up, m.data_ptr<float>(), (T*)onehot.data_ptr(),
-                         s.data_ptr<float>(), nrows, K, (float)unimix);
+                         s.data_ptr<float>(), nrows, K, (float)unimix, (long)K, 1);
   });
   return {m, onehot, s};
 }
@@ -1181,10 +1183,20 @@ std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optio
 void cat_st_fwd_o(const torch::Tensor& raw, const torch::Tensor& urand, double unimix, torch::Tensor m,
                   torch::Tensor onehot, torch::Tensor s) {
   CHECK_IN(raw);
-  TORCH_CHECK(m.is_contiguous() && onehot.is_contiguous() && s.is_contiguous(), "cat_st_fwd_o outputs");
+  TORCH_CHECK(m.is_contiguous() && s.is_contiguous(), "cat_st_fwd_o outputs");
   TORCH_CHECK(onehot.scalar_type() == raw.scalar_type(), "onehot dtype must match raw");
   int K = (int)raw.size(-1);
   long nrows = raw.numel() / K;
+  // onehot is either contiguous (any shape) or a 2-D [B, S*K] row-strided
+  // slice of a stacked buffer (fast imagination writes z straight into the
+  // trajectory tensor)
+  long ohs = (long)K, spb = 1;
+  if (!onehot.is_contiguous()) {
+    TORCH_CHECK(onehot.dim() == 2 && onehot.stride(1) == 1 && nrows % onehot.size(0) == 0,
+                "cat_st_fwd_o: onehot must be contiguous or a row-strided 2-D slice");
+    spb = nrows / onehot.size(0);
+    ohs = onehot.stride(0);
+  }
   const int rows_per_block = kBlock / 64;
   int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
@@ -1192,7 +1204,7 @@ void cat_st_fwd_o(const torch::Tensor& raw, const torch::Tensor& urand, double u
     using T = scalar_t;
     hipLaunchKernelGGL((cat_st_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
                        (const T*)raw.data_ptr(), urand.data_ptr<float>(), m.data_ptr<float>(),
-                       (T*)onehot.data_ptr(), s.data_ptr<float>(), nrows, K, (float)unimix);
+                       (T*)onehot.data_ptr(), s.data_ptr<float>(), nrows, K, (float)unimix, ohs, spb);
   });
 }
 

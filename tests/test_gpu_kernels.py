@@ -358,3 +358,59 @@ def test_fused_rssm_scan_matches_module_loop():
         assert torch.allclose(p.grad, g1, atol=2e-3, rtol=2e-3), (
             n, (p.grad - g1).abs().max().item()
         )
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_fast_imagination_matches_module_loop():
+    """imagine.imagine_rollout vs the module-based imagination loop:
+    identical trajectories/actions given the same uniforms (fp32)."""
+    from sheeprl_amd.algos.dreamer_v3.agent import RSSM, RecurrentModel, Actor
+    from sheeprl_amd.algos.dreamer_v3.imagine import imagine_applicable, imagine_rollout
+    from sheeprl_amd.models import MLP
+
+    B, A, H, S, K, DU, P, HZ = 7, 5, 16, 4, 4, 16, 20, 6
+    SK = S * K
+    torch.manual_seed(3)
+    rssm = RSSM(
+        RecurrentModel(SK + A, H, DU),
+        MLP(24 + H, SK, [P], activation="silu", layer_norm=True, layer_norm_eps=1e-3),
+        MLP(H, SK, [P], activation="silu", layer_norm=True, layer_norm_eps=1e-3),
+        discrete=K,
+        unimix=0.01,
+    ).cuda()
+    actor = Actor(SK + H, [A], False, dense_units=DU, mlp_layers=2, unimix=0.01).cuda()
+    assert imagine_applicable(rssm, actor)
+
+    z0 = torch.rand(B, SK, device="cuda")
+    h0 = torch.randn(B, H, device="cuda")
+
+    # module-loop reference (same structure as dreamer_v3.train's fallback)
+    torch.manual_seed(11)
+    with torch.no_grad():
+        prior = z0.view(1, B, SK)
+        rec = h0.view(1, B, H)
+        latent = torch.cat((prior, rec), -1)
+        traj_ref = [latent[0]]
+        act = torch.cat(actor(latent)[0], dim=-1)
+        acts_ref = [act[0]]
+        for i in range(1, HZ + 1):
+            prior, rec = rssm.imagination(prior, rec, act)
+            prior = prior.view(1, B, SK)
+            latent = torch.cat((prior, rec), -1)
+            traj_ref.append(latent[0])
+            act = torch.cat(actor(latent)[0], dim=-1)
+            acts_ref.append(act[0])
+    traj_ref = torch.stack(traj_ref)
+    acts_ref = torch.stack(acts_ref)
+
+    # replicate the module loop's philox consumption order
+    torch.manual_seed(11)
+    ua, ut = [torch.rand(1, B, A, device="cuda")[0]], []
+    for i in range(HZ):
+        ut.append(torch.rand(B, S, K, device="cuda"))
+        ua.append(torch.rand(1, B, A, device="cuda")[0])
+    traj, acts = imagine_rollout(rssm, actor, z0, h0, HZ,
+                                 urand_t=torch.stack(ut), urand_a=torch.stack(ua))
+    assert torch.allclose(traj, traj_ref, atol=1e-4, rtol=1e-4), (traj - traj_ref).abs().max()
+    assert torch.equal(acts, acts_ref)
