@@ -1597,6 +1597,532 @@ torch::Tensor chlast_bias_sum(const torch::Tensor& g, long C) {
   return out;
 }
 
+// ---------------------------------------------------------------------------
+// persistent fused RSSM scan — building blocks
+// ---------------------------------------------------------------------------
+// MFMA fragment types for v_mfma_f32_16x16x32_bf16 (gfx950): 8 bf16 per lane
+// for A/B (A: row=lane&15, k=(lane>>4)*8+e; B: col=lane&15, same k), 4 fp32
+// accumulators per lane (C/D: col=lane&15, row=(lane>>4)*4+reg).
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+// Micro-kernel used by the layout unit test: Y[16,N] = X[16,K] @ W[N,K]^T.
+__global__ void pk_gemm16_test_kernel(const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ W,
+                                      float* __restrict__ Y, int N, int K) {
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int ncol0 = (blockIdx.x * (int)(blockDim.x >> 6) + wv) * 16;
+  if (ncol0 >= N) return;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  for (int k0 = 0; k0 < K; k0 += 32) {
+    bf16x8 a = *(const bf16x8*)(X + (long)arow * K + k0 + kgrp * 8);
+    bf16x8 b = *(const bf16x8*)(W + (long)(ncol0 + arow) * K + k0 + kgrp * 8);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) Y[(long)(kgrp * 4 + r) * N + ncol0 + (lane & 15)] = acc[r];
+}
+
+torch::Tensor pk_gemm16_test(const torch::Tensor& x, const torch::Tensor& w) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && w.is_contiguous());
+  TORCH_CHECK(x.size(0) == 16 && x.size(1) == w.size(1) && w.size(1) % 32 == 0 && w.size(0) % 64 == 0);
+  int K = (int)x.size(1), N = (int)w.size(0);
+  auto y = torch::empty({16, N}, x.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(pk_gemm16_test_kernel, dim3(N / 64), dim3(256), 0, stream.stream(),
+                     (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)w.data_ptr(),
+                     y.data_ptr<float>(), N, K);
+  return y;
+}
+
+// grid-wide barrier for persistent kernels: all workgroups must be resident
+// (grid <= #CUs).  counter/gen live in a small global workspace.
+__device__ __forceinline__ void pk_grid_barrier(int* counter, volatile int* gen, int nwg) {
+  __syncthreads();
+  __threadfence();
+  if (threadIdx.x == 0) {
+    const int g = *gen;
+    if (atomicAdd(counter, 1) == nwg - 1) {
+      *counter = 0;
+      __threadfence();
+      atomicAdd((int*)gen, 1);
+    } else {
+      while (*gen == g) __builtin_amdgcn_s_sleep(1);
+    }
+  }
+  __syncthreads();
+  __threadfence();
+}
+
+// barrier smoke test: NWG workgroups increment a per-round slot ITERS times;
+// result[i] must equal nwg for every round.
+__global__ void pk_barrier_test_kernel(int* ws, int* out, int nwg, int iters) {
+  for (int i = 0; i < iters; ++i) {
+    if (threadIdx.x == 0) atomicAdd(&out[i], 1);
+    pk_grid_barrier(ws, (volatile int*)(ws + 1), nwg);
+    // after the barrier every WG must see the full count of round i
+    if (threadIdx.x == 0 && out[i] != nwg) out[iters] = 1 + i;  // flag failure
+    pk_grid_barrier(ws, (volatile int*)(ws + 1), nwg);
+  }
+}
+
+torch::Tensor pk_barrier_test(long nwg, long iters) {
+  auto ws = torch::zeros({2}, torch::TensorOptions().dtype(at::kInt).device(at::kCUDA));
+  auto out = torch::zeros({iters + 1}, torch::TensorOptions().dtype(at::kInt).device(at::kCUDA));
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(pk_barrier_test_kernel, dim3((int)nwg), dim3(256), 0, stream.stream(),
+                     ws.data_ptr<int>(), out.data_ptr<int>(), (int)nwg, (int)iters);
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// Persistent fused RSSM scan FORWARD: the whole T-step recurrence in ONE
+// kernel launch.  The per-step launch sequence (3 masked-lerp, 3-4 GEMMs,
+// 2 LN, GRU gates, categorical-ST ~ 10 kernels x ~5 us latency each) is
+// replaced by grid-resident workgroups that cycle through five phases per
+// step, separated by a device-wide barrier (~2.7 us):
+//   PA: G1 = X @ W1^T            (X assembled inline from the episode resets)
+//   PB: Y  = [h', silu(LN(G1))] @ W2^T
+//   PB2: GRU gate math -> h_t    (elementwise, needs the full-row LN sums)
+//   PC: G3 = [h_t, embed] @ W3^T
+//   PD: RAW = silu(LN(G3)) @ W4^T + b4 -> unimix softmax + gumbel one-hot
+// GEMMs run on v_mfma_f32_16x16x32_bf16 with M=16 row tiles (the DV3 batch),
+// one 16-col tile per wave, 64-col tile per workgroup; weights stream from
+// L2.  All intermediates are written to the SAME stacked [T, B, *] buffers
+// the multi-kernel scan uses, so the hand-written backward is unchanged.
+// bf16-only, B <= 16; the Python wrapper falls back to the multi-kernel path
+// otherwise.
+
+struct PkParams {
+  const __hip_bfloat16 *actions, *f_all, *ih, *iz;
+  const __hip_bfloat16 *w1, *lnw1, *lnb1, *w2, *lnwg, *lnbg, *w3, *lnw3, *lnb3, *w4, *b4;
+  const float* urand;
+  __hip_bfloat16 *x_s, *g1_s, *hu_s, *y_s, *r_s, *g3_s, *p_s, *h_seq, *z_seq;
+  float *mr1, *mrg, *mr3, *m_seq, *s_s;  // mr*: [2,T,B]
+  float* ws;   // [2 slots][3 lns][2 sums][16 rows]
+  int* ibar;   // {counter, gen}
+  int T, B, A, H, SK, D, P, E, K1p, S, KD, nwg;
+  float unimix, eps;
+};
+
+__device__ __forceinline__ float pk_b2f(__hip_bfloat16 v) { return __bfloat162float(v); }
+__device__ __forceinline__ float pk_h2f(__bf16 v) { return (float)v; }
+
+// guarded, lerp-transformed load of X[m][k0..k0+7] (the G1 GEMM A operand):
+// X = [ (1-f)z_prev + f*iz , (1-f)a ] padded with zeros to K1p
+__device__ bf16x8 pk_load_x(const PkParams& P, int t, int m, int k0) {
+  bf16x8 out;
+  if (m >= P.B) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) out[e] = (__bf16)0.f;
+    return out;
+  }
+  const float f = pk_b2f(P.f_all[t * P.B + m]);
+  if (k0 + 8 <= P.SK) {
+    bf16x8 zi = *(const bf16x8*)(P.iz + (long)m * P.SK + k0);
+    if (t == 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) out[e] = (__bf16)(f * pk_h2f(zi[e]));
+    } else {
+      bf16x8 zp = *(const bf16x8*)(P.z_seq + ((long)(t - 1) * P.B + m) * P.SK + k0);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) out[e] = (__bf16)((1.f - f) * pk_h2f(zp[e]) + f * pk_h2f(zi[e]));
+    }
+    return out;
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int k = k0 + e;
+    float v = 0.f;
+    if (k < P.SK) {
+      float zp = t > 0 ? pk_b2f(P.z_seq[((long)(t - 1) * P.B + m) * P.SK + k]) : 0.f;
+      v = (1.f - f) * zp + f * pk_b2f(P.iz[(long)m * P.SK + k]);
+    } else if (k < P.SK + P.A) {
+      v = (1.f - f) * pk_b2f(P.actions[((long)t * P.B + m) * P.A + (k - P.SK)]);
+    }
+    out[e] = (__bf16)v;
+  }
+  return out;
+}
+
+// silu(LN(x)) for one element given the row mean / rstd
+__device__ __forceinline__ float pk_silu_ln(float x, float mean, float rstd, float w, float b) {
+  float z = (x - mean) * rstd * w + b;
+  return z / (1.f + expf(-z));
+}
+
+__global__ void __launch_bounds__(256) pk_scan_fwd_kernel(PkParams P) {
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  const int wg = blockIdx.x;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int tb = P.T * P.B;
+  extern __shared__ __attribute__((aligned(16))) float lds[];  // [16][64] raw tile (PD)
+
+  for (int t = 0; t < P.T; ++t) {
+    const int slot = t & 1;
+    float* acc_g1 = P.ws + ((slot * 3 + 0) * 2) * 16;
+    float* acc_y = P.ws + ((slot * 3 + 1) * 2) * 16;
+    float* acc_g3 = P.ws + ((slot * 3 + 2) * 2) * 16;
+
+    // ---------------- PA: G1 = X @ W1^T ----------------
+    if (wg < P.D / 64) {
+      const int ncol0 = wg * 64 + wv * 16;
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int k0 = 0; k0 < P.K1p; k0 += 32) {
+        bf16x8 a = pk_load_x(P, t, arow, k0 + kgrp * 8);
+        bf16x8 b;
+        {
+          const int wrow = ncol0 + arow;
+          const int k = k0 + kgrp * 8;
+          if (k + 8 <= P.SK + P.A) {
+            b = *(const bf16x8*)(P.w1 + (long)wrow * (P.SK + P.A) + k);
+          } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              b[e] = (k + e < P.SK + P.A) ? (__bf16)pk_b2f(P.w1[(long)wrow * (P.SK + P.A) + k + e]) : (__bf16)0.f;
+          }
+        }
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      }
+      // store g1 tile + row partial sums
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = kgrp * 4 + r;
+        if (m < P.B) P.g1_s[((long)t * P.B + m) * P.D + ncol0 + arow] = (__hip_bfloat16)acc[r];
+        float v = acc[r];
+        float v2 = v * v;
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          v += __shfl_xor(v, off, 64);
+          v2 += __shfl_xor(v2, off, 64);
+        }
+        if (arow == 0 && m < P.B) {
+          atomicAdd(&acc_g1[m], v);
+          atomicAdd(&acc_g1[16 + m], v2);
+        }
+      }
+    } else if (wg == P.D / 64) {
+      // assemble h' into hu_s[t][:, :H] (needed by PB and saved for backward)
+      for (int i = threadIdx.x; i < P.B * P.H; i += blockDim.x) {
+        const int m = i / P.H, j = i - m * P.H;
+        const float f = pk_b2f(P.f_all[t * P.B + m]);
+        float hp = t > 0 ? pk_b2f(P.h_seq[((long)(t - 1) * P.B + m) * P.H + j]) : 0.f;
+        float v = (1.f - f) * hp + f * pk_b2f(P.ih[(long)m * P.H + j]);
+        P.hu_s[((long)t * P.B + m) * (P.H + P.D) + j] = (__hip_bfloat16)v;
+      }
+    } else if (wg == P.D / 64 + 1) {
+      // store x_s[t] for the backward pass
+      for (int i = threadIdx.x; i < P.B * (P.SK + P.A); i += blockDim.x) {
+        const int m = i / (P.SK + P.A), k = i - m * (P.SK + P.A);
+        const float f = pk_b2f(P.f_all[t * P.B + m]);
+        float v;
+        if (k < P.SK) {
+          float zp = t > 0 ? pk_b2f(P.z_seq[((long)(t - 1) * P.B + m) * P.SK + k]) : 0.f;
+          v = (1.f - f) * zp + f * pk_b2f(P.iz[(long)m * P.SK + k]);
+        } else {
+          v = (1.f - f) * pk_b2f(P.actions[((long)t * P.B + m) * P.A + (k - P.SK)]);
+        }
+        P.x_s[(long)t * P.B * (P.SK + P.A) + i] = (__hip_bfloat16)v;
+      }
+    }
+    pk_grid_barrier(P.ibar, (volatile int*)(P.ibar + 1), P.nwg);
+
+    // ---------------- PB: Y = HU @ W2^T ----------------
+    {
+      const float m_g1 = acc_g1[arow] / P.D;
+      const float var_g1 = acc_g1[16 + arow] / P.D - m_g1 * m_g1;
+      const float rs_g1 = rsqrtf(fmaxf(var_g1, 0.f) + P.eps);
+      if (wg == 0 && wv == 0 && lane < P.B) {
+        P.mr1[(long)t * P.B + lane] = acc_g1[lane] / P.D;
+        float mm = acc_g1[lane] / P.D;
+        float vv = acc_g1[16 + lane] / P.D - mm * mm;
+        P.mr1[(long)tb + t * P.B + lane] = rsqrtf(fmaxf(vv, 0.f) + P.eps);
+      }
+      if (wg < (3 * P.H) / 64) {
+        const int ncol0 = wg * 64 + wv * 16;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        const int KK = P.H + P.D;
+        for (int k0 = 0; k0 < KK; k0 += 32) {
+          const int k = k0 + kgrp * 8;
+          bf16x8 a;
+          if (arow >= P.B) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) a[e] = (__bf16)0.f;
+          } else if (k + 8 <= P.H) {
+            a = *(const bf16x8*)(P.hu_s + ((long)t * P.B + arow) * (long)KK + k);
+          } else {
+            // u region: silu(LN(g1)) computed inline (k >= H; H%32==0 so no straddle)
+            bf16x8 g = *(const bf16x8*)(P.g1_s + ((long)t * P.B + arow) * P.D + (k - P.H));
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              const int j = k - P.H + e;
+              a[e] = (__bf16)pk_silu_ln(pk_h2f(g[e]), m_g1, rs_g1, pk_b2f(P.lnw1[j]), pk_b2f(P.lnb1[j]));
+            }
+          }
+          bf16x8 b = *(const bf16x8*)(P.w2 + (long)(ncol0 + arow) * KK + k);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = kgrp * 4 + r;
+          if (m < P.B) P.y_s[((long)t * P.B + m) * (3 * P.H) + ncol0 + arow] = (__hip_bfloat16)acc[r];
+          float v = acc[r], v2 = v * v;
+#pragma unroll
+          for (int off = 1; off < 16; off <<= 1) {
+            v += __shfl_xor(v, off, 64);
+            v2 += __shfl_xor(v2, off, 64);
+          }
+          if (arow == 0 && m < P.B) {
+            atomicAdd(&acc_y[m], v);
+            atomicAdd(&acc_y[16 + m], v2);
+          }
+        }
+      }
+      // save u into hu_s[t][:, H:]
+      if (wg < P.D / 64) {
+        for (int i = threadIdx.x; i < P.B * 64; i += blockDim.x) {
+          const int m = i >> 6, j0 = wg * 64 + (i & 63);
+          const float mm = acc_g1[m] / P.D;
+          const float vv = acc_g1[16 + m] / P.D - mm * mm;
+          const float rs = rsqrtf(fmaxf(vv, 0.f) + P.eps);
+          float g = pk_b2f(P.g1_s[((long)t * P.B + m) * P.D + j0]);
+          P.hu_s[((long)t * P.B + m) * (P.H + P.D) + P.H + j0] =
+              (__hip_bfloat16)pk_silu_ln(g, mm, rs, pk_b2f(P.lnw1[j0]), pk_b2f(P.lnb1[j0]));
+        }
+      }
+    }
+    pk_grid_barrier(P.ibar, (volatile int*)(P.ibar + 1), P.nwg);
+
+    // ---------------- PB2: GRU gates -> h_t ----------------
+    if (wg < P.H / 64) {
+      for (int i = threadIdx.x; i < P.B * 64; i += blockDim.x) {
+        const int m = i >> 6, j = wg * 64 + (i & 63);
+        const int D3 = 3 * P.H;
+        const float mean = acc_y[m] / D3;
+        const float var = acc_y[16 + m] / D3 - mean * mean;
+        const float rs = rsqrtf(fmaxf(var, 0.f) + P.eps);
+        const __hip_bfloat16* yr = P.y_s + ((long)t * P.B + m) * D3;
+        float zr = (pk_b2f(yr[j]) - mean) * rs * pk_b2f(P.lnwg[j]) + pk_b2f(P.lnbg[j]);
+        float zc = (pk_b2f(yr[P.H + j]) - mean) * rs * pk_b2f(P.lnwg[P.H + j]) + pk_b2f(P.lnbg[P.H + j]);
+        float zu = (pk_b2f(yr[2 * P.H + j]) - mean) * rs * pk_b2f(P.lnwg[2 * P.H + j]) + pk_b2f(P.lnbg[2 * P.H + j]);
+        float r = 1.f / (1.f + expf(-zr));
+        float c = tanhf(r * zc);
+        float u = 1.f / (1.f + expf(-(zu - 1.f)));
+        float hp = pk_b2f(P.hu_s[((long)t * P.B + m) * (P.H + P.D) + j]);
+        float hv = u * c + (1.f - u) * hp;
+        P.h_seq[((long)t * P.B + m) * P.H + j] = (__hip_bfloat16)hv;
+        P.r_s[((long)t * P.B + m) * (P.H + P.E) + j] = (__hip_bfloat16)hv;
+      }
+      if (wg == 0 && wv == 0 && lane < P.B) {
+        const int D3 = 3 * P.H;
+        float mm = acc_y[lane] / D3;
+        float vv = acc_y[16 + lane] / D3 - mm * mm;
+        P.mrg[(long)t * P.B + lane] = mm;
+        P.mrg[(long)tb + t * P.B + lane] = rsqrtf(fmaxf(vv, 0.f) + P.eps);
+      }
+    } else if (wg == P.H / 64) {
+      // zero the other slot's accumulators for step t+1
+      const int other = (t + 1) & 1;
+      for (int i = threadIdx.x; i < 3 * 2 * 16; i += blockDim.x) P.ws[other * 96 + i] = 0.f;
+    }
+    pk_grid_barrier(P.ibar, (volatile int*)(P.ibar + 1), P.nwg);
+
+    // ---------------- PC: G3 = R @ W3^T ----------------
+    if (wg < P.P / 64) {
+      const int ncol0 = wg * 64 + wv * 16;
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      const int KK = P.H + P.E;
+      for (int k0 = 0; k0 < KK; k0 += 32) {
+        const int k = k0 + kgrp * 8;
+        bf16x8 a;
+        if (arow >= P.B) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) a[e] = (__bf16)0.f;
+        } else {
+          a = *(const bf16x8*)(P.r_s + ((long)t * P.B + arow) * (long)KK + k);
+        }
+        bf16x8 b = *(const bf16x8*)(P.w3 + (long)(ncol0 + arow) * KK + k);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = kgrp * 4 + r;
+        if (m < P.B) P.g3_s[((long)t * P.B + m) * P.P + ncol0 + arow] = (__hip_bfloat16)acc[r];
+        float v = acc[r], v2 = v * v;
+#pragma unroll
+        for (int off = 1; off < 16; off <<= 1) {
+          v += __shfl_xor(v, off, 64);
+          v2 += __shfl_xor(v2, off, 64);
+        }
+        if (arow == 0 && m < P.B) {
+          atomicAdd(&acc_g3[m], v);
+          atomicAdd(&acc_g3[16 + m], v2);
+        }
+      }
+    }
+    pk_grid_barrier(P.ibar, (volatile int*)(P.ibar + 1), P.nwg);
+
+    // ---------------- PD: RAW = p @ W4^T + b4 -> categorical ST ----------------
+    {
+      const float m_g3 = acc_g3[arow] / P.P;
+      const float var_g3 = acc_g3[16 + arow] / P.P - m_g3 * m_g3;
+      const float rs_g3 = rsqrtf(fmaxf(var_g3, 0.f) + P.eps);
+      if (wg == 0 && wv == 0 && lane < P.B) {
+        float mm = acc_g3[lane] / P.P;
+        float vv = acc_g3[16 + lane] / P.P - mm * mm;
+        P.mr3[(long)t * P.B + lane] = mm;
+        P.mr3[(long)tb + t * P.B + lane] = rsqrtf(fmaxf(vv, 0.f) + P.eps);
+      }
+      if (wg < P.P / 64) {
+        // save p for the backward pass
+        for (int i = threadIdx.x; i < P.B * 64; i += blockDim.x) {
+          const int m = i >> 6, j0 = wg * 64 + (i & 63);
+          const float mm = acc_g3[m] / P.P;
+          const float vv = acc_g3[16 + m] / P.P - mm * mm;
+          const float rs = rsqrtf(fmaxf(vv, 0.f) + P.eps);
+          float g = pk_b2f(P.g3_s[((long)t * P.B + m) * P.P + j0]);
+          P.p_s[((long)t * P.B + m) * P.P + j0] =
+              (__hip_bfloat16)pk_silu_ln(g, mm, rs, pk_b2f(P.lnw3[j0]), pk_b2f(P.lnb3[j0]));
+        }
+      }
+      if (wg < P.SK / 64) {
+        const int ncol0 = wg * 64 + wv * 16;
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        for (int k0 = 0; k0 < P.P; k0 += 32) {
+          const int k = k0 + kgrp * 8;
+          bf16x8 a;
+          if (arow >= P.B) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) a[e] = (__bf16)0.f;
+          } else {
+            bf16x8 g = *(const bf16x8*)(P.g3_s + ((long)t * P.B + arow) * P.P + k);
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              a[e] = (__bf16)pk_silu_ln(pk_h2f(g[e]), m_g3, rs_g3, pk_b2f(P.lnw3[k + e]), pk_b2f(P.lnb3[k + e]));
+          }
+          bf16x8 b = *(const bf16x8*)(P.w4 + (long)(ncol0 + arow) * P.P + k);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+        }
+        // raw tile (+bias) to LDS, then per-(row, group) categorical
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int m = kgrp * 4 + r;
+          lds[m * 64 + wv * 16 + arow] = acc[r] + pk_b2f(P.b4[ncol0 + arow]);
+        }
+        __syncthreads();
+        const int groups_per_wg = 64 / P.KD;  // KD divides 64
+        const int tasks = 16 * groups_per_wg;
+        for (int task = threadIdx.x; task < tasks; task += blockDim.x) {
+          const int m = task / groups_per_wg;
+          if (m >= P.B) continue;
+          const int gloc = task - m * groups_per_wg;
+          const int col0 = gloc * P.KD;                 // within the WG tile
+          const int gcol0 = wg * 64 + col0;             // absolute column
+          const int sgrp = gcol0 / P.KD;                // stochastic group index
+          float lmax = -1e30f;
+          for (int j = 0; j < P.KD; ++j) lmax = fmaxf(lmax, lds[m * 64 + col0 + j]);
+          float lsum = 0.f;
+          for (int j = 0; j < P.KD; ++j) lsum += expf(lds[m * 64 + col0 + j] - lmax);
+          const float inv = 1.f / lsum;
+          float best = -1e30f;
+          int best_j = 0;
+          const float* ur = P.urand + (((long)t * P.B + m) * P.S + sgrp) * P.KD;
+          float* mr = P.m_seq + ((long)t * P.B + m) * P.SK + gcol0;
+          float* sr = P.s_s + ((long)t * P.B + m) * P.SK + gcol0;
+          __hip_bfloat16* zr = P.z_seq + ((long)t * P.B + m) * P.SK + gcol0;
+          for (int j = 0; j < P.KD; ++j) {
+            float sv = expf(lds[m * 64 + col0 + j] - lmax) * inv;
+            float pv = (1.f - P.unimix) * sv + P.unimix / P.KD;
+            float mv = logf(pv);
+            sr[j] = sv;
+            mr[j] = mv;
+            float tt = fmaxf(-logf(fmaxf(ur[j], 1e-20f)), 1e-20f);
+            float score = mv - logf(tt);
+            if (score > best) {
+              best = score;
+              best_j = j;
+            }
+          }
+          for (int j = 0; j < P.KD; ++j) zr[j] = (__hip_bfloat16)(j == best_j ? 1.f : 0.f);
+        }
+        __syncthreads();
+      }
+    }
+    pk_grid_barrier(P.ibar, (volatile int*)(P.ibar + 1), P.nwg);
+  }
+}
+
+void pk_scan_fwd(const torch::Tensor& actions, const torch::Tensor& f_all, const torch::Tensor& ih,
+                 const torch::Tensor& iz, const torch::Tensor& w1, const torch::Tensor& lnw1,
+                 const torch::Tensor& lnb1, const torch::Tensor& w2, const torch::Tensor& lnwg,
+                 const torch::Tensor& lnbg, const torch::Tensor& w3, const torch::Tensor& lnw3,
+                 const torch::Tensor& lnb3, const torch::Tensor& w4, const torch::Tensor& b4,
+                 const torch::Tensor& urand, torch::Tensor x_s, torch::Tensor g1_s, torch::Tensor hu_s,
+                 torch::Tensor y_s, torch::Tensor r_s, torch::Tensor g3_s, torch::Tensor p_s,
+                 torch::Tensor h_seq, torch::Tensor z_seq, torch::Tensor mr1, torch::Tensor mrg,
+                 torch::Tensor mr3, torch::Tensor m_seq, torch::Tensor s_s, torch::Tensor ws,
+                 torch::Tensor ibar, double unimix, double eps) {
+  PkParams P;
+  P.T = (int)h_seq.size(0);
+  P.B = (int)h_seq.size(1);
+  P.H = (int)h_seq.size(2);
+  P.SK = (int)z_seq.size(2);
+  P.A = (int)actions.size(2);
+  P.D = (int)g1_s.size(2);
+  P.P = (int)g3_s.size(2);
+  P.E = (int)r_s.size(2) - P.H;
+  P.K1p = ((P.SK + P.A + 31) / 32) * 32;
+  P.KD = (int)s_s.size(3);
+  P.S = P.SK / P.KD;
+  P.unimix = (float)unimix;
+  P.eps = (float)eps;
+  TORCH_CHECK(P.B <= 16 && w1.scalar_type() == at::kBFloat16, "pk_scan_fwd: bf16, B<=16 only");
+  TORCH_CHECK(P.D % 64 == 0 && P.H % 64 == 0 && P.P % 64 == 0 && P.SK % 64 == 0, "pk dims");
+  TORCH_CHECK((P.H + P.D) % 32 == 0 && (P.H + P.E) % 32 == 0 && P.P % 32 == 0 && P.SK % 32 == 0, "pk K dims");
+  TORCH_CHECK(P.KD <= 64 && 64 % P.KD == 0, "pk discrete size");
+  int nwg = std::max({3 * P.H / 64, P.D / 64 + 2, P.H / 64 + 1, P.P / 64, P.SK / 64});
+  TORCH_CHECK(nwg <= 128, "pk grid too large for co-residency");
+  P.nwg = nwg;
+  P.actions = (const __hip_bfloat16*)actions.data_ptr();
+  P.f_all = (const __hip_bfloat16*)f_all.data_ptr();
+  P.ih = (const __hip_bfloat16*)ih.data_ptr();
+  P.iz = (const __hip_bfloat16*)iz.data_ptr();
+  P.w1 = (const __hip_bfloat16*)w1.data_ptr();
+  P.lnw1 = (const __hip_bfloat16*)lnw1.data_ptr();
+  P.lnb1 = (const __hip_bfloat16*)lnb1.data_ptr();
+  P.w2 = (const __hip_bfloat16*)w2.data_ptr();
+  P.lnwg = (const __hip_bfloat16*)lnwg.data_ptr();
+  P.lnbg = (const __hip_bfloat16*)lnbg.data_ptr();
+  P.w3 = (const __hip_bfloat16*)w3.data_ptr();
+  P.lnw3 = (const __hip_bfloat16*)lnw3.data_ptr();
+  P.lnb3 = (const __hip_bfloat16*)lnb3.data_ptr();
+  P.w4 = (const __hip_bfloat16*)w4.data_ptr();
+  P.b4 = (const __hip_bfloat16*)b4.data_ptr();
+  P.urand = urand.data_ptr<float>();
+  P.x_s = (__hip_bfloat16*)x_s.data_ptr();
+  P.g1_s = (__hip_bfloat16*)g1_s.data_ptr();
+  P.hu_s = (__hip_bfloat16*)hu_s.data_ptr();
+  P.y_s = (__hip_bfloat16*)y_s.data_ptr();
+  P.r_s = (__hip_bfloat16*)r_s.data_ptr();
+  P.g3_s = (__hip_bfloat16*)g3_s.data_ptr();
+  P.p_s = (__hip_bfloat16*)p_s.data_ptr();
+  P.h_seq = (__hip_bfloat16*)h_seq.data_ptr();
+  P.z_seq = (__hip_bfloat16*)z_seq.data_ptr();
+  P.mr1 = mr1.data_ptr<float>();
+  P.mrg = mrg.data_ptr<float>();
+  P.mr3 = mr3.data_ptr<float>();
+  P.m_seq = m_seq.data_ptr<float>();
+  P.s_s = s_s.data_ptr<float>();
+  P.ws = ws.data_ptr<float>();
+  P.ibar = ibar.data_ptr<int>();
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  hipLaunchKernelGGL(pk_scan_fwd_kernel, dim3(nwg), dim3(256), 16 * 64 * sizeof(float), stream.stream(), P);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("symlog_fwd", [](const torch::Tensor& x) { return symmath<0>(x, c10::nullopt); });
   m.def("symlog_bwd", [](const torch::Tensor& x, const torch::Tensor& g) { return symmath<1>(x, g); });
@@ -1628,4 +2154,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ema_update", &ema_update);
   m.def("obs_norm", &obs_norm);
   m.def("chlast_bias_sum", &chlast_bias_sum);
+  m.def("pk_gemm16_test", &pk_gemm16_test);
+  m.def("pk_barrier_test", &pk_barrier_test);
+  m.def("pk_scan_fwd", &pk_scan_fwd);
 }

@@ -37,6 +37,8 @@ from __future__ import annotations
 
 from typing import Any, Tuple
 
+import os
+
 import torch
 from torch import Tensor
 
@@ -101,6 +103,38 @@ class _RSSMScan(torch.autograd.Function):
             urand_all = torch.rand(T, B, S, discrete, device=dev, dtype=torch.float32)
         ih = init_h[0]
         iz = init_z[0]
+
+        # persistent-kernel path: the whole T-step recurrence in ONE launch
+        # (5 device-wide barriers per step instead of ~10 kernel launches);
+        # writes the exact same stacked buffers, so backward is shared.
+        pk_ok = (
+            dt == torch.bfloat16
+            and B <= 16
+            and w1.dtype == dt
+            and D % 64 == 0 and H % 64 == 0 and P % 64 == 0 and SK % 64 == 0
+            and (H + D) % 32 == 0 and (H + E) % 32 == 0
+            and discrete <= 64 and 64 % discrete == 0
+            and os.environ.get("SHEEPRL_AMD_PK", "1") == "1"
+        )
+        if pk_ok:
+            ws = torch.zeros(192, device=dev, dtype=torch.float32)
+            ibar = torch.zeros(2, device=dev, dtype=torch.int32)
+            ext.pk_scan_fwd(
+                actions.contiguous(), f_all.contiguous(), ih.contiguous(), iz.contiguous(),
+                w1.contiguous(), lnw1.contiguous(), lnb1.contiguous(),
+                w2.contiguous(), lnwg.contiguous(), lnbg.contiguous(),
+                w3.contiguous(), lnw3.contiguous(), lnb3.contiguous(),
+                w4.contiguous(), b4.contiguous(), urand_all,
+                x_s, g1_s, hu_s, y_s, r_s, g3_s, p_s, h_seq, z_seq,
+                mr1_s, mrg_s, mr3_s, m_seq, s_s, ws, ibar, unimix, eps,
+            )
+            ctx.save_for_backward(
+                f_all, w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4,
+                x_s, g1_s, mr1_s, hu_s, y_s, mrg_s, r_s, g3_s, mr3_s, p_s, s_s,
+            )
+            ctx.dims = (T, B, E, A, H, SK, D, P, S, discrete)
+            ctx.unimix = unimix
+            return h_seq, z_seq, m_seq
         h0 = torch.zeros(B, H, device=dev, dtype=dt)
         z0 = torch.zeros(B, SK, device=dev, dtype=dt)
         w1t, w2t, w3t, w4t = w1.t(), w2.t(), w3.t(), w4.t()

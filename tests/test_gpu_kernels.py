@@ -414,3 +414,79 @@ def test_fast_imagination_matches_module_loop():
                                  urand_t=torch.stack(ut), urand_a=torch.stack(ua))
     assert torch.allclose(traj, traj_ref, atol=1e-4, rtol=1e-4), (traj - traj_ref).abs().max()
     assert torch.equal(acts, acts_ref)
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_persistent_scan_forward_matches_loop():
+    """pk_scan_fwd (single persistent kernel) vs the multi-kernel scan path:
+    same saved buffers and outputs (bf16, same urand)."""
+    import os
+    from sheeprl_amd.algos.dreamer_v3.agent import RSSM, RecurrentModel
+    from sheeprl_amd.models import MLP
+    from sheeprl_amd.ops import scan as scan_mod
+
+    T, B, E, A, H, S, K, DU, P = 4, 16, 64, 6, 64, 2, 32, 64, 64
+    SK = S * K
+    torch.manual_seed(0)
+    rssm = RSSM(
+        RecurrentModel(SK + A, H, DU),
+        MLP(E + H, SK, [P], activation="silu", layer_norm=True, layer_norm_eps=1e-3),
+        MLP(H, SK, [P], activation="silu", layer_norm=True, layer_norm_eps=1e-3),
+        discrete=K,
+        unimix=0.01,
+    ).cuda().to(torch.bfloat16)
+    embed = torch.randn(T, B, E, device="cuda", dtype=torch.bfloat16)
+    actions = torch.randn(T, B, A, device="cuda", dtype=torch.bfloat16)
+    is_first = (torch.rand(T, B, 1, device="cuda") < 0.3).float()
+    is_first[0] = 1.0
+    urand = torch.rand(T, B, S, K, device="cuda")
+    _ir, _ip = rssm.get_initial_states((1, B))
+    init = (_ir.contiguous(), _ip.contiguous())
+
+    def run(pk: str):
+        os.environ["SHEEPRL_AMD_PK"] = pk
+        try:
+            return scan_mod.rssm_scan(rssm, embed, actions, is_first, init, urand=urand)
+        finally:
+            os.environ.pop("SHEEPRL_AMD_PK", None)
+
+    h1, z1, m1 = run("0")
+    h2, z2, m2 = run("1")
+    # step 0 must agree tightly (no recurrence divergence yet)
+    assert torch.allclose(h2[0].float(), h1[0].float(), atol=3e-2, rtol=3e-2), \
+        (h2[0] - h1[0]).abs().max()
+    assert torch.allclose(m2[0], m1[0], atol=3e-2, rtol=3e-2), (m2[0] - m1[0]).abs().max()
+    mism = (z2[0] != z1[0]).any(-1).float().mean().item()
+    assert mism < 0.05, f"step-0 one-hot mismatch rate {mism}"
+
+    # internal consistency at every step: recompute step t from the pk run's
+    # own saved states with the multi-kernel ops
+    from sheeprl_amd.ops._ext import require_ext
+    ext = require_ext()
+    dt = torch.bfloat16
+    f_all = is_first.to(dt).reshape(T, B)
+    mlp = rssm.recurrent_model.mlp
+    gru = rssm.recurrent_model.rnn
+    rep = rssm.representation_model.model
+    for t in range(T):
+        f = f_all[t]
+        z_prev = z2[t - 1] if t > 0 else torch.zeros(B, SK, device="cuda", dtype=dt)
+        h_prev = h2[t - 1] if t > 0 else torch.zeros(B, H, device="cuda", dtype=dt)
+        a_eff = ext.masked_lerp_fwd(actions[t].contiguous(), None, f.reshape(-1))
+        z_in = ext.masked_lerp_fwd(z_prev.contiguous(), init[1].view(B, SK).to(dt).contiguous(), f.reshape(-1))
+        h_in = ext.masked_lerp_fwd(h_prev.contiguous(), init[0].view(B, H).to(dt).contiguous(), f.reshape(-1))
+        x = torch.cat((z_in, a_eff), -1)
+        g1 = x @ mlp.linear.weight.t()
+        u, _, _ = ext.ln_act_fwd(g1, mlp.ln_weight, mlp.ln_bias, 1e-3, True)
+        y = torch.cat((h_in, u), -1) @ gru.linear.weight.t()
+        h_t, _, _ = ext.gru_gates_fwd(y, h_in, gru.ln_weight, gru.ln_bias, 1e-3)
+        assert torch.allclose(h_t.float(), h2[t].float(), atol=5e-2, rtol=5e-2), \
+            (t, (h_t - h2[t]).abs().max())
+        r = torch.cat((h2[t], embed[t]), -1)
+        g3 = r @ rep[0].linear.weight.t()
+        p, _, _ = ext.ln_act_fwd(g3, rep[0].ln_weight, rep[0].ln_bias, 1e-3, True)
+        raw = torch.addmm(rep[1].linear.bias, p, rep[1].linear.weight.t())
+        m_ref, _, _ = ext.cat_st_fwd(raw.view(B, S, K), urand[t], 0.01, True)
+        assert torch.allclose(m_ref.view(B, SK), m2[t], atol=5e-2, rtol=5e-2), \
+            (t, (m_ref.view(B, SK) - m2[t]).abs().max())
