@@ -114,7 +114,7 @@ class _RSSMScan(torch.autograd.Function):
             and D % 64 == 0 and H % 64 == 0 and P % 64 == 0 and SK % 64 == 0
             and (H + D) % 32 == 0 and (H + E) % 32 == 0
             and discrete <= 64 and 64 % discrete == 0
-            and os.environ.get("SHEEPRL_AMD_PK", "1") == "1"
+            and os.environ.get("SHEEPRL_AMD_PK", "0") == "1"
         )
         if pk_ok:
             ws = torch.zeros(192, device=dev, dtype=torch.float32)
