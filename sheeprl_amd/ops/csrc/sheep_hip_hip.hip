@@ -376,6 +376,9 @@ void launch_ln_bwd_cl(int L, long R, hipStream_t st, const T* gy, const T* x, co
                       const float* mean, const float* rstd, T* gx, float* gw, float* gb, int D) {
   int rpb = (kBlock / 64) * (64 / L);
   int blocks = (int)std::min((R + rpb - 1) / rpb, (long)2048);
+  // each block flushes 2*D atomicAdds at the end; for mid-sized N that flush
+  // dominates — keep >=32 rows of real work per block
+  blocks = (int)std::min((long)blocks, std::max((long)64, (R + 31) / 32));
   size_t shmem = 2 * (size_t)D * sizeof(float);
 #define SHEEP_LN_BWD_CASE(LV)                                                                               \
   case LV:                                                                                                  \
@@ -1598,6 +1601,114 @@ torch::Tensor chlast_bias_sum(const torch::Tensor& g, long C) {
 }
 
 // ---------------------------------------------------------------------------
+// fused per-step episode-reset kernels for the RSSM scan: the three
+// masked-lerp launches of a step (z', a', h') collapse into ONE kernel
+// (regions decoded from the flat index), writing straight into the strided
+// column blocks of the stacked x_s / hu_s buffers.
+// ---------------------------------------------------------------------------
+
+template <typename T, bool T0>
+__global__ void scan_resets_fwd_kernel(const T* __restrict__ z_prev, const T* __restrict__ iz,
+                                       const T* __restrict__ h_prev, const T* __restrict__ ih,
+                                       const T* __restrict__ actions, const T* __restrict__ f,
+                                       T* __restrict__ x, long xs, T* __restrict__ hu, long hus, long B, int SK,
+                                       int A, int H) {
+  const long n = B * (long)(SK + A + H);
+  const int cols = SK + A + H;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
+    const long m = i / cols;
+    const int c = (int)(i - m * cols);
+    const float fb = ld(f, m);
+    if (c < SK) {
+      float zp = T0 ? 0.f : ld(z_prev, m * SK + c);
+      st(x, m * xs + c, (1.f - fb) * zp + fb * ld(iz, m * SK + c));
+    } else if (c < SK + A) {
+      st(x, m * xs + c, (1.f - fb) * ld(actions, m * A + (c - SK)));
+    } else {
+      const int j = c - SK - A;
+      float hp = T0 ? 0.f : ld(h_prev, m * H + j);
+      st(hu, m * hus + j, (1.f - fb) * hp + fb * ld(ih, m * H + j));
+    }
+  }
+}
+
+void scan_resets_fwd(const torch::Tensor& z_prev, const torch::Tensor& iz, const torch::Tensor& h_prev,
+                     const torch::Tensor& ih, const torch::Tensor& actions, const torch::Tensor& f,
+                     torch::Tensor x, torch::Tensor hu, bool t0) {
+  long B = actions.size(0);
+  int A = (int)actions.size(1);
+  int SK = (int)iz.size(1);
+  int H = (int)ih.size(1);
+  long n = B * (long)(SK + A + H);
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)1024);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, actions.scalar_type(), "scan_resets_fwd", [&] {
+    using T = scalar_t;
+    if (t0)
+      hipLaunchKernelGGL((scan_resets_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)z_prev.data_ptr(), (const T*)iz.data_ptr(), (const T*)h_prev.data_ptr(),
+                         (const T*)ih.data_ptr(), (const T*)actions.data_ptr(), (const T*)f.data_ptr(),
+                         (T*)x.data_ptr(), x.stride(0), (T*)hu.data_ptr(), hu.stride(0), B, SK, A, H);
+    else
+      hipLaunchKernelGGL((scan_resets_fwd_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                         (const T*)z_prev.data_ptr(), (const T*)iz.data_ptr(), (const T*)h_prev.data_ptr(),
+                         (const T*)ih.data_ptr(), (const T*)actions.data_ptr(), (const T*)f.data_ptr(),
+                         (T*)x.data_ptr(), x.stride(0), (T*)hu.data_ptr(), hu.stride(0), B, SK, A, H);
+  });
+}
+
+// backward of the fused resets: consumes gh_in (= ghu[:, :H] + ghp summed
+// in-kernel) and gx (strided [z', a'] block), producing the carries, the
+// action grad, and the fp32 init-state accumulators — one launch instead of
+// three.
+template <typename T>
+__global__ void scan_resets_bwd_kernel(const T* __restrict__ ghu, long ghus, const T* __restrict__ ghp,
+                                       const T* __restrict__ gx, long gxs, const T* __restrict__ f,
+                                       T* __restrict__ gh_carry, T* __restrict__ gz_carry,
+                                       T* __restrict__ ga, float* __restrict__ gih_acc,
+                                       float* __restrict__ giz_acc, long B, int SK, int A, int H) {
+  const long n = B * (long)(SK + A + H);
+  const int cols = SK + A + H;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n; i += (long)gridDim.x * blockDim.x) {
+    const long m = i / cols;
+    const int c = (int)(i - m * cols);
+    const float fb = ld(f, m);
+    if (c < SK) {
+      float g = ld(gx, m * gxs + c);
+      st(gz_carry, m * SK + c, (1.f - fb) * g);
+      giz_acc[m * SK + c] += fb * g;
+    } else if (c < SK + A) {
+      st(ga, m * A + (c - SK), (1.f - fb) * ld(gx, m * gxs + (c - 0)));
+    } else {
+      const int j = c - SK - A;
+      float g = ld(ghu, m * ghus + j) + ld(ghp, m * H + j);
+      st(gh_carry, m * H + j, (1.f - fb) * g);
+      gih_acc[m * H + j] += fb * g;
+    }
+  }
+}
+
+void scan_resets_bwd(const torch::Tensor& ghu, const torch::Tensor& ghp, const torch::Tensor& gx,
+                     const torch::Tensor& f, torch::Tensor gh_carry, torch::Tensor gz_carry, torch::Tensor ga,
+                     torch::Tensor gih_acc, torch::Tensor giz_acc) {
+  long B = ga.size(0);
+  int A = (int)ga.size(1);
+  int SK = (int)gz_carry.size(1);
+  int H = (int)gh_carry.size(1);
+  long n = B * (long)(SK + A + H);
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)1024);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, ga.scalar_type(), "scan_resets_bwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((scan_resets_bwd_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)ghu.data_ptr(), ghu.stride(0), (const T*)ghp.data_ptr(),
+                       (const T*)gx.data_ptr(), gx.stride(0), (const T*)f.data_ptr(),
+                       (T*)gh_carry.data_ptr(), (T*)gz_carry.data_ptr(), (T*)ga.data_ptr(),
+                       gih_acc.data_ptr<float>(), giz_acc.data_ptr<float>(), B, SK, A, H);
+  });
+}
+
+// ---------------------------------------------------------------------------
 // persistent fused RSSM scan — building blocks
 // ---------------------------------------------------------------------------
 // MFMA fragment types for v_mfma_f32_16x16x32_bf16 (gfx950): 8 bf16 per lane
@@ -1639,21 +1750,27 @@ torch::Tensor pk_gemm16_test(const torch::Tensor& x, const torch::Tensor& w) {
 
 // grid-wide barrier for persistent kernels: all workgroups must be resident
 // (grid <= #CUs).  counter/gen live in a small global workspace.
+// Cross-XCD visibility: every wave drains its stores (vmcnt), the arriving
+// thread issues an agent-scope RELEASE fence before the ticket and the
+// leaving threads an agent-scope ACQUIRE fence after the spin (per
+// cdna_hip_programming.md guideline 16 — cheaper than __threadfence()'s
+// full writeback on both sides).
 __device__ __forceinline__ void pk_grid_barrier(int* counter, volatile int* gen, int nwg) {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
-  __threadfence();
   if (threadIdx.x == 0) {
-    const int g = *gen;
-    if (atomicAdd(counter, 1) == nwg - 1) {
-      *counter = 0;
-      __threadfence();
-      atomicAdd((int*)gen, 1);
+    const int g = __hip_atomic_load((int*)gen, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+    if (__hip_atomic_fetch_add(counter, 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) == nwg - 1) {
+      __hip_atomic_store(counter, 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_fetch_add((int*)gen, 1, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
     } else {
-      while (*gen == g) __builtin_amdgcn_s_sleep(1);
+      while (__hip_atomic_load((int*)gen, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) == g) {
+      }
     }
+    __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
   }
   __syncthreads();
-  __threadfence();
 }
 
 // barrier smoke test: NWG workgroups increment a per-round slot ITERS times;
@@ -2157,4 +2274,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pk_gemm16_test", &pk_gemm16_test);
   m.def("pk_barrier_test", &pk_barrier_test);
   m.def("pk_scan_fwd", &pk_scan_fwd);
+  m.def("scan_resets_fwd", &scan_resets_fwd);
+  m.def("scan_resets_bwd", &scan_resets_bwd);
 }

@@ -142,9 +142,8 @@ class _RSSMScan(torch.autograd.Function):
             f = f_all[t]
             h_prev = h_seq[t - 1] if t > 0 else h0
             z_prev = z_seq[t - 1] if t > 0 else z0
-            ext.masked_lerp_fwd_o(actions[t], None, f, x_s[t, :, SK:])
-            ext.masked_lerp_fwd_o(z_prev, iz, f, x_s[t, :, :SK])
-            ext.masked_lerp_fwd_o(h_prev, ih, f, hu_s[t, :, :H])
+            # one kernel assembles the reset-masked GEMM inputs (z', a', h')
+            ext.scan_resets_fwd(z_prev, iz, h_prev, ih, actions[t], f, x_s[t], hu_s[t], t == 0)
             torch.mm(x_s[t], w1t, out=g1_s[t])
             ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
             torch.mm(hu_s[t], w2t, out=y_s[t])
@@ -224,9 +223,8 @@ class _RSSMScan(torch.autograd.Function):
             ext.ln_act_bwd_acc(ghu[:, H:], g1_s[t], lnw1, lnb1, mr1_s[0, t], mr1_s[1, t], True,
                                gg1_s[t], glnw1, glnb1)
             torch.mm(gg1_s[t], w1, out=gx)
-            ext.masked_lerp_bwd_acc(ghu[:, :H], ghp, f, gh_carry, gih_acc)
-            ext.masked_lerp_bwd_acc(gx[:, :SK], None, f, gz_carry, giz_acc)
-            ext.masked_lerp_bwd_acc(gx[:, SK:], None, f, g_actions[t], None)
+            # one kernel: carries, action grad, init-state accumulators
+            ext.scan_resets_bwd(ghu, ghp, gx, f, gh_carry, gz_carry, g_actions[t], gih_acc, giz_acc)
 
         # batched weight grads: one MFMA GEMM per weight over all T*B rows
         TB = T * B
