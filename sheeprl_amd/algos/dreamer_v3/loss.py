@@ -37,11 +37,20 @@ def reconstruction_loss(
     observation_loss = -sum(po[k].log_prob(observations[k]) for k in po.keys())
     reward_loss = -pr.log_prob(rewards)
     # KL balancing: dynamic = KL(sg(post) || prior), representation = KL(post || sg(prior))
-    kl = dyn_loss = categorical_kl(posteriors_logits.detach(), priors_logits)
-    free_nats = torch.full_like(dyn_loss, kl_free_nats)
-    dyn_loss = kl_dynamic * torch.maximum(dyn_loss, free_nats)
-    repr_loss = categorical_kl(posteriors_logits, priors_logits.detach())
-    repr_loss = kl_representation * torch.maximum(repr_loss, free_nats)
+    from sheeprl_amd import ops as _ops
+
+    if posteriors_logits.is_cuda and _ops.use_hip(posteriors_logits):
+        kl_dyn_v, kl_rep_v = _ops.kl_balanced(posteriors_logits.float(), priors_logits.float())
+        kl = dyn_loss = kl_dyn_v
+        free_nats = torch.full_like(dyn_loss, kl_free_nats)
+        dyn_loss = kl_dynamic * torch.maximum(dyn_loss, free_nats)
+        repr_loss = kl_representation * torch.maximum(kl_rep_v, free_nats)
+    else:
+        kl = dyn_loss = categorical_kl(posteriors_logits.detach(), priors_logits)
+        free_nats = torch.full_like(dyn_loss, kl_free_nats)
+        dyn_loss = kl_dynamic * torch.maximum(dyn_loss, free_nats)
+        repr_loss = categorical_kl(posteriors_logits, priors_logits.detach())
+        repr_loss = kl_representation * torch.maximum(repr_loss, free_nats)
     kl_loss = dyn_loss + repr_loss
     if pc is not None and continue_targets is not None:
         continue_loss = continue_scale_factor * -pc.log_prob(continue_targets)

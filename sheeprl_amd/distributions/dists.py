@@ -91,6 +91,7 @@ class TwoHotEncodingDistribution:
     ) -> None:
         self.logits = logits
         self._dims = tuple(-i for i in range(1, dims + 1))
+        self._low, self._high = low, high
         self.bins = _support(low, high, logits.shape[-1], logits.device)
 
     @property
@@ -107,6 +108,17 @@ class TwoHotEncodingDistribution:
 
     def log_prob(self, value: Tensor) -> Tensor:
         # value: [..., 1]; two-hot encode symlog(value) over bins
+        if (
+            self._dims == (-1,)
+            and self.logits.is_cuda
+            and self.logits.dtype == torch.float32
+            and not value.requires_grad
+            and ops.use_hip(self.logits)
+        ):
+            v = value.float()
+            if v.shape[-1] == 1:
+                v = v.squeeze(-1)
+            return ops.twohot_log_prob(self.logits, v, self._low, self._high)
         target = ops.twohot_from_support(ops.symlog(value.float()), self.bins)
         log_pred = self.logits - torch.logsumexp(self.logits, dim=-1, keepdim=True)
         return (target * log_pred).sum(self._dims)

@@ -9,7 +9,15 @@ from sheeprl_amd.ops.functional import (
     twohot_from_support,
 )
 from sheeprl_amd.ops.categorical import categorical_st
-from sheeprl_amd.ops.fused import ema_update_, gru_gates, layer_norm_act, masked_lerp, normalize_obs
+from sheeprl_amd.ops.fused import (
+    ema_update_,
+    gru_gates,
+    kl_balanced,
+    layer_norm_act,
+    masked_lerp,
+    normalize_obs,
+    twohot_log_prob,
+)
 
 __all__ = [
     "get_ext",
@@ -24,6 +32,8 @@ __all__ = [
     "gae",
     "lambda_values",
     "layer_norm_act",
+    "kl_balanced",
+    "twohot_log_prob",
     "categorical_st",
     "gru_gates",
     "ema_update_",

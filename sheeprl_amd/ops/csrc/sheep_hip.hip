@@ -1601,6 +1601,226 @@ torch::Tensor chlast_bias_sum(const torch::Tensor& g, long C) {
 }
 
 // ---------------------------------------------------------------------------
+// fused two-hot log-prob (DV3 reward/critic heads): logsumexp + uniform-bin
+// two-hot encoding of symlog(value) + cross-entropy in one kernel each way
+// (replaces the ~12-launch clamp/searchsorted/scatter/logsumexp chain of
+// TwoHotEncodingDistribution.log_prob, distribution.py:224-276).
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void twohot_idx(float x, float low, float high, int K, int& lo, int& hi,
+                                           float& w_lo, float& w_hi) {
+  const float step = (high - low) / (K - 1);
+  float a = fabsf(x);
+  float sx = (x >= 0.f ? 1.f : -1.f) * log1pf(a);  // symlog
+  sx = fminf(fmaxf(sx, low), high);
+  hi = (int)ceilf((sx - low) / step);
+  hi = min(max(hi, 0), K - 1);
+  lo = max(hi - 1, 0);
+  const float lo_v = low + lo * step;
+  const float hi_v = low + hi * step;
+  const float denom = fmaxf(hi_v - lo_v, 1e-8f);
+  w_hi = fminf(fmaxf((sx - lo_v) / denom, 0.f), 1.f);
+  w_lo = 1.f - w_hi;
+}
+
+__global__ void twohot_lp_fwd_kernel(const float* __restrict__ logits, const float* __restrict__ value,
+                                     float* __restrict__ out, float* __restrict__ lse_out, long N, int K,
+                                     float low, float high) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= N) return;
+  const float* lr = logits + row * (long)K;
+  float mx = -1e30f;
+  for (int j = lane; j < K; j += 64) mx = fmaxf(mx, lr[j]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+  float se = 0.f;
+  for (int j = lane; j < K; j += 64) se += expf(lr[j] - mx);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) se += __shfl_xor(se, off, 64);
+  const float lse = mx + logf(se);
+  if (lane == 0) {
+    int lo, hi;
+    float wl, wh;
+    twohot_idx(value[row], low, high, K, lo, hi, wl, wh);
+    out[row] = wl * (lr[lo] - lse) + wh * (lr[hi] - lse);
+    lse_out[row] = lse;
+  }
+}
+
+__global__ void twohot_lp_bwd_kernel(const float* __restrict__ g, const float* __restrict__ logits,
+                                     const float* __restrict__ value, const float* __restrict__ lse,
+                                     float* __restrict__ gl, long N, int K, float low, float high) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < N * (long)K;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / K;
+    const int j = (int)(i - row * K);
+    int lo, hi;
+    float wl, wh;
+    twohot_idx(value[row], low, high, K, lo, hi, wl, wh);
+    const float tw = (j == lo ? wl : 0.f) + (j == hi ? wh : 0.f);
+    gl[i] = g[row] * (tw - expf(logits[i] - lse[row]));
+  }
+}
+
+std::vector<torch::Tensor> twohot_lp_fwd(const torch::Tensor& logits, const torch::Tensor& value, double low,
+                                         double high) {
+  CHECK_IN(logits);
+  TORCH_CHECK(logits.scalar_type() == at::kFloat && value.scalar_type() == at::kFloat);
+  int K = (int)logits.size(-1);
+  long N = logits.numel() / K;
+  auto out = torch::empty({N}, logits.options());
+  auto lse = torch::empty({N}, logits.options());
+  const int rpb = kBlock / 64;
+  int blocks = (int)((N + rpb - 1) / rpb);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(twohot_lp_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     logits.data_ptr<float>(), value.data_ptr<float>(), out.data_ptr<float>(),
+                     lse.data_ptr<float>(), N, K, (float)low, (float)high);
+  return {out, lse};
+}
+
+torch::Tensor twohot_lp_bwd(const torch::Tensor& g, const torch::Tensor& logits, const torch::Tensor& value,
+                            const torch::Tensor& lse, double low, double high) {
+  int K = (int)logits.size(-1);
+  long N = logits.numel() / K;
+  auto gl = torch::empty_like(logits);
+  int blocks = (int)std::min((N * K + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(twohot_lp_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     g.data_ptr<float>(), logits.data_ptr<float>(), value.data_ptr<float>(),
+                     lse.data_ptr<float>(), gl.data_ptr<float>(), N, K, (float)low, (float)high);
+  return gl;
+}
+
+// ---------------------------------------------------------------------------
+// fused balanced categorical KL (DV3 world-model loss, loss.py:64-75): both
+// KL(sg(post)||prior) and KL(post||sg(prior)) share the forward VALUE; one
+// kernel computes the per-sample KL (summed over stoch groups), the backward
+// kernel routes the two incoming grads to prior/post respectively.
+// ---------------------------------------------------------------------------
+
+__global__ void klbal_fwd_kernel(const float* __restrict__ post, const float* __restrict__ prior,
+                                 float* __restrict__ out, float* __restrict__ kls, long NS, int S, int K) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);  // (n, s) pair
+  if (row >= NS) return;
+  const float* pr = post + row * (long)K;
+  const float* qr = prior + row * (long)K;
+  float mp = -1e30f, mq = -1e30f;
+  for (int j = lane; j < K; j += 64) {
+    mp = fmaxf(mp, pr[j]);
+    mq = fmaxf(mq, qr[j]);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    mp = fmaxf(mp, __shfl_xor(mp, off, 64));
+    mq = fmaxf(mq, __shfl_xor(mq, off, 64));
+  }
+  float sp = 0.f, sq = 0.f;
+  for (int j = lane; j < K; j += 64) {
+    sp += expf(pr[j] - mp);
+    sq += expf(qr[j] - mq);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sp += __shfl_xor(sp, off, 64);
+    sq += __shfl_xor(sq, off, 64);
+  }
+  const float lsep = mp + logf(sp), lseq = mq + logf(sq);
+  float kl = 0.f;
+  for (int j = lane; j < K; j += 64) {
+    const float lp = pr[j] - lsep;
+    const float lq = qr[j] - lseq;
+    kl += expf(lp) * (lp - lq);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) kl += __shfl_xor(kl, off, 64);
+  if (lane == 0) {
+    kls[row] = kl;
+    atomicAdd(&out[row / S], kl);
+  }
+}
+
+__global__ void klbal_bwd_kernel(const float* __restrict__ g_dyn, const float* __restrict__ g_rep,
+                                 const float* __restrict__ post, const float* __restrict__ prior,
+                                 const float* __restrict__ kls, float* __restrict__ g_post,
+                                 float* __restrict__ g_prior, long NS, int S, int K) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= NS) return;
+  const float* pr = post + row * (long)K;
+  const float* qr = prior + row * (long)K;
+  float mp = -1e30f, mq = -1e30f;
+  for (int j = lane; j < K; j += 64) {
+    mp = fmaxf(mp, pr[j]);
+    mq = fmaxf(mq, qr[j]);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    mp = fmaxf(mp, __shfl_xor(mp, off, 64));
+    mq = fmaxf(mq, __shfl_xor(mq, off, 64));
+  }
+  float sp = 0.f, sq = 0.f;
+  for (int j = lane; j < K; j += 64) {
+    sp += expf(pr[j] - mp);
+    sq += expf(qr[j] - mq);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    sp += __shfl_xor(sp, off, 64);
+    sq += __shfl_xor(sq, off, 64);
+  }
+  const float lsep = mp + logf(sp), lseq = mq + logf(sq);
+  const float gd = g_dyn[row / S], gr2 = g_rep[row / S];
+  const float klg = kls[row];
+  for (int j = lane; j < K; j += 64) {
+    const float lp = pr[j] - lsep;
+    const float lq = qr[j] - lseq;
+    const float p = expf(lp);
+    const float q = expf(lq);
+    g_prior[row * (long)K + j] = gd * (q - p);
+    g_post[row * (long)K + j] = gr2 * p * ((lp - lq) - klg);
+  }
+}
+
+std::vector<torch::Tensor> klbal_fwd(const torch::Tensor& post, const torch::Tensor& prior) {
+  CHECK_IN(post);
+  CHECK_IN(prior);
+  TORCH_CHECK(post.scalar_type() == at::kFloat && post.dim() >= 2);
+  int K = (int)post.size(-1);
+  int S = (int)post.size(-2);
+  long NS = post.numel() / K;
+  auto out = torch::zeros({NS / S}, post.options());
+  auto kls = torch::empty({NS}, post.options());
+  const int rpb = kBlock / 64;
+  int blocks = (int)((NS + rpb - 1) / rpb);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(klbal_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     post.data_ptr<float>(), prior.data_ptr<float>(), out.data_ptr<float>(),
+                     kls.data_ptr<float>(), NS, S, K);
+  return {out, kls};
+}
+
+std::vector<torch::Tensor> klbal_bwd(const torch::Tensor& g_dyn, const torch::Tensor& g_rep,
+                                     const torch::Tensor& post, const torch::Tensor& prior,
+                                     const torch::Tensor& kls) {
+  int K = (int)post.size(-1);
+  int S = (int)post.size(-2);
+  long NS = post.numel() / K;
+  auto g_post = torch::empty_like(post);
+  auto g_prior = torch::empty_like(prior);
+  const int rpb = kBlock / 64;
+  int blocks = (int)((NS + rpb - 1) / rpb);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(klbal_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     g_dyn.contiguous().data_ptr<float>(), g_rep.contiguous().data_ptr<float>(),
+                     post.data_ptr<float>(), prior.data_ptr<float>(), kls.data_ptr<float>(),
+                     g_post.data_ptr<float>(), g_prior.data_ptr<float>(), NS, S, K);
+  return {g_post, g_prior};
+}
+
+// ---------------------------------------------------------------------------
 // fused per-step episode-reset kernels for the RSSM scan: the three
 // masked-lerp launches of a step (z', a', h') collapse into ONE kernel
 // (regions decoded from the flat index), writing straight into the strided
@@ -2276,4 +2496,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pk_scan_fwd", &pk_scan_fwd);
   m.def("scan_resets_fwd", &scan_resets_fwd);
   m.def("scan_resets_bwd", &scan_resets_bwd);
+  m.def("twohot_lp_fwd", &twohot_lp_fwd);
+  m.def("twohot_lp_bwd", &twohot_lp_bwd);
+  m.def("klbal_fwd", &klbal_fwd);
+  m.def("klbal_bwd", &klbal_bwd);
 }

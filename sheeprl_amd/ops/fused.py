@@ -193,3 +193,65 @@ class _MaskedLerp(torch.autograd.Function):
 def masked_lerp(x: Tensor, init: Optional[Tensor], f: Tensor) -> Tensor:
     """Rows of ``x``/``init`` are lerped by the per-row mask ``f`` ([*, 1])."""
     return _MaskedLerp.apply(x, init, f)
+
+
+class _TwoHotLogProb(torch.autograd.Function):
+    """Fused two-hot cross-entropy: logsumexp + uniform-bin two-hot of
+    symlog(value) + inner product in one kernel each way (the torch chain is
+    ~12 launches; reference distribution.py:224-276 semantics)."""
+
+    @staticmethod
+    def forward(ctx, logits: Tensor, value: Tensor, low: float, high: float) -> Tensor:
+        ext = require_ext()
+        out, lse = ext.twohot_lp_fwd(logits.contiguous(), value.contiguous(), low, high)
+        ctx.save_for_backward(logits, value, lse)
+        ctx.lh = (low, high)
+        return out.view(logits.shape[:-1])
+
+    @staticmethod
+    def backward(ctx, g: Tensor):
+        ext = require_ext()
+        logits, value, lse = ctx.saved_tensors
+        low, high = ctx.lh
+        gl = ext.twohot_lp_bwd(g.contiguous().view(-1), logits, value, lse, low, high)
+        return gl.view(logits.shape), None, None, None
+
+
+def twohot_log_prob(logits: Tensor, value: Tensor, low: float = -20.0, high: float = 20.0) -> Tensor:
+    """log_prob of a TwoHotEncodingDistribution over a uniform symlog support.
+
+    ``logits`` [..., K] fp32, ``value`` [...] fp32 (no grad); returns [...].
+    """
+    return _TwoHotLogProb.apply(logits, value.float(), low, high)
+
+
+class _KLBalanced(torch.autograd.Function):
+    """Fused two-sided KL balancing (DV3 world-model loss, reference
+    loss.py:64-75): KL(sg(post)||prior) and KL(post||sg(prior)) share the
+    forward value; the two outputs route their gradients to prior and post
+    respectively."""
+
+    @staticmethod
+    def forward(ctx, post_logits: Tensor, prior_logits: Tensor):
+        ext = require_ext()
+        post = post_logits.contiguous()
+        prior = prior_logits.contiguous()
+        kl, kls = ext.klbal_fwd(post, prior)
+        ctx.save_for_backward(post, prior, kls)
+        out = kl.view(post_logits.shape[:-2])
+        return out, out.clone()
+
+    @staticmethod
+    def backward(ctx, g_dyn: Tensor, g_rep: Tensor):
+        ext = require_ext()
+        post, prior, kls = ctx.saved_tensors
+        g_post, g_prior = ext.klbal_bwd(g_dyn.contiguous().view(-1), g_rep.contiguous().view(-1),
+                                        post, prior, kls)
+        return g_post, g_prior
+
+
+def kl_balanced(post_logits: Tensor, prior_logits: Tensor):
+    """Returns (kl_dynamic, kl_representation): numerically the same
+    KL(post||prior) summed over the stoch dim, but the first backpropagates
+    only into ``prior_logits`` and the second only into ``post_logits``."""
+    return _KLBalanced.apply(post_logits, prior_logits)

@@ -490,3 +490,55 @@ def test_persistent_scan_forward_matches_loop():
         m_ref, _, _ = ext.cat_st_fwd(raw.view(B, S, K), urand[t], 0.01, True)
         assert torch.allclose(m_ref.view(B, SK), m2[t], atol=5e-2, rtol=5e-2), \
             (t, (m_ref.view(B, SK) - m2[t]).abs().max())
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_twohot_log_prob_fused_matches_torch():
+    from sheeprl_amd.distributions import TwoHotEncodingDistribution
+    from sheeprl_amd import ops
+
+    torch.manual_seed(0)
+    logits = torch.randn(7, 13, 255, device="cuda", requires_grad=True)
+    value = torch.randn(7, 13, 1, device="cuda") * 30  # exercises clamping too
+    # torch reference path
+    target = ops.twohot_from_support(ops.symlog(value.float()),
+                                     torch.linspace(-20, 20, 255, device="cuda"))
+    ref = (target * (logits - torch.logsumexp(logits, -1, keepdim=True))).sum(-1)
+    ref.sum().backward()
+    g_ref = logits.grad.clone()
+    logits.grad = None
+    out = ops.twohot_log_prob(logits, value.squeeze(-1))
+    assert torch.allclose(out, ref.detach(), atol=1e-4, rtol=1e-4), (out - ref).abs().max()
+    out.sum().backward()
+    assert torch.allclose(logits.grad, g_ref, atol=1e-4, rtol=1e-4), (logits.grad - g_ref).abs().max()
+    # integration through the distribution
+    d = TwoHotEncodingDistribution(logits.detach().clone().requires_grad_(True), dims=1)
+    lp = d.log_prob(value)
+    assert torch.allclose(lp, ref.detach(), atol=1e-4, rtol=1e-4)
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_kl_balanced_fused_matches_torch():
+    from sheeprl_amd.algos.dreamer_v3.loss import categorical_kl
+    from sheeprl_amd import ops
+
+    torch.manual_seed(1)
+    post = torch.randn(5, 9, 32, 32, device="cuda", requires_grad=True)
+    prior = torch.randn(5, 9, 32, 32, device="cuda", requires_grad=True)
+    w = torch.randn(5, 9, device="cuda")
+
+    dyn_ref = categorical_kl(post.detach(), prior)
+    rep_ref = categorical_kl(post, prior.detach())
+    (dyn_ref * w).sum().backward()
+    (rep_ref * (2 * w)).sum().backward()
+    gp_ref, gq_ref = post.grad.clone(), prior.grad.clone()
+    post.grad = prior.grad = None
+
+    dyn, rep = ops.kl_balanced(post, prior)
+    assert torch.allclose(dyn, dyn_ref.detach(), atol=1e-4, rtol=1e-4), (dyn - dyn_ref).abs().max()
+    assert torch.allclose(rep, rep_ref.detach(), atol=1e-4, rtol=1e-4)
+    ((dyn * w).sum() + (rep * (2 * w)).sum()).backward()
+    assert torch.allclose(post.grad, gp_ref, atol=1e-4, rtol=1e-4), (post.grad - gp_ref).abs().max()
+    assert torch.allclose(prior.grad, gq_ref, atol=1e-4, rtol=1e-4), (prior.grad - gq_ref).abs().max()
