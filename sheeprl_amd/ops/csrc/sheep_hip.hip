@@ -1937,6 +1937,267 @@ void scan_resets_bwd(const torch::Tensor& ghu, const torch::Tensor& ghp, const t
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
+// ---------------------------------------------------------------------------
+// M=16 GEMM family for the RSSM scan steps (per the CDNA4 guide's M<=16
+// recipe: operands straight to VGPRs, deep unroll, no LDS staging for the
+// streamed weight).  A is the [B<=16, K] activation (row-strided, unpadded:
+// the final K-chunk is guard-loaded), W is the torch [N, K] weight.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ bf16x8 g16_load8(const __hip_bfloat16* base, long row, long stride, int k, int K,
+                                            int nrows, int r) {
+  bf16x8 v;
+  if (r >= nrows) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) v[e] = (__bf16)0.f;
+    return v;
+  }
+  const __hip_bfloat16* p = base + row * stride + k;
+  if (k + 8 <= K) return *(const bf16x8*)p;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) v[e] = (k + e < K) ? (__bf16)__bfloat162float(p[e]) : (__bf16)0.f;
+  return v;
+}
+
+// plain multi-workgroup variant: C[16, N] (strided rows) = A @ W^T; one
+// 16-col tile per wave, grid = N/64 workgroups.
+__global__ void __launch_bounds__(256) g16_plain_kernel(const __hip_bfloat16* __restrict__ A, long as_,
+                                                        const __hip_bfloat16* __restrict__ W,
+                                                        __hip_bfloat16* __restrict__ C, long cs, int B, int N,
+                                                        int K) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int ncol0 = (blockIdx.x * (int)(blockDim.x >> 6) + (threadIdx.x >> 6)) * 16;
+  if (ncol0 >= N) return;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int Kp = ((K + 31) / 32) * 32;
+#pragma unroll 4
+  for (int k0 = 0; k0 < Kp; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8(A, arow, as_, k, K, B, arow);
+    bf16x8 b = g16_load8(W, ncol0 + arow, K, k, K, N, ncol0 + arow);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    if (m < B) C[(long)m * cs + ncol0 + arow] = (__hip_bfloat16)acc[r];
+  }
+}
+
+void g16_plain(const torch::Tensor& A, const torch::Tensor& W, torch::Tensor C) {
+  TORCH_CHECK(A.dim() == 2 && A.stride(1) == 1 && W.is_contiguous() && C.stride(1) == 1);
+  TORCH_CHECK(A.size(0) <= 16 && W.size(0) % 64 == 0 && A.scalar_type() == at::kBFloat16);
+  int B = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(g16_plain_kernel, dim3(N / 64), dim3(256), 0, stream.stream(),
+                     (const __hip_bfloat16*)A.data_ptr(), A.stride(0), (const __hip_bfloat16*)W.data_ptr(),
+                     (__hip_bfloat16*)C.data_ptr(), C.stride(0), B, N, K);
+}
+
+// single-workgroup GEMM + rowwise LN+SiLU epilogue (the recurrent-model MLP
+// step): writes the pre-LN GEMM result (saved for backward), mean/rstd, and
+// the activated output into a strided slice.  N <= 512 (LDS image).
+template <int TILES>
+__global__ void __launch_bounds__(256) g16_ln_silu_kernel(const __hip_bfloat16* __restrict__ A, long as_,
+                                                          const __hip_bfloat16* __restrict__ W,
+                                                          const __hip_bfloat16* __restrict__ lnw,
+                                                          const __hip_bfloat16* __restrict__ lnb,
+                                                          __hip_bfloat16* __restrict__ G, long gs,
+                                                          __hip_bfloat16* __restrict__ Y, long ys,
+                                                          float* __restrict__ mean_out,
+                                                          float* __restrict__ rstd_out, int B, int N, int K,
+                                                          float eps) {
+  extern __shared__ __attribute__((aligned(16))) float lds[];  // [16][N]
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  f32x4 acc[TILES];
+#pragma unroll
+  for (int t = 0; t < TILES; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  const int Kp = ((K + 31) / 32) * 32;
+  for (int k0 = 0; k0 < Kp; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8(A, arow, as_, k, K, B, arow);
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
+      const int ncol0 = (t * 4 + wv) * 16;
+      bf16x8 b = g16_load8(W, ncol0 + arow, K, k, K, N, ncol0 + arow);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < TILES; ++t) {
+    const int ncol0 = (t * 4 + wv) * 16;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) lds[(kgrp * 4 + r) * N + ncol0 + arow] = acc[t][r];
+  }
+  __syncthreads();
+  // rowwise LN stats from the LDS image (4 waves x 4 rows each)
+  __shared__ float mr[2][16];
+  for (int m = wv * 4; m < wv * 4 + 4; ++m) {
+    float s = 0.f, s2 = 0.f;
+    for (int j = lane; j < N; j += 64) {
+      float v = lds[m * N + j];
+      s += v;
+      s2 += v * v;
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s += __shfl_xor(s, off, 64);
+      s2 += __shfl_xor(s2, off, 64);
+    }
+    if (lane == 0) {
+      float mean = s / N;
+      float var = s2 / N - mean * mean;
+      mr[0][m] = mean;
+      mr[1][m] = rsqrtf(fmaxf(var, 0.f) + eps);
+      if (m < B) {
+        mean_out[m] = mean;
+        rstd_out[m] = mr[1][m];
+      }
+    }
+  }
+  __syncthreads();
+  // write G (pre-LN) and Y = silu(LN(G))
+  for (int i = threadIdx.x; i < B * N; i += blockDim.x) {
+    const int m = i / N, j = i - m * N;
+    const float v = lds[m * N + j];
+    G[(long)m * gs + j] = (__hip_bfloat16)v;
+    float z = (v - mr[0][m]) * mr[1][m] * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
+    Y[(long)m * ys + j] = (__hip_bfloat16)(z / (1.f + expf(-z)));
+  }
+}
+
+void g16_ln_silu(const torch::Tensor& A, const torch::Tensor& W, const torch::Tensor& lnw,
+                 const torch::Tensor& lnb, torch::Tensor G, torch::Tensor Y, torch::Tensor mean,
+                 torch::Tensor rstd, double eps) {
+  TORCH_CHECK(A.dim() == 2 && A.stride(1) == 1 && W.is_contiguous() && A.scalar_type() == at::kBFloat16);
+  int B = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
+  TORCH_CHECK(B <= 16 && N % 64 == 0 && N <= 512 && G.stride(1) == 1 && Y.stride(1) == 1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+#define SHEEP_G16LN_CASE(TV)                                                                              \
+  case TV:                                                                                                \
+    hipLaunchKernelGGL(g16_ln_silu_kernel<TV>, dim3(1), dim3(256), 16 * N * sizeof(float),                \
+                       stream.stream(), (const __hip_bfloat16*)A.data_ptr(), A.stride(0),                 \
+                       (const __hip_bfloat16*)W.data_ptr(), (const __hip_bfloat16*)lnw.data_ptr(),        \
+                       (const __hip_bfloat16*)lnb.data_ptr(), (__hip_bfloat16*)G.data_ptr(), G.stride(0), \
+                       (__hip_bfloat16*)Y.data_ptr(), Y.stride(0), mean.data_ptr<float>(),                \
+                       rstd.data_ptr<float>(), B, N, K, (float)eps);                                      \
+    break;
+  switch (N / 64) {
+    SHEEP_G16LN_CASE(1)
+    SHEEP_G16LN_CASE(2)
+    SHEEP_G16LN_CASE(4)
+    SHEEP_G16LN_CASE(8)
+    default:
+      TORCH_CHECK(false, "g16_ln_silu: unsupported N");
+  }
+#undef SHEEP_G16LN_CASE
+}
+
+// single-workgroup GEMM + bias + unimix categorical-ST epilogue (the
+// posterior head): raw = A @ W^T + b, then per-group softmax / log-prob /
+// gumbel one-hot straight into the stacked m/z/s buffers.  N <= 1024.
+template <int TILES>
+__global__ void __launch_bounds__(256) g16_cat_st_kernel(const __hip_bfloat16* __restrict__ A, long as_,
+                                                         const __hip_bfloat16* __restrict__ W,
+                                                         const __hip_bfloat16* __restrict__ bias,
+                                                         const float* __restrict__ urand,
+                                                         float* __restrict__ m_out,
+                                                         __hip_bfloat16* __restrict__ z_out,
+                                                         float* __restrict__ s_out, int B, int N, int K, int KD,
+                                                         float unimix) {
+  extern __shared__ __attribute__((aligned(16))) float lds[];  // [16][N]
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  f32x4 acc[TILES];
+#pragma unroll
+  for (int t = 0; t < TILES; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  const int Kp = ((K + 31) / 32) * 32;
+  for (int k0 = 0; k0 < Kp; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8(A, arow, as_, k, K, B, arow);
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
+      const int ncol0 = (t * 4 + wv) * 16;
+      bf16x8 b = g16_load8(W, ncol0 + arow, K, k, K, N, ncol0 + arow);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < TILES; ++t) {
+    const int ncol0 = (t * 4 + wv) * 16;
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      lds[(kgrp * 4 + r) * N + ncol0 + arow] = acc[t][r] + __bfloat162float(bias[ncol0 + arow]);
+  }
+  __syncthreads();
+  const int groups = N / KD;
+  for (int task = threadIdx.x; task < 16 * groups; task += blockDim.x) {
+    const int m = task / groups;
+    if (m >= B) continue;
+    const int g = task - m * groups;
+    const float* row = lds + m * N + g * KD;
+    float lmax = -1e30f;
+    for (int j = 0; j < KD; ++j) lmax = fmaxf(lmax, row[j]);
+    float lsum = 0.f;
+    for (int j = 0; j < KD; ++j) lsum += expf(row[j] - lmax);
+    const float inv = 1.f / lsum;
+    const float* ur = urand + ((long)m * groups + g) * KD;
+    float* mr_ = m_out + (long)m * N + g * KD;
+    float* sr = s_out + (long)m * N + g * KD;
+    __hip_bfloat16* zr = z_out + (long)m * N + g * KD;
+    float best = -1e30f;
+    int best_j = 0;
+    for (int j = 0; j < KD; ++j) {
+      float sv = expf(row[j] - lmax) * inv;
+      float pv = (1.f - unimix) * sv + unimix / KD;
+      float mv = logf(pv);
+      sr[j] = sv;
+      mr_[j] = mv;
+      float tt = fmaxf(-logf(fmaxf(ur[j], 1e-20f)), 1e-20f);
+      float score = mv - logf(tt);
+      if (score > best) {
+        best = score;
+        best_j = j;
+      }
+    }
+    for (int j = 0; j < KD; ++j) zr[j] = (__hip_bfloat16)(j == best_j ? 1.f : 0.f);
+  }
+}
+
+void g16_cat_st(const torch::Tensor& A, const torch::Tensor& W, const torch::Tensor& bias,
+                const torch::Tensor& urand, torch::Tensor m, torch::Tensor z, torch::Tensor s, long KD,
+                double unimix) {
+  TORCH_CHECK(A.dim() == 2 && A.stride(1) == 1 && W.is_contiguous() && A.scalar_type() == at::kBFloat16);
+  int B = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
+  TORCH_CHECK(B <= 16 && N % 64 == 0 && N <= 1024 && m.is_contiguous() && z.is_contiguous() &&
+              s.is_contiguous() && N % KD == 0);
+  auto stream = at::cuda::getCurrentCUDAStream();
+#define SHEEP_G16CS_CASE(TV)                                                                             \
+  case TV:                                                                                               \
+    hipLaunchKernelGGL(g16_cat_st_kernel<TV>, dim3(1), dim3(256), 16 * N * sizeof(float),                \
+                       stream.stream(), (const __hip_bfloat16*)A.data_ptr(), A.stride(0),                \
+                       (const __hip_bfloat16*)W.data_ptr(), (const __hip_bfloat16*)bias.data_ptr(),      \
+                       urand.data_ptr<float>(), m.data_ptr<float>(), (__hip_bfloat16*)z.data_ptr(),      \
+                       s.data_ptr<float>(), B, N, K, (int)KD, (float)unimix);                            \
+    break;
+  switch (N / 64) {
+    SHEEP_G16CS_CASE(2)
+    SHEEP_G16CS_CASE(4)
+    SHEEP_G16CS_CASE(8)
+    SHEEP_G16CS_CASE(16)
+    default:
+      TORCH_CHECK(false, "g16_cat_st: unsupported N");
+  }
+#undef SHEEP_G16CS_CASE
+}
+
 // Micro-kernel used by the layout unit test: Y[16,N] = X[16,K] @ W[N,K]^T.
 __global__ void pk_gemm16_test_kernel(const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ W,
                                       float* __restrict__ Y, int N, int K) {
@@ -2500,4 +2761,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("twohot_lp_bwd", &twohot_lp_bwd);
   m.def("klbal_fwd", &klbal_fwd);
   m.def("klbal_bwd", &klbal_bwd);
+  m.def("g16_plain", &g16_plain);
+  m.def("g16_ln_silu", &g16_ln_silu);
+  m.def("g16_cat_st", &g16_cat_st);
 }

@@ -137,6 +137,18 @@ class _RSSMScan(torch.autograd.Function):
             return h_seq, z_seq, m_seq
         h0 = torch.zeros(B, H, device=dev, dtype=dt)
         z0 = torch.zeros(B, SK, device=dev, dtype=dt)
+        # hand-written M=16 MFMA kernels (with fused LN / categorical-ST
+        # epilogues) replace hipblaslt for the step GEMMs when shapes allow —
+        # each fused call is one launch where torch needs two
+        g16 = (
+            dt == torch.bfloat16
+            and B <= 16
+            and w1.dtype == dt
+            and D % 64 == 0 and D <= 512
+            and (3 * H) % 64 == 0
+            and P % 64 == 0
+            and SK % 64 == 0 and SK <= 1024
+        )
         w1t, w2t, w3t, w4t = w1.t(), w2.t(), w3.t(), w4.t()
         for t in range(T):
             f = f_all[t]
@@ -144,6 +156,17 @@ class _RSSMScan(torch.autograd.Function):
             z_prev = z_seq[t - 1] if t > 0 else z0
             # one kernel assembles the reset-masked GEMM inputs (z', a', h')
             ext.scan_resets_fwd(z_prev, iz, h_prev, ih, actions[t], f, x_s[t], hu_s[t], t == 0)
+            if g16:
+                ext.g16_ln_silu(x_s[t], w1, lnw1, lnb1, g1_s[t], hu_s[t, :, H:],
+                                mr1_s[0, t], mr1_s[1, t], eps)
+                ext.g16_plain(hu_s[t], w2, y_s[t])
+                ext.gru_gates_fwd_o(y_s[t], hu_s[t, :, :H], lnwg, lnbg, eps,
+                                    h_seq[t], r_s[t, :, :H], mrg_s[0, t], mrg_s[1, t])
+                ext.g16_plain(r_s[t], w3, g3_s[t])
+                ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
+                ext.g16_cat_st(p_s[t], w4, b4, urand_all[t], m_seq[t], z_seq[t], s_s[t],
+                               discrete, unimix)
+                continue
             torch.mm(x_s[t], w1t, out=g1_s[t])
             ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
             torch.mm(hu_s[t], w2t, out=y_s[t])

@@ -542,3 +542,56 @@ def test_kl_balanced_fused_matches_torch():
     ((dyn * w).sum() + (rep * (2 * w)).sum()).backward()
     assert torch.allclose(post.grad, gp_ref, atol=1e-4, rtol=1e-4), (post.grad - gp_ref).abs().max()
     assert torch.allclose(prior.grad, gq_ref, atol=1e-4, rtol=1e-4), (prior.grad - gq_ref).abs().max()
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_g16_gemm_kernels_match_torch():
+    from sheeprl_amd.ops._ext import require_ext
+    ext = require_ext()
+    torch.manual_seed(0)
+    dt = torch.bfloat16
+    B, K, N = 16, 1030, 512
+    A = torch.randn(B, K, device="cuda", dtype=dt)
+    W = torch.randn(N, K, device="cuda", dtype=dt) * 0.05
+    C = torch.empty(B, N, device="cuda", dtype=dt)
+    ext.g16_plain(A, W, C)
+    ref = (A.float() @ W.float().t())
+    assert torch.allclose(C.float(), ref, atol=0.35, rtol=0.05), (C.float() - ref).abs().max()
+    # partial rows
+    A7 = A[:7].contiguous()
+    C7 = torch.empty(7, N, device="cuda", dtype=dt)
+    ext.g16_plain(A7, W, C7)
+    assert torch.allclose(C7.float(), ref[:7], atol=0.35, rtol=0.05)
+
+    # fused LN+SiLU epilogue
+    lnw = torch.randn(N, device="cuda", dtype=dt)
+    lnb = torch.randn(N, device="cuda", dtype=dt)
+    G = torch.empty(B, N, device="cuda", dtype=dt)
+    Y = torch.empty(B, N + 64, device="cuda", dtype=dt)[:, 32:32 + N]  # strided out
+    mean = torch.empty(B, device="cuda")
+    rstd = torch.empty(B, device="cuda")
+    ext.g16_ln_silu(A, W, lnw, lnb, G, Y, mean, rstd, 1e-3)
+    g_ref = torch.from_numpy(ref.cpu().numpy()).cuda()
+    gb = G.float()
+    ln = torch.nn.functional.layer_norm(gb, (N,), lnw.float(), lnb.float(), 1e-3)
+    y_ref = torch.nn.functional.silu(ln)
+    assert torch.allclose(G.float(), ref, atol=0.35, rtol=0.05)
+    assert torch.allclose(Y.float(), y_ref, atol=0.1, rtol=0.1), (Y.float() - y_ref).abs().max()
+
+    # fused categorical-ST epilogue
+    S, KD = 32, 32
+    N2, K2 = S * KD, 512
+    A2 = torch.randn(B, K2, device="cuda", dtype=dt)
+    W2 = torch.randn(N2, K2, device="cuda", dtype=dt) * 0.05
+    b2 = torch.randn(N2, device="cuda", dtype=dt)
+    ur = torch.rand(B, S, KD, device="cuda")
+    m = torch.empty(B, N2, device="cuda")
+    z = torch.empty(B, N2, device="cuda", dtype=dt)
+    sp = torch.empty(B, S, KD, device="cuda")
+    ext.g16_cat_st(A2, W2, b2, ur, m, z, sp, KD, 0.01)
+    raw_ref = (A2.float() @ W2.float().t() + b2.float()).view(B, S, KD)
+    m_ref, _, s_ref = ext.cat_st_fwd(raw_ref.to(dt).contiguous(), ur, 0.01, True)
+    assert torch.allclose(m.view(B, S, KD), m_ref, atol=2e-2, rtol=2e-2), (m.view(B,S,KD) - m_ref).abs().max()
+    assert torch.allclose(sp, s_ref, atol=2e-2, rtol=2e-2)
+    assert (z.view(B, S, KD).sum(-1) == 1).all()
