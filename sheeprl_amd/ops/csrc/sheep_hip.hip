@@ -1944,6 +1944,10 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 // the final K-chunk is guard-loaded), W is the torch [N, K] weight.
 // ---------------------------------------------------------------------------
 
+__device__ __forceinline__ bf16x8 g16_load8v(const __hip_bfloat16* base, long row, long stride, int k) {
+  return *(const bf16x8*)(base + row * stride + k);
+}
+
 __device__ __forceinline__ bf16x8 g16_load8(const __hip_bfloat16* base, long row, long stride, int k, int K,
                                             int nrows, int r) {
   bf16x8 v;
@@ -1963,6 +1967,7 @@ __device__ __forceinline__ bf16x8 g16_load8(const __hip_bfloat16* base, long row
 // 16-col tile per wave, grid = N/64 workgroups.
 __global__ void __launch_bounds__(256) g16_plain_kernel(const __hip_bfloat16* __restrict__ A, long as_,
                                                         const __hip_bfloat16* __restrict__ W,
+                                                        const __hip_bfloat16* __restrict__ bias,
                                                         __hip_bfloat16* __restrict__ C, long cs, int B, int N,
                                                         int K) {
   const int lane = threadIdx.x & 63;
@@ -1971,28 +1976,142 @@ __global__ void __launch_bounds__(256) g16_plain_kernel(const __hip_bfloat16* __
   const int ncol0 = (blockIdx.x * (int)(blockDim.x >> 6) + (threadIdx.x >> 6)) * 16;
   if (ncol0 >= N) return;
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  const int Kp = ((K + 31) / 32) * 32;
+  const int KU = K & ~31;
+  const int arow_c = arow < B ? arow : B - 1;  // clamp: garbage rows unused
+  // two-phase body: issue all 16 loads of a 256-deep K block, then run the 8
+  // MFMAs — keeps 16 loads in flight per wave (the "late vmcnt" idiom for
+  // M<=16 streamed-weight GEMMs, cdna_hip_programming.md §6)
+  const int KU8 = KU & ~255;
+  for (int k0 = 0; k0 < KU8; k0 += 256) {
+    bf16x8 af[8], bf[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int k = k0 + u * 32 + kgrp * 8;
+      af[u] = g16_load8v(A, arow_c, as_, k);
+      bf[u] = g16_load8v(W, ncol0 + arow, K, k);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf[u], acc, 0, 0, 0);
+  }
 #pragma unroll 4
-  for (int k0 = 0; k0 < Kp; k0 += 32) {
+  for (int k0 = KU8; k0 < KU; k0 += 32) {
     const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8v(A, arow_c, as_, k);
+    bf16x8 b = g16_load8v(W, ncol0 + arow, K, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  if (K & 31) {
+    const int k = KU + kgrp * 8;
     bf16x8 a = g16_load8(A, arow, as_, k, K, B, arow);
     bf16x8 b = g16_load8(W, ncol0 + arow, K, k, K, N, ncol0 + arow);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  const float bv = bias ? __bfloat162float(bias[ncol0 + arow]) : 0.f;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    if (m < B) C[(long)m * cs + ncol0 + arow] = (__hip_bfloat16)(acc[r] + bv);
+  }
+}
+
+// split-K variant for long-K scan GEMMs (K ~ 6k: the representation-model
+// input): KS K-slices per 64-col tile accumulate into an fp32 scratch with
+// device atomics; the LAST-arriving workgroup of each tile (ticket counter,
+// one agent-scope acquire per tile — the in-launch split-K reduction of
+// cdna_hip_programming.md §6) converts the tile to bf16 and resets the
+// scratch/ticket for the next call.
+__global__ void __launch_bounds__(256) g16_splitk_kernel(const __hip_bfloat16* __restrict__ A, long as_,
+                                                         const __hip_bfloat16* __restrict__ W,
+                                                         const __hip_bfloat16* __restrict__ bias,
+                                                         float* __restrict__ scratch, int* __restrict__ tickets,
+                                                         __hip_bfloat16* __restrict__ C, long cs, int B, int N,
+                                                         int K, int KS) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int ntiles = N / 64;
+  const int tile = blockIdx.x % ntiles;
+  const int slice = blockIdx.x / ntiles;
+  const int ncol0 = tile * 64 + (int)(threadIdx.x >> 6) * 16;
+  const int klen32 = (((K + KS - 1) / KS + 31) / 32) * 32;
+  const int kbeg = slice * klen32;
+  const int kend = min(kbeg + klen32, K);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int arow_c = arow < B ? arow : B - 1;
+  const int KU = kbeg + ((kend - kbeg) & ~31);
+  const int KU8 = kbeg + ((KU - kbeg) & ~255);
+  for (int k0 = kbeg; k0 < KU8; k0 += 256) {
+    bf16x8 af[8], bf[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int k = k0 + u * 32 + kgrp * 8;
+      af[u] = g16_load8v(A, arow_c, as_, k);
+      bf[u] = g16_load8v(W, ncol0 + arow, K, k);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf[u], acc, 0, 0, 0);
+  }
+#pragma unroll 4
+  for (int k0 = KU8; k0 < KU; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8v(A, arow_c, as_, k);
+    bf16x8 b = g16_load8v(W, ncol0 + arow, K, k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  if (KU < kend) {
+    const int k = KU + kgrp * 8;
+    bf16x8 a = g16_load8(A, arow, as_, k, kend, B, arow);
+    bf16x8 b = g16_load8(W, ncol0 + arow, K, k, kend, N, ncol0 + arow);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int m = kgrp * 4 + r;
-    if (m < B) C[(long)m * cs + ncol0 + arow] = (__hip_bfloat16)acc[r];
+    atomicAdd(&scratch[(long)m * N + ncol0 + arow], acc[r]);
+  }
+  // tile episode hand-off: last arriver converts + resets
+  __shared__ int last;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    int t = __hip_atomic_fetch_add(&tickets[tile], 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    last = (t == KS - 1) ? 1 : 0;
+    if (last) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  if (last) {
+    for (int i = threadIdx.x; i < 16 * 64; i += blockDim.x) {
+      const int m = i >> 6, j = tile * 64 + (i & 63);
+      const float v = scratch[(long)m * N + j];
+      if (m < B) C[(long)m * cs + j] = (__hip_bfloat16)(v + (bias ? __bfloat162float(bias[j]) : 0.f));
+      scratch[(long)m * N + j] = 0.f;
+    }
+    if (threadIdx.x == 0) __hip_atomic_store(&tickets[tile], 0, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
   }
 }
 
-void g16_plain(const torch::Tensor& A, const torch::Tensor& W, torch::Tensor C) {
+void g16_splitk(const torch::Tensor& A, const torch::Tensor& W, const c10::optional<torch::Tensor>& bias,
+                torch::Tensor scratch, torch::Tensor tickets, torch::Tensor C, long KS) {
+  TORCH_CHECK(A.dim() == 2 && A.stride(1) == 1 && W.is_contiguous() && C.stride(1) == 1);
+  TORCH_CHECK(A.size(0) <= 16 && W.size(0) % 64 == 0 && A.scalar_type() == at::kBFloat16);
+  int B = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
+  TORCH_CHECK(scratch.numel() >= 16 * (long)N && tickets.numel() >= N / 64);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const __hip_bfloat16* bp = bias.has_value() ? (const __hip_bfloat16*)bias->data_ptr() : nullptr;
+  hipLaunchKernelGGL(g16_splitk_kernel, dim3((N / 64) * (int)KS), dim3(256), 0, stream.stream(),
+                     (const __hip_bfloat16*)A.data_ptr(), A.stride(0), (const __hip_bfloat16*)W.data_ptr(), bp,
+                     scratch.data_ptr<float>(), tickets.data_ptr<int>(), (__hip_bfloat16*)C.data_ptr(),
+                     C.stride(0), B, N, K, (int)KS);
+}
+
+void g16_plain(const torch::Tensor& A, const torch::Tensor& W, const c10::optional<torch::Tensor>& bias,
+               torch::Tensor C) {
   TORCH_CHECK(A.dim() == 2 && A.stride(1) == 1 && W.is_contiguous() && C.stride(1) == 1);
   TORCH_CHECK(A.size(0) <= 16 && W.size(0) % 64 == 0 && A.scalar_type() == at::kBFloat16);
   int B = (int)A.size(0), K = (int)A.size(1), N = (int)W.size(0);
   auto stream = at::cuda::getCurrentCUDAStream();
+  const __hip_bfloat16* bp = bias.has_value() ? (const __hip_bfloat16*)bias->data_ptr() : nullptr;
   hipLaunchKernelGGL(g16_plain_kernel, dim3(N / 64), dim3(256), 0, stream.stream(),
-                     (const __hip_bfloat16*)A.data_ptr(), A.stride(0), (const __hip_bfloat16*)W.data_ptr(),
+                     (const __hip_bfloat16*)A.data_ptr(), A.stride(0), (const __hip_bfloat16*)W.data_ptr(), bp,
                      (__hip_bfloat16*)C.data_ptr(), C.stride(0), B, N, K);
 }
 
@@ -2017,9 +2136,21 @@ __global__ void __launch_bounds__(256) g16_ln_silu_kernel(const __hip_bfloat16* 
   f32x4 acc[TILES];
 #pragma unroll
   for (int t = 0; t < TILES; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
-  const int Kp = ((K + 31) / 32) * 32;
-  for (int k0 = 0; k0 < Kp; k0 += 32) {
+  const int KU = K & ~31;
+  const int arow_c = arow < B ? arow : B - 1;
+#pragma unroll 4
+  for (int k0 = 0; k0 < KU; k0 += 32) {
     const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8v(A, arow_c, as_, k);
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
+      const int ncol0 = (t * 4 + wv) * 16;
+      bf16x8 b = g16_load8v(W, ncol0 + arow, K, k);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+  }
+  if (K & 31) {
+    const int k = KU + kgrp * 8;
     bf16x8 a = g16_load8(A, arow, as_, k, K, B, arow);
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
@@ -2118,9 +2249,21 @@ __global__ void __launch_bounds__(256) g16_cat_st_kernel(const __hip_bfloat16* _
   f32x4 acc[TILES];
 #pragma unroll
   for (int t = 0; t < TILES; ++t) acc[t] = (f32x4){0.f, 0.f, 0.f, 0.f};
-  const int Kp = ((K + 31) / 32) * 32;
-  for (int k0 = 0; k0 < Kp; k0 += 32) {
+  const int KU = K & ~31;
+  const int arow_c = arow < B ? arow : B - 1;
+#pragma unroll 4
+  for (int k0 = 0; k0 < KU; k0 += 32) {
     const int k = k0 + kgrp * 8;
+    bf16x8 a = g16_load8v(A, arow_c, as_, k);
+#pragma unroll
+    for (int t = 0; t < TILES; ++t) {
+      const int ncol0 = (t * 4 + wv) * 16;
+      bf16x8 b = g16_load8v(W, ncol0 + arow, K, k);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
+    }
+  }
+  if (K & 31) {
+    const int k = KU + kgrp * 8;
     bf16x8 a = g16_load8(A, arow, as_, k, K, B, arow);
 #pragma unroll
     for (int t = 0; t < TILES; ++t) {
@@ -2188,6 +2331,7 @@ void g16_cat_st(const torch::Tensor& A, const torch::Tensor& W, const torch::Ten
                        s.data_ptr<float>(), B, N, K, (int)KD, (float)unimix);                            \
     break;
   switch (N / 64) {
+    SHEEP_G16CS_CASE(1)
     SHEEP_G16CS_CASE(2)
     SHEEP_G16CS_CASE(4)
     SHEEP_G16CS_CASE(8)
@@ -2762,6 +2906,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("klbal_fwd", &klbal_fwd);
   m.def("klbal_bwd", &klbal_bwd);
   m.def("g16_plain", &g16_plain);
+  m.def("g16_splitk", &g16_splitk);
   m.def("g16_ln_silu", &g16_ln_silu);
   m.def("g16_cat_st", &g16_cat_st);
 }

@@ -144,11 +144,14 @@ class _RSSMScan(torch.autograd.Function):
             dt == torch.bfloat16
             and B <= 16
             and w1.dtype == dt
-            and D % 64 == 0 and D <= 512
             and (3 * H) % 64 == 0
             and P % 64 == 0
-            and SK % 64 == 0 and SK <= 1024
+            and SK % 64 == 0
         )
+        if g16:
+            sk_scratch = torch.zeros(16, P, device=dev, dtype=torch.float32)
+            sk_tickets = torch.zeros(P // 64, device=dev, dtype=torch.int32)
+            sk_ks = max(1, min(8, (H + E) // 832))
         w1t, w2t, w3t, w4t = w1.t(), w2.t(), w3.t(), w4.t()
         for t in range(T):
             f = f_all[t]
@@ -157,15 +160,19 @@ class _RSSMScan(torch.autograd.Function):
             # one kernel assembles the reset-masked GEMM inputs (z', a', h')
             ext.scan_resets_fwd(z_prev, iz, h_prev, ih, actions[t], f, x_s[t], hu_s[t], t == 0)
             if g16:
-                ext.g16_ln_silu(x_s[t], w1, lnw1, lnb1, g1_s[t], hu_s[t, :, H:],
-                                mr1_s[0, t], mr1_s[1, t], eps)
-                ext.g16_plain(hu_s[t], w2, y_s[t])
+                # hand-written MFMA GEMMs where they beat hipblaslt at M=16:
+                # plain N-split for mid-K, in-launch split-K for the long-K
+                # representation input (measured 2x at these shapes)
+                torch.mm(x_s[t], w1t, out=g1_s[t])
+                ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
+                ext.g16_plain(hu_s[t], w2, None, y_s[t])
                 ext.gru_gates_fwd_o(y_s[t], hu_s[t, :, :H], lnwg, lnbg, eps,
                                     h_seq[t], r_s[t, :, :H], mrg_s[0, t], mrg_s[1, t])
-                ext.g16_plain(r_s[t], w3, g3_s[t])
+                ext.g16_splitk(r_s[t], w3, None, sk_scratch, sk_tickets, g3_s[t], sk_ks)
                 ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
-                ext.g16_cat_st(p_s[t], w4, b4, urand_all[t], m_seq[t], z_seq[t], s_s[t],
-                               discrete, unimix)
+                ext.g16_plain(p_s[t], w4, b4, raw)
+                ext.cat_st_fwd_o(raw.view(B, S, discrete), urand_all[t], unimix,
+                                 m_seq[t].view(B, S, discrete), z_seq[t].view(B, S, discrete), s_s[t])
                 continue
             torch.mm(x_s[t], w1t, out=g1_s[t])
             ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])

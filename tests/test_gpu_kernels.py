@@ -555,13 +555,13 @@ def test_g16_gemm_kernels_match_torch():
     A = torch.randn(B, K, device="cuda", dtype=dt)
     W = torch.randn(N, K, device="cuda", dtype=dt) * 0.05
     C = torch.empty(B, N, device="cuda", dtype=dt)
-    ext.g16_plain(A, W, C)
+    ext.g16_plain(A, W, None, C)
     ref = (A.float() @ W.float().t())
     assert torch.allclose(C.float(), ref, atol=0.35, rtol=0.05), (C.float() - ref).abs().max()
     # partial rows
     A7 = A[:7].contiguous()
     C7 = torch.empty(7, N, device="cuda", dtype=dt)
-    ext.g16_plain(A7, W, C7)
+    ext.g16_plain(A7, W, None, C7)
     assert torch.allclose(C7.float(), ref[:7], atol=0.35, rtol=0.05)
 
     # fused LN+SiLU epilogue
