@@ -66,8 +66,12 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
 
     total_steps = int(cfg.algo.total_steps)
     policy_step = 0
-    last_log = 0
-    last_checkpoint = 0
+    if cfg.checkpoint.resume_from:
+        _st = torch.load(cfg.checkpoint.resume_from, map_location="cpu", weights_only=False)
+        policy_step = int(_st.get("policy_step", 0))
+        del _st
+    last_log = policy_step
+    last_checkpoint = policy_step
     num_iters = max(1, total_steps // (rollout_steps * num_envs)) if not cfg.dry_run else 1
 
     obs, _ = envs.reset(seed=cfg.seed)
@@ -148,12 +152,21 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
             and policy_step - last_checkpoint >= cfg.checkpoint.every
             or (it == num_iters and cfg.checkpoint.save_last)
         ):
-            # the player holds the freshest synced weights; save locally (the
-            # trainer-side optimizer state is rebuilt on resume)
+            # the player holds the freshest synced weights; the lead trainer
+            # ships its optimizer state over the player<->trainer group so the
+            # checkpoint is complete (resume restores it on every trainer)
             last_checkpoint = policy_step
+            runtime.broadcast_object_list(["__send_opt__"], src=0, group=pt_group)
+            opt_payload: List[Any] = [None]
+            runtime.broadcast_object_list(opt_payload, src=1, group=pt_group)
             ckpt_path = os.path.join(log_dir, "checkpoint", f"ckpt_{policy_step}_0.ckpt")
             os.makedirs(os.path.dirname(ckpt_path), exist_ok=True)
-            torch.save({"agent": agent.state_dict(), "policy_step": policy_step}, ckpt_path)
+            torch.save(
+                {"agent": agent.state_dict(), "optimizer": opt_payload[0], "policy_step": policy_step},
+                ckpt_path,
+            )
+        else:
+            runtime.broadcast_object_list(["__noop__"], src=0, group=pt_group)
 
     # shutdown sentinel to the trainers (reference :344)
     runtime.scatter_object_list([None], [None] + [-1] * n_trainers, src=0, group=world_group)
@@ -176,11 +189,22 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
     probe.close()
 
     agent = PPOAgent(obs_space, action_space, cfg.algo).to(device)
+    resume_opt = None
+    if cfg.checkpoint.resume_from and rank == 1:
+        _st = torch.load(cfg.checkpoint.resume_from, map_location="cpu", weights_only=False)
+        agent.load_state_dict(_st["agent"])
+        resume_opt = _st.get("optimizer")
     gs = GradSync(agent, bucket_cap_mb=runtime.bucket_cap_mb, process_group=opt_group)
     gs.broadcast_params(src=1)
     agent._grad_sync = gs  # noqa: SLF001
     runtime._synced_modules.append(agent)
     optimizer = FusedAdam(agent.parameters(), lr=cfg.algo.optimizer.lr, eps=cfg.algo.optimizer.get("eps", 1e-8))
+    if cfg.checkpoint.resume_from:
+        # rank 1 saved its optimizer state; every trainer resumes from it
+        payload: List[Any] = [resume_opt]
+        runtime.broadcast_object_list(payload, src=1, group=opt_group)
+        if payload[0] is not None:
+            optimizer.load_state_dict(payload[0])
 
     flat = parameters_to_vector(agent.parameters()).detach()
     if rank == 1:
@@ -243,6 +267,20 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
                 "Loss/entropy_loss": losses["entropy"] / max(n_batches, 1),
             }
             runtime.broadcast_object_list([metrics], src=1, group=pt_group)
+            # checkpoint control: the player either requests the optimizer
+            # state for a complete checkpoint or sends a no-op
+            ctrl: List[Any] = [None]
+            runtime.broadcast_object_list(ctrl, src=0, group=pt_group)
+            if ctrl[0] == "__send_opt__":
+                opt_sd = optimizer.state_dict()
+                opt_sd = {
+                    "state": {
+                        k: {kk: (vv.cpu() if torch.is_tensor(vv) else vv) for kk, vv in v.items()}
+                        for k, v in opt_sd["state"].items()
+                    },
+                    "param_groups": opt_sd["param_groups"],
+                }
+                runtime.broadcast_object_list([opt_sd], src=1, group=pt_group)
 
 
 @register_algorithm(name="ppo_decoupled", decoupled=True)

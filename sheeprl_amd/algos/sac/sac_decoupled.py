@@ -80,6 +80,11 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
     policy_step = 0
     last_log = 0
     last_checkpoint = 0
+    trainer_state = None
+    if cfg.checkpoint.resume_from:
+        _st = torch.load(cfg.checkpoint.resume_from, map_location="cpu", weights_only=False)
+        last_checkpoint = int(_st.get("policy_step", 0))
+        del _st
 
     def flat_obs(o):
         return np.concatenate(
@@ -124,10 +129,17 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
             if gradient_steps > 0 and len(rb) > 0:
                 # one scatter round per training burst: each trainer gets its
                 # own stack of batches (reference :240-257)
+                want_ckpt = (
+                    cfg.checkpoint.every > 0
+                    and policy_step - last_checkpoint >= cfg.checkpoint.every
+                    or (iter_num == total_iters and cfg.checkpoint.save_last)
+                )
                 chunks: List[Any] = [None]
                 for _ in range(n_trainers):
                     s = rb.sample(cfg.algo.per_rank_batch_size * gradient_steps)
-                    chunks.append({k: v[0] for k, v in s.items()})
+                    c = {k: v[0] for k, v in s.items()}
+                    c["__ckpt__"] = want_ckpt
+                    chunks.append(c)
                 out: List[Any] = [None]
                 runtime.scatter_object_list(out, chunks, src=0, group=world_group)
                 runtime.broadcast(flat, src=1, group=pt_group)
@@ -136,6 +148,10 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
                 runtime.broadcast_object_list(payload, src=1, group=pt_group)
                 for k, v in (payload[0] or {}).items():
                     aggregator.update(k, v)
+                if want_ckpt:
+                    opt_payload: List[Any] = [None]
+                    runtime.broadcast_object_list(opt_payload, src=1, group=pt_group)
+                    trainer_state = opt_payload[0]
 
         if policy_step - last_log >= cfg.metric.log_every or iter_num == total_iters or cfg.dry_run:
             runtime.log_dict(aggregator.compute(), policy_step)
@@ -151,7 +167,11 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
             last_checkpoint = policy_step
             ckpt_path = os.path.join(log_dir, "checkpoint", f"ckpt_{policy_step}_0.ckpt")
             os.makedirs(os.path.dirname(ckpt_path), exist_ok=True)
-            torch.save({"actor": agent.actor.state_dict(), "policy_step": policy_step}, ckpt_path)
+            state = {"actor": agent.actor.state_dict(), "policy_step": policy_step}
+            if trainer_state is not None:
+                state.update(trainer_state)  # full agent + the three optimizer states
+            torch.save(state, ckpt_path)
+            trainer_state = None
 
     runtime.scatter_object_list([None], [None] + [-1] * n_trainers, src=0, group=world_group)
     envs.close()
@@ -170,6 +190,11 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
     probe.close()
 
     agent = _agent_for(cfg, obs_space, action_space, device)
+    resume_state = None
+    if cfg.checkpoint.resume_from and rank == 1:
+        resume_state = torch.load(cfg.checkpoint.resume_from, map_location="cpu", weights_only=False)
+        if "agent" in resume_state:
+            agent.load_state_dict(resume_state["agent"])
     gs = GradSync(agent, bucket_cap_mb=runtime.bucket_cap_mb, process_group=opt_group)
     gs.broadcast_params(src=1)
     agent._grad_sync = gs  # noqa: SLF001
@@ -179,6 +204,19 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
     qf_optimizer = FusedAdam(agent.qfs.parameters(), lr=cfg.algo.critic.optimizer.lr)
     actor_optimizer = FusedAdam(agent.actor.parameters(), lr=cfg.algo.actor.optimizer.lr)
     alpha_optimizer = FusedAdam([agent.log_alpha], lr=cfg.algo.alpha.optimizer.lr)
+    if cfg.checkpoint.resume_from:
+        # rank 1 loaded the checkpoint; every trainer resumes from its states
+        payload: List[Any] = [
+            {k: (resume_state or {}).get(k) for k in ("qf_optimizer", "actor_optimizer", "alpha_optimizer")}
+        ]
+        runtime.broadcast_object_list(payload, src=1, group=opt_group)
+        for name, opt in (
+            ("qf_optimizer", qf_optimizer),
+            ("actor_optimizer", actor_optimizer),
+            ("alpha_optimizer", alpha_optimizer),
+        ):
+            if payload[0].get(name):
+                opt.load_state_dict(payload[0][name])
     aggregator = MetricAggregator({k: "mean" for k in AGGREGATOR_KEYS})
 
     if rank == 1:
@@ -191,6 +229,7 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
         chunk = out[0]
         if isinstance(chunk, int) and chunk == -1:
             return
+        want_ckpt = bool(chunk.pop("__ckpt__", False))
         data = {k: torch.as_tensor(np.ascontiguousarray(v), device=device).float() for k, v in chunk.items()}
         bs = cfg.algo.per_rank_batch_size
         n = data["obs"].shape[0]
@@ -205,6 +244,24 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
             runtime.broadcast(parameters_to_vector(agent.actor.parameters()).detach(), src=1, group=pt_group)
             runtime.broadcast_object_list([aggregator.compute()], src=1, group=pt_group)
             aggregator.reset()
+            if want_ckpt:
+                def _cpu_sd(opt):
+                    sd = opt.state_dict()
+                    return {
+                        "state": {
+                            k: {kk: (vv.cpu() if torch.is_tensor(vv) else vv) for kk, vv in v.items()}
+                            for k, v in sd["state"].items()
+                        },
+                        "param_groups": sd["param_groups"],
+                    }
+
+                payload = {
+                    "agent": {k: v.cpu() for k, v in agent.state_dict().items()},
+                    "qf_optimizer": _cpu_sd(qf_optimizer),
+                    "actor_optimizer": _cpu_sd(actor_optimizer),
+                    "alpha_optimizer": _cpu_sd(alpha_optimizer),
+                }
+                runtime.broadcast_object_list([payload], src=1, group=pt_group)
 
 
 @register_algorithm(name="sac_decoupled", decoupled=True)

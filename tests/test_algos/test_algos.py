@@ -272,6 +272,63 @@ def test_ppo_decoupled(tmp_path):
     )
 
 
+@pytest.mark.timeout(420)
+def test_ppo_decoupled_checkpoint_resume(tmp_path):
+    args = standard_args(
+        tmp_path,
+        [
+            "exp=ppo_decoupled",
+            "env=dummy",
+            "algo.total_steps=64",
+            "algo.rollout_steps=8",
+            "algo.update_epochs=1",
+            "algo.per_rank_batch_size=8",
+            "algo.run_test=False",
+            "dry_run=False",
+            "checkpoint.save_last=True",
+        ],
+    )
+    _run(tmp_path, args, devices="2")
+    import glob
+
+    ckpts = glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True)
+    assert ckpts, "no decoupled checkpoint written"
+    import torch as _t
+
+    state = _t.load(ckpts[-1], map_location="cpu", weights_only=False)
+    assert "optimizer" in state and state["optimizer"] is not None, "trainer optimizer state missing"
+    _run(tmp_path, args + [f"checkpoint.resume_from={ckpts[-1]}"], devices="2")
+
+
+@pytest.mark.timeout(420)
+def test_sac_decoupled_checkpoint_resume(tmp_path):
+    args = standard_args(
+        tmp_path,
+        [
+            "exp=sac_decoupled",
+            "env=dummy",
+            "algo.total_steps=32",
+            "algo.learning_starts=4",
+            "algo.per_rank_batch_size=8",
+            "algo.replay_ratio=0.5",
+            "buffer.size=128",
+            "algo.run_test=False",
+            "dry_run=False",
+            "checkpoint.save_last=True",
+        ],
+    )
+    _run(tmp_path, args, devices="2")
+    import glob
+
+    ckpts = glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True)
+    assert ckpts, "no decoupled checkpoint written"
+    import torch as _t
+
+    state = _t.load(ckpts[-1], map_location="cpu", weights_only=False)
+    assert "qf_optimizer" in state, "trainer optimizer states missing"
+    _run(tmp_path, args + [f"checkpoint.resume_from={ckpts[-1]}"], devices="2")
+
+
 def test_decoupled_requires_multi_device(tmp_path):
     with pytest.raises(Exception):
         _run(
