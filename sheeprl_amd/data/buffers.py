@@ -403,9 +403,18 @@ class EpisodeBuffer:
         self._n_envs = n_envs
         self._obs_keys = tuple(obs_keys)
         self._prioritize_ends = prioritize_ends
+        self._memmap = memmap
+        self._memmap_dir = Path(memmap_dir) if memmap_dir is not None else None
+        if memmap and self._memmap_dir is not None:
+            self._memmap_dir.mkdir(parents=True, exist_ok=True)
+        self._episode_counter = 0
         self._episodes: List[Dict[str, np.ndarray]] = []
         self._open: List[Optional[Dict[str, List[np.ndarray]]]] = [None] * n_envs
         self._cum_lengths: List[int] = []
+
+    @property
+    def is_memmap(self) -> bool:
+        return self._memmap
 
     @property
     def buffer(self) -> List[Dict[str, np.ndarray]]:
@@ -466,14 +475,40 @@ class EpisodeBuffer:
         if length > self._buffer_size:
             ep = {k: v[-self._buffer_size :] for k, v in ep.items()}
             length = self._buffer_size
+        if self._memmap and self._memmap_dir is not None:
+            # spill the committed episode to disk (reference buffers.py:961-991):
+            # one MemmapArray file per key per episode; only the sampled
+            # windows are faulted back in
+            ep_dir = self._memmap_dir / f"episode_{self._episode_counter}"
+            ep_dir.mkdir(parents=True, exist_ok=True)
+            self._episode_counter += 1
+            spilled = {}
+            for k, v in ep.items():
+                ma = MemmapArray.from_array(v, filename=ep_dir / f"{k}.memmap")
+                ma.has_ownership = False  # eviction unlinks explicitly
+                spilled[k] = ma.array
+            ep = spilled
         self._episodes.append(ep)
-        # evict the oldest episodes (incl. memmap dirs in the reference,
+        # evict the oldest episodes (incl. memmap files, reference
         # buffers.py:993-1014) until total length fits
         while len(self) > self._buffer_size and len(self._episodes) > 1:
-            self._episodes.pop(0)
+            old_ep = self._episodes.pop(0)
+            self._evict_files(old_ep)
         if len(self) > self._buffer_size:
             ep = self._episodes[0]
-            self._episodes[0] = {k: v[-self._buffer_size :] for k, v in ep.items()}
+            self._episodes[0] = {k: np.asarray(v)[-self._buffer_size :] for k, v in ep.items()}
+
+    @staticmethod
+    def _evict_files(ep: Dict[str, np.ndarray]) -> None:
+        import os
+
+        for v in ep.values():
+            fn = getattr(v, "filename", None)
+            if fn:
+                try:
+                    os.unlink(fn)
+                except OSError:
+                    pass
 
     def sample(
         self,

@@ -172,3 +172,27 @@ class TestMemmap:
         rb.add(_mk_data(8, 1))
         assert rb.is_memmap
         assert (tmp_path / "rb" / "obs.memmap").exists()
+
+
+def test_episode_buffer_memmap(tmp_path):
+    import numpy as np
+    from sheeprl_amd.data import EpisodeBuffer
+
+    rb = EpisodeBuffer(40, 4, n_envs=1, obs_keys=("obs",), memmap=True, memmap_dir=tmp_path / "eb")
+    assert rb.is_memmap
+    rng = np.random.default_rng(0)
+    for i in range(12):
+        term = np.zeros((10, 1, 1), dtype=np.float32)
+        term[-1] = 1.0  # one 10-step episode per add
+        rb.add({"obs": rng.normal(size=(10, 1, 3)).astype(np.float32),
+                "terminated": term, "truncated": np.zeros_like(term)})
+    files = list((tmp_path / "eb").rglob("*.memmap"))
+    assert files, "no memmap spill files written"
+    # capacity 40 -> at most 4 live episodes; older files must be evicted
+    assert len(rb.buffer) <= 4
+    live = {getattr(v, "filename", None) for ep in rb.buffer for v in ep.values()}
+    on_disk = {str(f) for f in files}
+    assert live <= on_disk | {None}
+    assert len(on_disk) <= 3 * 5  # 3 keys x (4 live + transient)
+    s = rb.sample(6, n_samples=2)
+    assert s["obs"].shape == (2, 4, 6, 3)
