@@ -238,40 +238,19 @@ class _RSSMScan(torch.autograd.Function):
         gh_carry = torch.empty(B, H, device=dev, dtype=dt)   # written at t before read at t-1
         gz_carry = torch.empty(B, SK, device=dev, dtype=dt)
 
-        # dX GEMMs via the hand-written M=16 MFMA kernel where shapes allow:
-        # it wants the weight as [out, in] rows, so cache the transposes once
-        # per backward (three small copies, reused for all T steps)
-        g16b = (
-            dt == torch.bfloat16 and B <= 16
-            and P % 64 == 0 and (H + E) % 64 == 0 and (H + D) % 64 == 0
-        )
-        if g16b:
-            w4T = w4.t().contiguous()
-            w3T = w3.t().contiguous()
-            w2T = w2.t().contiguous()
-
         for t in range(T - 1, -1, -1):
             f = f_all[t]
             zc = gz_carry.view(B, S, discrete) if t < T - 1 else None
             hc = gh_carry if t < T - 1 else None
             ext.cat_st_bwd_o(g_m_seq[t].view(B, S, discrete), g_z_seq[t].view(B, S, discrete), zc,
                              s_s[t], unimix, graw_s[t].view(B, S, discrete))
-            if g16b:
-                ext.g16_plain(graw_s[t], w4T, None, gp)
-            else:
-                torch.mm(graw_s[t], w4, out=gp)
+            torch.mm(graw_s[t], w4, out=gp)
             ext.ln_act_bwd_acc(gp, g3_s[t], lnw3, lnb3, mr3_s[0, t], mr3_s[1, t], True,
                                gg3_s[t], glnw3, glnb3)
-            if g16b:
-                ext.g16_plain(gg3_s[t], w3T, None, gr_s[t])
-            else:
-                torch.mm(gg3_s[t], w3, out=gr_s[t])
+            torch.mm(gg3_s[t], w3, out=gr_s[t])
             ext.gru_gates_bwd_acc(g_h_seq[t], hc, gr_s[t, :, :H], y_s[t], hu_s[t, :, :H], lnwg, lnbg,
                                   mrg_s[0, t], mrg_s[1, t], gy_s[t], ghp, glnwg, glnbg)
-            if g16b:
-                ext.g16_plain(gy_s[t], w2T, None, ghu)
-            else:
-                torch.mm(gy_s[t], w2, out=ghu)
+            torch.mm(gy_s[t], w2, out=ghu)
             ext.ln_act_bwd_acc(ghu[:, H:], g1_s[t], lnw1, lnb1, mr1_s[0, t], mr1_s[1, t], True,
                                gg1_s[t], glnw1, glnb1)
             torch.mm(gg1_s[t], w1, out=gx)
