@@ -49,7 +49,9 @@ class CUDAGraphStep:
         torch.cuda.current_stream().wait_stream(side)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph, pool=pool):
+        # thread_local error mode: RCCL's proxy threads issue HIP calls of
+        # their own during capture; global mode would abort the capture
+        with torch.cuda.graph(self.graph, pool=pool, capture_error_mode="thread_local"):
             fn(self.static)
 
     def __call__(self, inputs: Dict[str, torch.Tensor]) -> None:
