@@ -507,6 +507,51 @@ class Actor(nn.Module):
         return tuple(actions_list), tuple(dists)
 
 
+class MinedojoActor(Actor):
+    """Action-masked actor for MineDojo (reference dreamer_v3/agent.py:848-933).
+
+    Head 0 (action type) is masked by ``mask["mask_action_type"]``; the
+    craft-argument head is masked per sample where the sampled action type is
+    15 (craft); the equip/place/destroy-argument head where the type is 16/17
+    (equip/place, ``mask_equip_place``) or 18 (destroy, ``mask_destroy``).
+    Masks apply AFTER the unimix transform, as -inf logits (probability 0),
+    vectorized over [T, B] instead of the reference's python loops.
+    """
+
+    def forward(
+        self, state: Tensor, greedy: bool = False, mask: Optional[Dict[str, Tensor]] = None
+    ) -> Tuple[Tuple[Tensor, ...], Tuple[Any, ...]]:
+        out = self.model(state)
+        actions_list: List[Tensor] = []
+        dists: List[Any] = []
+        functional_action: Optional[Tensor] = None
+        neg_inf = -torch.inf
+        for i, head in enumerate(self.mlp_heads):
+            logits = head(out).float()
+            K = logits.shape[-1]
+            mixed = torch.log((1 - self._unimix) * torch.softmax(logits, -1) + self._unimix / K)
+            if mask is not None:
+                if i == 0:
+                    mixed = torch.where(mask["mask_action_type"].expand_as(mixed), mixed, neg_inf)
+                elif i == 1:
+                    sel = (functional_action == 15).unsqueeze(-1)  # craft
+                    blocked = sel & ~mask["mask_craft_smelt"].expand_as(mixed)
+                    mixed = torch.where(blocked, neg_inf, mixed)
+                elif i == 2:
+                    sel_ep = ((functional_action == 16) | (functional_action == 17)).unsqueeze(-1)
+                    sel_d = (functional_action == 18).unsqueeze(-1)
+                    blocked = (sel_ep & ~mask["mask_equip_place"].expand_as(mixed)) | (
+                        sel_d & ~mask["mask_destroy"].expand_as(mixed)
+                    )
+                    mixed = torch.where(blocked, neg_inf, mixed)
+            dist = OneHotCategoricalST(logits=mixed)
+            dists.append(dist)
+            actions_list.append(dist.mode if greedy else dist.rsample())
+            if functional_action is None:
+                functional_action = actions_list[0].argmax(dim=-1)
+        return tuple(actions_list), tuple(dists)
+
+
 # ---------------------------------------------------------------------------
 # player
 # ---------------------------------------------------------------------------
@@ -704,7 +749,8 @@ def build_agent(
     )
     world_model = WorldModel(encoder, rssm, observation_model, reward_model, continue_model)
 
-    actor = Actor(
+    actor_cls = MinedojoActor if str(cfg.env.id).startswith("minedojo") else Actor
+    actor = actor_cls(
         latent_state_size=latent_state_size,
         actions_dim=actions_dim,
         is_continuous=is_continuous,

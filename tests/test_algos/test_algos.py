@@ -635,3 +635,26 @@ def test_eval_and_registration_cli(tmp_path):
         assert (tmp_path / "registry").exists()
     finally:
         os.chdir(cwd)
+
+
+def test_minedojo_actor_masking():
+    import torch
+    from sheeprl_amd.algos.dreamer_v3.agent import MinedojoActor
+
+    torch.manual_seed(0)
+    a = MinedojoActor(24, [19, 5, 7], False, dense_units=16, mlp_layers=2, unimix=0.01)
+    state = torch.randn(3, 4, 24)
+    mask = {
+        "mask_action_type": torch.ones(3, 4, 19, dtype=torch.bool),
+        "mask_craft_smelt": torch.zeros(3, 4, 5, dtype=torch.bool),
+        "mask_equip_place": torch.ones(3, 4, 7, dtype=torch.bool),
+        "mask_destroy": torch.zeros(3, 4, 7, dtype=torch.bool),
+    }
+    mask["mask_action_type"][..., 3] = False
+    mask["mask_craft_smelt"][..., 1] = True
+    for greedy in (False, True):
+        acts, dists = a(state, greedy=greedy, mask=mask)
+        assert (acts[0].argmax(-1) != 3).all()
+        crafted = acts[0].argmax(-1) == 15
+        if crafted.any():
+            assert (acts[1].argmax(-1)[crafted] == 1).all()
