@@ -1283,44 +1283,6 @@ __global__ void masked_lerp_bwd_kernel(const T* __restrict__ g, const T* __restr
   }
 }
 
-// Strided variant for the fused scan: x and y may be row-strided 2-D views
-// (slices of stacked [T, B, *] buffers); init (when present) is contiguous.
-template <typename T, bool HAS_INIT>
-__global__ void masked_lerp_fwd_s_kernel(const T* __restrict__ x, const T* __restrict__ init,
-                                         const T* __restrict__ f, T* __restrict__ y, long rows, int cols,
-                                         long xs, long ys) {
-  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < rows * (long)cols;
-       i += (long)gridDim.x * blockDim.x) {
-    const long b = i / cols;
-    const long c = i - b * cols;
-    float fb = ld(f, b);
-    float v = (1.f - fb) * ld(x, b * xs + c);
-    if (HAS_INIT) v += fb * ld(init, i);
-    st(y, b * ys + c, v);
-  }
-}
-
-void masked_lerp_fwd_o(const torch::Tensor& x, const c10::optional<torch::Tensor>& init,
-                       const torch::Tensor& f, torch::Tensor y) {
-  TORCH_CHECK(x.dim() == 2 && x.stride(1) == 1 && y.dim() == 2 && y.stride(1) == 1, "masked_lerp_fwd_o shapes");
-  long rows = x.size(0);
-  int cols = (int)x.size(1);
-  long n = rows * (long)cols;
-  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
-  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "masked_lerp_fwd_o", [&] {
-    using T = scalar_t;
-    if (init.has_value())
-      hipLaunchKernelGGL((masked_lerp_fwd_s_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
-                         (const T*)x.data_ptr(), (const T*)init->data_ptr(), (const T*)f.data_ptr(),
-                         (T*)y.data_ptr(), rows, cols, x.stride(0), y.stride(0));
-    else
-      hipLaunchKernelGGL((masked_lerp_fwd_s_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
-                         (const T*)x.data_ptr(), nullptr, (const T*)f.data_ptr(), (T*)y.data_ptr(), rows, cols,
-                         x.stride(0), y.stride(0));
-  });
-}
-
 torch::Tensor masked_lerp_fwd(const torch::Tensor& x, const c10::optional<torch::Tensor>& init,
                               const torch::Tensor& f) {
   CHECK_IN(x);
@@ -1340,50 +1302,6 @@ torch::Tensor masked_lerp_fwd(const torch::Tensor& x, const c10::optional<torch:
                          (const T*)x.data_ptr(), nullptr, (const T*)f.data_ptr(), (T*)y.data_ptr(), rows, cols);
   });
   return y;
-}
-
-// Accumulate-variant for the fused scan backward: g may be row-strided; gx is
-// a caller-provided contiguous output; gacc (optional) is a fp32 accumulator
-// of the same 2-D shape receiving the init-side gradient (f * g), summed
-// across scan steps without atomics (steps are stream-ordered).
-template <typename T, bool HAS_ACC>
-__global__ void masked_lerp_bwd_s_kernel(const T* __restrict__ g, const T* __restrict__ g2,
-                                         const T* __restrict__ f, T* __restrict__ gx,
-                                         float* __restrict__ gacc, long rows, int cols, long gs) {
-  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < rows * (long)cols;
-       i += (long)gridDim.x * blockDim.x) {
-    const long b = i / cols;
-    const long c = i - b * cols;
-    float fb = ld(f, b);
-    float gv = ld(g, b * gs + c) + (g2 ? ld(g2, i) : 0.f);
-    st(gx, i, (1.f - fb) * gv);
-    if (HAS_ACC) gacc[i] += fb * gv;
-  }
-}
-
-void masked_lerp_bwd_acc(const torch::Tensor& g, const c10::optional<torch::Tensor>& g2,
-                         const torch::Tensor& f, torch::Tensor gx,
-                         const c10::optional<torch::Tensor>& gacc) {
-  TORCH_CHECK(g.dim() == 2 && g.stride(1) == 1 && gx.is_contiguous(), "masked_lerp_bwd_acc shapes");
-  long rows = g.size(0);
-  int cols = (int)g.size(1);
-  long n = rows * (long)cols;
-  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
-  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
-  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, g.scalar_type(), "masked_lerp_bwd_acc", [&] {
-    using T = scalar_t;
-    const T* g2p = g2.has_value() ? (const T*)g2->data_ptr() : nullptr;
-    if (gacc.has_value()) {
-      TORCH_CHECK(gacc->scalar_type() == at::kFloat && gacc->is_contiguous(), "gacc must be contiguous fp32");
-      hipLaunchKernelGGL((masked_lerp_bwd_s_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
-                         (const T*)g.data_ptr(), g2p, (const T*)f.data_ptr(), (T*)gx.data_ptr(),
-                         gacc->data_ptr<float>(), rows, cols, g.stride(0));
-    } else {
-      hipLaunchKernelGGL((masked_lerp_bwd_s_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
-                         (const T*)g.data_ptr(), g2p, (const T*)f.data_ptr(), (T*)gx.data_ptr(), nullptr, rows,
-                         cols, g.stride(0));
-    }
-  });
 }
 
 std::vector<torch::Tensor> masked_lerp_bwd(const torch::Tensor& g, const torch::Tensor& f, bool has_init) {
@@ -2889,8 +2807,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_act_bwd_acc", &ln_act_bwd_acc);
   m.def("gru_gates_fwd_o", &gru_gates_fwd_o);
   m.def("gru_gates_bwd_acc", &gru_gates_bwd_acc);
-  m.def("masked_lerp_fwd_o", &masked_lerp_fwd_o);
-  m.def("masked_lerp_bwd_acc", &masked_lerp_bwd_acc);
   m.def("cat_st_fwd_o", &cat_st_fwd_o);
   m.def("cat_st_bwd_o", &cat_st_bwd_o);
   m.def("ema_update", &ema_update);
