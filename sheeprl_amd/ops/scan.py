@@ -137,9 +137,8 @@ class _RSSMScan(torch.autograd.Function):
             return h_seq, z_seq, m_seq
         h0 = torch.zeros(B, H, device=dev, dtype=dt)
         z0 = torch.zeros(B, SK, device=dev, dtype=dt)
-        # hand-written M=16 MFMA kernels (with fused LN / categorical-ST
-        # epilogues) replace hipblaslt for the step GEMMs when shapes allow —
-        # each fused call is one launch where torch needs two
+        # hand-written split-K M=16 MFMA kernel for the long-K
+        # representation GEMM (the one shape where it beats hipblaslt)
         g16 = (
             dt == torch.bfloat16
             and B <= 16
