@@ -277,10 +277,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
         state.get("world_model"), state.get("actor"), state.get("critic"), state.get("target_critic"),
     )
     world_optimizer = FusedAdam(world_model.parameters(), lr=cfg.algo.world_model.optimizer.lr,
-                                eps=cfg.algo.world_model.optimizer.eps)
-    actor_optimizer = FusedAdam(actor.parameters(), lr=cfg.algo.actor.optimizer.lr, eps=cfg.algo.actor.optimizer.eps)
+                                eps=cfg.algo.world_model.optimizer.eps,
+                                weight_decay=cfg.algo.world_model.optimizer.get("weight_decay", 0.0))
+    actor_optimizer = FusedAdam(actor.parameters(), lr=cfg.algo.actor.optimizer.lr, eps=cfg.algo.actor.optimizer.eps,
+                                weight_decay=cfg.algo.actor.optimizer.get("weight_decay", 0.0))
     critic_optimizer = FusedAdam(critic.parameters(), lr=cfg.algo.critic.optimizer.lr,
-                                 eps=cfg.algo.critic.optimizer.eps)
+                                 eps=cfg.algo.critic.optimizer.eps,
+                                 weight_decay=cfg.algo.critic.optimizer.get("weight_decay", 0.0))
 
     aggregator = MetricAggregator({k: "mean" for k in AGGREGATOR_KEYS})
 

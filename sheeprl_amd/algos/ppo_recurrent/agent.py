@@ -21,11 +21,30 @@ from sheeprl_amd.parallel import Runtime
 
 
 class RecurrentModel(nn.Module):
-    def __init__(self, input_size: int, lstm_hidden_size: int, dense_units: int, layer_norm: bool = True) -> None:
+    def __init__(
+        self,
+        input_size: int,
+        lstm_hidden_size: int,
+        dense_units: int,
+        layer_norm: bool = True,
+        pre_rnn_mlp: bool = False,
+        post_rnn_mlp: bool = False,
+    ) -> None:
         super().__init__()
-        self.pre_mlp = MLP(input_size, None, [dense_units], activation="relu", layer_norm=layer_norm)
-        self.lstm = nn.LSTM(dense_units, lstm_hidden_size, batch_first=False)
-        self.post_mlp = MLP(lstm_hidden_size, None, [lstm_hidden_size], activation="relu", layer_norm=layer_norm)
+        # pre/post MLPs around the LSTM are OFF by default, matching the
+        # reference's rnn.pre_rnn_mlp.apply / post_rnn_mlp.apply defaults
+        self.pre_mlp = (
+            MLP(input_size, None, [dense_units], activation="relu", layer_norm=layer_norm)
+            if pre_rnn_mlp
+            else nn.Identity()
+        )
+        lstm_in = dense_units if pre_rnn_mlp else input_size
+        self.lstm = nn.LSTM(lstm_in, lstm_hidden_size, batch_first=False)
+        self.post_mlp = (
+            MLP(lstm_hidden_size, None, [lstm_hidden_size], activation="relu", layer_norm=layer_norm)
+            if post_rnn_mlp
+            else nn.Identity()
+        )
         self.output_dim = lstm_hidden_size
         self.hidden_size = lstm_hidden_size
 
@@ -35,6 +54,10 @@ class RecurrentModel(nn.Module):
         out, states = self.lstm(x, states)
         shape = out.shape
         return self.post_mlp(out.reshape(-1, shape[-1])).view(*shape[:-1], -1), states
+
+    def _post(self, out: Tensor) -> Tensor:
+        shape = out.shape
+        return self.post_mlp(out.reshape(-1, shape[-1])).view(*shape[:-1], -1)
 
     def masked_scan(
         self, x: Tensor, is_first: Tensor, states: Tuple[Tensor, Tensor]
@@ -50,8 +73,7 @@ class RecurrentModel(nn.Module):
             out, (h, c) = self.lstm(x[t : t + 1], (h.contiguous(), c.contiguous()))
             outs.append(out)
         out = torch.cat(outs, dim=0)
-        shape = out.shape
-        return self.post_mlp(out.reshape(-1, shape[-1])).view(*shape[:-1], -1), (h, c)
+        return self._post(out), (h, c)
 
 
 class RecurrentPPOAgent(nn.Module):
@@ -94,6 +116,8 @@ class RecurrentPPOAgent(nn.Module):
             cfg_algo.rnn.lstm.hidden_size,
             cfg_algo.rnn.get("dense_units", 64),
             layer_norm=cfg_algo.rnn.get("layer_norm", True),
+            pre_rnn_mlp=cfg_algo.rnn.get("pre_rnn_mlp", {}).get("apply", False),
+            post_rnn_mlp=cfg_algo.rnn.get("post_rnn_mlp", {}).get("apply", False),
         )
         units = cfg_algo.actor.dense_units
         self.actor_torso = MLP(self.rnn.output_dim, None, [units] * cfg_algo.actor.mlp_layers, activation="relu")

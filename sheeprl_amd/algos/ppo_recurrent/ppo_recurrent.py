@@ -86,8 +86,10 @@ def main(runtime: Runtime, cfg: Any) -> None:
             for _ in range(rollout_steps):
                 t_obs = {k: v.unsqueeze(0) for k, v in prepare_obs(obs, cfg, device).items()}
                 # reset recurrent states for envs that restarted
+                # (reference ppo_recurrent.yaml: reset_recurrent_state_on_done)
                 mask = torch.as_tensor(1.0 - is_first_np, device=device).view(1, -1, 1).float()
-                states = (states[0] * mask, states[1] * mask)
+                if cfg.algo.get("reset_recurrent_state_on_done", True):
+                    states = (states[0] * mask, states[1] * mask)
                 prev_actions = prev_actions * mask
                 with torch.no_grad():
                     actions, logprobs, values, states = player.get_actions(t_obs, prev_actions, states)
