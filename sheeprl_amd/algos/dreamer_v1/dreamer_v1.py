@@ -359,6 +359,32 @@ def train(
         aggregator.update("Loss/value_loss", value_loss.detach())
 
 
+def _capture_train_step(
+    runtime, world_model, actor, critic,
+    world_optimizer, actor_optimizer, critic_optimizer, example_batch, cfg,
+):
+    """hipGraph-capture the DV1 gradient step; None on capture failure."""
+    from sheeprl_amd.parallel.graphs import CUDAGraphStep
+    from sheeprl_amd.utils.metric import MetricAggregator
+
+    def train_fn(batch):
+        was_disabled = MetricAggregator.disabled
+        MetricAggregator.disabled = True
+        try:
+            train(runtime, world_model, actor, critic, world_optimizer, actor_optimizer,
+                  critic_optimizer, batch, None, cfg)
+        finally:
+            MetricAggregator.disabled = was_disabled
+
+    try:
+        step = CUDAGraphStep(train_fn, example_batch, warmup=2)
+        runtime.print("[dreamer_v1] gradient step captured in a hipGraph")
+        return step
+    except Exception as e:  # noqa: BLE001
+        runtime.print(f"[dreamer_v1] hipGraph capture failed ({e}); eager training")
+        return None
+
+
 @register_algorithm(name="dreamer_v1")
 def main(runtime: Runtime, cfg: Any) -> None:
     device = runtime.device
@@ -407,6 +433,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
     learning_starts = cfg.algo.learning_starts // policy_steps_per_iter if not cfg.dry_run else 0
     policy_step = int(state.get("policy_step", 0))
     last_log = 0
+    graphed_step = None
+    _n_train_calls = 0
+    want_graphs = (
+        runtime.device.type == "cuda"
+        and cfg.algo.get("hipgraphs", True)
+        and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
+    )
     last_checkpoint = 0
     ratio = Ratio(cfg.algo.replay_ratio, pretrain_steps=cfg.algo.per_rank_pretrain_steps)
 
@@ -479,8 +512,19 @@ def main(runtime: Runtime, cfg: Any) -> None:
                             device=device,
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
-                        train(runtime, world_model, actor, critic, world_optimizer, actor_optimizer,
-                              critic_optimizer, batch, aggregator, cfg)
+                        _n_train_calls += 1
+                        if graphed_step is not None and _n_train_calls % 16 != 0:
+                            graphed_step(batch)
+                        else:
+                            train(runtime, world_model, actor, critic, world_optimizer, actor_optimizer,
+                                  critic_optimizer, batch, aggregator, cfg)
+                        if graphed_step is None and want_graphs and _n_train_calls >= 3:
+                            graphed_step = _capture_train_step(
+                                runtime, world_model, actor, critic,
+                                world_optimizer, actor_optimizer, critic_optimizer, batch, cfg,
+                            )
+                            if graphed_step is None:
+                                want_graphs = False
 
         if policy_step - last_log >= cfg.metric.log_every or iter_num == total_iters or cfg.dry_run:
             runtime.log_dict(aggregator.compute(), policy_step)

@@ -245,6 +245,39 @@ def train(
         aggregator.update("Loss/value_loss", value_loss.detach())
 
 
+def _capture_train_step(
+    runtime, world_model, actor, critic, target_critic,
+    world_optimizer, actor_optimizer, critic_optimizer,
+    example_batch, cfg, is_continuous, actions_dim,
+):
+    """hipGraph-capture the DV2 gradient step (without the hard
+    target-critic copy, which stays outside the graph on its own cadence);
+    returns a replayable callable or None when capture fails."""
+    from sheeprl_amd.parallel.graphs import CUDAGraphStep
+    from sheeprl_amd.utils.metric import MetricAggregator
+
+    def train_fn(batch):
+        was_disabled = MetricAggregator.disabled
+        MetricAggregator.disabled = True
+        try:
+            # cumulative_step=-1 never hits the modulo-zero target update
+            train(
+                runtime, world_model, actor, critic, target_critic,
+                world_optimizer, actor_optimizer, critic_optimizer,
+                batch, None, cfg, is_continuous, actions_dim, -1,
+            )
+        finally:
+            MetricAggregator.disabled = was_disabled
+
+    try:
+        step = CUDAGraphStep(train_fn, example_batch, warmup=2)
+        runtime.print("[dreamer_v2] gradient step captured in a hipGraph")
+        return step
+    except Exception as e:  # noqa: BLE001
+        runtime.print(f"[dreamer_v2] hipGraph capture failed ({e}); eager training")
+        return None
+
+
 @register_algorithm(name="dreamer_v2")
 def main(runtime: Runtime, cfg: Any) -> None:
     from sheeprl_amd.algos.dreamer_v3.dreamer_v3 import main as dv3_main  # noqa: F401 (shape reference)
@@ -309,6 +342,12 @@ def main(runtime: Runtime, cfg: Any) -> None:
     last_checkpoint = 0
     ratio = Ratio(cfg.algo.replay_ratio, pretrain_steps=cfg.algo.per_rank_pretrain_steps)
     cumulative_steps = 0
+    graphed_step = None
+    want_graphs = (
+        runtime.device.type == "cuda"
+        and cfg.algo.get("hipgraphs", True)
+        and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
+    )
 
     import torch.nn.functional as F  # noqa: F811
 
@@ -406,11 +445,28 @@ def main(runtime: Runtime, cfg: Any) -> None:
                             device=device,
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
-                        train(
-                            runtime, world_model, actor, critic, target_critic,
-                            world_optimizer, actor_optimizer, critic_optimizer,
-                            batch, aggregator, cfg, is_continuous, actions_dim, cumulative_steps,
-                        )
+                        use_eager = graphed_step is None or (cumulative_steps % 16 == 0)
+                        if use_eager:
+                            train(
+                                runtime, world_model, actor, critic, target_critic,
+                                world_optimizer, actor_optimizer, critic_optimizer,
+                                batch, aggregator, cfg, is_continuous, actions_dim, cumulative_steps,
+                            )
+                        else:
+                            graphed_step(batch)
+                            # the hard target copy runs outside the graph
+                            if cumulative_steps % cfg.algo.critic.per_rank_target_network_update_freq == 0:
+                                with torch.no_grad():
+                                    for tp, p in zip(target_critic.parameters(), critic.parameters()):
+                                        tp.data.copy_(p.data)
+                        if graphed_step is None and want_graphs and cumulative_steps >= 3:
+                            graphed_step = _capture_train_step(
+                                runtime, world_model, actor, critic, target_critic,
+                                world_optimizer, actor_optimizer, critic_optimizer,
+                                batch, cfg, is_continuous, actions_dim,
+                            )
+                            if graphed_step is None:
+                                want_graphs = False
 
         if policy_step - last_log >= cfg.metric.log_every or iter_num == total_iters or cfg.dry_run:
             runtime.log_dict(aggregator.compute(), policy_step)
