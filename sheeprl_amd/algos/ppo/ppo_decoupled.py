@@ -164,6 +164,7 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
             torch.save(
                 {"agent": agent.state_dict(), "optimizer": opt_payload[0], "policy_step": policy_step},
                 ckpt_path,
+                pickle_protocol=4,
             )
         else:
             runtime.broadcast_object_list(["__noop__"], src=0, group=pt_group)

@@ -170,7 +170,7 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
             state = {"actor": agent.actor.state_dict(), "policy_step": policy_step}
             if trainer_state is not None:
                 state.update(trainer_state)  # full agent + the three optimizer states
-            torch.save(state, ckpt_path)
+            torch.save(state, ckpt_path, pickle_protocol=4)
             trainer_state = None
 
     runtime.scatter_object_list([None], [None] + [-1] * n_trainers, src=0, group=world_group)

@@ -304,7 +304,7 @@ class Runtime:
             else:
                 payload[k] = v
         os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
-        torch.save(payload, path)
+        torch.save(payload, path, pickle_protocol=4)
         self.barrier()
 
     def load(self, path: str, map_location: Any = "cpu") -> Dict[str, Any]:

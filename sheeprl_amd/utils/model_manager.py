@@ -59,7 +59,7 @@ class ModelManager:
         version = (self.get_latest_version(name) or 0) + 1
         vdir = self._model_dir(name) / f"v{version}"
         vdir.mkdir(parents=True, exist_ok=True)
-        torch.save(state_dict, vdir / "model.pt")
+        torch.save(state_dict, vdir / "model.pt", pickle_protocol=4)
         with open(vdir / "meta.yaml", "w") as f:
             yaml.safe_dump(
                 {
