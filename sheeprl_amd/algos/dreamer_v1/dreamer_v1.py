@@ -56,6 +56,12 @@ AGGREGATOR_KEYS = {
 MODELS_TO_REGISTER = {"world_model", "actor", "critic"}
 
 
+
+def _unit_scale(x):
+    """Scale=1 as a device tensor: td.Normal(x, 1) materializes the python
+    scalar with a pageable H2D copy, which is illegal inside hipGraph capture."""
+    return torch.ones((), device=x.device, dtype=x.dtype)
+
 def compute_stochastic_state(info: Tensor, min_std: float = 0.1):
     """(mean, std), sample — parity: dreamer_v1/utils.py:80-108."""
     mean, std = torch.chunk(info.float(), 2, -1)
@@ -295,8 +301,8 @@ def train(
     latent_states = torch.cat((stochastic_states, recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), 1), len(v.shape[2:])) for k, v in decoded.items()}
-    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), 1), 1)
+    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v.float())), len(v.shape[2:])) for k, v in decoded.items()}
+    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), _unit_scale(world_model.reward_model(latent_states).float())), 1)
 
     posteriors_dist = td.Independent(td.Normal(post_means, post_stds), 1)
     priors_dist = td.Independent(td.Normal(prior_means, prior_stds), 1)
@@ -341,7 +347,7 @@ def train(
         runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
     actor_optimizer.step()
 
-    qv = td.Independent(td.Normal(critic(imagined_latent_states.detach()).float(), 1), 1)
+    qv = td.Independent(td.Normal(critic(imagined_latent_states.detach()).float(), _unit_scale(critic(imagined_latent_states.detach()).float())), 1)
     critic_optimizer.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)

@@ -47,6 +47,12 @@ AGGREGATOR_KEYS = {
 MODELS_TO_REGISTER = {"world_model", "actor", "critic", "target_critic"}
 
 
+
+def _unit_scale(x):
+    """Scale=1 as a device tensor: td.Normal(x, 1) materializes the python
+    scalar with a pageable H2D copy, which is illegal inside hipGraph capture."""
+    return torch.ones((), device=x.device, dtype=x.dtype)
+
 def dv2_reconstruction_loss(
     po, observations, pr, rewards, priors_logits, posteriors_logits,
     kl_balancing_alpha=0.8, kl_free_nats=1.0, kl_free_avg=True, kl_regularizer=1.0,
@@ -135,8 +141,8 @@ def train(
     latent_states = torch.cat((posteriors.view(*posteriors.shape[:-2], -1), recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), 1), len(v.shape[2:])) for k, v in decoded.items()}
-    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), 1), 1)
+    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v.float())), len(v.shape[2:])) for k, v in decoded.items()}
+    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), _unit_scale(world_model.reward_model(latent_states).float())), 1)
     if cfg.algo.world_model.use_continues and world_model.continue_model:
         pc = td.Independent(td.Bernoulli(logits=world_model.continue_model(latent_states).float()), 1)
         continues_targets = (1 - data["terminated"]) * cfg.algo.gamma
@@ -221,7 +227,7 @@ def train(
     actor_optimizer.step()
 
     # critic: Normal log-prob regression on λ-values (dreamer_v2.py:340-356)
-    qv = td.Independent(td.Normal(critic(imagined_trajectories.detach()[:-1]).float(), 1), 1)
+    qv = td.Independent(td.Normal(critic(imagined_trajectories.detach()[:-1]).float(), _unit_scale(critic(imagined_trajectories.detach()[:-1]).float())), 1)
     critic_optimizer.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[:-1, ..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)

@@ -51,6 +51,12 @@ AGGREGATOR_KEYS = {
 MODELS_TO_REGISTER = {"world_model", "ensembles", "actor_task", "critic_task", "actor_exploration", "critic_exploration"}
 
 
+
+def _unit_scale(x):
+    """Scale=1 as a device tensor: td.Normal(x, 1) materializes the python
+    scalar with a pageable H2D copy, which is illegal inside hipGraph capture."""
+    return torch.ones((), device=x.device, dtype=x.dtype)
+
 def _behaviour_update(
     runtime, cfg, world_model, actor, critic, target_critic, actor_opt, critic_opt,
     posteriors, recurrent_states, data, reward_fn, is_continuous, actions_dim, aggregator, tag,
@@ -122,7 +128,7 @@ def _behaviour_update(
         runtime.clip_gradients(actor, actor_opt, cfg.algo.actor.clip_gradients)
     actor_opt.step()
 
-    qv = td.Independent(td.Normal(critic(imagined_trajectories.detach()[:-1]).float(), 1), 1)
+    qv = td.Independent(td.Normal(critic(imagined_trajectories.detach()[:-1]).float(), _unit_scale(critic(imagined_trajectories.detach()[:-1]).float())), 1)
     critic_opt.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[:-1, ..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)
@@ -175,8 +181,8 @@ def train(
     latent_states = torch.cat((posteriors.view(*posteriors.shape[:-2], -1), recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), 1), len(v.shape[2:])) for k, v in decoded.items()}
-    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), 1), 1)
+    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v.float())), len(v.shape[2:])) for k, v in decoded.items()}
+    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), _unit_scale(world_model.reward_model(latent_states).float())), 1)
     if cfg.algo.world_model.use_continues and world_model.continue_model:
         pc = td.Independent(td.Bernoulli(logits=world_model.continue_model(latent_states).float()), 1)
         continues_targets = (1 - data["terminated"]) * cfg.algo.gamma
