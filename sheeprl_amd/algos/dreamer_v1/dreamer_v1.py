@@ -301,8 +301,10 @@ def train(
     latent_states = torch.cat((stochastic_states, recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v.float())), len(v.shape[2:])) for k, v in decoded.items()}
-    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), _unit_scale(world_model.reward_model(latent_states).float())), 1)
+    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v)), len(v.shape[2:])) for k, v in decoded.items()}
+    _rm_out = world_model.reward_model(latent_states).float()
+
+    pr = td.Independent(td.Normal(_rm_out, _unit_scale(_rm_out)), 1)
 
     posteriors_dist = td.Independent(td.Normal(post_means, post_stds), 1)
     priors_dist = td.Independent(td.Normal(prior_means, prior_stds), 1)
@@ -310,7 +312,7 @@ def train(
     observation_loss = -sum(po[k].log_prob(batch_obs[k].float()).mean() for k in po)
     reward_loss = -pr.log_prob(data["rewards"]).mean()
     kl = td.kl_divergence(posteriors_dist, priors_dist).mean()
-    state_loss = torch.max(kl, torch.tensor(cfg.algo.world_model.kl_free_nats, device=device))
+    state_loss = torch.clamp(kl, min=cfg.algo.world_model.kl_free_nats)
     rec_loss = cfg.algo.world_model.kl_regularizer * state_loss + observation_loss + reward_loss
 
     world_optimizer.zero_grad(set_to_none=True)
@@ -347,7 +349,10 @@ def train(
         runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
     actor_optimizer.step()
 
-    qv = td.Independent(td.Normal(critic(imagined_latent_states.detach()).float(), _unit_scale(critic(imagined_latent_states.detach()).float())), 1)
+    _qv_out = critic(imagined_latent_states.detach()).float()
+
+
+    qv = td.Independent(td.Normal(_qv_out, _unit_scale(_qv_out)), 1)
     critic_optimizer.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)

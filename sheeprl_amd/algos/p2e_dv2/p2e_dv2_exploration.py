@@ -128,7 +128,10 @@ def _behaviour_update(
         runtime.clip_gradients(actor, actor_opt, cfg.algo.actor.clip_gradients)
     actor_opt.step()
 
-    qv = td.Independent(td.Normal(critic(imagined_trajectories.detach()[:-1]).float(), _unit_scale(critic(imagined_trajectories.detach()[:-1]).float())), 1)
+    _qv_out = critic(imagined_trajectories.detach()[:-1]).float()
+
+
+    qv = td.Independent(td.Normal(_qv_out, _unit_scale(_qv_out)), 1)
     critic_opt.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[:-1, ..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)
@@ -181,8 +184,10 @@ def train(
     latent_states = torch.cat((posteriors.view(*posteriors.shape[:-2], -1), recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v.float())), len(v.shape[2:])) for k, v in decoded.items()}
-    pr = td.Independent(td.Normal(world_model.reward_model(latent_states).float(), _unit_scale(world_model.reward_model(latent_states).float())), 1)
+    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v)), len(v.shape[2:])) for k, v in decoded.items()}
+    _rm_out = world_model.reward_model(latent_states).float()
+
+    pr = td.Independent(td.Normal(_rm_out, _unit_scale(_rm_out)), 1)
     if cfg.algo.world_model.use_continues and world_model.continue_model:
         pc = td.Independent(td.Bernoulli(logits=world_model.continue_model(latent_states).float()), 1)
         continues_targets = (1 - data["terminated"]) * cfg.algo.gamma
