@@ -125,6 +125,17 @@ def _make_atari(env_id: str = "MsPacmanNoFrameskip-v4", **kwargs: Any) -> Env:
     return _GymnasiumAdapter(gymnasium.make(env_id, render_mode="rgb_array"))
 
 
+def _make_gym(env_id: str, **kwargs: Any) -> Env:
+    """Generic gymnasium env (the reference's env=gym / env=mujoco configs)."""
+    try:
+        import gymnasium  # type: ignore
+    except ImportError as e:
+        raise ImportError("env backend 'gym' needs gymnasium (not in this image)") from e
+    return _GymnasiumAdapter(gymnasium.make(env_id, render_mode="rgb_array"))
+
+
+register_env("gym", _make_gym)
+register_env("mujoco", _make_gym)
 register_env("dmc", _make_dmc)
 register_env("crafter", _make_crafter)
 register_env("atari", _make_atari)

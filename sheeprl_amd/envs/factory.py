@@ -47,10 +47,27 @@ def make_env(
     def thunk() -> Env:
         env_cfg = cfg.env
         env_id = env_cfg.id
-        if env_id not in _ENV_BUILDERS:
-            raise ValueError(f"unknown env id '{env_id}'; known: {sorted(_ENV_BUILDERS)}")
         kwargs = dict(env_cfg.get("wrapper_kwargs", {}) or {})
-        env: Env = _ENV_BUILDERS[env_id](**kwargs)
+        if env_id in _ENV_BUILDERS:
+            env: Env = _ENV_BUILDERS[env_id](**kwargs)
+        else:
+            # route by explicit backend (env/atari.yaml etc.) or id prefix to
+            # the external-package adapters (sheeprl_amd/envs/external.py)
+            backend = env_cfg.get("backend", None)
+            if backend is None:
+                pfx = {
+                    "dmc_": "dmc", "crafter_": "crafter", "minedojo_": "minedojo",
+                    "minerl_": "minerl", "diambra_": "diambra", "supermario_": "super_mario_bros",
+                }
+                for k, v in pfx.items():
+                    if str(env_id).startswith(k):
+                        backend = v
+                        break
+                if backend is None and ("NoFrameskip" in str(env_id) or str(env_id).startswith("ALE/")):
+                    backend = "atari"
+            if backend is None or backend not in _ENV_BUILDERS:
+                raise ValueError(f"unknown env id '{env_id}'; known: {sorted(_ENV_BUILDERS)}")
+            env = _ENV_BUILDERS[backend](env_id, **kwargs)
 
         if env_cfg.get("mask_velocities", False):
             env = wrappers.MaskVelocityWrapper(env, env_id)
