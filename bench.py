@@ -236,7 +236,10 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         step_data["is_first"] = np.zeros_like(step_data["terminated"])
         dones = np.logical_or(term, trunc)
         if dones.any():
-            player.init_states(np.nonzero(dones)[0].tolist())
+            # player states were produced under inference_mode; resetting them
+            # in place must happen under it too
+            with torch.inference_mode():
+                player.init_states(np.nonzero(dones)[0].tolist())
         # --- one gradient step (replay_ratio=1 at num_envs=1)
         from sheeprl_amd import ops as _ops
 
