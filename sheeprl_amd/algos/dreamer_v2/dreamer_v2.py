@@ -436,7 +436,10 @@ def main(runtime: Runtime, cfg: Any) -> None:
                 step_data["terminated"][:, dones_idxes] = 0.0
                 step_data["truncated"][:, dones_idxes] = 0.0
                 step_data["is_first"][:, dones_idxes] = 1.0
-                player.init_states(dones_idxes)
+                # player states are inference tensors (created under the
+                # action-selection inference_mode); reset them under it too
+                with torch.inference_mode():
+                    player.init_states(dones_idxes)
 
         if isinstance(rb, EnvIndependentReplayBuffer):
             rb_ready = any(len(b) >= cfg.algo.per_rank_sequence_length for b in rb.buffer)

@@ -558,7 +558,10 @@ def main(runtime: Runtime, cfg: Any) -> None:
                 step_data["terminated"][:, dones_idxes] = 0.0
                 step_data["truncated"][:, dones_idxes] = 0.0
                 step_data["is_first"][:, dones_idxes] = 1.0
-                player.init_states(dones_idxes)
+                # player states are inference tensors (created under the
+                # action-selection inference_mode); reset them under it too
+                with torch.inference_mode():
+                    player.init_states(dones_idxes)
 
         # training phase, replay-ratio driven.  On CUDA the whole gradient
         # step is hipGraph-captured after warmup (algo.hip_graphs, default

@@ -345,7 +345,10 @@ def main(runtime: Runtime, cfg: Any) -> None:
             step_data["truncated"] = np.asarray(truncated, np.float32).reshape(1, num_envs, 1)
             dones_idxes = np.nonzero(dones)[0].tolist()
             if dones_idxes:
-                player.init_states(dones_idxes)
+                # player states are inference tensors (created under the
+                # action-selection inference_mode); reset them under it too
+                with torch.inference_mode():
+                    player.init_states(dones_idxes)
 
         rb_ready = any(len(b) >= cfg.algo.per_rank_sequence_length for b in rb.buffer)
         if iter_num >= learning_starts and rb_ready:

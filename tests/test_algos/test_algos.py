@@ -151,6 +151,42 @@ def test_dreamer_v3(tmp_path, devices):
 
 
 @pytest.mark.timeout(300)
+def test_dreamer_v3_episode_boundary_resets(tmp_path):
+    # partial player resets (inference-tensor in-place update) only trigger
+    # when an episode ends mid-run: force short episodes with several envs
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=dreamer_v3",
+                "algo=dreamer_v3_S",
+                "algo.dense_units=8",
+                "algo.mlp_layers=1",
+                "algo.world_model.encoder.cnn_channels_multiplier=2",
+                "algo.world_model.recurrent_model.recurrent_state_size=8",
+                "algo.world_model.transition_model.hidden_size=8",
+                "algo.world_model.representation_model.hidden_size=8",
+                "algo.world_model.discrete_size=4",
+                "algo.world_model.stochastic_size=4",
+                "algo.per_rank_batch_size=2",
+                "algo.per_rank_sequence_length=4",
+                "algo.horizon=3",
+                "algo.mlp_keys.encoder=[state]",
+                "algo.total_steps=48",
+                "algo.learning_starts=8",
+                "algo.replay_ratio=0.25",
+                "env.num_envs=2",
+                "env.max_episode_steps=5",
+                "buffer.size=128",
+                "algo.run_test=False",
+                "dry_run=False",
+            ],
+        ),
+    )
+
+
+@pytest.mark.timeout(300)
 def test_dreamer_v3_checkpoint_resume(tmp_path):
     args = standard_args(
         tmp_path,
