@@ -187,6 +187,30 @@ def train(runtime, world_model, ensembles, actor_task, critic_task, actor_explor
         aggregator.update("Loss/ensemble_loss", ens_loss.detach())
 
 
+def _capture_train_step(args, example_batch):
+    """hipGraph-capture the P2E-DV1 gradient step; None on failure."""
+    from sheeprl_amd.parallel.graphs import CUDAGraphStep
+    from sheeprl_amd.utils.metric import MetricAggregator
+
+    runtime, cfg = args[0], args[-1]
+
+    def train_fn(batch):
+        was_disabled = MetricAggregator.disabled
+        MetricAggregator.disabled = True
+        try:
+            train(*args[:-1], batch, None, cfg)
+        finally:
+            MetricAggregator.disabled = was_disabled
+
+    try:
+        step = CUDAGraphStep(train_fn, example_batch, warmup=2)
+        runtime.print("[p2e_dv1] gradient step captured in a hipGraph")
+        return step
+    except Exception as e:  # noqa: BLE001
+        runtime.print(f"[p2e_dv1] hipGraph capture failed ({e}); eager training")
+        return None
+
+
 @register_algorithm(name="p2e_dv1_exploration")
 def main(runtime: Runtime, cfg: Any) -> None:
     device = runtime.device
@@ -290,6 +314,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
     learning_starts = cfg.algo.learning_starts // policy_steps_per_iter if not cfg.dry_run else 0
     policy_step = int(state.get("policy_step", 0))
     last_log = 0
+    graphed_step = None
+    _n_train_calls = 0
+    want_graphs = (
+        runtime.device.type == "cuda"
+        and cfg.algo.get("hipgraphs", True)
+        and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
+    )
     last_checkpoint = 0
     ratio = Ratio(cfg.algo.replay_ratio, pretrain_steps=cfg.algo.per_rank_pretrain_steps)
 
@@ -363,12 +394,25 @@ def main(runtime: Runtime, cfg: Any) -> None:
                             device=device,
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
-                        train(
-                            runtime, world_model, ensembles, actor_task, critic_task,
-                            actor_exploration, critic_exploration, world_optimizer, ensemble_optimizer,
-                            actor_task_opt, critic_task_opt, actor_expl_opt, critic_expl_opt,
-                            batch, aggregator, cfg,
-                        )
+                        _n_train_calls += 1
+                        if graphed_step is not None and _n_train_calls % 16 != 0:
+                            graphed_step(batch)
+                        else:
+                            train(
+                                runtime, world_model, ensembles, actor_task, critic_task,
+                                actor_exploration, critic_exploration, world_optimizer, ensemble_optimizer,
+                                actor_task_opt, critic_task_opt, actor_expl_opt, critic_expl_opt,
+                                batch, aggregator, cfg,
+                            )
+                        if graphed_step is None and want_graphs and _n_train_calls >= 3:
+                            graphed_step = _capture_train_step(
+                                (runtime, world_model, ensembles, actor_task, critic_task,
+                                 actor_exploration, critic_exploration, world_optimizer, ensemble_optimizer,
+                                 actor_task_opt, critic_task_opt, actor_expl_opt, critic_expl_opt, cfg),
+                                batch,
+                            )
+                            if graphed_step is None:
+                                want_graphs = False
 
         if policy_step - last_log >= cfg.metric.log_every or iter_num == total_iters or cfg.dry_run:
             runtime.log_dict(aggregator.compute(), policy_step)

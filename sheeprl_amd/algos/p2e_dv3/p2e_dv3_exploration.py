@@ -332,6 +332,32 @@ def train(
         aggregator.update("Loss/value_loss_task", value_loss_task.detach())
 
 
+def _capture_train_step(train_args):
+    """hipGraph-capture the P2E-DV3 gradient step (unconditional in-place
+    EMA/moments updates make the whole step replayable); None on failure."""
+    from sheeprl_amd.parallel.graphs import CUDAGraphStep
+    from sheeprl_amd.utils.metric import MetricAggregator
+
+    runtime = train_args[0]
+    pre, post = train_args[:15], train_args[16:]
+
+    def train_fn(batch):
+        was_disabled = MetricAggregator.disabled
+        MetricAggregator.disabled = True
+        try:
+            train(*pre, batch, None, *post[1:])
+        finally:
+            MetricAggregator.disabled = was_disabled
+
+    try:
+        step = CUDAGraphStep(train_fn, train_args[15], warmup=2)
+        runtime.print("[p2e_dv3] gradient step captured in a hipGraph")
+        return step
+    except Exception as e:  # noqa: BLE001
+        runtime.print(f"[p2e_dv3] hipGraph capture failed ({e}); eager training")
+        return None
+
+
 @register_algorithm(name="p2e_dv3_exploration")
 def main(runtime: Runtime, cfg: Any) -> None:
     device = runtime.device
@@ -400,6 +426,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
     learning_starts = cfg.algo.learning_starts // policy_steps_per_iter if not cfg.dry_run else 0
     policy_step = int(state.get("policy_step", 0))
     last_log = 0
+    graphed_step = None
+    _n_train_calls = 0
+    want_graphs = (
+        runtime.device.type == "cuda"
+        and cfg.algo.get("hipgraphs", True)
+        and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
+    )
     last_checkpoint = 0
     ratio = Ratio(cfg.algo.replay_ratio, pretrain_steps=cfg.algo.per_rank_pretrain_steps)
 
@@ -493,13 +526,27 @@ def main(runtime: Runtime, cfg: Any) -> None:
                             device=device,
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
-                        train(
-                            runtime, world_model, ensembles, actor_task, critic_task, target_critic_task,
-                            actor_exploration, critics_exploration, world_optimizer, ensemble_optimizer,
-                            actor_task_optimizer, critic_task_optimizer, actor_expl_optimizer,
-                            critics_expl_optimizers, moments_task, batch, aggregator, cfg,
-                            is_continuous, actions_dim,
-                        )
+                        _n_train_calls += 1
+                        if graphed_step is not None and _n_train_calls % 16 != 0:
+                            graphed_step(batch)
+                        else:
+                            train(
+                                runtime, world_model, ensembles, actor_task, critic_task, target_critic_task,
+                                actor_exploration, critics_exploration, world_optimizer, ensemble_optimizer,
+                                actor_task_optimizer, critic_task_optimizer, actor_expl_optimizer,
+                                critics_expl_optimizers, moments_task, batch, aggregator, cfg,
+                                is_continuous, actions_dim,
+                            )
+                        if graphed_step is None and want_graphs and _n_train_calls >= 3:
+                            graphed_step = _capture_train_step((
+                                runtime, world_model, ensembles, actor_task, critic_task, target_critic_task,
+                                actor_exploration, critics_exploration, world_optimizer, ensemble_optimizer,
+                                actor_task_optimizer, critic_task_optimizer, actor_expl_optimizer,
+                                critics_expl_optimizers, moments_task, batch, aggregator, cfg,
+                                is_continuous, actions_dim,
+                            ))
+                            if graphed_step is None:
+                                want_graphs = False
 
         if policy_step - last_log >= cfg.metric.log_every or iter_num == total_iters or cfg.dry_run:
             runtime.log_dict(aggregator.compute(), policy_step)
