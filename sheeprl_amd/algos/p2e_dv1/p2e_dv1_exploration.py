@@ -155,7 +155,7 @@ def train(runtime, world_model, ensembles, actor_task, critic_task, actor_explor
     # ensembles on the continuous stochastic state
     ensemble_optimizer.zero_grad(set_to_none=True)
     ens_loss = 0.0
-    ens_input = torch.cat((stochastic_states.detach(), recurrent_states.detach(), data["actions"].detach()), -1)
+    ens_input = torch.cat((stochastic_states.detach(), recurrent_states.detach(), data["actions"].detach().to(dtype)), -1)
     target_next = stochastic_states.detach()[1:].float()
     for ens in ensembles:
         out = ens(ens_input)[:-1].float()

@@ -213,7 +213,7 @@ def train(
     ensemble_optimizer.zero_grad(set_to_none=True)
     ens_loss = 0.0
     ens_input = torch.cat(
-        (posteriors.view(*posteriors.shape[:-2], -1).detach(), recurrent_states.detach(), data["actions"].detach()),
+        (posteriors.view(*posteriors.shape[:-2], -1).detach(), recurrent_states.detach(), data["actions"].detach().to(dtype)),
         -1,
     )
     target_next = posteriors.view(sequence_length, batch_size, -1).detach()[1:].float()

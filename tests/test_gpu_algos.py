@@ -143,3 +143,13 @@ def test_gpu_dreamer_v3_bf16(tmp_path):
     """bf16-true DV3 on GPU through the real CLI (incl. graph capture path)."""
     _run(tmp_path, ["exp=dreamer_v3", "algo.mlp_keys.encoder=[state]", *TINY_DV],
          precision="bf16")
+
+
+@requires_gpu
+@pytest.mark.timeout(300)
+def test_gpu_p2e_dv3_bf16(tmp_path):
+    """bf16-true Plan2Explore-DV3 on GPU (exercises the ensemble dtype path)."""
+    _run(tmp_path, ["exp=p2e_dv3_exploration", "algo.mlp_keys.encoder=[state]",
+                    "algo.ensembles.n=2", "algo.ensembles.dense_units=8",
+                    "algo.ensembles.mlp_layers=1", *TINY_DV],
+         precision="bf16")
