@@ -428,9 +428,12 @@ def main(runtime: Runtime, cfg: Any) -> None:
     last_log = 0
     graphed_step = None
     _n_train_calls = 0
+    # p2e capture is OPT-IN for now (algo.hipgraphs=true): the captured
+    # multi-critic step segfaulted in round-1 validation; the single-family
+    # Dreamer mains capture by default
     want_graphs = (
         runtime.device.type == "cuda"
-        and cfg.algo.get("hipgraphs", True)
+        and cfg.algo.get("hipgraphs", False)
         and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
     )
     last_checkpoint = 0
