@@ -447,8 +447,8 @@ def main(runtime: Runtime, cfg: Any) -> None:
     graphed_step = None
     _n_train_calls = 0
     want_graphs = (
-        runtime.device.type == "cuda"
-        and cfg.algo.get("hipgraphs", True)
+        (runtime.device.type == "cuda" or os.environ.get("SHEEPRL_AMD_FORCE_GRAPHS") == "1")
+        and cfg.algo.get("hip_graphs", True)
         and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
     )
     last_checkpoint = 0

@@ -354,8 +354,8 @@ def main(runtime: Runtime, cfg: Any) -> None:
     cumulative_steps = 0
     graphed_step = None
     want_graphs = (
-        runtime.device.type == "cuda"
-        and cfg.algo.get("hipgraphs", True)
+        (runtime.device.type == "cuda" or os.environ.get("SHEEPRL_AMD_FORCE_GRAPHS") == "1")
+        and cfg.algo.get("hip_graphs", True)
         and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
     )
 

@@ -459,7 +459,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
 
     want_graphs = (
         bool(cfg.algo.get("hip_graphs", True))
-        and device.type == "cuda"
+        and (device.type == "cuda" or os.environ.get("SHEEPRL_AMD_FORCE_GRAPHS") == "1")
         and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
     )
     graphed_step = None

@@ -320,8 +320,8 @@ def main(runtime: Runtime, cfg: Any) -> None:
     # multi-critic step segfaulted in round-1 validation; the single-family
     # Dreamer mains capture by default
     want_graphs = (
-        runtime.device.type == "cuda"
-        and cfg.algo.get("hipgraphs", False)
+        (runtime.device.type == "cuda" or os.environ.get("SHEEPRL_AMD_FORCE_GRAPHS") == "1")
+        and cfg.algo.get("hip_graphs", False)
         and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
     )
     last_checkpoint = 0

@@ -694,3 +694,43 @@ def test_minedojo_actor_masking():
         crafted = acts[0].argmax(-1) == 15
         if crafted.any():
             assert (acts[1].argmax(-1)[crafted] == 1).all()
+
+
+def test_graph_capture_plumbing_cpu(tmp_path, monkeypatch):
+    """Exercise every main's hipGraph-capture wiring on CPU with an eager
+    stand-in for CUDAGraphStep — catches argument-plumbing regressions in the
+    GPU-only code path."""
+    from sheeprl_amd.parallel import graphs as graphs_mod
+
+    class EagerStep:
+        def __init__(self, fn, example_inputs, warmup=3, pool=None):
+            self.fn = fn
+            for _ in range(1):
+                fn(example_inputs)
+
+        def __call__(self, inputs):
+            self.fn(inputs)
+
+    monkeypatch.setattr(graphs_mod, "CUDAGraphStep", EagerStep)
+    monkeypatch.setenv("SHEEPRL_AMD_FORCE_GRAPHS", "1")
+    tiny = [
+        "algo.dense_units=8", "algo.mlp_layers=1",
+        "algo.world_model.encoder.cnn_channels_multiplier=2",
+        "algo.world_model.recurrent_model.recurrent_state_size=8",
+        "algo.world_model.transition_model.hidden_size=8",
+        "algo.world_model.representation_model.hidden_size=8",
+        "algo.world_model.discrete_size=4", "algo.world_model.stochastic_size=4",
+        "algo.per_rank_batch_size=2", "algo.per_rank_sequence_length=4",
+        "algo.horizon=3", "algo.total_steps=20", "algo.learning_starts=4",
+        "algo.replay_ratio=0.5", "buffer.size=64", "algo.run_test=False",
+        "algo.mlp_keys.encoder=[state]",
+    ]
+    for exp, extra in [
+        ("dreamer_v3", []),
+        ("dreamer_v2", ["algo.per_rank_pretrain_steps=1"]),
+        ("p2e_dv3_exploration", ["algo.hip_graphs=true", "algo.ensembles.n=2",
+                                 "algo.ensembles.dense_units=8", "algo.ensembles.mlp_layers=1"]),
+    ]:
+        _run(tmp_path, [f"exp={exp}", "env=dummy", "runtime.accelerator=cpu", "dry_run=False",
+                        "checkpoint.every=0", "metric.log_every=100", "env.num_envs=2", "seed=0",
+                        "algo.cnn_keys.encoder=[]", *tiny, *extra])
