@@ -155,6 +155,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
         with timer("Time/train_time"):
             train(runtime, agent, optimizer, data, aggregator, cfg)
 
+        if cfg.algo.get("anneal_lr", False):
+            frac = 1.0 - (it - 1) / num_iters
+            for pg in optimizer.param_groups:
+                pg["lr"] = cfg.algo.optimizer.lr * frac
+
         if policy_step - last_log >= cfg.metric.log_every or it == num_iters or cfg.dry_run:
             metrics = aggregator.compute()
             runtime.log_dict(metrics, policy_step)

@@ -75,6 +75,8 @@ def main(runtime: Runtime, cfg: Any) -> None:
     states = agent.initial_states(num_envs, device)
     is_first_np = np.ones((num_envs,), dtype=np.float32)
 
+    initial_ent_coef = float(cfg.algo.ent_coef)
+    initial_clip_coef = float(cfg.algo.clip_coef)
     for it in range(1, num_iters + 1):
         rollout: Dict[str, list] = {k: [] for k in
                                     ["actions", "prev_actions", "logprobs", "values", "rewards", "dones", "is_first"]}
@@ -171,6 +173,15 @@ def main(runtime: Runtime, cfg: Any) -> None:
                     aggregator.update("Loss/value_loss", vl.detach())
                     aggregator.update("Loss/entropy_loss", el.detach())
 
+
+        if cfg.algo.get("anneal_lr", False):
+            frac = 1.0 - (it - 1) / num_iters
+            for pg in optimizer.param_groups:
+                pg["lr"] = cfg.algo.optimizer.lr * frac
+        if cfg.algo.get("anneal_ent_coef", False):
+            cfg.algo.ent_coef = polynomial_decay(it, initial=initial_ent_coef, final=0.0, max_decay_steps=num_iters)
+        if cfg.algo.get("anneal_clip_coef", False):
+            cfg.algo.clip_coef = polynomial_decay(it, initial=initial_clip_coef, final=0.0, max_decay_steps=num_iters)
         if policy_step - last_log >= cfg.metric.log_every or it == num_iters or cfg.dry_run:
             runtime.log_dict(aggregator.compute(), policy_step)
             aggregator.reset()
