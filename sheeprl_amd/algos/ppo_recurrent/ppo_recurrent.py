@@ -22,6 +22,7 @@ from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
 from sheeprl_amd.utils.registry import register_algorithm, register_evaluation
 from sheeprl_amd.utils.timer import timer
+from sheeprl_amd.utils.utils import polynomial_decay
 MODELS_TO_REGISTER = {"agent"}
 
 
