@@ -120,8 +120,28 @@ class WorldModelV1(nn.Module):
         self.reward_model = reward_model
 
 
+def add_exploration_noise(actions, amount, is_continuous):
+    """Exploration noise on sampled actions (reference dreamer_v1/agent.py
+    :558-574): Gaussian-perturb-and-clip for continuous actions,
+    epsilon-random one-hot swap for discrete heads."""
+    if amount <= 0:
+        return actions
+    if is_continuous:
+        a = torch.cat(list(actions), -1)
+        a = torch.clip(torch.normal(a, amount), -1, 1)
+        return (a,)
+    out = []
+    for act in actions:
+        rand_idx = torch.randint(0, act.shape[-1], act.shape[:-1], device=act.device)
+        rand_onehot = torch.nn.functional.one_hot(rand_idx, act.shape[-1]).to(act.dtype)
+        swap = (torch.rand(act.shape[:1], device=act.device) < amount).view(-1, *([1] * (act.dim() - 1)))
+        out.append(torch.where(swap, rand_onehot, act))
+    return tuple(out)
+
+
 class PlayerDV1(nn.Module):
-    def __init__(self, encoder, rssm, actor, actions_dim, num_envs, stochastic_size, recurrent_state_size, device):
+    def __init__(self, encoder, rssm, actor, actions_dim, num_envs, stochastic_size, recurrent_state_size, device,
+                 expl_amount: float = 0.0, expl_min: float = 0.0, expl_decay: float = 0.0):
         super().__init__()
         self.encoder = encoder
         self.rssm = rssm
@@ -131,6 +151,24 @@ class PlayerDV1(nn.Module):
         self.stochastic_size = stochastic_size
         self.recurrent_state_size = recurrent_state_size
         self.device = device
+        self.expl_amount = expl_amount
+        self.expl_min = expl_min
+        self.expl_decay = expl_decay
+
+    def _expl(self, step: int) -> float:
+        amount = self.expl_amount
+        if self.expl_decay:
+            amount *= 0.5 ** (float(step) / self.expl_decay)
+        return max(amount, self.expl_min)
+
+    @torch.no_grad()
+    def get_exploration_actions(self, obs, step: int = 0, mask=None):
+        actions = self.get_actions(obs, greedy=False, mask=mask)
+        amount = self._expl(step)
+        if amount > 0:
+            actions = add_exploration_noise(actions, amount, self.actor.is_continuous)
+            self.actions = torch.cat(list(actions), -1).to(self.actions.dtype)
+        return actions
 
     @torch.no_grad()
     def init_states(self, reset_envs: Optional[Sequence[int]] = None) -> None:
@@ -251,7 +289,10 @@ def build_agent(
     actor = runtime.setup_module(actor)
     critic = runtime.setup_module(critic)
     player = PlayerDV1(world_model.encoder, world_model.rssm, actor, actions_dim, cfg.env.num_envs,
-                       stochastic_size, recurrent_state_size, runtime.device)
+                       stochastic_size, recurrent_state_size, runtime.device,
+                       expl_amount=cfg.algo.actor.get("expl_amount", 0.0),
+                       expl_min=cfg.algo.actor.get("expl_min", 0.0),
+                       expl_decay=cfg.algo.actor.get("expl_decay", 0.0))
     return world_model, actor, critic, player
 
 
@@ -481,7 +522,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
                         ).reshape(num_envs, -1)
                 else:
                     torch_obs = prepare_obs(runtime, obs, cnn_keys=cnn_keys, num_envs=num_envs)
-                    acts = player.get_actions(torch_obs)
+                    acts = player.get_exploration_actions(torch_obs, step=policy_step)
                     actions = torch.cat(acts, -1).view(num_envs, -1).float().cpu().numpy()
                     if is_continuous:
                         real_actions = actions

@@ -319,6 +319,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
         runtime, actions_dim, is_continuous, cfg, obs_space,
         state.get("world_model"), state.get("actor"), state.get("critic"), state.get("target_critic"),
     )
+    player.set_exploration(
+        cfg.algo.actor.get("expl_amount", 0.0),
+        cfg.algo.actor.get("expl_min", 0.0),
+        cfg.algo.actor.get("expl_decay", 0.0),
+    )
     world_optimizer = FusedAdam(world_model.parameters(), lr=cfg.algo.world_model.optimizer.lr,
                                 eps=cfg.algo.world_model.optimizer.eps,
                                 weight_decay=cfg.algo.world_model.optimizer.get("weight_decay", 0.0))
@@ -389,7 +394,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
                         ).reshape(num_envs, -1)
                 else:
                     torch_obs = prepare_obs(runtime, obs, cnn_keys=cnn_keys, num_envs=num_envs)
-                    acts = player.get_actions(torch_obs)
+                    acts = player.get_exploration_actions(torch_obs, step=policy_step)
                     actions = torch.cat(acts, -1).view(num_envs, -1).float().cpu().numpy()
                     if is_continuous:
                         real_actions = actions

@@ -584,6 +584,30 @@ class PlayerDV3(nn.Module):
         self.decoupled_rssm = isinstance(rssm, DecoupledRSSM)
 
     @torch.no_grad()
+    def set_exploration(self, amount: float = 0.0, minimum: float = 0.0, decay: float = 0.0) -> None:
+        """Configure DV1/DV2-style exploration noise for
+        :meth:`get_exploration_actions` (reference dreamer_v2/agent.py:663)."""
+        self._expl_amount = amount
+        self._expl_min = minimum
+        self._expl_decay = decay
+
+    def _expl(self, step: int) -> float:
+        amount = getattr(self, "_expl_amount", 0.0)
+        if getattr(self, "_expl_decay", 0.0):
+            amount *= 0.5 ** (float(step) / self._expl_decay)
+        return max(amount, getattr(self, "_expl_min", 0.0))
+
+    @torch.no_grad()
+    def get_exploration_actions(self, obs, step: int = 0, mask=None):
+        from sheeprl_amd.algos.dreamer_v1.dreamer_v1 import add_exploration_noise
+
+        actions = self.get_actions(obs, greedy=False, mask=mask)
+        amount = self._expl(step)
+        if amount > 0:
+            actions = add_exploration_noise(actions, amount, self.actor.is_continuous)
+            self.actions = torch.cat(list(actions), -1).to(self.actions.dtype)
+        return actions
+
     def init_states(self, reset_envs: Optional[Sequence[int]] = None) -> None:
         dtype = next(self.rssm.parameters()).dtype
         if reset_envs is None or len(reset_envs) == 0:
