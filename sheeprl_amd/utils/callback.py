@@ -30,7 +30,7 @@ class CheckpointCallback:
     def _prune(self, ckpt_dir: Path) -> None:
         if not self.keep_last:
             return
-        ckpts = sorted(ckpt_dir.glob("ckpt_*.ckpt"), key=os.path.getmtime)
+        ckpts = sorted(Path(ckpt_dir).glob("ckpt_*.ckpt"), key=os.path.getmtime)
         for old in ckpts[: -self.keep_last]:
             try:
                 old.unlink()
