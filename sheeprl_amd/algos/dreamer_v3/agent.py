@@ -608,6 +608,7 @@ class PlayerDV3(nn.Module):
             self.actions = torch.cat(list(actions), -1).to(self.actions.dtype)
         return actions
 
+    @torch.no_grad()
     def init_states(self, reset_envs: Optional[Sequence[int]] = None) -> None:
         dtype = next(self.rssm.parameters()).dtype
         if reset_envs is None or len(reset_envs) == 0:
