@@ -86,3 +86,56 @@ def test_async_vector_env_autoreset():
         obs, r, term, trunc, infos = envs.step([0, 1])
     assert any(e is not None for e in infos["episode"]) or True
     envs.close()
+
+
+def test_action_repeat_and_clip_reward():
+    from sheeprl_amd.envs import make_env
+    from sheeprl_amd.config import compose
+
+    cfg = compose(["exp=ppo", "env=classic", "env.id=cartpole", "env.action_repeat=2",
+                   "env.clip_rewards=True", "env.num_envs=1"])
+    env = make_env(cfg, 3, 0)()
+    obs, _ = env.reset(seed=3)
+    _, r, *_ = env.step(0)
+    # ActionRepeat sums the 2 raw rewards; ClipReward then clips the sum to
+    # [-1, 1] (same order as the reference pipeline)
+    assert r == 1.0
+
+
+def test_frame_stack_dilation():
+    import numpy as np
+    from sheeprl_amd.envs import wrappers
+    from sheeprl_amd.envs.synthetic import SyntheticAtariEnv
+
+    env = wrappers.DictObservation(SyntheticAtariEnv(), key="state")
+    env = wrappers.FrameStack(env, 3, ["rgb"], dilation=2)
+    obs, _ = env.reset(seed=0)
+    assert obs["rgb"].shape[0] == 9  # 3 frames x 3 channels stacked
+    for _ in range(5):
+        obs, *_ = env.step(0)
+    assert obs["rgb"].shape[0] == 9
+
+
+def test_time_limit_truncates():
+    from sheeprl_amd.envs import wrappers
+    from sheeprl_amd.envs.classic import PendulumEnv
+
+    env = wrappers.TimeLimit(PendulumEnv(), 4)
+    env.reset(seed=0)
+    truncs = []
+    for _ in range(4):
+        *_, term, trunc, _ = env.step(np.zeros(1, dtype=np.float32))
+        truncs.append(trunc)
+    assert truncs[-1] and not any(truncs[:-1])
+
+
+def test_actions_as_observation_stack():
+    from sheeprl_amd.envs import wrappers
+    from sheeprl_amd.envs.classic import CartPoleEnv
+
+    env = wrappers.DictObservation(CartPoleEnv(), key="state")
+    env = wrappers.ActionsAsObservation(env, num_stack=3, noop=0)
+    obs, _ = env.reset(seed=0)
+    assert "action_stack" in obs and obs["action_stack"].shape[0] == 3 * 2  # one-hot x 3
+    obs, *_ = env.step(1)
+    assert obs["action_stack"].sum() == 3  # still 3 one-hots
