@@ -112,12 +112,13 @@ class RSSMV1(nn.Module):
 
 
 class WorldModelV1(nn.Module):
-    def __init__(self, encoder, rssm, observation_model, reward_model):
+    def __init__(self, encoder, rssm, observation_model, reward_model, continue_model=None):
         super().__init__()
         self.encoder = encoder
         self.rssm = rssm
         self.observation_model = observation_model
         self.reward_model = reward_model
+        self.continue_model = continue_model
 
 
 def add_exploration_noise(actions, amount, is_continuous):
@@ -258,7 +259,13 @@ def build_agent(
     observation_model = MultiDecoder(cnn_decoder, mlp_decoder)
     reward_model = MLP(latent_state_size, 1, [wm_cfg.reward_model.dense_units] * wm_cfg.reward_model.mlp_layers,
                        activation=act, layer_norm=False)
-    world_model = WorldModelV1(encoder, rssm, observation_model, reward_model)
+    continue_model = None
+    if cfg.algo.world_model.get("use_continues", False):
+        dm = cfg.algo.world_model.get("discount_model", {})
+        continue_model = MLP(latent_state_size, 1,
+                             [dm.get("dense_units", 400)] * dm.get("mlp_layers", 4),
+                             activation=act, layer_norm=False)
+    world_model = WorldModelV1(encoder, rssm, observation_model, reward_model, continue_model)
 
     actor = Actor(
         latent_state_size=latent_state_size,
@@ -354,7 +361,17 @@ def train(
     reward_loss = -pr.log_prob(data["rewards"]).mean()
     kl = td.kl_divergence(posteriors_dist, priors_dist).mean()
     state_loss = torch.clamp(kl, min=cfg.algo.world_model.kl_free_nats)
-    rec_loss = cfg.algo.world_model.kl_regularizer * state_loss + observation_loss + reward_loss
+    continue_loss = torch.zeros((), device=device)
+    if cfg.algo.world_model.get("use_continues", False) and world_model.continue_model is not None:
+        qc_out = world_model.continue_model(latent_states).float()
+        qc = td.Independent(td.Bernoulli(logits=qc_out), 1)
+        continues_targets = (1 - data["terminated"]) * cfg.algo.gamma
+        continue_loss = cfg.algo.world_model.get("continue_scale_factor", 10.0) * -qc.log_prob(
+            continues_targets
+        ).mean()
+    rec_loss = (
+        cfg.algo.world_model.kl_regularizer * state_loss + observation_loss + reward_loss + continue_loss
+    )
 
     world_optimizer.zero_grad(set_to_none=True)
     runtime.backward(rec_loss)
@@ -376,7 +393,10 @@ def train(
 
     predicted_values = critic(imagined_latent_states).float()
     predicted_rewards = world_model.reward_model(imagined_latent_states).float()
-    continues = torch.ones_like(predicted_rewards) * cfg.algo.gamma
+    if cfg.algo.world_model.get("use_continues", False) and world_model.continue_model is not None:
+        continues = torch.sigmoid(world_model.continue_model(imagined_latent_states).float()) * cfg.algo.gamma
+    else:
+        continues = torch.ones_like(predicted_rewards) * cfg.algo.gamma
     # λ-values with v_{t+1} alignment and bootstrap = last value
     next_values = torch.cat((predicted_values[1:], predicted_values[-1:]), dim=0)
     lambda_values = ops.lambda_values(predicted_rewards, next_values, continues, cfg.algo.lmbda)
