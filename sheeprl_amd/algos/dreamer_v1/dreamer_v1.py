@@ -366,7 +366,7 @@ def train(
         qc_out = world_model.continue_model(latent_states).float()
         qc = td.Independent(td.Bernoulli(logits=qc_out), 1)
         continues_targets = (1 - data["terminated"]) * cfg.algo.gamma
-        continue_loss = cfg.algo.world_model.get("continue_scale_factor", 10.0) * -qc.log_prob(
+        continue_loss = cfg.algo.world_model.get("continue_scale_factor", 1.0) * -qc.log_prob(
             continues_targets
         ).mean()
     rec_loss = (
