@@ -59,10 +59,10 @@ def _setup(cfg, rank: int, world_size: int):
     from sheeprl_amd.algos.dreamer_v3.agent import build_agent
     from sheeprl_amd.algos.dreamer_v3.utils import Moments, prepare_obs
     from sheeprl_amd.data import EnvIndependentReplayBuffer, SequentialReplayBuffer
-    from sheeprl_amd.envs import spaces, vectorize_env
+    from sheeprl_amd.envs import vectorize_env
     from sheeprl_amd.optim import FusedAdam
     from sheeprl_amd.parallel import Runtime
-    from sheeprl_amd.utils.utils import Ratio, seed_everything
+    from sheeprl_amd.utils.utils import seed_everything
 
     seed_everything(cfg.seed + rank)
     torch.backends.cudnn.benchmark = True  # MIOpen find mode: avoid naive-conv fallback
@@ -261,8 +261,6 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
 
     if os.environ.get("SHEEPRL_AMD_PHASE_TIMING"):
         # coarse phase split over a few synchronized steps
-        import numpy as _np
-
         phases = {"env": 0.0, "sample": 0.0, "train": 0.0}
         for _ in range(5):
             torch.cuda.synchronize()

@@ -290,8 +290,6 @@ def _capture_train_step(
 
 @register_algorithm(name="dreamer_v2")
 def main(runtime: Runtime, cfg: Any) -> None:
-    from sheeprl_amd.algos.dreamer_v3.dreamer_v3 import main as dv3_main  # noqa: F401 (shape reference)
-
     device = runtime.device
     log_dir = get_log_dir(runtime, cfg.root_dir, cfg.run_name)
     logger = get_logger(runtime, cfg, log_dir)
