@@ -187,6 +187,39 @@ def test_dreamer_v3_episode_boundary_resets(tmp_path):
 
 
 @pytest.mark.timeout(300)
+def test_dreamer_v3_decoupled_rssm(tmp_path):
+    # the sequence-parallel representation variant (reference agent.py:501)
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=dreamer_v3",
+                "algo.world_model.decoupled_rssm=True",
+                "algo.dense_units=8",
+                "algo.mlp_layers=1",
+                "algo.world_model.encoder.cnn_channels_multiplier=2",
+                "algo.world_model.recurrent_model.recurrent_state_size=8",
+                "algo.world_model.transition_model.hidden_size=8",
+                "algo.world_model.representation_model.hidden_size=8",
+                "algo.world_model.discrete_size=4",
+                "algo.world_model.stochastic_size=4",
+                "algo.per_rank_batch_size=2",
+                "algo.per_rank_sequence_length=4",
+                "algo.horizon=3",
+                "algo.mlp_keys.encoder=[state]",
+                "algo.total_steps=24",
+                "algo.learning_starts=4",
+                "algo.replay_ratio=0.5",
+                "buffer.size=64",
+                "algo.run_test=False",
+                "dry_run=False",
+            ],
+        ),
+    )
+
+
+@pytest.mark.timeout(300)
 def test_dreamer_v3_checkpoint_resume(tmp_path):
     args = standard_args(
         tmp_path,

@@ -66,3 +66,42 @@ def test_gradsync_matches_ddp_average():
     for p in ps:
         p.join(120)
     assert all(p.exitcode == 0 for p in ps)
+
+
+def _collectives_worker(rank, world_size, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from sheeprl_amd.parallel import Runtime
+
+        rt = Runtime(devices=1, accelerator="cpu")
+
+        # object scatter with equal-size chunks (the decoupled rollout path)
+        out = [None]
+        chunks = [None] + [{"a": i + 1} for i in range(world_size - 1)] if rank == 0 else None
+        rt.scatter_object_list(out, chunks, src=0)
+        if rank > 0:
+            assert out[0] == {"a": rank}
+
+        # object broadcast
+        payload = [{"x": 42}] if rank == 0 else [None]
+        rt.broadcast_object_list(payload, src=0)
+        assert payload[0] == {"x": 42}
+
+        # tensor all_reduce through the wrapper (mean semantics)
+        t = torch.tensor([float(rank + 1)])
+        out_t = rt.all_reduce(t, op="sum")
+        assert out_t.item() == sum(range(1, world_size + 1))
+        assert rt.all_reduce(t, op="mean").item() == pytest.approx(1.5)
+
+        # gather_object to rank 0
+        gathered = rt.gather_object({"r": rank}, dst=0)
+        if rank == 0:
+            assert [g["r"] for g in gathered] == list(range(world_size))
+    finally:
+        dist.destroy_process_group()
+
+
+def test_runtime_object_collectives_world2():
+    mp.spawn(_collectives_worker, args=(2, 29581), nprocs=2, join=True)
