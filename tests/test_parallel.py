@@ -75,7 +75,9 @@ def _collectives_worker(rank, world_size, port):
     try:
         from sheeprl_amd.parallel import Runtime
 
-        rt = Runtime(devices=1, accelerator="cpu")
+        rt = Runtime(devices=world_size, accelerator="cpu")
+        rt.global_rank = rank
+        rt.world_size = world_size
 
         # object scatter with equal-size chunks (the decoupled rollout path)
         out = [None]
