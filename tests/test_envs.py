@@ -116,7 +116,7 @@ def test_frame_stack_dilation():
     assert obs["rgb"].shape[0] == 9
 
 
-def test_time_limit_truncates():
+def test_time_limit_truncates_pendulum():
     from sheeprl_amd.envs import wrappers
     from sheeprl_amd.envs.classic import PendulumEnv
 
