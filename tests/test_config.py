@@ -47,3 +47,20 @@ def test_exp_child_group_override_beats_base():
     cfg = compose(["exp=dreamer_v3_100k_ms_pacman"])
     assert cfg.algo.world_model.recurrent_model.recurrent_state_size == 512
     assert cfg.algo.dense_units == 512
+
+
+def test_registry_completeness():
+    """Every registered algorithm has an evaluator and a composable config."""
+    import sheeprl_amd.algos  # noqa: F401
+    from sheeprl_amd.config import compose
+    from sheeprl_amd.utils.registry import algorithm_registry, evaluation_registry
+
+    assert len(algorithm_registry) == 17
+    missing = set(algorithm_registry) - set(evaluation_registry)
+    assert not missing, f"algorithms without evaluators: {missing}"
+    for name in algorithm_registry:
+        extra = []
+        if name.endswith("_finetuning"):
+            extra = ["checkpoint.exploration_ckpt_path=/tmp/x.ckpt"]
+        cfg = compose([f"exp={name}", "env=dummy", *extra])
+        assert cfg.algo.name == name
