@@ -230,17 +230,34 @@ def available_agents() -> None:
             print(f"{name:24s} {entry.module} decoupled={entry.decoupled}")
 
 
+_USAGE = """usage: python -m sheeprl_amd exp=<experiment> [group=value] [a.b.c=value ...]
+       python -m sheeprl_amd eval checkpoint_path=<ckpt> [overrides...]
+       python -m sheeprl_amd register checkpoint_path=<ckpt> [overrides...]
+       python -m sheeprl_amd agents
+
+Experiments live in sheeprl_amd/configs/exp/ (e.g. exp=dreamer_v3, exp=ppo);
+run `python -m sheeprl_amd agents` to list the registered algorithms."""
+
+
 def main() -> None:
     warnings.filterwarnings("ignore", category=UserWarning, module="torch.distributed")
     argv = sys.argv[1:]
-    if argv and argv[0] == "eval":
+    if not argv or argv[0] in ("-h", "--help", "help"):
+        print(_USAGE)
+        return
+    if argv[0] == "eval":
         evaluation(argv[1:])
-    elif argv and argv[0] == "register":
+    elif argv[0] == "register":
         registration(argv[1:])
-    elif argv and argv[0] == "agents":
+    elif argv[0] == "agents":
         available_agents()
     else:
-        run(argv)
+        try:
+            run(argv)
+        except ValueError as e:
+            if "missing required config values" in str(e):
+                raise SystemExit(f"{e}\n\n{_USAGE}") from None
+            raise
 
 
 if __name__ == "__main__":
