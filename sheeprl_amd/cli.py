@@ -111,6 +111,7 @@ class _ReproducibleEntry:
 
     def __call__(self, runtime: Runtime, cfg: DotDict):
         _import_algorithms()
+        _import_external_modules(cfg)
         fn = algorithm_registry[self.algo_name].entrypoint
         seed_everything(cfg.seed + runtime.global_rank)
         torch.set_num_threads(int(cfg.get("num_threads", 1)))
@@ -128,9 +129,21 @@ class _ReproducibleEntry:
         return fn(runtime, cfg)
 
 
+def _import_external_modules(cfg: DotDict) -> None:
+    """Import user modules listed in ``import_modules`` so out-of-tree
+    ``@register_algorithm`` entrypoints are registered (the reference's
+    external-algorithm howto flow; combine with SHEEPRL_AMD_SEARCH_PATH for
+    the configs)."""
+    import importlib
+
+    for mod in cfg.get("import_modules", []) or []:
+        importlib.import_module(mod)
+
+
 def run_algorithm(cfg: DotDict) -> None:
     """Registry lookup + launch (parity: cli.py:60-199)."""
     _import_algorithms()
+    _import_external_modules(cfg)
     if cfg.metric.get("log_level", 1) <= 0 or cfg.metric.get("disable_timer", False):
         timer.disabled = True
     runtime = _build_runtime(cfg)

@@ -64,3 +64,42 @@ def test_registry_completeness():
             extra = ["checkpoint.exploration_ckpt_path=/tmp/x.ckpt"]
         cfg = compose([f"exp={name}", "env=dummy", *extra])
         assert cfg.algo.name == name
+
+
+def test_external_algorithm_via_search_path(tmp_path, monkeypatch):
+    """End-to-end external-algorithm flow: configs from SHEEPRL_AMD_SEARCH_PATH,
+    the module imported via the import_modules config key, entrypoint resolved
+    from the registry (docs/register_external_algorithm.md)."""
+    import subprocess
+    import sys
+    import textwrap
+
+    pkg = tmp_path / "myext"
+    pkg.mkdir()
+    (pkg / "__init__.py").write_text("")
+    (pkg / "algo.py").write_text(textwrap.dedent("""
+        from sheeprl_amd.utils.registry import register_algorithm
+
+        @register_algorithm(name="my_algo")
+        def main(runtime, cfg):
+            print("MY_ALGO_RAN", cfg.algo.total_steps)
+    """))
+    cfgdir = tmp_path / "configs"
+    (cfgdir / "algo").mkdir(parents=True)
+    (cfgdir / "exp").mkdir()
+    (cfgdir / "algo" / "my_algo.yaml").write_text(
+        "name: my_algo\ntotal_steps: 7\nper_rank_batch_size: 1\n"
+    )
+    (cfgdir / "exp" / "my_algo.yaml").write_text(
+        "defaults:\n  - override /algo: my_algo\nimport_modules: [myext.algo]\n"
+    )
+    env = dict(**__import__("os").environ)
+    env["SHEEPRL_AMD_SEARCH_PATH"] = str(cfgdir)
+    repo_root = __import__("os").path.dirname(__import__("os").path.dirname(__import__("os").path.abspath(__file__)))
+    env["PYTHONPATH"] = f"{tmp_path}:{repo_root}:{env.get('PYTHONPATH', '')}"
+    out = subprocess.run(
+        [sys.executable, "-m", "sheeprl_amd", "exp=my_algo", "env=dummy",
+         "buffer.size=8", "runtime.devices=1", "runtime.accelerator=cpu"],
+        capture_output=True, text=True, env=env, cwd=str(tmp_path), timeout=240,
+    )
+    assert "MY_ALGO_RAN 7" in out.stdout, out.stdout + out.stderr
