@@ -153,6 +153,8 @@ def run_algorithm(cfg: DotDict) -> None:
 def run(args: Optional[List[str]] = None) -> None:
     argv = list(sys.argv[1:] if args is None else args)
     cfg = compose(argv)
+    _import_algorithms()
+    _import_external_modules(cfg)
     if cfg.checkpoint.get("resume_from"):
         cfg = resume_from_checkpoint(cfg)
     check_configs(cfg)
