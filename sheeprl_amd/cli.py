@@ -167,6 +167,7 @@ def eval_algorithm(cfg: DotDict) -> None:
     """Load ckpt on a single-device runtime and dispatch the evaluator
     (parity: cli.py:202-268)."""
     _import_algorithms()
+    _import_external_modules(cfg)
     name = cfg.algo.name
     if name not in evaluation_registry:
         raise ValueError(f"no evaluation registered for '{name}'")
