@@ -595,3 +595,24 @@ def test_g16_gemm_kernels_match_torch():
     assert torch.allclose(m.view(B, S, KD), m_ref, atol=2e-2, rtol=2e-2), (m.view(B,S,KD) - m_ref).abs().max()
     assert torch.allclose(sp, s_ref, atol=2e-2, rtol=2e-2)
     assert (z.view(B, S, KD).sum(-1) == 1).all()
+
+
+@requires_gpu
+@pytest.mark.timeout(300)
+def test_mfma_fragment_layout_and_grid_barrier():
+    """The v_mfma_f32_16x16x32_bf16 fragment mapping (A: row=lane&15,
+    k=(lane>>4)*8+e; C/D: col=lane&15, row=(lane>>4)*4+reg) and the
+    device-wide barrier used by the persistent-scan experiment."""
+    from sheeprl_amd.ops._ext import require_ext
+
+    ext = require_ext()
+    torch.manual_seed(0)
+    x = torch.randn(16, 96, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(128, 96, device="cuda", dtype=torch.bfloat16)
+    y = ext.pk_gemm16_test(x, w)
+    ref = x.float() @ w.float().t()
+    assert (y - ref).abs().max() < 0.2
+
+    out = ext.pk_barrier_test(24, 100)
+    assert out[-1].item() == 0          # no round observed a partial count
+    assert bool((out[:-1] == 24).all())
