@@ -39,7 +39,12 @@ class DevicePrefetcher:
         self._q: "queue.Queue" = queue.Queue(maxsize=depth)
         self._stop = threading.Event()
         self._err: Optional[BaseException] = None
-        self._pinned: Dict[str, torch.Tensor] = {}
+        # ring of pinned staging sets: a slot is only reused after depth+2
+        # batches, by which time its async H2D must have been consumed — a
+        # single set would let the worker overwrite host memory still being
+        # copied by a previous non_blocking transfer
+        self._pinned_ring = [dict() for _ in range(depth + 2)]
+        self._ring_idx = 0
         self._thread = threading.Thread(target=self._worker, daemon=True)
         self._thread.start()
 
@@ -47,12 +52,14 @@ class DevicePrefetcher:
         if not self.use_cuda:
             return host_batch, None
         out: Dict[str, torch.Tensor] = {}
+        pinned = self._pinned_ring[self._ring_idx]
+        self._ring_idx = (self._ring_idx + 1) % len(self._pinned_ring)
         with torch.cuda.stream(self.stream):
             for k, v in host_batch.items():
-                pin = self._pinned.get(k)
+                pin = pinned.get(k)
                 if pin is None or pin.shape != v.shape or pin.dtype != v.dtype:
                     pin = torch.empty_like(v, pin_memory=True)
-                    self._pinned[k] = pin
+                    pinned[k] = pin
                 pin.copy_(v)
                 out[k] = pin.to(self.device, non_blocking=True)
             event = torch.cuda.Event()
