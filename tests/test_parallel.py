@@ -101,6 +101,14 @@ def _collectives_worker(rank, world_size, port):
         gathered = rt.gather_object({"r": rank}, dst=0)
         if rank == 0:
             assert [g["r"] for g in gathered] == list(range(world_size))
+
+        # rank-independent metric aggregation (all-gathered per-rank values)
+        from sheeprl_amd.utils.metric import RankIndependentMetricAggregator
+
+        agg = RankIndependentMetricAggregator({"m": "mean"})
+        agg.update("m", float(rank + 1))
+        per_rank = agg.compute()
+        assert [d["m"] for d in per_rank] == [1.0, 2.0][:world_size]
     finally:
         dist.destroy_process_group()
 
