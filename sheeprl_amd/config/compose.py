@@ -86,7 +86,7 @@ def _load_group_file(group: str, name: str, _seen: Optional[set] = None) -> dict
     out: dict = {}
     for entry in defaults:
         if isinstance(entry, str):
-            if entry in ("_self_", "default") and entry == "_self_":
+            if entry == "_self_":
                 continue
             merge(out, _load_group_file(group, entry, _seen))
         elif isinstance(entry, dict):

@@ -103,3 +103,11 @@ def test_external_algorithm_via_search_path(tmp_path, monkeypatch):
         capture_output=True, text=True, env=env, cwd=str(tmp_path), timeout=240,
     )
     assert "MY_ALGO_RAN 7" in out.stdout, out.stdout + out.stderr
+
+
+def test_cli_group_selection_beats_exp_override():
+    # dreamer_v3_100k_ms_pacman's defaults select env=atari; an explicit CLI
+    # group selection must win over the exp's override (compose.py:173-179).
+    cfg = compose(["exp=dreamer_v3_100k_ms_pacman", "env=dummy", "env.id=dummy"])
+    assert cfg.env.id == "dummy"
+    assert cfg.algo.name == "dreamer_v3"
