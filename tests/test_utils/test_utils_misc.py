@@ -89,3 +89,52 @@ def test_checkpoint_keep_last(tmp_path):
         cb._prune(str(ckdir))
     left = sorted(p.name for p in ckdir.glob("*.ckpt"))
     assert len(left) == 2 and "ckpt_4_0.ckpt" in left
+
+
+def test_ckpt_rb_truncation_mark_and_restore():
+    """The buffer-consistency trick (callback.py:40-71): the slot before the
+    write head is marked truncated for the snapshot, then restored."""
+    from sheeprl_amd.data.buffers import EnvIndependentReplayBuffer, EpisodeBuffer, ReplayBuffer
+    from sheeprl_amd.utils.callback import CheckpointCallback
+
+    rb = ReplayBuffer(8, 2)
+    step = {
+        "obs": np.zeros((1, 2, 3), np.float32),
+        "truncated": np.zeros((1, 2, 1), np.float32),
+    }
+    for _ in range(3):
+        rb.add(step)
+    restore = CheckpointCallback._ckpt_rb(rb)
+    marked = np.asarray(rb["truncated"][(rb._pos - 1) % rb.buffer_size])
+    assert (marked == 1).all()
+    CheckpointCallback._restore_rb(rb, restore)
+    assert (np.asarray(rb["truncated"][(rb._pos - 1) % rb.buffer_size]) == 0).all()
+
+    # EnvIndependent recursion: every sub-buffer marked then restored
+    eib = EnvIndependentReplayBuffer(8, n_envs=2)
+    sub_step = {
+        "obs": np.zeros((1, 1, 3), np.float32),
+        "truncated": np.zeros((1, 1, 1), np.float32),
+    }
+    for _ in range(2):
+        eib.add({k: np.repeat(v, 2, axis=1) for k, v in sub_step.items()})
+    restore = CheckpointCallback._ckpt_rb(eib)
+    for b in eib.buffer:
+        assert (np.asarray(b["truncated"][(b._pos - 1) % b.buffer_size]) == 1).all()
+    CheckpointCallback._restore_rb(eib, restore)
+    for b in eib.buffer:
+        assert (np.asarray(b["truncated"][(b._pos - 1) % b.buffer_size]) == 0).all()
+
+    # EpisodeBuffer: open (unfinished) episodes dropped from the snapshot,
+    # put back afterwards
+    eb = EpisodeBuffer(40, 2, n_envs=1, obs_keys=("obs",))
+    eb.add({
+        "obs": np.zeros((3, 1, 2), np.float32),
+        "terminated": np.zeros((3, 1, 1), np.float32),
+        "truncated": np.zeros((3, 1, 1), np.float32),
+    })
+    assert any(o is not None for o in eb._open)
+    restore = CheckpointCallback._ckpt_rb(eb)
+    assert all(o is None for o in eb._open)
+    CheckpointCallback._restore_rb(eb, restore)
+    assert any(o is not None for o in eb._open)
