@@ -402,7 +402,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
     last_log = 0
     graphed_step = None
     _n_train_calls = 0
-    # p2e capture is OPT-IN for now (algo.hipgraphs=true): the captured
+    # p2e capture is OPT-IN for now (algo.hip_graphs=true): the captured
     # multi-critic step segfaulted in round-1 validation; the single-family
     # Dreamer mains capture by default
     want_graphs = (
