@@ -78,3 +78,36 @@ def test_ppo_policy_loss_directions():
     loss3 = ppo_policy_loss(lp3, torch.tensor([0.0]), torch.tensor([-1.0]), clip_coef=0.2)
     loss3.backward()
     assert abs(lp3.grad.item()) < 1e-8
+
+
+def test_dv3_kl_balancing_gradient_routing():
+    """dyn term must push the PRIOR toward the (detached) posterior; repr term
+    must push the POSTERIOR toward the (detached) prior; free-nats clamp kills
+    gradients when KL is below the floor (dreamer_v3/loss.py:39-55)."""
+    from sheeprl_amd.algos.dreamer_v3.loss import categorical_kl
+
+    torch.manual_seed(0)
+    post = torch.nn.Parameter(torch.randn(4, 8, 6))
+    prior = torch.nn.Parameter(torch.randn(4, 8, 6))
+
+    dyn = categorical_kl(post.detach(), prior)
+    dyn.mean().backward()
+    assert post.grad is None and prior.grad is not None and prior.grad.abs().sum() > 0
+    g_prior = prior.grad.clone()
+    prior.grad = None
+
+    rep = categorical_kl(post, prior.detach())
+    rep.mean().backward()
+    assert prior.grad is None and post.grad is not None and post.grad.abs().sum() > 0
+
+    # a gradient step on the prior reduces the dynamic KL
+    with torch.no_grad():
+        prior2 = prior - 5.0 * g_prior
+    assert categorical_kl(post.detach(), prior2).mean() < dyn.mean()
+
+    # free nats: identical distributions (KL=0 < floor) give zero gradient
+    same = torch.nn.Parameter(torch.randn(4, 8, 6))
+    kl = categorical_kl(same.detach(), same)
+    clamped = torch.maximum(kl, torch.full_like(kl, 1.0))
+    clamped.mean().backward()
+    assert same.grad.abs().max() == 0
