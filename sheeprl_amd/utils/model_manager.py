@@ -117,11 +117,21 @@ def register_models_from_checkpoint(
     state = torch.load(ckpt_path, map_location="cpu", weights_only=False)
     algo_name = cfg.algo.name
     base = algo_name.replace("_exploration", "").replace("_finetuning", "").replace("_decoupled", "")
-    try:
-        utils_mod = importlib.import_module(f"sheeprl_amd.algos.{base}.utils")
-        to_register = getattr(utils_mod, "MODELS_TO_REGISTER", None)
-    except ImportError:
-        to_register = None
+    to_register = None
+    # MODELS_TO_REGISTER lives either in the algo's utils module or in the
+    # entrypoint module itself (e.g. a2c/a2c.py, p2e_dv3/p2e_dv3_exploration.py)
+    for mod_name in (
+        f"sheeprl_amd.algos.{base}.utils",
+        f"sheeprl_amd.algos.{base}.{algo_name}",
+        f"sheeprl_amd.algos.{base}.{base}",
+    ):
+        try:
+            mod = importlib.import_module(mod_name)
+        except ImportError:
+            continue
+        to_register = getattr(mod, "MODELS_TO_REGISTER", None)
+        if to_register:
+            break
     if not to_register:
         to_register = {k for k in state if isinstance(state[k], dict) and any("weight" in kk for kk in state[k])}
     manager = ModelManager(registry_dir)
