@@ -49,6 +49,7 @@ setup(
         "console_scripts": [
             "sheeprl-amd=sheeprl_amd.cli:main",
             "sheeprl-amd-eval=sheeprl_amd.cli:evaluation",
+            "sheeprl-amd-registration=sheeprl_amd.cli:registration",
             "sheeprl-amd-agents=sheeprl_amd.cli:available_agents",
         ]
     },
