@@ -1,30 +1,21 @@
-"""Wall-clock benchmark harness (parity: the reference's benchmarks/
-benchmark.py): times a full CLI training run of any *_benchmarks experiment.
+"""Throughput benchmark driver (parity: benchmarks/benchmark.py): run one of
+the *_benchmarks experiment presets through the CLI and report wall time.
 
-    python benchmarks/benchmark.py exp=ppo_benchmarks [overrides...]
-    python benchmarks/benchmark.py exp=dreamer_v3_benchmarks runtime=gpu-bf16
+    python benchmarks/benchmark.py exp=ppo_benchmarks
+    python benchmarks/benchmark.py exp=dreamer_v3_benchmarks
+
+The flagship MI355X benchmark (DreamerV3 env-frames/s vs BASELINE.md) lives
+in `bench.py` at the repo root; this script is for quick cross-algorithm
+comparisons on the reference's own benchmark presets.
 """
-
-from __future__ import annotations
 
 import sys
 import time
-from pathlib import Path
-
-sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 from sheeprl_amd.cli import run
 
-
-def main() -> None:
-    args = sys.argv[1:]
-    if not any(a.startswith("exp=") for a in args):
-        raise SystemExit("usage: python benchmarks/benchmark.py exp=<name>_benchmarks [overrides...]")
-    t0 = time.perf_counter()
-    run(args)
-    elapsed = time.perf_counter() - t0
-    print(f"\n[benchmark] {' '.join(args)} -> {elapsed:.2f} s")
-
-
 if __name__ == "__main__":
-    main()
+    args = sys.argv[1:] or ["exp=ppo_benchmarks"]
+    tic = time.perf_counter()
+    run(args)
+    print(f"\nbenchmark wall time: {time.perf_counter() - tic:.2f} s")
