@@ -21,7 +21,7 @@ from typing import Any, Dict, List
 
 import numpy as np
 import torch
-from torch.nn.utils import parameters_to_vector, vector_to_parameters
+from sheeprl_amd.parallel import flat_to_params, params_to_flat
 
 from sheeprl_amd.algos.ppo.agent import PPOAgent, PPOPlayer
 from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, value_loss
@@ -54,9 +54,9 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
     ppo_player = PPOPlayer(agent.feature_extractor, agent.actor, agent.critic)
 
     # initial weights come from the lead trainer (rank 1)
-    flat = parameters_to_vector(agent.parameters()).detach()
+    flat = params_to_flat(agent.parameters()).detach()
     runtime.broadcast(flat, src=1, group=pt_group)
-    vector_to_parameters(flat, agent.parameters())
+    flat_to_params(flat, agent.parameters())
 
     n_trainers = runtime.world_size - 1
     rollout_steps = cfg.algo.rollout_steps
@@ -133,7 +133,7 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
 
         # receive updated weights from the lead trainer
         runtime.broadcast(flat, src=1, group=pt_group)
-        vector_to_parameters(flat, agent.parameters())
+        flat_to_params(flat, agent.parameters())
 
         # receive trainer metrics
         payload: List[Any] = [None]
@@ -207,7 +207,7 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
         if payload[0] is not None:
             optimizer.load_state_dict(payload[0])
 
-    flat = parameters_to_vector(agent.parameters()).detach()
+    flat = params_to_flat(agent.parameters()).detach()
     if rank == 1:
         runtime.broadcast(flat, src=1, group=pt_group)
 
@@ -260,7 +260,7 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
                 n_batches += 1
 
         if rank == 1:
-            flat = parameters_to_vector(agent.parameters()).detach()
+            flat = params_to_flat(agent.parameters()).detach()
             runtime.broadcast(flat, src=1, group=pt_group)
             metrics = {
                 "Loss/policy_loss": losses["policy"] / max(n_batches, 1),

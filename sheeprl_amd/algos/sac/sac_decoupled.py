@@ -13,7 +13,7 @@ from typing import Any, List
 
 import numpy as np
 import torch
-from torch.nn.utils import parameters_to_vector, vector_to_parameters
+from sheeprl_amd.parallel import flat_to_params, params_to_flat
 
 from sheeprl_amd.algos.sac.agent import SACAgent, SACPlayer
 from sheeprl_amd.algos.sac.sac import train as sac_train
@@ -66,9 +66,9 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
 
     agent = _agent_for(cfg, obs_space, action_space, device)
     sac_player = SACPlayer(agent.actor)
-    flat = parameters_to_vector(agent.actor.parameters()).detach()
+    flat = params_to_flat(agent.actor.parameters()).detach()
     runtime.broadcast(flat, src=1, group=pt_group)
-    vector_to_parameters(flat, agent.actor.parameters())
+    flat_to_params(flat, agent.actor.parameters())
 
     n_trainers = runtime.world_size - 1
     rb = ReplayBuffer(int(cfg.buffer.size), num_envs, obs_keys=("obs",))
@@ -143,7 +143,7 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
                 out: List[Any] = [None]
                 runtime.scatter_object_list(out, chunks, src=0, group=world_group)
                 runtime.broadcast(flat, src=1, group=pt_group)
-                vector_to_parameters(flat, agent.actor.parameters())
+                flat_to_params(flat, agent.actor.parameters())
                 payload: List[Any] = [None]
                 runtime.broadcast_object_list(payload, src=1, group=pt_group)
                 for k, v in (payload[0] or {}).items():
@@ -220,7 +220,7 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
     aggregator = MetricAggregator({k: "mean" for k in AGGREGATOR_KEYS})
 
     if rank == 1:
-        runtime.broadcast(parameters_to_vector(agent.actor.parameters()).detach(), src=1, group=pt_group)
+        runtime.broadcast(params_to_flat(agent.actor.parameters()).detach(), src=1, group=pt_group)
 
     update = 0
     while True:
@@ -241,7 +241,7 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
                 batch, aggregator, update, cfg, cfg.env.num_envs,
             )
         if rank == 1:
-            runtime.broadcast(parameters_to_vector(agent.actor.parameters()).detach(), src=1, group=pt_group)
+            runtime.broadcast(params_to_flat(agent.actor.parameters()).detach(), src=1, group=pt_group)
             runtime.broadcast_object_list([aggregator.compute()], src=1, group=pt_group)
             aggregator.reset()
             if want_ckpt:

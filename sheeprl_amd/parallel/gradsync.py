@@ -53,7 +53,9 @@ class _Bucket:
         for i, p in enumerate(self.params):
             dst = self.flat[self.offsets[i] : self.offsets[i] + self.numels[i]]
             if self.seen[i] and p.grad is not None:
-                dst.copy_(p.grad.detach().view(-1))
+                # reshape, not view: channels_last conv grads have permuted
+                # strides and are not viewable as 1-D
+                dst.copy_(p.grad.detach().reshape(-1))
             else:
                 dst.zero_()
         self.flat.div_(world_size)

@@ -351,3 +351,22 @@ def get_single_device_runtime(runtime: Runtime) -> Runtime:
     r.world_size = 1
     r._launched = True
     return r
+
+
+def params_to_flat(parameters) -> torch.Tensor:
+    """Flatten parameters into one vector (layout-safe replacement for
+    torch.nn.utils.parameters_to_vector, which view(-1)s and raises on
+    channels_last conv weights — the setup_module layout on GPU boxes)."""
+    return torch.cat([p.detach().reshape(-1) for p in parameters])
+
+
+def flat_to_params(vec: torch.Tensor, parameters) -> None:
+    """Copy a flat vector back into parameters IN PLACE, preserving each
+    parameter's memory format (vector_to_parameters would swap the param
+    storage to a standard-layout tensor)."""
+    off = 0
+    with torch.no_grad():
+        for p in parameters:
+            n = p.numel()
+            p.copy_(vec[off : off + n].view(p.shape))
+            off += n

@@ -115,3 +115,64 @@ def _collectives_worker(rank, world_size, port):
 
 def test_runtime_object_collectives_world2():
     mp.spawn(_collectives_worker, args=(2, 29581), nprocs=2, join=True)
+
+
+class ConvNet(nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.c1 = nn.Conv2d(3, 8, 3, padding=1)
+        self.c2 = nn.Conv2d(8, 4, 3, padding=1)
+
+    def forward(self, x):
+        return self.c2(torch.relu(self.c1(x))).mean((-1, -2))
+
+
+def _chlast_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(0)
+        net = ConvNet().to(memory_format=torch.channels_last)
+        gs = GradSync(net, bucket_cap_mb=1)
+        gs.broadcast_params(src=0)
+        torch.manual_seed(200 + rank)
+        x = torch.randn(4, 3, 8, 8).to(memory_format=torch.channels_last)
+        net(x).pow(2).mean().backward()
+        gs.finalize()  # channels_last grads must flatten into the buckets
+        torch.manual_seed(0)
+        ref = ConvNet()
+        for rp, p in zip(ref.parameters(), net.parameters()):
+            rp.data.copy_(p.data)
+        torch.manual_seed(200 + rank)
+        ref(torch.randn(4, 3, 8, 8)).pow(2).mean().backward()
+        for (n, p), rp in zip(net.named_parameters(), ref.parameters()):
+            g = rp.grad.clone()
+            dist.all_reduce(g)
+            g /= world
+            assert torch.allclose(p.grad, g, atol=1e-6), f"mismatch on {n}"
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_gradsync_channels_last_convs():
+    """Conv grads under channels_last (the setup_module path on GPU boxes)
+    must bucket correctly — view(-1) on permuted strides would raise."""
+    mp.spawn(_chlast_worker, args=(2, 29591), nprocs=2, join=True)
+
+
+def test_flat_params_roundtrip_channels_last():
+    """params_to_flat/flat_to_params must survive channels_last conv weights
+    and preserve the receiver's memory format (runtime.py helpers used for
+    decoupled weight transport)."""
+    from sheeprl_amd.parallel import flat_to_params, params_to_flat
+
+    torch.manual_seed(0)
+    src = ConvNet().to(memory_format=torch.channels_last)
+    dst = ConvNet().to(memory_format=torch.channels_last)
+    vec = params_to_flat(src.parameters())
+    flat_to_params(vec, dst.parameters())
+    for a, b in zip(src.parameters(), dst.parameters()):
+        assert torch.equal(a, b)
+    assert dst.c1.weight.is_contiguous(memory_format=torch.channels_last)
