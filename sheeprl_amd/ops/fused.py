@@ -146,7 +146,13 @@ def ema_update_(target_params: Iterable[Tensor], source_params: Iterable[Tensor]
     """t <- tau * p + (1 - tau) * t, multi-tensor."""
     tgt = list(target_params)
     src = list(source_params)
-    if tgt and use_hip(tgt[0]):
+    if (
+        tgt
+        and use_hip(tgt[0])
+        # the HIP kernel walks data_ptr flat: both sides must share the same
+        # dense layout (true for deepcopy/convert pairs; guard the rest)
+        and all(t.stride() == s.stride() for t, s in zip(tgt, src))
+    ):
         require_ext().ema_update(tgt, src, float(tau))
         return
     torch._foreach_mul_(tgt, 1.0 - tau)

@@ -185,3 +185,17 @@ def test_ratio_scheduler():
     st = r.state_dict()
     r2 = Ratio(0.5).load_state_dict(st)
     assert r2(12) == 1
+
+
+def test_ema_update_mixed_layout_falls_back_correctly():
+    """ema_update_ must stay correct when target and source layouts differ
+    (the HIP fast path requires identical strides; mismatches use foreach)."""
+    import torch
+
+    from sheeprl_amd.ops.fused import ema_update_
+
+    t = torch.nn.Conv2d(4, 8, 3).weight.detach().clone().contiguous(memory_format=torch.channels_last)
+    s = torch.randn_like(t).contiguous()  # standard layout
+    expect = 0.1 * s + 0.9 * t
+    ema_update_([t], [s], tau=0.1)
+    assert torch.allclose(t, expect, atol=1e-6)
