@@ -124,7 +124,11 @@ class FusedAdam(torch.optim.Optimizer):
                     True,
                 )
                 self._zeroes_grads_in_kernel = True
-            elif use_hip(params[0]):
+            elif use_hip(params[0]) and all(
+                # the dev kernel walks p flat and contiguous()-copies g: both
+                # must be standard-contiguous or the orders diverge
+                p.is_contiguous() and g.is_contiguous() for p, g in zip(params, grads)
+            ):
                 st0 = self.state[group["params"][0]]
                 if "step_t" not in st0 or st0["step_t"].device != params[0].device:
                     st0["step_t"] = torch.zeros(1, dtype=torch.float32, device=params[0].device)

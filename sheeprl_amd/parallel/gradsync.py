@@ -68,7 +68,9 @@ class _Bucket:
         for i, p in enumerate(self.params):
             src = self.flat[self.offsets[i] : self.offsets[i] + self.numels[i]].view(p.shape)
             if p.grad is None:
-                p.grad = src.clone()
+                # empty_like preserves the param's memory format so the grad
+                # layout matches (keeps fused-optimizer fast paths eligible)
+                p.grad = torch.empty_like(p).copy_(src)
             else:
                 p.grad.detach().copy_(src)
         self.ready = 0
