@@ -145,9 +145,20 @@ class GradSync:
         self._any_ready = False
 
     def broadcast_params(self, src: int = 0) -> None:
+        """One flat broadcast per dtype instead of per-tensor (init-time sync;
+        also layout-safe: flattening copies channels_last tensors densely)."""
         with torch.no_grad():
+            by_dtype = {}
             for p in self.module.parameters():
-                dist.broadcast(p.data, src=src, group=self.group)
+                by_dtype.setdefault(p.dtype, []).append(p.data)
             for b in self.module.buffers():
                 if b.dtype.is_floating_point or b.dtype in (torch.int32, torch.int64, torch.uint8, torch.bool):
-                    dist.broadcast(b.data, src=src, group=self.group)
+                    by_dtype.setdefault(b.dtype, []).append(b.data)
+            for ts in by_dtype.values():
+                flat = torch.cat([t.reshape(-1) for t in ts])
+                dist.broadcast(flat, src=src, group=self.group)
+                off = 0
+                for t in ts:
+                    n = t.numel()
+                    t.copy_(flat[off : off + n].view(t.shape))
+                    off += n
