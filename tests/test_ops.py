@@ -221,3 +221,38 @@ def test_fused_adam_mixed_layout_grads_match_torch_adam():
         opt1.step()
         opt2.step()
     assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_rmsprop_tf_semantics():
+    """TF-style RMSprop: ones-init square_avg and eps INSIDE the sqrt
+    (reference optim/rmsprop_tf.py:63-156); verified against a hand rollout."""
+    import torch
+
+    from sheeprl_amd.optim import RMSpropTF
+
+    torch.manual_seed(0)
+    p = torch.nn.Parameter(torch.tensor([1.0, -2.0]))
+    g = torch.tensor([0.5, 0.25])
+    lr, alpha, eps = 0.1, 0.9, 1e-10
+    opt = RMSpropTF([p], lr=lr, alpha=alpha, eps=eps)
+    v = torch.ones(2)
+    expect = p.detach().clone()
+    for _ in range(3):
+        p.grad = g.clone()
+        opt.step()
+        v = alpha * v + (1 - alpha) * g * g
+        expect = expect - lr * g / (v + eps).sqrt()
+    assert torch.allclose(p, expect, atol=1e-6)
+
+    # momentum variant accumulates buf = mom*buf + g/sqrt(v+eps)
+    p2 = torch.nn.Parameter(torch.tensor([1.0, -2.0]))
+    opt2 = RMSpropTF([p2], lr=lr, alpha=alpha, eps=eps, momentum=0.5)
+    v2, buf = torch.ones(2), torch.zeros(2)
+    expect2 = p2.detach().clone()
+    for _ in range(3):
+        p2.grad = g.clone()
+        opt2.step()
+        v2 = alpha * v2 + (1 - alpha) * g * g
+        buf = 0.5 * buf + g / (v2 + eps).sqrt()
+        expect2 = expect2 - lr * buf
+    assert torch.allclose(p2, expect2, atol=1e-6)
