@@ -176,3 +176,32 @@ def test_flat_params_roundtrip_channels_last():
     for a, b in zip(src.parameters(), dst.parameters()):
         assert torch.equal(a, b)
     assert dst.c1.weight.is_contiguous(memory_format=torch.channels_last)
+
+
+def _moments_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from sheeprl_amd.algos.dreamer_v3.utils import Moments
+        from sheeprl_amd.parallel import Runtime
+
+        rt = Runtime(devices=world, accelerator="cpu")
+        rt.global_rank = rank
+        rt.world_size = world
+        m = Moments(decay=0.0)  # decay 0: buffers become this batch's quantiles
+        x = torch.arange(10, dtype=torch.float32) + rank * 10  # rank0: 0..9, rank1: 10..19
+        low, invscale = m(x, rt)
+        # quantiles of the GLOBAL gathered sample 0..19, identical on all ranks
+        full = torch.arange(20, dtype=torch.float32)
+        assert torch.allclose(low, torch.quantile(full, 0.05), atol=1e-5)
+        assert torch.allclose(low + invscale, torch.quantile(full, 0.95), atol=1e-5)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_moments_percentiles_are_global(tmp_path):
+    """DV3 return normalization must use the ALL-GATHERED sample so every
+    rank scales advantages identically (reference dreamer_v3/utils.py:56-63)."""
+    mp.spawn(_moments_worker, args=(2, 29601), nprocs=2, join=True)
