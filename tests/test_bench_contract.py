@@ -46,3 +46,20 @@ def test_bench_two_rank_gloo_emits_single_json(tmp_path):
     assert rec["value"] > 0 and rec["higher_is_better"] is True
     assert rec["scaling"] == "weak" and rec["data"] == "synthetic"
     assert "ms_per_step" in rec and "config" in rec
+
+
+@pytest.mark.timeout(600)
+def test_bench_single_rank_default(tmp_path):
+    """Plain `python bench.py` (the driver's N=1 run) with tiny overrides."""
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = f"{repo}:{env.get('PYTHONPATH', '')}"
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"), "--steps", "2", "--warmup", "1", *TINY],
+        capture_output=True, text=True, env=env, cwd=str(tmp_path), timeout=570,
+    )
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    json_lines = [l for l in out.stdout.splitlines() if l.strip().startswith("{")]
+    assert len(json_lines) == 1, out.stdout
+    rec = json.loads(json_lines[0])
+    assert rec["n_gpus"] == 1 and rec["metric"] == "env_frames_per_sec"
