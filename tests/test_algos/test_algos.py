@@ -767,3 +767,27 @@ def test_graph_capture_plumbing_cpu(tmp_path, monkeypatch):
         _run(tmp_path, [f"exp={exp}", "env=dummy", "runtime.accelerator=cpu", "dry_run=False",
                         "checkpoint.every=0", "metric.log_every=100", "env.num_envs=2", "seed=0",
                         "algo.cnn_keys.encoder=[]", *tiny, *extra])
+
+
+@pytest.mark.timeout(420)
+def test_ppo_decoupled_two_trainers(tmp_path):
+    """world=3: player + TWO trainers — exercises the trainers' DDP group
+    all-reduce, equal-size chunk padding across trainers and the rank-1
+    metric/weight broadcasts (reference Join-context case)."""
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=ppo_decoupled",
+                "env=dummy",
+                "algo.total_steps=64",
+                "algo.rollout_steps=9",  # odd size -> uneven chunks get padded
+                "algo.update_epochs=1",
+                "algo.per_rank_batch_size=4",
+                "algo.run_test=False",
+                "dry_run=False",
+            ],
+        ),
+        devices="3",
+    )
