@@ -791,3 +791,27 @@ def test_ppo_decoupled_two_trainers(tmp_path):
         ),
         devices="3",
     )
+
+
+@pytest.mark.timeout(420)
+def test_sac_decoupled_two_trainers(tmp_path):
+    """world=3 SAC decoupled: sample-chunk scatter across two trainers."""
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=sac_decoupled",
+                "env=dummy",
+                "env.id=dummy_continuous",
+                "algo.total_steps=32",
+                "algo.learning_starts=8",
+                "algo.per_rank_batch_size=6",
+                "algo.replay_ratio=0.5",
+                "buffer.size=256",
+                "algo.run_test=False",
+                "dry_run=False",
+            ],
+        ),
+        devices="3",
+    )
