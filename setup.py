@@ -20,7 +20,10 @@ try:
     from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
     csrc = Path(__file__).parent / "sheeprl_amd" / "ops" / "csrc"
-    sources = sorted(str(p) for p in csrc.glob("*.hip")) + sorted(str(p) for p in csrc.glob("*.cpp"))
+    # exclude torch-hipify outputs (*_hip.hip) — generated at build time
+    sources = sorted(
+        str(p) for p in csrc.glob("*.hip") if not p.name.endswith("_hip.hip")
+    ) + sorted(str(p) for p in csrc.glob("*.cpp"))
     if sources:
         ext_modules.append(
             CUDAExtension(

@@ -40,7 +40,8 @@ def value_loss(
     v_loss_unclipped = (new_values - returns) ** 2
     v_clipped = old_values + torch.clamp(new_values - old_values, -clip_coef, clip_coef)
     v_loss_clipped = (v_clipped - returns) ** 2
-    v_loss = torch.max(v_loss_unclipped, v_loss_clipped)
+    # reference scales the clipped branch by 0.5 (sheeprl/algos/ppo/loss.py:61)
+    v_loss = 0.5 * torch.max(v_loss_unclipped, v_loss_clipped)
     if reduction == "mean":
         return v_loss.mean()
     if reduction == "sum":

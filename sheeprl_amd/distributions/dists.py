@@ -239,7 +239,19 @@ class TruncatedNormal(td.Distribution):
         return self._normal.log_prob(value) - torch.log(z.clamp_min(1e-8))
 
     def entropy(self) -> Tensor:
-        return self._normal.entropy()
+        # exact truncated-normal entropy (reference distribution.py:64 + log_scale):
+        # H = log(sqrt(2*pi*e)*scale) + log Z - (b*phi(b) - a*phi(a)) / (2Z)
+        # with a,b the standardized bounds and phi the standard-normal pdf
+        a = (torch.as_tensor(self.low, dtype=self.loc.dtype, device=self.loc.device) - self.loc) / self.scale
+        b = (torch.as_tensor(self.high, dtype=self.loc.dtype, device=self.loc.device) - self.loc) / self.scale
+        inv_sqrt_2pi = 1.0 / math.sqrt(2.0 * math.pi)
+        phi_a = torch.exp(-0.5 * a * a) * inv_sqrt_2pi
+        phi_b = torch.exp(-0.5 * b * b) * inv_sqrt_2pi
+        big_a = 0.5 * (1 + torch.erf(a / math.sqrt(2.0)))
+        big_b = 0.5 * (1 + torch.erf(b / math.sqrt(2.0)))
+        z = (big_b - big_a).clamp_min(torch.finfo(self.loc.dtype).eps)
+        const = 0.5 * math.log(2.0 * math.pi * math.e)
+        return const + self.scale.log() + z.log() - 0.5 * (b * phi_b - a * phi_a) / z
 
 
 class TanhNormal(td.Distribution):

@@ -77,6 +77,11 @@ class FusedAdam(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        # re-derived every step from the path actually taken: only the
+        # multi-tensor kernel zeroes grads in place, so a single group (or
+        # step) falling back to the dev/eager path must re-enable the training
+        # loop's zero_grad or gradients would accumulate across steps
+        all_groups_zeroed_in_kernel = True
         for group in self.param_groups:
             params, grads, exp_avgs, exp_avg_sqs = [], [], [], []
             for p in group["params"]:
@@ -123,7 +128,6 @@ class FusedAdam(torch.optim.Optimizer):
                     group["lr"], beta1, beta2, group["eps"], group["weight_decay"],
                     True,
                 )
-                self._zeroes_grads_in_kernel = True
             elif use_hip(params[0]) and all(
                 # the dev kernel walks p flat and contiguous()-copies g: both
                 # must be standard-contiguous or the orders diverge
@@ -133,6 +137,7 @@ class FusedAdam(torch.optim.Optimizer):
                 if "step_t" not in st0 or st0["step_t"].device != params[0].device:
                     st0["step_t"] = torch.zeros(1, dtype=torch.float32, device=params[0].device)
                     st0["step_t"].fill_(float(step - 1))
+                all_groups_zeroed_in_kernel = False
                 require_ext().adam_step_dev(
                     params,
                     grads,
@@ -146,6 +151,7 @@ class FusedAdam(torch.optim.Optimizer):
                     group["weight_decay"],
                 )
             else:
+                all_groups_zeroed_in_kernel = False
                 for p, g, m, v in zip(params, grads, exp_avgs, exp_avg_sqs):
                     gf = g.float()
                     if group["weight_decay"] != 0:
@@ -155,6 +161,7 @@ class FusedAdam(torch.optim.Optimizer):
                     denom = (v / bc2).sqrt_().add_(group["eps"])
                     upd = (m / bc1) / denom
                     p.add_((-group["lr"] * upd).to(p.dtype))
+        self._zeroes_grads_in_kernel = all_groups_zeroed_in_kernel
         return loss
 
 

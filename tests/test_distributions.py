@@ -91,3 +91,19 @@ def test_symlog_mse_distributions():
     assert torch.allclose(d.log_prob(x), torch.zeros(4), atol=1e-5)  # perfect prediction
     m = MSEDistribution(x, dims=1)
     assert torch.allclose(m.log_prob(x), torch.zeros(4), atol=1e-6)
+
+
+def test_truncated_normal_entropy_numeric():
+    # exact truncated entropy vs numeric integration of -p log p over [low, high]
+    import numpy as np
+    from scipy import stats
+
+    loc = torch.tensor([0.0, 0.5, -0.8])
+    scale = torch.tensor([1.0, 0.3, 2.5])
+    d = TruncatedNormal(loc, scale)
+    ent = d.entropy()
+    for i in range(3):
+        a = (-1.0 - loc[i].item()) / scale[i].item()
+        b = (1.0 - loc[i].item()) / scale[i].item()
+        ref = stats.truncnorm.entropy(a, b, loc=loc[i].item(), scale=scale[i].item())
+        assert abs(ent[i].item() - float(ref)) < 1e-4, (i, ent[i].item(), ref)

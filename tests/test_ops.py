@@ -256,3 +256,35 @@ def test_rmsprop_tf_semantics():
         buf = 0.5 * buf + g / (v2 + eps).sqrt()
         expect2 = expect2 - lr * buf
     assert torch.allclose(p2, expect2, atol=1e-6)
+
+
+def test_fused_adam_zero_grad_unlatches_on_fallback():
+    """The in-kernel grad-zeroing flag must be re-derived every step: if a step
+    falls off the multi-tensor path (here: the CPU eager path), zero_grad must
+    go back to actually zeroing grads."""
+    from sheeprl_amd.optim import FusedAdam
+
+    p = torch.nn.Parameter(torch.randn(8))
+    opt = FusedAdam([p], lr=1e-2)
+    # simulate a previous step having taken the in-kernel-zeroing path
+    opt._zeroes_grads_in_kernel = True
+    p.grad = torch.ones(8)
+    opt.step()  # eager CPU path -> must clear the latch
+    assert opt._zeroes_grads_in_kernel is False
+    opt.zero_grad(set_to_none=False)
+    assert p.grad is not None and p.grad.abs().sum().item() == 0.0
+
+
+def test_ppo_value_loss_clipped_half_factor():
+    """Clipped value loss carries the reference's 0.5 factor
+    (sheeprl/algos/ppo/loss.py:61)."""
+    from sheeprl_amd.algos.ppo.loss import value_loss
+
+    torch.manual_seed(0)
+    nv, ov, ret = torch.randn(32), torch.randn(32), torch.randn(32)
+    v = value_loss(nv, ov, ret, clip_coef=0.2, clip_vloss=True)
+    unc = (nv - ret) ** 2
+    cl = (ov + (nv - ov).clamp(-0.2, 0.2) - ret) ** 2
+    assert torch.allclose(v, 0.5 * torch.max(unc, cl).mean())
+    # unclipped branch stays plain MSE
+    assert torch.allclose(value_loss(nv, ov, ret, 0.2, False), torch.nn.functional.mse_loss(nv, ret))
