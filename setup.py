@@ -20,9 +20,12 @@ try:
     from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
     csrc = Path(__file__).parent / "sheeprl_amd" / "ops" / "csrc"
-    # exclude torch-hipify outputs (*_hip.hip) — generated at build time
+    # exclude torch-hipify outputs: X_hip.hip generated from an existing X.hip
+    def _is_hipify_output(p: Path) -> bool:
+        return p.name.endswith("_hip.hip") and (p.parent / (p.name[: -len("_hip.hip")] + ".hip")).exists()
+
     sources = sorted(
-        str(p) for p in csrc.glob("*.hip") if not p.name.endswith("_hip.hip")
+        str(p) for p in csrc.glob("*.hip") if not _is_hipify_output(p)
     ) + sorted(str(p) for p in csrc.glob("*.cpp"))
     if sources:
         ext_modules.append(

@@ -2783,6 +2783,844 @@ void pk_scan_fwd(const torch::Tensor& actions, const torch::Tensor& f_all, const
   hipLaunchKernelGGL(pk_scan_fwd_kernel, dim3(nwg), dim3(256), 16 * 64 * sizeof(float), stream.stream(), P);
 }
 
+// ---------------------------------------------------------------------------
+// scan2: multi-workgroup fused phase kernels for the RSSM scan (round 2).
+//
+// The round-1 scan runs ~9 launches per step (4 hipblaslt GEMMs at the
+// ~5.8 us M=16 latency floor + LN/gate/sample epilogue kernels).  These
+// kernels fuse each phase into ONE launch: the GEMM runs on the g16 MFMA
+// skeleton with the A-operand staged in LDS, and the epilogue (LayerNorm,
+// Hafner gates, categorical-ST, episode resets) runs in the same launch.
+// LayerNorm statistics that span the N-split workgroups are exchanged with
+// the in-launch last-arriver pattern (atomicAdd partials -> ticket ->
+// relaxed poll -> ONE agent acquire -> plain reads) per the CDNA4 guide;
+// backward kernels avoid any sync by recomputing the small row-local
+// gradients per workgroup (the "recompute beats cross-WG traffic" rule for
+// B<=16 rows).  Grids are <= H/16 <= 256 workgroups, always co-resident,
+// so the bounded spin cannot deadlock.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void sc2_flush_and_arrive(int* ticket) {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (threadIdx.x == 0) __hip_atomic_fetch_add(ticket, 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__device__ __forceinline__ void sc2_wait(int* ticket, int nwg) {
+  if (threadIdx.x == 0) {
+    int spins = 0;
+    // bounded spin: never hang the box (a lost ticket produces wrong numbers
+    // that the numerics tests catch, not a dead GPU)
+    while (__hip_atomic_load(ticket, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) < nwg &&
+           spins < (1 << 27)) {
+      __builtin_amdgcn_s_sleep(2);
+      ++spins;
+    }
+  }
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  __syncthreads();
+}
+
+// cooperative vectorized copy of a row-strided bf16 [B, C] into LDS [16][KP]
+// (zero-padded rows >= B / cols >= C); requires C % 8 == 0 and ss % 8 == 0.
+__device__ __forceinline__ void sc2_stage_vec(const __hip_bfloat16* src, long ss, int B, int C,
+                                              __hip_bfloat16* dst, int KP) {
+  const int CV = C >> 3, KPV = KP >> 3;
+  uint4* d4 = (uint4*)dst;
+  for (int i = threadIdx.x; i < 16 * KPV; i += blockDim.x) {
+    const int m = i / KPV, c = i - m * KPV;
+    uint4 v = {0u, 0u, 0u, 0u};
+    if (m < B && c < CV) v = *(const uint4*)(src + (long)m * ss + ((long)c << 3));
+    d4[i] = v;
+  }
+}
+
+// one 16x16 C tile of  A[16, K] @ W[N, K]^T  with A staged in LDS ([16][KP],
+// KP = K rounded up to 32, zero-padded) and W streamed; wrow clamped so
+// ragged-N grids stay in bounds (garbage cols are never written).
+__device__ __forceinline__ f32x4 sc2_gemm_tile(const __hip_bfloat16* ldsA, int KP,
+                                               const __hip_bfloat16* W, long ws_, int K, int N,
+                                               int ncol0) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow = min(ncol0 + arow, N - 1);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int KU = K & ~31;
+  const int KU8 = KU & ~255;
+  for (int k0 = 0; k0 < KU8; k0 += 256) {
+    bf16x8 af[8], bf[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int k = k0 + u * 32 + kgrp * 8;
+      af[u] = *(const bf16x8*)(ldsA + arow * KP + k);
+      bf[u] = *(const bf16x8*)(W + (long)wrow * ws_ + k);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf[u], acc, 0, 0, 0);
+  }
+#pragma unroll 4
+  for (int k0 = KU8; k0 < KU; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = *(const bf16x8*)(ldsA + arow * KP + k);
+    bf16x8 b = *(const bf16x8*)(W + (long)wrow * ws_ + k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  if (K & 31) {
+    const int k = KU + kgrp * 8;
+    bf16x8 a = *(const bf16x8*)(ldsA + arow * KP + k);  // zero-padded: safe
+    bf16x8 b;
+    const __hip_bfloat16* p = W + (long)wrow * ws_ + k;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) b[e] = (k + e < K) ? (__bf16)__bfloat162float(p[e]) : (__bf16)0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  return acc;
+}
+
+// per-wave reduce of a per-(row, col) value over the wave's 16 cols; lanes
+// with arow == 0 then hold the per-row result (rows kgrp*4 + r).
+__device__ __forceinline__ float sc2_colsum(float v) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// scan2 forward phase 1/3: C = A @ W^T, then rowwise LN (+SiLU) across the
+// whole grid via the ticket exchange.  RESETS=true additionally assembles the
+// episode-reset-masked GEMM input x = [z', a'] (phase 1) in LDS and persists
+// x / h' for the backward.
+// ---------------------------------------------------------------------------
+template <bool RESETS, bool T0>
+__global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
+    const __hip_bfloat16* __restrict__ a_in, long as_,      // staged A (phase 3) or z_prev (phase 1)
+    const __hip_bfloat16* __restrict__ iz,                  // phase 1 only
+    const __hip_bfloat16* __restrict__ h_prev,
+    const __hip_bfloat16* __restrict__ ih,
+    const __hip_bfloat16* __restrict__ act,                 // [B, A]
+    const __hip_bfloat16* __restrict__ f,                   // [B]
+    const __hip_bfloat16* __restrict__ W, const __hip_bfloat16* __restrict__ lnw,
+    const __hip_bfloat16* __restrict__ lnb,
+    __hip_bfloat16* __restrict__ x_out, long xs,            // phase 1: x_s[t]
+    __hip_bfloat16* __restrict__ hu_out, long hus,          // phase 1: hu_s[t] ([:, :H] = h', [:, H:] = u); phase 3: p out
+    __hip_bfloat16* __restrict__ g_out, long gs,            // pre-LN GEMM result
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    float* __restrict__ ws2, int* __restrict__ ticket,
+    int B, int SK, int A, int H, int N, int K, float eps, int hu_off) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int KP = (K + 31) & ~31;
+  __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;                  // [16][KP]
+  float* lds_sum = (float*)(smem + ((16 * KP * 2 + 15) & ~15));  // [32]
+  if (threadIdx.x < 32) lds_sum[threadIdx.x] = 0.f;
+  if (RESETS) {
+    // x = [(1-f) z_prev + f iz, (1-f) a]; rows >= B and cols >= K zero
+    for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
+      const int m = i / KP, c = i - m * KP;
+      float v = 0.f;
+      if (m < B && c < K) {
+        const float fb = __bfloat162float(f[m]);
+        if (c < SK) {
+          float zp = T0 ? 0.f : __bfloat162float(a_in[(long)m * SK + c]);
+          v = (1.f - fb) * zp + fb * __bfloat162float(iz[(long)m * SK + c]);
+        } else {
+          v = (1.f - fb) * __bfloat162float(act[(long)m * A + (c - SK)]);
+        }
+      }
+      ldsA[i] = __float2bfloat16(v);
+    }
+    __syncthreads();
+    if (blockIdx.x == 0) {
+      // persist x for backward + h' = (1-f) h_prev + f ih into hu[:, :H]
+      for (int i = threadIdx.x; i < B * K; i += blockDim.x) {
+        const int m = i / K, c = i - m * K;
+        x_out[(long)m * xs + c] = ldsA[m * KP + c];
+      }
+      for (int i = threadIdx.x; i < B * H; i += blockDim.x) {
+        const int m = i / H, j = i - m * H;
+        const float fb = __bfloat162float(f[m]);
+        float hp = T0 ? 0.f : __bfloat162float(h_prev[(long)m * H + j]);
+        st(hu_out, (long)m * hus + j, (1.f - fb) * hp + fb * __bfloat162float(ih[(long)m * H + j]));
+      }
+    }
+  } else {
+    if ((K & 7) == 0 && (as_ & 7) == 0) {
+      sc2_stage_vec(a_in, as_, B, K, ldsA, KP);
+    } else {
+      for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
+        const int m = i / KP, c = i - m * KP;
+        ldsA[i] = (m < B && c < K) ? a_in[(long)m * as_ + c] : (__hip_bfloat16)0.f;
+      }
+    }
+    __syncthreads();
+  }
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int ncol0 = (blockIdx.x * 4 + wv) * 16;
+  f32x4 acc = sc2_gemm_tile(ldsA, KP, W, K, K, N, ncol0);
+  // write pre-LN G and accumulate row partials
+  float s = 0.f, s2 = 0.f;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    if (m < B && ncol0 + arow < N) g_out[(long)m * gs + ncol0 + arow] = __float2bfloat16(acc[r]);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    float v = (ncol0 + arow < N) ? acc[r] : 0.f;
+    float cs = sc2_colsum(v);
+    float cs2 = sc2_colsum(v * v);
+    if (arow == 0) {
+      atomicAdd(&lds_sum[kgrp * 4 + r], cs);
+      atomicAdd(&lds_sum[16 + kgrp * 4 + r], cs2);
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x < 32) atomicAdd(&ws2[threadIdx.x], lds_sum[threadIdx.x]);
+  sc2_flush_and_arrive(ticket);
+  sc2_wait(ticket, gridDim.x);
+  __shared__ float mr_[2][16];
+  if (threadIdx.x < 16) {
+    float mean = ws2[threadIdx.x] / N;
+    float var = ws2[16 + threadIdx.x] / N - mean * mean;
+    mr_[0][threadIdx.x] = mean;
+    mr_[1][threadIdx.x] = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (blockIdx.x == 0 && threadIdx.x < B) {
+      mean_out[threadIdx.x] = mean;
+      rstd_out[threadIdx.x] = mr_[1][threadIdx.x];
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const int j = ncol0 + arow;
+    if (m < B && j < N) {
+      float z = (acc[r] - mr_[0][m]) * mr_[1][m] * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
+      z = z / (1.f + expf(-z));
+      st(hu_out, (long)m * hus + hu_off + j, z);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scan2 forward phase 2: y = hu @ W2^T (three gate stripes per h column),
+// LN over 3H via the ticket exchange, then the Hafner GRU gates.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) scan2_gru_kernel(
+    const __hip_bfloat16* __restrict__ hu, long hus,  // [B, H+D] (h' | u)
+    const __hip_bfloat16* __restrict__ W2,            // [3H, H+D]
+    const __hip_bfloat16* __restrict__ lnw, const __hip_bfloat16* __restrict__ lnb,
+    __hip_bfloat16* __restrict__ y_out, long ys2,     // pre-LN y
+    __hip_bfloat16* __restrict__ h_out,               // h_seq[t] [B, H] contiguous
+    __hip_bfloat16* __restrict__ h_out2, long h2s,    // r_s[t][:, :H]
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    float* __restrict__ ws2, int* __restrict__ ticket, int B, int H, int D, float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int K = H + D;
+  const int KP = (K + 31) & ~31;
+  __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;
+  float* lds_sum = (float*)(smem + ((16 * KP * 2 + 15) & ~15));
+  if (threadIdx.x < 32) lds_sum[threadIdx.x] = 0.f;
+  if ((K & 7) == 0 && (hus & 7) == 0) {
+    sc2_stage_vec(hu, hus, B, K, ldsA, KP);
+  } else {
+    for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
+      const int m = i / KP, c = i - m * KP;
+      ldsA[i] = (m < B && c < K) ? hu[(long)m * hus + c] : (__hip_bfloat16)0.f;
+    }
+  }
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int c0 = (blockIdx.x * 4 + wv) * 16;  // h-column tile
+  f32x4 accg[3];
+  float s = 0.f, s2 = 0.f;
+#pragma unroll
+  for (int g = 0; g < 3; ++g) {
+    accg[g] = sc2_gemm_tile(ldsA, KP, W2, K, K, 3 * H, g * H + c0);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = kgrp * 4 + r;
+      if (m < B) y_out[(long)m * ys2 + g * H + c0 + arow] = __float2bfloat16(accg[g][r]);
+      float cs = sc2_colsum(accg[g][r]);
+      float cs2 = sc2_colsum(accg[g][r] * accg[g][r]);
+      if (arow == 0) {
+        atomicAdd(&lds_sum[kgrp * 4 + r], cs);
+        atomicAdd(&lds_sum[16 + kgrp * 4 + r], cs2);
+      }
+    }
+  }
+  __syncthreads();
+  if (threadIdx.x < 32) atomicAdd(&ws2[threadIdx.x], lds_sum[threadIdx.x]);
+  sc2_flush_and_arrive(ticket);
+  sc2_wait(ticket, gridDim.x);
+  const int DD = 3 * H;
+  __shared__ float mr_[2][16];
+  if (threadIdx.x < 16) {
+    float mean = ws2[threadIdx.x] / DD;
+    float var = ws2[16 + threadIdx.x] / DD - mean * mean;
+    mr_[0][threadIdx.x] = mean;
+    mr_[1][threadIdx.x] = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (blockIdx.x == 0 && threadIdx.x < B) {
+      mean_out[threadIdx.x] = mean;
+      rstd_out[threadIdx.x] = mr_[1][threadIdx.x];
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const int j = c0 + arow;
+    if (m < B) {
+      const float mean = mr_[0][m], rstd = mr_[1][m];
+      float zr = (accg[0][r] - mean) * rstd * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
+      float zc = (accg[1][r] - mean) * rstd * __bfloat162float(lnw[H + j]) + __bfloat162float(lnb[H + j]);
+      float zu = (accg[2][r] - mean) * rstd * __bfloat162float(lnw[2 * H + j]) + __bfloat162float(lnb[2 * H + j]);
+      float rg = 1.f / (1.f + expf(-zr));
+      float cg = tanhf(rg * zc);
+      float ug = 1.f / (1.f + expf(-(zu - 1.f)));
+      float hp = __bfloat162float(ldsA[m * KP + j]);  // h' lives in hu[:, :H]
+      float hv = ug * cg + (1.f - ug) * hp;
+      h_out[(long)m * H + j] = __float2bfloat16(hv);
+      h_out2[(long)m * h2s + j] = __float2bfloat16(hv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scan2 forward phase 4: raw = p @ W4^T + b4, then the unimix categorical-ST
+// head per KD-group (no cross-WG traffic: each WG owns whole groups).
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) scan2_catst_kernel(
+    const __hip_bfloat16* __restrict__ p_in, long ps,  // [B, P]
+    const __hip_bfloat16* __restrict__ W4,             // [SK, P]
+    const __hip_bfloat16* __restrict__ b4, const float* __restrict__ urand,
+    float* __restrict__ m_out, __hip_bfloat16* __restrict__ z_out, float* __restrict__ s_out,
+    int B, int P, int SK, int KD, float unimix) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int KP = (P + 31) & ~31;
+  __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;
+  float* raw = (float*)(smem + ((16 * KP * 2 + 15) & ~15));  // [16][64]
+  if ((P & 7) == 0 && (ps & 7) == 0) {
+    sc2_stage_vec(p_in, ps, B, P, ldsA, KP);
+  } else {
+    for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
+      const int m = i / KP, c = i - m * KP;
+      ldsA[i] = (m < B && c < P) ? p_in[(long)m * ps + c] : (__hip_bfloat16)0.f;
+    }
+  }
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int ncol0 = (blockIdx.x * 4 + wv) * 16;
+  f32x4 acc = sc2_gemm_tile(ldsA, KP, W4, P, P, SK, ncol0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    raw[m * 64 + (ncol0 & 63) + arow] = acc[r] + __bfloat162float(b4[ncol0 + arow]);
+  }
+  __syncthreads();
+  const int gpw = 64 / KD;  // groups per WG per row
+  for (int task = threadIdx.x; task < 16 * gpw; task += blockDim.x) {
+    const int m = task / gpw;
+    if (m >= B) continue;
+    const int gl = task - m * gpw;
+    const float* row = raw + m * 64 + gl * KD;
+    const long gcol = (long)blockIdx.x * 64 + gl * KD;  // global col of the group
+    float lmax = -1e30f;
+    for (int j = 0; j < KD; ++j) lmax = fmaxf(lmax, row[j]);
+    float lsum = 0.f;
+    for (int j = 0; j < KD; ++j) lsum += expf(row[j] - lmax);
+    const float inv = 1.f / lsum;
+    const float* ur = urand + (long)m * SK + gcol;
+    float* mro = m_out + (long)m * SK + gcol;
+    float* sro = s_out + (long)m * SK + gcol;
+    __hip_bfloat16* zro = z_out + (long)m * SK + gcol;
+    float best = -1e30f;
+    int best_j = 0;
+    for (int j = 0; j < KD; ++j) {
+      float sv = expf(row[j] - lmax) * inv;
+      float pv = (1.f - unimix) * sv + unimix / KD;
+      float mv = logf(pv);
+      sro[j] = sv;
+      mro[j] = mv;
+      float tt = fmaxf(-logf(fmaxf(ur[j], 1e-20f)), 1e-20f);
+      float score = mv - logf(tt);
+      if (score > best) {
+        best = score;
+        best_j = j;
+      }
+    }
+    for (int j = 0; j < KD; ++j) zro[j] = (__hip_bfloat16)(j == best_j ? 1.f : 0.f);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scan2 backward phase 4 (reverse order: runs first): categorical-ST backward
+// recomputed per WG into LDS, then the gp = graw @ W4t GEMM tile.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) scan2_b4_kernel(
+    const float* __restrict__ gm,                 // [B, SK] fp32
+    const __hip_bfloat16* __restrict__ gon,       // g_z_seq[t]
+    const __hip_bfloat16* __restrict__ gon2,      // carry or null
+    const float* __restrict__ s_saved,            // [B, SK]
+    const __hip_bfloat16* __restrict__ W4t,       // [P, SK] (pre-transposed)
+    __hip_bfloat16* __restrict__ graw_out, long gws,  // graw_s[t] (WG 0 writes)
+    __hip_bfloat16* __restrict__ gp_out,          // [B, P] scratch
+    int B, int SK, int P, int KD, float unimix) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int KP = SK;  // SK % 64 == 0
+  __hip_bfloat16* graw = (__hip_bfloat16*)smem;  // [16][SK]
+  const int lane = threadIdx.x & 63;
+  const int groups = SK / KD;
+  const int gpercall = (int)blockDim.x / KD;
+  for (int task = threadIdx.x / KD; task < 16 * groups; task += gpercall) {
+    const int m = task / groups;
+    const int g = task - m * groups;
+    const int j = (threadIdx.x % KD);
+    const long idx = (long)m * SK + g * KD + j;
+    float t = 0.f, sj = 0.f;
+    if (m < B) {
+      sj = s_saved[idx];
+      float pj = (1.f - unimix) * sj + unimix / KD;
+      t = gm[idx] / pj + __bfloat162float(gon[idx]) + (gon2 ? __bfloat162float(gon2[idx]) : 0.f);
+    }
+    float acc = t * sj;
+#pragma unroll
+    for (int off = 1; off < 64; off <<= 1) {
+      if (off < KD) acc += __shfl_xor(acc, off, 64);
+    }
+    graw[m * KP + g * KD + j] = __float2bfloat16(m < B ? (1.f - unimix) * sj * (t - acc) : 0.f);
+  }
+  __syncthreads();
+  if (blockIdx.x == 0) {
+    for (int i = threadIdx.x; i < B * SK; i += blockDim.x) {
+      const int m = i / SK, c = i - m * SK;
+      graw_out[(long)m * gws + c] = graw[m * KP + c];
+    }
+  }
+  const int wv = threadIdx.x >> 6;
+  const int ncol0 = (blockIdx.x * 4 + wv) * 16;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  f32x4 acc = sc2_gemm_tile(graw, KP, W4t, SK, SK, P, ncol0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    if (m < B && ncol0 + arow < P) gp_out[(long)m * P + ncol0 + arow] = __float2bfloat16(acc[r]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scan2 backward LN(+SiLU) phase (phases 3 and 1): recompute the row-local
+// LayerNorm backward into LDS per WG (no sync), then the next GEMM tile.
+// Stripe WGs additionally persist gg and flush the LN affine grads.
+// RESETS=false -> phase 3 (gr output, strided); RESETS=true -> phase 1
+// (gx output with the episode-reset backward fused: carries + action grad +
+// init-state accumulators).
+// ---------------------------------------------------------------------------
+template <bool RESETS>
+__global__ void __launch_bounds__(256) scan2_blnsilu_kernel(
+    const __hip_bfloat16* __restrict__ gy_in, long gys,  // grad wrt activated out [B, P]
+    const __hip_bfloat16* __restrict__ g_in, long gis,   // saved pre-LN GEMM result
+    const __hip_bfloat16* __restrict__ lnw, const __hip_bfloat16* __restrict__ lnb,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const __hip_bfloat16* __restrict__ Wt,               // [N, P] pre-transposed next-GEMM weight
+    const __hip_bfloat16* __restrict__ f,                // [B] (RESETS)
+    __hip_bfloat16* __restrict__ gg_out, long ggs,       // gg stripe persist
+    float* __restrict__ glnw, float* __restrict__ glnb,  // fp32 accumulators
+    __hip_bfloat16* __restrict__ out, long outs,         // gr_s[t] (phase 3) / unused (RESETS)
+    __hip_bfloat16* __restrict__ gz_carry, float* __restrict__ giz_acc,
+    __hip_bfloat16* __restrict__ ga_out,                 // [B, A] (RESETS)
+    int B, int P, int N, int SK, int A) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int KP = P;  // P % 64 == 0
+  __hip_bfloat16* gg = (__hip_bfloat16*)smem;                      // [16][P]
+  float* s12 = (float*)(smem + ((16 * KP * 2 + 15) & ~15));        // [2][16]
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  // pass A: per-row sums (wave wv owns rows 4wv..4wv+3) + gz into LDS
+  for (int mr2 = 0; mr2 < 4; ++mr2) {
+    const int m = wv * 4 + mr2;
+    float s1 = 0.f, s2 = 0.f;
+    if (m < B) {
+      const float mn = mean[m], rs = rstd[m];
+      for (int j = lane; j < P; j += 64) {
+        float xh = (__bfloat162float(g_in[(long)m * gis + j]) - mn) * rs;
+        float w = __bfloat162float(lnw[j]);
+        float z = xh * w + __bfloat162float(lnb[j]);
+        float sig = 1.f / (1.f + expf(-z));
+        float gz = __bfloat162float(gy_in[(long)m * gys + j]) * sig * (1.f + z * (1.f - sig));
+        gg[m * KP + j] = __float2bfloat16(gz);  // temporarily gz
+        float gxh = gz * w;
+        s1 += gxh;
+        s2 += gxh * xh;
+      }
+      s1 = wave_sum(s1);
+      s2 = wave_sum(s2);
+    } else {
+      for (int j = lane; j < P; j += 64) gg[m * KP + j] = (__hip_bfloat16)0.f;
+    }
+    if (lane == 0) {
+      s12[m] = s1 / P;
+      s12[16 + m] = s2 / P;
+    }
+  }
+  __syncthreads();
+  // LN affine grads: stripe WGs flush one atomicAdd per column
+  for (int stripe = blockIdx.x; stripe < P / 64; stripe += gridDim.x) {
+    const int j = stripe * 64 + threadIdx.x % 64;
+    if (threadIdx.x < 64) {
+      float sw = 0.f, sb = 0.f;
+      for (int m = 0; m < B; ++m) {
+        float gz = __bfloat162float(gg[m * KP + j]);
+        float xh = (__bfloat162float(g_in[(long)m * gis + j]) - mean[m]) * rstd[m];
+        sw += gz * xh;
+        sb += gz;
+      }
+      atomicAdd(&glnw[j], sw);
+      atomicAdd(&glnb[j], sb);
+    }
+  }
+  __syncthreads();
+  // pass B: finalize gg in place
+  for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
+    const int m = i / KP, j = i - m * KP;
+    if (m < B) {
+      float gz = __bfloat162float(gg[i]);
+      float xh = (__bfloat162float(g_in[(long)m * gis + j]) - mean[m]) * rstd[m];
+      float gv = (gz * __bfloat162float(lnw[j]) - s12[m] - xh * s12[16 + m]) * rstd[m];
+      gg[i] = __float2bfloat16(gv);
+    }
+  }
+  __syncthreads();
+  // persist gg stripes (for the batched weight-grad GEMMs)
+  for (int stripe = blockIdx.x; stripe < P / 64; stripe += gridDim.x) {
+    for (int i = threadIdx.x; i < B * 64; i += blockDim.x) {
+      const int m = i / 64, j = stripe * 64 + (i & 63);
+      gg_out[(long)m * ggs + j] = gg[m * KP + j];
+    }
+  }
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int ncol0 = (blockIdx.x * 4 + wv) * 16;
+  f32x4 acc = sc2_gemm_tile(gg, KP, Wt, P, P, N, ncol0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const int c = ncol0 + arow;
+    if (m < B && c < N) {
+      if (RESETS) {
+        const float fb = __bfloat162float(f[m]);
+        const float gv = acc[r];
+        if (c < SK) {
+          gz_carry[(long)m * SK + c] = __float2bfloat16((1.f - fb) * gv);
+          giz_acc[(long)m * SK + c] += fb * gv;
+        } else if (c < SK + A) {
+          ga_out[(long)m * A + (c - SK)] = __float2bfloat16((1.f - fb) * gv);
+        }
+      } else {
+        out[(long)m * outs + c] = __float2bfloat16(acc[r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scan2 backward phase 2: full Hafner-gate + LN backward recomputed per WG
+// into LDS, then the ghu = gy @ W2t GEMM tile with the reset backward for the
+// h columns fused into the epilogue.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256) scan2_bgru_kernel(
+    const __hip_bfloat16* __restrict__ gh,        // g_h_seq[t] [B, H]
+    const __hip_bfloat16* __restrict__ gh2,       // carry or null
+    const __hip_bfloat16* __restrict__ gh3, long gh3s,  // gr_s[t][:, :H]
+    const __hip_bfloat16* __restrict__ y_in, long ys2,  // saved pre-LN y [B, 3H]
+    const __hip_bfloat16* __restrict__ hu, long hus,    // h' in [:, :H]
+    const __hip_bfloat16* __restrict__ lnw, const __hip_bfloat16* __restrict__ lnb,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    const __hip_bfloat16* __restrict__ W2t,       // [H+D, 3H]
+    const __hip_bfloat16* __restrict__ f,         // [B]
+    __hip_bfloat16* __restrict__ gy_out, long gys,      // gy_s[t]
+    float* __restrict__ glnw, float* __restrict__ glnb,
+    __hip_bfloat16* __restrict__ gh_carry,        // [B, H]
+    float* __restrict__ gih_acc,                  // [B, H] fp32
+    __hip_bfloat16* __restrict__ ghu_out,         // [B, H+D] scratch (u part consumed by b1)
+    int B, int H, int D) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int DD = 3 * H;
+  __hip_bfloat16* gybuf = (__hip_bfloat16*)smem;                     // [16][3H]
+  __hip_bfloat16* ghp = (__hip_bfloat16*)(smem + 16 * DD * 2);       // [16][H]
+  float* s12 = (float*)(smem + ((16 * DD * 2 + 16 * H * 2 + 15) & ~15));  // [2][16]
+  const int lane = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  for (int mr2 = 0; mr2 < 4; ++mr2) {
+    const int m = wv * 4 + mr2;
+    float s1 = 0.f, s2 = 0.f;
+    if (m < B) {
+      const float mn = mean[m], rs = rstd[m];
+      for (int j = lane; j < H; j += 64) {
+        float xh_r = (__bfloat162float(y_in[(long)m * ys2 + j]) - mn) * rs;
+        float xh_c = (__bfloat162float(y_in[(long)m * ys2 + H + j]) - mn) * rs;
+        float xh_u = (__bfloat162float(y_in[(long)m * ys2 + 2 * H + j]) - mn) * rs;
+        float zr = xh_r * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
+        float zc = xh_c * __bfloat162float(lnw[H + j]) + __bfloat162float(lnb[H + j]);
+        float zu = xh_u * __bfloat162float(lnw[2 * H + j]) + __bfloat162float(lnb[2 * H + j]);
+        float r = 1.f / (1.f + expf(-zr));
+        float rc = r * zc;
+        float c = tanhf(rc);
+        float u = 1.f / (1.f + expf(-(zu - 1.f)));
+        float g = __bfloat162float(gh[(long)m * H + j]) + (gh2 ? __bfloat162float(gh2[(long)m * H + j]) : 0.f) +
+                  __bfloat162float(gh3[(long)m * gh3s + j]);
+        float hp = __bfloat162float(hu[(long)m * hus + j]);
+        float gu = g * (c - hp);
+        float gc = g * u;
+        float gzu = gu * u * (1.f - u);
+        float grc = gc * (1.f - c * c);
+        float gzc = grc * r;
+        float gr = grc * zc;
+        float gzr = gr * r * (1.f - r);
+        ghp[m * H + j] = __float2bfloat16(g * (1.f - u));
+        gybuf[m * DD + j] = __float2bfloat16(gzr);
+        gybuf[m * DD + H + j] = __float2bfloat16(gzc);
+        gybuf[m * DD + 2 * H + j] = __float2bfloat16(gzu);
+        float gxh_r = gzr * __bfloat162float(lnw[j]);
+        float gxh_c = gzc * __bfloat162float(lnw[H + j]);
+        float gxh_u = gzu * __bfloat162float(lnw[2 * H + j]);
+        s1 += gxh_r + gxh_c + gxh_u;
+        s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
+      }
+      s1 = wave_sum(s1);
+      s2 = wave_sum(s2);
+    } else {
+      for (int j = lane; j < DD; j += 64) gybuf[m * DD + j] = (__hip_bfloat16)0.f;
+    }
+    if (lane == 0) {
+      s12[m] = s1 / DD;
+      s12[16 + m] = s2 / DD;
+    }
+  }
+  __syncthreads();
+  // LN affine grads per column stripe
+  for (int stripe = blockIdx.x; stripe < DD / 64; stripe += gridDim.x) {
+    const int j = stripe * 64 + threadIdx.x % 64;
+    if (threadIdx.x < 64) {
+      float sw = 0.f, sb = 0.f;
+      for (int m = 0; m < B; ++m) {
+        float gz = __bfloat162float(gybuf[m * DD + j]);
+        float xh = (__bfloat162float(y_in[(long)m * ys2 + j]) - mean[m]) * rstd[m];
+        sw += gz * xh;
+        sb += gz;
+      }
+      atomicAdd(&glnw[j], sw);
+      atomicAdd(&glnb[j], sb);
+    }
+  }
+  __syncthreads();
+  // finalize gy in place
+  for (int i = threadIdx.x; i < 16 * DD; i += blockDim.x) {
+    const int m = i / DD, j = i - m * DD;
+    if (m < B) {
+      float gz = __bfloat162float(gybuf[i]);
+      float xh = (__bfloat162float(y_in[(long)m * ys2 + j]) - mean[m]) * rstd[m];
+      gybuf[i] = __float2bfloat16((gz * __bfloat162float(lnw[j]) - s12[m] - xh * s12[16 + m]) * rstd[m]);
+    }
+  }
+  __syncthreads();
+  for (int stripe = blockIdx.x; stripe < DD / 64; stripe += gridDim.x) {
+    for (int i = threadIdx.x; i < B * 64; i += blockDim.x) {
+      const int m = i / 64, j = stripe * 64 + (i & 63);
+      gy_out[(long)m * gys + j] = gybuf[m * DD + j];
+    }
+  }
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int ncol0 = (blockIdx.x * 4 + wv) * 16;
+  f32x4 acc = sc2_gemm_tile(gybuf, DD, W2t, DD, DD, H + D, ncol0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const int c = ncol0 + arow;
+    if (m < B && c < H + D) {
+      if (c < H) {
+        const float fb = __bfloat162float(f[m]);
+        float g2 = acc[r] + __bfloat162float(ghp[m * H + c]);
+        gh_carry[(long)m * H + c] = __float2bfloat16((1.f - fb) * g2);
+        gih_acc[(long)m * H + c] += fb * g2;
+      } else {
+        ghu_out[(long)m * (H + D) + c] = __float2bfloat16(acc[r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static inline const __hip_bfloat16* sc2_bp(const torch::Tensor& t) {
+  return (const __hip_bfloat16*)t.data_ptr();
+}
+static inline __hip_bfloat16* sc2_bpm(torch::Tensor& t) { return (__hip_bfloat16*)t.data_ptr(); }
+
+void scan2_f1(const c10::optional<torch::Tensor>& z_prev, const torch::Tensor& iz,
+              const c10::optional<torch::Tensor>& h_prev, const torch::Tensor& ih,
+              const torch::Tensor& act, const torch::Tensor& f, const torch::Tensor& W1,
+              const torch::Tensor& lnw, const torch::Tensor& lnb, torch::Tensor x_out,
+              torch::Tensor hu_out, torch::Tensor g_out, torch::Tensor mean, torch::Tensor rstd,
+              torch::Tensor ws2, torch::Tensor ticket, double eps) {
+  const int B = (int)act.size(0), A = (int)act.size(1), SK = (int)iz.size(1), H = (int)ih.size(1);
+  const int N = (int)W1.size(0), K = (int)W1.size(1);
+  TORCH_CHECK(B <= 16 && N % 64 == 0 && K == SK + A && W1.is_contiguous());
+  const int KP = (K + 31) & ~31;
+  const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bool t0 = !z_prev.has_value();
+  const __hip_bfloat16* zp = t0 ? nullptr : sc2_bp(*z_prev);
+  const __hip_bfloat16* hp = h_prev.has_value() ? sc2_bp(*h_prev) : nullptr;
+#define SC2_F1_LAUNCH(T0V)                                                                            \
+  hipLaunchKernelGGL((scan2_lnsilu_kernel<true, T0V>), dim3(N / 64), dim3(256), shmem, stream.stream(), \
+                     zp, 0, sc2_bp(iz), hp, sc2_bp(ih), sc2_bp(act), sc2_bp(f), sc2_bp(W1),           \
+                     sc2_bp(lnw), sc2_bp(lnb), sc2_bpm(x_out), x_out.stride(0), sc2_bpm(hu_out),      \
+                     hu_out.stride(0), sc2_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(),       \
+                     rstd.data_ptr<float>(), ws2.data_ptr<float>(), ticket.data_ptr<int>(), B, SK, A, \
+                     H, N, K, (float)eps, H)
+  if (t0)
+    SC2_F1_LAUNCH(true);
+  else
+    SC2_F1_LAUNCH(false);
+#undef SC2_F1_LAUNCH
+}
+
+void scan2_f3(const torch::Tensor& a_in, const torch::Tensor& W3, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, torch::Tensor p_out, torch::Tensor g_out, torch::Tensor mean,
+              torch::Tensor rstd, torch::Tensor ws2, torch::Tensor ticket, double eps) {
+  const int B = (int)a_in.size(0);
+  const int N = (int)W3.size(0), K = (int)W3.size(1);
+  TORCH_CHECK(B <= 16 && N % 64 == 0 && a_in.size(1) == K && W3.is_contiguous());
+  const int KP = (K + 31) & ~31;
+  const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
+  TORCH_CHECK(shmem <= 160 * 1024, "scan2_f3: K too large for LDS");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((scan2_lnsilu_kernel<false, false>), dim3(N / 64), dim3(256), shmem, stream.stream(),
+                     sc2_bp(a_in), a_in.stride(0), nullptr, nullptr, nullptr, nullptr, nullptr,
+                     sc2_bp(W3), sc2_bp(lnw), sc2_bp(lnb), nullptr, 0, sc2_bpm(p_out), p_out.stride(0),
+                     sc2_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     ws2.data_ptr<float>(), ticket.data_ptr<int>(), B, 0, 0, 0, N, K, (float)eps, 0);
+}
+
+void scan2_f2(const torch::Tensor& hu, const torch::Tensor& W2, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, torch::Tensor y_out, torch::Tensor h_out, torch::Tensor h_out2,
+              torch::Tensor mean, torch::Tensor rstd, torch::Tensor ws2, torch::Tensor ticket,
+              double eps) {
+  const int B = (int)hu.size(0), K = (int)hu.size(1);
+  const int H = (int)h_out.size(1), D = K - H;
+  TORCH_CHECK(B <= 16 && H % 64 == 0 && W2.size(0) == 3 * H && W2.size(1) == K && W2.is_contiguous());
+  TORCH_CHECK(h_out.is_contiguous());
+  const int KP = (K + 31) & ~31;
+  const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(scan2_gru_kernel, dim3(H / 64), dim3(256), shmem, stream.stream(), sc2_bp(hu),
+                     hu.stride(0), sc2_bp(W2), sc2_bp(lnw), sc2_bp(lnb), sc2_bpm(y_out), y_out.stride(0),
+                     sc2_bpm(h_out), sc2_bpm(h_out2), h_out2.stride(0), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), ws2.data_ptr<float>(), ticket.data_ptr<int>(), B, H, D,
+                     (float)eps);
+}
+
+void scan2_f4(const torch::Tensor& p_in, const torch::Tensor& W4, const torch::Tensor& b4,
+              const torch::Tensor& urand, torch::Tensor m_out, torch::Tensor z_out, torch::Tensor s_out,
+              long KD, double unimix) {
+  const int B = (int)p_in.size(0), P = (int)p_in.size(1);
+  const int SK = (int)W4.size(0);
+  TORCH_CHECK(B <= 16 && SK % 64 == 0 && W4.size(1) == P && W4.is_contiguous());
+  TORCH_CHECK(KD <= 64 && 64 % KD == 0 && m_out.is_contiguous() && z_out.is_contiguous() && s_out.is_contiguous());
+  const int KP = (P + 31) & ~31;
+  const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 16 * 64 * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(scan2_catst_kernel, dim3(SK / 64), dim3(256), shmem, stream.stream(), sc2_bp(p_in),
+                     p_in.stride(0), sc2_bp(W4), sc2_bp(b4), urand.data_ptr<float>(),
+                     m_out.data_ptr<float>(), sc2_bpm(z_out), s_out.data_ptr<float>(), B, P, SK, (int)KD,
+                     (float)unimix);
+}
+
+void scan2_b4(const torch::Tensor& gm, const torch::Tensor& gon, const c10::optional<torch::Tensor>& gon2,
+              const torch::Tensor& s_saved, const torch::Tensor& W4t, torch::Tensor graw_out,
+              torch::Tensor gp_out, long KD, double unimix) {
+  const int B = (int)gon.size(0), SK = (int)gon.size(1);
+  const int P = (int)W4t.size(0);
+  TORCH_CHECK(B <= 16 && SK % 64 == 0 && P % 64 == 0 && W4t.size(1) == SK && W4t.is_contiguous());
+  TORCH_CHECK(KD <= 64 && (KD & (KD - 1)) == 0 && gp_out.is_contiguous());
+  const size_t shmem = 16 * SK * 2;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const __hip_bfloat16* g2 = gon2.has_value() ? sc2_bp(*gon2) : nullptr;
+  hipLaunchKernelGGL(scan2_b4_kernel, dim3(P / 64), dim3(256), shmem, stream.stream(),
+                     gm.data_ptr<float>(), sc2_bp(gon), g2, s_saved.data_ptr<float>(), sc2_bp(W4t),
+                     sc2_bpm(graw_out), graw_out.stride(0), sc2_bpm(gp_out), B, SK, P, (int)KD,
+                     (float)unimix);
+}
+
+void scan2_b3(const torch::Tensor& gy_in, const torch::Tensor& g_in, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, const torch::Tensor& mean, const torch::Tensor& rstd,
+              const torch::Tensor& Wt, torch::Tensor gg_out, torch::Tensor glnw, torch::Tensor glnb,
+              torch::Tensor out) {
+  const int B = (int)gy_in.size(0), P = (int)gy_in.size(1);
+  const int N = (int)Wt.size(0);
+  TORCH_CHECK(B <= 16 && P % 64 == 0 && Wt.size(1) == P && Wt.is_contiguous());
+  const size_t shmem = ((16 * P * 2 + 15) & ~15) + 32 * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((scan2_blnsilu_kernel<false>), dim3((N + 63) / 64), dim3(256), shmem, stream.stream(),
+                     sc2_bp(gy_in), gy_in.stride(0), sc2_bp(g_in), g_in.stride(0), sc2_bp(lnw),
+                     sc2_bp(lnb), mean.data_ptr<float>(), rstd.data_ptr<float>(), sc2_bp(Wt), nullptr,
+                     sc2_bpm(gg_out), gg_out.stride(0), glnw.data_ptr<float>(), glnb.data_ptr<float>(),
+                     sc2_bpm(out), out.stride(0), nullptr, nullptr, nullptr, B, P, N, 0, 0);
+}
+
+void scan2_b1(const torch::Tensor& gy_in, const torch::Tensor& g_in, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, const torch::Tensor& mean, const torch::Tensor& rstd,
+              const torch::Tensor& Wt, const torch::Tensor& f, torch::Tensor gg_out, torch::Tensor glnw,
+              torch::Tensor glnb, torch::Tensor gz_carry, torch::Tensor giz_acc, torch::Tensor ga_out) {
+  const int B = (int)gy_in.size(0), P = (int)gy_in.size(1);
+  const int N = (int)Wt.size(0);
+  const int SK = (int)gz_carry.size(1), A = (int)ga_out.size(1);
+  TORCH_CHECK(B <= 16 && P % 64 == 0 && Wt.size(1) == P && Wt.is_contiguous() && N == SK + A);
+  const size_t shmem = ((16 * P * 2 + 15) & ~15) + 32 * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((scan2_blnsilu_kernel<true>), dim3((N + 63) / 64), dim3(256), shmem, stream.stream(),
+                     sc2_bp(gy_in), gy_in.stride(0), sc2_bp(g_in), g_in.stride(0), sc2_bp(lnw),
+                     sc2_bp(lnb), mean.data_ptr<float>(), rstd.data_ptr<float>(), sc2_bp(Wt), sc2_bp(f),
+                     sc2_bpm(gg_out), gg_out.stride(0), glnw.data_ptr<float>(), glnb.data_ptr<float>(),
+                     nullptr, 0, sc2_bpm(gz_carry), giz_acc.data_ptr<float>(), sc2_bpm(ga_out), B, P, N,
+                     SK, A);
+}
+
+void scan2_b2(const torch::Tensor& gh, const c10::optional<torch::Tensor>& gh2, const torch::Tensor& gh3,
+              const torch::Tensor& y_in, const torch::Tensor& hu, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, const torch::Tensor& mean, const torch::Tensor& rstd,
+              const torch::Tensor& W2t, const torch::Tensor& f, torch::Tensor gy_out, torch::Tensor glnw,
+              torch::Tensor glnb, torch::Tensor gh_carry, torch::Tensor gih_acc, torch::Tensor ghu_out) {
+  const int B = (int)gh.size(0), H = (int)gh.size(1);
+  const int HD = (int)W2t.size(0), D = HD - H;
+  TORCH_CHECK(B <= 16 && H % 64 == 0 && HD % 64 == 0 && W2t.size(1) == 3 * H && W2t.is_contiguous());
+  TORCH_CHECK(gh.is_contiguous() && gh_carry.is_contiguous() && ghu_out.is_contiguous());
+  const size_t shmem = ((16 * 3 * H * 2 + 16 * H * 2 + 15) & ~15) + 32 * sizeof(float);
+  TORCH_CHECK(shmem <= 160 * 1024, "scan2_b2: H too large for LDS");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const __hip_bfloat16* g2 = gh2.has_value() ? sc2_bp(*gh2) : nullptr;
+  hipLaunchKernelGGL(scan2_bgru_kernel, dim3(HD / 64), dim3(256), shmem, stream.stream(), sc2_bp(gh), g2,
+                     sc2_bp(gh3), gh3.stride(0), sc2_bp(y_in), y_in.stride(0), sc2_bp(hu), hu.stride(0),
+                     sc2_bp(lnw), sc2_bp(lnb), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     sc2_bp(W2t), sc2_bp(f), sc2_bpm(gy_out), gy_out.stride(0), glnw.data_ptr<float>(),
+                     glnb.data_ptr<float>(), sc2_bpm(gh_carry), gih_acc.data_ptr<float>(),
+                     sc2_bpm(ghu_out), B, H, D);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("symlog_fwd", [](const torch::Tensor& x) { return symmath<0>(x, c10::nullopt); });
   m.def("symlog_bwd", [](const torch::Tensor& x, const torch::Tensor& g) { return symmath<1>(x, g); });
@@ -2821,6 +3659,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("twohot_lp_bwd", &twohot_lp_bwd);
   m.def("klbal_fwd", &klbal_fwd);
   m.def("klbal_bwd", &klbal_bwd);
+  m.def("scan2_f1", &scan2_f1);
+  m.def("scan2_f2", &scan2_f2);
+  m.def("scan2_f3", &scan2_f3);
+  m.def("scan2_f4", &scan2_f4);
+  m.def("scan2_b4", &scan2_b4);
+  m.def("scan2_b3", &scan2_b3);
+  m.def("scan2_b2", &scan2_b2);
+  m.def("scan2_b1", &scan2_b1);
   m.def("g16_plain", &g16_plain);
   m.def("g16_splitk", &g16_splitk);
   m.def("g16_ln_silu", &g16_ln_silu);

@@ -135,6 +135,46 @@ class _RSSMScan(torch.autograd.Function):
             ctx.dims = (T, B, E, A, H, SK, D, P, S, discrete)
             ctx.unimix = unimix
             return h_seq, z_seq, m_seq
+        # round-2 fused-phase path: ONE multi-workgroup GEMM+epilogue kernel
+        # per phase (4 launches/step instead of ~9; LN stats exchanged with
+        # the in-launch ticket pattern).  Falls back to the launch-per-op path
+        # for shapes outside the kernel contracts.
+        v2_ok = (
+            dt == torch.bfloat16
+            and B <= 16
+            and w1.dtype == dt
+            and D % 64 == 0 and H % 64 == 0 and P % 64 == 0 and SK % 64 == 0
+            and discrete <= 64 and (discrete & (discrete - 1)) == 0
+            and max(D, 3 * H, P, SK) // 64 <= 256
+            and 16 * (((H + E) + 31) & ~31) * 2 + 160 <= 160 * 1024
+            and 16 * 3 * H * 2 + 16 * H * 2 + 160 <= 160 * 1024
+            and os.environ.get("SHEEPRL_AMD_SCAN_V2", "1") == "1"
+        )
+        if v2_ok:
+            ws_f = torch.zeros(T, 3, 32, device=dev, dtype=torch.float32)
+            tk_f = torch.zeros(T, 3, device=dev, dtype=torch.int32)
+            iz_c = iz.contiguous()
+            ih_c = ih.contiguous()
+            for t in range(T):
+                zp = None if t == 0 else z_seq[t - 1]
+                hp = None if t == 0 else h_seq[t - 1]
+                ext.scan2_f1(zp, iz_c, hp, ih_c, actions[t], f_all[t], w1, lnw1, lnb1,
+                             x_s[t], hu_s[t], g1_s[t], mr1_s[0, t], mr1_s[1, t],
+                             ws_f[t, 0], tk_f[t, 0], eps)
+                ext.scan2_f2(hu_s[t], w2, lnwg, lnbg, y_s[t], h_seq[t], r_s[t, :, :H],
+                             mrg_s[0, t], mrg_s[1, t], ws_f[t, 1], tk_f[t, 1], eps)
+                ext.scan2_f3(r_s[t], w3, lnw3, lnb3, p_s[t], g3_s[t], mr3_s[0, t],
+                             mr3_s[1, t], ws_f[t, 2], tk_f[t, 2], eps)
+                ext.scan2_f4(p_s[t], w4, b4, urand_all[t].reshape(B, SK), m_seq[t],
+                             z_seq[t], s_s[t].view(B, SK), discrete, unimix)
+            ctx.save_for_backward(
+                f_all, w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4,
+                x_s, g1_s, mr1_s, hu_s, y_s, mrg_s, r_s, g3_s, mr3_s, p_s, s_s,
+            )
+            ctx.dims = (T, B, E, A, H, SK, D, P, S, discrete)
+            ctx.unimix = unimix
+            ctx.v2 = True
+            return h_seq, z_seq, m_seq
         h0 = torch.zeros(B, H, device=dev, dtype=dt)
         z0 = torch.zeros(B, SK, device=dev, dtype=dt)
         # hand-written split-K M=16 MFMA kernel for the long-K
@@ -191,6 +231,7 @@ class _RSSMScan(torch.autograd.Function):
         )
         ctx.dims = (T, B, E, A, H, SK, D, P, S, discrete)
         ctx.unimix = unimix
+        ctx.v2 = False
         return h_seq, z_seq, m_seq
 
     @staticmethod
@@ -237,6 +278,35 @@ class _RSSMScan(torch.autograd.Function):
         gh_carry = torch.empty(B, H, device=dev, dtype=dt)   # written at t before read at t-1
         gz_carry = torch.empty(B, SK, device=dev, dtype=dt)
 
+        if getattr(ctx, "v2", False):
+            # fused-phase backward: 4 launches/step; the per-WG recompute of
+            # the small row-local gradients replaces all cross-WG traffic.
+            # Weight transposes once per step so the bwd GEMMs read k-contig.
+            w1t = w1.t().contiguous()
+            w2t = w2.t().contiguous()
+            w3t = w3.t().contiguous()
+            w4t = w4.t().contiguous()
+            for t in range(T - 1, -1, -1):
+                zc = gz_carry if t < T - 1 else None
+                hc = gh_carry if t < T - 1 else None
+                ext.scan2_b4(g_m_seq[t].view(B, SK), g_z_seq[t].view(B, SK), zc,
+                             s_s[t].view(B, SK), w4t, graw_s[t], gp, discrete, unimix)
+                ext.scan2_b3(gp, g3_s[t], lnw3, lnb3, mr3_s[0, t], mr3_s[1, t], w3t,
+                             gg3_s[t], glnw3, glnb3, gr_s[t])
+                ext.scan2_b2(g_h_seq[t], hc, gr_s[t, :, :H], y_s[t], hu_s[t, :, :H],
+                             lnwg, lnbg, mrg_s[0, t], mrg_s[1, t], w2t, f_all[t],
+                             gy_s[t], glnwg, glnbg, gh_carry, gih_acc, ghu)
+                ext.scan2_b1(ghu[:, H:], g1_s[t], lnw1, lnb1, mr1_s[0, t], mr1_s[1, t],
+                             w1t, f_all[t], gg1_s[t], glnw1, glnb1, gz_carry, giz_acc,
+                             g_actions[t])
+            return _scan_weight_grads(
+                ext, T, B, E, A, H, SK, D, P, dt, dev,
+                graw_s, gg3_s, gy_s, gg1_s, gr_s, g_actions,
+                p_s, r_s, hu_s, x_s,
+                gih_acc, giz_acc, glnw1, glnb1, glnwg, glnbg, glnw3, glnb3,
+                lnw1, lnb1, lnwg, lnbg, lnw3, lnb3,
+            )
+
         for t in range(T - 1, -1, -1):
             f = f_all[t]
             zc = gz_carry.view(B, S, discrete) if t < T - 1 else None
@@ -256,27 +326,43 @@ class _RSSMScan(torch.autograd.Function):
             # one kernel: carries, action grad, init-state accumulators
             ext.scan_resets_bwd(ghu, ghp, gx, f, gh_carry, gz_carry, g_actions[t], gih_acc, giz_acc)
 
-        # batched weight grads: one MFMA GEMM per weight over all T*B rows
-        TB = T * B
-        ones_row = torch.ones(1, TB, device=dev, dtype=dt)
-        gW4 = torch.mm(graw_s.view(TB, SK).t(), p_s.view(TB, P))
-        # bias grad as a GEMM against ones: torch's bf16 column-reduce sum(0)
-        # was measured at ~300 us for this shape; the GEMV path is ~5 us.
-        gb4 = torch.mm(ones_row, graw_s.view(TB, SK)).view(SK)
-        gW3 = torch.mm(gg3_s.view(TB, P).t(), r_s.view(TB, H + E))
-        gW2 = torch.mm(gy_s.view(TB, 3 * H).t(), hu_s.view(TB, H + D))
-        gW1 = torch.mm(gg1_s.view(TB, D).t(), x_s.view(TB, SK + A))
-        return (
-            gr_s[:, :, H:],                       # g_embed (strided view is fine)
-            g_actions, None,
-            gih_acc.to(dt).view(1, B, H),
-            giz_acc.to(dt).view(1, B, SK),
-            gW1, glnw1.to(lnw1.dtype), glnb1.to(lnb1.dtype),
-            gW2, glnwg.to(lnwg.dtype), glnbg.to(lnbg.dtype),
-            gW3, glnw3.to(lnw3.dtype), glnb3.to(lnb3.dtype),
-            gW4, gb4,
-            None, None, None, None,
+        return _scan_weight_grads(
+            ext, T, B, E, A, H, SK, D, P, dt, dev,
+            graw_s, gg3_s, gy_s, gg1_s, gr_s, g_actions,
+            p_s, r_s, hu_s, x_s,
+            gih_acc, giz_acc, glnw1, glnb1, glnwg, glnbg, glnw3, glnb3,
+            lnw1, lnb1, lnwg, lnbg, lnw3, lnb3,
         )
+
+
+def _scan_weight_grads(
+    ext, T, B, E, A, H, SK, D, P, dt, dev,
+    graw_s, gg3_s, gy_s, gg1_s, gr_s, g_actions,
+    p_s, r_s, hu_s, x_s,
+    gih_acc, giz_acc, glnw1, glnb1, glnwg, glnbg, glnw3, glnb3,
+    lnw1, lnb1, lnwg, lnbg, lnw3, lnb3,
+):
+    # batched weight grads: one MFMA GEMM per weight over all T*B rows
+    TB = T * B
+    ones_row = torch.ones(1, TB, device=dev, dtype=dt)
+    gW4 = torch.mm(graw_s.view(TB, SK).t(), p_s.view(TB, P))
+    # bias grad as a GEMM against ones: torch's bf16 column-reduce sum(0)
+    # was measured at ~300 us for this shape; the GEMV path is ~5 us.
+    gb4 = torch.mm(ones_row, graw_s.view(TB, SK)).view(SK)
+    gW3 = torch.mm(gg3_s.view(TB, P).t(), r_s.view(TB, H + E))
+    gW2 = torch.mm(gy_s.view(TB, 3 * H).t(), hu_s.view(TB, H + D))
+    gW1 = torch.mm(gg1_s.view(TB, D).t(), x_s.view(TB, SK + A))
+    return (
+        gr_s[:, :, H:],                       # g_embed (strided view is fine)
+        g_actions, None,
+        gih_acc.to(dt).view(1, B, H),
+        giz_acc.to(dt).view(1, B, SK),
+        gW1, glnw1.to(lnw1.dtype), glnb1.to(lnb1.dtype),
+        gW2, glnwg.to(lnwg.dtype), glnbg.to(lnbg.dtype),
+        gW3, glnw3.to(lnw3.dtype), glnb3.to(lnb3.dtype),
+        gW4, gb4,
+        None, None, None, None,
+    )
 
 
 def rssm_scan(

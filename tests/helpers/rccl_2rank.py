@@ -61,8 +61,10 @@ def main() -> None:
     rank = int(os.environ["RANK"])
     world = int(os.environ["WORLD_SIZE"])
     assert world == 2
-    torch.cuda.set_device(0)  # both ranks share the single GPU
-    device = torch.device("cuda", 0)
+    # one rank per device when the box has >= 2 (RCCL refuses co-located
+    # ranks: "Duplicate GPU detected" — measured on ROCm 7.2 / NCCL 2.26)
+    torch.cuda.set_device(rank % torch.cuda.device_count())
+    device = torch.device("cuda", torch.cuda.current_device())
     dist.init_process_group("nccl", rank=rank, world_size=world)
 
     from sheeprl_amd.parallel.gradsync import GradSync
