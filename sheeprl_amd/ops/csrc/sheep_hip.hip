@@ -2835,9 +2835,50 @@ __device__ __forceinline__ void sc2_stage_vec(const __hip_bfloat16* src, long ss
   }
 }
 
-// one 16x16 C tile of  A[16, K] @ W[N, K]^T  with A staged in LDS ([16][KP],
-// KP = K rounded up to 32, zero-padded) and W streamed; wrow clamped so
-// ragged-N grids stay in bounds (garbage cols are never written).
+// one 16x16 C tile of  A[16, K] @ W[N, K]^T  over k in [kbeg, kend) with A
+// staged in LDS ([16][KP], KP = K rounded up to 32, zero-padded) and W
+// streamed; wrow clamped so ragged-N grids stay in bounds (garbage cols are
+// never written).
+__device__ __forceinline__ f32x4 sc2_gemm_tile_range(const __hip_bfloat16* ldsA, int KP,
+                                                     const __hip_bfloat16* W, long ws_, int K, int N,
+                                                     int ncol0, int kbeg, int kend) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow = min(ncol0 + arow, N - 1);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int KU = kbeg + ((kend - kbeg) & ~31);
+  const int KU8 = kbeg + ((KU - kbeg) & ~255);
+  for (int k0 = kbeg; k0 < KU8; k0 += 256) {
+    bf16x8 af[8], bf[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int k = k0 + u * 32 + kgrp * 8;
+      af[u] = *(const bf16x8*)(ldsA + arow * KP + k);
+      bf[u] = *(const bf16x8*)(W + (long)wrow * ws_ + k);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf[u], acc, 0, 0, 0);
+  }
+#pragma unroll 4
+  for (int k0 = KU8; k0 < KU; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = *(const bf16x8*)(ldsA + arow * KP + k);
+    bf16x8 b = *(const bf16x8*)(W + (long)wrow * ws_ + k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  if (KU < kend) {
+    const int k = KU + kgrp * 8;
+    bf16x8 a = *(const bf16x8*)(ldsA + arow * KP + k);  // zero-padded: safe
+    bf16x8 b;
+    const __hip_bfloat16* p = W + (long)wrow * ws_ + k;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) b[e] = (k + e < kend) ? (__bf16)__bfloat162float(p[e]) : (__bf16)0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  return acc;
+}
+
 __device__ __forceinline__ f32x4 sc2_gemm_tile(const __hip_bfloat16* ldsA, int KP,
                                                const __hip_bfloat16* W, long ws_, int K, int N,
                                                int ncol0) {
@@ -2892,7 +2933,10 @@ __device__ __forceinline__ float sc2_colsum(float v) {
 // episode-reset-masked GEMM input x = [z', a'] (phase 1) in LDS and persists
 // x / h' for the backward.
 // ---------------------------------------------------------------------------
-template <bool RESETS, bool T0>
+// KSPLIT=false: grid N/64, each of 4 waves owns a separate 16-col tile over
+// full K.  KSPLIT=true (long-K shapes): grid N/16, the 4 waves K-split ONE
+// 16-col tile and combine through LDS — 4x the weight-stream concurrency.
+template <bool RESETS, bool T0, bool KSPLIT>
 __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
     const __hip_bfloat16* __restrict__ a_in, long as_,      // staged A (phase 3) or z_prev (phase 1)
     const __hip_bfloat16* __restrict__ iz,                  // phase 1 only
@@ -2914,34 +2958,52 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
   float* lds_sum = (float*)(smem + ((16 * KP * 2 + 15) & ~15));  // [32]
   if (threadIdx.x < 32) lds_sum[threadIdx.x] = 0.f;
   if (RESETS) {
-    // x = [(1-f) z_prev + f iz, (1-f) a]; rows >= B and cols >= K zero
-    for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
-      const int m = i / KP, c = i - m * KP;
-      float v = 0.f;
-      if (m < B && c < K) {
+    // x = [(1-f) z_prev + f iz, (1-f) a]; rows >= B and cols >= K zero.
+    // Vectorized: SK % 8 == 0 (host-checked), the action/pad tail is scalar.
+    const int KPV = KP >> 3, SKV = SK >> 3;
+    uint4* d4 = (uint4*)ldsA;
+    for (int i = threadIdx.x; i < 16 * KPV; i += blockDim.x) {
+      const int m = i / KPV, cv = i - m * KPV;
+      LnVec<__hip_bfloat16, 8> v;
+      v.u = (uint4){0u, 0u, 0u, 0u};
+      if (m < B) {
         const float fb = __bfloat162float(f[m]);
-        if (c < SK) {
-          float zp = T0 ? 0.f : __bfloat162float(a_in[(long)m * SK + c]);
-          v = (1.f - fb) * zp + fb * __bfloat162float(iz[(long)m * SK + c]);
+        if (cv < SKV) {
+          bf16x8 izv = *(const bf16x8*)(iz + (long)m * SK + ((long)cv << 3));
+          if (T0) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) v.e[e] = __float2bfloat16(fb * (float)izv[e]);
+          } else {
+            bf16x8 zpv = *(const bf16x8*)(a_in + (long)m * SK + ((long)cv << 3));
+#pragma unroll
+            for (int e = 0; e < 8; ++e)
+              v.e[e] = __float2bfloat16((1.f - fb) * (float)zpv[e] + fb * (float)izv[e]);
+          }
         } else {
-          v = (1.f - fb) * __bfloat162float(act[(long)m * A + (c - SK)]);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int c = (cv << 3) + e;
+            if (c >= SK && c < SK + A)
+              v.e[e] = __float2bfloat16((1.f - fb) * __bfloat162float(act[(long)m * A + (c - SK)]));
+          }
         }
       }
-      ldsA[i] = __float2bfloat16(v);
+      d4[i] = v.u;
     }
     __syncthreads();
-    if (blockIdx.x == 0) {
-      // persist x for backward + h' = (1-f) h_prev + f ih into hu[:, :H]
-      for (int i = threadIdx.x; i < B * K; i += blockDim.x) {
-        const int m = i / K, c = i - m * K;
-        x_out[(long)m * xs + c] = ldsA[m * KP + c];
-      }
-      for (int i = threadIdx.x; i < B * H; i += blockDim.x) {
-        const int m = i / H, j = i - m * H;
-        const float fb = __bfloat162float(f[m]);
-        float hp = T0 ? 0.f : __bfloat162float(h_prev[(long)m * H + j]);
-        st(hu_out, (long)m * hus + j, (1.f - fb) * hp + fb * __bfloat162float(ih[(long)m * H + j]));
-      }
+    // persist x and h' for the backward, striped across the grid (every WG
+    // staged the identical LDS image — no single-WG straggler)
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < (long)B * K;
+         i += (long)gridDim.x * blockDim.x) {
+      const int m = (int)(i / K), c = (int)(i - (long)m * K);
+      x_out[(long)m * xs + c] = ldsA[m * KP + c];
+    }
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < (long)B * H;
+         i += (long)gridDim.x * blockDim.x) {
+      const int m = (int)(i / H), j = (int)(i - (long)m * H);
+      const float fb = __bfloat162float(f[m]);
+      float hp = T0 ? 0.f : __bfloat162float(h_prev[(long)m * H + j]);
+      st(hu_out, (long)m * hus + j, (1.f - fb) * hp + fb * __bfloat162float(ih[(long)m * H + j]));
     }
   } else {
     if ((K & 7) == 0 && (as_ & 7) == 0) {
@@ -2958,21 +3020,41 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
   const int arow = lane & 15;
   const int kgrp = lane >> 4;
   const int wv = threadIdx.x >> 6;
-  const int ncol0 = (blockIdx.x * 4 + wv) * 16;
-  f32x4 acc = sc2_gemm_tile(ldsA, KP, W, K, K, N, ncol0);
+  const int ncol0 = KSPLIT ? blockIdx.x * 16 : (blockIdx.x * 4 + wv) * 16;
+  f32x4 acc;
+  if (KSPLIT) {
+    const int KQ = (((K + 3) / 4) + 31) & ~31;
+    const int kbeg = min(wv * KQ, K);
+    const int kend = min(kbeg + KQ, K);
+    acc = sc2_gemm_tile_range(ldsA, KP, W, K, K, N, ncol0, kbeg, kend);
+    // combine the 4 K-slices through LDS (ldsA is dead after the GEMM)
+    __syncthreads();
+    float* comb = (float*)smem;  // [4][16][16]
+#pragma unroll
+    for (int r = 0; r < 4; ++r) comb[(wv * 16 + kgrp * 4 + r) * 16 + arow] = acc[r];
+    __syncthreads();
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = kgrp * 4 + r;
+      acc[r] = comb[(m)*16 + arow] + comb[(16 + m) * 16 + arow] + comb[(32 + m) * 16 + arow] +
+               comb[(48 + m) * 16 + arow];
+    }
+  } else {
+    acc = sc2_gemm_tile(ldsA, KP, W, K, K, N, ncol0);
+  }
+  const bool emit = !KSPLIT || wv == 0;  // KSPLIT: wave 0 owns the tile
   // write pre-LN G and accumulate row partials
-  float s = 0.f, s2 = 0.f;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int m = kgrp * 4 + r;
-    if (m < B && ncol0 + arow < N) g_out[(long)m * gs + ncol0 + arow] = __float2bfloat16(acc[r]);
+    if (emit && m < B && ncol0 + arow < N) g_out[(long)m * gs + ncol0 + arow] = __float2bfloat16(acc[r]);
   }
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     float v = (ncol0 + arow < N) ? acc[r] : 0.f;
     float cs = sc2_colsum(v);
     float cs2 = sc2_colsum(v * v);
-    if (arow == 0) {
+    if (emit && arow == 0) {
       atomicAdd(&lds_sum[kgrp * 4 + r], cs);
       atomicAdd(&lds_sum[16 + kgrp * 4 + r], cs2);
     }
@@ -2997,7 +3079,7 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
   for (int r = 0; r < 4; ++r) {
     const int m = kgrp * 4 + r;
     const int j = ncol0 + arow;
-    if (m < B && j < N) {
+    if (emit && m < B && j < N) {
       float z = (acc[r] - mr_[0][m]) * mr_[1][m] * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
       z = z / (1.f + expf(-z));
       st(hu_out, (long)m * hus + hu_off + j, z);
@@ -3179,34 +3261,35 @@ __global__ void __launch_bounds__(256) scan2_b4_kernel(
   const int KP = SK;  // SK % 64 == 0
   __hip_bfloat16* graw = (__hip_bfloat16*)smem;  // [16][SK]
   const int lane = threadIdx.x & 63;
-  const int groups = SK / KD;
-  const int gpercall = (int)blockDim.x / KD;
-  for (int task = threadIdx.x / KD; task < 16 * groups; task += gpercall) {
-    const int m = task / groups;
-    const int g = task - m * groups;
-    const int j = (threadIdx.x % KD);
-    const long idx = (long)m * SK + g * KD + j;
-    float t = 0.f, sj = 0.f;
-    if (m < B) {
-      sj = s_saved[idx];
-      float pj = (1.f - unimix) * sj + unimix / KD;
-      t = gm[idx] / pj + __bfloat162float(gon[idx]) + (gon2 ? __bfloat162float(gon2[idx]) : 0.f);
-    }
-    float acc = t * sj;
+  const int wv = threadIdx.x >> 6;
+  // wave-per-4-rows coalesced recompute; KD <= 64 groups align to lane
+  // halves, so the group softmax-grad dot reduces with in-wave shuffles
+  for (int mr2 = 0; mr2 < 4; ++mr2) {
+    const int m = wv * 4 + mr2;
+    for (int c0 = 0; c0 < SK; c0 += 64) {
+      const int c = c0 + lane;
+      float t = 0.f, sj = 0.f;
+      if (m < B) {
+        const long idx = (long)m * SK + c;
+        sj = s_saved[idx];
+        float pj = (1.f - unimix) * sj + unimix / KD;
+        t = gm[idx] / pj + __bfloat162float(gon[idx]) + (gon2 ? __bfloat162float(gon2[idx]) : 0.f);
+      }
+      float acc = t * sj;
 #pragma unroll
-    for (int off = 1; off < 64; off <<= 1) {
-      if (off < KD) acc += __shfl_xor(acc, off, 64);
+      for (int off = 1; off < 64; off <<= 1) {
+        if (off < KD) acc += __shfl_xor(acc, off, 64);
+      }
+      graw[m * KP + c] = __float2bfloat16(m < B ? (1.f - unimix) * sj * (t - acc) : 0.f);
     }
-    graw[m * KP + g * KD + j] = __float2bfloat16(m < B ? (1.f - unimix) * sj * (t - acc) : 0.f);
   }
   __syncthreads();
-  if (blockIdx.x == 0) {
-    for (int i = threadIdx.x; i < B * SK; i += blockDim.x) {
-      const int m = i / SK, c = i - m * SK;
-      graw_out[(long)m * gws + c] = graw[m * KP + c];
-    }
+  // persist graw striped across the grid (all WGs hold the same image)
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < (long)B * SK;
+       i += (long)gridDim.x * blockDim.x) {
+    const int m = (int)(i / SK), c = (int)(i - (long)m * SK);
+    graw_out[(long)m * gws + c] = graw[m * KP + c];
   }
-  const int wv = threadIdx.x >> 6;
   const int ncol0 = (blockIdx.x * 4 + wv) * 16;
   const int arow = lane & 15;
   const int kgrp = lane >> 4;
@@ -3246,27 +3329,37 @@ __global__ void __launch_bounds__(256) scan2_blnsilu_kernel(
   float* s12 = (float*)(smem + ((16 * KP * 2 + 15) & ~15));        // [2][16]
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
-  // pass A: per-row sums (wave wv owns rows 4wv..4wv+3) + gz into LDS
+  // pass A (vectorized): per-row sums (wave wv owns rows 4wv..4wv+3) + gz
   for (int mr2 = 0; mr2 < 4; ++mr2) {
     const int m = wv * 4 + mr2;
     float s1 = 0.f, s2 = 0.f;
     if (m < B) {
       const float mn = mean[m], rs = rstd[m];
-      for (int j = lane; j < P; j += 64) {
-        float xh = (__bfloat162float(g_in[(long)m * gis + j]) - mn) * rs;
-        float w = __bfloat162float(lnw[j]);
-        float z = xh * w + __bfloat162float(lnb[j]);
-        float sig = 1.f / (1.f + expf(-z));
-        float gz = __bfloat162float(gy_in[(long)m * gys + j]) * sig * (1.f + z * (1.f - sig));
-        gg[m * KP + j] = __float2bfloat16(gz);  // temporarily gz
-        float gxh = gz * w;
-        s1 += gxh;
-        s2 += gxh * xh;
+      for (int j0 = lane * 8; j0 < P; j0 += 512) {
+        bf16x8 gv = *(const bf16x8*)(g_in + (long)m * gis + j0);
+        bf16x8 gyv = *(const bf16x8*)(gy_in + (long)m * gys + j0);
+        bf16x8 lwv = *(const bf16x8*)(lnw + j0);
+        bf16x8 lbv = *(const bf16x8*)(lnb + j0);
+        bf16x8 ogz;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float xh = ((float)gv[e] - mn) * rs;
+          float w = (float)lwv[e];
+          float z = xh * w + (float)lbv[e];
+          float sig = 1.f / (1.f + expf(-z));
+          float gz = (float)gyv[e] * sig * (1.f + z * (1.f - sig));
+          ogz[e] = (__bf16)gz;
+          float gxh = gz * w;
+          s1 += gxh;
+          s2 += gxh * xh;
+        }
+        *(bf16x8*)(gg + m * KP + j0) = ogz;  // temporarily gz
       }
       s1 = wave_sum(s1);
       s2 = wave_sum(s2);
     } else {
-      for (int j = lane; j < P; j += 64) gg[m * KP + j] = (__hip_bfloat16)0.f;
+      bf16x8 zero{};
+      for (int j0 = lane * 8; j0 < P; j0 += 512) *(bf16x8*)(gg + m * KP + j0) = zero;
     }
     if (lane == 0) {
       s12[m] = s1 / P;
@@ -3274,9 +3367,9 @@ __global__ void __launch_bounds__(256) scan2_blnsilu_kernel(
     }
   }
   __syncthreads();
-  // LN affine grads: stripe WGs flush one atomicAdd per column
+  // LN affine grads: lane-per-column coalesced row loop, one stripe per WG
   for (int stripe = blockIdx.x; stripe < P / 64; stripe += gridDim.x) {
-    const int j = stripe * 64 + threadIdx.x % 64;
+    const int j = stripe * 64 + (threadIdx.x & 63);
     if (threadIdx.x < 64) {
       float sw = 0.f, sb = 0.f;
       for (int m = 0; m < B; ++m) {
@@ -3290,22 +3383,29 @@ __global__ void __launch_bounds__(256) scan2_blnsilu_kernel(
     }
   }
   __syncthreads();
-  // pass B: finalize gg in place
-  for (int i = threadIdx.x; i < 16 * KP; i += blockDim.x) {
-    const int m = i / KP, j = i - m * KP;
-    if (m < B) {
-      float gz = __bfloat162float(gg[i]);
-      float xh = (__bfloat162float(g_in[(long)m * gis + j]) - mean[m]) * rstd[m];
-      float gv = (gz * __bfloat162float(lnw[j]) - s12[m] - xh * s12[16 + m]) * rstd[m];
-      gg[i] = __float2bfloat16(gv);
+  // pass B: finalize gg in place (vectorized)
+  for (int iv = threadIdx.x; iv < 16 * (KP >> 3); iv += blockDim.x) {
+    const int m = iv / (KP >> 3), jv = iv - m * (KP >> 3);
+    if (m >= B) continue;
+    const int j0 = jv << 3;
+    bf16x8 gzv = *(const bf16x8*)(gg + m * KP + j0);
+    bf16x8 gv = *(const bf16x8*)(g_in + (long)m * gis + j0);
+    bf16x8 lwv = *(const bf16x8*)(lnw + j0);
+    bf16x8 outv;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float xh = ((float)gv[e] - mean[m]) * rstd[m];
+      outv[e] = (__bf16)(((float)gzv[e] * (float)lwv[e] - s12[m] - xh * s12[16 + m]) * rstd[m]);
     }
+    *(bf16x8*)(gg + m * KP + j0) = outv;
   }
   __syncthreads();
-  // persist gg stripes (for the batched weight-grad GEMMs)
+  // persist gg stripes (for the batched weight-grad GEMMs), vectorized
   for (int stripe = blockIdx.x; stripe < P / 64; stripe += gridDim.x) {
-    for (int i = threadIdx.x; i < B * 64; i += blockDim.x) {
-      const int m = i / 64, j = stripe * 64 + (i & 63);
-      gg_out[(long)m * ggs + j] = gg[m * KP + j];
+    for (int i = threadIdx.x; i < B * 8; i += blockDim.x) {
+      const int m = i / 8, jv = i & 7;
+      const int j0 = stripe * 64 + jv * 8;
+      *(bf16x8*)(gg_out + (long)m * ggs + j0) = *(const bf16x8*)(gg + m * KP + j0);
     }
   }
   const int arow = lane & 15;
@@ -3361,46 +3461,68 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
   float* s12 = (float*)(smem + ((16 * DD * 2 + 16 * H * 2 + 15) & ~15));  // [2][16]
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
+  // pass 1, wave per 4 rows, 8-wide vector loads: full Hafner-gate backward
+  // into LDS (gz, ghp) + the LN row sums
   for (int mr2 = 0; mr2 < 4; ++mr2) {
     const int m = wv * 4 + mr2;
     float s1 = 0.f, s2 = 0.f;
     if (m < B) {
       const float mn = mean[m], rs = rstd[m];
-      for (int j = lane; j < H; j += 64) {
-        float xh_r = (__bfloat162float(y_in[(long)m * ys2 + j]) - mn) * rs;
-        float xh_c = (__bfloat162float(y_in[(long)m * ys2 + H + j]) - mn) * rs;
-        float xh_u = (__bfloat162float(y_in[(long)m * ys2 + 2 * H + j]) - mn) * rs;
-        float zr = xh_r * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
-        float zc = xh_c * __bfloat162float(lnw[H + j]) + __bfloat162float(lnb[H + j]);
-        float zu = xh_u * __bfloat162float(lnw[2 * H + j]) + __bfloat162float(lnb[2 * H + j]);
-        float r = 1.f / (1.f + expf(-zr));
-        float rc = r * zc;
-        float c = tanhf(rc);
-        float u = 1.f / (1.f + expf(-(zu - 1.f)));
-        float g = __bfloat162float(gh[(long)m * H + j]) + (gh2 ? __bfloat162float(gh2[(long)m * H + j]) : 0.f) +
-                  __bfloat162float(gh3[(long)m * gh3s + j]);
-        float hp = __bfloat162float(hu[(long)m * hus + j]);
-        float gu = g * (c - hp);
-        float gc = g * u;
-        float gzu = gu * u * (1.f - u);
-        float grc = gc * (1.f - c * c);
-        float gzc = grc * r;
-        float gr = grc * zc;
-        float gzr = gr * r * (1.f - r);
-        ghp[m * H + j] = __float2bfloat16(g * (1.f - u));
-        gybuf[m * DD + j] = __float2bfloat16(gzr);
-        gybuf[m * DD + H + j] = __float2bfloat16(gzc);
-        gybuf[m * DD + 2 * H + j] = __float2bfloat16(gzu);
-        float gxh_r = gzr * __bfloat162float(lnw[j]);
-        float gxh_c = gzc * __bfloat162float(lnw[H + j]);
-        float gxh_u = gzu * __bfloat162float(lnw[2 * H + j]);
-        s1 += gxh_r + gxh_c + gxh_u;
-        s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
+      for (int j0 = lane * 8; j0 < H; j0 += 512) {
+        bf16x8 yrv = *(const bf16x8*)(y_in + (long)m * ys2 + j0);
+        bf16x8 ycv = *(const bf16x8*)(y_in + (long)m * ys2 + H + j0);
+        bf16x8 yuv = *(const bf16x8*)(y_in + (long)m * ys2 + 2 * H + j0);
+        bf16x8 ghv = *(const bf16x8*)(gh + (long)m * H + j0);
+        bf16x8 gh3v = *(const bf16x8*)(gh3 + (long)m * gh3s + j0);
+        bf16x8 huv = *(const bf16x8*)(hu + (long)m * hus + j0);
+        bf16x8 lwr = *(const bf16x8*)(lnw + j0);
+        bf16x8 lwc = *(const bf16x8*)(lnw + H + j0);
+        bf16x8 lwu = *(const bf16x8*)(lnw + 2 * H + j0);
+        bf16x8 lbr = *(const bf16x8*)(lnb + j0);
+        bf16x8 lbc = *(const bf16x8*)(lnb + H + j0);
+        bf16x8 lbu = *(const bf16x8*)(lnb + 2 * H + j0);
+        bf16x8 gh2v{};
+        if (gh2) gh2v = *(const bf16x8*)(gh2 + (long)m * H + j0);
+        bf16x8 ozr, ozc, ozu, ophp;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          float xh_r = ((float)yrv[e] - mn) * rs;
+          float xh_c = ((float)ycv[e] - mn) * rs;
+          float xh_u = ((float)yuv[e] - mn) * rs;
+          float zr = xh_r * (float)lwr[e] + (float)lbr[e];
+          float zc = xh_c * (float)lwc[e] + (float)lbc[e];
+          float zu = xh_u * (float)lwu[e] + (float)lbu[e];
+          float r = 1.f / (1.f + expf(-zr));
+          float c = tanhf(r * zc);
+          float u = 1.f / (1.f + expf(-(zu - 1.f)));
+          float g = (float)ghv[e] + (gh2 ? (float)gh2v[e] : 0.f) + (float)gh3v[e];
+          float gu = g * (c - (float)huv[e]);
+          float gc = g * u;
+          float gzu = gu * u * (1.f - u);
+          float grc = gc * (1.f - c * c);
+          float gzc = grc * r;
+          float gr = grc * zc;
+          float gzr = gr * r * (1.f - r);
+          ophp[e] = (__bf16)(g * (1.f - u));
+          ozr[e] = (__bf16)gzr;
+          ozc[e] = (__bf16)gzc;
+          ozu[e] = (__bf16)gzu;
+          float gxh_r = gzr * (float)lwr[e];
+          float gxh_c = gzc * (float)lwc[e];
+          float gxh_u = gzu * (float)lwu[e];
+          s1 += gxh_r + gxh_c + gxh_u;
+          s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
+        }
+        *(bf16x8*)(ghp + m * H + j0) = ophp;
+        *(bf16x8*)(gybuf + m * DD + j0) = ozr;
+        *(bf16x8*)(gybuf + m * DD + H + j0) = ozc;
+        *(bf16x8*)(gybuf + m * DD + 2 * H + j0) = ozu;
       }
       s1 = wave_sum(s1);
       s2 = wave_sum(s2);
     } else {
-      for (int j = lane; j < DD; j += 64) gybuf[m * DD + j] = (__hip_bfloat16)0.f;
+      bf16x8 zero{};
+      for (int j0 = lane * 8; j0 < DD; j0 += 512) *(bf16x8*)(gybuf + m * DD + j0) = zero;
     }
     if (lane == 0) {
       s12[m] = s1 / DD;
@@ -3408,9 +3530,9 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
     }
   }
   __syncthreads();
-  // LN affine grads per column stripe
+  // LN affine grads: lane-per-column coalesced row loop, one stripe per WG
   for (int stripe = blockIdx.x; stripe < DD / 64; stripe += gridDim.x) {
-    const int j = stripe * 64 + threadIdx.x % 64;
+    const int j = stripe * 64 + (threadIdx.x & 63);
     if (threadIdx.x < 64) {
       float sw = 0.f, sb = 0.f;
       for (int m = 0; m < B; ++m) {
@@ -3424,20 +3546,28 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
     }
   }
   __syncthreads();
-  // finalize gy in place
-  for (int i = threadIdx.x; i < 16 * DD; i += blockDim.x) {
-    const int m = i / DD, j = i - m * DD;
-    if (m < B) {
-      float gz = __bfloat162float(gybuf[i]);
-      float xh = (__bfloat162float(y_in[(long)m * ys2 + j]) - mean[m]) * rstd[m];
-      gybuf[i] = __float2bfloat16((gz * __bfloat162float(lnw[j]) - s12[m] - xh * s12[16 + m]) * rstd[m]);
+  // finalize gy in place (vectorized)
+  for (int iv = threadIdx.x; iv < 16 * (DD >> 3); iv += blockDim.x) {
+    const int m = iv / (DD >> 3), jv = iv - m * (DD >> 3);
+    if (m >= B) continue;
+    const int j0 = jv << 3;
+    bf16x8 gzv = *(const bf16x8*)(gybuf + m * DD + j0);
+    bf16x8 yv = *(const bf16x8*)(y_in + (long)m * ys2 + j0);
+    bf16x8 lwv = *(const bf16x8*)(lnw + j0);
+    bf16x8 outv;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      float xh = ((float)yv[e] - mean[m]) * rstd[m];
+      outv[e] = (__bf16)(((float)gzv[e] * (float)lwv[e] - s12[m] - xh * s12[16 + m]) * rstd[m]);
     }
+    *(bf16x8*)(gybuf + m * DD + j0) = outv;
   }
   __syncthreads();
   for (int stripe = blockIdx.x; stripe < DD / 64; stripe += gridDim.x) {
-    for (int i = threadIdx.x; i < B * 64; i += blockDim.x) {
-      const int m = i / 64, j = stripe * 64 + (i & 63);
-      gy_out[(long)m * gys + j] = gybuf[m * DD + j];
+    for (int i = threadIdx.x; i < B * 8; i += blockDim.x) {
+      const int m = i / 8, jv = i & 7;
+      const int j0 = stripe * 64 + jv * 8;
+      *(bf16x8*)(gy_out + (long)m * gys + j0) = *(const bf16x8*)(gybuf + m * DD + j0);
     }
   }
   const int arow = lane & 15;
@@ -3478,7 +3608,7 @@ void scan2_f1(const c10::optional<torch::Tensor>& z_prev, const torch::Tensor& i
               torch::Tensor ws2, torch::Tensor ticket, double eps) {
   const int B = (int)act.size(0), A = (int)act.size(1), SK = (int)iz.size(1), H = (int)ih.size(1);
   const int N = (int)W1.size(0), K = (int)W1.size(1);
-  TORCH_CHECK(B <= 16 && N % 64 == 0 && K == SK + A && W1.is_contiguous());
+  TORCH_CHECK(B <= 16 && N % 64 == 0 && K == SK + A && SK % 8 == 0 && W1.is_contiguous());
   const int KP = (K + 31) & ~31;
   const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -3486,7 +3616,8 @@ void scan2_f1(const c10::optional<torch::Tensor>& z_prev, const torch::Tensor& i
   const __hip_bfloat16* zp = t0 ? nullptr : sc2_bp(*z_prev);
   const __hip_bfloat16* hp = h_prev.has_value() ? sc2_bp(*h_prev) : nullptr;
 #define SC2_F1_LAUNCH(T0V)                                                                            \
-  hipLaunchKernelGGL((scan2_lnsilu_kernel<true, T0V>), dim3(N / 64), dim3(256), shmem, stream.stream(), \
+  hipLaunchKernelGGL((scan2_lnsilu_kernel<true, T0V, false>), dim3(N / 64), dim3(256), shmem,         \
+                     stream.stream(),                                                                 \
                      zp, 0, sc2_bp(iz), hp, sc2_bp(ih), sc2_bp(act), sc2_bp(f), sc2_bp(W1),           \
                      sc2_bp(lnw), sc2_bp(lnb), sc2_bpm(x_out), x_out.stride(0), sc2_bpm(hu_out),      \
                      hu_out.stride(0), sc2_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(),       \
@@ -3509,11 +3640,24 @@ void scan2_f3(const torch::Tensor& a_in, const torch::Tensor& W3, const torch::T
   const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "scan2_f3: K too large for LDS");
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL((scan2_lnsilu_kernel<false, false>), dim3(N / 64), dim3(256), shmem, stream.stream(),
-                     sc2_bp(a_in), a_in.stride(0), nullptr, nullptr, nullptr, nullptr, nullptr,
-                     sc2_bp(W3), sc2_bp(lnw), sc2_bp(lnb), nullptr, 0, sc2_bpm(p_out), p_out.stride(0),
-                     sc2_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     ws2.data_ptr<float>(), ticket.data_ptr<int>(), B, 0, 0, 0, N, K, (float)eps, 0);
+  // long-K shapes (the representation GEMM, K = H+E ~ 4.6k at S): K-split the
+  // 4 waves over ONE 16-col tile and run N/16 workgroups — 4x the
+  // weight-stream concurrency of the 4-tile layout
+  const bool ksplit = K >= 3072 && N / 16 <= 256;
+  if (ksplit)
+    hipLaunchKernelGGL((scan2_lnsilu_kernel<false, false, true>), dim3(N / 16), dim3(256), shmem,
+                       stream.stream(),
+                       sc2_bp(a_in), a_in.stride(0), nullptr, nullptr, nullptr, nullptr, nullptr,
+                       sc2_bp(W3), sc2_bp(lnw), sc2_bp(lnb), nullptr, 0, sc2_bpm(p_out), p_out.stride(0),
+                       sc2_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       ws2.data_ptr<float>(), ticket.data_ptr<int>(), B, 0, 0, 0, N, K, (float)eps, 0);
+  else
+    hipLaunchKernelGGL((scan2_lnsilu_kernel<false, false, false>), dim3(N / 64), dim3(256), shmem,
+                       stream.stream(),
+                       sc2_bp(a_in), a_in.stride(0), nullptr, nullptr, nullptr, nullptr, nullptr,
+                       sc2_bp(W3), sc2_bp(lnw), sc2_bp(lnb), nullptr, 0, sc2_bpm(p_out), p_out.stride(0),
+                       sc2_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       ws2.data_ptr<float>(), ticket.data_ptr<int>(), B, 0, 0, 0, N, K, (float)eps, 0);
 }
 
 void scan2_f2(const torch::Tensor& hu, const torch::Tensor& W2, const torch::Tensor& lnw,
@@ -3573,6 +3717,8 @@ void scan2_b3(const torch::Tensor& gy_in, const torch::Tensor& g_in, const torch
   const int B = (int)gy_in.size(0), P = (int)gy_in.size(1);
   const int N = (int)Wt.size(0);
   TORCH_CHECK(B <= 16 && P % 64 == 0 && Wt.size(1) == P && Wt.is_contiguous());
+  TORCH_CHECK(gy_in.stride(0) % 8 == 0 && g_in.stride(0) % 8 == 0 && gg_out.stride(0) % 8 == 0,
+              "scan2 blnsilu: 16B-aligned rows required");
   const size_t shmem = ((16 * P * 2 + 15) & ~15) + 32 * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL((scan2_blnsilu_kernel<false>), dim3((N + 63) / 64), dim3(256), shmem, stream.stream(),
@@ -3609,6 +3755,8 @@ void scan2_b2(const torch::Tensor& gh, const c10::optional<torch::Tensor>& gh2, 
   const int HD = (int)W2t.size(0), D = HD - H;
   TORCH_CHECK(B <= 16 && H % 64 == 0 && HD % 64 == 0 && W2t.size(1) == 3 * H && W2t.is_contiguous());
   TORCH_CHECK(gh.is_contiguous() && gh_carry.is_contiguous() && ghu_out.is_contiguous());
+  TORCH_CHECK(gh3.stride(0) % 8 == 0 && y_in.stride(0) % 8 == 0 && hu.stride(0) % 8 == 0 &&
+              gy_out.stride(0) % 8 == 0, "scan2_b2: 16B-aligned rows required");
   const size_t shmem = ((16 * 3 * H * 2 + 16 * H * 2 + 15) & ~15) + 32 * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "scan2_b2: H too large for LDS");
   auto stream = at::cuda::getCurrentCUDAStream();

@@ -145,6 +145,7 @@ class _RSSMScan(torch.autograd.Function):
             and w1.dtype == dt
             and D % 64 == 0 and H % 64 == 0 and P % 64 == 0 and SK % 64 == 0
             and discrete <= 64 and (discrete & (discrete - 1)) == 0
+            and (H + E) % 8 == 0
             and max(D, 3 * H, P, SK) // 64 <= 256
             and 16 * (((H + E) + 31) & ~31) * 2 + 160 <= 160 * 1024
             and 16 * 3 * H * 2 + 16 * H * 2 + 160 <= 160 * 1024

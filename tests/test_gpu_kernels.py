@@ -650,8 +650,6 @@ def test_scan_v2_fused_phases_match_v1(dims):
     is_first = (torch.rand(T, B, 1, device="cuda") < 0.3).float()
     is_first[0] = 1.0
     urand = torch.rand(T, B, S, K, device="cuda")
-    _ir, _ip = rssm.get_initial_states((1, B))
-    init = (_ir.contiguous(), _ip.contiguous())
 
     gh = torch.randn(T, B, H, device="cuda", dtype=torch.bfloat16)
     gz = torch.randn(T, B, SK, device="cuda", dtype=torch.bfloat16)
@@ -662,6 +660,10 @@ def test_scan_v2_fused_phases_match_v1(dims):
         try:
             for p in rssm.parameters():
                 p.grad = None
+            # fresh initial-state graph per run (its autograd graph is freed
+            # by each backward)
+            _ir, _ip = rssm.get_initial_states((1, B))
+            init = (_ir.contiguous(), _ip.contiguous())
             h, z, m = scan_mod.rssm_scan(rssm, embed, actions, is_first, init, urand=urand)
             torch.autograd.backward([h, z, m], [gh, gz, gm])
             grads = {n: p.grad.clone() for n, p in rssm.named_parameters() if p.grad is not None}
