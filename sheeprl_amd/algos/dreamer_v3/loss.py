@@ -53,7 +53,17 @@ def reconstruction_loss(
         repr_loss = kl_representation * torch.maximum(repr_loss, free_nats)
     kl_loss = dyn_loss + repr_loss
     if pc is not None and continue_targets is not None:
-        continue_loss = continue_scale_factor * -pc.log_prob(continue_targets)
+        logits = getattr(getattr(pc, "base_dist", None), "logits", None)
+        if (
+            logits is not None
+            and logits.is_cuda
+            and logits.shape == continue_targets.shape
+            and _ops.use_hip(logits)
+        ):
+            # fused Bernoulli NLL (one kernel each way)
+            continue_loss = continue_scale_factor * -_ops.bernoulli_log_prob(logits, continue_targets, 1)
+        else:
+            continue_loss = continue_scale_factor * -pc.log_prob(continue_targets)
     else:
         continue_loss = torch.zeros_like(reward_loss)
     rec_loss = (kl_regularizer * kl_loss + observation_loss + reward_loss + continue_loss).mean()

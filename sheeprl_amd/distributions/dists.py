@@ -40,6 +40,16 @@ class SymlogDistribution:
         return ops.symexp(self._mode)
 
     def log_prob(self, value: Tensor) -> Tensor:
+        if (
+            self._agg == "sum"
+            and self._mode.is_cuda
+            and value.shape == self._mode.shape
+            and not value.requires_grad
+            and ops.use_hip(self._mode)
+        ):
+            # one reduction kernel + one elementwise backward (symlog of the
+            # target folded into the kernel)
+            return ops.symlog_mse_log_prob(self._mode, value, len(self._dims))
         distance = -((self._mode - ops.symlog(value)) ** 2)
         if self._agg == "mean":
             return distance.mean(self._dims)
@@ -61,6 +71,14 @@ class MSEDistribution:
         return self._mode
 
     def log_prob(self, value: Tensor) -> Tensor:
+        if (
+            self._agg == "sum"
+            and self._mode.is_cuda
+            and value.shape == self._mode.shape
+            and not value.requires_grad
+            and ops.use_hip(self._mode)
+        ):
+            return ops.mse_log_prob(self._mode, value, len(self._dims))
         distance = -((self._mode - value) ** 2)
         if self._agg == "mean":
             return distance.mean(self._dims)

@@ -1,9 +1,12 @@
 from sheeprl_amd.ops._ext import get_ext, has_ext, require_ext, use_hip
 from sheeprl_amd.ops.functional import (
+    bernoulli_log_prob,
     gae,
     lambda_values,
+    mse_log_prob,
     symexp,
     symlog,
+    symlog_mse_log_prob,
     two_hot_decoder,
     two_hot_encoder,
     twohot_from_support,
@@ -31,6 +34,9 @@ __all__ = [
     "twohot_from_support",
     "gae",
     "lambda_values",
+    "mse_log_prob",
+    "symlog_mse_log_prob",
+    "bernoulli_log_prob",
     "layer_norm_act",
     "kl_balanced",
     "twohot_log_prob",

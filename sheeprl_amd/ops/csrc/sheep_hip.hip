@@ -2879,6 +2879,55 @@ __device__ __forceinline__ f32x4 sc2_gemm_tile_range(const __hip_bfloat16* ldsA,
   return acc;
 }
 
+// same tile over k in [kbeg, kend) with A read DIRECTLY from global memory
+// (row stride as_, B rows, zero beyond) — used by the K-split mode where each
+// wave touches a distinct K-quarter exactly once (LDS staging buys nothing).
+__device__ __forceinline__ f32x4 sc2_gemm_tile_range_g(const __hip_bfloat16* A, long as_, int B,
+                                                       const __hip_bfloat16* W, long ws_, int K,
+                                                       int N, int ncol0, int kbeg, int kend) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wrow = min(ncol0 + arow, N - 1);
+  const int arow_c = arow < B ? arow : B - 1;  // clamp: garbage rows unused
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int KU = kbeg + ((kend - kbeg) & ~31);
+  const int KU8 = kbeg + ((KU - kbeg) & ~255);
+  for (int k0 = kbeg; k0 < KU8; k0 += 256) {
+    bf16x8 af[8], bf[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int k = k0 + u * 32 + kgrp * 8;
+      af[u] = *(const bf16x8*)(A + (long)arow_c * as_ + k);
+      bf[u] = *(const bf16x8*)(W + (long)wrow * ws_ + k);
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u) acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[u], bf[u], acc, 0, 0, 0);
+  }
+#pragma unroll 4
+  for (int k0 = KU8; k0 < KU; k0 += 32) {
+    const int k = k0 + kgrp * 8;
+    bf16x8 a = *(const bf16x8*)(A + (long)arow_c * as_ + k);
+    bf16x8 b = *(const bf16x8*)(W + (long)wrow * ws_ + k);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  if (KU < kend) {
+    const int k = KU + kgrp * 8;
+    bf16x8 a, b;
+    const __hip_bfloat16* ap = A + (long)arow_c * as_ + k;
+    const __hip_bfloat16* p = W + (long)wrow * ws_ + k;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      a[e] = (k + e < kend) ? (__bf16)__bfloat162float(ap[e]) : (__bf16)0.f;
+      b[e] = (k + e < kend) ? (__bf16)__bfloat162float(p[e]) : (__bf16)0.f;
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  // note: C rows m >= B carry clamped-row garbage; every caller guards
+  // its writes and statistics with m < B (same contract as the LDS path)
+  return acc;
+}
+
 __device__ __forceinline__ f32x4 sc2_gemm_tile(const __hip_bfloat16* ldsA, int KP,
                                                const __hip_bfloat16* W, long ws_, int K, int N,
                                                int ncol0) {
@@ -2953,9 +3002,11 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
     float* __restrict__ ws2, int* __restrict__ ticket,
     int B, int SK, int A, int H, int N, int K, float eps, int hu_off) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int KP = (K + 31) & ~31;
-  __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;                  // [16][KP]
-  float* lds_sum = (float*)(smem + ((16 * KP * 2 + 15) & ~15));  // [32]
+  // +8-element row pad: a 2-KB-multiple row stride lands every A-fragment
+  // lane on the same LDS bank group (multi-way ds_read_b128 conflicts)
+  const int KP = (((K + 31) & ~31) + 8);
+  __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;                  // [16][KP] (unused in KSPLIT)
+  float* lds_sum = (float*)(smem + (KSPLIT ? 4096 : ((16 * KP * 2 + 15) & ~15)));  // [32]
   if (threadIdx.x < 32) lds_sum[threadIdx.x] = 0.f;
   if (RESETS) {
     // x = [(1-f) z_prev + f iz, (1-f) a]; rows >= B and cols >= K zero.
@@ -3005,7 +3056,7 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
       float hp = T0 ? 0.f : __bfloat162float(h_prev[(long)m * H + j]);
       st(hu_out, (long)m * hus + j, (1.f - fb) * hp + fb * __bfloat162float(ih[(long)m * H + j]));
     }
-  } else {
+  } else if (!KSPLIT) {
     if ((K & 7) == 0 && (as_ & 7) == 0) {
       sc2_stage_vec(a_in, as_, B, K, ldsA, KP);
     } else {
@@ -3015,6 +3066,8 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
       }
     }
     __syncthreads();
+  } else {
+    __syncthreads();  // lds_sum zero-fill visible before atomics
   }
   const int lane = threadIdx.x & 63;
   const int arow = lane & 15;
@@ -3026,10 +3079,9 @@ __global__ void __launch_bounds__(256) scan2_lnsilu_kernel(
     const int KQ = (((K + 3) / 4) + 31) & ~31;
     const int kbeg = min(wv * KQ, K);
     const int kend = min(kbeg + KQ, K);
-    acc = sc2_gemm_tile_range(ldsA, KP, W, K, K, N, ncol0, kbeg, kend);
-    // combine the 4 K-slices through LDS (ldsA is dead after the GEMM)
+    acc = sc2_gemm_tile_range_g(a_in, as_, B, W, K, K, N, ncol0, kbeg, kend);
     __syncthreads();
-    float* comb = (float*)smem;  // [4][16][16]
+    float* comb = (float*)smem;  // [4][16][16] = 4 KB
 #pragma unroll
     for (int r = 0; r < 4; ++r) comb[(wv * 16 + kgrp * 4 + r) * 16 + arow] = acc[r];
     __syncthreads();
@@ -3102,7 +3154,7 @@ __global__ void __launch_bounds__(256) scan2_gru_kernel(
     float* __restrict__ ws2, int* __restrict__ ticket, int B, int H, int D, float eps) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int K = H + D;
-  const int KP = (K + 31) & ~31;
+  const int KP = ((K + 31) & ~31) + 8;  // +8: LDS bank de-phasing
   __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;
   float* lds_sum = (float*)(smem + ((16 * KP * 2 + 15) & ~15));
   if (threadIdx.x < 32) lds_sum[threadIdx.x] = 0.f;
@@ -3185,7 +3237,7 @@ __global__ void __launch_bounds__(256) scan2_catst_kernel(
     float* __restrict__ m_out, __hip_bfloat16* __restrict__ z_out, float* __restrict__ s_out,
     int B, int P, int SK, int KD, float unimix) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int KP = (P + 31) & ~31;
+  const int KP = ((P + 31) & ~31) + 8;  // +8: LDS bank de-phasing
   __hip_bfloat16* ldsA = (__hip_bfloat16*)smem;
   float* raw = (float*)(smem + ((16 * KP * 2 + 15) & ~15));  // [16][64]
   if ((P & 7) == 0 && (ps & 7) == 0) {
@@ -3258,8 +3310,8 @@ __global__ void __launch_bounds__(256) scan2_b4_kernel(
     __hip_bfloat16* __restrict__ gp_out,          // [B, P] scratch
     int B, int SK, int P, int KD, float unimix) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int KP = SK;  // SK % 64 == 0
-  __hip_bfloat16* graw = (__hip_bfloat16*)smem;  // [16][SK]
+  const int KP = SK + 8;  // +8: LDS bank de-phasing (SK % 64 == 0)
+  __hip_bfloat16* graw = (__hip_bfloat16*)smem;  // [16][SK+8]
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   // wave-per-4-rows coalesced recompute; KD <= 64 groups align to lane
@@ -3324,8 +3376,8 @@ __global__ void __launch_bounds__(256) scan2_blnsilu_kernel(
     __hip_bfloat16* __restrict__ ga_out,                 // [B, A] (RESETS)
     int B, int P, int N, int SK, int A) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  const int KP = P;  // P % 64 == 0
-  __hip_bfloat16* gg = (__hip_bfloat16*)smem;                      // [16][P]
+  const int KP = P + 8;  // +8: LDS bank de-phasing (P % 64 == 0)
+  __hip_bfloat16* gg = (__hip_bfloat16*)smem;                      // [16][P+8]
   float* s12 = (float*)(smem + ((16 * KP * 2 + 15) & ~15));        // [2][16]
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
@@ -3383,9 +3435,9 @@ __global__ void __launch_bounds__(256) scan2_blnsilu_kernel(
     }
   }
   __syncthreads();
-  // pass B: finalize gg in place (vectorized)
-  for (int iv = threadIdx.x; iv < 16 * (KP >> 3); iv += blockDim.x) {
-    const int m = iv / (KP >> 3), jv = iv - m * (KP >> 3);
+  // pass B: finalize gg in place (vectorized; skip the pad columns)
+  for (int iv = threadIdx.x; iv < 16 * (P >> 3); iv += blockDim.x) {
+    const int m = iv / (P >> 3), jv = iv - m * (P >> 3);
     if (m >= B) continue;
     const int j0 = jv << 3;
     bf16x8 gzv = *(const bf16x8*)(gg + m * KP + j0);
@@ -3456,9 +3508,10 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
     int B, int H, int D) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int DD = 3 * H;
-  __hip_bfloat16* gybuf = (__hip_bfloat16*)smem;                     // [16][3H]
-  __hip_bfloat16* ghp = (__hip_bfloat16*)(smem + 16 * DD * 2);       // [16][H]
-  float* s12 = (float*)(smem + ((16 * DD * 2 + 16 * H * 2 + 15) & ~15));  // [2][16]
+  const int DDP = DD + 8;  // +8: LDS bank de-phasing
+  __hip_bfloat16* gybuf = (__hip_bfloat16*)smem;                     // [16][3H+8]
+  __hip_bfloat16* ghp = (__hip_bfloat16*)(smem + 16 * DDP * 2);      // [16][H]
+  float* s12 = (float*)(smem + ((16 * DDP * 2 + 16 * H * 2 + 15) & ~15));  // [2][16]
   const int lane = threadIdx.x & 63;
   const int wv = threadIdx.x >> 6;
   // pass 1, wave per 4 rows, 8-wide vector loads: full Hafner-gate backward
@@ -3514,15 +3567,15 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
           s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
         }
         *(bf16x8*)(ghp + m * H + j0) = ophp;
-        *(bf16x8*)(gybuf + m * DD + j0) = ozr;
-        *(bf16x8*)(gybuf + m * DD + H + j0) = ozc;
-        *(bf16x8*)(gybuf + m * DD + 2 * H + j0) = ozu;
+        *(bf16x8*)(gybuf + m * DDP + j0) = ozr;
+        *(bf16x8*)(gybuf + m * DDP + H + j0) = ozc;
+        *(bf16x8*)(gybuf + m * DDP + 2 * H + j0) = ozu;
       }
       s1 = wave_sum(s1);
       s2 = wave_sum(s2);
     } else {
       bf16x8 zero{};
-      for (int j0 = lane * 8; j0 < DD; j0 += 512) *(bf16x8*)(gybuf + m * DD + j0) = zero;
+      for (int j0 = lane * 8; j0 < DD; j0 += 512) *(bf16x8*)(gybuf + m * DDP + j0) = zero;
     }
     if (lane == 0) {
       s12[m] = s1 / DD;
@@ -3536,7 +3589,7 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
     if (threadIdx.x < 64) {
       float sw = 0.f, sb = 0.f;
       for (int m = 0; m < B; ++m) {
-        float gz = __bfloat162float(gybuf[m * DD + j]);
+        float gz = __bfloat162float(gybuf[m * DDP + j]);
         float xh = (__bfloat162float(y_in[(long)m * ys2 + j]) - mean[m]) * rstd[m];
         sw += gz * xh;
         sb += gz;
@@ -3551,7 +3604,7 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
     const int m = iv / (DD >> 3), jv = iv - m * (DD >> 3);
     if (m >= B) continue;
     const int j0 = jv << 3;
-    bf16x8 gzv = *(const bf16x8*)(gybuf + m * DD + j0);
+    bf16x8 gzv = *(const bf16x8*)(gybuf + m * DDP + j0);
     bf16x8 yv = *(const bf16x8*)(y_in + (long)m * ys2 + j0);
     bf16x8 lwv = *(const bf16x8*)(lnw + j0);
     bf16x8 outv;
@@ -3560,20 +3613,20 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
       float xh = ((float)yv[e] - mean[m]) * rstd[m];
       outv[e] = (__bf16)(((float)gzv[e] * (float)lwv[e] - s12[m] - xh * s12[16 + m]) * rstd[m]);
     }
-    *(bf16x8*)(gybuf + m * DD + j0) = outv;
+    *(bf16x8*)(gybuf + m * DDP + j0) = outv;
   }
   __syncthreads();
   for (int stripe = blockIdx.x; stripe < DD / 64; stripe += gridDim.x) {
     for (int i = threadIdx.x; i < B * 8; i += blockDim.x) {
       const int m = i / 8, jv = i & 7;
       const int j0 = stripe * 64 + jv * 8;
-      *(bf16x8*)(gy_out + (long)m * gys + j0) = *(const bf16x8*)(gybuf + m * DD + j0);
+      *(bf16x8*)(gy_out + (long)m * gys + j0) = *(const bf16x8*)(gybuf + m * DDP + j0);
     }
   }
   const int arow = lane & 15;
   const int kgrp = lane >> 4;
   const int ncol0 = (blockIdx.x * 4 + wv) * 16;
-  f32x4 acc = sc2_gemm_tile(gybuf, DD, W2t, DD, DD, H + D, ncol0);
+  f32x4 acc = sc2_gemm_tile(gybuf, DDP, W2t, DD, DD, H + D, ncol0);
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int m = kgrp * 4 + r;
@@ -3589,6 +3642,122 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
       }
     }
   }
+}
+
+// ---------------------------------------------------------------------------
+// fused reconstruction-loss NLL kernels (SURVEY.md §2.8 item 6 extension):
+// one reduction launch forward + one elementwise launch backward instead of
+// the autograd sub/pow/sum/cast chains of MSEDistribution / SymlogDistribution
+// / Bernoulli log_prob (sheeprl/utils/distribution.py:152-221 semantics).
+//   MODE 0: log_prob = -sum_d (pred - tgt)^2
+//   MODE 1: log_prob = -sum_d (pred - symlog(tgt))^2
+//   MODE 2: log_prob =  sum_d (tgt * pred - softplus(pred))   (Bernoulli logits)
+// ---------------------------------------------------------------------------
+
+template <typename T, int MODE>
+__global__ void nll_fwd_kernel(const T* __restrict__ pred, const float* __restrict__ tgt,
+                               float* __restrict__ out, long R, long D) {
+  const int lane = threadIdx.x & 63;
+  const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
+  for (long r = wave; r < R; r += nwaves) {
+    const T* pr = pred + r * D;
+    const float* tr = tgt + r * D;
+    float acc = 0.f;
+    for (long j = lane; j < D; j += 64) {
+      float p = ld(pr, j), t = tr[j];
+      if (MODE == 0) {
+        float d = p - t;
+        acc -= d * d;
+      } else if (MODE == 1) {
+        float ts = copysignf(logf(fabsf(t) + 1.f), t);
+        float d = p - ts;
+        acc -= d * d;
+      } else {
+        acc += t * p - (fmaxf(p, 0.f) + log1pf(expf(-fabsf(p))));
+      }
+    }
+    acc = wave_sum(acc);
+    if (lane == 0) out[r] = acc;
+  }
+}
+
+template <typename T, int MODE>
+__global__ void nll_bwd_kernel(const float* __restrict__ g, const T* __restrict__ pred,
+                               const float* __restrict__ tgt, T* __restrict__ gpred, long R, long D) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < R * D;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / D;
+    float p = ld(pred, i), t = tgt[i];
+    float gr = g[r];
+    float gv;
+    if (MODE == 0) {
+      gv = gr * -2.f * (p - t);
+    } else if (MODE == 1) {
+      float ts = copysignf(logf(fabsf(t) + 1.f), t);
+      gv = gr * -2.f * (p - ts);
+    } else {
+      gv = gr * (t - 1.f / (1.f + expf(-p)));
+    }
+    st(gpred, i, gv);
+  }
+}
+
+torch::Tensor nll_fwd(const torch::Tensor& pred, const torch::Tensor& tgt, long D, long mode) {
+  CHECK_IN(pred);
+  TORCH_CHECK(tgt.is_contiguous() && tgt.scalar_type() == at::kFloat && tgt.numel() == pred.numel());
+  long R = pred.numel() / D;
+  auto out = torch::empty({R}, pred.options().dtype(at::kFloat));
+  int blocks = (int)std::min((R + 3) / 4, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, pred.scalar_type(), "nll_fwd", [&] {
+    using T = scalar_t;
+    switch (mode) {
+      case 0:
+        hipLaunchKernelGGL((nll_fwd_kernel<T, 0>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                           (const T*)pred.data_ptr(), tgt.data_ptr<float>(), out.data_ptr<float>(), R, D);
+        break;
+      case 1:
+        hipLaunchKernelGGL((nll_fwd_kernel<T, 1>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                           (const T*)pred.data_ptr(), tgt.data_ptr<float>(), out.data_ptr<float>(), R, D);
+        break;
+      default:
+        hipLaunchKernelGGL((nll_fwd_kernel<T, 2>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                           (const T*)pred.data_ptr(), tgt.data_ptr<float>(), out.data_ptr<float>(), R, D);
+    }
+  });
+  return out;
+}
+
+torch::Tensor nll_bwd(const torch::Tensor& g, const torch::Tensor& pred, const torch::Tensor& tgt,
+                      long D, long mode) {
+  CHECK_IN(pred);
+  TORCH_CHECK(g.is_contiguous() && g.scalar_type() == at::kFloat);
+  long R = pred.numel() / D;
+  auto gpred = torch::empty_like(pred);
+  long n = R * D;
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, pred.scalar_type(), "nll_bwd", [&] {
+    using T = scalar_t;
+    switch (mode) {
+      case 0:
+        hipLaunchKernelGGL((nll_bwd_kernel<T, 0>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                           g.data_ptr<float>(), (const T*)pred.data_ptr(), tgt.data_ptr<float>(),
+                           (T*)gpred.data_ptr(), R, D);
+        break;
+      case 1:
+        hipLaunchKernelGGL((nll_bwd_kernel<T, 1>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                           g.data_ptr<float>(), (const T*)pred.data_ptr(), tgt.data_ptr<float>(),
+                           (T*)gpred.data_ptr(), R, D);
+        break;
+      default:
+        hipLaunchKernelGGL((nll_bwd_kernel<T, 2>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                           g.data_ptr<float>(), (const T*)pred.data_ptr(), tgt.data_ptr<float>(),
+                           (T*)gpred.data_ptr(), R, D);
+    }
+  });
+  return gpred;
 }
 
 // ---------------------------------------------------------------------------
@@ -3609,7 +3778,7 @@ void scan2_f1(const c10::optional<torch::Tensor>& z_prev, const torch::Tensor& i
   const int B = (int)act.size(0), A = (int)act.size(1), SK = (int)iz.size(1), H = (int)ih.size(1);
   const int N = (int)W1.size(0), K = (int)W1.size(1);
   TORCH_CHECK(B <= 16 && N % 64 == 0 && K == SK + A && SK % 8 == 0 && W1.is_contiguous());
-  const int KP = (K + 31) & ~31;
+  const int KP = ((K + 31) & ~31) + 8;
   const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
   const bool t0 = !z_prev.has_value();
@@ -3636,14 +3805,15 @@ void scan2_f3(const torch::Tensor& a_in, const torch::Tensor& W3, const torch::T
   const int B = (int)a_in.size(0);
   const int N = (int)W3.size(0), K = (int)W3.size(1);
   TORCH_CHECK(B <= 16 && N % 64 == 0 && a_in.size(1) == K && W3.is_contiguous());
-  const int KP = (K + 31) & ~31;
-  const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
+  const int KP = ((K + 31) & ~31) + 8;
+  // long-K shapes (the representation GEMM, K = H+E ~ 4.6k at S): K-split the
+  // 4 waves over ONE 16-col tile over GLOBAL A and run N/16 workgroups — 4x
+  // the weight-stream concurrency, no LDS staging
+  const bool ksplit = K >= 3072 && N / 16 <= 256 && (K % 8) == 0 && (a_in.stride(0) % 8) == 0;
+  const size_t shmem = ksplit ? (4096 + 32 * sizeof(float))
+                              : ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "scan2_f3: K too large for LDS");
   auto stream = at::cuda::getCurrentCUDAStream();
-  // long-K shapes (the representation GEMM, K = H+E ~ 4.6k at S): K-split the
-  // 4 waves over ONE 16-col tile and run N/16 workgroups — 4x the
-  // weight-stream concurrency of the 4-tile layout
-  const bool ksplit = K >= 3072 && N / 16 <= 256;
   if (ksplit)
     hipLaunchKernelGGL((scan2_lnsilu_kernel<false, false, true>), dim3(N / 16), dim3(256), shmem,
                        stream.stream(),
@@ -3668,7 +3838,7 @@ void scan2_f2(const torch::Tensor& hu, const torch::Tensor& W2, const torch::Ten
   const int H = (int)h_out.size(1), D = K - H;
   TORCH_CHECK(B <= 16 && H % 64 == 0 && W2.size(0) == 3 * H && W2.size(1) == K && W2.is_contiguous());
   TORCH_CHECK(h_out.is_contiguous());
-  const int KP = (K + 31) & ~31;
+  const int KP = ((K + 31) & ~31) + 8;
   const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 32 * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(scan2_gru_kernel, dim3(H / 64), dim3(256), shmem, stream.stream(), sc2_bp(hu),
@@ -3685,7 +3855,7 @@ void scan2_f4(const torch::Tensor& p_in, const torch::Tensor& W4, const torch::T
   const int SK = (int)W4.size(0);
   TORCH_CHECK(B <= 16 && SK % 64 == 0 && W4.size(1) == P && W4.is_contiguous());
   TORCH_CHECK(KD <= 64 && 64 % KD == 0 && m_out.is_contiguous() && z_out.is_contiguous() && s_out.is_contiguous());
-  const int KP = (P + 31) & ~31;
+  const int KP = ((P + 31) & ~31) + 8;
   const size_t shmem = ((16 * KP * 2 + 15) & ~15) + 16 * 64 * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(scan2_catst_kernel, dim3(SK / 64), dim3(256), shmem, stream.stream(), sc2_bp(p_in),
@@ -3701,7 +3871,7 @@ void scan2_b4(const torch::Tensor& gm, const torch::Tensor& gon, const c10::opti
   const int P = (int)W4t.size(0);
   TORCH_CHECK(B <= 16 && SK % 64 == 0 && P % 64 == 0 && W4t.size(1) == SK && W4t.is_contiguous());
   TORCH_CHECK(KD <= 64 && (KD & (KD - 1)) == 0 && gp_out.is_contiguous());
-  const size_t shmem = 16 * SK * 2;
+  const size_t shmem = 16 * (SK + 8) * 2;
   auto stream = at::cuda::getCurrentCUDAStream();
   const __hip_bfloat16* g2 = gon2.has_value() ? sc2_bp(*gon2) : nullptr;
   hipLaunchKernelGGL(scan2_b4_kernel, dim3(P / 64), dim3(256), shmem, stream.stream(),
@@ -3719,7 +3889,7 @@ void scan2_b3(const torch::Tensor& gy_in, const torch::Tensor& g_in, const torch
   TORCH_CHECK(B <= 16 && P % 64 == 0 && Wt.size(1) == P && Wt.is_contiguous());
   TORCH_CHECK(gy_in.stride(0) % 8 == 0 && g_in.stride(0) % 8 == 0 && gg_out.stride(0) % 8 == 0,
               "scan2 blnsilu: 16B-aligned rows required");
-  const size_t shmem = ((16 * P * 2 + 15) & ~15) + 32 * sizeof(float);
+  const size_t shmem = ((16 * (P + 8) * 2 + 15) & ~15) + 32 * sizeof(float);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL((scan2_blnsilu_kernel<false>), dim3((N + 63) / 64), dim3(256), shmem, stream.stream(),
                      sc2_bp(gy_in), gy_in.stride(0), sc2_bp(g_in), g_in.stride(0), sc2_bp(lnw),
@@ -3757,7 +3927,7 @@ void scan2_b2(const torch::Tensor& gh, const c10::optional<torch::Tensor>& gh2, 
   TORCH_CHECK(gh.is_contiguous() && gh_carry.is_contiguous() && ghu_out.is_contiguous());
   TORCH_CHECK(gh3.stride(0) % 8 == 0 && y_in.stride(0) % 8 == 0 && hu.stride(0) % 8 == 0 &&
               gy_out.stride(0) % 8 == 0, "scan2_b2: 16B-aligned rows required");
-  const size_t shmem = ((16 * 3 * H * 2 + 16 * H * 2 + 15) & ~15) + 32 * sizeof(float);
+  const size_t shmem = ((16 * (3 * H + 8) * 2 + 16 * H * 2 + 15) & ~15) + 32 * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "scan2_b2: H too large for LDS");
   auto stream = at::cuda::getCurrentCUDAStream();
   const __hip_bfloat16* g2 = gh2.has_value() ? sc2_bp(*gh2) : nullptr;
@@ -3807,6 +3977,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("twohot_lp_bwd", &twohot_lp_bwd);
   m.def("klbal_fwd", &klbal_fwd);
   m.def("klbal_bwd", &klbal_bwd);
+  m.def("nll_fwd", &nll_fwd);
+  m.def("nll_bwd", &nll_bwd);
   m.def("scan2_f1", &scan2_f1);
   m.def("scan2_f2", &scan2_f2);
   m.def("scan2_f3", &scan2_f3);

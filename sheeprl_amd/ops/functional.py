@@ -222,3 +222,45 @@ def lambda_values(rewards: Tensor, values: Tensor, continues: Tensor, lmbda: flo
     c = continues.reshape(shape[0], -1)
     out = _LambdaValues.apply(r, v, c, lmbda)
     return out.reshape(shape)
+
+
+# ---------------------------------------------------------------------------
+# fused reconstruction-loss log_probs (one reduction fwd + one elementwise
+# bwd instead of the autograd sub/pow/sum chains)
+# ---------------------------------------------------------------------------
+
+class _FusedNLL(torch.autograd.Function):
+    """log_prob kernels over trailing dims D of a contiguous pred tensor:
+    mode 0 = MSE (-sum (p-t)^2), mode 1 = symlog-MSE, mode 2 = Bernoulli
+    logits (sum (t*p - softplus(p)))."""
+
+    @staticmethod
+    def forward(ctx, pred: Tensor, target: Tensor, dims: int, mode: int) -> Tensor:
+        D = 1
+        for s in pred.shape[len(pred.shape) - dims:]:
+            D *= s
+        predc = pred.contiguous()
+        tgt = target.detach().float().contiguous()
+        ctx.save_for_backward(predc, tgt)
+        ctx.D = D
+        ctx.mode = mode
+        out = require_ext().nll_fwd(predc, tgt, D, mode)
+        return out.view(pred.shape[: len(pred.shape) - dims])
+
+    @staticmethod
+    def backward(ctx, gy: Tensor):
+        pred, tgt = ctx.saved_tensors
+        gpred = require_ext().nll_bwd(gy.contiguous().float().view(-1), pred, tgt, ctx.D, ctx.mode)
+        return gpred, None, None, None
+
+
+def mse_log_prob(pred: Tensor, target: Tensor, dims: int) -> Tensor:
+    return _FusedNLL.apply(pred, target, dims, 0)
+
+
+def symlog_mse_log_prob(pred: Tensor, target: Tensor, dims: int) -> Tensor:
+    return _FusedNLL.apply(pred, target, dims, 1)
+
+
+def bernoulli_log_prob(logits: Tensor, target: Tensor, dims: int) -> Tensor:
+    return _FusedNLL.apply(logits, target, dims, 2)
