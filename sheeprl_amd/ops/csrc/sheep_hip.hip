@@ -3645,6 +3645,224 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// fused behaviour-learning losses (the DV3 actor/critic loss sections,
+// sheeprl/algos/dreamer_v3/dreamer_v3.py:262-325): each is ONE kernel per
+// direction instead of the ~30-launch autograd elementwise chain.
+// ---------------------------------------------------------------------------
+
+// mean of a TwoHotEncodingDistribution: symexp(sum softmax(logits) * bins)
+// (inference-only: used where the value is consumed detached)
+template <typename T>
+__global__ void twohot_mean_kernel(const T* __restrict__ logits, float* __restrict__ out, long N, int K,
+                                   float low, float high) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= N) return;
+  const T* lr = logits + row * (long)K;
+  float mx = -1e30f;
+  for (int j = lane; j < K; j += 64) mx = fmaxf(mx, ld(lr, j));
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+  const float step = (high - low) / (K - 1);
+  float se = 0.f, dot = 0.f;
+  for (int j = lane; j < K; j += 64) {
+    float e = expf(ld(lr, j) - mx);
+    se += e;
+    dot += e * (low + j * step);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    se += __shfl_xor(se, off, 64);
+    dot += __shfl_xor(dot, off, 64);
+  }
+  if (lane == 0) {
+    float m = dot / se;
+    out[row] = (m >= 0.f ? 1.f : -1.f) * (expf(fabsf(m)) - 1.f);  // symexp
+  }
+}
+
+torch::Tensor twohot_mean(const torch::Tensor& logits, double low, double high) {
+  CHECK_IN(logits);
+  int K = (int)logits.size(-1);
+  long N = logits.numel() / K;
+  auto out = torch::empty({N}, logits.options().dtype(at::kFloat));
+  const int rpb = kBlock / 64;
+  int blocks = (int)((N + rpb - 1) / rpb);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, logits.scalar_type(), "twohot_mean", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(twohot_mean_kernel<T>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)logits.data_ptr(), out.data_ptr<float>(), N, K, (float)low, (float)high);
+  });
+  return out;
+}
+
+// REINFORCE actor loss (discrete, single head):
+//   loss = -(1/(HZ*F)) sum_{t<HZ,f} disc * (sum_a m*act*adv - ent_coef * sum_a e^m m)
+// m are the fused categorical head's NORMALIZED log-probs.
+template <typename T>
+__global__ void reinforce_fwd_kernel(const float* __restrict__ m, const T* __restrict__ act,
+                                     const float* __restrict__ adv, const float* __restrict__ disc,
+                                     float* __restrict__ out, long HZF, int A, float ent_coef) {
+  __shared__ float lds[9];
+  float acc = 0.f;
+  for (long r = blockIdx.x * (long)blockDim.x + threadIdx.x; r < HZF;
+       r += (long)gridDim.x * blockDim.x) {
+    const float* mr = m + r * A;
+    const T* ar = act + r * A;
+    float logp = 0.f, ent = 0.f;
+    for (int a = 0; a < A; ++a) {
+      float mv = mr[a];
+      logp += mv * ld(ar, a);
+      ent -= expf(mv) * mv;
+    }
+    acc += disc[r] * (logp * adv[r] + ent_coef * ent);
+  }
+  acc = block_sum(acc, lds);
+  if (threadIdx.x == 0) atomicAdd(out, -acc / HZF);
+}
+
+template <typename T>
+__global__ void reinforce_bwd_kernel(const float* __restrict__ g, const float* __restrict__ m,
+                                     const T* __restrict__ act, const float* __restrict__ adv,
+                                     const float* __restrict__ disc, float* __restrict__ gm, long HZF,
+                                     long TOT, int A, float ent_coef) {
+  const float gs = g[0];
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < TOT * A;
+       i += (long)gridDim.x * blockDim.x) {
+    const long r = i / A;
+    if (r >= HZF) {
+      gm[i] = 0.f;  // the t = HZ row contributes nothing
+      continue;
+    }
+    float mv = m[i];
+    float dent = -expf(mv) * (1.f + mv);
+    gm[i] = gs * (-1.f / HZF) * disc[r] * (ld(act, i) * adv[r] + ent_coef * dent);
+  }
+}
+
+std::vector<torch::Tensor> reinforce_fwd(const torch::Tensor& m, const torch::Tensor& act,
+                                         const torch::Tensor& adv, const torch::Tensor& disc, long HZF,
+                                         double ent_coef) {
+  CHECK_IN(m);
+  TORCH_CHECK(m.scalar_type() == at::kFloat && adv.is_contiguous() && disc.is_contiguous());
+  int A = (int)m.size(-1);
+  auto out = torch::zeros({}, m.options());
+  int blocks = (int)std::min((HZF + kBlock - 1) / kBlock, (long)1024);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, act.scalar_type(), "reinforce_fwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(reinforce_fwd_kernel<T>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       m.data_ptr<float>(), (const T*)act.data_ptr(), adv.data_ptr<float>(),
+                       disc.data_ptr<float>(), out.data_ptr<float>(), HZF, A, (float)ent_coef);
+  });
+  return {out};
+}
+
+torch::Tensor reinforce_bwd(const torch::Tensor& g, const torch::Tensor& m, const torch::Tensor& act,
+                            const torch::Tensor& adv, const torch::Tensor& disc, long HZF,
+                            double ent_coef) {
+  int A = (int)m.size(-1);
+  long TOT = m.numel() / A;
+  auto gm = torch::empty_like(m);
+  int blocks = (int)std::min((TOT * A + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, act.scalar_type(), "reinforce_bwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(reinforce_bwd_kernel<T>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       g.data_ptr<float>(), m.data_ptr<float>(), (const T*)act.data_ptr(),
+                       adv.data_ptr<float>(), disc.data_ptr<float>(), gm.data_ptr<float>(), HZF, TOT, A,
+                       (float)ent_coef);
+  });
+  return gm;
+}
+
+// critic loss: mean_r disc[r] * (-lp(t1[r]) - lp(t2[r])) over shared logits
+// (two two-hot cross-entropies per row in one pass; bwd recomputes softmax)
+__global__ void vloss2_fwd_kernel(const float* __restrict__ logits, const float* __restrict__ t1,
+                                  const float* __restrict__ t2, const float* __restrict__ disc,
+                                  float* __restrict__ out, float* __restrict__ lse_out, long N, int K,
+                                  float low, float high) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= N) return;
+  const float* lr = logits + row * (long)K;
+  float mx = -1e30f;
+  for (int j = lane; j < K; j += 64) mx = fmaxf(mx, lr[j]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+  float se = 0.f;
+  for (int j = lane; j < K; j += 64) se += expf(lr[j] - mx);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) se += __shfl_xor(se, off, 64);
+  const float lse = mx + logf(se);
+  if (lane == 0) {
+    int lo1, hi1, lo2, hi2;
+    float wl1, wh1, wl2, wh2;
+    twohot_idx(t1[row], low, high, K, lo1, hi1, wl1, wh1);
+    twohot_idx(t2[row], low, high, K, lo2, hi2, wl2, wh2);
+    float lp1 = wl1 * (lr[lo1] - lse) + wh1 * (lr[hi1] - lse);
+    float lp2 = wl2 * (lr[lo2] - lse) + wh2 * (lr[hi2] - lse);
+    lse_out[row] = lse;
+    atomicAdd(out, disc[row] * (-(lp1)-lp2) / N);
+  }
+}
+
+__global__ void vloss2_bwd_kernel(const float* __restrict__ g, const float* __restrict__ logits,
+                                  const float* __restrict__ t1, const float* __restrict__ t2,
+                                  const float* __restrict__ disc, const float* __restrict__ lse,
+                                  float* __restrict__ gl, long N, int K, float low, float high) {
+  const float gs = g[0];
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < N * (long)K;
+       i += (long)gridDim.x * blockDim.x) {
+    const long row = i / K;
+    const int j = (int)(i - row * K);
+    int lo1, hi1, lo2, hi2;
+    float wl1, wh1, wl2, wh2;
+    twohot_idx(t1[row], low, high, K, lo1, hi1, wl1, wh1);
+    twohot_idx(t2[row], low, high, K, lo2, hi2, wl2, wh2);
+    const float tw = (j == lo1 ? wl1 : 0.f) + (j == hi1 ? wh1 : 0.f) + (j == lo2 ? wl2 : 0.f) +
+                     (j == hi2 ? wh2 : 0.f);
+    gl[i] = gs * disc[row] * (2.f * expf(logits[i] - lse[row]) - tw) / N;
+  }
+}
+
+std::vector<torch::Tensor> vloss2_fwd(const torch::Tensor& logits, const torch::Tensor& t1,
+                                      const torch::Tensor& t2, const torch::Tensor& disc, double low,
+                                      double high) {
+  CHECK_IN(logits);
+  TORCH_CHECK(logits.scalar_type() == at::kFloat && t1.is_contiguous() && t2.is_contiguous() &&
+              disc.is_contiguous());
+  int K = (int)logits.size(-1);
+  long N = logits.numel() / K;
+  auto out = torch::zeros({}, logits.options());
+  auto lse = torch::empty({N}, logits.options());
+  const int rpb = kBlock / 64;
+  int blocks = (int)((N + rpb - 1) / rpb);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(vloss2_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     logits.data_ptr<float>(), t1.data_ptr<float>(), t2.data_ptr<float>(),
+                     disc.data_ptr<float>(), out.data_ptr<float>(), lse.data_ptr<float>(), N, K,
+                     (float)low, (float)high);
+  return {out, lse};
+}
+
+torch::Tensor vloss2_bwd(const torch::Tensor& g, const torch::Tensor& logits, const torch::Tensor& t1,
+                         const torch::Tensor& t2, const torch::Tensor& disc, const torch::Tensor& lse,
+                         double low, double high) {
+  int K = (int)logits.size(-1);
+  long N = logits.numel() / K;
+  auto gl = torch::empty_like(logits);
+  int blocks = (int)std::min((N * K + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(vloss2_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     g.data_ptr<float>(), logits.data_ptr<float>(), t1.data_ptr<float>(),
+                     t2.data_ptr<float>(), disc.data_ptr<float>(), lse.data_ptr<float>(),
+                     gl.data_ptr<float>(), N, K, (float)low, (float)high);
+  return gl;
+}
+
+// ---------------------------------------------------------------------------
 // fused reconstruction-loss NLL kernels (SURVEY.md §2.8 item 6 extension):
 // one reduction launch forward + one elementwise launch backward instead of
 // the autograd sub/pow/sum/cast chains of MSEDistribution / SymlogDistribution
@@ -3979,6 +4197,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("klbal_bwd", &klbal_bwd);
   m.def("nll_fwd", &nll_fwd);
   m.def("nll_bwd", &nll_bwd);
+  m.def("twohot_mean", &twohot_mean);
+  m.def("reinforce_fwd", &reinforce_fwd);
+  m.def("reinforce_bwd", &reinforce_bwd);
+  m.def("vloss2_fwd", &vloss2_fwd);
+  m.def("vloss2_bwd", &vloss2_bwd);
   m.def("scan2_f1", &scan2_f1);
   m.def("scan2_f2", &scan2_f2);
   m.def("scan2_f3", &scan2_f3);
