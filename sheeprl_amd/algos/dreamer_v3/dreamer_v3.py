@@ -221,68 +221,115 @@ def train(
             actions = torch.cat(actor(imagined_latent_state.detach())[0], dim=-1).to(dtype)
             imagined_actions[i] = actions
 
-    predicted_values = TwoHotEncodingDistribution(critic(imagined_trajectories).float(), dims=1).mean
-    predicted_rewards = TwoHotEncodingDistribution(world_model.reward_model(imagined_trajectories).float(), dims=1).mean
-    continues = td.Independent(
-        BernoulliSafeMode(logits=world_model.continue_model(imagined_trajectories).float()), 1
-    ).mode
-    true_continue = (1 - data["terminated"]).flatten().reshape(1, -1, 1)
-    continues = torch.cat((true_continue, continues[1:]))
+    # fast behaviour-loss path (discrete single-head on HIP): for a REINFORCE
+    # actor nothing backpropagates through the imagined values/returns, so the
+    # value/reward/continue heads run inference-only with fused twohot means,
+    # and each loss is ONE kernel per direction instead of the autograd chain
+    fast_losses = use_fast_imagine and len(actions_dim) == 1
+    if fast_losses:
+        with torch.no_grad():
+            predicted_values = ops.twohot_mean(critic(imagined_trajectories))
+            predicted_rewards = ops.twohot_mean(world_model.reward_model(imagined_trajectories))
+            continues = (world_model.continue_model(imagined_trajectories) > 0).float()
+            true_continue = (1 - data["terminated"]).flatten().reshape(1, -1, 1)
+            continues = torch.cat((true_continue, continues[1:]))
+            lambda_values = compute_lambda_values(
+                predicted_rewards[1:], predicted_values[1:], continues[1:] * cfg.algo.gamma,
+                lmbda=cfg.algo.lmbda,
+            )
+            discount = torch.cumprod(continues * cfg.algo.gamma, dim=0) / cfg.algo.gamma
 
-    lambda_values = compute_lambda_values(
-        predicted_rewards[1:], predicted_values[1:], continues[1:] * cfg.algo.gamma, lmbda=cfg.algo.lmbda
-    )
-
-    with torch.no_grad():
-        discount = torch.cumprod(continues * cfg.algo.gamma, dim=0) / cfg.algo.gamma
-
-    # actor loss
-    actor_optimizer.zero_grad(set_to_none=True)
-    policies = actor(imagined_trajectories.detach())[1]
-    baseline = predicted_values[:-1]
-    offset, invscale = moments(lambda_values, runtime)
-    normed_lambda_values = (lambda_values - offset) / invscale
-    normed_baseline = (baseline - offset) / invscale
-    advantage = normed_lambda_values - normed_baseline
-    if is_continuous:
-        objective = advantage
-    else:
-        objective = (
-            torch.stack(
-                [
-                    p.log_prob(imgnd_act.detach().float()).unsqueeze(-1)[:-1]
-                    for p, imgnd_act in zip(policies, torch.split(imagined_actions, list(actions_dim), dim=-1))
-                ],
-                dim=-1,
-            ).sum(dim=-1)
-            * advantage.detach()
+        # actor loss (advantage offsets cancel: (λ-off)/s - (v-off)/s = (λ-v)/s)
+        actor_optimizer.zero_grad(set_to_none=True)
+        policies = actor(imagined_trajectories.detach())[1]
+        offset, invscale = moments(lambda_values, runtime)
+        with torch.no_grad():
+            advantage = (lambda_values - predicted_values[:-1]) / invscale
+        policy_loss = ops.reinforce_loss(
+            policies[0].logits, imagined_actions, advantage, discount[:-1], cfg.algo.actor.ent_coef
         )
-    try:
-        entropy = cfg.algo.actor.ent_coef * torch.stack([p.entropy() for p in policies], -1).sum(dim=-1)
-    except NotImplementedError:
-        entropy = torch.zeros_like(objective)
-    policy_loss = -torch.mean(discount[:-1].detach() * (objective + entropy.unsqueeze(dim=-1)[:-1]))
-    runtime.backward(policy_loss)
-    actor_grads = None
-    if cfg.algo.actor.clip_gradients and cfg.algo.actor.clip_gradients > 0:
-        actor_grads = runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
-    actor_optimizer.step()
+        runtime.backward(policy_loss)
+        actor_grads = None
+        if cfg.algo.actor.clip_gradients and cfg.algo.actor.clip_gradients > 0:
+            actor_grads = runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
+        actor_optimizer.step()
 
-    # critic loss (with EMA-critic regularizer)
-    qv = TwoHotEncodingDistribution(critic(imagined_trajectories.detach()[:-1]).float(), dims=1)
-    predicted_target_values = TwoHotEncodingDistribution(
-        target_critic(imagined_trajectories.detach()[:-1]).float(), dims=1
-    ).mean
+        # critic loss: two two-hot CEs over shared logits in one kernel
+        critic_optimizer.zero_grad(set_to_none=True)
+        qv_logits = critic(imagined_trajectories.detach()[:-1]).float()
+        with torch.no_grad():
+            predicted_target_values = ops.twohot_mean(target_critic(imagined_trajectories.detach()[:-1]))
+        value_loss = ops.critic_twohot_loss(
+            qv_logits, lambda_values, predicted_target_values, discount[:-1]
+        )
+        runtime.backward(value_loss)
+        critic_grads = None
+        if cfg.algo.critic.clip_gradients and cfg.algo.critic.clip_gradients > 0:
+            critic_grads = runtime.clip_gradients(critic, critic_optimizer, cfg.algo.critic.clip_gradients)
+        critic_optimizer.step()
+    else:
+        predicted_values = TwoHotEncodingDistribution(critic(imagined_trajectories).float(), dims=1).mean
+        predicted_rewards = TwoHotEncodingDistribution(world_model.reward_model(imagined_trajectories).float(), dims=1).mean
+        continues = td.Independent(
+            BernoulliSafeMode(logits=world_model.continue_model(imagined_trajectories).float()), 1
+        ).mode
+        true_continue = (1 - data["terminated"]).flatten().reshape(1, -1, 1)
+        continues = torch.cat((true_continue, continues[1:]))
 
-    critic_optimizer.zero_grad(set_to_none=True)
-    value_loss = -qv.log_prob(lambda_values.detach())
-    value_loss = value_loss - qv.log_prob(predicted_target_values.detach())
-    value_loss = torch.mean(value_loss * discount[:-1].squeeze(-1))
-    runtime.backward(value_loss)
-    critic_grads = None
-    if cfg.algo.critic.clip_gradients and cfg.algo.critic.clip_gradients > 0:
-        critic_grads = runtime.clip_gradients(critic, critic_optimizer, cfg.algo.critic.clip_gradients)
-    critic_optimizer.step()
+        lambda_values = compute_lambda_values(
+            predicted_rewards[1:], predicted_values[1:], continues[1:] * cfg.algo.gamma, lmbda=cfg.algo.lmbda
+        )
+
+        with torch.no_grad():
+            discount = torch.cumprod(continues * cfg.algo.gamma, dim=0) / cfg.algo.gamma
+
+        # actor loss
+        actor_optimizer.zero_grad(set_to_none=True)
+        policies = actor(imagined_trajectories.detach())[1]
+        baseline = predicted_values[:-1]
+        offset, invscale = moments(lambda_values, runtime)
+        normed_lambda_values = (lambda_values - offset) / invscale
+        normed_baseline = (baseline - offset) / invscale
+        advantage = normed_lambda_values - normed_baseline
+        if is_continuous:
+            objective = advantage
+        else:
+            objective = (
+                torch.stack(
+                    [
+                        p.log_prob(imgnd_act.detach().float()).unsqueeze(-1)[:-1]
+                        for p, imgnd_act in zip(policies, torch.split(imagined_actions, list(actions_dim), dim=-1))
+                    ],
+                    dim=-1,
+                ).sum(dim=-1)
+                * advantage.detach()
+            )
+        try:
+            entropy = cfg.algo.actor.ent_coef * torch.stack([p.entropy() for p in policies], -1).sum(dim=-1)
+        except NotImplementedError:
+            entropy = torch.zeros_like(objective)
+        policy_loss = -torch.mean(discount[:-1].detach() * (objective + entropy.unsqueeze(dim=-1)[:-1]))
+        runtime.backward(policy_loss)
+        actor_grads = None
+        if cfg.algo.actor.clip_gradients and cfg.algo.actor.clip_gradients > 0:
+            actor_grads = runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
+        actor_optimizer.step()
+
+        # critic loss (with EMA-critic regularizer)
+        qv = TwoHotEncodingDistribution(critic(imagined_trajectories.detach()[:-1]).float(), dims=1)
+        predicted_target_values = TwoHotEncodingDistribution(
+            target_critic(imagined_trajectories.detach()[:-1]).float(), dims=1
+        ).mean
+
+        critic_optimizer.zero_grad(set_to_none=True)
+        value_loss = -qv.log_prob(lambda_values.detach())
+        value_loss = value_loss - qv.log_prob(predicted_target_values.detach())
+        value_loss = torch.mean(value_loss * discount[:-1].squeeze(-1))
+        runtime.backward(value_loss)
+        critic_grads = None
+        if cfg.algo.critic.clip_gradients and cfg.algo.critic.clip_gradients > 0:
+            critic_grads = runtime.clip_gradients(critic, critic_optimizer, cfg.algo.critic.clip_gradients)
+        critic_optimizer.step()
 
     if aggregator and not MetricAggregator.disabled:
         aggregator.update("Loss/world_model_loss", rec_loss.detach())

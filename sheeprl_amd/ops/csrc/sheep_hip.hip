@@ -3645,6 +3645,45 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// tiled 2-D transpose (64x64 LDS tiles, padded rows): torch's .t().contiguous()
+// on bf16 weights is an uncoalesced 2-byte strided copy; the scan2 backward
+// transposes ~10 MB of weights per step and needs this to be bandwidth-bound.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void transpose2d_kernel(const T* __restrict__ in, T* __restrict__ out, int R, int C) {
+  __shared__ T tile[64][72];  // +8 pad: conflict-free transposed reads
+  const int tr = blockIdx.y * 64;
+  const int tc = blockIdx.x * 64;
+  for (int i = threadIdx.x; i < 64 * 64; i += blockDim.x) {
+    const int r = i >> 6, c = i & 63;
+    T v = (T)0;
+    if (tr + r < R && tc + c < C) v = in[(long)(tr + r) * C + tc + c];
+    tile[r][c] = v;
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 64 * 64; i += blockDim.x) {
+    const int r2 = i >> 6, c2 = i & 63;  // output row tile = input cols
+    if (tc + r2 < C && tr + c2 < R) out[(long)(tc + r2) * R + tr + c2] = tile[c2][r2];
+  }
+}
+
+torch::Tensor transpose2d(const torch::Tensor& in) {
+  CHECK_IN(in);
+  TORCH_CHECK(in.dim() == 2);
+  const int R = (int)in.size(0), C = (int)in.size(1);
+  auto out = torch::empty({C, R}, in.options());
+  dim3 grid((C + 63) / 64, (R + 63) / 64);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, in.scalar_type(), "transpose2d", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL(transpose2d_kernel<T>, grid, dim3(256), 0, stream.stream(),
+                       (const T*)in.data_ptr(), (T*)out.data_ptr(), R, C);
+  });
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // fused behaviour-learning losses (the DV3 actor/critic loss sections,
 // sheeprl/algos/dreamer_v3/dreamer_v3.py:262-325): each is ONE kernel per
 // direction instead of the ~30-launch autograd elementwise chain.
@@ -4195,6 +4234,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("twohot_lp_bwd", &twohot_lp_bwd);
   m.def("klbal_fwd", &klbal_fwd);
   m.def("klbal_bwd", &klbal_bwd);
+  m.def("transpose2d", &transpose2d);
   m.def("nll_fwd", &nll_fwd);
   m.def("nll_bwd", &nll_bwd);
   m.def("twohot_mean", &twohot_mean);

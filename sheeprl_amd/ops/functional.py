@@ -264,3 +264,82 @@ def symlog_mse_log_prob(pred: Tensor, target: Tensor, dims: int) -> Tensor:
 
 def bernoulli_log_prob(logits: Tensor, target: Tensor, dims: int) -> Tensor:
     return _FusedNLL.apply(logits, target, dims, 2)
+
+
+# ---------------------------------------------------------------------------
+# fused behaviour-learning losses (DV3 actor/critic sections)
+# ---------------------------------------------------------------------------
+
+def twohot_mean(logits: Tensor, low: float = -20.0, high: float = 20.0) -> Tensor:
+    """Inference-only mean of a TwoHotEncodingDistribution:
+    symexp(softmax(logits) @ bins).  Returns [..., 1] fp32."""
+    assert not logits.requires_grad, "twohot_mean is inference-only"
+    out = require_ext().twohot_mean(logits.contiguous(), low, high)
+    return out.view(*logits.shape[:-1], 1)
+
+
+class _ReinforceLoss(torch.autograd.Function):
+    """Discrete single-head DV3 actor loss:
+    -(1/(HZ*F)) sum disc * (logp(a) * adv + ent_coef * entropy), with
+    normalized log-probs ``m`` [HZ+1, F, A] (the t=HZ row is unused, matching
+    the reference's [:-1] slicing)."""
+
+    @staticmethod
+    def forward(ctx, m: Tensor, act: Tensor, adv: Tensor, disc: Tensor, ent_coef: float) -> Tensor:
+        A = m.shape[-1]
+        HZF = adv.numel()
+        mc = m.contiguous()
+        actc = act.detach().contiguous()
+        advc = adv.detach().float().reshape(-1).contiguous()
+        discc = disc.detach().float().reshape(-1).contiguous()
+        (out,) = require_ext().reinforce_fwd(mc.view(-1, A), actc.view(-1, A), advc, discc, HZF, ent_coef)
+        ctx.save_for_backward(mc, actc, advc, discc)
+        ctx.meta = (HZF, ent_coef)
+        return out
+
+    @staticmethod
+    def backward(ctx, gy: Tensor):
+        m, act, adv, disc = ctx.saved_tensors
+        HZF, ent_coef = ctx.meta
+        A = m.shape[-1]
+        gm = require_ext().reinforce_bwd(
+            gy.contiguous().float().view(1), m.view(-1, A), act.view(-1, A), adv, disc, HZF, ent_coef
+        )
+        return gm.view(m.shape), None, None, None, None
+
+
+def reinforce_loss(m: Tensor, act: Tensor, adv: Tensor, disc: Tensor, ent_coef: float) -> Tensor:
+    return _ReinforceLoss.apply(m, act, adv, disc, ent_coef)
+
+
+class _CriticTwohotLoss(torch.autograd.Function):
+    """mean(disc * (-log_prob(t1) - log_prob(t2))) over shared two-hot critic
+    logits [R, K] (fp32)."""
+
+    @staticmethod
+    def forward(ctx, logits: Tensor, t1: Tensor, t2: Tensor, disc: Tensor,
+                low: float, high: float) -> Tensor:
+        K = logits.shape[-1]
+        lc = logits.contiguous()
+        t1c = t1.detach().float().reshape(-1).contiguous()
+        t2c = t2.detach().float().reshape(-1).contiguous()
+        dc = disc.detach().float().reshape(-1).contiguous()
+        out, lse = require_ext().vloss2_fwd(lc.view(-1, K), t1c, t2c, dc, low, high)
+        ctx.save_for_backward(lc, t1c, t2c, dc, lse)
+        ctx.meta = (low, high)
+        return out
+
+    @staticmethod
+    def backward(ctx, gy: Tensor):
+        logits, t1, t2, disc, lse = ctx.saved_tensors
+        low, high = ctx.meta
+        K = logits.shape[-1]
+        gl = require_ext().vloss2_bwd(
+            gy.contiguous().float().view(1), logits.view(-1, K), t1, t2, disc, lse, low, high
+        )
+        return gl.view(logits.shape), None, None, None, None, None
+
+
+def critic_twohot_loss(logits: Tensor, t1: Tensor, t2: Tensor, disc: Tensor,
+                       low: float = -20.0, high: float = 20.0) -> Tensor:
+    return _CriticTwohotLoss.apply(logits, t1, t2, disc, low, high)

@@ -283,10 +283,10 @@ class _RSSMScan(torch.autograd.Function):
             # fused-phase backward: 4 launches/step; the per-WG recompute of
             # the small row-local gradients replaces all cross-WG traffic.
             # Weight transposes once per step so the bwd GEMMs read k-contig.
-            w1t = w1.t().contiguous()
-            w2t = w2.t().contiguous()
-            w3t = w3.t().contiguous()
-            w4t = w4.t().contiguous()
+            w1t = ext.transpose2d(w1)
+            w2t = ext.transpose2d(w2)
+            w3t = ext.transpose2d(w3)
+            w4t = ext.transpose2d(w4)
             for t in range(T - 1, -1, -1):
                 zc = gz_carry if t < T - 1 else None
                 hc = gh_carry if t < T - 1 else None

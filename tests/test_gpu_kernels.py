@@ -742,3 +742,63 @@ def test_fused_nll_log_probs_match_eager():
     assert torch.allclose(m.log_prob(tgt), ops.mse_log_prob(pred, tgt, 3))
     sd = SymlogDistribution(v, dims=1)
     assert torch.allclose(sd.log_prob(tv), ops.symlog_mse_log_prob(v, tv, 1))
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_fused_behaviour_losses_match_eager():
+    """twohot_mean / reinforce_loss / critic_twohot_loss vs the eager DV3
+    formulas (values + gradients)."""
+    from sheeprl_amd import ops
+    from sheeprl_amd.distributions import TwoHotEncodingDistribution
+
+    torch.manual_seed(0)
+    HZ, F, A, K = 7, 64, 6, 255
+
+    # twohot_mean
+    logits = torch.randn(HZ, F, K, device="cuda")
+    ref = TwoHotEncodingDistribution(logits, dims=1).mean
+    got = ops.twohot_mean(logits)
+    assert torch.allclose(got, ref, atol=1e-4, rtol=1e-4), (got - ref).abs().max()
+
+    # reinforce loss
+    m = torch.log_softmax(torch.randn(HZ + 1, F, A, device="cuda"), -1).requires_grad_()
+    onehot = torch.nn.functional.one_hot(torch.randint(0, A, (HZ + 1, F), device="cuda"), A).to(torch.bfloat16)
+    adv = torch.randn(HZ, F, 1, device="cuda")
+    disc = torch.rand(HZ, F, 1, device="cuda")
+    ent_coef = 3e-4
+    loss = ops.reinforce_loss(m, onehot, adv, disc, ent_coef)
+    loss.backward()
+    g_fused = m.grad.clone()
+    m.grad = None
+    logp = (m * onehot.float()).sum(-1, keepdim=True)[:-1]
+    entropy = -(m.exp() * m).sum(-1, keepdim=True)
+    ref_loss = -torch.mean(disc * (logp * adv + ent_coef * entropy[:-1]))
+    assert torch.allclose(loss, ref_loss, atol=1e-5, rtol=1e-5), (loss, ref_loss)
+    ref_loss.backward()
+    assert torch.allclose(g_fused, m.grad, atol=1e-6), (g_fused - m.grad).abs().max()
+
+    # critic two-hot loss
+    ql = torch.randn(HZ, F, K, device="cuda").requires_grad_()
+    lam = torch.randn(HZ, F, 1, device="cuda") * 3
+    tv = torch.randn(HZ, F, 1, device="cuda") * 3
+    vl = ops.critic_twohot_loss(ql, lam, tv, disc)
+    vl.backward()
+    gq_fused = ql.grad.clone()
+    ql.grad = None
+    qv = TwoHotEncodingDistribution(ql, dims=1)
+    ref_vl = torch.mean((-qv.log_prob(lam) - qv.log_prob(tv)) * disc.squeeze(-1))
+    assert torch.allclose(vl, ref_vl, atol=1e-4, rtol=1e-4), (vl, ref_vl)
+    ref_vl.backward()
+    assert torch.allclose(gq_fused, ql.grad, atol=1e-6, rtol=1e-4), (gq_fused - ql.grad).abs().max()
+
+
+@requires_gpu
+@pytest.mark.timeout(300)
+def test_transpose2d_matches_torch():
+    from sheeprl_amd.ops import require_ext
+
+    ext = require_ext()
+    for R, C in [(512, 1030), (1536, 1024), (512, 4608), (1024, 512), (7, 3)]:
+        x = torch.randn(R, C, device="cuda", dtype=torch.bfloat16)
+        assert torch.equal(ext.transpose2d(x), x.t().contiguous())
