@@ -352,6 +352,171 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     return result
 
 
+def run_bench_sac(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
+    """SAC on a HalfCheetah-shaped synthetic env (17-dim obs, 6-dim actions),
+    bf16 on GPU — BASELINE.json config #2.  One step = one vectorized env
+    interaction + the replay-ratio-driven gradient step (batch 256), with the
+    whole gradient step captured in a hipGraph."""
+    import torch.distributed as dist
+
+    from sheeprl_amd.algos.sac.agent import build_agent as build_sac
+    from sheeprl_amd.algos.sac.sac import train as sac_train
+    from sheeprl_amd.config import compose
+    from sheeprl_amd.data import ReplayBuffer
+    from sheeprl_amd.envs import vectorize_env
+    from sheeprl_amd.optim import FusedAdam
+    from sheeprl_amd.parallel import Runtime
+    from sheeprl_amd.utils.metric import MetricAggregator
+    from sheeprl_amd.utils.utils import seed_everything
+
+    rank = int(os.environ.get("RANK", 0))
+    world_size = int(os.environ.get("WORLD_SIZE", n_gpus))
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    if device_type == "cuda":
+        from sheeprl_amd.ops import has_ext
+
+        if not has_ext():
+            raise RuntimeError("HIP extension _sheep_hip is not built — refusing to bench the eager fallback")
+    cfg = compose([
+        "exp=sac",
+        "env=dummy",
+        "env.id=dummy_continuous",
+        "env.num_envs=4",
+        "env.max_episode_steps=1000",
+        "algo.mlp_keys.encoder=[state]",
+        f"runtime.accelerator={device_type}",
+        "runtime.precision=" + ("bf16" if device_type == "cuda" else "fp32"),
+        "metric.log_level=0",
+        "metric.disable_timer=True",
+        "checkpoint.every=0",
+        "checkpoint.save_last=False",
+        "buffer.size=65536",
+        "buffer.memmap=False",
+        "algo.run_test=False",
+    ] + list(overrides))
+    cfg.env.wrapper_kwargs = {"state_dim": 17, "action_dim": 6, "n_steps": 1000}
+    seed_everything(cfg.seed + rank)
+    runtime = Runtime(devices=world_size, accelerator=cfg.runtime.accelerator, precision=cfg.runtime.precision)
+    runtime.world_size = world_size
+    runtime.global_rank = rank
+    runtime.local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world_size > 1:
+        runtime._init_process_group(init_method="env://")
+    else:
+        runtime._setup_device()
+    device = runtime.device
+    num_envs = cfg.env.num_envs
+
+    envs = vectorize_env(cfg, cfg.seed, rank)
+    obs_space = envs.single_observation_space
+    action_space = envs.single_action_space
+    agent, player = build_sac(runtime, cfg, obs_space, action_space, None)
+    qf_optimizer = FusedAdam(agent.qfs.parameters(), lr=cfg.algo.critic.optimizer.lr)
+    actor_optimizer = FusedAdam(agent.actor.parameters(), lr=cfg.algo.actor.optimizer.lr)
+    alpha_optimizer = FusedAdam([agent.log_alpha], lr=cfg.algo.alpha.optimizer.lr)
+    rb = ReplayBuffer(int(cfg.buffer.size), num_envs, obs_keys=("obs",), memmap=False)
+
+    obs_np, _ = envs.reset(seed=cfg.seed)
+    obs = np.asarray(obs_np["state"], np.float32).reshape(num_envs, -1)
+    rng = np.random.default_rng(0)
+    bs = cfg.algo.per_rank_batch_size
+
+    def env_step(actions):
+        nonlocal obs
+        nxt, rewards, terms, truncs, _ = envs.step(actions)
+        nxt_flat = np.asarray(nxt["state"], np.float32).reshape(num_envs, -1)
+        rb.add({
+            "obs": obs[None],
+            "next_obs": nxt_flat[None],
+            "actions": actions[None].astype(np.float32),
+            "rewards": np.asarray(rewards, np.float32).reshape(1, num_envs, 1),
+            "dones": np.logical_or(terms, truncs).astype(np.float32).reshape(1, num_envs, 1),
+        })
+        obs = nxt_flat
+
+    # random prefill (untimed)
+    for _ in range(max(bs // num_envs + 2, 70)):
+        env_step(rng.uniform(-1, 1, size=(num_envs, 6)).astype(np.float32))
+
+    aggregator = MetricAggregator({})
+    MetricAggregator.disabled = True
+    update = [0]
+
+    def train_fn(batch):
+        update[0] += 1
+        sac_train(runtime, agent, actor_optimizer, qf_optimizer, alpha_optimizer,
+                  batch, None, update[0], cfg, num_envs * world_size)
+
+    graphed = None
+    if device_type == "cuda" and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1":
+        from sheeprl_amd.parallel.graphs import CUDAGraphStep
+
+        example = {k: v[0] for k, v in rb.sample_tensors(bs, n_samples=1, device=device).items()}
+        try:
+            graphed = CUDAGraphStep(train_fn, example, warmup=3)
+            if rank == 0:
+                print("[bench] SAC train step captured in a hipGraph", file=sys.stderr)
+        except Exception as e:  # noqa: BLE001
+            graphed = None
+            if rank == 0:
+                print(f"[bench] hipGraph capture failed ({e}); running eager", file=sys.stderr)
+
+    def one_iter():
+        with torch.no_grad():
+            t_obs = torch.as_tensor(obs, device=device, dtype=torch.float32)
+            actions = player.get_actions(t_obs).float().cpu().numpy()
+        env_step(actions)
+        batch = {k: v[0] for k, v in rb.sample_tensors(bs, n_samples=1, device=device).items()}
+        if graphed is not None:
+            graphed(batch)
+        else:
+            train_fn(batch)
+
+    for _ in range(warmup):
+        one_iter()
+    if runtime.is_distributed:
+        dist.barrier()
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        one_iter()
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    if runtime.is_distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    if runtime.is_distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if runtime.backend == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    envs.close()
+    fps = steps * num_envs * world_size / elapsed
+    return {
+        "metric": "env_frames_per_sec",
+        "value": round(fps, 3),
+        "unit": "frames/s",
+        "n_gpus": world_size,
+        "steps": steps,
+        "warmup": warmup,
+        "ms_per_step": round(elapsed / steps * 1000, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if device_type == "cuda" else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "sac",
+            "global_batch": bs * world_size,
+            "seq_len": 1,
+            "parallelism": f"dp{world_size}",
+            "benchmark": "SAC HalfCheetah-shape (17-dim obs, 6-dim act, synthetic env)",
+            "grad_steps_per_policy_step": cfg.algo.replay_ratio,
+            "action_repeat": 1,
+        },
+    }
+
+
 def smoke_step() -> None:
     """One tiny forward+backward of the flagship model on cuda:0 (driver
     contract: __graft_entry__.smoke)."""
@@ -372,12 +537,17 @@ def smoke_step() -> None:
 def main() -> None:
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=30)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--algo", choices=["dreamer_v3", "sac"], default="dreamer_v3",
+                   help="flagship DV3 Atari-100K (default) or SAC HalfCheetah-shape (BASELINE #2)")
     p.add_argument("--override", action="append", default=[])
     args = p.parse_args()
     rank = int(os.environ.get("RANK", 0))
-    result = run_bench(args.gpus, args.steps, args.warmup, args.override)
+    if args.algo == "sac":
+        result = run_bench_sac(args.gpus, args.steps, args.warmup, args.override)
+    else:
+        result = run_bench(args.gpus, args.steps, args.warmup, args.override)
     if rank == 0:
         print(json.dumps(result))
 

@@ -3787,7 +3787,7 @@ std::vector<torch::Tensor> reinforce_fwd(const torch::Tensor& m, const torch::Te
   TORCH_CHECK(m.scalar_type() == at::kFloat && adv.is_contiguous() && disc.is_contiguous());
   int A = (int)m.size(-1);
   auto out = torch::zeros({}, m.options());
-  int blocks = (int)std::min((HZF + kBlock - 1) / kBlock, (long)1024);
+  int blocks = (int)std::min((HZF + kBlock - 1) / kBlock, (long)240);
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, act.scalar_type(), "reinforce_fwd", [&] {
     using T = scalar_t;
@@ -3822,29 +3822,37 @@ __global__ void vloss2_fwd_kernel(const float* __restrict__ logits, const float*
                                   const float* __restrict__ t2, const float* __restrict__ disc,
                                   float* __restrict__ out, float* __restrict__ lse_out, long N, int K,
                                   float low, float high) {
+  __shared__ float lds[9];
   const int lane = threadIdx.x & 63;
-  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  if (row >= N) return;
-  const float* lr = logits + row * (long)K;
-  float mx = -1e30f;
-  for (int j = lane; j < K; j += 64) mx = fmaxf(mx, lr[j]);
+  const long wave0 = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
+  float acc = 0.f;
+  for (long row = wave0; row < N; row += nwaves) {
+    const float* lr = logits + row * (long)K;
+    float mx = -1e30f;
+    for (int j = lane; j < K; j += 64) mx = fmaxf(mx, lr[j]);
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
-  float se = 0.f;
-  for (int j = lane; j < K; j += 64) se += expf(lr[j] - mx);
+    for (int off = 32; off > 0; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+    float se = 0.f;
+    for (int j = lane; j < K; j += 64) se += expf(lr[j] - mx);
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) se += __shfl_xor(se, off, 64);
-  const float lse = mx + logf(se);
-  if (lane == 0) {
-    int lo1, hi1, lo2, hi2;
-    float wl1, wh1, wl2, wh2;
-    twohot_idx(t1[row], low, high, K, lo1, hi1, wl1, wh1);
-    twohot_idx(t2[row], low, high, K, lo2, hi2, wl2, wh2);
-    float lp1 = wl1 * (lr[lo1] - lse) + wh1 * (lr[hi1] - lse);
-    float lp2 = wl2 * (lr[lo2] - lse) + wh2 * (lr[hi2] - lse);
-    lse_out[row] = lse;
-    atomicAdd(out, disc[row] * (-(lp1)-lp2) / N);
+    for (int off = 32; off > 0; off >>= 1) se += __shfl_xor(se, off, 64);
+    const float lse = mx + logf(se);
+    if (lane == 0) {
+      int lo1, hi1, lo2, hi2;
+      float wl1, wh1, wl2, wh2;
+      twohot_idx(t1[row], low, high, K, lo1, hi1, wl1, wh1);
+      twohot_idx(t2[row], low, high, K, lo2, hi2, wl2, wh2);
+      float lp1 = wl1 * (lr[lo1] - lse) + wh1 * (lr[hi1] - lse);
+      float lp2 = wl2 * (lr[lo2] - lse) + wh2 * (lr[hi2] - lse);
+      lse_out[row] = lse;
+      acc += disc[row] * (-(lp1)-lp2) / N;
+    }
   }
+  // one atomic per block instead of one per row (a single fp32 address
+  // saturates at ~88 atomics/us)
+  acc = block_sum(acc, lds);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
 }
 
 __global__ void vloss2_bwd_kernel(const float* __restrict__ g, const float* __restrict__ logits,
@@ -3877,7 +3885,7 @@ std::vector<torch::Tensor> vloss2_fwd(const torch::Tensor& logits, const torch::
   auto out = torch::zeros({}, logits.options());
   auto lse = torch::empty({N}, logits.options());
   const int rpb = kBlock / 64;
-  int blocks = (int)((N + rpb - 1) / rpb);
+  int blocks = (int)std::min((N + rpb - 1) / rpb, (long)480);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(vloss2_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
                      logits.data_ptr<float>(), t1.data_ptr<float>(), t2.data_ptr<float>(),

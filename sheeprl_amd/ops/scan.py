@@ -149,7 +149,7 @@ class _RSSMScan(torch.autograd.Function):
             and max(D, 3 * H, P, SK) // 64 <= 256
             and 16 * (((H + E) + 31) & ~31) * 2 + 160 <= 160 * 1024
             and 16 * 3 * H * 2 + 16 * H * 2 + 160 <= 160 * 1024
-            and os.environ.get("SHEEPRL_AMD_SCAN_V2", "1") == "1"
+            and os.environ.get("SHEEPRL_AMD_SCAN_V2", "0") == "1"
         )
         if v2_ok:
             ws_f = torch.zeros(T, 3, 32, device=dev, dtype=torch.float32)
