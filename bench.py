@@ -447,11 +447,19 @@ def run_bench_sac(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         sac_train(runtime, agent, actor_optimizer, qf_optimizer, alpha_optimizer,
                   batch, None, update[0], cfg, num_envs * world_size)
 
+    pdt = runtime.param_dtype
+
+    def cast_batch(batch):
+        # model runs in param_dtype (bf16 on GPU); rewards/dones stay fp32
+        for k in ("obs", "next_obs", "actions"):
+            batch[k] = batch[k].to(pdt)
+        return batch
+
     graphed = None
     if device_type == "cuda" and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1":
         from sheeprl_amd.parallel.graphs import CUDAGraphStep
 
-        example = {k: v[0] for k, v in rb.sample_tensors(bs, n_samples=1, device=device).items()}
+        example = cast_batch({k: v[0] for k, v in rb.sample_tensors(bs, n_samples=1, device=device).items()})
         try:
             graphed = CUDAGraphStep(train_fn, example, warmup=3)
             if rank == 0:
@@ -463,10 +471,10 @@ def run_bench_sac(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
 
     def one_iter():
         with torch.no_grad():
-            t_obs = torch.as_tensor(obs, device=device, dtype=torch.float32)
+            t_obs = torch.as_tensor(obs, device=device, dtype=torch.float32).to(pdt)
             actions = player.get_actions(t_obs).float().cpu().numpy()
         env_step(actions)
-        batch = {k: v[0] for k, v in rb.sample_tensors(bs, n_samples=1, device=device).items()}
+        batch = cast_batch({k: v[0] for k, v in rb.sample_tensors(bs, n_samples=1, device=device).items()})
         if graphed is not None:
             graphed(batch)
         else:

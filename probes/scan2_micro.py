@@ -231,3 +231,54 @@ timeit("scan2_b2", pb2)
 timeit("v1 b2 (grubwd+mm)", v1_b2)
 timeit("scan2_b1", pb1)
 timeit("v1 b1 (lnbwd+mm+resets)", v1_b1)
+
+# ---- scan3 split-K variants (generation tickets: monotonic per call)
+scr1 = torch.zeros(16, D, device=dev, dtype=torch.float32)
+scr2 = torch.zeros(16, 3 * H, device=dev, dtype=torch.float32)
+scr3 = torch.zeros(16, P, device=dev, dtype=torch.float32)
+scr4 = torch.zeros(16, SK, device=dev, dtype=torch.float32)
+scrb4 = torch.zeros(16, P, device=dev, dtype=torch.float32)
+tk3 = torch.zeros(4, 256, device=dev, dtype=torch.int32)
+tk2g = torch.zeros(4, device=dev, dtype=torch.int32)
+tkb4 = torch.zeros(P // 16, device=dev, dtype=torch.int32)
+ws3 = torch.zeros(4, 32, device=dev, dtype=torch.float32)
+gen_c = [0, 0, 0, 0, 0]
+
+
+def s3f1():
+    gen_c[0] += 1
+    ws3[0].zero_()
+    ext.scan3_f1(z_prev, iz, h_prev, ih, act, f, w1, lnw1, lnb1, x_s, hu_s, g1_s,
+                 mr[0], mr[1], scr1, tk3[0], tk2g[0:1], ws3[0], eps, gen_c[0])
+
+
+def s3f2():
+    gen_c[1] += 1
+    ws3[1].zero_()
+    ext.scan3_f2(hu_s, w2, lnwg, lnbg, y_s, h_seq, r_s[:, :H], mr[2], mr[3],
+                 scr2, tk3[1], tk2g[1:2], ws3[1], eps, gen_c[1])
+
+
+def s3f3():
+    gen_c[2] += 1
+    ws3[2].zero_()
+    ext.scan3_f3(r_s, w3, lnw3, lnb3, p_s, g3_s, mr[4], mr[5], scr3, tk3[2],
+                 tk2g[2:3], ws3[2], eps, gen_c[2])
+
+
+def s3f4():
+    gen_c[3] += 1
+    ext.scan3_f4(p_s, w4, b4, urand, m_seq, z_seq, s_s, scr4, tk3[3], KD, unimix, gen_c[3])
+
+
+def s3b4():
+    gen_c[4] += 1
+    ext.scan3_b4(gm_in, gz_in, gz_c, s_s, w4tc, graw_o, gp_o, scrb4, tkb4, KD, unimix, gen_c[4])
+
+
+print("== scan3 split-K variants ==")
+timeit("scan3_f1", s3f1)
+timeit("scan3_f2", s3f2)
+timeit("scan3_f3", s3f3)
+timeit("scan3_f4", s3f4)
+timeit("scan3_b4", s3b4)

@@ -3645,6 +3645,636 @@ __global__ void __launch_bounds__(256) scan2_bgru_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// scan3: split-K fused phase kernels (round-2 second pass).  The scan2
+// kernels proved ~2-4x slower than hipblaslt's M=16 GEMMs in the real graph:
+// with only N/64 workgroups the cold weight stream is latency-bound (too few
+// outstanding loads chip-wide).  scan3 adopts hipblaslt-grade parallelism —
+// grid = ntiles x KS workgroups, each computing one 16-col tile over a K
+// slice (further quartered across its 4 waves), accumulated into an fp32
+// scratch with device atomics.  Per tile, a generation ticket elects the
+// LAST-arriving workgroup, which runs the epilogue (cat-ST head) or joins the
+// second, ntiles-wide ticket round for the grid-spanning LayerNorm.
+// Generation counting (arrivals compared against gen*KS / gen*ntiles) makes
+// tickets monotonic — no reset races between the T graph-captured launches.
+// ---------------------------------------------------------------------------
+
+// wave-level combine of the 4 K-quarter partials through LDS; afterwards all
+// four waves hold the full k-slice tile.
+__device__ __forceinline__ f32x4 scan3_combine(f32x4 acc, float* comb) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) comb[(wv * 16 + kgrp * 4 + r) * 16 + arow] = acc[r];
+  __syncthreads();
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    acc[r] = comb[m * 16 + arow] + comb[(16 + m) * 16 + arow] + comb[(32 + m) * 16 + arow] +
+             comb[(48 + m) * 16 + arow];
+  }
+  return acc;
+}
+
+// wave-0 accumulates the combined tile into the fp32 scratch; generation
+// ticket elects the tile's last arriver (which gets an agent acquire).
+__device__ __forceinline__ bool scan3_commit(const f32x4& acc, float* scratch, int N, int col0,
+                                             int* tile_ticket, int arrivals) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  if (wv == 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) atomicAdd(&scratch[(long)(kgrp * 4 + r) * N + col0 + arow], acc[r]);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  __shared__ int last_;
+  if (threadIdx.x == 0) {
+    int t = __hip_atomic_fetch_add(tile_ticket, 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    last_ = (t == arrivals - 1) ? 1 : 0;
+    if (last_) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  return last_ != 0;
+}
+
+// wave-0-only second round for the grid-spanning LayerNorm: publish this
+// tile's partial sums, wait for every tile's last arriver, return row stats.
+// mr_ is a 32-float LDS area ([0..15] mean, [16..31] rstd).
+__device__ __forceinline__ void scan3_ln_round2(const f32x4& v, int N_total, int B, float* ws2,
+                                                int* ticket2, int wait_count, float eps, float* mr_) {
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  float s = 0.f;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    float cs = sc2_colsum(v[r]);
+    float cs2 = sc2_colsum(v[r] * v[r]);
+    if (arow == 0) {
+      atomicAdd(&ws2[kgrp * 4 + r], cs);
+      atomicAdd(&ws2[16 + kgrp * 4 + r], cs2);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (lane == 0) {
+    __hip_atomic_fetch_add(ticket2, 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    int spins = 0;
+    while (__hip_atomic_load(ticket2, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) < wait_count &&
+           spins < (1 << 27)) {
+      __builtin_amdgcn_s_sleep(2);
+      ++spins;
+    }
+  }
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  if (lane < 16) {
+    float mean = ws2[lane] / N_total;
+    float var = ws2[16 + lane] / N_total - mean * mean;
+    mr_[lane] = mean;
+    mr_[16 + lane] = rsqrtf(fmaxf(var, 0.f) + eps);
+  }
+}
+
+// phases 1 & 3: C = A @ W^T + grid LayerNorm + SiLU.  RESETS assembles the
+// episode-reset-masked x on the fly (z'/a') and persists x / h'.
+template <bool RESETS, bool T0>
+__global__ void __launch_bounds__(256) scan3_lnsilu_kernel(
+    const __hip_bfloat16* __restrict__ a_in, long as_,
+    const __hip_bfloat16* __restrict__ iz, const __hip_bfloat16* __restrict__ h_prev,
+    const __hip_bfloat16* __restrict__ ih, const __hip_bfloat16* __restrict__ act,
+    const __hip_bfloat16* __restrict__ f, const __hip_bfloat16* __restrict__ W,
+    const __hip_bfloat16* __restrict__ lnw, const __hip_bfloat16* __restrict__ lnb,
+    __hip_bfloat16* __restrict__ x_out, long xs, __hip_bfloat16* __restrict__ hu_out, long hus,
+    __hip_bfloat16* __restrict__ g_out, long gs, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, float* __restrict__ scratch, int* __restrict__ tickets,
+    int* __restrict__ ticket2, float* __restrict__ ws2, int B, int SK, int A, int H, int N, int K,
+    float eps, int hu_off, int gen) {
+  __shared__ float comb[4 * 16 * 16];
+  __shared__ float mr_[32];
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int tile = blockIdx.x;
+  const int KS = gridDim.y;
+  const int col0 = tile * 16;
+  const int KQ = (((K + KS - 1) / KS) + 31) & ~31;
+  const int kbeg0 = min((int)blockIdx.y * KQ, K);
+  const int kend0 = min(kbeg0 + KQ, K);
+  const int KQW = (((kend0 - kbeg0 + 3) / 4) + 31) & ~31;
+  const int kbeg = min(kbeg0 + wv * KQW, kend0);
+  const int kend = min(kbeg + KQW, kend0);
+  f32x4 acc;
+  if (RESETS) {
+    // on-the-fly reset-masked A: [z', a'] built per fragment load
+    const int wrow = min(col0 + arow, N - 1);
+    acc = (f32x4){0.f, 0.f, 0.f, 0.f};
+    const int arow_c = arow < B ? arow : B - 1;
+    const float fb = __bfloat162float(f[arow_c]);
+    for (int k0 = kbeg; k0 < kend; k0 += 32) {
+      const int k = k0 + kgrp * 8;
+      bf16x8 a, b;
+      if (k + 8 <= SK) {
+        bf16x8 izv = *(const bf16x8*)(iz + (long)arow_c * SK + k);
+        if (T0) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e) a[e] = (__bf16)(fb * (float)izv[e]);
+        } else {
+          bf16x8 zpv = *(const bf16x8*)(a_in + (long)arow_c * SK + k);
+#pragma unroll
+          for (int e = 0; e < 8; ++e) a[e] = (__bf16)((1.f - fb) * (float)zpv[e] + fb * (float)izv[e]);
+        }
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int kk = k + e;
+          float v = 0.f;
+          if (kk < SK) {
+            float zp = T0 ? 0.f : __bfloat162float(a_in[(long)arow_c * SK + kk]);
+            v = (1.f - fb) * zp + fb * __bfloat162float(iz[(long)arow_c * SK + kk]);
+          } else if (kk < SK + A) {
+            v = (1.f - fb) * __bfloat162float(act[(long)arow_c * A + (kk - SK)]);
+          }
+          a[e] = (__bf16)v;
+        }
+      }
+      const __hip_bfloat16* p = W + (long)wrow * K + k;
+      if (k + 8 <= K) {
+        b = *(const bf16x8*)p;
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) b[e] = (k + e < K) ? (__bf16)__bfloat162float(p[e]) : (__bf16)0.f;
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+    // persist x (K-striped by tile-0 slices) and h' (tile 0, slice 0)
+    if (tile == 0) {
+      for (int i = threadIdx.x; i < B * (kend0 - kbeg0); i += blockDim.x) {
+        const int m = i / (kend0 - kbeg0), kk = kbeg0 + i % (kend0 - kbeg0);
+        float v;
+        if (kk < SK) {
+          const float fm = __bfloat162float(f[m]);
+          float zp = T0 ? 0.f : __bfloat162float(a_in[(long)m * SK + kk]);
+          v = (1.f - fm) * zp + fm * __bfloat162float(iz[(long)m * SK + kk]);
+        } else {
+          const float fm = __bfloat162float(f[m]);
+          v = (1.f - fm) * __bfloat162float(act[(long)m * A + (kk - SK)]);
+        }
+        x_out[(long)m * xs + kk] = __float2bfloat16(v);
+      }
+      if (blockIdx.y == 0) {
+        for (int i = threadIdx.x; i < B * H; i += blockDim.x) {
+          const int m = i / H, j = i - m * H;
+          const float fm = __bfloat162float(f[m]);
+          float hp = T0 ? 0.f : __bfloat162float(h_prev[(long)m * H + j]);
+          st(hu_out, (long)m * hus + j, (1.f - fm) * hp + fm * __bfloat162float(ih[(long)m * H + j]));
+        }
+      }
+    }
+  } else {
+    acc = sc2_gemm_tile_range_g(a_in, as_, B, W, K, K, N, col0, kbeg, kend);
+  }
+  acc = scan3_combine(acc, comb);
+  if (!scan3_commit(acc, scratch, N, col0, &tickets[tile], gen * KS)) return;
+  if (wv != 0) return;
+  // wave-0 round 2: read the final tile, grid LayerNorm, epilogue, reset
+  f32x4 v;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) v[r] = scratch[(long)(kgrp * 4 + r) * N + col0 + arow];
+  scan3_ln_round2(v, N, B, ws2, ticket2, gen * gridDim.x, eps, mr_);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const int j = col0 + arow;
+    if (m < B) {
+      g_out[(long)m * gs + j] = __float2bfloat16(v[r]);
+      float z = (v[r] - mr_[m]) * mr_[16 + m] * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
+      z = z / (1.f + expf(-z));
+      st(hu_out, (long)m * hus + hu_off + j, z);
+    }
+    scratch[(long)(kgrp * 4 + r) * N + col0 + arow] = 0.f;
+  }
+  if (tile == 0 && lane < 16 && lane < B) {
+    mean_out[lane] = mr_[lane];
+    rstd_out[lane] = mr_[16 + lane];
+  }
+}
+
+// phase 2: the GRU projection (3 gate stripes per h column) + grid LayerNorm
+// over 3H + Hafner gates.
+__global__ void __launch_bounds__(256) scan3_gru_kernel(
+    const __hip_bfloat16* __restrict__ hu, long hus, const __hip_bfloat16* __restrict__ W2,
+    const __hip_bfloat16* __restrict__ lnw, const __hip_bfloat16* __restrict__ lnb,
+    __hip_bfloat16* __restrict__ y_out, long ys2, __hip_bfloat16* __restrict__ h_out,
+    __hip_bfloat16* __restrict__ h_out2, long h2s, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, float* __restrict__ scratch, int* __restrict__ tickets,
+    int* __restrict__ ticket2, float* __restrict__ ws2, int B, int H, int D, float eps, int gen) {
+  __shared__ float comb[4 * 16 * 16];
+  __shared__ float mr_[32];
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int tile = blockIdx.x;
+  const int KS = gridDim.y;
+  const int K = H + D;
+  const int col0 = tile * 16;
+  const int KQ = (((K + KS - 1) / KS) + 31) & ~31;
+  const int kbeg0 = min((int)blockIdx.y * KQ, K);
+  const int kend0 = min(kbeg0 + KQ, K);
+  const int KQW = (((kend0 - kbeg0 + 3) / 4) + 31) & ~31;
+  const int kbeg = min(kbeg0 + wv * KQW, kend0);
+  const int kend = min(kbeg + KQW, kend0);
+  f32x4 accg[3];
+#pragma unroll
+  for (int g = 0; g < 3; ++g) {
+    accg[g] = sc2_gemm_tile_range_g(hu, hus, B, W2, K, K, 3 * H, g * H + col0, kbeg, kend);
+    accg[g] = scan3_combine(accg[g], comb);
+    __syncthreads();
+  }
+  // commit all 3 stripes, one ticket
+  if (wv == 0) {
+#pragma unroll
+    for (int g = 0; g < 3; ++g) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        atomicAdd(&scratch[(long)(kgrp * 4 + r) * 3 * H + g * H + col0 + arow], accg[g][r]);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  __shared__ int last_;
+  if (threadIdx.x == 0) {
+    int t = __hip_atomic_fetch_add(&tickets[tile], 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    last_ = (t == gen * KS - 1) ? 1 : 0;
+    if (last_) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  if (!last_ || wv != 0) return;
+  const int DD = 3 * H;
+  f32x4 vg[3];
+  float s1p = 0.f;
+#pragma unroll
+  for (int g = 0; g < 3; ++g) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) vg[g][r] = scratch[(long)(kgrp * 4 + r) * DD + g * H + col0 + arow];
+  }
+  // partial sums over the 3 stripes
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    float s = vg[0][r] + vg[1][r] + vg[2][r];
+    float s2 = vg[0][r] * vg[0][r] + vg[1][r] * vg[1][r] + vg[2][r] * vg[2][r];
+    float cs = sc2_colsum(s);
+    float cs2 = sc2_colsum(s2);
+    if (arow == 0) {
+      atomicAdd(&ws2[kgrp * 4 + r], cs);
+      atomicAdd(&ws2[16 + kgrp * 4 + r], cs2);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  if (lane == 0) {
+    __hip_atomic_fetch_add(ticket2, 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    int spins = 0;
+    while (__hip_atomic_load(ticket2, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT) < gen * (int)gridDim.x &&
+           spins < (1 << 27)) {
+      __builtin_amdgcn_s_sleep(2);
+      ++spins;
+    }
+  }
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  if (lane < 16) {
+    float mean = ws2[lane] / DD;
+    float var = ws2[16 + lane] / DD - mean * mean;
+    mr_[lane] = mean;
+    mr_[16 + lane] = rsqrtf(fmaxf(var, 0.f) + eps);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const int j = col0 + arow;
+    if (m < B) {
+      // persist pre-LN y and run the Hafner gates
+      y_out[(long)m * ys2 + j] = __float2bfloat16(vg[0][r]);
+      y_out[(long)m * ys2 + H + j] = __float2bfloat16(vg[1][r]);
+      y_out[(long)m * ys2 + 2 * H + j] = __float2bfloat16(vg[2][r]);
+      const float mean = mr_[m], rstd = mr_[16 + m];
+      float zr = (vg[0][r] - mean) * rstd * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
+      float zc = (vg[1][r] - mean) * rstd * __bfloat162float(lnw[H + j]) + __bfloat162float(lnb[H + j]);
+      float zu = (vg[2][r] - mean) * rstd * __bfloat162float(lnw[2 * H + j]) + __bfloat162float(lnb[2 * H + j]);
+      float rg = 1.f / (1.f + expf(-zr));
+      float cg = tanhf(rg * zc);
+      float ug = 1.f / (1.f + expf(-(zu - 1.f)));
+      float hp = __bfloat162float(hu[(long)m * hus + j]);
+      float hv = ug * cg + (1.f - ug) * hp;
+      h_out[(long)m * H + j] = __float2bfloat16(hv);
+      h_out2[(long)m * h2s + j] = __float2bfloat16(hv);
+    }
+#pragma unroll
+    for (int g = 0; g < 3; ++g) scratch[(long)(kgrp * 4 + r) * DD + g * H + col0 + arow] = 0.f;
+  }
+  if (tile == 0 && lane < 16 && lane < B) {
+    mean_out[lane] = mr_[lane];
+    rstd_out[lane] = mr_[16 + lane];
+  }
+}
+
+// phase 4: raw = p @ W4^T + b4 + the unimix categorical-ST head.  Tiles are
+// 64 columns wide (whole KD groups per epilogue owner); the 4 waves each own
+// one 16-col subtile over the workgroup's K slice.
+__global__ void __launch_bounds__(256) scan3_catst_kernel(
+    const __hip_bfloat16* __restrict__ p_in, long ps, const __hip_bfloat16* __restrict__ W4,
+    const __hip_bfloat16* __restrict__ b4, const float* __restrict__ urand,
+    float* __restrict__ m_out, __hip_bfloat16* __restrict__ z_out, float* __restrict__ s_out,
+    float* __restrict__ scratch, int* __restrict__ tickets, int B, int P, int SK, int KD,
+    float unimix, int gen) {
+  __shared__ float raw[16][72];
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int tile = blockIdx.x;  // 64-col tile
+  const int KS = gridDim.y;
+  const int col0 = tile * 64 + wv * 16;
+  const int KQ = (((P + KS - 1) / KS) + 31) & ~31;
+  const int kbeg = min((int)blockIdx.y * KQ, P);
+  const int kend = min(kbeg + KQ, P);
+  f32x4 acc = sc2_gemm_tile_range_g(p_in, ps, B, W4, P, P, SK, col0, kbeg, kend);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) atomicAdd(&scratch[(long)(kgrp * 4 + r) * SK + col0 + arow], acc[r]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  __shared__ int last_;
+  if (threadIdx.x == 0) {
+    int t = __hip_atomic_fetch_add(&tickets[tile], 1, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    last_ = (t == gen * KS - 1) ? 1 : 0;
+    if (last_) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+  }
+  __syncthreads();
+  if (!last_) return;
+  for (int i = threadIdx.x; i < 16 * 64; i += blockDim.x) {
+    const int m = i >> 6, c = i & 63;
+    const long sidx = (long)m * SK + tile * 64 + c;
+    raw[m][c] = scratch[sidx] + __bfloat162float(b4[tile * 64 + c]);
+    scratch[sidx] = 0.f;
+  }
+  __syncthreads();
+  const int gpw = 64 / KD;
+  for (int task = threadIdx.x; task < 16 * gpw; task += blockDim.x) {
+    const int m = task / gpw;
+    if (m >= B) continue;
+    const int gl = task - m * gpw;
+    const float* row = &raw[m][gl * KD];
+    const long gcol = (long)tile * 64 + gl * KD;
+    float lmax = -1e30f;
+    for (int j = 0; j < KD; ++j) lmax = fmaxf(lmax, row[j]);
+    float lsum = 0.f;
+    for (int j = 0; j < KD; ++j) lsum += expf(row[j] - lmax);
+    const float inv = 1.f / lsum;
+    const float* ur = urand + (long)m * SK + gcol;
+    float* mro = m_out + (long)m * SK + gcol;
+    float* sro = s_out + (long)m * SK + gcol;
+    __hip_bfloat16* zro = z_out + (long)m * SK + gcol;
+    float best = -1e30f;
+    int best_j = 0;
+    for (int j = 0; j < KD; ++j) {
+      float sv = expf(row[j] - lmax) * inv;
+      float pv = (1.f - unimix) * sv + unimix / KD;
+      float mv = logf(pv);
+      sro[j] = sv;
+      mro[j] = mv;
+      float tt = fmaxf(-logf(fmaxf(ur[j], 1e-20f)), 1e-20f);
+      float score = mv - logf(tt);
+      if (score > best) {
+        best = score;
+        best_j = j;
+      }
+    }
+    for (int j = 0; j < KD; ++j) zro[j] = (__hip_bfloat16)(j == best_j ? 1.f : 0.f);
+  }
+}
+
+// backward phase 4: per-slice categorical-ST backward into LDS, split-K
+// gp = graw @ W4t into scratch, last arriver converts.
+__global__ void __launch_bounds__(256) scan3_b4_kernel(
+    const float* __restrict__ gm, const __hip_bfloat16* __restrict__ gon,
+    const __hip_bfloat16* __restrict__ gon2, const float* __restrict__ s_saved,
+    const __hip_bfloat16* __restrict__ W4t, __hip_bfloat16* __restrict__ graw_out, long gws,
+    __hip_bfloat16* __restrict__ gp_out, float* __restrict__ scratch, int* __restrict__ tickets,
+    int B, int SK, int P, int KD, float unimix, int gen) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int lane = threadIdx.x & 63;
+  const int arow = lane & 15;
+  const int kgrp = lane >> 4;
+  const int wv = threadIdx.x >> 6;
+  const int tile = blockIdx.x;
+  const int KS = gridDim.y;
+  const int col0 = tile * 16;
+  // K slice of SK, aligned to max(KD, 32) so KD groups stay whole
+  const int align = KD > 32 ? KD : 32;
+  const int KQ = (((SK + KS - 1) / KS) + align - 1) & ~(align - 1);
+  const int kbeg0 = min((int)blockIdx.y * KQ, SK);
+  const int kend0 = min(kbeg0 + KQ, SK);
+  const int KL = kend0 - kbeg0;
+  const int KLP = ((KL + 31) & ~31) + 8;
+  __hip_bfloat16* graw = (__hip_bfloat16*)smem;  // [16][KLP]
+  float* comb = (float*)(smem + ((16 * KLP * 2 + 15) & ~15));
+  // slice-local categorical-ST backward (wave-per-4-rows, coalesced)
+  for (int mr2 = 0; mr2 < 4; ++mr2) {
+    const int m = wv * 4 + mr2;
+    for (int c0 = 0; c0 < KL; c0 += 64) {
+      const int c = c0 + lane;
+      float t = 0.f, sj = 0.f;
+      if (m < B && c < KL) {
+        const long idx = (long)m * SK + kbeg0 + c;
+        sj = s_saved[idx];
+        float pj = (1.f - unimix) * sj + unimix / KD;
+        t = gm[idx] / pj + __bfloat162float(gon[idx]) + (gon2 ? __bfloat162float(gon2[idx]) : 0.f);
+      }
+      float acc = t * sj;
+#pragma unroll
+      for (int off = 1; off < 64; off <<= 1) {
+        if (off < KD) acc += __shfl_xor(acc, off, 64);
+      }
+      if (c < KLP) graw[m * KLP + c] = __float2bfloat16((m < B && c < KL) ? (1.f - unimix) * sj * (t - acc) : 0.f);
+    }
+    // zero the 32-align padding tail
+    for (int c = KL + lane; c < KLP; c += 64) graw[m * KLP + c] = (__hip_bfloat16)0.f;
+  }
+  __syncthreads();
+  if (tile == 0) {
+    for (int i = threadIdx.x; i < B * KL; i += blockDim.x) {
+      const int m = i / KL, c = i - m * KL;
+      graw_out[(long)m * gws + kbeg0 + c] = graw[m * KLP + c];
+    }
+  }
+  // GEMM over the slice (waves split KL four ways from LDS)
+  const int KQW = (((KL + 3) / 4) + 31) & ~31;
+  const int kb = min(wv * KQW, KL);
+  const int ke = min(kb + KQW, KL);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  {
+    const int wrow = min(col0 + arow, P - 1);
+    for (int k0 = kb; k0 < ke; k0 += 32) {
+      const int k = k0 + kgrp * 8;
+      bf16x8 a = *(const bf16x8*)(graw + arow * KLP + k);
+      bf16x8 b;
+      const __hip_bfloat16* p = W4t + (long)wrow * SK + kbeg0 + k;
+      if (kbeg0 + k + 8 <= SK && k + 8 <= ke) {
+        b = *(const bf16x8*)p;
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          b[e] = (k + e < ke) ? (__bf16)__bfloat162float(p[e]) : (__bf16)0.f;
+      }
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+    }
+  }
+  acc = scan3_combine(acc, comb);
+  if (!scan3_commit(acc, scratch, P, col0, &tickets[tile], gen * KS)) return;
+  if (wv != 0) return;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int m = kgrp * 4 + r;
+    const long sidx = (long)m * P + col0 + arow;
+    if (m < B) gp_out[(long)m * P + col0 + arow] = __float2bfloat16(scratch[sidx]);
+    scratch[sidx] = 0.f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// scan3 host wrappers
+// ---------------------------------------------------------------------------
+
+static inline const __hip_bfloat16* s3_bp(const torch::Tensor& t) {
+  return (const __hip_bfloat16*)t.data_ptr();
+}
+static inline __hip_bfloat16* s3_bpm(torch::Tensor& t) { return (__hip_bfloat16*)t.data_ptr(); }
+
+static inline int s3_ks(int ntiles, int K) {
+  // target 96-256 workgroups, K-slices of >= ~128
+  int ks = 192 / ntiles;
+  ks = std::max(1, std::min(8, ks));
+  while (ks > 1 && (K + ks - 1) / ks < 96) --ks;
+  return ks;
+}
+
+void scan3_f1(const c10::optional<torch::Tensor>& z_prev, const torch::Tensor& iz,
+              const c10::optional<torch::Tensor>& h_prev, const torch::Tensor& ih,
+              const torch::Tensor& act, const torch::Tensor& f, const torch::Tensor& W1,
+              const torch::Tensor& lnw, const torch::Tensor& lnb, torch::Tensor x_out,
+              torch::Tensor hu_out, torch::Tensor g_out, torch::Tensor mean, torch::Tensor rstd,
+              torch::Tensor scratch, torch::Tensor tickets, torch::Tensor ticket2, torch::Tensor ws2,
+              double eps, long gen) {
+  const int B = (int)act.size(0), A = (int)act.size(1), SK = (int)iz.size(1), H = (int)ih.size(1);
+  const int N = (int)W1.size(0), K = (int)W1.size(1);
+  TORCH_CHECK(B <= 16 && N % 16 == 0 && K == SK + A && SK % 8 == 0 && W1.is_contiguous());
+  const int ntiles = N / 16;
+  TORCH_CHECK(scratch.numel() >= 16 * (long)N && tickets.numel() >= ntiles);
+  const int KS = s3_ks(ntiles, K);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bool t0 = !z_prev.has_value();
+  const __hip_bfloat16* zp = t0 ? nullptr : s3_bp(*z_prev);
+  const __hip_bfloat16* hp = h_prev.has_value() ? s3_bp(*h_prev) : nullptr;
+#define S3_F1(T0V)                                                                                     \
+  hipLaunchKernelGGL((scan3_lnsilu_kernel<true, T0V>), dim3(ntiles, KS), dim3(256), 0,                 \
+                     stream.stream(), zp, 0, s3_bp(iz), hp, s3_bp(ih), s3_bp(act), s3_bp(f),           \
+                     s3_bp(W1), s3_bp(lnw), s3_bp(lnb), s3_bpm(x_out), x_out.stride(0),                \
+                     s3_bpm(hu_out), hu_out.stride(0), s3_bpm(g_out), g_out.stride(0),                 \
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), scratch.data_ptr<float>(),        \
+                     tickets.data_ptr<int>(), ticket2.data_ptr<int>(), ws2.data_ptr<float>(), B, SK,   \
+                     A, H, N, K, (float)eps, H, (int)gen)
+  if (t0)
+    S3_F1(true);
+  else
+    S3_F1(false);
+#undef S3_F1
+}
+
+void scan3_f3(const torch::Tensor& a_in, const torch::Tensor& W3, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, torch::Tensor p_out, torch::Tensor g_out, torch::Tensor mean,
+              torch::Tensor rstd, torch::Tensor scratch, torch::Tensor tickets, torch::Tensor ticket2,
+              torch::Tensor ws2, double eps, long gen) {
+  const int B = (int)a_in.size(0);
+  const int N = (int)W3.size(0), K = (int)W3.size(1);
+  TORCH_CHECK(B <= 16 && N % 16 == 0 && a_in.size(1) == K && W3.is_contiguous());
+  TORCH_CHECK((K % 8) == 0 && (a_in.stride(0) % 8) == 0, "scan3_f3: 16B-aligned A rows");
+  const int ntiles = N / 16;
+  const int KS = s3_ks(ntiles, K);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((scan3_lnsilu_kernel<false, false>), dim3(ntiles, KS), dim3(256), 0,
+                     stream.stream(), s3_bp(a_in), a_in.stride(0), nullptr, nullptr, nullptr, nullptr,
+                     nullptr, s3_bp(W3), s3_bp(lnw), s3_bp(lnb), nullptr, 0, s3_bpm(p_out),
+                     p_out.stride(0), s3_bpm(g_out), g_out.stride(0), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), scratch.data_ptr<float>(), tickets.data_ptr<int>(),
+                     ticket2.data_ptr<int>(), ws2.data_ptr<float>(), B, 0, 0, 0, N, K, (float)eps, 0,
+                     (int)gen);
+}
+
+void scan3_f2(const torch::Tensor& hu, const torch::Tensor& W2, const torch::Tensor& lnw,
+              const torch::Tensor& lnb, torch::Tensor y_out, torch::Tensor h_out, torch::Tensor h_out2,
+              torch::Tensor mean, torch::Tensor rstd, torch::Tensor scratch, torch::Tensor tickets,
+              torch::Tensor ticket2, torch::Tensor ws2, double eps, long gen) {
+  const int B = (int)hu.size(0), K = (int)hu.size(1);
+  const int H = (int)h_out.size(1), D = K - H;
+  TORCH_CHECK(B <= 16 && H % 16 == 0 && W2.size(0) == 3 * H && W2.size(1) == K && W2.is_contiguous());
+  TORCH_CHECK(h_out.is_contiguous() && (hu.stride(0) % 8) == 0 && (K % 8) == 0);
+  const int ntiles = H / 16;
+  TORCH_CHECK(scratch.numel() >= 16 * 3 * (long)H && tickets.numel() >= ntiles);
+  const int KS = s3_ks(ntiles, K);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(scan3_gru_kernel, dim3(ntiles, KS), dim3(256), 0, stream.stream(), s3_bp(hu),
+                     hu.stride(0), s3_bp(W2), s3_bp(lnw), s3_bp(lnb), s3_bpm(y_out), y_out.stride(0),
+                     s3_bpm(h_out), s3_bpm(h_out2), h_out2.stride(0), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), scratch.data_ptr<float>(), tickets.data_ptr<int>(),
+                     ticket2.data_ptr<int>(), ws2.data_ptr<float>(), B, H, D, (float)eps, (int)gen);
+}
+
+void scan3_f4(const torch::Tensor& p_in, const torch::Tensor& W4, const torch::Tensor& b4,
+              const torch::Tensor& urand, torch::Tensor m_out, torch::Tensor z_out, torch::Tensor s_out,
+              torch::Tensor scratch, torch::Tensor tickets, long KD, double unimix, long gen) {
+  const int B = (int)p_in.size(0), P = (int)p_in.size(1);
+  const int SK = (int)W4.size(0);
+  TORCH_CHECK(B <= 16 && SK % 64 == 0 && W4.size(1) == P && W4.is_contiguous());
+  TORCH_CHECK(KD <= 64 && 64 % KD == 0 && m_out.is_contiguous() && z_out.is_contiguous() && s_out.is_contiguous());
+  TORCH_CHECK((P % 8) == 0 && (p_in.stride(0) % 8) == 0);
+  const int ntiles = SK / 64;
+  TORCH_CHECK(scratch.numel() >= 16 * (long)SK && tickets.numel() >= ntiles);
+  const int KS = s3_ks(ntiles, P);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(scan3_catst_kernel, dim3(ntiles, KS), dim3(256), 0, stream.stream(), s3_bp(p_in),
+                     p_in.stride(0), s3_bp(W4), s3_bp(b4), urand.data_ptr<float>(),
+                     m_out.data_ptr<float>(), s3_bpm(z_out), s_out.data_ptr<float>(),
+                     scratch.data_ptr<float>(), tickets.data_ptr<int>(), B, P, SK, (int)KD,
+                     (float)unimix, (int)gen);
+}
+
+void scan3_b4(const torch::Tensor& gm, const torch::Tensor& gon, const c10::optional<torch::Tensor>& gon2,
+              const torch::Tensor& s_saved, const torch::Tensor& W4t, torch::Tensor graw_out,
+              torch::Tensor gp_out, torch::Tensor scratch, torch::Tensor tickets, long KD,
+              double unimix, long gen) {
+  const int B = (int)gon.size(0), SK = (int)gon.size(1);
+  const int P = (int)W4t.size(0);
+  TORCH_CHECK(B <= 16 && SK % 64 == 0 && P % 16 == 0 && W4t.size(1) == SK && W4t.is_contiguous());
+  TORCH_CHECK(KD <= 64 && (KD & (KD - 1)) == 0 && gp_out.is_contiguous());
+  const int ntiles = P / 16;
+  TORCH_CHECK(scratch.numel() >= 16 * (long)P && tickets.numel() >= ntiles);
+  const int KS = s3_ks(ntiles, SK);
+  const int align = KD > 32 ? (int)KD : 32;
+  const int KQ = (((SK + KS - 1) / KS) + align - 1) & ~(align - 1);
+  const int KLP = ((KQ + 31) & ~31) + 8;
+  const size_t shmem = ((16 * KLP * 2 + 15) & ~15) + 4 * 16 * 16 * sizeof(float);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const __hip_bfloat16* g2 = gon2.has_value() ? s3_bp(*gon2) : nullptr;
+  hipLaunchKernelGGL(scan3_b4_kernel, dim3(ntiles, KS), dim3(256), shmem, stream.stream(),
+                     gm.data_ptr<float>(), s3_bp(gon), g2, s_saved.data_ptr<float>(), s3_bp(W4t),
+                     s3_bpm(graw_out), graw_out.stride(0), s3_bpm(gp_out), scratch.data_ptr<float>(),
+                     tickets.data_ptr<int>(), B, SK, P, (int)KD, (float)unimix, (int)gen);
+}
+
+// ---------------------------------------------------------------------------
 // tiled 2-D transpose (64x64 LDS tiles, padded rows): torch's .t().contiguous()
 // on bf16 weights is an uncoalesced 2-byte strided copy; the scan2 backward
 // transposes ~10 MB of weights per step and needs this to be bandwidth-bound.
@@ -4250,6 +4880,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reinforce_bwd", &reinforce_bwd);
   m.def("vloss2_fwd", &vloss2_fwd);
   m.def("vloss2_bwd", &vloss2_bwd);
+  m.def("scan3_f1", &scan3_f1);
+  m.def("scan3_f2", &scan3_f2);
+  m.def("scan3_f3", &scan3_f3);
+  m.def("scan3_f4", &scan3_f4);
+  m.def("scan3_b4", &scan3_b4);
   m.def("scan2_f1", &scan2_f1);
   m.def("scan2_f2", &scan2_f2);
   m.def("scan2_f3", &scan2_f3);

@@ -45,6 +45,12 @@ from torch import Tensor
 from sheeprl_amd.ops._ext import require_ext
 
 
+def _scan_impl() -> str:
+    """Fused-phase scan implementation: "v3" (split-K, default), "v2"
+    (N-split + ticket LN), or "v1" (launch-per-op with hipblaslt GEMMs)."""
+    return os.environ.get("SHEEPRL_AMD_SCAN_IMPL", "v3")
+
+
 class _RSSMScan(torch.autograd.Function):
     @staticmethod
     def forward(
@@ -149,25 +155,51 @@ class _RSSMScan(torch.autograd.Function):
             and max(D, 3 * H, P, SK) // 64 <= 256
             and 16 * (((H + E) + 31) & ~31) * 2 + 160 <= 160 * 1024
             and 16 * 3 * H * 2 + 16 * H * 2 + 160 <= 160 * 1024
-            and os.environ.get("SHEEPRL_AMD_SCAN_V2", "0") == "1"
+            and _scan_impl() in ("v2", "v3")
         )
         if v2_ok:
-            ws_f = torch.zeros(T, 3, 32, device=dev, dtype=torch.float32)
-            tk_f = torch.zeros(T, 3, device=dev, dtype=torch.int32)
             iz_c = iz.contiguous()
             ih_c = ih.contiguous()
-            for t in range(T):
-                zp = None if t == 0 else z_seq[t - 1]
-                hp = None if t == 0 else h_seq[t - 1]
-                ext.scan2_f1(zp, iz_c, hp, ih_c, actions[t], f_all[t], w1, lnw1, lnb1,
-                             x_s[t], hu_s[t], g1_s[t], mr1_s[0, t], mr1_s[1, t],
-                             ws_f[t, 0], tk_f[t, 0], eps)
-                ext.scan2_f2(hu_s[t], w2, lnwg, lnbg, y_s[t], h_seq[t], r_s[t, :, :H],
-                             mrg_s[0, t], mrg_s[1, t], ws_f[t, 1], tk_f[t, 1], eps)
-                ext.scan2_f3(r_s[t], w3, lnw3, lnb3, p_s[t], g3_s[t], mr3_s[0, t],
-                             mr3_s[1, t], ws_f[t, 2], tk_f[t, 2], eps)
-                ext.scan2_f4(p_s[t], w4, b4, urand_all[t].reshape(B, SK), m_seq[t],
-                             z_seq[t], s_s[t].view(B, SK), discrete, unimix)
+            if _scan_impl() == "v3":
+                # split-K fused phases: hipblaslt-grade workgroup parallelism
+                # with generation tickets (no per-step state resets)
+                ws_f = torch.zeros(T, 3, 32, device=dev, dtype=torch.float32)
+                tk3 = torch.zeros(3, 256, device=dev, dtype=torch.int32)
+                tk2 = torch.zeros(3, device=dev, dtype=torch.int32)
+                tk4 = torch.zeros(SK // 64, device=dev, dtype=torch.int32)
+                scr1 = torch.zeros(16, D, device=dev, dtype=torch.float32)
+                scr2 = torch.zeros(16, 3 * H, device=dev, dtype=torch.float32)
+                scr3 = torch.zeros(16, P, device=dev, dtype=torch.float32)
+                scr4 = torch.zeros(16, SK, device=dev, dtype=torch.float32)
+                for t in range(T):
+                    gen = t + 1
+                    zp = None if t == 0 else z_seq[t - 1]
+                    hp = None if t == 0 else h_seq[t - 1]
+                    ext.scan3_f1(zp, iz_c, hp, ih_c, actions[t], f_all[t], w1, lnw1, lnb1,
+                                 x_s[t], hu_s[t], g1_s[t], mr1_s[0, t], mr1_s[1, t],
+                                 scr1, tk3[0], tk2[0:1], ws_f[t, 0], eps, gen)
+                    ext.scan3_f2(hu_s[t], w2, lnwg, lnbg, y_s[t], h_seq[t], r_s[t, :, :H],
+                                 mrg_s[0, t], mrg_s[1, t], scr2, tk3[1], tk2[1:2],
+                                 ws_f[t, 1], eps, gen)
+                    ext.scan3_f3(r_s[t], w3, lnw3, lnb3, p_s[t], g3_s[t], mr3_s[0, t],
+                                 mr3_s[1, t], scr3, tk3[2], tk2[2:3], ws_f[t, 2], eps, gen)
+                    ext.scan3_f4(p_s[t], w4, b4, urand_all[t].reshape(B, SK), m_seq[t],
+                                 z_seq[t], s_s[t].view(B, SK), scr4, tk4, discrete, unimix, gen)
+            else:
+                ws_f = torch.zeros(T, 3, 32, device=dev, dtype=torch.float32)
+                tk_f = torch.zeros(T, 3, device=dev, dtype=torch.int32)
+                for t in range(T):
+                    zp = None if t == 0 else z_seq[t - 1]
+                    hp = None if t == 0 else h_seq[t - 1]
+                    ext.scan2_f1(zp, iz_c, hp, ih_c, actions[t], f_all[t], w1, lnw1, lnb1,
+                                 x_s[t], hu_s[t], g1_s[t], mr1_s[0, t], mr1_s[1, t],
+                                 ws_f[t, 0], tk_f[t, 0], eps)
+                    ext.scan2_f2(hu_s[t], w2, lnwg, lnbg, y_s[t], h_seq[t], r_s[t, :, :H],
+                                 mrg_s[0, t], mrg_s[1, t], ws_f[t, 1], tk_f[t, 1], eps)
+                    ext.scan2_f3(r_s[t], w3, lnw3, lnb3, p_s[t], g3_s[t], mr3_s[0, t],
+                                 mr3_s[1, t], ws_f[t, 2], tk_f[t, 2], eps)
+                    ext.scan2_f4(p_s[t], w4, b4, urand_all[t].reshape(B, SK), m_seq[t],
+                                 z_seq[t], s_s[t].view(B, SK), discrete, unimix)
             ctx.save_for_backward(
                 f_all, w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4,
                 x_s, g1_s, mr1_s, hu_s, y_s, mrg_s, r_s, g3_s, mr3_s, p_s, s_s,
@@ -175,6 +207,7 @@ class _RSSMScan(torch.autograd.Function):
             ctx.dims = (T, B, E, A, H, SK, D, P, S, discrete)
             ctx.unimix = unimix
             ctx.v2 = True
+            ctx.impl = _scan_impl()
             return h_seq, z_seq, m_seq
         h0 = torch.zeros(B, H, device=dev, dtype=dt)
         z0 = torch.zeros(B, SK, device=dev, dtype=dt)
@@ -287,11 +320,19 @@ class _RSSMScan(torch.autograd.Function):
             w2t = ext.transpose2d(w2)
             w3t = ext.transpose2d(w3)
             w4t = ext.transpose2d(w4)
+            if ctx.impl == "v3":
+                scr_b4 = torch.zeros(16, P, device=dev, dtype=torch.float32)
+                tkb4 = torch.zeros(P // 16, device=dev, dtype=torch.int32)
             for t in range(T - 1, -1, -1):
                 zc = gz_carry if t < T - 1 else None
                 hc = gh_carry if t < T - 1 else None
-                ext.scan2_b4(g_m_seq[t].view(B, SK), g_z_seq[t].view(B, SK), zc,
-                             s_s[t].view(B, SK), w4t, graw_s[t], gp, discrete, unimix)
+                if ctx.impl == "v3":
+                    ext.scan3_b4(g_m_seq[t].view(B, SK), g_z_seq[t].view(B, SK), zc,
+                                 s_s[t].view(B, SK), w4t, graw_s[t], gp, scr_b4, tkb4,
+                                 discrete, unimix, T - t)
+                else:
+                    ext.scan2_b4(g_m_seq[t].view(B, SK), g_z_seq[t].view(B, SK), zc,
+                                 s_s[t].view(B, SK), w4t, graw_s[t], gp, discrete, unimix)
                 ext.scan2_b3(gp, g3_s[t], lnw3, lnb3, mr3_s[0, t], mr3_s[1, t], w3t,
                              gg3_s[t], glnw3, glnb3, gr_s[t])
                 ext.scan2_b2(g_h_seq[t], hc, gr_s[t, :, :H], y_s[t], hu_s[t, :, :H],

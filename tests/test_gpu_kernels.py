@@ -626,10 +626,11 @@ def test_mfma_fragment_layout_and_grid_barrier():
     (4, 16, 72, 6, 64, 2, 32, 64, 64),
     (3, 12, 60, 5, 64, 4, 16, 64, 128),
 ])
-def test_scan_v2_fused_phases_match_v1(dims):
-    """Round-2 fused-phase scan (4 multi-WG GEMM+epilogue launches per step,
-    ticket-exchange LN stats, recompute backward) vs the round-1
-    launch-per-op scan: same outputs and same gradients (bf16)."""
+@pytest.mark.parametrize("impl", ["v2", "v3"])
+def test_scan_v2_fused_phases_match_v1(dims, impl):
+    """Fused-phase scans (v2: N-split + ticket LN; v3: split-K + generation
+    tickets) vs the round-1 launch-per-op scan: same outputs and gradients
+    (bf16)."""
     import os
     from sheeprl_amd.algos.dreamer_v3.agent import RSSM, RecurrentModel
     from sheeprl_amd.models import MLP
@@ -656,7 +657,7 @@ def test_scan_v2_fused_phases_match_v1(dims):
     gm = torch.randn(T, B, SK, device="cuda")
 
     def run(v2: str):
-        os.environ["SHEEPRL_AMD_SCAN_V2"] = v2
+        os.environ["SHEEPRL_AMD_SCAN_IMPL"] = impl if v2 == "1" else "v1"
         try:
             for p in rssm.parameters():
                 p.grad = None
@@ -669,7 +670,7 @@ def test_scan_v2_fused_phases_match_v1(dims):
             grads = {n: p.grad.clone() for n, p in rssm.named_parameters() if p.grad is not None}
             return h.detach(), z.detach(), m.detach(), grads
         finally:
-            os.environ.pop("SHEEPRL_AMD_SCAN_V2", None)
+            os.environ.pop("SHEEPRL_AMD_SCAN_IMPL", None)
 
     h1, z1, m1, g1 = run("0")
     h2, z2, m2, g2 = run("1")
