@@ -66,7 +66,9 @@ class SACActor(nn.Module):
         """
         dist = self._dist(obs)
         y, logp = dist.rsample_with_log_prob()
-        action = y * self.action_scale + self.action_bias
+        # sampling/log-prob math runs in fp32 for stability; the action feeds
+        # the (possibly bf16) critics in the module dtype
+        action = (y * self.action_scale + self.action_bias).to(obs.dtype)
         log_prob = (logp - torch.log(self.action_scale.float()).expand_as(logp)).sum(-1, keepdim=True)
         return action, log_prob
 

@@ -57,7 +57,7 @@ def train(
         min_target = target_qs.min(dim=-1, keepdim=True).values - agent.alpha * next_logp
         next_qf_value = rewards + not_dones * cfg.algo.gamma * min_target
     qs = agent.get_q_values(obs, actions)
-    qf_loss = critic_loss(qs, next_qf_value, num_critics)
+    qf_loss = critic_loss(qs, next_qf_value.to(qs.dtype), num_critics)
     qf_optimizer.zero_grad(set_to_none=True)
     runtime.backward(qf_loss)
     qf_optimizer.step()

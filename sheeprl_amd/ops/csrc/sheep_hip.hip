@@ -4145,6 +4145,114 @@ __global__ void __launch_bounds__(256) scan3_b4_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// fused LSTM cell gates (ppo_recurrent): the
+// reference steps nn.LSTM per timestep in a Python loop; here the input
+// projection batches over T outside and each step is one gates kernel.
+// torch gate order i|f|g|o:  c' = σ(f)c + σ(i)tanh(g);  h = σ(o)tanh(c').
+// The episode-reset mask is folded in: c_used = (1-first)*c_prev.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void lstm_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ c_prev,
+                                      const T* __restrict__ first, T* __restrict__ h_out,
+                                      T* __restrict__ c_out, long N, int H) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < N * (long)H;
+       i += (long)gridDim.x * blockDim.x) {
+    const long n = i / H;
+    const int j = (int)(i - n * H);
+    const T* yr = y + n * 4 * (long)H;
+    float ig = 1.f / (1.f + expf(-ld(yr, j)));
+    float fg = 1.f / (1.f + expf(-ld(yr, H + j)));
+    float gg = tanhf(ld(yr, 2 * H + j));
+    float og = 1.f / (1.f + expf(-ld(yr, 3 * H + j)));
+    float cp = ld(c_prev, i) * (first ? (1.f - ld(first, n)) : 1.f);
+    float c = fg * cp + ig * gg;
+    st(c_out, i, c);
+    st(h_out, i, og * tanhf(c));
+  }
+}
+
+template <typename T>
+__global__ void lstm_gates_bwd_kernel(const T* __restrict__ gh, const T* __restrict__ gh2,
+                                      const T* __restrict__ first2, const T* __restrict__ gc_carry,
+                                      const T* __restrict__ y, const T* __restrict__ c_prev,
+                                      const T* __restrict__ c_out, const T* __restrict__ first,
+                                      T* __restrict__ gy, T* __restrict__ gc_prev, long N, int H) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < N * (long)H;
+       i += (long)gridDim.x * blockDim.x) {
+    const long n = i / H;
+    const int j = (int)(i - n * H);
+    const T* yr = y + n * 4 * (long)H;
+    float ig = 1.f / (1.f + expf(-ld(yr, j)));
+    float fg = 1.f / (1.f + expf(-ld(yr, H + j)));
+    float gg = tanhf(ld(yr, 2 * H + j));
+    float og = 1.f / (1.f + expf(-ld(yr, 3 * H + j)));
+    float mask = first ? (1.f - ld(first, n)) : 1.f;
+    float cp = ld(c_prev, i) * mask;
+    float c = ld(c_out, i);
+    float tc = tanhf(c);
+    float g = ld(gh, i) + (gh2 ? ld(gh2, i) * (first2 ? (1.f - ld(first2, n)) : 1.f) : 0.f);
+    float gc = g * og * (1.f - tc * tc) + (gc_carry ? ld(gc_carry, i) : 0.f);
+    float go = g * tc;
+    float gi = gc * gg;
+    float gf = gc * cp;
+    float gg_ = gc * ig;
+    T* gyr = gy + n * 4 * (long)H;
+    st(gyr, j, gi * ig * (1.f - ig));
+    st(gyr, H + j, gf * fg * (1.f - fg));
+    st(gyr, 2 * H + j, gg_ * (1.f - gg * gg));
+    st(gyr, 3 * H + j, go * og * (1.f - og));
+    st(gc_prev, i, gc * fg * mask);
+  }
+}
+
+void lstm_gates_fwd(const torch::Tensor& y, const torch::Tensor& c_prev,
+                    const c10::optional<torch::Tensor>& first, torch::Tensor h_out,
+                    torch::Tensor c_out) {
+  CHECK_IN(y);
+  TORCH_CHECK(c_prev.is_contiguous() && h_out.is_contiguous() && c_out.is_contiguous());
+  long N = c_prev.size(0) * (c_prev.dim() > 2 ? c_prev.size(1) : 1);
+  N = c_prev.numel() / c_prev.size(-1);
+  int H = (int)c_prev.size(-1);
+  long n = N * H;
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "lstm_gates_fwd", [&] {
+    using T = scalar_t;
+    const T* fp = first.has_value() ? (const T*)first->data_ptr() : nullptr;
+    hipLaunchKernelGGL(lstm_gates_fwd_kernel<T>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)y.data_ptr(), (const T*)c_prev.data_ptr(), fp, (T*)h_out.data_ptr(),
+                       (T*)c_out.data_ptr(), N, H);
+  });
+}
+
+void lstm_gates_bwd(const torch::Tensor& gh, const c10::optional<torch::Tensor>& gh2,
+                    const c10::optional<torch::Tensor>& first2,
+                    const c10::optional<torch::Tensor>& gc_carry, const torch::Tensor& y,
+                    const torch::Tensor& c_prev, const torch::Tensor& c_out,
+                    const c10::optional<torch::Tensor>& first, torch::Tensor gy,
+                    torch::Tensor gc_prev) {
+  CHECK_IN(y);
+  TORCH_CHECK(gy.is_contiguous() && gc_prev.is_contiguous());
+  long N = c_prev.numel() / c_prev.size(-1);
+  int H = (int)c_prev.size(-1);
+  long n = N * H;
+  int blocks = (int)std::min((n + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "lstm_gates_bwd", [&] {
+    using T = scalar_t;
+    const T* fp = first.has_value() ? (const T*)first->data_ptr() : nullptr;
+    const T* g2 = gh2.has_value() ? (const T*)gh2->data_ptr() : nullptr;
+    const T* f2 = first2.has_value() ? (const T*)first2->data_ptr() : nullptr;
+    const T* gcc = gc_carry.has_value() ? (const T*)gc_carry->data_ptr() : nullptr;
+    hipLaunchKernelGGL(lstm_gates_bwd_kernel<T>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)gh.data_ptr(), g2, f2, gcc, (const T*)y.data_ptr(),
+                       (const T*)c_prev.data_ptr(), (const T*)c_out.data_ptr(), fp,
+                       (T*)gy.data_ptr(), (T*)gc_prev.data_ptr(), N, H);
+  });
+}
+
+// ---------------------------------------------------------------------------
 // scan3 host wrappers
 // ---------------------------------------------------------------------------
 
@@ -4880,6 +4988,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reinforce_bwd", &reinforce_bwd);
   m.def("vloss2_fwd", &vloss2_fwd);
   m.def("vloss2_bwd", &vloss2_bwd);
+  m.def("lstm_gates_fwd", &lstm_gates_fwd);
+  m.def("lstm_gates_bwd", &lstm_gates_bwd);
   m.def("scan3_f1", &scan3_f1);
   m.def("scan3_f2", &scan3_f2);
   m.def("scan3_f3", &scan3_f3);

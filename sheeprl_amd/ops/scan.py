@@ -46,9 +46,11 @@ from sheeprl_amd.ops._ext import require_ext
 
 
 def _scan_impl() -> str:
-    """Fused-phase scan implementation: "v3" (split-K, default), "v2"
-    (N-split + ticket LN), or "v1" (launch-per-op with hipblaslt GEMMs)."""
-    return os.environ.get("SHEEPRL_AMD_SCAN_IMPL", "v3")
+    """Fused-phase scan implementation: "v1" (launch-per-op with hipblaslt
+    GEMMs — fastest measured in-graph; the guide's "many small graphed
+    launches win" rule), "v2" (N-split + ticket LN) or "v3" (split-K +
+    generation tickets), kept as measured experiments."""
+    return os.environ.get("SHEEPRL_AMD_SCAN_IMPL", "v1")
 
 
 class _RSSMScan(torch.autograd.Function):
