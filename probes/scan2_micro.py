@@ -282,3 +282,32 @@ timeit("scan3_f2", s3f2)
 timeit("scan3_f3", s3f3)
 timeit("scan3_f4", s3f4)
 timeit("scan3_b4", s3b4)
+
+# ---- weight-grad GEMM variants (the 5 batched [out, TB] x [TB, in] GEMMs
+# measured ~100 us each in-graph as mm(g.t(), x))
+TB = 1024
+gg3_big = bf(TB, P)
+r_big = bf(TB, H + E)
+gy_big = bf(TB, 3 * H)
+hu_big = bf(TB, H + D)
+
+
+def wg_tn():
+    torch.mm(gg3_big.t(), r_big)
+    torch.mm(gy_big.t(), hu_big)
+
+
+def wg_tr_nn():
+    torch.mm(ext.transpose2d(gg3_big), r_big)
+    torch.mm(ext.transpose2d(gy_big), hu_big)
+
+
+def wg_nt_t():
+    torch.mm(r_big.t(), gg3_big).t()
+    torch.mm(hu_big.t(), gy_big).t()
+
+
+print("== weight-grad GEMM variants (2 GEMMs each) ==")
+timeit("mm(g.t(), x)  TN", wg_tn)
+timeit("transpose2d + NN", wg_tr_nn)
+timeit("mm(x.t(), g).t()", wg_nt_t)

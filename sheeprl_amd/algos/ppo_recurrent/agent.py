@@ -62,8 +62,17 @@ class RecurrentModel(nn.Module):
     def masked_scan(
         self, x: Tensor, is_first: Tensor, states: Tuple[Tensor, Tensor]
     ) -> Tuple[Tensor, Tuple[Tensor, Tensor]]:
-        """Step the LSTM over T with per-step state reset where is_first=1."""
+        """Step the LSTM over T with per-step state reset where is_first=1.
+
+        On HIP the whole scan runs through the fused kernels (one batched
+        input GEMM + one gates launch per step, hand-written backward) —
+        SURVEY.md §2.8 item 12."""
         x = self.pre_mlp(x)
+        from sheeprl_amd.ops.lstm import lstm_scan, lstm_scan_applicable
+
+        if lstm_scan_applicable(self.lstm, x):
+            out, states = lstm_scan(x, is_first, states, self.lstm)
+            return self._post(out), states
         outs: List[Tensor] = []
         h, c = states
         for t in range(x.shape[0]):

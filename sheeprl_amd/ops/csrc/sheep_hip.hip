@@ -4658,27 +4658,41 @@ torch::Tensor vloss2_bwd(const torch::Tensor& g, const torch::Tensor& logits, co
 // ---------------------------------------------------------------------------
 
 template <typename T, int MODE>
+__device__ __forceinline__ float nll_elem(float p, float t) {
+  if (MODE == 0) {
+    float d = p - t;
+    return -d * d;
+  } else if (MODE == 1) {
+    float ts = copysignf(logf(fabsf(t) + 1.f), t);
+    float d = p - ts;
+    return -(p - ts) * d;
+  }
+  return t * p - (fmaxf(p, 0.f) + log1pf(expf(-fabsf(p))));
+}
+
+template <typename T, int MODE>
 __global__ void nll_fwd_kernel(const T* __restrict__ pred, const float* __restrict__ tgt,
                                float* __restrict__ out, long R, long D) {
   const int lane = threadIdx.x & 63;
   const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
+  constexpr int V = 16 / sizeof(T);  // pred vector width (8 bf16 / 4 fp32)
   for (long r = wave; r < R; r += nwaves) {
     const T* pr = pred + r * D;
     const float* tr = tgt + r * D;
     float acc = 0.f;
-    for (long j = lane; j < D; j += 64) {
-      float p = ld(pr, j), t = tr[j];
-      if (MODE == 0) {
-        float d = p - t;
-        acc -= d * d;
-      } else if (MODE == 1) {
-        float ts = copysignf(logf(fabsf(t) + 1.f), t);
-        float d = p - ts;
-        acc -= d * d;
-      } else {
-        acc += t * p - (fmaxf(p, 0.f) + log1pf(expf(-fabsf(p))));
+    if ((D & (V - 1)) == 0) {
+      for (long j0 = (long)lane * V; j0 < D; j0 += 64 * V) {
+        LnVec<T, V> pv;
+        pv.u = *(const uint4*)(pr + j0);
+        float tv[V];
+#pragma unroll
+        for (int e = 0; e < V; ++e) tv[e] = tr[j0 + e];
+#pragma unroll
+        for (int e = 0; e < V; ++e) acc += nll_elem<T, MODE>(ld(pv.e, e), tv[e]);
       }
+    } else {
+      for (long j = lane; j < D; j += 64) acc += nll_elem<T, MODE>(ld(pr, j), tr[j]);
     }
     acc = wave_sum(acc);
     if (lane == 0) out[r] = acc;
@@ -4711,7 +4725,8 @@ torch::Tensor nll_fwd(const torch::Tensor& pred, const torch::Tensor& tgt, long 
   TORCH_CHECK(tgt.is_contiguous() && tgt.scalar_type() == at::kFloat && tgt.numel() == pred.numel());
   long R = pred.numel() / D;
   auto out = torch::empty({R}, pred.options().dtype(at::kFloat));
-  int blocks = (int)std::min((R + 3) / 4, (long)2048);
+  // enough waves to cover the big image-reconstruction rows with slack
+  int blocks = (int)std::min(std::max((R + 3) / 4, (R * D + 32767) / 32768), (long)2048);
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, pred.scalar_type(), "nll_fwd", [&] {
     using T = scalar_t;

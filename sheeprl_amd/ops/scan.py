@@ -224,7 +224,15 @@ class _RSSMScan(torch.autograd.Function):
             and SK % 64 == 0
         )
         g16 = g16 and (H + E) >= 4096  # split-K pays off only at long K
-        if g16:
+        # long-K representation phase: the scan3 split-K GEMM+LN kernel
+        # measured 12.5 us in-graph vs 16.5 for splitk+ln_act (two launches)
+        g16_v3 = g16 and P % 16 == 0 and (H + E) % 8 == 0 and r_s.stride(2) == 1
+        if g16_v3:
+            s3_scr = torch.zeros(16, P, device=dev, dtype=torch.float32)
+            s3_tk = torch.zeros(max(P // 16, 1), device=dev, dtype=torch.int32)
+            s3_tk2 = torch.zeros(1, device=dev, dtype=torch.int32)
+            s3_ws = torch.zeros(T, 32, device=dev, dtype=torch.float32)
+        elif g16:
             sk_scratch = torch.zeros(16, P, device=dev, dtype=torch.float32)
             sk_tickets = torch.zeros(P // 64, device=dev, dtype=torch.int32)
             sk_ks = max(1, min(8, (H + E) // 832))
@@ -236,16 +244,17 @@ class _RSSMScan(torch.autograd.Function):
             # one kernel assembles the reset-masked GEMM inputs (z', a', h')
             ext.scan_resets_fwd(z_prev, iz, h_prev, ih, actions[t], f, x_s[t], hu_s[t], t == 0)
             if g16:
-                # split-K MFMA kernel for the long-K representation GEMM (the
-                # one shape where hand-written beats hipblaslt's in-graph
-                # pick: 10.5 us vs 12.5 us); hipblaslt keeps the mid-K GEMMs
                 torch.mm(x_s[t], w1t, out=g1_s[t])
                 ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
                 torch.mm(hu_s[t], w2t, out=y_s[t])
                 ext.gru_gates_fwd_o(y_s[t], hu_s[t, :, :H], lnwg, lnbg, eps,
                                     h_seq[t], r_s[t, :, :H], mrg_s[0, t], mrg_s[1, t])
-                ext.g16_splitk(r_s[t], w3, None, sk_scratch, sk_tickets, g3_s[t], sk_ks)
-                ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
+                if g16_v3:
+                    ext.scan3_f3(r_s[t], w3, lnw3, lnb3, p_s[t], g3_s[t], mr3_s[0, t],
+                                 mr3_s[1, t], s3_scr, s3_tk, s3_tk2, s3_ws[t], eps, t + 1)
+                else:
+                    ext.g16_splitk(r_s[t], w3, None, sk_scratch, sk_tickets, g3_s[t], sk_ks)
+                    ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
                 torch.addmm(b4, p_s[t], w4t, out=raw)
                 ext.cat_st_fwd_o(raw.view(B, S, discrete), urand_all[t], unimix,
                                  m_seq[t].view(B, S, discrete), z_seq[t].view(B, S, discrete), s_s[t])

@@ -650,7 +650,12 @@ def test_scan_v2_fused_phases_match_v1(dims, impl):
     actions = torch.randn(T, B, A, device="cuda", dtype=torch.bfloat16)
     is_first = (torch.rand(T, B, 1, device="cuda") < 0.3).float()
     is_first[0] = 1.0
-    urand = torch.rand(T, B, S, K, device="cuda")
+    # deterministic ST samples: per group one uniform is driven to ~1 so its
+    # gumbel noise dominates any ulp-level logit difference between the
+    # implementations (atomic float sums are order-nondeterministic)
+    urand = torch.full((T, B, S, K), 0.3, device="cuda")
+    pick = torch.randint(0, K, (T, B, S), device="cuda")
+    urand.scatter_(-1, pick.unsqueeze(-1), 1.0 - 1e-7)
 
     gh = torch.randn(T, B, H, device="cuda", dtype=torch.bfloat16)
     gz = torch.randn(T, B, SK, device="cuda", dtype=torch.bfloat16)
@@ -803,3 +808,55 @@ def test_transpose2d_matches_torch():
     for R, C in [(512, 1030), (1536, 1024), (512, 4608), (1024, 512), (7, 3)]:
         x = torch.randn(R, C, device="cuda", dtype=torch.bfloat16)
         assert torch.equal(ext.transpose2d(x), x.t().contiguous())
+
+
+@requires_gpu
+@pytest.mark.timeout(600)
+def test_fused_lstm_scan_matches_nn_lstm():
+    """ops.lstm.lstm_scan vs stepping nn.LSTM per-t with masked resets
+    (values + input/weight/state gradients, fp32)."""
+    from sheeprl_amd.ops.lstm import lstm_scan, lstm_scan_applicable
+
+    torch.manual_seed(0)
+    T, B, F, H = 9, 5, 12, 64
+    lstm = torch.nn.LSTM(F, H).cuda()
+    x = torch.randn(T, B, F, device="cuda", requires_grad=True)
+    is_first = (torch.rand(T, B, 1, device="cuda") < 0.25).float()
+    h0 = torch.randn(1, B, H, device="cuda", requires_grad=True)
+    c0 = torch.randn(1, B, H, device="cuda", requires_grad=True)
+    assert lstm_scan_applicable(lstm, x)
+
+    def eager():
+        outs = []
+        h, c = h0, c0
+        for t in range(T):
+            mask = (1.0 - is_first[t]).view(1, -1, 1)
+            h = h * mask
+            c = c * mask
+            out, (h, c) = lstm(x[t : t + 1], (h.contiguous(), c.contiguous()))
+            outs.append(out)
+        return torch.cat(outs, 0), h, c
+
+    ref, ref_h, ref_c = eager()
+    g = torch.randn_like(ref)
+    gh = torch.randn_like(ref_h)
+    gc = torch.randn_like(ref_c)
+    torch.autograd.backward([ref, ref_h, ref_c], [g, gh, gc])
+    ref_grads = {n: p.grad.clone() for n, p in lstm.named_parameters()}
+    gx_ref, gh0_ref, gc0_ref = x.grad.clone(), h0.grad.clone(), c0.grad.clone()
+    for p in lstm.parameters():
+        p.grad = None
+    x.grad = h0.grad = c0.grad = None
+
+    out, (h_last, c_last) = lstm_scan(x, is_first, (h0[0], c0[0]), lstm)
+    assert torch.allclose(out, ref, atol=1e-5, rtol=1e-5), (out - ref).abs().max()
+    assert torch.allclose(h_last, ref_h, atol=1e-5, rtol=1e-5)
+    assert torch.allclose(c_last, ref_c, atol=1e-5, rtol=1e-5)
+    torch.autograd.backward([out, h_last, c_last], [g, gh, gc])
+    assert torch.allclose(x.grad, gx_ref, atol=1e-4, rtol=1e-4), (x.grad - gx_ref).abs().max()
+    assert torch.allclose(h0.grad, gh0_ref, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(c0.grad, gc0_ref, atol=1e-4, rtol=1e-4)
+    for n, p in lstm.named_parameters():
+        assert torch.allclose(p.grad, ref_grads[n], atol=1e-3, rtol=1e-3), (
+            n, (p.grad - ref_grads[n]).abs().max()
+        )
