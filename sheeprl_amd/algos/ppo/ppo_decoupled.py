@@ -9,9 +9,14 @@ optimization group 1..N-1), flat-parameter weight sync :119-127/:302-305/
 MI355X mapping: the player rank pins env interaction +
 inference on its GPU; trainers DDP-train over an RCCL communicator spanning
 ranks 1..N-1; the flat-parameter broadcast player<->rank-1 rides one xGMI
-link.  Rollout chunks are padded to equal sizes so every trainer runs the
-same number of minibatch steps (replaces the reference's Join context —
-bucketed all-reduce needs matched call counts).
+link.  Rollout chunks travel as ONE device tensor per trainer
+(``Runtime.scatter_tensor`` — RCCL point-to-point over xGMI; the reference
+pickles python objects), with a 2-int control broadcast carrying the
+data/shutdown flag and the true row count.  Chunks are UNEVEN (reference
+semantics): trainers that exhaust their rows early run Join-equivalent
+zero-gradient sync rounds (``GradSync.sync_zero``) so bucketed all-reduce
+call counts stay matched while replicas remain bit-identical.  Objects are
+only used for the low-rate control plane (schema/metrics/checkpoints).
 """
 
 from __future__ import annotations
@@ -74,6 +79,10 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
     last_checkpoint = policy_step
     num_iters = max(1, total_steps // (rollout_steps * num_envs)) if not cfg.dry_run else 1
 
+    # tensor data plane setup: collectives ride RCCL on GPU, gloo on CPU
+    comm_dev = device if runtime.backend == "nccl" else torch.device("cpu")
+    schema = None
+
     obs, _ = envs.reset(seed=cfg.seed)
     for it in range(1, num_iters + 1):
         with timer("Time/env_interaction_time"):
@@ -121,15 +130,33 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
         data["returns"] = returns.reshape(-1, 1).cpu().numpy()
         data["advantages"] = advantages.reshape(-1, 1).cpu().numpy()
 
-        # scatter equal-size chunks to the trainers (pad by wrap-around)
+        # tensor data plane: uneven contiguous chunks packed into one flat
+        # fp32 tensor per trainer (equal buffer sizes for the scatter; each
+        # trainer slices its true row count from the control broadcast)
         n = data["returns"].shape[0]
-        per = int(np.ceil(n / n_trainers))
-        chunks: List[Any] = [None]
+        if schema is None:
+            schema = [(k, tuple(int(x) for x in data[k].shape[1:])) for k in sorted(data)]
+            runtime.broadcast_object_list([schema], src=0, group=world_group)
+        row_w = sum(int(np.prod(sh)) for _, sh in schema)
+        base, extra = divmod(n, n_trainers)
+        sizes = [base + (1 if t < extra else 0) for t in range(n_trainers)]
+        rows_max = max(sizes)
+        ctrl = torch.tensor([0, n], dtype=torch.int64, device=comm_dev)
+        runtime.broadcast(ctrl, src=0, group=world_group)
+        bufs = [torch.zeros(rows_max, row_w, dtype=torch.float32, device=comm_dev)]
+        start = 0
         for t in range(n_trainers):
-            idx = np.arange(t * per, (t + 1) * per) % n
-            chunks.append({k: v[idx] for k, v in data.items()})
-        out: List[Any] = [None]
-        runtime.scatter_object_list(out, chunks, src=0, group=world_group)
+            buf = torch.zeros(rows_max, row_w, dtype=torch.float32, device=comm_dev)
+            off = 0
+            for k, sh in schema:
+                w = int(np.prod(sh))
+                buf[: sizes[t], off : off + w] = torch.as_tensor(
+                    data[k][start : start + sizes[t]].reshape(sizes[t], w), device=comm_dev
+                )
+                off += w
+            bufs.append(buf)
+            start += sizes[t]
+        runtime.scatter_tensor(bufs[0], bufs, src=0, group=world_group)
 
         # receive updated weights from the lead trainer
         runtime.broadcast(flat, src=1, group=pt_group)
@@ -169,8 +196,11 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
         else:
             runtime.broadcast_object_list(["__noop__"], src=0, group=pt_group)
 
-    # shutdown sentinel to the trainers (reference :344)
-    runtime.scatter_object_list([None], [None] + [-1] * n_trainers, src=0, group=world_group)
+    # shutdown sentinel to the trainers (reference :344): control flag 1
+    if schema is None:
+        schema = []
+        runtime.broadcast_object_list([schema], src=0, group=world_group)
+    runtime.broadcast(torch.tensor([1, 0], dtype=torch.int64, device=comm_dev), src=0, group=world_group)
     envs.close()
     if cfg.algo.run_test:
         reward = test(ppo_player, make_env(cfg, cfg.seed, 0), cfg, log_dir, device)
@@ -212,31 +242,58 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
         runtime.broadcast(flat, src=1, group=pt_group)
 
     update = 0
+    comm_dev = device if runtime.backend == "nccl" else torch.device("cpu")
+    schema = None
+    n_trainers = runtime.world_size - 1
     while True:
-        out: List[Any] = [None]
-        runtime.scatter_object_list(out, None, src=0, group=world_group)
-        chunk = out[0]
-        if isinstance(chunk, int) and chunk == -1:  # shutdown (reference :463)
-            if rank == 1 and cfg.checkpoint.save_last:
-                pass
+        if schema is None:
+            payload: List[Any] = [None]
+            runtime.broadcast_object_list(payload, src=0, group=world_group)
+            schema = payload[0]
+        ctrl = torch.zeros(2, dtype=torch.int64, device=comm_dev)
+        runtime.broadcast(ctrl, src=0, group=world_group)
+        if int(ctrl[0].item()) == 1:  # shutdown (reference :463)
             return
+        n = int(ctrl[1].item())
+        base, extra = divmod(n, n_trainers)
+        sizes = [base + (1 if t < extra else 0) for t in range(n_trainers)]
+        my_rows = sizes[rank - 1]
+        rows_max = max(sizes)
+        row_w = sum(int(np.prod(sh)) for _, sh in schema)
         update += 1
-        data = {
-            k: torch.as_tensor(v, device=device) for k, v in chunk.items()
-        }
+        buf = torch.empty(rows_max, row_w, dtype=torch.float32, device=comm_dev)
+        runtime.scatter_tensor(buf, None, src=0, group=world_group)
+        if buf.device != device:
+            buf = buf.to(device)
+        data = {}
+        off = 0
+        for k, sh in schema:
+            w = int(np.prod(sh))
+            data[k] = buf[:my_rows, off : off + w].reshape(my_rows, *sh)
+            off += w
         obs = {k[len("obs_") :]: data[k] for k in data if k.startswith("obs_")}
         if not agent.is_continuous:
             data["actions"] = data["actions"].long()
 
         losses = {"policy": 0.0, "value": 0.0, "entropy": 0.0}
         n_batches = 0
-        idxs = np.arange(data["returns"].shape[0])
+        idxs = np.arange(my_rows)
         bs = cfg.algo.per_rank_batch_size
+        rounds_per_epoch = max(1, int(np.ceil(rows_max / bs)))
         for _ in range(cfg.algo.update_epochs):
             np.random.shuffle(idxs)
-            for start in range(0, len(idxs), bs):
+            for rnd in range(rounds_per_epoch):
+                start = rnd * bs
+                if start >= my_rows:
+                    # Join-equivalent: this trainer is out of rows; shadow the
+                    # peers' bucket all-reduces with zero grads and apply the
+                    # averaged step so replicas stay identical
+                    gs.sync_zero()
+                    if cfg.algo.max_grad_norm and cfg.algo.max_grad_norm > 0:
+                        runtime.clip_gradients(agent, optimizer, max_norm=cfg.algo.max_grad_norm)
+                    optimizer.step()
+                    continue
                 sel = idxs[start : start + bs]
-                # fixed batch count across trainers: chunks are padded equal
                 batch_obs = {k: v[sel] for k, v in obs.items()}
                 adv = data["advantages"][sel]
                 if cfg.algo.normalize_advantages and adv.numel() > 1:

@@ -93,6 +93,8 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
 
     obs_np, _ = envs.reset(seed=cfg.seed)
     obs = flat_obs(obs_np)
+    comm_dev = device if runtime.backend == "nccl" else torch.device("cpu")
+    schema = None
 
     for iter_num in range(1, total_iters + 1):
         policy_step += num_envs
@@ -134,14 +136,31 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
                     and policy_step - last_checkpoint >= cfg.checkpoint.every
                     or (iter_num == total_iters and cfg.checkpoint.save_last)
                 )
-                chunks: List[Any] = [None]
+                rows = cfg.algo.per_rank_batch_size * gradient_steps
+                samples = []
                 for _ in range(n_trainers):
-                    s = rb.sample(cfg.algo.per_rank_batch_size * gradient_steps)
-                    c = {k: v[0] for k, v in s.items()}
-                    c["__ckpt__"] = want_ckpt
-                    chunks.append(c)
-                out: List[Any] = [None]
-                runtime.scatter_object_list(out, chunks, src=0, group=world_group)
+                    s = rb.sample(rows)
+                    samples.append({k: v[0] for k, v in s.items()})
+                if schema is None:
+                    schema = [(k, tuple(int(x) for x in samples[0][k].shape[1:])) for k in sorted(samples[0])]
+                    runtime.broadcast_object_list([schema], src=0, group=world_group)
+                row_w = sum(int(np.prod(sh)) for _, sh in schema)
+                runtime.broadcast(
+                    torch.tensor([0, rows, int(want_ckpt)], dtype=torch.int64, device=comm_dev),
+                    src=0, group=world_group,
+                )
+                bufs = [torch.zeros(rows, row_w, dtype=torch.float32, device=comm_dev)]
+                for c in samples:
+                    buf = torch.empty(rows, row_w, dtype=torch.float32, device=comm_dev)
+                    off = 0
+                    for k, sh in schema:
+                        w = int(np.prod(sh))
+                        buf[:, off : off + w] = torch.as_tensor(
+                            np.ascontiguousarray(c[k]).reshape(rows, w), device=comm_dev
+                        )
+                        off += w
+                    bufs.append(buf)
+                runtime.scatter_tensor(bufs[0], bufs, src=0, group=world_group)
                 runtime.broadcast(flat, src=1, group=pt_group)
                 flat_to_params(flat, agent.actor.parameters())
                 payload: List[Any] = [None]
@@ -173,7 +192,10 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
             torch.save(state, ckpt_path, pickle_protocol=4)
             trainer_state = None
 
-    runtime.scatter_object_list([None], [None] + [-1] * n_trainers, src=0, group=world_group)
+    if schema is None:
+        schema = []
+        runtime.broadcast_object_list([schema], src=0, group=world_group)
+    runtime.broadcast(torch.tensor([1, 0, 0], dtype=torch.int64, device=comm_dev), src=0, group=world_group)
     envs.close()
     if cfg.algo.run_test:
         reward = test(sac_player, make_env(cfg, cfg.seed, 0), cfg, device)
@@ -223,16 +245,32 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
         runtime.broadcast(params_to_flat(agent.actor.parameters()).detach(), src=1, group=pt_group)
 
     update = 0
+    comm_dev = device if runtime.backend == "nccl" else torch.device("cpu")
+    schema = None
     while True:
-        out: List[Any] = [None]
-        runtime.scatter_object_list(out, None, src=0, group=world_group)
-        chunk = out[0]
-        if isinstance(chunk, int) and chunk == -1:
+        if schema is None:
+            payload0: List[Any] = [None]
+            runtime.broadcast_object_list(payload0, src=0, group=world_group)
+            schema = payload0[0]
+        ctrl = torch.zeros(3, dtype=torch.int64, device=comm_dev)
+        runtime.broadcast(ctrl, src=0, group=world_group)
+        if int(ctrl[0].item()) == 1:
             return
-        want_ckpt = bool(chunk.pop("__ckpt__", False))
-        data = {k: torch.as_tensor(np.ascontiguousarray(v), device=device).float() for k, v in chunk.items()}
+        rows = int(ctrl[1].item())
+        want_ckpt = bool(int(ctrl[2].item()))
+        row_w = sum(int(np.prod(sh)) for _, sh in schema)
+        buf = torch.empty(rows, row_w, dtype=torch.float32, device=comm_dev)
+        runtime.scatter_tensor(buf, None, src=0, group=world_group)
+        if buf.device != device:
+            buf = buf.to(device)
+        data = {}
+        off = 0
+        for k, sh in schema:
+            w = int(np.prod(sh))
+            data[k] = buf[:, off : off + w].reshape(rows, *sh)
+            off += w
         bs = cfg.algo.per_rank_batch_size
-        n = data["obs"].shape[0]
+        n = rows
         for start in range(0, n, bs):
             batch = {k: v[start : start + bs] for k, v in data.items()}
             update += 1

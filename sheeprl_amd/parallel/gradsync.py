@@ -90,6 +90,8 @@ class GradSync:
         self.world_size = world_size or dist.get_world_size(process_group)
         self.enabled = True
         self._any_ready = False
+        self._launch_order: List[int] = []  # bucket launch order of the last real round
+        self._any_launched = False
 
         params = [p for p in module.parameters() if p.requires_grad]
         cap = bucket_cap_mb * 1024 * 1024
@@ -125,6 +127,10 @@ class GradSync:
         self._any_ready = True
         bucket = self._param_bucket[id(p)]
         if bucket.mark_ready(p):
+            if not self._any_launched:
+                self._launch_order = []
+                self._any_launched = True
+            self._launch_order.append(self.buckets.index(bucket))
             bucket.launch(self.group, self.world_size)
 
     def finalize(self) -> None:
@@ -136,13 +142,35 @@ class GradSync:
         """
         if not self.enabled or not self._any_ready:
             return
-        for b in self.buckets:
+        for i, b in enumerate(self.buckets):
             if b.handle is None and b.ready > 0:
+                self._launch_order.append(i)
                 b.launch(self.group, self.world_size)
         for b in self.buckets:
             if b.handle is not None:
                 b.finish()
         self._any_ready = False
+        self._any_launched = False
+
+    def sync_zero(self) -> None:
+        """Join-equivalent shadow round for a trainer that ran out of local
+        minibatches (uneven rollout chunks, reference ppo_decoupled.py:497-499):
+        launches every bucket's all-reduce with ZERO local gradients, in the
+        same bucket order the peers' backward produces, waits, and writes the
+        averaged peer gradients back.  Followed by optimizer.step() this keeps
+        every replica bit-identical (unlike torch's Join, which leaves joined
+        ranks stale)."""
+        if not self.enabled or not dist.is_initialized():
+            return
+        # replay the last real round's bucket launch order so the zero round's
+        # collectives match the active peers' issue order
+        order = self._launch_order if len(self._launch_order) == len(self.buckets) else list(range(len(self.buckets)))
+        for i in order:
+            b = self.buckets[i]
+            b.flat.zero_()
+            b.handle = dist.all_reduce(b.flat, group=self.group, async_op=True)
+        for b in self.buckets:
+            b.finish()
 
     def broadcast_params(self, src: int = 0) -> None:
         """One flat broadcast per dtype instead of per-tensor (init-time sync;

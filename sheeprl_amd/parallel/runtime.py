@@ -276,6 +276,18 @@ class Runtime:
             out[0] = inputs[0]
         return out
 
+    def scatter_tensor(self, out: torch.Tensor, inputs: Optional[List[torch.Tensor]], src: int = 0,
+                       group: Any = None) -> torch.Tensor:
+        """Device-tensor scatter (equal shapes): RCCL point-to-point over xGMI
+        on GPU, gloo on CPU — the decoupled rollout data plane
+        (reference ppo_decoupled.py:294-299 used pickled objects)."""
+        if not self.is_distributed:
+            if inputs:
+                out.copy_(inputs[1] if len(inputs) > 1 else inputs[0])
+            return out
+        dist.scatter(out, scatter_list=inputs if self.global_rank == src else None, src=src, group=group)
+        return out
+
     def gather_object(self, obj: Any, dst: int = 0, group: Any = None) -> Optional[List[Any]]:
         if not self.is_distributed:
             return [obj]
