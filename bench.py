@@ -105,6 +105,9 @@ def _setup(cfg, rank: int, world_size: int):
         n_envs=cfg.env.num_envs,
         obs_keys=list(cfg.algo.cnn_keys.encoder),
         buffer_cls=SequentialReplayBuffer,
+        # pinned host ring: device-readable zero-copy for the HIP replay
+        # gather
+        pinned=runtime.use_cuda,
     )
     return runtime, envs, (world_model, actor, critic, target_critic, player), (
         world_optimizer, actor_optimizer, critic_optimizer), moments, rb
@@ -187,6 +190,17 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     use_prefetch = os.environ.get("SHEEPRL_AMD_PREFETCH", "0") == "1"
     prefetcher = DevicePrefetcher(_sample_host, device, depth=1) if use_prefetch else None
 
+    # device-side replay gather: HIP kernel pulls sequence windows from the
+    # pinned ring straight into HBM on a side stream, one batch ahead
+    gatherer = None
+    if device_type == "cuda" and os.environ.get("SHEEPRL_AMD_DEVICE_GATHER", "1") == "1" and prefetcher is None:
+        from sheeprl_amd.data.gather import DeviceReplayGather
+
+        try:
+            gatherer = DeviceReplayGather(rb, cfg.algo.per_rank_batch_size, seq_len, device)
+        except Exception as e:  # noqa: BLE001
+            print(f"[bench] device gather unavailable ({e}); host sampling", file=sys.stderr)
+
     def train_fn(batch):
         train(
             runtime, world_model, actor, critic, target_critic,
@@ -226,6 +240,9 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             with prefetcher.lock:
                 rb.add(step_data)
         else:
+            if gatherer is not None:
+                # a pending gather may read the ring row this add overwrites
+                gatherer.sync()
             rb.add(step_data)
         obs, rewards, term, trunc, infos = envs.step(real_actions)
         for k in cfg.algo.cnn_keys.encoder:
@@ -246,6 +263,8 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         _ops.ema_update_(list(target_critic.parameters()), list(critic.parameters()), cfg.algo.critic.tau)
         if prefetcher is not None:
             batch = prefetcher.next()
+        elif gatherer is not None:
+            batch = {k: v[0] for k, v in gatherer.next().items()}
         else:
             s = rb.sample_tensors(cfg.algo.per_rank_batch_size, sequence_length=seq_len, n_samples=1, device=device)
             batch = {k: v[0] for k, v in s.items()}

@@ -40,9 +40,10 @@ def build_arm(precision: str, fused: bool, seed: int = 11):
     ], "cuda")
     cfg.seed = seed
     runtime, envs, models, optims, moments, rb = bench._setup(cfg, 0, 1)
+    n_act = int(envs.single_action_space.n)
     obs, step_data = bench._prefill(cfg, envs, rb, n_steps=96)
     envs.close()
-    return cfg, runtime, models, optims, moments, rb
+    return cfg, runtime, models, optims, moments, rb, n_act
 
 
 @torch.no_grad()
@@ -114,7 +115,7 @@ def run(steps: int = 400, out_path: str = "profiles/loss_parity.json") -> dict:
         "eager_fp32": ("fp32", False, False),
         "fused_bf16_graphed": ("bf16", True, True),
     }.items():
-        cfg, runtime, models, optims, moments, rb = build_arm(precision, fused)
+        cfg, runtime, models, optims, moments, rb, n_act = build_arm(precision, fused)
         torch.manual_seed(123)
         sample_g = torch.Generator()
         sample_g.manual_seed(99)
@@ -122,7 +123,7 @@ def run(steps: int = 400, out_path: str = "profiles/loss_parity.json") -> dict:
         wo, ao, co = optims
         seq = cfg.algo.per_rank_sequence_length
         bsz = cfg.algo.per_rank_batch_size
-        actions_dim = [6]
+        actions_dim = [n_act]
         if eval_batch is None:
             s = rb.sample_tensors(bsz, sequence_length=seq, n_samples=1, device=runtime.device)
             eval_batch = {k: v[0].clone() for k, v in s.items()}

@@ -196,8 +196,8 @@ def train(
         lambda_values = compute_lambda_values(
             reward[1:], predicted_values[1:], continues[1:] * cfg.algo.gamma, lmbda=cfg.algo.lmbda
         )
-        critic["lambda_values"] = lambda_values
-        critic["continues"] = continues
+        critic["lambda_values"] = lambda_values.detach()
+        critic["continues"] = continues.detach()
         baseline = predicted_values[:-1]
         offset, invscale = critic["moments"](lambda_values, runtime)
         advantages.append(((lambda_values - offset) / invscale - (baseline - offset) / invscale)

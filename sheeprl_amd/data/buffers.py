@@ -62,6 +62,7 @@ class ReplayBuffer:
         obs_keys: Sequence[str] = ("observations",),
         memmap: bool = False,
         memmap_dir: Optional[str | Path] = None,
+        pinned: bool = False,
         **kwargs: Any,
     ) -> None:
         if buffer_size <= 0:
@@ -72,6 +73,12 @@ class ReplayBuffer:
         self._n_envs = n_envs
         self._obs_keys = tuple(obs_keys)
         self._memmap = memmap
+        # pinned host storage: the ring becomes device-readable zero-copy so
+        # the HIP replay-gather kernel can pull sequence windows straight into
+        # HBM; the numpy view shares the pinned
+        # torch storage, so every host-side add/sample path is unchanged
+        self._pinned = bool(pinned) and not memmap and torch.cuda.is_available()
+        self._pinned_t: Dict[str, torch.Tensor] = {}
         self._memmap_dir = Path(memmap_dir) if memmap_dir is not None else None
         if memmap and self._memmap_dir is not None:
             self._memmap_dir.mkdir(parents=True, exist_ok=True)
@@ -125,6 +132,16 @@ class ReplayBuffer:
         if self._memmap:
             fn = None if self._memmap_dir is None else self._memmap_dir / f"{key}.memmap"
             self._buf[key] = MemmapArray(shape, dtype, filename=fn)
+        elif self._pinned:
+            tdtype = {"uint8": torch.uint8, "float32": torch.float32, "float64": torch.float64,
+                      "int64": torch.int64, "int32": torch.int32, "bool": torch.bool}.get(
+                np.dtype(dtype).name)
+            if tdtype is None:
+                self._buf[key] = np.zeros(shape, dtype=dtype)
+            else:
+                t = torch.zeros(shape, dtype=tdtype, pin_memory=True)
+                self._pinned_t[key] = t
+                self._buf[key] = t.numpy()
         else:
             self._buf[key] = np.zeros(shape, dtype=dtype)
 
@@ -266,6 +283,7 @@ class SequentialReplayBuffer(ReplayBuffer):
         else:
             starts = np.random.randint(0, valid - L + 1, size=total)
         env_idxes = np.random.randint(0, self._n_envs, size=total)
+        self._last_picks = (starts.copy(), env_idxes.copy())  # device-gather hook
         win = (starts[:, None] + np.arange(L)[None, :]) % self._buffer_size  # [total, L]
         out: Dict[str, np.ndarray] = {}
         for k, v in self._buf.items():
@@ -404,6 +422,8 @@ class EpisodeBuffer:
         self._obs_keys = tuple(obs_keys)
         self._prioritize_ends = prioritize_ends
         self._memmap = memmap
+        self._pinned = bool(kwargs.get("pinned", False)) and not memmap and torch.cuda.is_available()
+        self._pinned_t: Dict[str, torch.Tensor] = {}
         self._memmap_dir = Path(memmap_dir) if memmap_dir is not None else None
         if memmap and self._memmap_dir is not None:
             self._memmap_dir.mkdir(parents=True, exist_ok=True)

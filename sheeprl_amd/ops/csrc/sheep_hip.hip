@@ -4145,6 +4145,64 @@ __global__ void __launch_bounds__(256) scan3_b4_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// device-side replay gather (BASELINE north star):
+// sequence windows are gathered by a HIP kernel reading the PINNED host ring
+// directly over PCIe (zero-copy) into HBM on a side stream — replacing numpy
+// fancy-indexing + host->device copies (reference buffers.py:493-511).
+// ptrs[s] = host base address of sample s's ring array for this key;
+// out[l, s, :] = ring[(starts[s] + l) % cap, :]  (time-major, matching the
+// [n_samples, L, batch, ...] layout sample_tensors produces).
+// ---------------------------------------------------------------------------
+
+template <int UNIT>
+__global__ void replay_gather_kernel(const long* __restrict__ ptrs, const long* __restrict__ starts,
+                                     unsigned char* __restrict__ out, long S, long L, long cap,
+                                     long row_bytes) {
+  const long row_units = row_bytes / UNIT;
+  const long total = S * L * row_units;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    const long u = i % row_units;
+    const long sl = i / row_units;
+    const long s = sl % S;
+    const long l = sl / S;
+    const long src_row = (starts[s] + l) % cap;
+    const unsigned char* src = (const unsigned char*)ptrs[s] + src_row * row_bytes + u * UNIT;
+    unsigned char* dst = out + (l * S + s) * row_bytes + u * UNIT;
+    if (UNIT == 16)
+      *(uint4*)dst = *(const uint4*)src;
+    else if (UNIT == 4)
+      *(unsigned int*)dst = *(const unsigned int*)src;
+    else
+      *dst = *src;
+  }
+}
+
+void replay_gather(const torch::Tensor& ptrs, const torch::Tensor& starts, torch::Tensor out, long L,
+                   long cap, long row_bytes) {
+  TORCH_CHECK(ptrs.is_cuda() && starts.is_cuda() && out.is_cuda() && out.is_contiguous());
+  TORCH_CHECK(ptrs.scalar_type() == at::kLong && starts.scalar_type() == at::kLong);
+  const long S = ptrs.numel();
+  TORCH_CHECK((long)out.numel() * out.element_size() == S * L * row_bytes, "replay_gather size");
+  const int unit = (row_bytes % 16 == 0) ? 16 : (row_bytes % 4 == 0 ? 4 : 1);
+  const long total = S * L * (row_bytes / unit);
+  int blocks = (int)std::min((total + kBlock - 1) / kBlock, (long)2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  if (unit == 16)
+    hipLaunchKernelGGL(replay_gather_kernel<16>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       ptrs.data_ptr<long>(), starts.data_ptr<long>(), (unsigned char*)out.data_ptr(),
+                       S, L, cap, row_bytes);
+  else if (unit == 4)
+    hipLaunchKernelGGL(replay_gather_kernel<4>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       ptrs.data_ptr<long>(), starts.data_ptr<long>(), (unsigned char*)out.data_ptr(),
+                       S, L, cap, row_bytes);
+  else
+    hipLaunchKernelGGL(replay_gather_kernel<1>, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       ptrs.data_ptr<long>(), starts.data_ptr<long>(), (unsigned char*)out.data_ptr(),
+                       S, L, cap, row_bytes);
+}
+
+// ---------------------------------------------------------------------------
 // fused LSTM cell gates (ppo_recurrent): the
 // reference steps nn.LSTM per timestep in a Python loop; here the input
 // projection batches over T outside and each step is one gates kernel.
@@ -5003,6 +5061,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("reinforce_bwd", &reinforce_bwd);
   m.def("vloss2_fwd", &vloss2_fwd);
   m.def("vloss2_bwd", &vloss2_bwd);
+  m.def("replay_gather", &replay_gather);
   m.def("lstm_gates_fwd", &lstm_gates_fwd);
   m.def("lstm_gates_bwd", &lstm_gates_bwd);
   m.def("scan3_f1", &scan3_f1);
