@@ -428,12 +428,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
     last_log = 0
     graphed_step = None
     _n_train_calls = 0
-    # p2e capture is OPT-IN for now (algo.hip_graphs=true): the captured
-    # multi-critic step segfaulted in round-1 validation; the single-family
-    # Dreamer mains capture by default
+    # round-2: capture is ON by default — the round-1 segfault was the
+    # behaviour tensors (lambda_values/continues) stored with live autograd
+    # graphs across steps, keeping stale AccumulateGrad nodes alive into the
+    # capture stream; they are stored detached now
     want_graphs = (
         (runtime.device.type == "cuda" or os.environ.get("SHEEPRL_AMD_FORCE_GRAPHS") == "1")
-        and cfg.algo.get("hip_graphs", False)
+        and cfg.algo.get("hip_graphs", True)
         and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1"
     )
     last_checkpoint = 0
