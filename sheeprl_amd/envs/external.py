@@ -139,7 +139,28 @@ register_env("mujoco", _make_gym)
 register_env("dmc", _make_dmc)
 register_env("crafter", _make_crafter)
 register_env("atari", _make_atari)
-register_env("minedojo", _missing("minedojo", "minedojo"))
-register_env("minerl", _missing("minerl", "minerl"))
-register_env("diambra", _missing("diambra", "diambra"))
+def _make_minedojo(**kwargs: Any) -> Env:
+    from sheeprl_amd.envs.minedojo_adapter import MineDojoAdapter
+
+    return MineDojoAdapter(**kwargs)
+
+
+def _make_minerl(**kwargs: Any) -> Env:
+    from sheeprl_amd.envs.minerl_adapter import MineRLAdapter
+
+    return MineRLAdapter(**kwargs)
+
+
+def _make_diambra(**kwargs: Any) -> Env:
+    from sheeprl_amd.envs.diambra_adapter import DiambraAdapter
+
+    return DiambraAdapter(**kwargs)
+
+
+# full adapters (action flattening / masks / settings plumbing implemented and
+# unit-tested against fakes); they raise an actionable ImportError when the
+# backing simulator package is absent from the image
+register_env("minedojo", _make_minedojo)
+register_env("minerl", _make_minerl)
+register_env("diambra", _make_diambra)
 register_env("super_mario_bros", _missing("super_mario_bros", "gym-super-mario-bros"))

@@ -172,3 +172,213 @@ def test_minedojo_masks(mdj):
     assert not obs["mask_equip_place"][ITEMS.index("oak_log")]
     assert obs["mask_destroy"][ITEMS.index("oak_log")]
     assert list(obs["mask_craft_smelt"]) == [True, True, False]
+
+
+# --------------------------------------------------------------------------
+# DIAMBRA
+# --------------------------------------------------------------------------
+
+class _FakeSpaceTypes:
+    DISCRETE = "discrete-type"
+    MULTI_DISCRETE = "multi-discrete-type"
+
+
+class _FakeRoles:
+    P1 = "p1-role"
+    P2 = "p2-role"
+
+
+class _FakeDiscrete:
+    def __init__(self, n):
+        self.n = n
+
+
+class _FakeMultiDiscrete:
+    def __init__(self, nvec):
+        self.nvec = np.asarray(nvec)
+
+
+class _FakeDictSpace:
+    def __init__(self, d):
+        self.spaces = d
+
+    def __getitem__(self, k):
+        return self.spaces[k]
+
+
+class FakeArenaEnv:
+    def __init__(self):
+        self.action_space = _FakeDiscrete(12)
+        self.observation_space = _FakeDictSpace({
+            "frame": spaces.Box(0, 255, (64, 64, 1), np.uint8),
+            "stage": _FakeDiscrete(4),
+            "P1_oppChar": _FakeMultiDiscrete([20, 20]),
+        })
+        self.steps = []
+
+    def reset(self, *, seed=None, options=None):
+        return {"frame": np.zeros((64, 64, 1), np.uint8), "stage": 2,
+                "P1_oppChar": np.array([3, 7])}, {}
+
+    def step(self, action):
+        self.steps.append(action)
+        done = {"env_done": len(self.steps) >= 2}
+        return ({"frame": np.zeros((64, 64, 1), np.uint8), "stage": 1,
+                 "P1_oppChar": np.array([1, 2])}, 1.5, False, False, done)
+
+    def close(self):
+        pass
+
+
+class FakeArena:
+    SpaceTypes = _FakeSpaceTypes
+    Roles = _FakeRoles
+
+    def __init__(self):
+        self.made = None
+        self.env = FakeArenaEnv()
+
+    def make(self, id, settings, wrappers, rank=0, render_mode="rgb_array", log_level=0):
+        self.made = {"id": id, "settings": dict(settings), "wrappers": dict(wrappers), "rank": rank}
+        return self.env
+
+
+def test_diambra_settings_plumbing():
+    from sheeprl_amd.envs.diambra_adapter import DiambraAdapter
+
+    arena = FakeArena()
+    env = DiambraAdapter("doapp", action_space="DISCRETE", screen_size=64, repeat_action=4,
+                         diambra_settings={"role": "P2", "difficulty": 3, "frame_shape": (9, 9, 9)},
+                         diambra_wrappers={"stack_actions": 2, "flatten": False},
+                         backend=arena)
+    st = arena.made["settings"]
+    assert st["game_id"] == "doapp"
+    assert st["role"] == _FakeRoles.P2
+    assert st["action_space"] == _FakeSpaceTypes.DISCRETE
+    assert st["step_ratio"] == 1          # sticky actions force engine ratio 1
+    assert st["frame_shape"] == (64, 64, 0)  # managed by the adapter
+    wr = arena.made["wrappers"]
+    assert wr["flatten"] is True and wr["repeat_action"] == 4 and wr["stack_actions"] == 2
+    assert isinstance(env.action_space, spaces.Discrete) and env.action_space.n == 12
+
+
+def test_diambra_obs_and_done_folding():
+    from sheeprl_amd.envs.diambra_adapter import DiambraAdapter
+
+    arena = FakeArena()
+    env = DiambraAdapter("doapp", backend=arena)
+    obs, info = env.reset()
+    assert info["env_domain"] == "DIAMBRA"
+    assert obs["stage"].shape == (1,) and obs["stage"][0] == 2  # Discrete -> Box[1]
+    assert obs["P1_oppChar"].shape == (2,)
+    # numpy discrete actions are unwrapped to python ints for the engine
+    env.step(np.array([5]))
+    assert arena.env.steps[-1] == 5 and isinstance(arena.env.steps[-1], int)
+    # env_done folds into terminated
+    _, r, terminated, truncated, _ = env.step(np.array(3))
+    assert terminated and not truncated and r == 1.5
+
+
+# --------------------------------------------------------------------------
+# MineRL
+# --------------------------------------------------------------------------
+
+class _FakeEnum:
+    def __init__(self, values):
+        self.values = values
+
+
+class FakeMineRLSim:
+    def __init__(self):
+        self.action_space = {
+            "attack": object(),
+            "forward": object(),
+            "jump": object(),
+            "camera": object(),
+            "craft": _FakeEnum(["none", "planks", "stick"]),
+            "equip": _FakeEnum(["none", "wooden_pickaxe"]),
+        }
+        self.observation_space = {
+            "inventory": {"dirt": 0, "planks": 0, "stick": 0, "air": 0, "wooden_pickaxe": 0},
+            "compass": {"angle": 0.0},
+            "equipped_items": {"mainhand": {"type": _FakeEnum(["air", "wooden_pickaxe", "other"])}},
+        }
+        self.steps = []
+        self._pov = np.zeros((64, 64, 3), np.uint8)
+
+    def _obs(self):
+        return {
+            "pov": self._pov,
+            "life_stats": {"life": 20.0, "food": 18.0, "air": 300.0},
+            "inventory": {"dirt": 5, "planks": 0, "stick": 2, "air": 1, "wooden_pickaxe": 1},
+            "compass": {"angle": np.array(42.0)},
+            "equipped_items": {"mainhand": {"type": "strange_item"}},
+        }
+
+    def reset(self):
+        return self._obs()
+
+    def step(self, action):
+        self.steps.append(copy.deepcopy(action))
+        return self._obs(), 0.5, False, {}
+
+
+import copy  # noqa: E402
+
+MINERL_ITEMS = ["air", "dirt", "planks", "stick", "wooden_pickaxe"]
+
+
+@pytest.fixture()
+def mrl():
+    from sheeprl_amd.envs.minerl_adapter import MineRLAdapter
+
+    sim = FakeMineRLSim()
+    env = MineRLAdapter(sim_env=sim, all_items=MINERL_ITEMS, sticky_attack=2, sticky_jump=2,
+                        break_speed_multiplier=1)
+    return env, sim
+
+
+def test_minerl_action_table(mrl):
+    env, sim = mrl
+    # 1 noop + attack + forward + jump + 4 camera + 2 craft + 1 equip = 11
+    assert env.action_space.n == 11
+    env.reset()
+    env.step(np.array([0]))
+    a = sim.steps[-1]
+    assert a["attack"] == 0 and a["forward"] == 0 and tuple(np.asarray(a["camera"])) == (0.0, 0.0)
+    env.step(np.array([3]))  # jump auto-adds forward
+    a = sim.steps[-1]
+    assert a["jump"] == 1 and a["forward"] == 1
+    env.step(np.array([8]))  # first craft value ("planks")
+    assert sim.steps[-1]["craft"] == "planks"
+    env.step(np.array([9]))
+    assert sim.steps[-1]["craft"] == "stick"
+
+
+def test_minerl_sticky_and_pitch(mrl):
+    env, sim = mrl
+    env.reset()
+    env.step(np.array([1]))          # attack -> sticky 2
+    assert sim.steps[-1]["attack"] == 1
+    env.step(np.array([3]))          # jump while sticky attack: jump suppressed
+    assert sim.steps[-1]["attack"] == 1 and sim.steps[-1]["jump"] == 0
+    # camera pitch limiting: -60..60 — the fifth -15 nudge would pass -60
+    for _ in range(4):
+        env.step(np.array([4]))      # pitch -15 (reaches exactly -60)
+    assert tuple(np.asarray(sim.steps[-1]["camera"])) == (-15.0, 0.0)
+    env.step(np.array([4]))
+    assert tuple(np.asarray(sim.steps[-1]["camera"])) == (0.0, 0.0)
+
+
+def test_minerl_obs(mrl):
+    env, sim = mrl
+    obs, _ = env.reset()
+    assert obs["rgb"].shape == (3, 64, 64)
+    assert obs["inventory"][MINERL_ITEMS.index("dirt")] == 5
+    assert obs["inventory"][MINERL_ITEMS.index("air")] == 1
+    assert np.allclose(obs["life_stats"], [20.0, 18.0, 300.0])
+    assert obs["compass"].shape == (1,) and obs["compass"][0] == 42.0
+    # unknown equipped item falls back to "air"
+    assert obs["equipment"][MINERL_ITEMS.index("air")] == 1
+    # max_inventory tracks the running max
+    assert np.array_equal(obs["max_inventory"], obs["inventory"])
