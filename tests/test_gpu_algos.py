@@ -153,3 +153,38 @@ def test_gpu_p2e_dv3_bf16(tmp_path):
                     "algo.ensembles.n=2", "algo.ensembles.dense_units=8",
                     "algo.ensembles.mlp_layers=1", *TINY_DV],
          precision="bf16")
+
+
+@requires_gpu
+@pytest.mark.timeout(300)
+def test_gpu_p2e_dv1_graphs(tmp_path):
+    """P2E-DV1 exploration with the (round-2 default) hipGraph capture."""
+    _run(tmp_path, ["exp=p2e_dv1_exploration", "algo.mlp_keys.encoder=[state]",
+                    "algo.ensembles.n=2", "algo.ensembles.dense_units=8",
+                    "algo.ensembles.mlp_layers=1",
+                    "algo.world_model.encoder.dense_units=8",
+                    "algo.world_model.recurrent_model.dense_units=8",
+                    "algo.world_model.observation_model.dense_units=8",
+                    "algo.world_model.reward_model.dense_units=8",
+                    "algo.actor.dense_units=8", "algo.actor.mlp_layers=1",
+                    "algo.critic.dense_units=8", "algo.critic.mlp_layers=1",
+                    "algo.world_model.stochastic_size=4",
+                    "algo.world_model.transition_model.hidden_size=8",
+                    "algo.world_model.representation_model.hidden_size=8",
+                    "algo.world_model.encoder.cnn_channels_multiplier=2",
+                    "algo.world_model.recurrent_model.recurrent_state_size=8",
+                    "algo.per_rank_batch_size=2", "algo.per_rank_sequence_length=4",
+                    "algo.horizon=3", "algo.total_steps=16", "algo.learning_starts=4",
+                    "algo.replay_ratio=0.5", "buffer.size=64", "algo.run_test=False"],
+         precision="bf16")
+
+
+@requires_gpu
+@pytest.mark.timeout(300)
+def test_gpu_p2e_dv2_graphs(tmp_path):
+    """P2E-DV2 exploration with the (round-2 default) hipGraph capture."""
+    _run(tmp_path, ["exp=p2e_dv2_exploration", "algo.mlp_keys.encoder=[state]",
+                    "algo.ensembles.n=2", "algo.ensembles.dense_units=8",
+                    "algo.ensembles.mlp_layers=1", "algo.per_rank_pretrain_steps=1",
+                    *TINY_DV],
+         precision="bf16")
