@@ -199,6 +199,8 @@ __global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restri
                                      float eps) {
   constexpr int V = 16 / sizeof(T);
   constexpr int RPW = 64 / L;
+  constexpr int UNROLL = 4;  // row-blocks in flight per wave: one 16-B load
+                             // per lane is latency-bound at ~3 TB/s
   const int lane = threadIdx.x & 63;
   const int lig = lane % L;
   const int grp = lane / L;
@@ -210,41 +212,50 @@ __global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restri
     wv[e] = ld(w, lig * V + e);
     bv[e] = ld(b, lig * V + e);
   }
-  for (long rb = wave * RPW; rb < R; rb += nwaves * RPW) {
-    const long row = rb + grp;
-    const bool active = row < R;
-    LnVec<T, V> xv;
-    float s = 0.f, s2 = 0.f;
-    if (active) {
-      xv.u = *reinterpret_cast<const uint4*>(x + row * (long)D + lig * V);
+  const long stride = nwaves * RPW;
+  for (long rb = wave * RPW; rb < R; rb += stride * UNROLL) {
+    LnVec<T, V> xv[UNROLL];
+    long rows[UNROLL];
 #pragma unroll
-      for (int e = 0; e < V; ++e) {
-        float v = ld(xv.e, e);
-        s += v;
-        s2 += v * v;
-      }
+    for (int u = 0; u < UNROLL; ++u) {
+      rows[u] = rb + u * stride + grp;
+      if (rows[u] < R) xv[u].u = *reinterpret_cast<const uint4*>(x + rows[u] * (long)D + lig * V);
     }
 #pragma unroll
-    for (int off = 1; off < L; off <<= 1) {
-      s += __shfl_xor(s, off, 64);
-      s2 += __shfl_xor(s2, off, 64);
-    }
-    if (active) {
-      float mean = s / D;
-      float var = s2 / D - mean * mean;
-      float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
-      if (lig == 0) {
-        mean_out[row] = mean;
-        rstd_out[row] = rstd;
-      }
-      LnVec<T, V> yv;
+    for (int u = 0; u < UNROLL; ++u) {
+      const long row = rows[u];
+      const bool active = row < R;
+      float s = 0.f, s2 = 0.f;
+      if (active) {
 #pragma unroll
-      for (int e = 0; e < V; ++e) {
-        float z = (ld(xv.e, e) - mean) * rstd * wv[e] + bv[e];
-        if (SILU) z = z / (1.f + expf(-z));
-        st(yv.e, e, z);
+        for (int e = 0; e < V; ++e) {
+          float v = ld(xv[u].e, e);
+          s += v;
+          s2 += v * v;
+        }
       }
-      *reinterpret_cast<uint4*>(y + row * (long)D + lig * V) = yv.u;
+#pragma unroll
+      for (int off = 1; off < L; off <<= 1) {
+        s += __shfl_xor(s, off, 64);
+        s2 += __shfl_xor(s2, off, 64);
+      }
+      if (active) {
+        float mean = s / D;
+        float var = s2 / D - mean * mean;
+        float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+        if (lig == 0) {
+          mean_out[row] = mean;
+          rstd_out[row] = rstd;
+        }
+        LnVec<T, V> yv;
+#pragma unroll
+        for (int e = 0; e < V; ++e) {
+          float z = (ld(xv[u].e, e) - mean) * rstd * wv[e] + bv[e];
+          if (SILU) z = z / (1.f + expf(-z));
+          st(yv.e, e, z);
+        }
+        *reinterpret_cast<uint4*>(y + row * (long)D + lig * V) = yv.u;
+      }
     }
   }
 }
@@ -275,46 +286,58 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
     gwa[e] = 0.f;
     gba[e] = 0.f;
   }
-  for (long rb = wave * RPW; rb < R; rb += nwaves * RPW) {
-    const long row = rb + grp;
-    const bool active = row < R;
-    float xh[V], gz[V];
-    float s1 = 0.f, s2 = 0.f;
-    float rs = 0.f;
-    if (active) {
-      LnVec<T, V> xv, gv;
-      xv.u = *reinterpret_cast<const uint4*>(x + row * (long)D + lig * V);
-      gv.u = *reinterpret_cast<const uint4*>(gy + row * (long)D + lig * V);
-      const float m = mean[row];
-      rs = rstd[row];
+  const long stride = nwaves * RPW;
+  constexpr int UNROLL = 2;  // 2 row-blocks x 2 tensors = 4 loads in flight
+  for (long rb = wave * RPW; rb < R; rb += stride * UNROLL) {
+    LnVec<T, V> xv[UNROLL], gv[UNROLL];
+    long rows[UNROLL];
 #pragma unroll
-      for (int e = 0; e < V; ++e) {
-        xh[e] = (ld(xv.e, e) - m) * rs;
-        float g = ld(gv.e, e);
-        if (SILU) {
-          float z = xh[e] * wv[e] + bv[e];
-          float sig = 1.f / (1.f + expf(-z));
-          g *= sig * (1.f + z * (1.f - sig));
-        }
-        gz[e] = g;
-        gwa[e] += g * xh[e];
-        gba[e] += g;
-        float gxhat = g * wv[e];
-        s1 += gxhat;
-        s2 += gxhat * xh[e];
+    for (int u = 0; u < UNROLL; ++u) {
+      rows[u] = rb + u * stride + grp;
+      if (rows[u] < R) {
+        xv[u].u = *reinterpret_cast<const uint4*>(x + rows[u] * (long)D + lig * V);
+        gv[u].u = *reinterpret_cast<const uint4*>(gy + rows[u] * (long)D + lig * V);
       }
     }
 #pragma unroll
-    for (int off = 1; off < L; off <<= 1) {
-      s1 += __shfl_xor(s1, off, 64);
-      s2 += __shfl_xor(s2, off, 64);
-    }
-    if (active) {
-      const float S1 = s1 / D, S2 = s2 / D;
-      LnVec<T, V> ov;
+    for (int u = 0; u < UNROLL; ++u) {
+      const long row = rows[u];
+      const bool active = row < R;
+      float xh[V], gz[V];
+      float s1 = 0.f, s2 = 0.f;
+      float rs = 0.f;
+      if (active) {
+        const float m = mean[row];
+        rs = rstd[row];
 #pragma unroll
-      for (int e = 0; e < V; ++e) st(ov.e, e, (gz[e] * wv[e] - S1 - xh[e] * S2) * rs);
-      *reinterpret_cast<uint4*>(gx + row * (long)D + lig * V) = ov.u;
+        for (int e = 0; e < V; ++e) {
+          xh[e] = (ld(xv[u].e, e) - m) * rs;
+          float g = ld(gv[u].e, e);
+          if (SILU) {
+            float z = xh[e] * wv[e] + bv[e];
+            float sig = 1.f / (1.f + expf(-z));
+            g *= sig * (1.f + z * (1.f - sig));
+          }
+          gz[e] = g;
+          gwa[e] += g * xh[e];
+          gba[e] += g;
+          float gxhat = g * wv[e];
+          s1 += gxhat;
+          s2 += gxhat * xh[e];
+        }
+      }
+#pragma unroll
+      for (int off = 1; off < L; off <<= 1) {
+        s1 += __shfl_xor(s1, off, 64);
+        s2 += __shfl_xor(s2, off, 64);
+      }
+      if (active) {
+        const float S1 = s1 / D, S2 = s2 / D;
+        LnVec<T, V> ov;
+#pragma unroll
+        for (int e = 0; e < V; ++e) st(ov.e, e, (gz[e] * wv[e] - S1 - xh[e] * S2) * rs);
+        *reinterpret_cast<uint4*>(gx + row * (long)D + lig * V) = ov.u;
+      }
     }
   }
   // fold groups within the wave (lanes sharing lane%L hold the same columns)
