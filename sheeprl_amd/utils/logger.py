@@ -18,12 +18,20 @@ from typing import Any, Dict, List, Optional
 
 
 class JsonlLogger:
-    def __init__(self, log_dir: str | Path) -> None:
+    def __init__(self, log_dir: str | Path, tensorboard: bool = True) -> None:
         self.log_dir = Path(log_dir)
         self.log_dir.mkdir(parents=True, exist_ok=True)
         self._path = self.log_dir / "metrics.jsonl"
         self._fh = open(self._path, "a", buffering=1)
         self._keys: List[str] = []
+        # parity with the reference's TensorBoard default logger
+        # (sheeprl/utils/logger.py:12): the in-repo tfevents writer needs no
+        # tensorboard package (utils/tboard.py)
+        self._tb = None
+        if tensorboard:
+            from sheeprl_amd.utils.tboard import TensorBoardWriter
+
+            self._tb = TensorBoardWriter(str(self.log_dir))
 
     @property
     def name(self) -> str:
@@ -37,6 +45,11 @@ class JsonlLogger:
             except (TypeError, ValueError):
                 rec[k] = str(v)
         self._fh.write(json.dumps(rec) + "\n")
+        if self._tb is not None:
+            self._tb.add_scalars(
+                {k: v for k, v in rec.items() if k != "step" and isinstance(v, float)},
+                int(step or 0),
+            )
 
     def log_hyperparams(self, params: Dict[str, Any]) -> None:
         with open(self.log_dir / "hparams.json", "w") as f:
@@ -61,6 +74,8 @@ class JsonlLogger:
     def close(self) -> None:
         self.finalize()
         self._fh.close()
+        if self._tb is not None:
+            self._tb.close()
 
 
 def get_log_dir(runtime: Any, root_dir: str, run_name: str, share: bool = True) -> str:
@@ -85,4 +100,5 @@ def get_logger(runtime: Any, cfg: Any, log_dir: str | Path) -> Optional[JsonlLog
         return None
     if cfg.metric.log_level <= 0:
         return None
-    return JsonlLogger(log_dir)
+    tb = bool(cfg.metric.get("tensorboard", True)) if hasattr(cfg.metric, "get") else True
+    return JsonlLogger(log_dir, tensorboard=tb)
