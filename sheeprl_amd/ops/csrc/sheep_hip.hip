@@ -1430,6 +1430,68 @@ void adam_step_mt(const torch::Tensor& ptrs, const torch::Tensor& sizes, const t
   });
 }
 
+// TF-style RMSprop, one launch for the whole optimizer (same chunk-table
+// scheme as adam_mt): v <- rho v + (1-rho) g^2; update g / sqrt(v + eps)
+// (eps INSIDE the sqrt); optional momentum and centered variants
+// (reference sheeprl/optim/rmsprop_tf.py:63-156).  ptrs rows:
+// [p, g, square_avg, momentum_buf, grad_avg] (unused slots 0).
+template <typename T>
+__global__ void rmsprop_mt_kernel(const long* __restrict__ ptrs, const long* __restrict__ sizes,
+                                  const int* __restrict__ ctid, const long* __restrict__ coff, long C,
+                                  float lr, float alpha, float eps, float wd, float momentum,
+                                  int centered, int zero_grad) {
+  const long c = blockIdx.x;
+  if (c >= C) return;
+  const int k = ctid[c];
+  const long off = coff[c];
+  const long n = sizes[k];
+  T* p = (T*)ptrs[5 * k];
+  T* g = (T*)ptrs[5 * k + 1];
+  float* sq = (float*)ptrs[5 * k + 2];
+  float* mb = (float*)ptrs[5 * k + 3];
+  float* ga = (float*)ptrs[5 * k + 4];
+  const long end = (off + kAdamChunk < n) ? off + kAdamChunk : n;
+  for (long i = off + threadIdx.x; i < end; i += blockDim.x) {
+    float gf = ld(g, i);
+    float pf = ld(p, i);
+    if (wd != 0.f) gf += wd * pf;
+    float s = sq[i] + (1.f - alpha) * (gf * gf - sq[i]);
+    sq[i] = s;
+    float avg;
+    if (centered) {
+      float a = ga[i] + (1.f - alpha) * (gf - ga[i]);
+      ga[i] = a;
+      avg = sqrtf(s - a * a + eps);  // reference does not clamp (rmsprop_tf.py:205)
+    } else {
+      avg = sqrtf(s + eps);
+    }
+    float upd;
+    if (momentum > 0.f) {
+      float b = momentum * mb[i] + gf / avg;
+      mb[i] = b;
+      upd = b;
+    } else {
+      upd = gf / avg;
+    }
+    st(p, i, pf - lr * upd);
+    if (zero_grad) st(g, i, 0.f);
+  }
+}
+
+void rmsprop_step_mt(const torch::Tensor& ptrs, const torch::Tensor& sizes, const torch::Tensor& ctid,
+                     const torch::Tensor& coff, const torch::Tensor& proto, double lr, double alpha,
+                     double eps, double wd, double momentum, bool centered, bool zero_grad) {
+  long C = ctid.numel();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, proto.scalar_type(), "rmsprop_mt", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((rmsprop_mt_kernel<T>), dim3((int)C), dim3(kBlock), 0, stream.stream(),
+                       ptrs.data_ptr<long>(), sizes.data_ptr<long>(), ctid.data_ptr<int>(),
+                       coff.data_ptr<long>(), C, (float)lr, (float)alpha, (float)eps, (float)wd,
+                       (float)momentum, centered ? 1 : 0, zero_grad ? 1 : 0);
+  });
+}
+
 void adam_step_dev(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
                    std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs, torch::Tensor step_t, double lr,
                    double b1, double b2, double eps, double wd) {
@@ -5053,6 +5115,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step);
   m.def("adam_step_dev", &adam_step_dev);
   m.def("adam_step_mt", &adam_step_mt);
+  m.def("rmsprop_step_mt", &rmsprop_step_mt);
   m.def("cat_st_fwd", &cat_st_fwd);
   m.def("cat_st_bwd", &cat_st_bwd);
   m.def("masked_lerp_fwd", &masked_lerp_fwd);

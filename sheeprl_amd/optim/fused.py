@@ -170,6 +170,8 @@ class RMSpropTF(torch.optim.Optimizer):
     (eps INSIDE the sqrt), square_avg initialized to ONES, momentum optional.
     """
 
+    _CHUNK = 4096  # must match kAdamChunk in the HIP kernel
+
     def __init__(
         self,
         params: Iterable,
@@ -182,6 +184,33 @@ class RMSpropTF(torch.optim.Optimizer):
     ) -> None:
         defaults = dict(lr=lr, alpha=alpha, eps=eps, weight_decay=weight_decay, momentum=momentum, centered=centered)
         super().__init__(params, defaults)
+        self._mt_cache = {}
+
+    def _chunk_table(self, group, params, grads, sqs, mbs, gas):
+        dev = params[0].device
+        ptr_sig = [p.data_ptr() for p in params] + [g.data_ptr() for g in grads]
+        ent = self._mt_cache.get(id(group))
+        if ent is not None and ent["sig"] == ptr_sig:
+            return ent
+        ptrs, sizes, ctid, coff = [], [], [], []
+        for k, (p, g, sq, mb, ga) in enumerate(zip(params, grads, sqs, mbs, gas)):
+            ptrs.append([p.data_ptr(), g.data_ptr(), sq.data_ptr(),
+                         mb.data_ptr() if mb is not None else 0,
+                         ga.data_ptr() if ga is not None else 0])
+            n = p.numel()
+            sizes.append(n)
+            for off in range(0, n, self._CHUNK):
+                ctid.append(k)
+                coff.append(off)
+        ent = {
+            "sig": ptr_sig,
+            "ptrs": torch.tensor(ptrs, dtype=torch.int64, device=dev),
+            "sizes": torch.tensor(sizes, dtype=torch.int64, device=dev),
+            "ctid": torch.tensor(ctid, dtype=torch.int32, device=dev),
+            "coff": torch.tensor(coff, dtype=torch.int64, device=dev),
+        }
+        self._mt_cache[id(group)] = ent
+        return ent
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -190,6 +219,31 @@ class RMSpropTF(torch.optim.Optimizer):
             with torch.enable_grad():
                 loss = closure()
         for group in self.param_groups:
+            # one-launch multi-tensor HIP path (same chunk-table scheme as
+            # FusedAdam)
+            mt = [p for p in group["params"] if p.grad is not None]
+            if mt and use_hip(mt[0]) and all(
+                p.dtype == mt[0].dtype and p.is_contiguous() and p.grad.is_contiguous() for p in mt
+            ):
+                sqs, mbs, gas = [], [], []
+                for p in mt:
+                    state = self.state[p]
+                    if len(state) == 0:
+                        state["square_avg"] = torch.ones_like(p, dtype=torch.float32)
+                        if group["momentum"] > 0:
+                            state["momentum_buffer"] = torch.zeros_like(p, dtype=torch.float32)
+                        if group["centered"]:
+                            state["grad_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                    sqs.append(state["square_avg"])
+                    mbs.append(state.get("momentum_buffer"))
+                    gas.append(state.get("grad_avg"))
+                ent = self._chunk_table(group, mt, [p.grad for p in mt], sqs, mbs, gas)
+                require_ext().rmsprop_step_mt(
+                    ent["ptrs"], ent["sizes"], ent["ctid"], ent["coff"], mt[0],
+                    group["lr"], group["alpha"], group["eps"], group["weight_decay"],
+                    group["momentum"], bool(group["centered"]), False,
+                )
+                continue
             for p in group["params"]:
                 if p.grad is None:
                     continue

@@ -899,3 +899,29 @@ def test_device_replay_gather_matches_host():
             ref = sub[win, 0]
             got = out[k][0, :, s].cpu().numpy()
             assert np.array_equal(got, ref), (k, s)
+
+
+@requires_gpu
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("momentum,centered", [(0.0, False), (0.9, False), (0.9, True)])
+def test_rmsprop_tf_mt_gpu(momentum, centered):
+    """Multi-tensor RMSpropTF kernel vs the eager CPU rollout."""
+    from sheeprl_amd.optim import RMSpropTF
+
+    torch.manual_seed(4)
+    p_ref = [torch.nn.Parameter(torch.randn(700)), torch.nn.Parameter(torch.randn(33, 5))]
+    p_gpu = [torch.nn.Parameter(p.detach().clone().cuda()) for p in p_ref]
+    o_ref = RMSpropTF(p_ref, lr=1e-2, momentum=momentum, centered=centered, weight_decay=1e-3)
+    o_gpu = RMSpropTF(p_gpu, lr=1e-2, momentum=momentum, centered=centered, weight_decay=1e-3)
+    for _ in range(8):
+        gs = [torch.randn_like(p) for p in p_ref]
+        for p, g in zip(p_ref, gs):
+            p.grad = g.clone()
+        for p, g in zip(p_gpu, gs):
+            p.grad = g.cuda()
+        o_ref.step()
+        o_gpu.step()
+    for a, b in zip(p_ref, p_gpu):
+        assert torch.allclose(b.detach().cpu(), a.detach(), atol=1e-5, rtol=1e-5), (
+            (b.detach().cpu() - a.detach()).abs().max()
+        )
