@@ -201,6 +201,58 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
         except Exception as e:  # noqa: BLE001
             print(f"[bench] device gather unavailable ({e}); host sampling", file=sys.stderr)
 
+    class GraphedPlayer:
+        """hipGraph-captured env-interaction forward: the per-step player
+        chain (encoder -> GRU cell -> representation -> actor) is ~30 small
+        kernels on a latency chain (~1.4 ms/step measured); one replay brings
+        it to a single graph launch.  States live in fixed buffers; episode
+        resets write them in place between replays."""
+
+        def __init__(self, player, torch_obs):
+            self.player = player
+            self.obs_keys = list(torch_obs.keys())
+            self.static_obs = {k: v.clone() for k, v in torch_obs.items()}
+            self.h = player.recurrent_state.clone()
+            self.z = player.stochastic_state.clone()
+            self.a = player.actions.clone()
+            self.out = None
+            rssm, actor_m, enc = player.rssm, player.actor, player.encoder
+
+            @torch.inference_mode()
+            def fwd():
+                emb = enc(self.static_obs)
+                rec = rssm.recurrent_model(torch.cat((self.z, self.a), -1), self.h)
+                _, stoch = rssm._representation(rec, emb)
+                z_new = stoch.view(*stoch.shape[:-2], -1)
+                acts, _ = actor_m(torch.cat((z_new, rec), -1))
+                a_new = torch.cat(acts, -1).to(z_new.dtype)
+                self.h.copy_(rec)
+                self.z.copy_(z_new)
+                self.a.copy_(a_new)
+                return a_new
+
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                for _ in range(2):
+                    out = fwd()
+            torch.cuda.current_stream().wait_stream(stream)
+            self.graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph):
+                self.out = fwd()
+
+        def get_actions(self, torch_obs):
+            for k in self.obs_keys:
+                self.static_obs[k].copy_(torch_obs[k])
+            self.graph.replay()
+            return [self.out[0]]
+
+        def init_states(self, reset_envs):
+            rec, stoch = self.player.rssm.get_initial_states((1, len(reset_envs)))
+            self.a[:, reset_envs] = 0.0
+            self.h[:, reset_envs] = rec.to(self.h.dtype)
+            self.z[:, reset_envs] = stoch.reshape(1, len(reset_envs), -1).to(self.z.dtype)
+
     def train_fn(batch):
         train(
             runtime, world_model, actor, critic, target_critic,
@@ -228,11 +280,25 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             if rank == 0:
                 print(f"[bench] hipGraph capture failed ({e}); running eager", file=sys.stderr)
 
+    # graph-capture the player forward too (opt-out: SHEEPRL_AMD_NO_GRAPHS)
+    gplayer = None
+    if graphed is not None and not player.actor.is_continuous and len(actions_dim) == 1:
+        try:
+            with torch.no_grad():
+                _tobs = prepare_obs(runtime, obs, cnn_keys=list(cfg.algo.cnn_keys.encoder), num_envs=num_envs)
+                gplayer = GraphedPlayer(player, _tobs)
+            if rank == 0:
+                print("[bench] player forward captured in a hipGraph", file=sys.stderr)
+        except Exception as e:  # noqa: BLE001
+            gplayer = None
+            if rank == 0:
+                print(f"[bench] player graph capture failed ({e}); eager player", file=sys.stderr)
+
     def one_iter(obs, step_data):
         # --- env interaction (the real policy forward + env step + buffer add)
         with torch.inference_mode():
             torch_obs = prepare_obs(runtime, obs, cnn_keys=list(cfg.algo.cnn_keys.encoder), num_envs=num_envs)
-            acts = player.get_actions(torch_obs)
+            acts = (gplayer or player).get_actions(torch_obs)
             actions = torch.cat(acts, -1).view(num_envs, -1).float().cpu().numpy()
             real_actions = torch.stack([a.argmax(dim=-1) for a in acts], dim=-1).view(num_envs).cpu().numpy()
         step_data["actions"] = actions[None]
@@ -256,7 +322,7 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             # player states were produced under inference_mode; resetting them
             # in place must happen under it too
             with torch.inference_mode():
-                player.init_states(np.nonzero(dones)[0].tolist())
+                (gplayer or player).init_states(np.nonzero(dones)[0].tolist())
         # --- one gradient step (replay_ratio=1 at num_envs=1)
         from sheeprl_amd import ops as _ops
 
