@@ -34,7 +34,7 @@ from sheeprl_amd.config import save_config
 from sheeprl_amd.data import EnvIndependentReplayBuffer, SequentialReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
 from sheeprl_amd.models import MLP, MultiDecoder, MultiEncoder
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -486,11 +486,9 @@ def main(runtime: Runtime, cfg: Any) -> None:
         runtime, actions_dim, is_continuous, cfg, obs_space,
         state.get("world_model"), state.get("actor"), state.get("critic"),
     )
-    world_optimizer = FusedAdam(world_model.parameters(), lr=cfg.algo.world_model.optimizer.lr,
-                                eps=cfg.algo.world_model.optimizer.eps)
-    actor_optimizer = FusedAdam(actor.parameters(), lr=cfg.algo.actor.optimizer.lr, eps=cfg.algo.actor.optimizer.eps)
-    critic_optimizer = FusedAdam(critic.parameters(), lr=cfg.algo.critic.optimizer.lr,
-                                 eps=cfg.algo.critic.optimizer.eps)
+    world_optimizer = make_optimizer(world_model.parameters(), cfg.algo.world_model.optimizer)
+    actor_optimizer = make_optimizer(actor.parameters(), cfg.algo.actor.optimizer)
+    critic_optimizer = make_optimizer(critic.parameters(), cfg.algo.critic.optimizer)
 
     aggregator = MetricAggregator({k: "mean" for k in AGGREGATOR_KEYS})
     rb = EnvIndependentReplayBuffer(

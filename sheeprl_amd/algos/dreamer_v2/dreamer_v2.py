@@ -24,7 +24,7 @@ from sheeprl_amd.algos.dreamer_v3.utils import prepare_obs, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import EnvIndependentReplayBuffer, EpisodeBuffer, SequentialReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -322,14 +322,9 @@ def main(runtime: Runtime, cfg: Any) -> None:
         cfg.algo.actor.get("expl_min", 0.0),
         cfg.algo.actor.get("expl_decay", 0.0),
     )
-    world_optimizer = FusedAdam(world_model.parameters(), lr=cfg.algo.world_model.optimizer.lr,
-                                eps=cfg.algo.world_model.optimizer.eps,
-                                weight_decay=cfg.algo.world_model.optimizer.get("weight_decay", 0.0))
-    actor_optimizer = FusedAdam(actor.parameters(), lr=cfg.algo.actor.optimizer.lr, eps=cfg.algo.actor.optimizer.eps,
-                                weight_decay=cfg.algo.actor.optimizer.get("weight_decay", 0.0))
-    critic_optimizer = FusedAdam(critic.parameters(), lr=cfg.algo.critic.optimizer.lr,
-                                 eps=cfg.algo.critic.optimizer.eps,
-                                 weight_decay=cfg.algo.critic.optimizer.get("weight_decay", 0.0))
+    world_optimizer = make_optimizer(world_model.parameters(), cfg.algo.world_model.optimizer)
+    actor_optimizer = make_optimizer(actor.parameters(), cfg.algo.actor.optimizer)
+    critic_optimizer = make_optimizer(critic.parameters(), cfg.algo.critic.optimizer)
 
     aggregator = MetricAggregator({k: "mean" for k in AGGREGATOR_KEYS})
 

@@ -44,7 +44,7 @@ from sheeprl_amd.distributions import (
     TwoHotEncodingDistribution,
 )
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -432,18 +432,9 @@ def main(runtime: Runtime, cfg: Any) -> None:
         state.get("target_critic"),
     )
 
-    world_optimizer = FusedAdam(
-        world_model.parameters(),
-        lr=cfg.algo.world_model.optimizer.lr,
-        eps=cfg.algo.world_model.optimizer.eps,
-        weight_decay=cfg.algo.world_model.optimizer.get("weight_decay", 0) or 0,
-    )
-    actor_optimizer = FusedAdam(
-        actor.parameters(), lr=cfg.algo.actor.optimizer.lr, eps=cfg.algo.actor.optimizer.eps
-    )
-    critic_optimizer = FusedAdam(
-        critic.parameters(), lr=cfg.algo.critic.optimizer.lr, eps=cfg.algo.critic.optimizer.eps
-    )
+    world_optimizer = make_optimizer(world_model.parameters(), cfg.algo.world_model.optimizer)
+    actor_optimizer = make_optimizer(actor.parameters(), cfg.algo.actor.optimizer)
+    critic_optimizer = make_optimizer(critic.parameters(), cfg.algo.critic.optimizer)
     if "world_optimizer" in state:
         world_optimizer.load_state_dict(state["world_optimizer"])
         actor_optimizer.load_state_dict(state["actor_optimizer"])

@@ -26,7 +26,7 @@ from sheeprl_amd.config import save_config
 from sheeprl_amd.data import EnvIndependentReplayBuffer, SequentialReplayBuffer
 from sheeprl_amd.distributions import BernoulliSafeMode, MSEDistribution, SymlogDistribution, TwoHotEncodingDistribution
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -394,18 +394,13 @@ def main(runtime: Runtime, cfg: Any) -> None:
         state.get("actor_exploration"), state.get("critics_exploration"),
     )
 
-    world_optimizer = FusedAdam(world_model.parameters(), lr=cfg.algo.world_model.optimizer.lr,
-                                eps=cfg.algo.world_model.optimizer.eps)
-    ensemble_optimizer = FusedAdam(ensembles.parameters(), lr=cfg.algo.ensembles.optimizer.lr,
-                                   eps=cfg.algo.ensembles.optimizer.eps)
-    actor_task_optimizer = FusedAdam(actor_task.parameters(), lr=cfg.algo.actor.optimizer.lr,
-                                     eps=cfg.algo.actor.optimizer.eps)
-    critic_task_optimizer = FusedAdam(critic_task.parameters(), lr=cfg.algo.critic.optimizer.lr,
-                                      eps=cfg.algo.critic.optimizer.eps)
-    actor_expl_optimizer = FusedAdam(actor_exploration.parameters(), lr=cfg.algo.actor.optimizer.lr,
-                                     eps=cfg.algo.actor.optimizer.eps)
+    world_optimizer = make_optimizer(world_model.parameters(), cfg.algo.world_model.optimizer)
+    ensemble_optimizer = make_optimizer(ensembles.parameters(), cfg.algo.ensembles.optimizer)
+    actor_task_optimizer = make_optimizer(actor_task.parameters(), cfg.algo.actor.optimizer)
+    critic_task_optimizer = make_optimizer(critic_task.parameters(), cfg.algo.critic.optimizer)
+    actor_expl_optimizer = make_optimizer(actor_exploration.parameters(), cfg.algo.actor.optimizer)
     critics_expl_optimizers = {
-        name: FusedAdam(c["module"].parameters(), lr=cfg.algo.critic.optimizer.lr, eps=cfg.algo.critic.optimizer.eps)
+        name: make_optimizer(c["module"].parameters(), cfg.algo.critic.optimizer)
         for name, c in critics_exploration.items()
     }
     moments_task = Moments(

@@ -23,7 +23,7 @@ from sheeprl_amd.data import ReplayBuffer
 from sheeprl_amd.envs import make_env, vectorize_env
 from sheeprl_amd.envs import spaces
 from sheeprl_amd.ops import gae as compute_gae
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -102,12 +102,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
         state = runtime.load(cfg.checkpoint.resume_from)
 
     agent, player = build_agent(runtime, obs_space, action_space, cfg, state.get("agent"))
-    optimizer = FusedAdam(
-        agent.parameters(),
-        lr=cfg.algo.optimizer.lr,
-        eps=cfg.algo.optimizer.get("eps", 1e-8),
-        betas=tuple(cfg.algo.optimizer.get("betas", (0.9, 0.999))),
-    )
+    optimizer = make_optimizer(agent.parameters(), cfg.algo.optimizer)
     if "optimizer" in state:
         optimizer.load_state_dict(state["optimizer"])
 

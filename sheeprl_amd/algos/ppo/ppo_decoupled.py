@@ -35,7 +35,7 @@ from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
 from sheeprl_amd.ops import gae as compute_gae
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.parallel.gradsync import GradSync
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
@@ -229,7 +229,7 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
     gs.broadcast_params(src=1)
     agent._grad_sync = gs  # noqa: SLF001
     runtime._synced_modules.append(agent)
-    optimizer = FusedAdam(agent.parameters(), lr=cfg.algo.optimizer.lr, eps=cfg.algo.optimizer.get("eps", 1e-8))
+    optimizer = make_optimizer(agent.parameters(), cfg.algo.optimizer)
     if cfg.checkpoint.resume_from:
         # rank 1 saved its optimizer state; every trainer resumes from it
         payload: List[Any] = [resume_opt]

@@ -22,7 +22,7 @@ from sheeprl_amd.algos.sac.utils import AGGREGATOR_KEYS, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -112,12 +112,9 @@ def main(runtime: Runtime, cfg: Any) -> None:
         state = runtime.load(cfg.checkpoint.resume_from)
 
     agent, player = build_agent(runtime, cfg, obs_space, action_space, state.get("agent"))
-    qf_optimizer = FusedAdam(agent.qfs.parameters(), lr=cfg.algo.critic.optimizer.lr,
-                             eps=cfg.algo.critic.optimizer.get("eps", 1e-8))
-    actor_optimizer = FusedAdam(agent.actor.parameters(), lr=cfg.algo.actor.optimizer.lr,
-                                eps=cfg.algo.actor.optimizer.get("eps", 1e-8))
-    alpha_optimizer = FusedAdam([agent.log_alpha], lr=cfg.algo.alpha.optimizer.lr,
-                                eps=cfg.algo.alpha.optimizer.get("eps", 1e-8))
+    qf_optimizer = make_optimizer(agent.qfs.parameters(), cfg.algo.critic.optimizer)
+    actor_optimizer = make_optimizer(agent.actor.parameters(), cfg.algo.actor.optimizer)
+    alpha_optimizer = make_optimizer([agent.log_alpha], cfg.algo.alpha.optimizer)
     for name, opt in (("qf_optimizer", qf_optimizer), ("actor_optimizer", actor_optimizer),
                       ("alpha_optimizer", alpha_optimizer)):
         if name in state:

@@ -21,7 +21,7 @@ from sheeprl_amd.algos.sac.utils import AGGREGATOR_KEYS, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.parallel.gradsync import GradSync
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
@@ -223,9 +223,9 @@ def trainer(runtime: Runtime, cfg: Any, world_group, pt_group, opt_group) -> Non
     runtime._synced_modules.append(agent)
     runtime._default_group = opt_group  # scalar all-reduces stay on the trainer group
 
-    qf_optimizer = FusedAdam(agent.qfs.parameters(), lr=cfg.algo.critic.optimizer.lr)
-    actor_optimizer = FusedAdam(agent.actor.parameters(), lr=cfg.algo.actor.optimizer.lr)
-    alpha_optimizer = FusedAdam([agent.log_alpha], lr=cfg.algo.alpha.optimizer.lr)
+    qf_optimizer = make_optimizer(agent.qfs.parameters(), cfg.algo.critic.optimizer)
+    actor_optimizer = make_optimizer(agent.actor.parameters(), cfg.algo.actor.optimizer)
+    alpha_optimizer = make_optimizer([agent.log_alpha], cfg.algo.alpha.optimizer)
     if cfg.checkpoint.resume_from:
         # rank 1 loaded the checkpoint; every trainer resumes from its states
         payload: List[Any] = [

@@ -22,7 +22,7 @@ from sheeprl_amd.algos.sac_ae.agent import build_agent
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
-from sheeprl_amd.optim import FusedAdam
+from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -139,11 +139,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
         state = runtime.load(cfg.checkpoint.resume_from)
 
     agent, player = build_agent(runtime, cfg, obs_space, action_space, state.get("agent"))
-    encoder_optimizer = FusedAdam(agent.encoder.parameters(), lr=cfg.algo.encoder.optimizer.lr)
-    decoder_optimizer = FusedAdam(agent.decoder.parameters(), lr=cfg.algo.decoder.optimizer.lr)
-    qf_optimizer = FusedAdam(agent.qfs.parameters(), lr=cfg.algo.critic.optimizer.lr)
-    actor_optimizer = FusedAdam(agent.actor.parameters(), lr=cfg.algo.actor.optimizer.lr)
-    alpha_optimizer = FusedAdam([agent.log_alpha], lr=cfg.algo.alpha.optimizer.lr)
+    encoder_optimizer = make_optimizer(agent.encoder.parameters(), cfg.algo.encoder.optimizer)
+    decoder_optimizer = make_optimizer(agent.decoder.parameters(), cfg.algo.decoder.optimizer)
+    qf_optimizer = make_optimizer(agent.qfs.parameters(), cfg.algo.critic.optimizer)
+    actor_optimizer = make_optimizer(agent.actor.parameters(), cfg.algo.actor.optimizer)
+    alpha_optimizer = make_optimizer([agent.log_alpha], cfg.algo.alpha.optimizer)
 
     aggregator = MetricAggregator({k: "mean" for k in AGGREGATOR_KEYS})
     num_envs = cfg.env.num_envs

@@ -1,3 +1,3 @@
-from sheeprl_amd.optim.fused import FusedAdam, RMSpropTF
+from sheeprl_amd.optim.fused import FusedAdam, RMSpropTF, make_optimizer
 
-__all__ = ["FusedAdam", "RMSpropTF"]
+__all__ = ["FusedAdam", "RMSpropTF", "make_optimizer"]

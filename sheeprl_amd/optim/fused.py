@@ -165,6 +165,44 @@ class FusedAdam(torch.optim.Optimizer):
         return loss
 
 
+def make_optimizer(params: Iterable, opt_cfg) -> torch.optim.Optimizer:
+    """Build the optimizer named by an algo config's ``optimizer`` block.
+
+    Parity: the reference selects the optimizer class per module via its
+    ``configs/optim/{adam,rmsprop,sgd}.yaml`` Hydra group
+    (sheeprl/configs/optim/adam.yaml:2, rmsprop.yaml:2); here the same choice
+    is the ``name`` field of each algo's ``optimizer`` block
+    (adam | rmsprop_tf | sgd).  Unknown names fail loudly.
+    """
+    name = str(opt_cfg.get("name", "adam")).lower()
+    lr = float(opt_cfg.lr)
+    wd = float(opt_cfg.get("weight_decay", 0.0) or 0.0)  # configs may carry null
+    if name in ("adam", "fused_adam", "adamw"):
+        return FusedAdam(
+            params, lr=lr,
+            betas=tuple(opt_cfg.get("betas", (0.9, 0.999))),
+            eps=float(opt_cfg.get("eps", 1e-8)),
+            weight_decay=wd,
+        )
+    if name in ("rmsprop", "rmsprop_tf", "rmsproptf"):
+        return RMSpropTF(
+            params, lr=lr,
+            alpha=float(opt_cfg.get("alpha", 0.9)),
+            eps=float(opt_cfg.get("eps", 1e-10)),
+            weight_decay=wd,
+            momentum=float(opt_cfg.get("momentum", 0.0)),
+            centered=bool(opt_cfg.get("centered", False)),
+        )
+    if name == "sgd":
+        return torch.optim.SGD(
+            params, lr=lr,
+            momentum=float(opt_cfg.get("momentum", 0.0)),
+            weight_decay=wd,
+            nesterov=bool(opt_cfg.get("nesterov", False)),
+        )
+    raise ValueError(f"unknown optimizer name {name!r} (expected adam | rmsprop_tf | sgd)")
+
+
 class RMSpropTF(torch.optim.Optimizer):
     """TF-style RMSprop: v <- rho v + (1-rho) g^2; update = g / sqrt(v + eps)
     (eps INSIDE the sqrt), square_avg initialized to ONES, momentum optional.

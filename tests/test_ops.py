@@ -288,3 +288,25 @@ def test_ppo_value_loss_clipped_half_factor():
     assert torch.allclose(v, 0.5 * torch.max(unc, cl).mean())
     # unclipped branch stays plain MSE
     assert torch.allclose(value_loss(nv, ov, ret, 0.2, False), torch.nn.functional.mse_loss(nv, ret))
+
+
+def test_make_optimizer_honours_config_name():
+    """The algo configs' ``optimizer.name`` field selects the optimizer class
+    (reference parity: sheeprl configs/optim Hydra group)."""
+    from sheeprl_amd.optim import FusedAdam, RMSpropTF, make_optimizer
+    from sheeprl_amd.utils.dotdict import DotDict
+
+    p = [torch.nn.Parameter(torch.randn(4))]
+    o = make_optimizer(p, DotDict({"name": "adam", "lr": 3e-4, "eps": 1e-5}))
+    assert isinstance(o, FusedAdam)
+    assert o.param_groups[0]["lr"] == 3e-4 and o.param_groups[0]["eps"] == 1e-5
+
+    o = make_optimizer(p, DotDict({"name": "rmsprop_tf", "lr": 1e-2, "alpha": 0.95, "momentum": 0.9}))
+    assert isinstance(o, RMSpropTF)
+    assert o.param_groups[0]["alpha"] == 0.95 and o.param_groups[0]["momentum"] == 0.9
+
+    o = make_optimizer(p, DotDict({"name": "sgd", "lr": 1e-1, "weight_decay": None}))
+    assert isinstance(o, torch.optim.SGD) and o.param_groups[0]["weight_decay"] == 0.0
+
+    with pytest.raises(ValueError):
+        make_optimizer(p, DotDict({"name": "lamb", "lr": 1e-3}))
