@@ -1492,6 +1492,71 @@ void rmsprop_step_mt(const torch::Tensor& ptrs, const torch::Tensor& sizes, cons
   });
 }
 
+// Fused gradient clipping by global norm (torch.nn.utils.clip_grad_norm_
+// semantics: coef = max_norm / (norm + 1e-6), clamped to 1) over the same
+// flat chunk table as the optimizers; ptrs rows here are just [g].  Two
+// launches replace the ~10-launch foreach path: (1) block-sum + one
+// atomicAdd/block of the squared norm into out[0] (out[1] pre-zeroed holds
+// the step's generation; out zeroed host-side / by a fill before), (2) a
+// second kernel — the launch boundary orders it after ALL norm blocks —
+// scales every grad in place and lets block 0 publish the norm to out[1].
+template <typename T>
+__global__ void gradsq_mt_kernel(const long* __restrict__ ptrs, const long* __restrict__ sizes,
+                                 const int* __restrict__ ctid, const long* __restrict__ coff, long C,
+                                 float* __restrict__ out) {
+  __shared__ float lds[16];
+  const long c = blockIdx.x;
+  if (c >= C) return;
+  const int k = ctid[c];
+  const long off = coff[c];
+  const long n = sizes[k];
+  const T* g = (const T*)ptrs[k];
+  const long end = (off + kAdamChunk < n) ? off + kAdamChunk : n;
+  float s = 0.f;
+  for (long i = off + threadIdx.x; i < end; i += blockDim.x) {
+    float gf = ld(g, i);
+    s += gf * gf;
+  }
+  s = block_sum(s, lds);
+  if (threadIdx.x == 0) atomicAdd(out, s);
+}
+
+template <typename T>
+__global__ void clip_apply_mt_kernel(const long* __restrict__ ptrs, const long* __restrict__ sizes,
+                                     const int* __restrict__ ctid, const long* __restrict__ coff, long C,
+                                     const float* __restrict__ sq, float* __restrict__ norm_out,
+                                     float max_norm) {
+  const long c = blockIdx.x;
+  if (c >= C) return;
+  const float norm = sqrtf(sq[0]);
+  if (c == 0 && threadIdx.x == 0) norm_out[0] = norm;
+  float coef = max_norm / (norm + 1e-6f);
+  if (coef >= 1.f) return;  // no-op when already under the bound
+  const int k = ctid[c];
+  const long off = coff[c];
+  const long n = sizes[k];
+  T* g = (T*)ptrs[k];
+  const long end = (off + kAdamChunk < n) ? off + kAdamChunk : n;
+  for (long i = off + threadIdx.x; i < end; i += blockDim.x) st(g, i, ld(g, i) * coef);
+}
+
+void clip_grad_norm_mt(const torch::Tensor& ptrs, const torch::Tensor& sizes, const torch::Tensor& ctid,
+                       const torch::Tensor& coff, torch::Tensor out, const torch::Tensor& proto,
+                       double max_norm) {
+  long C = ctid.numel();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, proto.scalar_type(), "clip_grad_norm_mt", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((gradsq_mt_kernel<T>), dim3((int)C), dim3(kBlock), 0, stream.stream(),
+                       ptrs.data_ptr<long>(), sizes.data_ptr<long>(), ctid.data_ptr<int>(),
+                       coff.data_ptr<long>(), C, out.data_ptr<float>());
+    hipLaunchKernelGGL((clip_apply_mt_kernel<T>), dim3((int)C), dim3(kBlock), 0, stream.stream(),
+                       ptrs.data_ptr<long>(), sizes.data_ptr<long>(), ctid.data_ptr<int>(),
+                       coff.data_ptr<long>(), C, out.data_ptr<float>(), out.data_ptr<float>() + 1,
+                       (float)max_norm);
+  });
+}
+
 void adam_step_dev(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
                    std::vector<torch::Tensor> ms, std::vector<torch::Tensor> vs, torch::Tensor step_t, double lr,
                    double b1, double b2, double eps, double wd) {
@@ -5116,6 +5181,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step_dev", &adam_step_dev);
   m.def("adam_step_mt", &adam_step_mt);
   m.def("rmsprop_step_mt", &rmsprop_step_mt);
+  m.def("clip_grad_norm_mt", &clip_grad_norm_mt);
   m.def("cat_st_fwd", &cat_st_fwd);
   m.def("cat_st_bwd", &cat_st_bwd);
   m.def("masked_lerp_fwd", &masked_lerp_fwd);

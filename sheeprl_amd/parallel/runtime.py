@@ -222,8 +222,60 @@ class Runtime:
             torch.nn.utils.clip_grad_value_(params, clip_val)
             return None
         if max_norm is not None:
+            fused = self._clip_grad_norm_fused(optimizer, params, float(max_norm))
+            if fused is not None:
+                return fused
             return torch.nn.utils.clip_grad_norm_(params, max_norm, error_if_nonfinite=error_if_nonfinite)
         return None
+
+    def _clip_grad_norm_fused(self, optimizer, params, max_norm: float) -> Optional[torch.Tensor]:
+        """Global-norm clipping as 3 launches (zero-fill, squared-norm
+        accumulate, scale) over the optimizers' flat chunk table, replacing
+        the ~10-launch torch foreach path inside the captured train step.
+        Grad pointers are stable step-to-step (FusedAdam keeps grads alive
+        for in-kernel zeroing) so the table is built once and re-validated
+        by pointer signature."""
+        from sheeprl_amd.ops._ext import get_ext, use_hip
+
+        grads = [p.grad for p in params if p.grad is not None]
+        if not grads or not use_hip(grads[0]):
+            return None
+        ext = get_ext()
+        if ext is None or not hasattr(ext, "clip_grad_norm_mt"):
+            return None
+        dt = grads[0].dtype
+        if any(g.dtype != dt or not (g.is_contiguous() or g.is_contiguous(memory_format=torch.channels_last))
+               for g in grads):
+            return None
+        cache = getattr(self, "_clip_cache", None)
+        if cache is None:
+            cache = self._clip_cache = {}
+        sig = [g.data_ptr() for g in grads]
+        ent = cache.get(id(optimizer))
+        if ent is None or ent["sig"] != sig:
+            chunk = 4096  # kAdamChunk
+            dev = grads[0].device
+            ptrs, sizes, ctid, coff = [], [], [], []
+            for k, g in enumerate(grads):
+                ptrs.append(g.data_ptr())
+                n = g.numel()
+                sizes.append(n)
+                for off in range(0, n, chunk):
+                    ctid.append(k)
+                    coff.append(off)
+            ent = {
+                "sig": sig,
+                "ptrs": torch.tensor(ptrs, dtype=torch.int64, device=dev),
+                "sizes": torch.tensor(sizes, dtype=torch.int64, device=dev),
+                "ctid": torch.tensor(ctid, dtype=torch.int32, device=dev),
+                "coff": torch.tensor(coff, dtype=torch.int64, device=dev),
+                "out": torch.zeros(2, dtype=torch.float32, device=dev),
+            }
+            cache[id(optimizer)] = ent
+        out = ent["out"]
+        out.zero_()
+        ext.clip_grad_norm_mt(ent["ptrs"], ent["sizes"], ent["ctid"], ent["coff"], out, grads[0], max_norm)
+        return out[1]
 
     # ------------------------------------------------------------------
     # collectives

@@ -925,3 +925,42 @@ def test_rmsprop_tf_mt_gpu(momentum, centered):
         assert torch.allclose(b.detach().cpu(), a.detach(), atol=1e-5, rtol=1e-5), (
             (b.detach().cpu() - a.detach()).abs().max()
         )
+
+
+@requires_gpu
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("scale", [0.01, 100.0])  # above / below the bound
+def test_fused_clip_grad_norm(dtype, scale):
+    """Runtime._clip_grad_norm_fused vs torch.nn.utils.clip_grad_norm_ (same
+    coef = max_norm/(norm+1e-6) rule, clamped to 1)."""
+    from sheeprl_amd.optim import FusedAdam
+    from sheeprl_amd.parallel import Runtime
+
+    torch.manual_seed(0)
+    params = [torch.nn.Parameter(torch.randn(n, device="cuda", dtype=dtype) * scale)
+              for n in (7, 4096, 9000)]
+    ref = [p.detach().clone() for p in params]
+    for p in params:
+        p.grad = p.detach().clone()
+    for r in ref:
+        r.requires_grad_(True)
+        r.grad = r.detach().clone()
+
+    rt = Runtime.__new__(Runtime)  # clipping needs no process-group state
+    opt = FusedAdam(params, lr=0.0)
+    max_norm = 10.0
+    got = rt.clip_gradients(None, opt, max_norm=max_norm)
+    expect = torch.nn.utils.clip_grad_norm_(ref, max_norm)
+    assert got is not None
+    tol = 1e-3 if dtype is torch.float32 else 2e-2
+    assert torch.allclose(got.float().cpu(), expect.float().cpu(), rtol=tol), (got, expect)
+    for p, r in zip(params, ref):
+        assert torch.allclose(p.grad.float(), r.grad.float(), rtol=tol, atol=tol * max(1.0, scale))
+
+    # second call must reuse the cached table and stay correct
+    for p, r in zip(params, ref):
+        p.grad.copy_(torch.randn_like(p) * scale)
+        r.grad.copy_(p.grad.to(r.dtype))
+    got2 = rt.clip_gradients(None, opt, max_norm=max_norm)
+    expect2 = torch.nn.utils.clip_grad_norm_(ref, max_norm)
+    assert torch.allclose(got2.float().cpu(), expect2.float().cpu(), rtol=tol)
