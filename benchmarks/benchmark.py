@@ -9,8 +9,11 @@ in `bench.py` at the repo root; this script is for quick cross-algorithm
 comparisons on the reference's own benchmark presets.
 """
 
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 from sheeprl_amd.cli import run
 

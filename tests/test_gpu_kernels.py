@@ -687,12 +687,14 @@ def test_scan_v2_fused_phases_match_v1(dims, impl):
     assert set(g1) == set(g2)
     for n in g1:
         a, b = g1[n].float(), g2[n].float()
-        denom = a.abs().max().clamp_min(1e-3)
-        rel = (a - b).abs().max() / denom
-        # the learned initial states accumulate tiny f-masked grads over T;
-        # v1 rounds the h-side GEMM output to bf16 before the add while the
-        # fused epilogues accumulate in fp32 — measured up to ~15% relative
-        # on these near-zero grads (absolute diffs are bf16-ulp level)
+        # norm-relative error: robust to single-element cancellation noise
+        # (per-element sums of ± bf16 terms over T×B rows can be near zero
+        # while the tensor is large); v1 rounds the h-side GEMM output to
+        # bf16 before the add while the fused epilogues accumulate in fp32,
+        # so element-max rel diffs of ~15% on such entries are rounding, not
+        # bugs — the energy of the difference is what distinguishes an
+        # implementation error
+        rel = (a - b).norm() / a.norm().clamp_min(1e-3)
         tol = 2.5e-1 if "initial" in n else 5e-2
         assert rel < tol, (n, rel.item(), a.abs().max().item())
 
