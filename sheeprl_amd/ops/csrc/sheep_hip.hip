@@ -1358,6 +1358,106 @@ __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __rest
   for (int j = lane; j < K; j += 64) st(oh, j, j == best_j ? 1.f : 0.f);
 }
 
+// cat_st forward fused with the NEXT scan step's reset-masked input
+// assembly: the z' block of x_{t+1} is the one-hot this kernel just sampled
+// (masked in-register, no extra read), each (b,s) wave also writes its
+// H/S-slice of the h' block of hu_{t+1} from h_t, and the s==S-1 wave adds
+// the masked-action tail — removing the standalone scan_resets_fwd launch
+// from steps 1..T-1 (~5 us/step on the DV3-S bench).
+template <typename T>
+__global__ void cat_st_resets_fwd_kernel(
+    const T* __restrict__ raw, const float* __restrict__ urand, float* __restrict__ m_out,
+    T* __restrict__ onehot, float* __restrict__ s_out, long nrows, int K, float unimix,
+    const T* __restrict__ iz, const T* __restrict__ h_cur, const T* __restrict__ ih,
+    const T* __restrict__ act, const T* __restrict__ f, T* __restrict__ x_next, long xs,
+    T* __restrict__ hu_next, long hus, int S, int A, int H) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= nrows) return;
+  const long b = row / S;
+  const int s = (int)(row - b * S);
+  const T* L = raw + row * (long)K;
+  float* mr = m_out + row * (long)K;
+  float* sr = s_out + row * (long)K;
+  T* oh = onehot + row * (long)K;
+  const float fb = ld(f, b);
+  const int SK = S * K;
+  float lmax = -1e30f;
+  for (int j = lane; j < K; j += 64) lmax = fmaxf(lmax, ld(L, j));
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) lmax = fmaxf(lmax, __shfl_xor(lmax, off, 64));
+  float lsum = 0.f;
+  for (int j = lane; j < K; j += 64) lsum += expf(ld(L, j) - lmax);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, 64);
+  const float inv = 1.f / lsum;
+  float best = -1e30f;
+  int best_j = 0;
+  for (int j = lane; j < K; j += 64) {
+    float sv = expf(ld(L, j) - lmax) * inv;
+    float p = (1.f - unimix) * sv + unimix / K;
+    float m = logf(p);
+    sr[j] = sv;
+    mr[j] = m;
+    float u = urand[row * (long)K + j];
+    float t = fmaxf(-logf(fmaxf(u, 1e-20f)), 1e-20f);
+    float score = m - logf(t);
+    if (score > best) {
+      best = score;
+      best_j = j;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    float ob = __shfl_xor(best, off, 64);
+    int oj = __shfl_xor(best_j, off, 64);
+    if (ob > best || (ob == best && oj < best_j)) {
+      best = ob;
+      best_j = oj;
+    }
+  }
+  for (int j = lane; j < K; j += 64) {
+    const float z = j == best_j ? 1.f : 0.f;
+    st(oh, j, z);
+    st(x_next, b * xs + s * K + j, (1.f - fb) * z + fb * ld(iz, b * (long)SK + s * K + j));
+  }
+  if (s == S - 1)
+    for (int j = lane; j < A; j += 64) st(x_next, b * xs + SK + j, (1.f - fb) * ld(act, b * (long)A + j));
+  const int HS = H / S;  // caller guarantees H % S == 0
+  for (int j = lane; j < HS; j += 64) {
+    const long c = (long)s * HS + j;
+    st(hu_next, b * hus + c, (1.f - fb) * ld(h_cur, b * (long)H + c) + fb * ld(ih, b * (long)H + c));
+  }
+}
+
+void cat_st_resets_fwd(const torch::Tensor& raw, const torch::Tensor& urand, double unimix,
+                       torch::Tensor m, torch::Tensor onehot, torch::Tensor s_out,
+                       const torch::Tensor& iz, const torch::Tensor& h_cur, const torch::Tensor& ih,
+                       const torch::Tensor& act, const torch::Tensor& f, torch::Tensor x_next,
+                       torch::Tensor hu_next) {
+  CHECK_IN(raw);
+  TORCH_CHECK(m.is_contiguous() && s_out.is_contiguous() && onehot.is_contiguous(), "cat_st_resets outputs");
+  int K = (int)raw.size(-1);
+  int S = (int)raw.size(-2);
+  long nrows = raw.numel() / K;
+  int A = (int)act.size(1);
+  int H = (int)ih.size(1);
+  TORCH_CHECK(H % S == 0, "cat_st_resets_fwd requires H % S == 0");
+  TORCH_CHECK(S * K == (int)iz.size(1), "iz must be [B, S*K]");
+  const int rows_per_block = kBlock / 64;
+  int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, raw.scalar_type(), "cat_st_resets_fwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((cat_st_resets_fwd_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       (const T*)raw.data_ptr(), urand.data_ptr<float>(), m.data_ptr<float>(),
+                       (T*)onehot.data_ptr(), s_out.data_ptr<float>(), nrows, K, (float)unimix,
+                       (const T*)iz.data_ptr(), (const T*)h_cur.data_ptr(), (const T*)ih.data_ptr(),
+                       (const T*)act.data_ptr(), (const T*)f.data_ptr(), (T*)x_next.data_ptr(),
+                       x_next.stride(0), (T*)hu_next.data_ptr(), hu_next.stride(0), S, A, H);
+  });
+}
+
 // backward: given gm (grad wrt m=log p) and gon (grad wrt the ST sample whose
 // gradient path is p), produce grad wrt raw logits.
 // g_L_j = (1-unimix) * s_j * [ (gm_j/p_j + gon_j) - sum_i (gm_i/p_i + gon_i) * s_i ]
@@ -5395,6 +5495,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step_mt", &adam_step_mt);
   m.def("rmsprop_step_mt", &rmsprop_step_mt);
   m.def("clip_grad_norm_mt", &clip_grad_norm_mt);
+  m.def("cat_st_resets_fwd", &cat_st_resets_fwd);
   m.def("cat_st_fwd", &cat_st_fwd);
   m.def("cat_st_bwd", &cat_st_bwd);
   m.def("masked_lerp_fwd", &masked_lerp_fwd);

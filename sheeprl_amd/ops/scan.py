@@ -237,12 +237,15 @@ class _RSSMScan(torch.autograd.Function):
             sk_tickets = torch.zeros(P // 64, device=dev, dtype=torch.int32)
             sk_ks = max(1, min(8, (H + E) // 832))
         w1t, w2t, w3t, w4t = w1.t(), w2.t(), w3.t(), w4.t()
+        # cat_st fused with the NEXT step's reset assembly removes the
+        # standalone scan_resets_fwd launch from steps 1..T-1
+        fuse_resets = hasattr(ext, "cat_st_resets_fwd") and H % S == 0
+        # step 0's inputs come from the standalone reset kernel (zero priors)
+        ext.scan_resets_fwd(z0, iz, h0, ih, actions[0], f_all[0], x_s[0], hu_s[0], True)
         for t in range(T):
-            f = f_all[t]
-            h_prev = h_seq[t - 1] if t > 0 else h0
-            z_prev = z_seq[t - 1] if t > 0 else z0
-            # one kernel assembles the reset-masked GEMM inputs (z', a', h')
-            ext.scan_resets_fwd(z_prev, iz, h_prev, ih, actions[t], f, x_s[t], hu_s[t], t == 0)
+            if t > 0 and not fuse_resets:
+                ext.scan_resets_fwd(z_seq[t - 1], iz, h_seq[t - 1], ih, actions[t], f_all[t],
+                                    x_s[t], hu_s[t], False)
             if g16:
                 torch.mm(x_s[t], w1t, out=g1_s[t])
                 ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
@@ -256,19 +259,23 @@ class _RSSMScan(torch.autograd.Function):
                     ext.g16_splitk(r_s[t], w3, None, sk_scratch, sk_tickets, g3_s[t], sk_ks)
                     ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
                 torch.addmm(b4, p_s[t], w4t, out=raw)
+            else:
+                torch.mm(x_s[t], w1t, out=g1_s[t])
+                ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
+                torch.mm(hu_s[t], w2t, out=y_s[t])
+                ext.gru_gates_fwd_o(y_s[t], hu_s[t, :, :H], lnwg, lnbg, eps,
+                                    h_seq[t], r_s[t, :, :H], mrg_s[0, t], mrg_s[1, t])
+                torch.mm(r_s[t], w3t, out=g3_s[t])
+                ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
+                torch.addmm(b4, p_s[t], w4t, out=raw)
+            if fuse_resets and t + 1 < T:
+                ext.cat_st_resets_fwd(raw.view(B, S, discrete), urand_all[t], unimix,
+                                      m_seq[t].view(B, S, discrete), z_seq[t], s_s[t],
+                                      iz, h_seq[t], ih, actions[t + 1], f_all[t + 1],
+                                      x_s[t + 1], hu_s[t + 1])
+            else:
                 ext.cat_st_fwd_o(raw.view(B, S, discrete), urand_all[t], unimix,
                                  m_seq[t].view(B, S, discrete), z_seq[t].view(B, S, discrete), s_s[t])
-                continue
-            torch.mm(x_s[t], w1t, out=g1_s[t])
-            ext.ln_act_fwd_o(g1_s[t], lnw1, lnb1, eps, True, hu_s[t, :, H:], mr1_s[0, t], mr1_s[1, t])
-            torch.mm(hu_s[t], w2t, out=y_s[t])
-            ext.gru_gates_fwd_o(y_s[t], hu_s[t, :, :H], lnwg, lnbg, eps,
-                                h_seq[t], r_s[t, :, :H], mrg_s[0, t], mrg_s[1, t])
-            torch.mm(r_s[t], w3t, out=g3_s[t])
-            ext.ln_act_fwd_o(g3_s[t], lnw3, lnb3, eps, True, p_s[t], mr3_s[0, t], mr3_s[1, t])
-            torch.addmm(b4, p_s[t], w4t, out=raw)
-            ext.cat_st_fwd_o(raw.view(B, S, discrete), urand_all[t], unimix,
-                             m_seq[t].view(B, S, discrete), z_seq[t].view(B, S, discrete), s_s[t])
 
         ctx.save_for_backward(
             f_all, w1, lnw1, lnb1, w2, lnwg, lnbg, w3, lnw3, lnb3, w4,
