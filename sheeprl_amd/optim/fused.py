@@ -150,6 +150,23 @@ class FusedAdam(torch.optim.Optimizer):
                     group["eps"],
                     group["weight_decay"],
                 )
+            elif all(p.dtype == torch.float32 and g.dtype == torch.float32 for p, g in zip(params, grads)):
+                # foreach eager path (CPU / non-HIP): one C++ multi-tensor op
+                # per stage instead of a Python loop per parameter — the loop
+                # measured ~3 ms/step on the CPU SAC wall-clock benchmark
+                all_groups_zeroed_in_kernel = False
+                if group["weight_decay"] != 0:
+                    grads = torch._foreach_add(grads, params, alpha=group["weight_decay"])
+                torch._foreach_mul_(exp_avgs, beta1)
+                torch._foreach_add_(exp_avgs, grads, alpha=1 - beta1)
+                torch._foreach_mul_(exp_avg_sqs, beta2)
+                torch._foreach_addcmul_(exp_avg_sqs, grads, grads, value=1 - beta2)
+                denom = torch._foreach_div(exp_avg_sqs, bc2)
+                torch._foreach_sqrt_(denom)
+                torch._foreach_add_(denom, group["eps"])
+                upd = torch._foreach_div(exp_avgs, bc1)
+                torch._foreach_div_(upd, denom)
+                torch._foreach_add_(params, upd, alpha=-group["lr"])
             else:
                 all_groups_zeroed_in_kernel = False
                 for p, g, m, v in zip(params, grads, exp_avgs, exp_avg_sqs):
