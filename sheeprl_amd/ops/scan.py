@@ -239,7 +239,8 @@ class _RSSMScan(torch.autograd.Function):
         w1t, w2t, w3t, w4t = w1.t(), w2.t(), w3.t(), w4.t()
         # cat_st fused with the NEXT step's reset assembly removes the
         # standalone scan_resets_fwd launch from steps 1..T-1
-        fuse_resets = hasattr(ext, "cat_st_resets_fwd") and H % S == 0
+        fuse_resets = (hasattr(ext, "cat_st_resets_fwd") and H % S == 0
+                       and os.environ.get("SHEEPRL_AMD_NO_CATRESETS", "0") != "1")
         # step 0's inputs come from the standalone reset kernel (zero priors)
         ext.scan_resets_fwd(z0, iz, h0, ih, actions[0], f_all[0], x_s[0], hu_s[0], True)
         for t in range(T):
