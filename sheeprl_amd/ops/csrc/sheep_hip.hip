@@ -1490,6 +1490,86 @@ __global__ void cat_st_bwd_kernel(const float* __restrict__ gm, const T* __restr
   }
 }
 
+// cat_st backward for step t fused with scan_resets_bwd of step t+1: the
+// z-carry entering this step's ST backward is (1-f_{t+1}) * gx_{t+1} — computed
+// in-register instead of via a standalone launch; the same kernel emits the
+// h-carry for this step's GRU backward, the action grad of t+1, and the
+// masked init-state accumulator updates.
+template <typename T>
+__global__ void cat_st_resets_bwd_kernel(
+    const float* __restrict__ gm, const T* __restrict__ gon, const float* __restrict__ s_saved,
+    T* __restrict__ graw, long nrows, int K, float unimix,
+    const T* __restrict__ ghu, long ghus, const T* __restrict__ ghp,
+    const T* __restrict__ gx, long gxs, const T* __restrict__ f,
+    T* __restrict__ gh_carry, T* __restrict__ ga, float* __restrict__ gih_acc,
+    float* __restrict__ giz_acc, int S, int A, int H) {
+  const int lane = threadIdx.x & 63;
+  const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= nrows) return;
+  const long b = row / S;
+  const int s = (int)(row - b * S);
+  const int SK = S * K;
+  const float fb = ld(f, b);
+  const float* gmr = gm + row * (long)K;
+  const T* gor = gon + row * (long)K;
+  const float* sr = s_saved + row * (long)K;
+  T* gr = graw + row * (long)K;
+  float acc = 0.f;
+  for (int j = lane; j < K; j += 64) {
+    float sj = sr[j];
+    float pj = (1.f - unimix) * sj + unimix / K;
+    float gxv = ld(gx, b * gxs + s * K + j);
+    giz_acc[b * (long)SK + s * K + j] += fb * gxv;
+    float t = gmr[j] / pj + ld(gor, j) + (1.f - fb) * gxv;
+    acc += t * sj;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_xor(acc, off, 64);
+  for (int j = lane; j < K; j += 64) {
+    float sj = sr[j];
+    float pj = (1.f - unimix) * sj + unimix / K;
+    float gxv = ld(gx, b * gxs + s * K + j);
+    float t = gmr[j] / pj + ld(gor, j) + (1.f - fb) * gxv;
+    st(gr, j, (1.f - unimix) * sj * (t - acc));
+  }
+  if (s == S - 1)
+    for (int j = lane; j < A; j += 64) st(ga, b * (long)A + j, (1.f - fb) * ld(gx, b * gxs + SK + j));
+  const int HS = H / S;  // caller guarantees H % S == 0
+  for (int j = lane; j < HS; j += 64) {
+    const long c = (long)s * HS + j;
+    float g = ld(ghu, b * ghus + c) + ld(ghp, b * (long)H + c);
+    st(gh_carry, b * (long)H + c, (1.f - fb) * g);
+    gih_acc[b * (long)H + c] += fb * g;
+  }
+}
+
+void cat_st_resets_bwd(const torch::Tensor& gm, const torch::Tensor& gon, const torch::Tensor& s_saved,
+                       double unimix, torch::Tensor graw, const torch::Tensor& ghu,
+                       const torch::Tensor& ghp, const torch::Tensor& gx, const torch::Tensor& f,
+                       torch::Tensor gh_carry, torch::Tensor ga, torch::Tensor gih_acc,
+                       torch::Tensor giz_acc) {
+  CHECK_IN(gm);
+  int K = (int)gm.size(-1);
+  int S = (int)gm.size(-2);
+  long nrows = gm.numel() / K;
+  int A = (int)ga.size(1);
+  int H = (int)gh_carry.size(1);
+  TORCH_CHECK(H % S == 0, "cat_st_resets_bwd requires H % S == 0");
+  const int rows_per_block = kBlock / 64;
+  int blocks = (int)((nrows + rows_per_block - 1) / rows_per_block);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto gonc = gon.contiguous();
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, graw.scalar_type(), "cat_st_resets_bwd", [&] {
+    using T = scalar_t;
+    hipLaunchKernelGGL((cat_st_resets_bwd_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                       gm.data_ptr<float>(), (const T*)gonc.data_ptr(), s_saved.data_ptr<float>(),
+                       (T*)graw.data_ptr(), nrows, K, (float)unimix, (const T*)ghu.data_ptr(),
+                       ghu.stride(0), (const T*)ghp.data_ptr(), (const T*)gx.data_ptr(), gx.stride(0),
+                       (const T*)f.data_ptr(), (T*)gh_carry.data_ptr(), (T*)ga.data_ptr(),
+                       gih_acc.data_ptr<float>(), giz_acc.data_ptr<float>(), S, A, H);
+  });
+}
+
 std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optional<torch::Tensor>& urand,
                                       double unimix, bool sample) {
   CHECK_IN(raw);
@@ -5496,6 +5576,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsprop_step_mt", &rmsprop_step_mt);
   m.def("clip_grad_norm_mt", &clip_grad_norm_mt);
   m.def("cat_st_resets_fwd", &cat_st_resets_fwd);
+  m.def("cat_st_resets_bwd", &cat_st_resets_bwd);
   m.def("cat_st_fwd", &cat_st_fwd);
   m.def("cat_st_bwd", &cat_st_bwd);
   m.def("masked_lerp_fwd", &masked_lerp_fwd);

@@ -368,12 +368,24 @@ class _RSSMScan(torch.autograd.Function):
                 lnw1, lnb1, lnwg, lnbg, lnw3, lnb3,
             )
 
+        # cat_st backward fused with the NEXT (already-processed) step's reset
+        # backward: the z-carry is computed in-register from gx_{t+1}, and the
+        # h-carry/action-grad/init-accumulator writes ride along — removing
+        # the standalone scan_resets_bwd launch from steps T-2..1
+        fuse_b = (hasattr(ext, "cat_st_resets_bwd") and H % S == 0
+                  and os.environ.get("SHEEPRL_AMD_NO_CATRESETS", "0") != "1")
         for t in range(T - 1, -1, -1):
             f = f_all[t]
-            zc = gz_carry.view(B, S, discrete) if t < T - 1 else None
             hc = gh_carry if t < T - 1 else None
-            ext.cat_st_bwd_o(g_m_seq[t].view(B, S, discrete), g_z_seq[t].view(B, S, discrete), zc,
-                             s_s[t], unimix, graw_s[t].view(B, S, discrete))
+            if fuse_b and t < T - 1:
+                ext.cat_st_resets_bwd(g_m_seq[t].view(B, S, discrete),
+                                      g_z_seq[t].view(B, S, discrete), s_s[t], unimix,
+                                      graw_s[t].view(B, S, discrete), ghu, ghp, gx, f_all[t + 1],
+                                      gh_carry, g_actions[t + 1], gih_acc, giz_acc)
+            else:
+                zc = gz_carry.view(B, S, discrete) if t < T - 1 else None
+                ext.cat_st_bwd_o(g_m_seq[t].view(B, S, discrete), g_z_seq[t].view(B, S, discrete), zc,
+                                 s_s[t], unimix, graw_s[t].view(B, S, discrete))
             torch.mm(graw_s[t], w4, out=gp)
             ext.ln_act_bwd_acc(gp, g3_s[t], lnw3, lnb3, mr3_s[0, t], mr3_s[1, t], True,
                                gg3_s[t], glnw3, glnb3)
@@ -385,7 +397,10 @@ class _RSSMScan(torch.autograd.Function):
                                gg1_s[t], glnw1, glnb1)
             torch.mm(gg1_s[t], w1, out=gx)
             # one kernel: carries, action grad, init-state accumulators
-            ext.scan_resets_bwd(ghu, ghp, gx, f, gh_carry, gz_carry, g_actions[t], gih_acc, giz_acc)
+            # (fused into the next iteration's cat_st backward when fuse_b,
+            # except t=0 whose carries are the init-state path)
+            if not fuse_b or t == 0:
+                ext.scan_resets_bwd(ghu, ghp, gx, f, gh_carry, gz_carry, g_actions[t], gih_acc, giz_acc)
 
         return _scan_weight_grads(
             ext, T, B, E, A, H, SK, D, P, dt, dev,
