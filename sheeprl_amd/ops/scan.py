@@ -374,6 +374,15 @@ class _RSSMScan(torch.autograd.Function):
         # the standalone scan_resets_bwd launch from steps T-2..1
         fuse_b = (hasattr(ext, "cat_st_resets_bwd") and H % S == 0
                   and os.environ.get("SHEEPRL_AMD_NO_CATRESETS", "0") != "1")
+        # interleaved weight-grad GEMMs (experiment, default OFF): issuing a
+        # partial [out, chunk*B] x [chunk*B, in] GEMM every `chunk` reverse
+        # steps — hoping to read slices still hot in L2/MALL — measured
+        # SLOWER than the whole-sequence GEMMs (6.14 -> 6.22/6.30/6.70 ms
+        # per graphed scan fwd+bwd at chunk 32/16/8, probes/
+        # wgrad_chunk_perf.py): the K=T*B GEMM shape amortizes better than
+        # the warmth gain, and the partial sums add launches.
+        wchunk = int(os.environ.get("SHEEPRL_AMD_WGRAD_CHUNK", "0"))
+        wparts = {k: [] for k in ("w4", "w3", "w2", "w1")} if wchunk > 0 else None
         for t in range(T - 1, -1, -1):
             f = f_all[t]
             hc = gh_carry if t < T - 1 else None
@@ -401,13 +410,24 @@ class _RSSMScan(torch.autograd.Function):
             # except t=0 whose carries are the init-state path)
             if not fuse_b or t == 0:
                 ext.scan_resets_bwd(ghu, ghp, gx, f, gh_carry, gz_carry, g_actions[t], gih_acc, giz_acc)
+            if wparts is not None and t % wchunk == 0:
+                sl = slice(t, min(t + wchunk, T))
+                n = (sl.stop - sl.start) * B
+                wparts["w4"].append(torch.mm(graw_s[sl].view(n, SK).t(), p_s[sl].view(n, P)))
+                wparts["w3"].append(torch.mm(gg3_s[sl].view(n, P).t(), r_s[sl].view(n, H + E)))
+                wparts["w2"].append(torch.mm(gy_s[sl].view(n, 3 * H).t(), hu_s[sl].view(n, H + D)))
+                wparts["w1"].append(torch.mm(gg1_s[sl].view(n, D).t(), x_s[sl].view(n, SK + A)))
 
+        gws = None
+        if wparts is not None:
+            gws = {k: v[0] if len(v) == 1 else torch.stack(v).sum(0) for k, v in wparts.items()}
         return _scan_weight_grads(
             ext, T, B, E, A, H, SK, D, P, dt, dev,
             graw_s, gg3_s, gy_s, gg1_s, gr_s, g_actions,
             p_s, r_s, hu_s, x_s,
             gih_acc, giz_acc, glnw1, glnb1, glnwg, glnbg, glnw3, glnb3,
             lnw1, lnb1, lnwg, lnbg, lnw3, lnb3,
+            gws=gws,
         )
 
 
@@ -417,17 +437,19 @@ def _scan_weight_grads(
     p_s, r_s, hu_s, x_s,
     gih_acc, giz_acc, glnw1, glnb1, glnwg, glnbg, glnw3, glnb3,
     lnw1, lnb1, lnwg, lnbg, lnw3, lnb3,
+    gws=None,
 ):
     # batched weight grads: one MFMA GEMM per weight over all T*B rows
+    # (or the summed interleaved partials computed inside the reverse loop)
     TB = T * B
     ones_row = torch.ones(1, TB, device=dev, dtype=dt)
-    gW4 = torch.mm(graw_s.view(TB, SK).t(), p_s.view(TB, P))
+    gW4 = gws["w4"] if gws else torch.mm(graw_s.view(TB, SK).t(), p_s.view(TB, P))
     # bias grad as a GEMM against ones: torch's bf16 column-reduce sum(0)
     # was measured at ~300 us for this shape; the GEMV path is ~5 us.
     gb4 = torch.mm(ones_row, graw_s.view(TB, SK)).view(SK)
-    gW3 = torch.mm(gg3_s.view(TB, P).t(), r_s.view(TB, H + E))
-    gW2 = torch.mm(gy_s.view(TB, 3 * H).t(), hu_s.view(TB, H + D))
-    gW1 = torch.mm(gg1_s.view(TB, D).t(), x_s.view(TB, SK + A))
+    gW3 = gws["w3"] if gws else torch.mm(gg3_s.view(TB, P).t(), r_s.view(TB, H + E))
+    gW2 = gws["w2"] if gws else torch.mm(gy_s.view(TB, 3 * H).t(), hu_s.view(TB, H + D))
+    gW1 = gws["w1"] if gws else torch.mm(gg1_s.view(TB, D).t(), x_s.view(TB, SK + A))
     return (
         gr_s[:, :, H:],                       # g_embed (strided view is fine)
         g_actions, None,
