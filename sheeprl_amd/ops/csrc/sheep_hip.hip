@@ -459,8 +459,13 @@ void launch_ln_bwd_cl(int L, long R, hipStream_t st, const T* gy, const T* x, co
   int rpb = (kBlock / 64) * (64 / L);
   int blocks = (int)std::min((R + rpb - 1) / rpb, (long)2048);
   // each block flushes 2*D atomicAdds at the end; for mid-sized N that flush
-  // dominates — keep >=32 rows of real work per block
-  blocks = (int)std::min((long)blocks, std::max((long)64, (R + 31) / 32));
+  // dominates — keep >=32 rows of real work per block.  The parallelism
+  // floor is tunable for measurement (SHEEPRL_AMD_LN_BWD_FLOOR, default 64).
+  static const long kFloor = [] {
+    const char* s = getenv("SHEEPRL_AMD_LN_BWD_FLOOR");
+    return s ? atol(s) : 64L;
+  }();
+  blocks = (int)std::min((long)blocks, std::max(kFloor, (R + 31) / 32));
   size_t shmem = 2 * (size_t)D * sizeof(float);
 #define SHEEP_LN_BWD_CASE(LV)                                                                               \
   case LV:                                                                                                  \
