@@ -610,6 +610,206 @@ def run_bench_sac(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
     }
 
 
+def run_bench_ppo(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
+    """PPO on the synthetic Atari-shaped env (3x64x64 uint8, 9 actions),
+    bf16 on GPU.  One step = one full PPO iteration: rollout_steps env
+    interactions per env + GAE + update_epochs x minibatch SGD, the
+    minibatch step captured in a hipGraph (minibatch shapes are constant).
+    BASELINE.json names PPO as part of the headline metric; the reference's
+    published PPO row is the CPU wall-clock benchmark (benchmarks/RESULTS.md)."""
+    import torch.distributed as dist
+
+    from sheeprl_amd.algos.ppo.agent import build_agent as build_ppo
+    from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, value_loss
+    from sheeprl_amd.algos.ppo.utils import prepare_obs
+    from sheeprl_amd.config import compose
+    from sheeprl_amd.envs import vectorize_env
+    from sheeprl_amd.optim import make_optimizer
+    from sheeprl_amd.parallel import Runtime
+    from sheeprl_amd.ops import gae as compute_gae
+    from sheeprl_amd.utils.utils import seed_everything
+
+    rank = int(os.environ.get("RANK", 0))
+    world_size = int(os.environ.get("WORLD_SIZE", n_gpus))
+    device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    if device_type == "cuda":
+        from sheeprl_amd.ops import has_ext
+
+        if not has_ext():
+            raise RuntimeError("HIP extension _sheep_hip is not built — refusing to bench the eager fallback")
+    cfg = compose([
+        "exp=ppo",
+        "env=synthetic_atari",
+        "env.num_envs=8",
+        "algo.cnn_keys.encoder=[rgb]",
+        "algo.mlp_keys.encoder=[]",
+        "algo.rollout_steps=128",
+        "algo.per_rank_batch_size=256",
+        "algo.update_epochs=4",
+        f"runtime.accelerator={device_type}",
+        "runtime.precision=" + ("bf16" if device_type == "cuda" else "fp32"),
+        "metric.log_level=0",
+        "metric.disable_timer=True",
+        "checkpoint.every=0",
+        "checkpoint.save_last=False",
+        "algo.run_test=False",
+        "env.sync_env=True",
+    ] + list(overrides))
+    seed_everything(cfg.seed + rank)
+    runtime = Runtime(devices=world_size, accelerator=cfg.runtime.accelerator, precision=cfg.runtime.precision)
+    runtime.world_size = world_size
+    runtime.global_rank = rank
+    runtime.local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    if world_size > 1:
+        runtime._init_process_group(init_method="env://")
+    else:
+        runtime._setup_device()
+    device = runtime.device
+    num_envs = cfg.env.num_envs
+    T = cfg.algo.rollout_steps
+    bs = cfg.algo.per_rank_batch_size
+    rows = T * num_envs
+
+    envs = vectorize_env(cfg, cfg.seed, rank)
+    obs_space = envs.single_observation_space
+    action_space = envs.single_action_space
+    agent, player = build_ppo(runtime, obs_space, action_space, cfg, None)
+    optimizer = make_optimizer(agent.parameters(), cfg.algo.optimizer)
+
+    # device-side rollout storage (obs stay uint8; encoder normalizes on GPU)
+    store = {
+        "obs_rgb": torch.empty(T, num_envs, *obs_space["rgb"].shape, device=device, dtype=torch.uint8),
+        "actions": torch.empty(T, num_envs, 1, device=device, dtype=torch.float32),
+        "logprobs": torch.empty(T, num_envs, 1, device=device, dtype=torch.float32),
+        "values": torch.empty(T, num_envs, 1, device=device, dtype=torch.float32),
+        "rewards": torch.empty(T, num_envs, 1, device=device, dtype=torch.float32),
+        "dones": torch.empty(T, num_envs, 1, device=device, dtype=torch.float32),
+    }
+    obs, _ = envs.reset(seed=cfg.seed + rank * num_envs)
+
+    normalize = cfg.algo.normalize_advantages
+
+    def mb_train(batch):
+        obs_b = {"rgb": batch["obs_rgb"]}
+        adv = batch["advantages"]
+        if normalize:
+            adv = (adv - adv.mean()) / (adv.std() + 1e-8)
+        _, logprobs, entropy, new_values = agent(obs_b, batch["actions"])
+        pg_loss = policy_loss(logprobs, batch["logprobs"], adv, cfg.algo.clip_coef, cfg.algo.loss_reduction)
+        v_loss = value_loss(new_values, batch["values"], batch["returns"], cfg.algo.clip_coef,
+                            cfg.algo.clip_vloss, cfg.algo.loss_reduction)
+        ent_loss = entropy_loss(entropy, cfg.algo.loss_reduction)
+        loss = pg_loss + cfg.algo.vf_coef * v_loss + cfg.algo.ent_coef * ent_loss
+        optimizer.zero_grad(set_to_none=True)
+        runtime.backward(loss)
+        if cfg.algo.max_grad_norm and cfg.algo.max_grad_norm > 0:
+            runtime.clip_gradients(agent, optimizer, max_norm=cfg.algo.max_grad_norm)
+        optimizer.step()
+
+    def rollout():
+        nonlocal obs
+        for t in range(T):
+            t_obs = prepare_obs(obs, cfg, device)
+            with torch.no_grad():
+                actions, logprobs, values = player.get_actions(t_obs)
+            env_actions = actions.cpu().numpy()[..., 0]
+            next_obs, rewards, terms, truncs, _ = envs.step(env_actions)
+            store["obs_rgb"][t].copy_(t_obs["rgb"])
+            store["actions"][t].copy_(actions.float())
+            store["logprobs"][t].copy_(logprobs.float())
+            store["values"][t].copy_(values.float())
+            store["rewards"][t].copy_(torch.as_tensor(rewards, dtype=torch.float32).reshape(num_envs, 1).to(device))
+            store["dones"][t].copy_(torch.as_tensor(
+                np.logical_or(terms, truncs).astype(np.float32)).reshape(num_envs, 1).to(device))
+            obs = next_obs
+        with torch.no_grad():
+            next_values = player.get_values(prepare_obs(obs, cfg, device))
+        returns, advantages = compute_gae(
+            store["rewards"], store["values"], store["dones"].bool(), next_values,
+            T, cfg.algo.gamma, cfg.algo.gae_lambda)
+        data = {k: v.reshape(rows, *v.shape[2:]) for k, v in store.items()}
+        data["returns"] = returns.reshape(rows, 1)
+        data["advantages"] = advantages.reshape(rows, 1)
+        data["actions"] = data["actions"].long()
+        return data
+
+    graphed = None
+    if device_type == "cuda" and os.environ.get("SHEEPRL_AMD_NO_GRAPHS", "0") != "1":
+        from sheeprl_amd.parallel.graphs import CUDAGraphStep
+
+        data = rollout()
+        example = {k: v[:bs].clone() for k, v in data.items()}
+        try:
+            graphed = CUDAGraphStep(mb_train, example, warmup=3)
+            if rank == 0:
+                print("[bench] PPO minibatch step captured in a hipGraph", file=sys.stderr)
+        except Exception as e:  # noqa: BLE001
+            graphed = None
+            if rank == 0:
+                print(f"[bench] hipGraph capture failed ({e}); running eager", file=sys.stderr)
+
+    gen = torch.Generator(device=device)
+    gen.manual_seed(cfg.seed + rank)
+
+    def one_iter():
+        data = rollout()
+        for _ in range(cfg.algo.update_epochs):
+            perm = torch.randperm(rows, device=device, generator=gen)
+            for s in range(0, rows, bs):
+                idx = perm[s:s + bs]
+                batch = {k: v[idx] for k, v in data.items()}
+                if graphed is not None:
+                    graphed(batch)
+                else:
+                    mb_train(batch)
+
+    for _ in range(warmup):
+        one_iter()
+    if runtime.is_distributed:
+        dist.barrier()
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        one_iter()
+    if device_type == "cuda":
+        torch.cuda.synchronize()
+    if runtime.is_distributed:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+    if runtime.is_distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if runtime.backend == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+    envs.close()
+    fps = steps * rows * world_size / elapsed
+    return {
+        "metric": "env_frames_per_sec",
+        "value": round(fps, 3),
+        "unit": "frames/s",
+        "n_gpus": world_size,
+        "steps": steps,
+        "warmup": warmup,
+        "ms_per_step": round(elapsed / steps * 1000, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if device_type == "cuda" else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "ppo_pixel",
+            "global_batch": bs * world_size,
+            "seq_len": 1,
+            "parallelism": f"dp{world_size}",
+            "benchmark": f"PPO pixels (3x64x64 synthetic Atari shape, rollout {T} x {num_envs} envs, "
+                         f"{cfg.algo.update_epochs} epochs)",
+            "rollout_steps": T,
+            "update_epochs": cfg.algo.update_epochs,
+            "action_repeat": 1,
+        },
+    }
+
+
 def smoke_step() -> None:
     """One tiny forward+backward of the flagship model on cuda:0 (driver
     contract: __graft_entry__.smoke)."""
@@ -632,13 +832,16 @@ def main() -> None:
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=100)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--algo", choices=["dreamer_v3", "sac"], default="dreamer_v3",
-                   help="flagship DV3 Atari-100K (default) or SAC HalfCheetah-shape (BASELINE #2)")
+    p.add_argument("--algo", choices=["dreamer_v3", "sac", "ppo"], default="dreamer_v3",
+                   help="flagship DV3 Atari-100K (default), SAC HalfCheetah-shape (BASELINE #2), "
+                        "or PPO pixels (synthetic Atari shape)")
     p.add_argument("--override", action="append", default=[])
     args = p.parse_args()
     rank = int(os.environ.get("RANK", 0))
     if args.algo == "sac":
         result = run_bench_sac(args.gpus, args.steps, args.warmup, args.override)
+    elif args.algo == "ppo":
+        result = run_bench_ppo(args.gpus, args.steps, args.warmup, args.override)
     else:
         result = run_bench(args.gpus, args.steps, args.warmup, args.override)
     if rank == 0:

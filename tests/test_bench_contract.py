@@ -63,3 +63,23 @@ def test_bench_single_rank_default(tmp_path):
     assert len(json_lines) == 1, out.stdout
     rec = json.loads(json_lines[0])
     assert rec["n_gpus"] == 1 and rec["metric"] == "env_frames_per_sec"
+
+
+@pytest.mark.timeout(600)
+def test_bench_ppo_algo_single_rank(tmp_path):
+    """`python bench.py --algo ppo` emits one valid contract JSON line."""
+    env = dict(os.environ)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env["PYTHONPATH"] = f"{repo}:{env.get('PYTHONPATH', '')}"
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "bench.py"), "--algo", "ppo", "--steps", "1",
+         "--warmup", "0", "--override", "algo.rollout_steps=8", "--override", "env.num_envs=2",
+         "--override", "algo.per_rank_batch_size=8", "--override", "algo.update_epochs=1"],
+        capture_output=True, text=True, env=env, cwd=str(tmp_path), timeout=570,
+    )
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    json_lines = [l for l in out.stdout.splitlines() if l.strip().startswith("{")]
+    assert len(json_lines) == 1, out.stdout
+    rec = json.loads(json_lines[0])
+    assert rec["metric"] == "env_frames_per_sec" and rec["config"]["model"] == "ppo_pixel"
+    assert rec["value"] > 0 and rec["scaling"] == "weak"
