@@ -35,7 +35,10 @@ class PixelEncoder(nn.Module):
 
     def forward(self, obs: Dict[str, Tensor]) -> Tensor:
         x = torch.cat([obs[k] for k in self.keys], dim=-3)
-        x = ops.normalize_obs(x) if x.dtype == torch.uint8 else x
+        if x.dtype == torch.uint8:
+            x = ops.normalize_obs(x)
+        # the fused obs_norm emits fp32; bf16-true modules need the cast
+        x = x.to(self.cnn.conv[0].weight.dtype)
         return cnn_forward(self.cnn, x, self.input_dim)
 
 
