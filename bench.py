@@ -696,7 +696,8 @@ def run_bench_ppo(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
             adv = (adv - adv.mean()) / (adv.std() + 1e-8)
         _, logprobs, entropy, new_values = agent(obs_b, batch["actions"])
         pg_loss = policy_loss(logprobs, batch["logprobs"], adv, cfg.algo.clip_coef, cfg.algo.loss_reduction)
-        v_loss = value_loss(new_values, batch["values"], batch["returns"], cfg.algo.clip_coef,
+        # bf16-true agent emits bf16 values; targets are stored fp32
+        v_loss = value_loss(new_values.float(), batch["values"], batch["returns"], cfg.algo.clip_coef,
                             cfg.algo.clip_vloss, cfg.algo.loss_reduction)
         ent_loss = entropy_loss(entropy, cfg.algo.loss_reduction)
         loss = pg_loss + cfg.algo.vf_coef * v_loss + cfg.algo.ent_coef * ent_loss

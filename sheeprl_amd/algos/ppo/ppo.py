@@ -57,7 +57,7 @@ def train(
             _, logprobs, entropy, new_values = agent(obs, batch["actions"])
             pg_loss = policy_loss(logprobs, batch["logprobs"], adv, cfg.algo.clip_coef, cfg.algo.loss_reduction)
             v_loss = value_loss(
-                new_values,
+                new_values.float(),  # bf16-true agent vs fp32 stored targets
                 batch["values"],
                 batch["returns"],
                 cfg.algo.clip_coef,
