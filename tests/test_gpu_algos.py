@@ -188,3 +188,28 @@ def test_gpu_p2e_dv2_graphs(tmp_path):
                     "algo.ensembles.mlp_layers=1", "algo.per_rank_pretrain_steps=1",
                     *TINY_DV],
          precision="bf16")
+
+
+@requires_gpu
+@pytest.mark.timeout(420)
+def test_gpu_ppo_bench_graphs(tmp_path):
+    """bench --algo ppo captures both the player forward and the minibatch
+    step in hipGraphs and emits a valid JSON line (bf16)."""
+    import json as _json
+    import subprocess
+    import sys as _sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = f"{repo}:{env.get('PYTHONPATH', '')}"
+    out = subprocess.run(
+        [_sys.executable, os.path.join(repo, "bench.py"), "--algo", "ppo", "--steps", "3",
+         "--warmup", "1", "--override", "algo.rollout_steps=32", "--override", "env.num_envs=4",
+         "--override", "algo.per_rank_batch_size=64", "--override", "algo.update_epochs=2"],
+        capture_output=True, text=True, env=env, cwd=str(tmp_path), timeout=390,
+    )
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    assert "minibatch step captured" in out.stderr, out.stderr[-800:]
+    assert "player forward captured" in out.stderr, out.stderr[-800:]
+    rec = _json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][0])
+    assert rec["config"]["model"] == "ppo_pixel" and rec["value"] > 0
