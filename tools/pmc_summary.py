@@ -20,3 +20,12 @@ for k, c in rows[:28]:
     wc = c["SQ_WAVE_CYCLES"] or 1.0
     print(f"{k} | {cnt[k]} | {100*c['SQ_WAIT_ANY']/wc:.0f} | {100*c['SQ_WAIT_INST_ANY']/wc:.0f} | "
           f"{100*c['SQ_ACTIVE_INST_ANY']/wc:.0f} | {c['SQ_INSTS_MFMA']/max(cnt[k],1):.0f}", file=out)
+
+print("\n-- steady-state kernels of interest --", file=out)
+import re
+pat = re.compile(r"Cijk|ln_act|gru_gates|cat_st|scan3|adam_mt|replay_gather|nll_|reinforce|vloss")
+for k, c in rows:
+    if pat.search(k):
+        wc = c["SQ_WAVE_CYCLES"] or 1.0
+        print(f"{k} | {cnt[k]} | {100*c['SQ_WAIT_ANY']/wc:.0f} | {100*c['SQ_WAIT_INST_ANY']/wc:.0f} | "
+              f"{100*c['SQ_ACTIVE_INST_ANY']/wc:.0f} | {c['SQ_INSTS_MFMA']/max(cnt[k],1):.0f}", file=out)
