@@ -487,6 +487,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
     policy_steps_per_iter = int(num_envs * world_size)
     total_iters = int(cfg.algo.total_steps // policy_steps_per_iter) if not cfg.dry_run else 1
     learning_starts = cfg.algo.learning_starts // policy_steps_per_iter if not cfg.dry_run else 0
+    # replay-ratio accounting starts AFTER the prefill (parity:
+    # sheeprl dreamer_v3.py:661, sac.py:301 — the reference subtracts the
+    # prefill policy steps before asking Ratio how many grad steps are owed,
+    # otherwise the first train iteration pays a learning_starts-sized backlog)
+    prefill_steps = max(learning_starts - 1, 0) * policy_steps_per_iter
     if cfg.checkpoint.resume_from:
         cfg.algo.per_rank_batch_size = state["batch_size"] // world_size
     start_iter = int(state.get("iter_num", 1))
@@ -618,7 +623,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
                 ep[next(iter(ep))].shape[0] >= cfg.algo.per_rank_sequence_length for ep in rb.buffer
             )
         if iter_num >= learning_starts and rb_ready:
-            per_rank_gradient_steps = ratio(policy_step / world_size)
+            per_rank_gradient_steps = ratio((policy_step - prefill_steps) / world_size)
             if per_rank_gradient_steps > 0:
                 with timer("Time/train_time"):
                     for _ in range(per_rank_gradient_steps):

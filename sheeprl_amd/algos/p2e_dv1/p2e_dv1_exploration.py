@@ -306,6 +306,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
     policy_steps_per_iter = int(num_envs * world_size)
     total_iters = int(cfg.algo.total_steps // policy_steps_per_iter) if not cfg.dry_run else 1
     learning_starts = cfg.algo.learning_starts // policy_steps_per_iter if not cfg.dry_run else 0
+    # replay-ratio accounting starts AFTER the prefill (parity:
+    # sheeprl dreamer_v3.py:661, sac.py:301 — the reference subtracts the
+    # prefill policy steps before asking Ratio how many grad steps are owed,
+    # otherwise the first train iteration pays a learning_starts-sized backlog)
+    prefill_steps = max(learning_starts - 1, 0) * policy_steps_per_iter
     policy_step = int(state.get("policy_step", 0))
     last_log = 0
     graphed_step = None
@@ -381,7 +386,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
 
         rb_ready = any(len(b) >= cfg.algo.per_rank_sequence_length for b in rb.buffer)
         if iter_num >= learning_starts and rb_ready:
-            n_steps = ratio(policy_step / world_size)
+            n_steps = ratio((policy_step - prefill_steps) / world_size)
             if n_steps > 0:
                 with timer("Time/train_time"):
                     for _ in range(n_steps):

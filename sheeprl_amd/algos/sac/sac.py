@@ -141,6 +141,11 @@ def main(runtime: Runtime, cfg: Any) -> None:
     policy_steps_per_iter = int(num_envs * world_size)
     total_iters = int(cfg.algo.total_steps // policy_steps_per_iter) if not cfg.dry_run else 1
     learning_starts = cfg.algo.learning_starts // policy_steps_per_iter if not cfg.dry_run else 0
+    # replay-ratio accounting starts AFTER the prefill (parity:
+    # sheeprl dreamer_v3.py:661, sac.py:301 — the reference subtracts the
+    # prefill policy steps before asking Ratio how many grad steps are owed,
+    # otherwise the first train iteration pays a learning_starts-sized backlog)
+    prefill_steps = max(learning_starts - 1, 0) * policy_steps_per_iter
     policy_step = int(state.get("policy_step", 0))
     last_log = int(state.get("last_log", 0))
     last_checkpoint = int(state.get("last_checkpoint", 0))
@@ -192,7 +197,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
                     aggregator.update("Game/ep_len_avg", float(ep["l"][0]))
 
         if iter_num >= learning_starts:
-            per_rank_gradient_steps = ratio(policy_step / world_size)
+            per_rank_gradient_steps = ratio((policy_step - prefill_steps) / world_size)
             if per_rank_gradient_steps > 0 and len(rb) >= 1:
                 with timer("Time/train_time"):
                     # sample all batches at once, share across ranks (sac.py:306-337)

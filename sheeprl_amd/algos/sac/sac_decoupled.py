@@ -77,6 +77,7 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
 
     total_iters = int(cfg.algo.total_steps // num_envs) if not cfg.dry_run else 1
     learning_starts = cfg.algo.learning_starts // num_envs if not cfg.dry_run else 0
+    prefill_steps = max(learning_starts - 1, 0) * num_envs  # see sac.py ratio note
     policy_step = 0
     last_log = 0
     last_checkpoint = 0
@@ -127,7 +128,7 @@ def player(runtime: Runtime, cfg: Any, world_group, pt_group) -> None:
                     aggregator.update("Game/ep_len_avg", float(ep["l"][0]))
 
         if iter_num >= learning_starts:
-            gradient_steps = ratio(policy_step)
+            gradient_steps = ratio(policy_step - prefill_steps)
             if gradient_steps > 0 and len(rb) > 0:
                 # one scatter round per training burst: each trainer gets its
                 # own stack of batches (reference :240-257)
