@@ -57,6 +57,9 @@ MODELS_TO_REGISTER = {"world_model", "actor", "critic"}
 
 
 
+from sheeprl_amd.algos.dreamer_v2.dreamer_v2 import _FastUnitNormal
+
+
 def _unit_scale(x):
     """Scale=1 as a device tensor: td.Normal(x, 1) materializes the python
     scalar with a pageable H2D copy, which is illegal inside hipGraph capture."""
@@ -349,10 +352,10 @@ def train(
     latent_states = torch.cat((stochastic_states, recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v)), len(v.shape[2:])) for k, v in decoded.items()}
+    po = {k: _FastUnitNormal(v.float(), len(v.shape[2:])) for k, v in decoded.items()}
     _rm_out = world_model.reward_model(latent_states).float()
 
-    pr = td.Independent(td.Normal(_rm_out, _unit_scale(_rm_out)), 1)
+    pr = _FastUnitNormal(_rm_out, 1)
 
     posteriors_dist = td.Independent(td.Normal(post_means, post_stds), 1)
     priors_dist = td.Independent(td.Normal(prior_means, prior_stds), 1)
@@ -413,7 +416,7 @@ def train(
     _qv_out = critic(imagined_latent_states.detach()).float()
 
 
-    qv = td.Independent(td.Normal(_qv_out, _unit_scale(_qv_out)), 1)
+    qv = _FastUnitNormal(_qv_out, 1)
     critic_optimizer.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)

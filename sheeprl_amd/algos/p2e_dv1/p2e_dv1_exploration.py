@@ -26,6 +26,7 @@ from sheeprl_amd.distributions import MSEDistribution
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
 from sheeprl_amd.models import MLP
 from sheeprl_amd.optim import make_optimizer
+from sheeprl_amd.algos.dreamer_v2.dreamer_v2 import _FastUnitNormal
 from sheeprl_amd.parallel import Runtime
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
@@ -90,7 +91,7 @@ def _dv1_behaviour(runtime, cfg, world_model, actor, critic, actor_opt, critic_o
     _qv_out = critic(imagined_latent_states.detach()).float()
 
 
-    qv = td.Independent(td.Normal(_qv_out, _unit_scale(_qv_out)), 1)
+    qv = _FastUnitNormal(_qv_out, 1)
     critic_opt.zero_grad(set_to_none=True)
     value_loss = -torch.mean(discount[..., 0] * qv.log_prob(lambda_values.detach()))
     runtime.backward(value_loss)
@@ -135,10 +136,10 @@ def train(runtime, world_model, ensembles, actor_task, critic_task, actor_explor
     latent_states = torch.cat((stochastic_states, recurrent_states), -1)
 
     decoded = world_model.observation_model(latent_states)
-    po = {k: td.Independent(td.Normal(v.float(), _unit_scale(v)), len(v.shape[2:])) for k, v in decoded.items()}
+    po = {k: _FastUnitNormal(v.float(), len(v.shape[2:])) for k, v in decoded.items()}
     _rm_out = world_model.reward_model(latent_states).float()
 
-    pr = td.Independent(td.Normal(_rm_out, _unit_scale(_rm_out)), 1)
+    pr = _FastUnitNormal(_rm_out, 1)
     posteriors_dist = td.Independent(td.Normal(post_means, post_stds), 1)
     priors_dist = td.Independent(td.Normal(prior_means, prior_stds), 1)
     observation_loss = -sum(po[k].log_prob(batch_obs[k].float()).mean() for k in po)
