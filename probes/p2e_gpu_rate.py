@@ -1,0 +1,20 @@
+import sys, os, tempfile
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+"""P2E-DV3 exploration graphed gradient-step rate at default dims: run the
+real CLI loop for ~160 post-prefill steps and read the cumulative
+Time/train_time from the timer registry (p2e accumulates it but never
+resets)."""
+from sheeprl_amd.cli import run
+from sheeprl_amd.utils.timer import timer
+
+STEPS = 160
+tmp = tempfile.mkdtemp()
+run(["exp=p2e_dv3_exploration", "env=synthetic_atari", "runtime.accelerator=cuda",
+     "runtime.precision=bf16", "algo.replay_ratio=1", f"algo.total_steps={1024 + STEPS}",
+     "algo.learning_starts=1024", "metric.log_level=1", "metric.log_every=1000000",
+     "metric.disable_timer=False", "algo.run_test=False", "checkpoint.every=0",
+     "checkpoint.save_last=False", f"root_dir={tmp}"])
+t = timer.compute()
+tt = t.get("Time/train_time", 0.0)
+print(f"p2e_dv3_exploration: {tt / max(STEPS, 1) * 1000:.1f} ms per gradient step "
+      f"({STEPS} steps incl. capture warmup; cumulative train time {tt:.1f} s)")
