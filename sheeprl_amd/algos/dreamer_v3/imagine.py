@@ -121,26 +121,32 @@ def imagine_rollout(
             ext.ln_act_fwd_o(pre, blk.ln_weight, blk.ln_bias, blk.ln_eps, True, hid, mr[0], mr[1])
             x = hid
         torch.addmm(head.bias, x, head.weight.t(), out=alogits)
-        ext.cat_st_fwd_o(alogits, urand_a[i], a_unimix, am_tmp, acts[i], as_tmp)
+        # the sampled action ALSO lands in the next step's GEMM input slice
+        ext.cat_st_fwd_o(alogits, urand_a[i], a_unimix, am_tmp, acts[i], as_tmp, xs[:, SK:])
 
+    # the producing kernels write every next-step input in place (second /
+    # third strided outputs), so the per-step assembly copies are gone; only
+    # the i=1 inputs (z0/h0, not produced by a kernel here) are staged
     traj[0, :, :SK] = z0
     traj[0, :, SK:] = h0
     h_buf.copy_(h0)
+    xs[:, :SK].copy_(z0)
+    hu[:, :H].copy_(h0)
     actor_step(0)
     for i in range(1, HZ + 1):
         # --- RSSM imagination step (recurrent_model + transition) ---
-        xs[:, :SK].copy_(traj[i - 1][:, :SK])
-        xs[:, SK:].copy_(acts[i - 1])
         torch.mm(xs, w1.t(), out=g1)
         ext.ln_act_fwd_o(g1, mlp_block.ln_weight, mlp_block.ln_bias, eps, True, hu[:, H:], mr[0], mr[1])
-        hu[:, :H].copy_(h_buf)
         torch.mm(hu, w2.t(), out=y)
-        ext.gru_gates_fwd_o(y, hu[:, :H], gru.ln_weight, gru.ln_bias, eps, h_buf, traj[i][:, SK:], mr[0], mr[1])
+        # h_i goes to h_buf, the trajectory slice, AND next step's hu[:, :H]
+        # (the kernel reads each hu row element before overwriting it)
+        ext.gru_gates_fwd_o(y, hu[:, :H], gru.ln_weight, gru.ln_bias, eps, h_buf, traj[i][:, SK:],
+                            mr[0], mr[1], hu[:, :H])
         torch.mm(h_buf, wt1.t(), out=tg)
         ext.ln_act_fwd_o(tg, trans[0].ln_weight, trans[0].ln_bias, trans[0].ln_eps, True, tp, mr[0], mr[1])
         torch.addmm(bt2, tp, wt2.t(), out=traw)
         ext.cat_st_fwd_o(traw.view(B, S, K), urand_t[i - 1], unimix, m_tmp.view(B, S, K),
-                         traj[i][:, :SK], s_tmp)
+                         traj[i][:, :SK], s_tmp, xs[:, :SK])
         # --- policy on the new latent ---
         actor_step(i)
     return traj, acts

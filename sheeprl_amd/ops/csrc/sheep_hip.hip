@@ -844,7 +844,7 @@ template <typename T, typename TW>
 __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ h, const TW* __restrict__ w,
                                      const TW* __restrict__ b, T* __restrict__ hout, float* __restrict__ mean_out,
                                      float* __restrict__ rstd_out, int H, float eps, long hs,
-                                     T* __restrict__ hout2, long h2s) {
+                                     T* __restrict__ hout2, long h2s, T* __restrict__ hout3, long h3s) {
   __shared__ float lds[18];
   const long row = blockIdx.x;
   const int D = 3 * H;
@@ -852,6 +852,7 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
   const T* hr = h + row * hs;
   T* outr = hout + row * (long)H;
   T* outr2 = hout2 ? hout2 + row * h2s : nullptr;
+  T* outr3 = hout3 ? hout3 + row * h3s : nullptr;
   // single-read register-cached path (latency-bound at scan batch sizes):
   // cache element j at index (j - threadIdx.x)/blockDim, so the gates pass
   // finds positions j / H+j / 2H+j when H is a blockDim multiple
@@ -885,6 +886,7 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
       float hv = u * c + (1.f - u) * ld(hr, j);
       st(outr, j, hv);
       if (outr2) st(outr2, j, hv);
+      if (outr3) st(outr3, j, hv);
     }
     return;
   }
@@ -913,12 +915,14 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
     float hv = u * c + (1.f - u) * ld(hr, j);
     st(outr, j, hv);
     if (outr2) st(outr2, j, hv);
+    if (outr3) st(outr3, j, hv);
   }
 }
 
 void gru_gates_fwd_core(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
                         const torch::Tensor& b, double eps, torch::Tensor& hout, torch::Tensor& mean,
-                        torch::Tensor& rstd, long hs, void* hout2, long h2s) {
+                        torch::Tensor& rstd, long hs, void* hout2, long h2s, void* hout3 = nullptr,
+                        long h3s = 0) {
   CHECK_IN(y);
   TORCH_CHECK(y.dim() == 2 && h.dim() == 2 && y.size(1) == 3 * h.size(1), "gru_gates_fwd shapes");
   long N = y.size(0);
@@ -933,7 +937,7 @@ void gru_gates_fwd_core(const torch::Tensor& y, const torch::Tensor& h, const to
       hipLaunchKernelGGL((gru_gates_fwd_kernel<T, TW>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                          (const T*)y.data_ptr(), (const T*)h.data_ptr(), (const TW*)wc.data_ptr(),
                          (const TW*)bc.data_ptr(), (T*)hout.data_ptr(), mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), H, (float)eps, hs, (T*)hout2, h2s);
+                         rstd.data_ptr<float>(), H, (float)eps, hs, (T*)hout2, h2s, (T*)hout3, h3s);
     });
   });
 }
@@ -954,10 +958,19 @@ std::vector<torch::Tensor> gru_gates_fwd(const torch::Tensor& y, const torch::Te
 // stacked representation-input buffer.  mean/rstd are caller-provided slices.
 void gru_gates_fwd_o(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
                      const torch::Tensor& b, double eps, torch::Tensor hout, torch::Tensor hout2,
-                     torch::Tensor mean, torch::Tensor rstd) {
+                     torch::Tensor mean, torch::Tensor rstd,
+                     const c10::optional<torch::Tensor>& hout3 = c10::nullopt) {
   TORCH_CHECK(h.dim() == 2 && h.stride(1) == 1 && hout.is_contiguous(), "gru_gates_fwd_o shapes");
   TORCH_CHECK(hout2.dim() == 2 && hout2.stride(1) == 1 && hout2.scalar_type() == hout.scalar_type());
-  gru_gates_fwd_core(y, h, w, b, eps, hout, mean, rstd, h.stride(0), hout2.data_ptr(), hout2.stride(0));
+  void* h3 = nullptr;
+  long h3s = 0;
+  if (hout3.has_value()) {
+    TORCH_CHECK(hout3->dim() == 2 && hout3->stride(1) == 1 && hout3->scalar_type() == hout.scalar_type());
+    h3 = hout3->data_ptr();
+    h3s = hout3->stride(0);
+  }
+  gru_gates_fwd_core(y, h, w, b, eps, hout, mean, rstd, h.stride(0), hout2.data_ptr(), hout2.stride(0),
+                     h3, h3s);
 }
 
 template <typename T, typename TW>
@@ -1310,7 +1323,8 @@ void adam_step(std::vector<torch::Tensor> params, std::vector<torch::Tensor> gra
 template <typename T, bool SAMPLE>
 __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __restrict__ urand,
                                   float* __restrict__ m_out, T* __restrict__ onehot, float* __restrict__ s_out,
-                                  long nrows, int K, float unimix, long ohs, long spb) {
+                                  long nrows, int K, float unimix, long ohs, long spb,
+                                  T* __restrict__ onehot2, long oh2s) {
   const int lane = threadIdx.x & 63;
   const long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (row >= nrows) return;
@@ -1320,6 +1334,7 @@ __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __rest
   // onehot may be a [B, S*K] row-strided slice: row = b*spb + s maps to
   // b*ohs + s*K (ohs = S*K, spb = S reproduces the contiguous layout)
   T* oh = onehot + (row / spb) * ohs + (row % spb) * (long)K;
+  T* oh2 = onehot2 ? onehot2 + (row / spb) * oh2s + (row % spb) * (long)K : nullptr;
   // row max
   float lmax = -1e30f;
   for (int j = lane; j < K; j += 64) lmax = fmaxf(lmax, ld(L, j));
@@ -1360,7 +1375,11 @@ __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __rest
       best_j = oj;
     }
   }
-  for (int j = lane; j < K; j += 64) st(oh, j, j == best_j ? 1.f : 0.f);
+  for (int j = lane; j < K; j += 64) {
+    const float z = j == best_j ? 1.f : 0.f;
+    st(oh, j, z);
+    if (oh2) st(oh2, j, z);
+  }
 }
 
 // cat_st forward fused with the NEXT scan step's reset-masked input
@@ -1592,11 +1611,11 @@ std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optio
     if (sample)
       hipLaunchKernelGGL((cat_st_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
                          (const T*)raw.data_ptr(), up, m.data_ptr<float>(), (T*)onehot.data_ptr(),
-                         s.data_ptr<float>(), nrows, K, (float)unimix, (long)K, 1);
+                         s.data_ptr<float>(), nrows, K, (float)unimix, (long)K, 1, (T*)nullptr, 0L);
     else
       hipLaunchKernelGGL((cat_st_fwd_kernel<T, false>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
                          (const T*)raw.data_ptr(), up, m.data_ptr<float>(), (T*)onehot.data_ptr(),
-                         s.data_ptr<float>(), nrows, K, (float)unimix, (long)K, 1);
+                         s.data_ptr<float>(), nrows, K, (float)unimix, (long)K, 1, (T*)nullptr, 0L);
   });
   return {m, onehot, s};
 }
@@ -1605,7 +1624,8 @@ std::vector<torch::Tensor> cat_st_fwd(const torch::Tensor& raw, const c10::optio
 // slices of stacked [T, ...] buffers (all contiguous); onehot is written in
 // the compute dtype directly (no separate cast kernel).
 void cat_st_fwd_o(const torch::Tensor& raw, const torch::Tensor& urand, double unimix, torch::Tensor m,
-                  torch::Tensor onehot, torch::Tensor s) {
+                  torch::Tensor onehot, torch::Tensor s,
+                  const c10::optional<torch::Tensor>& onehot2 = c10::nullopt) {
   CHECK_IN(raw);
   TORCH_CHECK(m.is_contiguous() && s.is_contiguous(), "cat_st_fwd_o outputs");
   TORCH_CHECK(onehot.scalar_type() == raw.scalar_type(), "onehot dtype must match raw");
@@ -1626,9 +1646,20 @@ void cat_st_fwd_o(const torch::Tensor& raw, const torch::Tensor& urand, double u
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, raw.scalar_type(), "cat_st_fwd_o", [&] {
     using T = scalar_t;
+    T* oh2 = nullptr;
+    long oh2s = 0;
+    if (onehot2.has_value()) {
+      // same [rows-of-onehot, spb*K] geometry, its own row stride
+      TORCH_CHECK(onehot2->dim() == 2 && onehot2->stride(1) == 1 &&
+                      onehot2->scalar_type() == raw.scalar_type(),
+                  "cat_st_fwd_o: onehot2 must be a row-strided 2-D slice");
+      oh2 = (T*)onehot2->data_ptr();
+      oh2s = onehot2->stride(0);
+    }
     hipLaunchKernelGGL((cat_st_fwd_kernel<T, true>), dim3(blocks), dim3(kBlock), 0, stream.stream(),
                        (const T*)raw.data_ptr(), urand.data_ptr<float>(), m.data_ptr<float>(),
-                       (T*)onehot.data_ptr(), s.data_ptr<float>(), nrows, K, (float)unimix, ohs, spb);
+                       (T*)onehot.data_ptr(), s.data_ptr<float>(), nrows, K, (float)unimix, ohs, spb,
+                       oh2, oh2s);
   });
 }
 
@@ -5589,9 +5620,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // fused-scan variants (strided views / caller-provided outputs+accumulators)
   m.def("ln_act_fwd_o", &ln_act_fwd_o);
   m.def("ln_act_bwd_acc", &ln_act_bwd_acc);
-  m.def("gru_gates_fwd_o", &gru_gates_fwd_o);
+  m.def("gru_gates_fwd_o", &gru_gates_fwd_o, py::arg("y"), py::arg("h"), py::arg("w"), py::arg("b"),
+        py::arg("eps"), py::arg("hout"), py::arg("hout2"), py::arg("mean"), py::arg("rstd"),
+        py::arg("hout3") = c10::nullopt);
   m.def("gru_gates_bwd_acc", &gru_gates_bwd_acc);
-  m.def("cat_st_fwd_o", &cat_st_fwd_o);
+  m.def("cat_st_fwd_o", &cat_st_fwd_o, py::arg("raw"), py::arg("urand"), py::arg("unimix"),
+        py::arg("m"), py::arg("onehot"), py::arg("s"), py::arg("onehot2") = c10::nullopt);
   m.def("cat_st_bwd_o", &cat_st_bwd_o);
   m.def("ema_update", &ema_update);
   m.def("obs_norm", &obs_norm);
