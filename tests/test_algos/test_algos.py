@@ -706,6 +706,54 @@ def test_eval_and_registration_cli(tmp_path):
         os.chdir(cwd)
 
 
+@pytest.mark.timeout(300)
+def test_eval_cli_dreamer_v3(tmp_path):
+    """Train tiny DV3 with save_last, then run the `eval` CLI on its
+    checkpoint (parity: sheeprl-eval over the flagship family)."""
+    _run(
+        tmp_path,
+        standard_args(
+            tmp_path,
+            [
+                "exp=dreamer_v3",
+                "algo=dreamer_v3_S",
+                "algo.dense_units=8",
+                "algo.mlp_layers=1",
+                "algo.world_model.encoder.cnn_channels_multiplier=2",
+                "algo.world_model.recurrent_model.recurrent_state_size=8",
+                "algo.world_model.transition_model.hidden_size=8",
+                "algo.world_model.representation_model.hidden_size=8",
+                "algo.world_model.discrete_size=4",
+                "algo.world_model.stochastic_size=4",
+                "algo.per_rank_batch_size=2",
+                "algo.per_rank_sequence_length=4",
+                "algo.horizon=3",
+                "algo.mlp_keys.encoder=[state]",
+                "algo.total_steps=16",
+                "algo.learning_starts=4",
+                "algo.replay_ratio=0.5",
+                "buffer.size=64",
+                "algo.run_test=False",
+                "checkpoint.save_last=True",
+                "dry_run=False",
+            ],
+        ),
+    )
+    import glob
+    import os
+
+    from sheeprl_amd.cli import evaluation
+
+    ckpts = sorted(glob.glob(str(tmp_path / "logs" / "runs" / "**" / "ckpt_*.ckpt"), recursive=True))
+    assert ckpts
+    cwd = os.getcwd()
+    os.chdir(tmp_path)
+    try:
+        evaluation([f"checkpoint_path={ckpts[-1]}", "runtime.accelerator=cpu"])
+    finally:
+        os.chdir(cwd)
+
+
 def test_minedojo_actor_masking():
     import torch
     from sheeprl_amd.algos.dreamer_v3.agent import MinedojoActor
