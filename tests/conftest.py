@@ -31,3 +31,15 @@ def _destroy_process_group():
 def tmp_run_dir(tmp_path, monkeypatch):
     monkeypatch.chdir(tmp_path)
     return tmp_path
+
+
+@pytest.fixture(autouse=True)
+def _restore_global_metric_flags():
+    """metric.log_level=0 runs flip the CLASS-LEVEL timer/aggregator disable
+    flags; snapshot+restore them so test outcomes don't depend on ordering."""
+    from sheeprl_amd.utils.metric import MetricAggregator
+    from sheeprl_amd.utils.timer import timer
+
+    t, m = timer.disabled, MetricAggregator.disabled
+    yield
+    timer.disabled, MetricAggregator.disabled = t, m

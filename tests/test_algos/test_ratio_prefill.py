@@ -38,6 +38,13 @@ def test_ratio_counts_start_after_prefill(tmp_path, monkeypatch):
         return orig(*a, **k)
 
     monkeypatch.setattr(dv3, "train", counting_train)
+    # metric.log_level=0 flips the class-level timer/aggregator disable
+    # flags; restore them so later tests see the default state
+    from sheeprl_amd.utils.metric import MetricAggregator
+    from sheeprl_amd.utils.timer import timer
+
+    monkeypatch.setattr(timer, "disabled", timer.disabled, raising=False)
+    monkeypatch.setattr(MetricAggregator, "disabled", MetricAggregator.disabled, raising=False)
     run([
         "exp=dreamer_v3", "algo=dreamer_v3_S", "env=dummy", "runtime.accelerator=cpu",
         "env.num_envs=1", "seed=0",
