@@ -27,7 +27,7 @@ import torch.nn.functional as F
 from sheeprl_amd import ops
 from sheeprl_amd.algos.dreamer_v3.agent import build_agent
 from sheeprl_amd.algos.dreamer_v3.imagine import imagine_applicable, imagine_rollout
-from sheeprl_amd.algos.dreamer_v3.loss import reconstruction_loss
+from sheeprl_amd.algos.dreamer_v3.loss import reconstruction_loss, reconstruction_loss_forked
 from sheeprl_amd.algos.dreamer_v3.utils import (
     AGGREGATOR_KEYS,
     Moments,
@@ -46,6 +46,7 @@ from sheeprl_amd.distributions import (
 from sheeprl_amd.envs import make_env, spaces, vectorize_env
 from sheeprl_amd.optim import make_optimizer
 from sheeprl_amd.parallel import Runtime
+from sheeprl_amd.parallel.streams import Branches
 from sheeprl_amd.utils.logger import get_log_dir, get_logger
 from sheeprl_amd.utils.metric import MetricAggregator
 from sheeprl_amd.utils.registry import register_algorithm
@@ -132,48 +133,98 @@ def train(
                 recurrent_states[i] = recurrent_state
                 posteriors[i] = posterior
                 posteriors_logits[i] = posterior_logits
-        priors_logits = world_model.rssm.transition_logits(recurrent_states)
+        # the batched prior (transition) head is off the scan's critical
+        # path; it is computed at the loss site (inside a stream fork on GPU)
+        priors_logits = None
     latent_states = torch.cat((posteriors.view(*posteriors.shape[:-2], -1), recurrent_states), -1)
 
-    reconstructed_obs = world_model.observation_model(latent_states)
+    # Intra-graph concurrency (parallel/streams.py): the world-model head
+    # chains and the actor/critic behaviour phases are mutually independent;
+    # forking them onto side streams lets the captured hipGraph replay run
+    # them concurrently on idle CUs.  Enabled on single-rank GPU runs only —
+    # multi-rank backward interleaves GradSync collectives, which stay
+    # single-stream for deterministic enqueue order.
+    br = Branches(
+        device.type == "cuda"
+        and runtime.world_size == 1
+        and bool(cfg.algo.get("graph_streams", True))
+        and os.environ.get("SHEEPRL_AMD_GRAPH_STREAMS", "1") != "0"
+    )
+
     # on the HIP path the fused NLL kernels take bf16 predictions directly —
     # skipping the fp32 upcast halves the loss-section reads and removes two
     # cast launches per key (fwd + backward grad cast)
     _no_cast = device.type == "cuda" and ops.use_hip(latent_states)
-    _c = (lambda t: t) if _no_cast else (lambda t: t.float())
-    po = {
-        k: MSEDistribution(_c(reconstructed_obs[k]), dims=len(reconstructed_obs[k].shape[2:]))
-        for k in cfg.algo.cnn_keys.decoder
-    }
-    po.update(
-        {
-            k: SymlogDistribution(_c(reconstructed_obs[k]), dims=len(reconstructed_obs[k].shape[2:]))
-            for k in cfg.algo.mlp_keys.decoder
-        }
-    )
-    pr = TwoHotEncodingDistribution(world_model.reward_model(latent_states).float(), dims=1)
-    pc = td.Independent(BernoulliSafeMode(logits=world_model.continue_model(latent_states).float()), 1)
     continues_targets = 1 - data["terminated"]
 
-    priors_logits = priors_logits.view(*priors_logits.shape[:-1], stochastic_size, discrete_size)
-    posteriors_logits = posteriors_logits.view(*posteriors_logits.shape[:-1], stochastic_size, discrete_size)
-
     world_optimizer.zero_grad(set_to_none=True)
-    rec_loss, kl, state_loss, reward_loss, observation_loss, continue_loss = reconstruction_loss(
-        po,
-        {k: v.float() for k, v in batch_obs.items()},
-        pr,
-        data["rewards"],
-        priors_logits,
-        posteriors_logits,
-        cfg.algo.world_model.kl_dynamic,
-        cfg.algo.world_model.kl_representation,
-        cfg.algo.world_model.kl_free_nats,
-        cfg.algo.world_model.kl_regularizer,
-        pc,
-        continues_targets,
-        cfg.algo.world_model.continue_scale_factor,
-    )
+    if br.enabled:
+        (
+            rec_loss,
+            kl,
+            state_loss,
+            reward_loss,
+            observation_loss,
+            continue_loss,
+            priors_logits,
+            posteriors_logits,
+        ) = reconstruction_loss_forked(
+            br,
+            world_model,
+            latent_states,
+            recurrent_states,
+            batch_obs,
+            data["rewards"],
+            continues_targets,
+            posteriors_logits,
+            priors_logits,
+            stochastic_size,
+            discrete_size,
+            cfg.algo.cnn_keys.decoder,
+            cfg.algo.mlp_keys.decoder,
+            cfg.algo.world_model.kl_dynamic,
+            cfg.algo.world_model.kl_representation,
+            cfg.algo.world_model.kl_free_nats,
+            cfg.algo.world_model.kl_regularizer,
+            cfg.algo.world_model.continue_scale_factor,
+            _no_cast,
+        )
+    else:
+        if priors_logits is None:
+            priors_logits = world_model.rssm.transition_logits(recurrent_states)
+        reconstructed_obs = world_model.observation_model(latent_states)
+        _c = (lambda t: t) if _no_cast else (lambda t: t.float())
+        po = {
+            k: MSEDistribution(_c(reconstructed_obs[k]), dims=len(reconstructed_obs[k].shape[2:]))
+            for k in cfg.algo.cnn_keys.decoder
+        }
+        po.update(
+            {
+                k: SymlogDistribution(_c(reconstructed_obs[k]), dims=len(reconstructed_obs[k].shape[2:]))
+                for k in cfg.algo.mlp_keys.decoder
+            }
+        )
+        pr = TwoHotEncodingDistribution(world_model.reward_model(latent_states).float(), dims=1)
+        pc = td.Independent(BernoulliSafeMode(logits=world_model.continue_model(latent_states).float()), 1)
+
+        priors_logits = priors_logits.view(*priors_logits.shape[:-1], stochastic_size, discrete_size)
+        posteriors_logits = posteriors_logits.view(*posteriors_logits.shape[:-1], stochastic_size, discrete_size)
+
+        rec_loss, kl, state_loss, reward_loss, observation_loss, continue_loss = reconstruction_loss(
+            po,
+            {k: v.float() for k, v in batch_obs.items()},
+            pr,
+            data["rewards"],
+            priors_logits,
+            posteriors_logits,
+            cfg.algo.world_model.kl_dynamic,
+            cfg.algo.world_model.kl_representation,
+            cfg.algo.world_model.kl_free_nats,
+            cfg.algo.world_model.kl_regularizer,
+            pc,
+            continues_targets,
+            cfg.algo.world_model.continue_scale_factor,
+        )
     runtime.backward(rec_loss)
     world_model_grads = None
     if cfg.algo.world_model.clip_gradients and cfg.algo.world_model.clip_gradients > 0:
@@ -228,9 +279,14 @@ def train(
     fast_losses = use_fast_imagine and len(actions_dim) == 1
     if fast_losses:
         with torch.no_grad():
-            predicted_values = ops.twohot_mean(critic(imagined_trajectories))
-            predicted_rewards = ops.twohot_mean(world_model.reward_model(imagined_trajectories))
-            continues = (world_model.continue_model(imagined_trajectories) > 0).float()
+            # the three trajectory sweeps are independent — fork them
+            with br.fork():
+                predicted_values = ops.twohot_mean(critic(imagined_trajectories))
+            with br.fork():
+                predicted_rewards = ops.twohot_mean(world_model.reward_model(imagined_trajectories))
+            with br.fork():
+                continues = (world_model.continue_model(imagined_trajectories) > 0).float()
+            br.join()
             true_continue = (1 - data["terminated"]).flatten().reshape(1, -1, 1)
             continues = torch.cat((true_continue, continues[1:]))
             lambda_values = compute_lambda_values(
@@ -239,34 +295,43 @@ def train(
             )
             discount = torch.cumprod(continues * cfg.algo.gamma, dim=0) / cfg.algo.gamma
 
-        # actor loss (advantage offsets cancel: (λ-off)/s - (v-off)/s = (λ-v)/s)
-        actor_optimizer.zero_grad(set_to_none=True)
-        policies = actor(imagined_trajectories.detach())[1]
+        # Moments mutates EMA state and (multi-rank) all-gathers: keep it on
+        # the ambient stream, ahead of the actor/critic forks
         offset, invscale = moments(lambda_values, runtime)
-        with torch.no_grad():
-            advantage = (lambda_values - predicted_values[:-1]) / invscale
-        policy_loss = ops.reinforce_loss(
-            policies[0].logits, imagined_actions, advantage, discount[:-1], cfg.algo.actor.ent_coef
-        )
-        runtime.backward(policy_loss)
-        actor_grads = None
-        if cfg.algo.actor.clip_gradients and cfg.algo.actor.clip_gradients > 0:
-            actor_grads = runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
-        actor_optimizer.step()
 
-        # critic loss: two two-hot CEs over shared logits in one kernel
-        critic_optimizer.zero_grad(set_to_none=True)
-        qv_logits = critic(imagined_trajectories.detach()[:-1]).float()
-        with torch.no_grad():
-            predicted_target_values = ops.twohot_mean(target_critic(imagined_trajectories.detach()[:-1]))
-        value_loss = ops.critic_twohot_loss(
-            qv_logits, lambda_values, predicted_target_values, discount[:-1]
-        )
-        runtime.backward(value_loss)
-        critic_grads = None
-        if cfg.algo.critic.clip_gradients and cfg.algo.critic.clip_gradients > 0:
-            critic_grads = runtime.clip_gradients(critic, critic_optimizer, cfg.algo.critic.clip_gradients)
-        critic_optimizer.step()
+        # the whole actor phase and the whole critic phase (forward, backward,
+        # clip, optimizer step) touch disjoint parameters and read-only share
+        # the trajectories/λ-values: run them concurrently
+        with br.fork():
+            # actor loss (advantage offsets cancel: (λ-off)/s - (v-off)/s = (λ-v)/s)
+            actor_optimizer.zero_grad(set_to_none=True)
+            policies = actor(imagined_trajectories.detach())[1]
+            with torch.no_grad():
+                advantage = (lambda_values - predicted_values[:-1]) / invscale
+            policy_loss = ops.reinforce_loss(
+                policies[0].logits, imagined_actions, advantage, discount[:-1], cfg.algo.actor.ent_coef
+            )
+            runtime.backward(policy_loss)
+            actor_grads = None
+            if cfg.algo.actor.clip_gradients and cfg.algo.actor.clip_gradients > 0:
+                actor_grads = runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
+            actor_optimizer.step()
+
+        with br.fork():
+            # critic loss: two two-hot CEs over shared logits in one kernel
+            critic_optimizer.zero_grad(set_to_none=True)
+            qv_logits = critic(imagined_trajectories.detach()[:-1]).float()
+            with torch.no_grad():
+                predicted_target_values = ops.twohot_mean(target_critic(imagined_trajectories.detach()[:-1]))
+            value_loss = ops.critic_twohot_loss(
+                qv_logits, lambda_values, predicted_target_values, discount[:-1]
+            )
+            runtime.backward(value_loss)
+            critic_grads = None
+            if cfg.algo.critic.clip_gradients and cfg.algo.critic.clip_gradients > 0:
+                critic_grads = runtime.clip_gradients(critic, critic_optimizer, cfg.algo.critic.clip_gradients)
+            critic_optimizer.step()
+        br.join()
     else:
         predicted_values = TwoHotEncodingDistribution(critic(imagined_trajectories).float(), dims=1).mean
         predicted_rewards = TwoHotEncodingDistribution(world_model.reward_model(imagined_trajectories).float(), dims=1).mean
