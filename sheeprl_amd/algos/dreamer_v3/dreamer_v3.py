@@ -144,12 +144,12 @@ def train(
     # them concurrently on idle CUs.  Enabled on single-rank GPU runs only —
     # multi-rank backward interleaves GradSync collectives, which stay
     # single-stream for deterministic enqueue order.
-    br = Branches(
-        device.type == "cuda"
-        and runtime.world_size == 1
-        and bool(cfg.algo.get("graph_streams", True))
-        and os.environ.get("SHEEPRL_AMD_GRAPH_STREAMS", "1") != "0"
-    )
+    # region selection for A/B: "0"=off, "all", "heads" (world-model head
+    # forks only), "behaviour" (actor/critic + trajectory-sweep forks only)
+    _sm = os.environ.get("SHEEPRL_AMD_GRAPH_STREAMS", "0")
+    _stream_ok = device.type == "cuda" and runtime.world_size == 1 and bool(cfg.algo.get("graph_streams", True))
+    br = Branches(_stream_ok and _sm in ("1", "all", "heads"))
+    br_beh = Branches(_stream_ok and _sm in ("1", "all", "behaviour"))
 
     # on the HIP path the fused NLL kernels take bf16 predictions directly —
     # skipping the fp32 upcast halves the loss-section reads and removes two
@@ -280,13 +280,13 @@ def train(
     if fast_losses:
         with torch.no_grad():
             # the three trajectory sweeps are independent — fork them
-            with br.fork():
+            with br_beh.fork():
                 predicted_values = ops.twohot_mean(critic(imagined_trajectories))
-            with br.fork():
+            with br_beh.fork():
                 predicted_rewards = ops.twohot_mean(world_model.reward_model(imagined_trajectories))
-            with br.fork():
+            with br_beh.fork():
                 continues = (world_model.continue_model(imagined_trajectories) > 0).float()
-            br.join()
+            br_beh.join()
             true_continue = (1 - data["terminated"]).flatten().reshape(1, -1, 1)
             continues = torch.cat((true_continue, continues[1:]))
             lambda_values = compute_lambda_values(
@@ -302,7 +302,7 @@ def train(
         # the whole actor phase and the whole critic phase (forward, backward,
         # clip, optimizer step) touch disjoint parameters and read-only share
         # the trajectories/λ-values: run them concurrently
-        with br.fork():
+        with br_beh.fork():
             # actor loss (advantage offsets cancel: (λ-off)/s - (v-off)/s = (λ-v)/s)
             actor_optimizer.zero_grad(set_to_none=True)
             policies = actor(imagined_trajectories.detach())[1]
@@ -317,7 +317,7 @@ def train(
                 actor_grads = runtime.clip_gradients(actor, actor_optimizer, cfg.algo.actor.clip_gradients)
             actor_optimizer.step()
 
-        with br.fork():
+        with br_beh.fork():
             # critic loss: two two-hot CEs over shared logits in one kernel
             critic_optimizer.zero_grad(set_to_none=True)
             qv_logits = critic(imagined_trajectories.detach()[:-1]).float()
@@ -331,7 +331,7 @@ def train(
             if cfg.algo.critic.clip_gradients and cfg.algo.critic.clip_gradients > 0:
                 critic_grads = runtime.clip_gradients(critic, critic_optimizer, cfg.algo.critic.clip_gradients)
             critic_optimizer.step()
-        br.join()
+        br_beh.join()
     else:
         predicted_values = TwoHotEncodingDistribution(critic(imagined_trajectories).float(), dims=1).mean
         predicted_rewards = TwoHotEncodingDistribution(world_model.reward_model(imagined_trajectories).float(), dims=1).mean
