@@ -53,6 +53,14 @@ class Moments(nn.Module):
 
     def forward(self, x: Tensor, runtime: Runtime):
         gathered = runtime.all_gather(x.detach()).float()
+        if gathered.is_cuda and gathered.numel() <= 32768 and ops.use_hip(gathered):
+            # one-kernel LDS-sort quantiles + in-place EMA (§2.8 item 9);
+            # replaces the torch.quantile sort path + six elementwise launches
+            invscale = ops.moments_update(
+                gathered, self.low, self.high,
+                self._percentile_low, self._percentile_high, self._decay, self._max,
+            )
+            return self.low.detach().clone(), invscale.detach()
         low = torch.quantile(gathered, self._percentile_low)
         high = torch.quantile(gathered, self._percentile_high)
         # in-place EMA so the buffers keep fixed storage (hipGraph-replayable)

@@ -64,7 +64,19 @@ class SACActor(nn.Module):
         log pi(a) = logN(x) - sum log(scale * (1 - tanh(x)^2)); the tanh term
         is inside TanhNormal.rsample_with_log_prob, leaving the scale term.
         """
-        dist = self._dist(obs)
+        x = self.model(obs)
+        mean = self.fc_mean(x).float()
+        log_std = self.fc_logstd(x).float()
+        if mean.is_cuda and ops.use_hip(mean):
+            # fused sample: clamp/exp + reparam + tanh + rescale + summed
+            # log-prob in ONE kernel each way (§2.8 item 13); philox noise
+            # stays a torch op so the step remains hipGraph-capturable
+            eps = torch.randn_like(mean)
+            return ops.tanh_normal_sample(
+                mean, log_std, eps, self.action_scale.float(), self.action_bias.float(), obs,
+                LOG_STD_MIN, LOG_STD_MAX,
+            )
+        dist = TanhNormal(mean, log_std.clamp(LOG_STD_MIN, LOG_STD_MAX).exp())
         y, logp = dist.rsample_with_log_prob()
         # sampling/log-prob math runs in fp32 for stability; the action feeds
         # the (possibly bf16) critics in the module dtype

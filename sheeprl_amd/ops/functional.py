@@ -343,3 +343,47 @@ class _CriticTwohotLoss(torch.autograd.Function):
 def critic_twohot_loss(logits: Tensor, t1: Tensor, t2: Tensor, disc: Tensor,
                        low: float = -20.0, high: float = 20.0) -> Tensor:
     return _CriticTwohotLoss.apply(logits, t1, t2, disc, low, high)
+
+
+def moments_update(x: Tensor, low: Tensor, high: Tensor, p_low: float, p_high: float,
+                   decay: float, max_: float) -> Tensor:
+    """One-kernel Moments percentile-EMA update (SURVEY.md §2.8 item 9;
+    parity: sheeprl/algos/dreamer_v3/utils.py:56-63): LDS bitonic sort of the
+    gathered returns, linearly-interpolated quantiles (torch.quantile
+    "linear"), in-place EMA of the ``low``/``high`` buffers and
+    invscale = max(high - low, 1/max_).  Requires x.numel() <= 32768 (the
+    128 KB LDS sort cap) — callers fall back to torch.quantile beyond it."""
+    return require_ext().moments_update(x.reshape(-1).contiguous(), low, high,
+                                        p_low, p_high, decay, max_)
+
+
+class _TanhNormalSample(torch.autograd.Function):
+    """Fused reparameterized tanh-Normal sample + summed log-prob (the SAC
+    actor head; SURVEY.md §2.8 item 13, parity: sheeprl/algos/sac/agent.py:
+    123-142).  One kernel each way; backward recomputes std/x/y from inputs."""
+
+    @staticmethod
+    def forward(ctx, mean: Tensor, logstd: Tensor, eps: Tensor, scale: Tensor,
+                bias: Tensor, like: Tensor, lmin: float, lmax: float):
+        mc, lc, ec = mean.contiguous(), logstd.contiguous(), eps.contiguous()
+        sc, bc = scale.contiguous(), bias.contiguous()
+        action, logp = require_ext().tanh_normal_fwd(mc, lc, ec, sc, bc, like, lmin, lmax)
+        ctx.save_for_backward(mc, lc, ec, sc)
+        ctx.meta = (lmin, lmax)
+        return action, logp
+
+    @staticmethod
+    def backward(ctx, gaction: Tensor, glogp: Tensor):
+        mean, logstd, eps, scale = ctx.saved_tensors
+        lmin, lmax = ctx.meta
+        dmean, dlogstd = require_ext().tanh_normal_bwd(
+            gaction.contiguous(), glogp.reshape(-1).contiguous().float(),
+            mean, logstd, eps, scale, lmin, lmax,
+        )
+        return dmean, dlogstd, None, None, None, None, None, None
+
+
+def tanh_normal_sample(mean: Tensor, logstd: Tensor, eps: Tensor, scale: Tensor,
+                       bias: Tensor, like: Tensor, lmin: float = -5.0, lmax: float = 2.0):
+    """Returns (action in env range, per-row summed log-prob [..., 1])."""
+    return _TanhNormalSample.apply(mean, logstd, eps, scale, bias, like, lmin, lmax)

@@ -974,3 +974,88 @@ def test_fused_clip_grad_norm(dtype, scale):
     got2 = rt.clip_gradients(None, opt, max_norm=max_norm)
     expect2 = torch.nn.utils.clip_grad_norm_(ref, max_norm)
     assert torch.allclose(got2.float().cpu(), expect2.float().cpu(), rtol=tol)
+
+
+@requires_gpu
+def test_moments_update_matches_quantile():
+    """One-kernel Moments update (LDS bitonic sort + EMA) vs the eager
+    torch.quantile path, over several EMA iterations (§2.8 item 9)."""
+    torch.manual_seed(0)
+    decay, max_, pl, ph = 0.99, 1e8, 0.05, 0.95
+    low_a = torch.zeros((), device="cuda")
+    high_a = torch.zeros((), device="cuda")
+    low_b = torch.zeros((), device="cuda")
+    high_b = torch.zeros((), device="cuda")
+    for it in range(4):
+        n = [15360, 1000, 7, 32768][it]
+        x = torch.randn(n, device="cuda") * (it + 1)
+        # eager reference
+        ql = torch.quantile(x, pl)
+        qh = torch.quantile(x, ph)
+        low_a.mul_(decay).add_(ql, alpha=1 - decay)
+        high_a.mul_(decay).add_(qh, alpha=1 - decay)
+        inv_a = torch.clamp(high_a - low_a, min=1.0 / max_)
+        # fused kernel (updates buffers in place)
+        inv_b = ops.moments_update(x, low_b, high_b, pl, ph, decay, max_)
+        torch.testing.assert_close(low_a, low_b, atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(high_a, high_b, atol=1e-5, rtol=1e-5)
+        torch.testing.assert_close(inv_a, inv_b, atol=1e-5, rtol=1e-5)
+
+
+@requires_gpu
+def test_tanh_normal_fused_matches_eager():
+    """Fused SAC actor head (sample + summed log-prob, fwd+bwd) vs the eager
+    fp32 composition (§2.8 item 13)."""
+    import math
+
+    torch.manual_seed(1)
+    B, A = 257, 6
+    lmin, lmax = -5.0, 2.0
+    mean = (torch.randn(B, A, device="cuda") * 2).requires_grad_(True)
+    # logstd spread across and beyond the clamp bounds
+    logstd = (torch.randn(B, A, device="cuda") * 4).requires_grad_(True)
+    eps = torch.randn(B, A, device="cuda")
+    scale = torch.rand(A, device="cuda") + 0.5
+    bias = torch.randn(A, device="cuda")
+    like = torch.zeros(1, device="cuda", dtype=torch.float32)
+
+    # eager reference
+    m2 = mean.detach().clone().requires_grad_(True)
+    l2 = logstd.detach().clone().requires_grad_(True)
+    std = l2.clamp(lmin, lmax).exp()
+    x = m2 + std * eps
+    y = torch.tanh(x)
+    action_ref = y * scale + bias
+    logp = (
+        -0.5 * eps**2 - l2.clamp(lmin, lmax) - 0.5 * math.log(2 * math.pi)
+        - 2.0 * (math.log(2.0) - x - torch.nn.functional.softplus(-2.0 * x))
+        - torch.log(scale)
+    ).sum(-1, keepdim=True)
+
+    action, lp = ops.tanh_normal_sample(mean, logstd, eps, scale, bias, like, lmin, lmax)
+    torch.testing.assert_close(action, action_ref, atol=2e-4, rtol=2e-4)
+    torch.testing.assert_close(lp, logp, atol=3e-4, rtol=3e-4)
+
+    ga = torch.randn_like(action_ref)
+    glp = torch.randn_like(logp)
+    (action_ref * ga + logp * glp).sum().backward()
+    (action * ga + lp * glp).sum().backward()
+    torch.testing.assert_close(mean.grad, m2.grad, atol=3e-4, rtol=3e-4)
+    torch.testing.assert_close(logstd.grad, l2.grad, atol=3e-4, rtol=3e-4)
+
+
+@requires_gpu
+def test_sac_actor_fused_forward_bf16():
+    """SACActor.forward routes through the fused kernel on GPU and returns
+    the module-dtype action + fp32 log-prob with finite values."""
+    from sheeprl_amd.algos.sac.agent import SACActor
+
+    torch.manual_seed(2)
+    actor = SACActor(11, 3, action_low=-2 * torch.ones(3).numpy(), action_high=2 * torch.ones(3).numpy())
+    actor = actor.to(device="cuda", dtype=torch.bfloat16)
+    obs = torch.randn(64, 11, device="cuda", dtype=torch.bfloat16)
+    action, logp = actor(obs)
+    assert action.dtype == torch.bfloat16 and action.shape == (64, 3)
+    assert logp.dtype == torch.float32 and logp.shape == (64, 1)
+    assert action.abs().max() <= 2.0 + 1e-2
+    assert torch.isfinite(logp).all()
