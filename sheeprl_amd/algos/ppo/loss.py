@@ -55,3 +55,33 @@ def entropy_loss(entropy: Tensor, reduction: str = "mean") -> Tensor:
     if reduction == "sum":
         return -entropy.sum()
     return -entropy
+
+
+def ppo_losses(
+    new_logprobs: Tensor,
+    old_logprobs: Tensor,
+    advantages: Tensor,
+    new_values: Tensor,
+    old_values: Tensor,
+    returns: Tensor,
+    entropy: Tensor,
+    clip_coef: float,
+    clip_vloss: bool,
+    reduction: str = "mean",
+):
+    """(policy_loss, value_loss, entropy_loss) triple — ONE fused kernel each
+    way on GPU (ops.ppo_losses), the eager composition above elsewhere.
+    Value- and gradient-identical by construction (tested in
+    tests/test_gpu_kernels.py::test_ppo_losses_fused_matches_eager)."""
+    from sheeprl_amd import ops as _ops
+
+    if reduction in ("mean", "sum") and new_logprobs.is_cuda and _ops.use_hip(new_logprobs):
+        return _ops.ppo_losses(
+            new_logprobs, old_logprobs, advantages, new_values, old_values, returns,
+            entropy, clip_coef, clip_vloss, reduction,
+        )
+    return (
+        policy_loss(new_logprobs, old_logprobs, advantages, clip_coef, reduction),
+        value_loss(new_values, old_values, returns, clip_coef, clip_vloss, reduction),
+        entropy_loss(entropy, reduction),
+    )

@@ -16,7 +16,7 @@ import numpy as np
 import torch
 
 from sheeprl_amd.algos.ppo.agent import build_agent
-from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, value_loss
+from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, ppo_losses, value_loss
 from sheeprl_amd.algos.ppo.utils import AGGREGATOR_KEYS, prepare_obs, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer
@@ -55,16 +55,18 @@ def train(
             if normalize and adv.numel() > 1:
                 adv = (adv - adv.mean()) / (adv.std() + 1e-8)
             _, logprobs, entropy, new_values = agent(obs, batch["actions"])
-            pg_loss = policy_loss(logprobs, batch["logprobs"], adv, cfg.algo.clip_coef, cfg.algo.loss_reduction)
-            v_loss = value_loss(
+            pg_loss, v_loss, ent_loss = ppo_losses(
+                logprobs,
+                batch["logprobs"],
+                adv,
                 new_values.float(),  # bf16-true agent vs fp32 stored targets
                 batch["values"],
                 batch["returns"],
+                entropy,
                 cfg.algo.clip_coef,
                 cfg.algo.clip_vloss,
                 cfg.algo.loss_reduction,
             )
-            ent_loss = entropy_loss(entropy, cfg.algo.loss_reduction)
             loss = pg_loss + cfg.algo.vf_coef * v_loss + cfg.algo.ent_coef * ent_loss
             optimizer.zero_grad(set_to_none=True)
             runtime.backward(loss)

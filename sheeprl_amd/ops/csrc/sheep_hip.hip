@@ -5776,6 +5776,146 @@ std::vector<torch::Tensor> tanh_normal_bwd(const torch::Tensor& gaction, const t
   return {dmean, dlogstd};
 }
 
+// ---------------------------------------------------------------------------
+// Fused PPO losses (parity: sheeprl/algos/ppo/loss.py
+// — policy_loss :6, value_loss :45, entropy_loss :65).  One reduction kernel
+// computes the clipped policy loss, (optionally clipped) value loss and
+// entropy loss plus their weighted total; backward recomputes the branch
+// selections elementwise.  Gradient semantics match the torch composition
+// exactly, including torch.maximum's 0.5/0.5 split on ties — the UNCLIPPED
+// region makes the two policy branches equal, so ties are the common case,
+// not the edge case.
+__global__ void ppo_loss_fwd_kernel(const float* __restrict__ lp_new, const float* __restrict__ lp_old,
+                                    const float* __restrict__ adv, const float* __restrict__ v_new,
+                                    const float* __restrict__ v_old, const float* __restrict__ ret,
+                                    const float* __restrict__ ent, float* __restrict__ out, long N,
+                                    float clip, bool clip_vloss, float scale) {
+  __shared__ float lds[32];
+  float pg = 0.f, vl = 0.f, el = 0.f;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < N;
+       i += (long)gridDim.x * blockDim.x) {
+    const float r = __expf(lp_new[i] - lp_old[i]);
+    const float a = adv[i];
+    const float rc = fminf(fmaxf(r, 1.f - clip), 1.f + clip);
+    pg += fmaxf(-a * r, -a * rc);
+    const float dv = v_new[i] - ret[i];
+    if (clip_vloss) {
+      const float vc = v_old[i] + fminf(fmaxf(v_new[i] - v_old[i], -clip), clip) - ret[i];
+      vl += 0.5f * fmaxf(dv * dv, vc * vc);
+    } else {
+      vl += dv * dv;
+    }
+    el += -ent[i];
+  }
+  pg = wave_sum(pg);
+  vl = wave_sum(vl);
+  el = wave_sum(el);
+  const int lane = threadIdx.x & 63, wid = threadIdx.x >> 6;
+  if (lane == 0) {
+    lds[wid] = pg;
+    lds[8 + wid] = vl;
+    lds[16 + wid] = el;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float p = 0.f, v = 0.f, e = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) {
+      p += lds[w];
+      v += lds[8 + w];
+      e += lds[16 + w];
+    }
+    atomicAdd(&out[0], p * scale);
+    atomicAdd(&out[1], v * scale);
+    atomicAdd(&out[2], e * scale);
+  }
+}
+
+__global__ void ppo_loss_bwd_kernel(const float* __restrict__ g3, const float* __restrict__ lp_new,
+                                    const float* __restrict__ lp_old, const float* __restrict__ adv,
+                                    const float* __restrict__ v_new, const float* __restrict__ v_old,
+                                    const float* __restrict__ ret, float* __restrict__ dlp,
+                                    float* __restrict__ dv_out, float* __restrict__ dent, long N,
+                                    float clip, bool clip_vloss, float scale) {
+  // g3 = {g_pg, g_v, g_ent} already weighted by the caller's combine coefs
+  const float gp = g3[0] * scale, gv = g3[1] * scale, ge = g3[2] * scale;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < N;
+       i += (long)gridDim.x * blockDim.x) {
+    const float r = __expf(lp_new[i] - lp_old[i]);
+    const float a = adv[i];
+    const float rc = fminf(fmaxf(r, 1.f - clip), 1.f + clip);
+    const float inside = (r >= 1.f - clip && r <= 1.f + clip) ? 1.f : 0.f;
+    const float l1 = -a * r, l2 = -a * rc;
+    float dr;
+    if (l1 > l2) {
+      dr = -a;
+    } else if (l2 > l1) {
+      dr = -a * inside;
+    } else {
+      dr = -a * 0.5f * (1.f + inside);
+    }
+    dlp[i] = gp * dr * r;  // dr/dlp_new = r
+    const float d = v_new[i] - ret[i];
+    if (clip_vloss) {
+      const float delta = v_new[i] - v_old[i];
+      const float din = (delta >= -clip && delta <= clip) ? 1.f : 0.f;
+      const float vc = v_old[i] + fminf(fmaxf(delta, -clip), clip) - ret[i];
+      const float u = d * d, c2 = vc * vc;
+      float g;
+      if (u > c2) {
+        g = 2.f * d;
+      } else if (c2 > u) {
+        g = 2.f * vc * din;
+      } else {
+        g = 0.5f * (2.f * d + 2.f * vc * din);
+      }
+      dv_out[i] = gv * 0.5f * g;
+    } else {
+      dv_out[i] = gv * 2.f * d;
+    }
+    dent[i] = -ge;
+  }
+}
+
+std::vector<torch::Tensor> ppo_loss_fwd(const torch::Tensor& lp_new, const torch::Tensor& lp_old,
+                                        const torch::Tensor& adv, const torch::Tensor& v_new,
+                                        const torch::Tensor& v_old, const torch::Tensor& ret,
+                                        const torch::Tensor& ent, double clip, bool clip_vloss,
+                                        bool mean) {
+  CHECK_IN(lp_new);
+  const long N = lp_new.numel();
+  TORCH_CHECK(lp_old.numel() == N && adv.numel() == N && v_new.numel() == N && v_old.numel() == N &&
+              ret.numel() == N && ent.numel() == N);
+  auto out = torch::zeros({3}, lp_new.options());
+  const float scale = mean ? 1.f / (float)N : 1.f;
+  const int blocks = (int)std::min<long>((N + kBlock - 1) / kBlock, 2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(ppo_loss_fwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     lp_new.data_ptr<float>(), lp_old.data_ptr<float>(), adv.data_ptr<float>(),
+                     v_new.data_ptr<float>(), v_old.data_ptr<float>(), ret.data_ptr<float>(),
+                     ent.data_ptr<float>(), out.data_ptr<float>(), N, (float)clip, clip_vloss, scale);
+  return {out};
+}
+
+std::vector<torch::Tensor> ppo_loss_bwd(const torch::Tensor& g3, const torch::Tensor& lp_new,
+                                        const torch::Tensor& lp_old, const torch::Tensor& adv,
+                                        const torch::Tensor& v_new, const torch::Tensor& v_old,
+                                        const torch::Tensor& ret, double clip, bool clip_vloss,
+                                        bool mean) {
+  const long N = lp_new.numel();
+  auto dlp = torch::empty_like(lp_new);
+  auto dv = torch::empty_like(v_new);
+  auto dent = torch::empty_like(lp_new);
+  const float scale = mean ? 1.f / (float)N : 1.f;
+  const int blocks = (int)std::min<long>((N + kBlock - 1) / kBlock, 2048);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(ppo_loss_bwd_kernel, dim3(blocks), dim3(kBlock), 0, stream.stream(),
+                     g3.data_ptr<float>(), lp_new.data_ptr<float>(), lp_old.data_ptr<float>(),
+                     adv.data_ptr<float>(), v_new.data_ptr<float>(), v_old.data_ptr<float>(),
+                     ret.data_ptr<float>(), dlp.data_ptr<float>(), dv.data_ptr<float>(),
+                     dent.data_ptr<float>(), N, (float)clip, clip_vloss, scale);
+  return {dlp, dv, dent};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("symlog_fwd", [](const torch::Tensor& x) { return symmath<0>(x, c10::nullopt); });
   m.def("symlog_bwd", [](const torch::Tensor& x, const torch::Tensor& g) { return symmath<1>(x, g); });
@@ -5830,6 +5970,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moments_update", &moments_update);
   m.def("tanh_normal_fwd", &tanh_normal_fwd);
   m.def("tanh_normal_bwd", &tanh_normal_bwd);
+  m.def("ppo_loss_fwd", &ppo_loss_fwd);
+  m.def("ppo_loss_bwd", &ppo_loss_bwd);
   m.def("vloss2_fwd", &vloss2_fwd);
   m.def("vloss2_bwd", &vloss2_bwd);
   m.def("replay_gather", &replay_gather);

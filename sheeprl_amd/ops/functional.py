@@ -387,3 +387,49 @@ def tanh_normal_sample(mean: Tensor, logstd: Tensor, eps: Tensor, scale: Tensor,
                        bias: Tensor, like: Tensor, lmin: float = -5.0, lmax: float = 2.0):
     """Returns (action in env range, per-row summed log-prob [..., 1])."""
     return _TanhNormalSample.apply(mean, logstd, eps, scale, bias, like, lmin, lmax)
+
+
+class _PPOLosses(torch.autograd.Function):
+    """Fused PPO clip losses (parity:
+    sheeprl/algos/ppo/loss.py:6-65): one reduction kernel returns the
+    [policy, value, entropy] loss triple; backward recomputes the branch
+    selections (incl. torch.maximum's 0.5/0.5 tie split — the unclipped
+    region ties the two policy branches, so this is the common case)."""
+
+    @staticmethod
+    def forward(ctx, lp_new: Tensor, lp_old: Tensor, adv: Tensor, v_new: Tensor,
+                v_old: Tensor, ret: Tensor, ent: Tensor, clip: float,
+                clip_vloss: bool, mean: bool) -> Tensor:
+        args = [t.detach().float().reshape(-1).contiguous()
+                for t in (lp_new, lp_old, adv, v_new, v_old, ret)]
+        entc = ent.detach().float().reshape(-1).contiguous()
+        (out3,) = require_ext().ppo_loss_fwd(*args, entc, clip, clip_vloss, mean)
+        ctx.save_for_backward(*args)
+        ctx.meta = (clip, clip_vloss, mean,
+                    (lp_new.dtype, lp_new.shape), (v_new.dtype, v_new.shape),
+                    (ent.dtype, ent.shape))
+        return out3
+
+    @staticmethod
+    def backward(ctx, g3: Tensor):
+        clip, clip_vloss, mean, lp_m, v_m, e_m = ctx.meta
+        dlp, dv, dent = require_ext().ppo_loss_bwd(
+            g3.contiguous().float(), *ctx.saved_tensors, clip, clip_vloss, mean
+        )
+        return (
+            dlp.view(lp_m[1]).to(lp_m[0]), None, None,
+            dv.view(v_m[1]).to(v_m[0]), None, None,
+            dent.view(e_m[1]).to(e_m[0]), None, None, None,
+        )
+
+
+def ppo_losses(new_logprobs: Tensor, old_logprobs: Tensor, advantages: Tensor,
+               new_values: Tensor, old_values: Tensor, returns: Tensor, entropy: Tensor,
+               clip_coef: float, clip_vloss: bool, reduction: str = "mean"):
+    """Returns the (policy_loss, value_loss, entropy_loss) triple, fused on
+    GPU for reduction 'mean'/'sum'.  Callers fall back to the eager
+    composition off-GPU (sheeprl_amd/algos/ppo/loss.py)."""
+    out3 = _PPOLosses.apply(new_logprobs, old_logprobs, advantages, new_values,
+                            old_values, returns, entropy, clip_coef, clip_vloss,
+                            reduction == "mean")
+    return out3[0], out3[1], out3[2]

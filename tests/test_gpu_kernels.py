@@ -1059,3 +1059,41 @@ def test_sac_actor_fused_forward_bf16():
     assert logp.dtype == torch.float32 and logp.shape == (64, 1)
     assert action.abs().max() <= 2.0 + 1e-2
     assert torch.isfinite(logp).all()
+
+
+@requires_gpu
+@pytest.mark.parametrize("clip_vloss", [False, True])
+@pytest.mark.parametrize("reduction", ["mean", "sum"])
+def test_ppo_losses_fused_matches_eager(clip_vloss, reduction):
+    """Fused PPO loss triple (one kernel each way) vs the eager composition,
+    values AND gradients — including torch.maximum tie semantics (the
+    unclipped region ties the policy branches)."""
+    from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, value_loss
+
+    torch.manual_seed(3)
+    N = 1024
+    lp_new = (torch.randn(N, device="cuda") * 0.2).requires_grad_(True)
+    lp_old = torch.randn(N, device="cuda") * 0.2
+    adv = torch.randn(N, device="cuda")
+    v_new = torch.randn(N, 1, device="cuda").requires_grad_(True)
+    v_old = torch.randn(N, 1, device="cuda")
+    ret = torch.randn(N, 1, device="cuda")
+    ent = torch.randn(N, device="cuda").requires_grad_(True)
+    clip = 0.2
+
+    lp2 = lp_new.detach().clone().requires_grad_(True)
+    v2 = v_new.detach().clone().requires_grad_(True)
+    e2 = ent.detach().clone().requires_grad_(True)
+    pg_ref = policy_loss(lp2, lp_old, adv, clip, reduction)
+    vl_ref = value_loss(v2, v_old, ret, clip, clip_vloss, reduction)
+    el_ref = entropy_loss(e2, reduction)
+    (pg_ref + 0.5 * vl_ref + 0.01 * el_ref).backward()
+
+    pg, vl, el = ops.ppo_losses(lp_new, lp_old, adv, v_new, v_old, ret, ent, clip, clip_vloss, reduction)
+    torch.testing.assert_close(pg, pg_ref, atol=1e-4, rtol=1e-4)
+    torch.testing.assert_close(vl, vl_ref, atol=1e-4, rtol=1e-4)
+    torch.testing.assert_close(el, el_ref, atol=1e-4, rtol=1e-4)
+    (pg + 0.5 * vl + 0.01 * el).backward()
+    torch.testing.assert_close(lp_new.grad, lp2.grad, atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(v_new.grad, v2.grad, atol=1e-5, rtol=1e-4)
+    torch.testing.assert_close(ent.grad, e2.grad, atol=1e-6, rtol=1e-5)
