@@ -40,7 +40,8 @@ def main() -> None:
     cnt = collections.Counter()
     for s, e, n, grid, wg in rows:
         if t0 < s <= t1:
-            key = (n.split("(")[0][:70], grid, wg)
+            nn = n.replace("(anonymous namespace)::", "").replace("void ", "")
+            key = (nn.split("(")[0][:70], grid, wg)
             agg[key] += e - s
             cnt[key] += 1
     tot = sum(agg.values())
