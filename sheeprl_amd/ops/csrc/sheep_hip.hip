@@ -141,7 +141,7 @@ torch::Tensor symmath(const torch::Tensor& x, const c10::optional<torch::Tensor>
 template <typename T, typename TW, bool SILU>
 __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict__ w,
                                   const TW* __restrict__ b, T* __restrict__ y, float* __restrict__ mean_out,
-                                  float* __restrict__ rstd_out, int D, float eps, long ys) {
+                                  float* __restrict__ rstd_out, int D, float eps, long ys, bool cached) {
   __shared__ float lds[18];
   const long row = blockIdx.x;
   const T* xr = x + row * (long)D;
@@ -149,8 +149,11 @@ __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict_
   // register-cached single-read path: with the scan's tiny row counts the
   // kernel is latency-bound, so one global pass + one fused sum/sumsq
   // reduction (instead of mean pass, barrier, var pass, barrier) is ~1/3
-  // fewer round trips; 16 floats/thread covers D <= 4096 at 256 threads
-  if (D <= 16 * (int)blockDim.x) {
+  // fewer round trips; 16 floats/thread covers D <= 4096 at 256 threads.
+  // The cache array lives in scratch (runtime-bounded indexing), which is
+  // invisible at <=64 rows but dominates at behaviour-batch row counts —
+  // the host gates `cached` by N (measured: 16384x1024 fwd 48.9 us cached).
+  if (cached && D <= 16 * (int)blockDim.x) {
     float cache[16];
     float s = 0.f, s2 = 0.f;
     int cnt = 0;
@@ -257,12 +260,14 @@ __global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restri
                                      float* __restrict__ mean_out, float* __restrict__ rstd_out, long R, int D,
                                      float eps) {
   constexpr int V = 16 / sizeof(T);
-  constexpr int RPW = 64 / L;
+  constexpr int RPW = 64 / L;  // floor for non-pow2 L (12/24: conv widths
+                               // 96/192 bf16); lanes past RPW*L idle
   constexpr int UNROLL = 4;  // row-blocks in flight per wave: one 16-B load
                              // per lane is latency-bound at ~3 TB/s
   const int lane = threadIdx.x & 63;
   const int lig = lane % L;
   const int grp = lane / L;
+  const bool lane_ok = grp < RPW;
   const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
   float wv[V], bv[V];
@@ -278,12 +283,12 @@ __global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restri
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
       rows[u] = rb + u * stride + grp;
-      if (rows[u] < R) xv[u].u = *reinterpret_cast<const uint4*>(x + rows[u] * (long)D + lig * V);
+      if (lane_ok && rows[u] < R) xv[u].u = *reinterpret_cast<const uint4*>(x + rows[u] * (long)D + lig * V);
     }
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
       const long row = rows[u];
-      const bool active = row < R;
+      const bool active = lane_ok && row < R;
       float s = 0.f, s2 = 0.f;
       if (active) {
 #pragma unroll
@@ -293,10 +298,23 @@ __global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restri
           s2 += v * v;
         }
       }
+      if constexpr ((L & (L - 1)) == 0) {
 #pragma unroll
-      for (int off = 1; off < L; off <<= 1) {
-        s += __shfl_xor(s, off, 64);
-        s2 += __shfl_xor(s2, off, 64);
+        for (int off = 1; off < L; off <<= 1) {
+          s += __shfl_xor(s, off, 64);
+          s2 += __shfl_xor(s2, off, 64);
+        }
+      } else {
+        // non-pow2 group: gather-loop sum over the group's L lanes
+        const int base = grp * L;
+        float a = s, a2 = s2;
+        s = 0.f;
+        s2 = 0.f;
+#pragma unroll
+        for (int i = 0; i < L; ++i) {
+          s += __shfl(a, (base + i) & 63, 64);
+          s2 += __shfl(a2, (base + i) & 63, 64);
+        }
       }
       if (active) {
         float mean = s / D;
@@ -326,7 +344,7 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
                                      T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb, long R,
                                      int D) {
   constexpr int V = 16 / sizeof(T);
-  constexpr int RPW = 64 / L;
+  constexpr int RPW = 64 / L;  // floor for non-pow2 L; lanes past RPW*L idle
   extern __shared__ __attribute__((aligned(16))) float smem[];  // [2*D]
   float* gw_s = smem;
   float* gb_s = smem + D;
@@ -335,6 +353,7 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
   const int lane = threadIdx.x & 63;
   const int lig = lane % L;
   const int grp = lane / L;
+  const bool lane_ok = grp < RPW;
   const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
   float wv[V], bv[V], gwa[V], gba[V];
@@ -353,7 +372,7 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
       rows[u] = rb + u * stride + grp;
-      if (rows[u] < R) {
+      if (lane_ok && rows[u] < R) {
         xv[u].u = *reinterpret_cast<const uint4*>(x + rows[u] * (long)D + lig * V);
         gv[u].u = *reinterpret_cast<const uint4*>(gy + rows[u] * (long)D + lig * V);
       }
@@ -361,7 +380,7 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
 #pragma unroll
     for (int u = 0; u < UNROLL; ++u) {
       const long row = rows[u];
-      const bool active = row < R;
+      const bool active = lane_ok && row < R;
       float xh[V], gz[V];
       float s1 = 0.f, s2 = 0.f;
       float rs = 0.f;
@@ -385,10 +404,22 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
           s2 += gxhat * xh[e];
         }
       }
+      if constexpr ((L & (L - 1)) == 0) {
 #pragma unroll
-      for (int off = 1; off < L; off <<= 1) {
-        s1 += __shfl_xor(s1, off, 64);
-        s2 += __shfl_xor(s2, off, 64);
+        for (int off = 1; off < L; off <<= 1) {
+          s1 += __shfl_xor(s1, off, 64);
+          s2 += __shfl_xor(s2, off, 64);
+        }
+      } else {
+        const int base = grp * L;
+        float a1 = s1, a2 = s2;
+        s1 = 0.f;
+        s2 = 0.f;
+#pragma unroll
+        for (int i = 0; i < L; ++i) {
+          s1 += __shfl(a1, (base + i) & 63, 64);
+          s2 += __shfl(a2, (base + i) & 63, 64);
+        }
       }
       if (active) {
         const float S1 = s1 / D, S2 = s2 / D;
@@ -400,12 +431,26 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
     }
   }
   // fold groups within the wave (lanes sharing lane%L hold the same columns)
+  if constexpr ((L & (L - 1)) == 0) {
 #pragma unroll
-  for (int off = L; off < 64; off <<= 1) {
+    for (int off = L; off < 64; off <<= 1) {
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        gwa[e] += __shfl_xor(gwa[e], off, 64);
+        gba[e] += __shfl_xor(gba[e], off, 64);
+      }
+    }
+  } else {
 #pragma unroll
     for (int e = 0; e < V; ++e) {
-      gwa[e] += __shfl_xor(gwa[e], off, 64);
-      gba[e] += __shfl_xor(gba[e], off, 64);
+      float aw = gwa[e], ab = gba[e];
+      gwa[e] = 0.f;
+      gba[e] = 0.f;
+#pragma unroll
+      for (int g = 0; g < RPW; ++g) {
+        gwa[e] += __shfl(aw, (lig + g * L) & 63, 64);
+        gba[e] += __shfl(ab, (lig + g * L) & 63, 64);
+      }
     }
   }
   if (grp == 0) {
@@ -428,8 +473,12 @@ int ln_cl_lanes(int D, long stride) {
   constexpr int V = 16 / sizeof(T);
   if (stride != D || D % V != 0) return 0;
   int L = D / V;
-  if (L < 2 || L > 64 || (L & (L - 1)) != 0) return 0;
-  return L;
+  if (L < 2 || L > 64) return 0;
+  if ((L & (L - 1)) == 0) return L;
+  // non-pow2 lane groups supported for the conv channel widths (96/192
+  // bf16 -> L=12/24): floor(64/L) rows per wave, tail lanes idle
+  if (L == 12 || L == 24) return L;
+  return 0;
 }
 
 template <typename T, typename TW, bool SILU>
@@ -446,7 +495,9 @@ void launch_ln_fwd_cl(int L, long R, hipStream_t st, const T* x, const TW* w, co
     SHEEP_LN_FWD_CASE(2)
     SHEEP_LN_FWD_CASE(4)
     SHEEP_LN_FWD_CASE(8)
+    SHEEP_LN_FWD_CASE(12)
     SHEEP_LN_FWD_CASE(16)
+    SHEEP_LN_FWD_CASE(24)
     SHEEP_LN_FWD_CASE(32)
     SHEEP_LN_FWD_CASE(64)
   }
@@ -476,7 +527,9 @@ void launch_ln_bwd_cl(int L, long R, hipStream_t st, const T* gy, const T* x, co
     SHEEP_LN_BWD_CASE(2)
     SHEEP_LN_BWD_CASE(4)
     SHEEP_LN_BWD_CASE(8)
+    SHEEP_LN_BWD_CASE(12)
     SHEEP_LN_BWD_CASE(16)
+    SHEEP_LN_BWD_CASE(24)
     SHEEP_LN_BWD_CASE(32)
     SHEEP_LN_BWD_CASE(64)
   }
@@ -520,11 +573,13 @@ void ln_act_fwd_core(const torch::Tensor& x, const torch::Tensor& w, const torch
       } else if (silu)
         hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, true>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                            (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
-                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps, ys);
+                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps, ys,
+                           N <= 64);
       else
         hipLaunchKernelGGL((ln_act_fwd_kernel<T, TW, false>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
                            (const T*)x.data_ptr(), (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
-                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps, ys);
+                           (T*)y.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps, ys,
+                           N <= 64);
     });
   });
 }
@@ -561,7 +616,7 @@ template <typename T, typename TW, bool SILU>
 __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict__ x, const TW* __restrict__ w,
                                   const TW* __restrict__ b, const float* __restrict__ mean,
                                   const float* __restrict__ rstd, T* __restrict__ gx, float* __restrict__ gw,
-                                  float* __restrict__ gb, long N, int D, long gys) {
+                                  float* __restrict__ gb, long N, int D, long gys, bool cached) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* lds = smem;          // 9 floats for block_sum
   float* gw_acc = smem + 32;  // [D] (block_sum2 scratch precedes)
@@ -574,7 +629,9 @@ __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict_
   // register-cached single-read path: xhat and post-SILU gz persist across
   // the reduction (no second read of x/gy, no SILU recompute); S1/S2 reduce
   // in one fused barrier round.  8 floats each covers D <= 2048 at 256 thr.
-  if (D <= 8 * (int)blockDim.x) {
+  // Scratch-backed caches (see ln_act_fwd_kernel): host gates `cached` by N
+  // (measured: the behaviour MLP bwd at 16384 rows ran 93.9 us cached).
+  if (cached && D <= 8 * (int)blockDim.x) {
     float cxh[8], cgz[8];
     for (long row = blockIdx.x; row < N; row += gridDim.x) {
       const T* xr = x + row * (long)D;
@@ -791,7 +848,7 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
   }
   size_t shmem = (32 + 2 * (size_t)D) * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "ln_act_bwd: D too large for LDS accumulation");
-  int blocks = (int)std::min(N, (long)512);
+  int blocks = (int)std::min(N, (long)2048);
   auto stream = at::cuda::getCurrentCUDAStream();
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_bwd", [&] {
     using T = scalar_t;
@@ -801,12 +858,12 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
         hipLaunchKernelGGL((ln_act_bwd_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
                            (const T*)gy.data_ptr(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                            (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
+                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, N <= 64);
       else
         hipLaunchKernelGGL((ln_act_bwd_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
                            (const T*)gy.data_ptr(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                            (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
+                           (T*)gx.data_ptr(), gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, N <= 64);
     });
   });
 }
@@ -844,7 +901,8 @@ template <typename T, typename TW>
 __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restrict__ h, const TW* __restrict__ w,
                                      const TW* __restrict__ b, T* __restrict__ hout, float* __restrict__ mean_out,
                                      float* __restrict__ rstd_out, int H, float eps, long hs,
-                                     T* __restrict__ hout2, long h2s, T* __restrict__ hout3, long h3s) {
+                                     T* __restrict__ hout2, long h2s, T* __restrict__ hout3, long h3s,
+                                     bool cached) {
   __shared__ float lds[18];
   const long row = blockIdx.x;
   const int D = 3 * H;
@@ -856,7 +914,7 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
   // single-read register-cached path (latency-bound at scan batch sizes):
   // cache element j at index (j - threadIdx.x)/blockDim, so the gates pass
   // finds positions j / H+j / 2H+j when H is a blockDim multiple
-  if (D <= 16 * (int)blockDim.x && (H % (int)blockDim.x) == 0) {
+  if (cached && D <= 16 * (int)blockDim.x && (H % (int)blockDim.x) == 0) {
     float cache[16];
     float s = 0.f, s2 = 0.f;
     int cnt = 0;
@@ -919,6 +977,56 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
   }
 }
 
+// Column-chunked ("wide") variant for few rows x large H (the XL scan:
+// B=16, H=4096 put only 16 workgroups on 256 CUs — measured 30-60 us/call
+// while doing ~1 MB of traffic).  Grid (N, C): each block redundantly
+// streams the full row for the LN statistics (reads are L2/MALL hits across
+// the row's blocks) and then computes/writes only its H-chunk — same launch
+// count, ~C x the parallelism.
+template <typename T, typename TW>
+__global__ void gru_gates_fwd_wide_kernel(const T* __restrict__ y, const T* __restrict__ h,
+                                          const TW* __restrict__ w, const TW* __restrict__ b,
+                                          T* __restrict__ hout, float* __restrict__ mean_out,
+                                          float* __restrict__ rstd_out, int H, float eps, long hs,
+                                          T* __restrict__ hout2, long h2s, T* __restrict__ hout3, long h3s) {
+  __shared__ float lds[18];
+  const long row = blockIdx.x;
+  const int D = 3 * H;
+  const T* yr = y + row * (long)D;
+  const T* hr = h + row * hs;
+  T* outr = hout + row * (long)H;
+  T* outr2 = hout2 ? hout2 + row * h2s : nullptr;
+  T* outr3 = hout3 ? hout3 + row * h3s : nullptr;
+  float s = 0.f, s2 = 0.f;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    float v = ld(yr, j);
+    s += v;
+    s2 += v * v;
+  }
+  block_sum2(s, s2, lds);
+  const float mean = s / D;
+  const float var = s2 / D - mean * mean;
+  const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+  if (blockIdx.y == 0 && threadIdx.x == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  const int W = (H + (int)gridDim.y - 1) / (int)gridDim.y;
+  const int j0 = (int)blockIdx.y * W, j1 = min(j0 + W, H);
+  for (int j = j0 + (int)threadIdx.x; j < j1; j += blockDim.x) {
+    float zr = ((ld(yr, j) - mean) * rstd) * ld(w, j) + ld(b, j);
+    float zc = ((ld(yr, H + j) - mean) * rstd) * ld(w, H + j) + ld(b, H + j);
+    float zu = ((ld(yr, 2 * H + j) - mean) * rstd) * ld(w, 2 * H + j) + ld(b, 2 * H + j);
+    float r = 1.f / (1.f + expf(-zr));
+    float c = tanhf(r * zc);
+    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    float hv = u * c + (1.f - u) * ld(hr, j);
+    st(outr, j, hv);
+    if (outr2) st(outr2, j, hv);
+    if (outr3) st(outr3, j, hv);
+  }
+}
+
 void gru_gates_fwd_core(const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
                         const torch::Tensor& b, double eps, torch::Tensor& hout, torch::Tensor& mean,
                         torch::Tensor& rstd, long hs, void* hout2, long h2s, void* hout3 = nullptr,
@@ -930,14 +1038,26 @@ void gru_gates_fwd_core(const torch::Tensor& y, const torch::Tensor& h, const to
   auto wc = w.contiguous();
   auto bc = b.contiguous();
   auto stream = at::cuda::getCurrentCUDAStream();
+  // wide dispatch: few rows x large H leave the chip idle on the row-per-
+  // block kernel; chunk columns so N*C workgroups >= ~384
+  const bool wide = N <= 64 && H >= 2048;
+  const int C = wide ? std::min<int>(std::max<int>(1, 384 / (int)N), (H + 255) / 256) : 1;
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_fwd", [&] {
     using T = scalar_t;
     AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "gru_gates_fwd_w", [&] {
       using TW = scalar_t;
-      hipLaunchKernelGGL((gru_gates_fwd_kernel<T, TW>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
-                         (const T*)y.data_ptr(), (const T*)h.data_ptr(), (const TW*)wc.data_ptr(),
-                         (const TW*)bc.data_ptr(), (T*)hout.data_ptr(), mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), H, (float)eps, hs, (T*)hout2, h2s, (T*)hout3, h3s);
+      if (wide)
+        hipLaunchKernelGGL((gru_gates_fwd_wide_kernel<T, TW>), dim3((int)N, C), dim3(kBlock), 0,
+                           stream.stream(), (const T*)y.data_ptr(), (const T*)h.data_ptr(),
+                           (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), (T*)hout.data_ptr(),
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(), H, (float)eps, hs, (T*)hout2,
+                           h2s, (T*)hout3, h3s);
+      else
+        hipLaunchKernelGGL((gru_gates_fwd_kernel<T, TW>), dim3((int)N), dim3(kBlock), 0, stream.stream(),
+                           (const T*)y.data_ptr(), (const T*)h.data_ptr(), (const TW*)wc.data_ptr(),
+                           (const TW*)bc.data_ptr(), (T*)hout.data_ptr(), mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), H, (float)eps, hs, (T*)hout2, h2s, (T*)hout3, h3s,
+                           N <= 64);
     });
   });
 }
@@ -979,7 +1099,7 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
                                      const T* __restrict__ h, const TW* __restrict__ w, const TW* __restrict__ b,
                                      const float* __restrict__ mean, const float* __restrict__ rstd,
                                      T* __restrict__ gy, T* __restrict__ ghprev, float* __restrict__ gw,
-                                     float* __restrict__ gb, long N, int H, long hs) {
+                                     float* __restrict__ gb, long N, int H, long hs, bool cached) {
   extern __shared__ __attribute__((aligned(16))) float smem[];
   float* lds = smem;
   float* gw_acc = smem + 32;       // [3H] (block_sum2 scratch precedes)
@@ -995,7 +1115,7 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
   // written once with its final value (the generic path stores gz into gy
   // and re-reads it); S1/S2 reduce together in one barrier round
   const int hstep = H / (int)blockDim.x;
-  if ((H % (int)blockDim.x) == 0 && hstep <= 4) {
+  if (cached && (H % (int)blockDim.x) == 0 && hstep <= 4) {
     float cxh[12], cgz[12];
     for (long row = blockIdx.x; row < N; row += gridDim.x) {
       const T* yr = y + row * (long)D;
@@ -1126,6 +1246,107 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
   }
 }
 
+// Column-chunked backward (see gru_gates_fwd_wide_kernel): each block
+// recomputes the gate math over the FULL row to form the LN-backward sums
+// S1/S2 (redundant across the row's C blocks; all reads are row-local and
+// L2-shared), then writes gy/ghprev and accumulates gw/gb only for its
+// H-chunk.  gw/gb LDS images cover the chunk, flushed with one atomicAdd
+// per column per block.
+template <typename T, typename TW>
+__global__ void gru_gates_bwd_wide_kernel(const T* __restrict__ gh, const T* __restrict__ gh2,
+                                          const T* __restrict__ gh3, long gh3s, const T* __restrict__ y,
+                                          const T* __restrict__ h, const TW* __restrict__ w,
+                                          const TW* __restrict__ b, const float* __restrict__ mean,
+                                          const float* __restrict__ rstd, T* __restrict__ gy,
+                                          T* __restrict__ ghprev, float* __restrict__ gw,
+                                          float* __restrict__ gb, long N, int H, long hs) {
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* lds = smem;
+  const int W = (H + (int)gridDim.y - 1) / (int)gridDim.y;
+  float* gw_acc = smem + 32;           // [3W]
+  float* gb_acc = gw_acc + 3 * W;      // [3W]
+  const int D = 3 * H;
+  const long row = blockIdx.x;
+  const int j0 = (int)blockIdx.y * W, j1 = min(j0 + W, H);
+  for (int j = threadIdx.x; j < 3 * W; j += blockDim.x) {
+    gw_acc[j] = 0.f;
+    gb_acc[j] = 0.f;
+  }
+  __syncthreads();
+  const T* yr = y + row * (long)D;
+  const T* hr = h + row * hs;
+  const T* ghr = gh + row * (long)H;
+  const T* gh2r = gh2 ? gh2 + row * (long)H : nullptr;
+  const T* gh3r = gh3 ? gh3 + row * gh3s : nullptr;
+  T* gyr = gy + row * (long)D;
+  T* ghp = ghprev + row * (long)H;
+  const float m = mean[row], rs = rstd[row];
+  float s1 = 0.f, s2 = 0.f;
+  for (int j = threadIdx.x; j < H; j += blockDim.x) {
+    float xh_r = (ld(yr, j) - m) * rs;
+    float xh_c = (ld(yr, H + j) - m) * rs;
+    float xh_u = (ld(yr, 2 * H + j) - m) * rs;
+    float zr = xh_r * ld(w, j) + ld(b, j);
+    float zc = xh_c * ld(w, H + j) + ld(b, H + j);
+    float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
+    float r = 1.f / (1.f + expf(-zr));
+    float c = tanhf(r * zc);
+    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
+    float gu = g * (c - ld(hr, j));
+    float gc = g * u;
+    float gzu = gu * u * (1.f - u);
+    float grc = gc * (1.f - c * c);
+    float gzc = grc * r;
+    float gzr = grc * zc * r * (1.f - r);
+    float gxh_r = gzr * ld(w, j);
+    float gxh_c = gzc * ld(w, H + j);
+    float gxh_u = gzu * ld(w, 2 * H + j);
+    s1 += gxh_r + gxh_c + gxh_u;
+    s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
+  }
+  block_sum2(s1, s2, lds);
+  const float S1 = s1 / D, S2 = s2 / D;
+  for (int j = j0 + (int)threadIdx.x; j < j1; j += blockDim.x) {
+    float xh_r = (ld(yr, j) - m) * rs;
+    float xh_c = (ld(yr, H + j) - m) * rs;
+    float xh_u = (ld(yr, 2 * H + j) - m) * rs;
+    float zr = xh_r * ld(w, j) + ld(b, j);
+    float zc = xh_c * ld(w, H + j) + ld(b, H + j);
+    float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
+    float r = 1.f / (1.f + expf(-zr));
+    float c = tanhf(r * zc);
+    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
+    float gu = g * (c - ld(hr, j));
+    float gc = g * u;
+    float gzu = gu * u * (1.f - u);
+    float grc = gc * (1.f - c * c);
+    float gzc = grc * r;
+    float gzr = grc * zc * r * (1.f - r);
+    st(ghp, j, g * (1.f - u));
+    const int jc = j - j0;
+    gw_acc[jc] += gzr * xh_r;
+    gb_acc[jc] += gzr;
+    gw_acc[W + jc] += gzc * xh_c;
+    gb_acc[W + jc] += gzc;
+    gw_acc[2 * W + jc] += gzu * xh_u;
+    gb_acc[2 * W + jc] += gzu;
+    st(gyr, j, (gzr * ld(w, j) - S1 - xh_r * S2) * rs);
+    st(gyr, H + j, (gzc * ld(w, H + j) - S1 - xh_c * S2) * rs);
+    st(gyr, 2 * H + j, (gzu * ld(w, 2 * H + j) - S1 - xh_u * S2) * rs);
+  }
+  __syncthreads();
+  for (int jc = threadIdx.x; jc < j1 - j0; jc += blockDim.x) {
+    atomicAdd(&gw[j0 + jc], gw_acc[jc]);
+    atomicAdd(&gb[j0 + jc], gb_acc[jc]);
+    atomicAdd(&gw[H + j0 + jc], gw_acc[W + jc]);
+    atomicAdd(&gb[H + j0 + jc], gb_acc[W + jc]);
+    atomicAdd(&gw[2 * H + j0 + jc], gw_acc[2 * W + jc]);
+    atomicAdd(&gb[2 * H + j0 + jc], gb_acc[2 * W + jc]);
+  }
+}
+
 void gru_gates_bwd_core(const torch::Tensor& gh, const void* gh2, const void* gh3, long gh3s,
                         const torch::Tensor& y, const torch::Tensor& h, const torch::Tensor& w,
                         const torch::Tensor& b, const torch::Tensor& mean, const torch::Tensor& rstd,
@@ -1137,20 +1358,33 @@ void gru_gates_bwd_core(const torch::Tensor& gh, const void* gh2, const void* gh
   int H = (int)(y.size(1) / 3);
   auto wc = w.contiguous();
   auto bc = b.contiguous();
-  size_t shmem = (32 + 6 * (size_t)H) * sizeof(float);
-  TORCH_CHECK(shmem <= 160 * 1024, "gru_gates_bwd: H too large for LDS accumulation");
-  int blocks = (int)std::min(N, (long)512);
   auto stream = at::cuda::getCurrentCUDAStream();
+  const bool wide = N <= 64 && H >= 2048;
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_bwd", [&] {
     using T = scalar_t;
     AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "gru_gates_bwd_w", [&] {
       using TW = scalar_t;
+      if (wide) {
+        const int C = std::min<int>(std::max<int>(1, 384 / (int)N), (H + 255) / 256);
+        const int W = (H + C - 1) / C;
+        size_t shmem = (32 + 6 * (size_t)W) * sizeof(float);
+        hipLaunchKernelGGL((gru_gates_bwd_wide_kernel<T, TW>), dim3((int)N, C), dim3(kBlock), shmem,
+                           stream.stream(), (const T*)gh.data_ptr(), (const T*)gh2, (const T*)gh3, gh3s,
+                           (const T*)y.data_ptr(), (const T*)h.data_ptr(), (const TW*)wc.data_ptr(),
+                           (const TW*)bc.data_ptr(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                           (T*)gy.data_ptr(), (T*)ghprev.data_ptr(), gw.data_ptr<float>(),
+                           gb.data_ptr<float>(), N, H, hs);
+        return;
+      }
+      size_t shmem = (32 + 6 * (size_t)H) * sizeof(float);
+      TORCH_CHECK(shmem <= 160 * 1024, "gru_gates_bwd: H too large for LDS accumulation");
+      int blocks = (int)std::min(N, (long)512);
       hipLaunchKernelGGL((gru_gates_bwd_kernel<T, TW>), dim3(blocks), dim3(kBlock), shmem, stream.stream(),
                          (const T*)gh.data_ptr(), (const T*)gh2, (const T*)gh3, gh3s,
                          (const T*)y.data_ptr(), (const T*)h.data_ptr(),
                          (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
                          rstd.data_ptr<float>(), (T*)gy.data_ptr(), (T*)ghprev.data_ptr(), gw.data_ptr<float>(),
-                         gb.data_ptr<float>(), N, H, hs);
+                         gb.data_ptr<float>(), N, H, hs, N <= 64);
     });
   });
 }

@@ -47,7 +47,7 @@ def test_symlog_grad_gpu():
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
-@pytest.mark.parametrize("D", [32, 64, 255, 512, 1536, 4096])
+@pytest.mark.parametrize("D", [32, 64, 96, 192, 255, 512, 1536, 4096])
 def test_ln_act(dtype, D):
     torch.manual_seed(0)
     N = 64
@@ -75,7 +75,7 @@ def test_ln_act(dtype, D):
 
 @requires_gpu
 @pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
-@pytest.mark.parametrize("H", [8, 512, 1024])
+@pytest.mark.parametrize("H", [8, 512, 1024, 4096])
 def test_gru_gates(dtype, H):
     torch.manual_seed(1)
     B = 48
@@ -1097,3 +1097,51 @@ def test_ppo_losses_fused_matches_eager(clip_vloss, reduction):
     torch.testing.assert_close(lp_new.grad, lp2.grad, atol=1e-5, rtol=1e-4)
     torch.testing.assert_close(v_new.grad, v2.grad, atol=1e-5, rtol=1e-4)
     torch.testing.assert_close(ent.grad, e2.grad, atol=1e-6, rtol=1e-5)
+
+
+@requires_gpu
+def test_ln_act_large_row_count():
+    """Behaviour-MLP shape ([16384, 1024]): the non-register-cached many-row
+    path (scratch-spill gate) must stay numerically identical."""
+    torch.manual_seed(7)
+    N, D = 16384, 1024
+    x = torch.randn(N, D)
+    w = torch.rand(D) + 0.5
+    b = torch.randn(D)
+    ref = ops.layer_norm_act(x.requires_grad_(), w.requires_grad_(), b.requires_grad_(), 1e-3, "silu")
+    xg = x.detach().cuda().requires_grad_()
+    wg = w.detach().cuda().requires_grad_()
+    bg = b.detach().cuda().requires_grad_()
+    got = ops.layer_norm_act(xg, wg, bg, 1e-3, "silu")
+    torch.testing.assert_close(got.cpu(), ref.detach(), atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.cuda())
+    torch.testing.assert_close(xg.grad.cpu(), x.grad, atol=1e-4, rtol=1e-3)
+    # weight grads sum over 16384 rows
+    torch.testing.assert_close(wg.grad.cpu(), w.grad, atol=0.3, rtol=1e-2)
+    torch.testing.assert_close(bg.grad.cpu(), b.grad, atol=0.3, rtol=1e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("D", [96, 192])
+def test_ln_act_conv_channel_widths_bf16(D):
+    """Non-power-of-2 lane groups (L=12/24) on the channels-last conv widths,
+    many rows (the XL conv LayerNorm shape)."""
+    torch.manual_seed(8)
+    N = 4096 + 7
+    x = torch.randn(N, D)
+    w = torch.rand(D) + 0.5
+    b = torch.randn(D)
+    ref = ops.layer_norm_act(x.requires_grad_(), w.requires_grad_(), b.requires_grad_(), 1e-3, "silu")
+    xg = x.detach().bfloat16().cuda().requires_grad_()
+    wg = w.detach().cuda().requires_grad_()
+    bg = b.detach().cuda().requires_grad_()
+    got = ops.layer_norm_act(xg, wg, bg, 1e-3, "silu")
+    torch.testing.assert_close(got.cpu().float(), ref.detach(), atol=3e-2, rtol=3e-2)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.bfloat16().cuda())
+    torch.testing.assert_close(xg.grad.cpu().float(), x.grad, atol=1e-1, rtol=1e-1)
+    torch.testing.assert_close(wg.grad.cpu(), w.grad, atol=0.6, rtol=5e-2)
+    torch.testing.assert_close(bg.grad.cpu(), b.grad, atol=0.6, rtol=5e-2)
