@@ -467,6 +467,131 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
   }
 }
 
+// Vectorized block-per-row LN kernels for the behaviour-MLP shapes (many
+// rows x D <= 256*V): one 16-B load per thread covers the whole row, the
+// row lives in STATICALLY-sized registers (no scratch spill — the dynamic
+// register caches of ln_act_fwd_kernel measured 49-127 us on [16384, 1024]),
+// one fused sum/sumsq barrier round, vector store.  ~4x the effective
+// bandwidth of the scalar per-row path.
+template <typename T, typename TW, bool SILU>
+__global__ void ln_act_fwd_v_kernel(const T* __restrict__ x, const TW* __restrict__ w,
+                                    const TW* __restrict__ b, T* __restrict__ y,
+                                    float* __restrict__ mean_out, float* __restrict__ rstd_out, long N,
+                                    int D, float eps, long ys) {
+  constexpr int V = 16 / sizeof(T);
+  __shared__ float lds[18];
+  const int nth = (D + V - 1) / V;  // threads holding a vector (D % V == 0)
+  const int j0 = threadIdx.x * V;
+  const bool own = threadIdx.x < nth;
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + row * (long)D;
+    T* yr = y + row * ys;
+    LnVec<T, V> xv;
+    float s = 0.f, s2 = 0.f;
+    if (own) {
+      xv.u = *reinterpret_cast<const uint4*>(xr + j0);
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        float v = ld(xv.e, e);
+        s += v;
+        s2 += v * v;
+      }
+    }
+    block_sum2(s, s2, lds);
+    const float mean = s / D;
+    const float var = s2 / D - mean * mean;
+    const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
+    if (threadIdx.x == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+    if (own) {
+      LnVec<T, V> yv;
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        float z = (ld(xv.e, e) - mean) * rstd * ld(w, j0 + e) + ld(b, j0 + e);
+        if (SILU) z = z / (1.f + expf(-z));
+        st(yv.e, e, z);
+      }
+      *reinterpret_cast<uint4*>(yr + j0) = yv.u;
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T, typename TW, bool SILU>
+__global__ void ln_act_bwd_v_kernel(const T* __restrict__ gy, const T* __restrict__ x,
+                                    const TW* __restrict__ w, const TW* __restrict__ b,
+                                    const float* __restrict__ mean, const float* __restrict__ rstd,
+                                    T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb,
+                                    long N, int D, long gys) {
+  constexpr int V = 16 / sizeof(T);
+  extern __shared__ __attribute__((aligned(16))) float smem[];
+  float* lds = smem;
+  float* gw_acc = smem + 32;   // [D]
+  float* gb_acc = gw_acc + D;  // [D]
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    gw_acc[j] = 0.f;
+    gb_acc[j] = 0.f;
+  }
+  __syncthreads();
+  const int nth = (D + V - 1) / V;
+  const int j0 = threadIdx.x * V;
+  const bool own = threadIdx.x < nth;
+  float wv[V], bv[V];
+  if (own) {
+#pragma unroll
+    for (int e = 0; e < V; ++e) {
+      wv[e] = ld(w, j0 + e);
+      bv[e] = ld(b, j0 + e);
+    }
+  }
+  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+    const T* xr = x + row * (long)D;
+    const T* gr = gy + row * gys;
+    T* gxr = gx + row * (long)D;
+    const float m = mean[row], rs = rstd[row];
+    float xh[V], gz[V];
+    float s1 = 0.f, s2 = 0.f;
+    if (own) {
+      LnVec<T, V> xv, gv;
+      xv.u = *reinterpret_cast<const uint4*>(xr + j0);
+      gv.u = *reinterpret_cast<const uint4*>(gr + j0);
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        xh[e] = (ld(xv.e, e) - m) * rs;
+        float g = ld(gv.e, e);
+        if (SILU) {
+          float z = xh[e] * wv[e] + bv[e];
+          float sig = 1.f / (1.f + expf(-z));
+          g *= sig * (1.f + z * (1.f - sig));
+        }
+        gz[e] = g;
+        float gxhat = g * wv[e];
+        s1 += gxhat;
+        s2 += gxhat * xh[e];
+      }
+    }
+    block_sum2(s1, s2, lds);
+    const float S1 = s1 / D, S2 = s2 / D;
+    if (own) {
+      LnVec<T, V> ov;
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        gw_acc[j0 + e] += gz[e] * xh[e];
+        gb_acc[j0 + e] += gz[e];
+        st(ov.e, e, (gz[e] * wv[e] - S1 - xh[e] * S2) * rs);
+      }
+      *reinterpret_cast<uint4*>(gxr + j0) = ov.u;
+    }
+    __syncthreads();
+  }
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    atomicAdd(&gw[j], gw_acc[j]);
+    atomicAdd(&gb[j], gb_acc[j]);
+  }
+}
+
 // returns L (lanes per row) when the vectorized channels-last path applies
 template <typename T>
 int ln_cl_lanes(int D, long stride) {
@@ -558,6 +683,21 @@ void ln_act_fwd_core(const torch::Tensor& x, const torch::Tensor& w, const torch
           launch_ln_fwd_cl<T, TW, false>(L, N, stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                                          (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
                                          rstd.data_ptr<float>(), D, (float)eps);
+      } else if (N > 64 && D % (16 / (int)sizeof(T)) == 0 && D <= kBlock * (16 / (int)sizeof(T)) &&
+                 ys % (16 / (int)sizeof(T)) == 0) {
+        // vectorized block-per-row (behaviour-MLP shapes): 16-B lane loads,
+        // row in statically-sized registers, one fused reduction round
+        int blocks = (int)std::min(N, (long)2048);
+        if (silu)
+          hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), 0,
+                             stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                             (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), N, D, (float)eps, ys);
+        else
+          hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), 0,
+                             stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                             (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), N, D, (float)eps, ys);
       } else if (small) {
         int blocks = (int)std::min((N + 3) / 4, (long)2048);
         if (silu)
@@ -848,8 +988,37 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
   }
   size_t shmem = (32 + 2 * (size_t)D) * sizeof(float);
   TORCH_CHECK(shmem <= 160 * 1024, "ln_act_bwd: D too large for LDS accumulation");
-  int blocks = (int)std::min(N, (long)2048);
+  int blocks = (int)std::min(N, (long)512);
   auto stream = at::cuda::getCurrentCUDAStream();
+  {
+    bool done = false;
+    AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_bwd_v", [&] {
+      using T = scalar_t;
+      AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_bwd_v_w", [&] {
+        using TW = scalar_t;
+        constexpr int V = 16 / sizeof(T);
+        if (N > 64 && D % V == 0 && D <= kBlock * V && gys % V == 0) {
+          // vectorized block-per-row; 128 blocks bound the gw/gb atomic
+          // flush (blocks x D adds) while the row loop amortizes LDS init
+          int vblocks = (int)std::min(N, (long)128);
+          if (silu)
+            hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true>), dim3(vblocks), dim3(kBlock), shmem,
+                               stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                               (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
+                               gb.data_ptr<float>(), N, D, gys);
+          else
+            hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, false>), dim3(vblocks), dim3(kBlock), shmem,
+                               stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                               (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
+                               gb.data_ptr<float>(), N, D, gys);
+          done = true;
+        }
+      });
+    });
+    if (done) return;
+  }
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "ln_act_bwd", [&] {
     using T = scalar_t;
     AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_bwd_w", [&] {
@@ -1041,7 +1210,7 @@ void gru_gates_fwd_core(const torch::Tensor& y, const torch::Tensor& h, const to
   // wide dispatch: few rows x large H leave the chip idle on the row-per-
   // block kernel; chunk columns so N*C workgroups >= ~384
   const bool wide = N <= 64 && H >= 2048;
-  const int C = wide ? std::min<int>(std::max<int>(1, 384 / (int)N), (H + 255) / 256) : 1;
+  const int C = wide ? std::min<int>(std::max<int>(1, 512 / (int)N), (H + 127) / 128) : 1;
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, y.scalar_type(), "gru_gates_fwd", [&] {
     using T = scalar_t;
     AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "gru_gates_fwd_w", [&] {
@@ -1365,7 +1534,7 @@ void gru_gates_bwd_core(const torch::Tensor& gh, const void* gh2, const void* gh
     AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "gru_gates_bwd_w", [&] {
       using TW = scalar_t;
       if (wide) {
-        const int C = std::min<int>(std::max<int>(1, 384 / (int)N), (H + 255) / 256);
+        const int C = std::min<int>(std::max<int>(1, 512 / (int)N), (H + 127) / 128);
         const int W = (H + C - 1) / C;
         size_t shmem = (32 + 6 * (size_t)W) * sizeof(float);
         hipLaunchKernelGGL((gru_gates_bwd_wide_kernel<T, TW>), dim3((int)N, C), dim3(kBlock), shmem,
