@@ -467,128 +467,179 @@ __global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restri
   }
 }
 
-// Vectorized block-per-row LN kernels for the behaviour-MLP shapes (many
-// rows x D <= 256*V): one 16-B load per thread covers the whole row, the
-// row lives in STATICALLY-sized registers (no scratch spill — the dynamic
-// register caches of ln_act_fwd_kernel measured 49-127 us on [16384, 1024]),
-// one fused sum/sumsq barrier round, vector store.  ~4x the effective
-// bandwidth of the scalar per-row path.
-template <typename T, typename TW, bool SILU>
+// Vectorized wave-per-row LN kernels for the behaviour-MLP shapes (many
+// rows x 512 < D <= 2048): each lane holds K 16-B vectors of its row, so a
+// row needs nlanes = D/(K*V) <= 64 lanes and the LN statistics reduce with
+// wave shuffles ONLY — no LDS, no barriers in the row loop (a first
+// block-per-row version paid a block_sum2 barrier round per row and ran
+// 127 us on [16384, 1024]).  gw/gb accumulate in registers over the row
+// loop (each lane owns fixed columns), fold across the block's waves in
+// LDS once, and flush with one atomicAdd per column per block — blocks
+// bounded so the flush stays off the critical path.
+template <typename T, typename TW, bool SILU, int K>
 __global__ void ln_act_fwd_v_kernel(const T* __restrict__ x, const TW* __restrict__ w,
                                     const TW* __restrict__ b, T* __restrict__ y,
                                     float* __restrict__ mean_out, float* __restrict__ rstd_out, long N,
                                     int D, float eps, long ys) {
   constexpr int V = 16 / sizeof(T);
-  __shared__ float lds[18];
-  const int nth = (D + V - 1) / V;  // threads holding a vector (D % V == 0)
-  const int j0 = threadIdx.x * V;
-  const bool own = threadIdx.x < nth;
-  for (long row = blockIdx.x; row < N; row += gridDim.x) {
-    const T* xr = x + row * (long)D;
-    T* yr = y + row * ys;
-    LnVec<T, V> xv;
-    float s = 0.f, s2 = 0.f;
-    if (own) {
-      xv.u = *reinterpret_cast<const uint4*>(xr + j0);
+  const int lane = threadIdx.x & 63;
+  const int nlanes = D / (K * V);
+  const bool own = lane < nlanes;
+  const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
+  float wv[K * V], bv[K * V];
+  if (own) {
+#pragma unroll
+    for (int k = 0; k < K; ++k)
 #pragma unroll
       for (int e = 0; e < V; ++e) {
-        float v = ld(xv.e, e);
-        s += v;
-        s2 += v * v;
+        wv[k * V + e] = ld(w, (k * nlanes + lane) * V + e);
+        bv[k * V + e] = ld(b, (k * nlanes + lane) * V + e);
+      }
+  }
+  for (long row = wave; row < N; row += nwaves) {
+    const T* xr = x + row * (long)D;
+    T* yr = y + row * ys;
+    LnVec<T, V> xv[K];
+    float s = 0.f, s2 = 0.f;
+    if (own) {
+#pragma unroll
+      for (int k = 0; k < K; ++k) {
+        xv[k].u = *reinterpret_cast<const uint4*>(xr + (k * nlanes + lane) * V);
+#pragma unroll
+        for (int e = 0; e < V; ++e) {
+          float v = ld(xv[k].e, e);
+          s += v;
+          s2 += v * v;
+        }
       }
     }
-    block_sum2(s, s2, lds);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s += __shfl_xor(s, off, 64);
+      s2 += __shfl_xor(s2, off, 64);
+    }
     const float mean = s / D;
     const float var = s2 / D - mean * mean;
     const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
-    if (threadIdx.x == 0) {
+    if (lane == 0) {
       mean_out[row] = mean;
       rstd_out[row] = rstd;
     }
     if (own) {
-      LnVec<T, V> yv;
 #pragma unroll
-      for (int e = 0; e < V; ++e) {
-        float z = (ld(xv.e, e) - mean) * rstd * ld(w, j0 + e) + ld(b, j0 + e);
-        if (SILU) z = z / (1.f + expf(-z));
-        st(yv.e, e, z);
+      for (int k = 0; k < K; ++k) {
+        LnVec<T, V> yv;
+#pragma unroll
+        for (int e = 0; e < V; ++e) {
+          float z = (ld(xv[k].e, e) - mean) * rstd * wv[k * V + e] + bv[k * V + e];
+          if (SILU) z = z / (1.f + expf(-z));
+          st(yv.e, e, z);
+        }
+        *reinterpret_cast<uint4*>(yr + (k * nlanes + lane) * V) = yv.u;
       }
-      *reinterpret_cast<uint4*>(yr + j0) = yv.u;
     }
-    __syncthreads();
   }
 }
 
-template <typename T, typename TW, bool SILU>
+template <typename T, typename TW, bool SILU, int K>
 __global__ void ln_act_bwd_v_kernel(const T* __restrict__ gy, const T* __restrict__ x,
                                     const TW* __restrict__ w, const TW* __restrict__ b,
                                     const float* __restrict__ mean, const float* __restrict__ rstd,
                                     T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb,
                                     long N, int D, long gys) {
   constexpr int V = 16 / sizeof(T);
-  extern __shared__ __attribute__((aligned(16))) float smem[];
-  float* lds = smem;
-  float* gw_acc = smem + 32;   // [D]
-  float* gb_acc = gw_acc + D;  // [D]
-  for (int j = threadIdx.x; j < D; j += blockDim.x) {
-    gw_acc[j] = 0.f;
-    gb_acc[j] = 0.f;
-  }
+  extern __shared__ __attribute__((aligned(16))) float smem[];  // [2*D]
+  float* gw_s = smem;
+  float* gb_s = smem + D;
+  for (int j = threadIdx.x; j < 2 * D; j += blockDim.x) smem[j] = 0.f;
   __syncthreads();
-  const int nth = (D + V - 1) / V;
-  const int j0 = threadIdx.x * V;
-  const bool own = threadIdx.x < nth;
-  float wv[V], bv[V];
+  const int lane = threadIdx.x & 63;
+  const int nlanes = D / (K * V);
+  const bool own = lane < nlanes;
+  const long wave = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const long nwaves = (long)gridDim.x * (blockDim.x >> 6);
+  float wv[K * V], bv[K * V], gwa[K * V], gba[K * V];
+#pragma unroll
+  for (int k = 0; k < K * V; ++k) {
+    gwa[k] = 0.f;
+    gba[k] = 0.f;
+  }
   if (own) {
 #pragma unroll
-    for (int e = 0; e < V; ++e) {
-      wv[e] = ld(w, j0 + e);
-      bv[e] = ld(b, j0 + e);
-    }
+    for (int k = 0; k < K; ++k)
+#pragma unroll
+      for (int e = 0; e < V; ++e) {
+        wv[k * V + e] = ld(w, (k * nlanes + lane) * V + e);
+        bv[k * V + e] = ld(b, (k * nlanes + lane) * V + e);
+      }
   }
-  for (long row = blockIdx.x; row < N; row += gridDim.x) {
+  for (long row = wave; row < N; row += nwaves) {
     const T* xr = x + row * (long)D;
     const T* gr = gy + row * gys;
     T* gxr = gx + row * (long)D;
     const float m = mean[row], rs = rstd[row];
-    float xh[V], gz[V];
+    float xh[K * V], gz[K * V];
     float s1 = 0.f, s2 = 0.f;
     if (own) {
-      LnVec<T, V> xv, gv;
-      xv.u = *reinterpret_cast<const uint4*>(xr + j0);
-      gv.u = *reinterpret_cast<const uint4*>(gr + j0);
 #pragma unroll
-      for (int e = 0; e < V; ++e) {
-        xh[e] = (ld(xv.e, e) - m) * rs;
-        float g = ld(gv.e, e);
-        if (SILU) {
-          float z = xh[e] * wv[e] + bv[e];
-          float sig = 1.f / (1.f + expf(-z));
-          g *= sig * (1.f + z * (1.f - sig));
+      for (int k = 0; k < K; ++k) {
+        LnVec<T, V> xv, gv;
+        xv.u = *reinterpret_cast<const uint4*>(xr + (k * nlanes + lane) * V);
+        gv.u = *reinterpret_cast<const uint4*>(gr + (k * nlanes + lane) * V);
+#pragma unroll
+        for (int e = 0; e < V; ++e) {
+          const int c = k * V + e;
+          xh[c] = (ld(xv.e, e) - m) * rs;
+          float g = ld(gv.e, e);
+          if (SILU) {
+            float z = xh[c] * wv[c] + bv[c];
+            float sig = 1.f / (1.f + expf(-z));
+            g *= sig * (1.f + z * (1.f - sig));
+          }
+          gz[c] = g;
+          float gxhat = g * wv[c];
+          s1 += gxhat;
+          s2 += gxhat * xh[c];
         }
-        gz[e] = g;
-        float gxhat = g * wv[e];
-        s1 += gxhat;
-        s2 += gxhat * xh[e];
       }
     }
-    block_sum2(s1, s2, lds);
-    const float S1 = s1 / D, S2 = s2 / D;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s1 += __shfl_xor(s1, off, 64);
+      s2 += __shfl_xor(s2, off, 64);
+    }
     if (own) {
-      LnVec<T, V> ov;
+      const float S1 = s1 / D, S2 = s2 / D;
+#pragma unroll
+      for (int k = 0; k < K; ++k) {
+        LnVec<T, V> ov;
+#pragma unroll
+        for (int e = 0; e < V; ++e) {
+          const int c = k * V + e;
+          gwa[c] += gz[c] * xh[c];
+          gba[c] += gz[c];
+          st(ov.e, e, (gz[c] * wv[c] - S1 - xh[c] * S2) * rs);
+        }
+        *reinterpret_cast<uint4*>(gxr + (k * nlanes + lane) * V) = ov.u;
+      }
+    }
+  }
+  // fold the block's waves through LDS (all waves own the same columns)
+  if (own) {
+#pragma unroll
+    for (int k = 0; k < K; ++k)
 #pragma unroll
       for (int e = 0; e < V; ++e) {
-        gw_acc[j0 + e] += gz[e] * xh[e];
-        gb_acc[j0 + e] += gz[e];
-        st(ov.e, e, (gz[e] * wv[e] - S1 - xh[e] * S2) * rs);
+        const int j = (k * nlanes + lane) * V + e;
+        atomicAdd(&gw_s[j], gwa[k * V + e]);
+        atomicAdd(&gb_s[j], gba[k * V + e]);
       }
-      *reinterpret_cast<uint4*>(gxr + j0) = ov.u;
-    }
-    __syncthreads();
   }
+  __syncthreads();
   for (int j = threadIdx.x; j < D; j += blockDim.x) {
-    atomicAdd(&gw[j], gw_acc[j]);
-    atomicAdd(&gb[j], gb_acc[j]);
+    atomicAdd(&gw[j], gw_s[j]);
+    atomicAdd(&gb[j], gb_s[j]);
   }
 }
 
@@ -683,21 +734,40 @@ void ln_act_fwd_core(const torch::Tensor& x, const torch::Tensor& w, const torch
           launch_ln_fwd_cl<T, TW, false>(L, N, stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
                                          (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
                                          rstd.data_ptr<float>(), D, (float)eps);
-      } else if (N > 64 && D % (16 / (int)sizeof(T)) == 0 && D <= kBlock * (16 / (int)sizeof(T)) &&
-                 ys % (16 / (int)sizeof(T)) == 0) {
-        // vectorized block-per-row (behaviour-MLP shapes): 16-B lane loads,
-        // row in statically-sized registers, one fused reduction round
-        int blocks = (int)std::min(N, (long)2048);
-        if (silu)
-          hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, true>), dim3(blocks), dim3(kBlock), 0,
-                             stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
-                             (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
-                             rstd.data_ptr<float>(), N, D, (float)eps, ys);
-        else
-          hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, false>), dim3(blocks), dim3(kBlock), 0,
-                             stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
-                             (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
-                             rstd.data_ptr<float>(), N, D, (float)eps, ys);
+      } else if (int K = (N > 64 && ys % (16 / (int)sizeof(T)) == 0)
+                              ? ((D % (2 * (16 / (int)sizeof(T))) == 0 && D / (2 * (16 / (int)sizeof(T))) <= 64)
+                                     ? 2
+                                     : ((D % (4 * (16 / (int)sizeof(T))) == 0 &&
+                                         D / (4 * (16 / (int)sizeof(T))) <= 64)
+                                            ? 4
+                                            : 0))
+                              : 0) {
+        // vectorized wave-per-row (behaviour-MLP shapes): no barriers in
+        // the row loop, wave-shuffle LN stats, 16-B lane loads
+        int blocks = (int)std::min((N + 3) / 4, (long)2048);
+        if (silu) {
+          if (K == 2)
+            hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, true, 2>), dim3(blocks), dim3(kBlock), 0,
+                               stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                               (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), N, D, (float)eps, ys);
+          else
+            hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, true, 4>), dim3(blocks), dim3(kBlock), 0,
+                               stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                               (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), N, D, (float)eps, ys);
+        } else {
+          if (K == 2)
+            hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, false, 2>), dim3(blocks), dim3(kBlock), 0,
+                               stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                               (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), N, D, (float)eps, ys);
+          else
+            hipLaunchKernelGGL((ln_act_fwd_v_kernel<T, TW, false, 4>), dim3(blocks), dim3(kBlock), 0,
+                               stream.stream(), (const T*)x.data_ptr(), (const TW*)wc.data_ptr(),
+                               (const TW*)bc.data_ptr(), (T*)y.data_ptr(), mean.data_ptr<float>(),
+                               rstd.data_ptr<float>(), N, D, (float)eps, ys);
+        }
       } else if (small) {
         int blocks = (int)std::min((N + 3) / 4, (long)2048);
         if (silu)
@@ -997,22 +1067,44 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
       AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, wc.scalar_type(), "ln_act_bwd_v_w", [&] {
         using TW = scalar_t;
         constexpr int V = 16 / sizeof(T);
-        if (N > 64 && D % V == 0 && D <= kBlock * V && gys % V == 0) {
-          // vectorized block-per-row; 128 blocks bound the gw/gb atomic
-          // flush (blocks x D adds) while the row loop amortizes LDS init
-          int vblocks = (int)std::min(N, (long)128);
-          if (silu)
-            hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true>), dim3(vblocks), dim3(kBlock), shmem,
-                               stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
-                               (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
-                               rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
-                               gb.data_ptr<float>(), N, D, gys);
-          else
-            hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, false>), dim3(vblocks), dim3(kBlock), shmem,
-                               stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
-                               (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(), mean.data_ptr<float>(),
-                               rstd.data_ptr<float>(), (T*)gx.data_ptr(), gw.data_ptr<float>(),
-                               gb.data_ptr<float>(), N, D, gys);
+        int K = 0;
+        if (N > 64 && gys % V == 0) {
+          if (D % (2 * V) == 0 && D / (2 * V) <= 64)
+            K = 2;
+          else if (D % (4 * V) == 0 && D / (4 * V) <= 64)
+            K = 4;
+        }
+        if (K) {
+          // vectorized wave-per-row; bounded blocks keep the per-block
+          // gw/gb atomic flush (blocks x D adds) off the critical path
+          int vblocks = (int)std::min((N + 3) / 4, (long)384);
+          if (silu) {
+            if (K == 2)
+              hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true, 2>), dim3(vblocks), dim3(kBlock),
+                                 shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                                 (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
+                                 mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
+            else
+              hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true, 4>), dim3(vblocks), dim3(kBlock),
+                                 shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                                 (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
+                                 mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
+          } else {
+            if (K == 2)
+              hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, false, 2>), dim3(vblocks), dim3(kBlock),
+                                 shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                                 (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
+                                 mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
+            else
+              hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, false, 4>), dim3(vblocks), dim3(kBlock),
+                                 shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
+                                 (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
+                                 mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
+          }
           done = true;
         }
       });

@@ -1100,11 +1100,13 @@ def test_ppo_losses_fused_matches_eager(clip_vloss, reduction):
 
 
 @requires_gpu
-def test_ln_act_large_row_count():
-    """Behaviour-MLP shape ([16384, 1024]): the non-register-cached many-row
-    path (scratch-spill gate) must stay numerically identical."""
+@pytest.mark.parametrize("D", [640, 768, 1024, 1536, 2048])
+def test_ln_act_large_row_count(D):
+    """Behaviour-MLP shapes (many rows x 512<D<=2048): the vectorized
+    wave-per-row path must stay numerically identical, including partial
+    lane coverage (D=640/768) and K=4 rows (D=1536/2048)."""
     torch.manual_seed(7)
-    N, D = 16384, 1024
+    N = 16384 if D == 1024 else 4096
     x = torch.randn(N, D)
     w = torch.rand(D) + 0.5
     b = torch.randn(D)
