@@ -158,6 +158,18 @@ def run_bench(n_gpus: int, steps: int, warmup: int, overrides=()) -> dict:
 
         if not has_ext():
             raise RuntimeError("HIP extension _sheep_hip is not built — refusing to bench the eager fallback")
+        # hipBLASLt algorithm tuning during the (untimed) warmup: the scan's
+        # M=16 long-K GEMMs gain ~1 ms/step on XL from better algo picks
+        # (neutral on S, measured).  Single-rank only — tuning sweeps would
+        # desync multi-rank warmup; opt out with SHEEPRL_AMD_TUNABLEOP=0.
+        if world_size == 1 and os.environ.get("SHEEPRL_AMD_TUNABLEOP", "1") != "0":
+            try:
+                import torch.cuda.tunable as tunable
+
+                tunable.set_filename(os.environ.get("TMPDIR", "/tmp") + "/sheeprl_tunableop.csv")
+                tunable.enable(True)
+            except Exception as e:  # noqa: BLE001
+                print(f"[bench] TunableOp unavailable ({e})", file=sys.stderr)
 
     runtime, envs, models, optims, moments, rb = _setup(cfg, rank, world_size)
     world_model, actor, critic, target_critic, player = models

@@ -497,15 +497,26 @@ __global__ void ln_act_fwd_v_kernel(const T* __restrict__ x, const TW* __restric
         bv[k * V + e] = ld(b, (k * nlanes + lane) * V + e);
       }
   }
+  LnVec<T, V> xv[K], xn[K];
+  if (own && wave < N) {
+#pragma unroll
+    for (int k = 0; k < K; ++k)
+      xv[k].u = *reinterpret_cast<const uint4*>(x + wave * (long)D + (k * nlanes + lane) * V);
+  }
   for (long row = wave; row < N; row += nwaves) {
-    const T* xr = x + row * (long)D;
     T* yr = y + row * ys;
-    LnVec<T, V> xv[K];
+    // issue the NEXT row's loads before this row's math so the row loop
+    // keeps >1 memory latency in flight per wave (no barriers to hide it)
+    const long nrow = row + nwaves;
+    if (own && nrow < N) {
+#pragma unroll
+      for (int k = 0; k < K; ++k)
+        xn[k].u = *reinterpret_cast<const uint4*>(x + nrow * (long)D + (k * nlanes + lane) * V);
+    }
     float s = 0.f, s2 = 0.f;
     if (own) {
 #pragma unroll
       for (int k = 0; k < K; ++k) {
-        xv[k].u = *reinterpret_cast<const uint4*>(xr + (k * nlanes + lane) * V);
 #pragma unroll
         for (int e = 0; e < V; ++e) {
           float v = ld(xv[k].e, e);
@@ -539,6 +550,8 @@ __global__ void ln_act_fwd_v_kernel(const T* __restrict__ x, const TW* __restric
         *reinterpret_cast<uint4*>(yr + (k * nlanes + lane) * V) = yv.u;
       }
     }
+#pragma unroll
+    for (int k = 0; k < K; ++k) xv[k] = xn[k];
   }
 }
 
@@ -574,24 +587,35 @@ __global__ void ln_act_bwd_v_kernel(const T* __restrict__ gy, const T* __restric
         bv[k * V + e] = ld(b, (k * nlanes + lane) * V + e);
       }
   }
+  LnVec<T, V> xv[K], gv[K], xn[K], gn[K];
+  if (own && wave < N) {
+#pragma unroll
+    for (int k = 0; k < K; ++k) {
+      xv[k].u = *reinterpret_cast<const uint4*>(x + wave * (long)D + (k * nlanes + lane) * V);
+      gv[k].u = *reinterpret_cast<const uint4*>(gy + wave * gys + (k * nlanes + lane) * V);
+    }
+  }
   for (long row = wave; row < N; row += nwaves) {
-    const T* xr = x + row * (long)D;
-    const T* gr = gy + row * gys;
     T* gxr = gx + row * (long)D;
+    const long nrow = row + nwaves;
+    if (own && nrow < N) {
+#pragma unroll
+      for (int k = 0; k < K; ++k) {
+        xn[k].u = *reinterpret_cast<const uint4*>(x + nrow * (long)D + (k * nlanes + lane) * V);
+        gn[k].u = *reinterpret_cast<const uint4*>(gy + nrow * gys + (k * nlanes + lane) * V);
+      }
+    }
     const float m = mean[row], rs = rstd[row];
     float xh[K * V], gz[K * V];
     float s1 = 0.f, s2 = 0.f;
     if (own) {
 #pragma unroll
       for (int k = 0; k < K; ++k) {
-        LnVec<T, V> xv, gv;
-        xv.u = *reinterpret_cast<const uint4*>(xr + (k * nlanes + lane) * V);
-        gv.u = *reinterpret_cast<const uint4*>(gr + (k * nlanes + lane) * V);
 #pragma unroll
         for (int e = 0; e < V; ++e) {
           const int c = k * V + e;
-          xh[c] = (ld(xv.e, e) - m) * rs;
-          float g = ld(gv.e, e);
+          xh[c] = (ld(xv[k].e, e) - m) * rs;
+          float g = ld(gv[k].e, e);
           if (SILU) {
             float z = xh[c] * wv[c] + bv[c];
             float sig = 1.f / (1.f + expf(-z));
@@ -623,6 +647,11 @@ __global__ void ln_act_bwd_v_kernel(const T* __restrict__ gy, const T* __restric
         }
         *reinterpret_cast<uint4*>(gxr + (k * nlanes + lane) * V) = ov.u;
       }
+    }
+#pragma unroll
+    for (int k = 0; k < K; ++k) {
+      xv[k] = xn[k];
+      gv[k] = gn[k];
     }
   }
   // fold the block's waves through LDS (all waves own the same columns)
