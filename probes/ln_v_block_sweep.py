@@ -8,7 +8,11 @@ limits from trace effects, and times the op at several row counts.
 Run on a GPU box:  python probes/ln_v_block_sweep.py
 """
 
+import sys
 import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 import torch
 
