@@ -697,28 +697,14 @@ __global__ void __launch_bounds__(kBlock) ln_act_bwd_v_kernel(const T* __restric
 // (accumulate semantics — the scan's _acc variants add across calls)
 __global__ void ln_part_reduce_kernel(const float* __restrict__ gpart, float* __restrict__ gw,
                                       float* __restrict__ gb, int nblocks, int D) {
-  // grid (j-blocks, bk-chunks): a single serial bk loop per thread issued
-  // one load per accumulate (vmcnt(0) each iteration) and measured 236 us;
-  // chunking the block dimension across gridDim.y and keeping 4
-  // independent accumulator chains keeps the strided loads in flight
-  const int j = blockIdx.x * blockDim.x + threadIdx.x;
-  if (j >= 2 * D) return;
-  const int per = (nblocks + (int)gridDim.y - 1) / (int)gridDim.y;
-  const int b0 = (int)blockIdx.y * per, b1 = min(b0 + per, nblocks);
-  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
-  int bk = b0;
-  for (; bk + 3 < b1; bk += 4) {
-    a0 += gpart[(size_t)bk * 2 * D + j];
-    a1 += gpart[(size_t)(bk + 1) * 2 * D + j];
-    a2 += gpart[(size_t)(bk + 2) * 2 * D + j];
-    a3 += gpart[(size_t)(bk + 3) * 2 * D + j];
+  for (int j = blockIdx.x * blockDim.x + threadIdx.x; j < 2 * D; j += gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int bk = 0; bk < nblocks; ++bk) acc += gpart[(size_t)bk * 2 * D + j];
+    if (j < D)
+      gw[j] += acc;
+    else
+      gb[j - D] += acc;
   }
-  for (; bk < b1; ++bk) a0 += gpart[(size_t)bk * 2 * D + j];
-  const float acc = (a0 + a1) + (a2 + a3);
-  if (j < D)
-    atomicAdd(&gw[j], acc);
-  else
-    atomicAdd(&gb[j - D], acc);
 }
 
 // returns L (lanes per row) when the vectorized channels-last path applies
@@ -1184,9 +1170,9 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
                                  mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
                                  gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, gpart.data_ptr<float>());
           }
-          hipLaunchKernelGGL(ln_part_reduce_kernel, dim3((2 * D + kBlock - 1) / kBlock, 8),
-                             dim3(kBlock), 0, stream.stream(), gpart.data_ptr<float>(),
-                             gw.data_ptr<float>(), gb.data_ptr<float>(), vblocks, D);
+          hipLaunchKernelGGL(ln_part_reduce_kernel, dim3(16), dim3(kBlock), 0, stream.stream(),
+                             gpart.data_ptr<float>(), gw.data_ptr<float>(), gb.data_ptr<float>(), vblocks,
+                             D);
           done = true;
         }
       });
