@@ -560,7 +560,7 @@ __global__ void __launch_bounds__(kBlock) ln_act_bwd_v_kernel(const T* __restric
                                     const TW* __restrict__ w, const TW* __restrict__ b,
                                     const float* __restrict__ mean, const float* __restrict__ rstd,
                                     T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb,
-                                    long N, int D, long gys, float* __restrict__ gpart) {
+                                    long N, int D, long gys) {
   constexpr int V = 16 / sizeof(T);
   extern __shared__ __attribute__((aligned(16))) float smem[];  // [2*D]
   float* gw_s = smem;
@@ -675,35 +675,9 @@ __global__ void __launch_bounds__(kBlock) ln_act_bwd_v_kernel(const T* __restric
       }
   }
   __syncthreads();
-  if (gpart) {
-    // partial-store flush: per-block row into a [blocks, 2D] workspace
-    // (plain stores; a tiny reduce kernel sums the block dimension) — the
-    // atomicAdd flush serialized at blocks x D adds and capped the useful
-    // block count at ~384
-    float* rowp = gpart + (size_t)blockIdx.x * 2 * D;
-    for (int j = threadIdx.x; j < D; j += blockDim.x) {
-      rowp[j] = gw_s[j];
-      rowp[D + j] = gb_s[j];
-    }
-  } else {
-    for (int j = threadIdx.x; j < D; j += blockDim.x) {
-      atomicAdd(&gw[j], gw_s[j]);
-      atomicAdd(&gb[j], gb_s[j]);
-    }
-  }
-}
-
-// sums the block dimension of a [B, 2D] partial workspace INTO gw/gb
-// (accumulate semantics — the scan's _acc variants add across calls)
-__global__ void ln_part_reduce_kernel(const float* __restrict__ gpart, float* __restrict__ gw,
-                                      float* __restrict__ gb, int nblocks, int D) {
-  for (int j = blockIdx.x * blockDim.x + threadIdx.x; j < 2 * D; j += gridDim.x * blockDim.x) {
-    float acc = 0.f;
-    for (int bk = 0; bk < nblocks; ++bk) acc += gpart[(size_t)bk * 2 * D + j];
-    if (j < D)
-      gw[j] += acc;
-    else
-      gb[j - D] += acc;
+  for (int j = threadIdx.x; j < D; j += blockDim.x) {
+    atomicAdd(&gw[j], gw_s[j]);
+    atomicAdd(&gb[j], gb_s[j]);
   }
 }
 
@@ -1139,40 +1113,36 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
             K = 4;
         }
         if (K) {
-          // vectorized wave-per-row; partial-store flush + reduce kernel
-          // lets the block count scale without an atomic storm
-          int vblocks = (int)std::min((N + 3) / 4, (long)1024);
-          auto gpart = torch::empty({(long)vblocks * 2 * D}, gw.options());
+          // vectorized wave-per-row; bounded blocks keep the per-block
+          // gw/gb atomic flush (blocks x D adds) off the critical path
+          int vblocks = (int)std::min((N + 3) / 4, (long)384);
           if (silu) {
             if (K == 2)
               hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true, 2>), dim3(vblocks), dim3(kBlock),
                                  shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                                  (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
                                  mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
-                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, gpart.data_ptr<float>());
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
             else
               hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true, 4>), dim3(vblocks), dim3(kBlock),
                                  shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                                  (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
                                  mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
-                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, gpart.data_ptr<float>());
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
           } else {
             if (K == 2)
               hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, false, 2>), dim3(vblocks), dim3(kBlock),
                                  shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                                  (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
                                  mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
-                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, gpart.data_ptr<float>());
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
             else
               hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, false, 4>), dim3(vblocks), dim3(kBlock),
                                  shmem, stream.stream(), (const T*)gy.data_ptr(), (const T*)x.data_ptr(),
                                  (const TW*)wc.data_ptr(), (const TW*)bc.data_ptr(),
                                  mean.data_ptr<float>(), rstd.data_ptr<float>(), (T*)gx.data_ptr(),
-                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys, gpart.data_ptr<float>());
+                                 gw.data_ptr<float>(), gb.data_ptr<float>(), N, D, gys);
           }
-          hipLaunchKernelGGL(ln_part_reduce_kernel, dim3(16), dim3(kBlock), 0, stream.stream(),
-                             gpart.data_ptr<float>(), gw.data_ptr<float>(), gb.data_ptr<float>(), vblocks,
-                             D);
           done = true;
         }
       });
