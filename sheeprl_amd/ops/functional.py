@@ -429,7 +429,10 @@ def ppo_losses(new_logprobs: Tensor, old_logprobs: Tensor, advantages: Tensor,
     """Returns the (policy_loss, value_loss, entropy_loss) triple, fused on
     GPU for reduction 'mean'/'sum'.  Callers fall back to the eager
     composition off-GPU (sheeprl_amd/algos/ppo/loss.py)."""
-    out3 = _PPOLosses.apply(new_logprobs, old_logprobs, advantages, new_values,
-                            old_values, returns, entropy, clip_coef, clip_vloss,
-                            reduction == "mean")
+    # only new_logprobs / new_values / entropy are differentiated (reference
+    # semantics: stored rollout quantities are constants); detach the rest so
+    # autograd never expects gradients the backward does not produce
+    out3 = _PPOLosses.apply(new_logprobs, old_logprobs.detach(), advantages.detach(),
+                            new_values, old_values.detach(), returns.detach(), entropy,
+                            clip_coef, clip_vloss, reduction == "mean")
     return out3[0], out3[1], out3[2]
