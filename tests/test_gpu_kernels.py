@@ -1147,3 +1147,52 @@ def test_ln_act_conv_channel_widths_bf16(D):
     torch.testing.assert_close(xg.grad.cpu().float(), x.grad, atol=1e-1, rtol=1e-1)
     torch.testing.assert_close(wg.grad.cpu(), w.grad, atol=0.6, rtol=5e-2)
     torch.testing.assert_close(bg.grad.cpu(), b.grad, atol=0.6, rtol=5e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("N,D", [(65, 1024), (63, 1024), (100, 2049), (64, 2112 * 3 // 3)])
+def test_ln_act_dispatch_boundaries(N, D):
+    """Shapes straddling the dispatch gates (register-cache N<=64, vec path
+    D%16==0, generic odd-D) must all match the CPU fp32 reference."""
+    torch.manual_seed(11)
+    x = torch.randn(N, D)
+    w = torch.rand(D) + 0.5
+    b = torch.randn(D)
+    ref = ops.layer_norm_act(x.requires_grad_(), w.requires_grad_(), b.requires_grad_(), 1e-3, "silu")
+    xg = x.detach().cuda().requires_grad_()
+    wg = w.detach().cuda().requires_grad_()
+    bg = b.detach().cuda().requires_grad_()
+    got = ops.layer_norm_act(xg, wg, bg, 1e-3, "silu")
+    torch.testing.assert_close(got.cpu(), ref.detach(), atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.cuda())
+    torch.testing.assert_close(xg.grad.cpu(), x.grad, atol=1e-4, rtol=1e-3)
+    torch.testing.assert_close(wg.grad.cpu(), w.grad, atol=0.2, rtol=1e-2)
+    torch.testing.assert_close(bg.grad.cpu(), b.grad, atol=0.2, rtol=1e-2)
+
+
+@requires_gpu
+@pytest.mark.parametrize("B,H", [(65, 2048), (64, 2112), (16, 2048)])
+def test_gru_gates_wide_boundaries(B, H):
+    """Wide-dispatch gate boundaries: just-over-64 rows (generic), H not a
+    multiple of the 128-column chunk, and the standard wide shape."""
+    torch.manual_seed(12)
+    y = torch.randn(B, 3 * H)
+    h = torch.randn(B, H)
+    w = torch.rand(3 * H) + 0.5
+    b = torch.randn(3 * H) * 0.1
+    ref = ops.gru_gates(y.requires_grad_(), h.requires_grad_(), w.requires_grad_(), b.requires_grad_(), 1e-3)
+    yg = y.detach().cuda().requires_grad_()
+    hg = h.detach().cuda().requires_grad_()
+    wg = w.detach().cuda().requires_grad_()
+    bg = b.detach().cuda().requires_grad_()
+    got = ops.gru_gates(yg, hg, wg, bg, 1e-3)
+    torch.testing.assert_close(got.cpu(), ref.detach(), atol=1e-4, rtol=1e-4)
+    g = torch.randn_like(ref)
+    ref.backward(g)
+    got.backward(g.cuda())
+    torch.testing.assert_close(yg.grad.cpu(), y.grad, atol=1e-4, rtol=1e-3)
+    torch.testing.assert_close(hg.grad.cpu(), h.grad, atol=1e-4, rtol=1e-3)
+    torch.testing.assert_close(wg.grad.cpu(), w.grad, atol=0.1, rtol=2e-2)
+    torch.testing.assert_close(bg.grad.cpu(), b.grad, atol=0.1, rtol=2e-2)
