@@ -1545,107 +1545,6 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
   }
 }
 
-// Column-chunked backward (see gru_gates_fwd_wide_kernel): each block
-// recomputes the gate math over the FULL row to form the LN-backward sums
-// S1/S2 (redundant across the row's C blocks; all reads are row-local and
-// L2-shared), then writes gy/ghprev and accumulates gw/gb only for its
-// H-chunk.  gw/gb LDS images cover the chunk, flushed with one atomicAdd
-// per column per block.
-template <typename T, typename TW>
-__global__ void gru_gates_bwd_wide_kernel(const T* __restrict__ gh, const T* __restrict__ gh2,
-                                          const T* __restrict__ gh3, long gh3s, const T* __restrict__ y,
-                                          const T* __restrict__ h, const TW* __restrict__ w,
-                                          const TW* __restrict__ b, const float* __restrict__ mean,
-                                          const float* __restrict__ rstd, T* __restrict__ gy,
-                                          T* __restrict__ ghprev, float* __restrict__ gw,
-                                          float* __restrict__ gb, long N, int H, long hs) {
-  extern __shared__ __attribute__((aligned(16))) float smem[];
-  float* lds = smem;
-  const int W = (H + (int)gridDim.y - 1) / (int)gridDim.y;
-  float* gw_acc = smem + 32;           // [3W]
-  float* gb_acc = gw_acc + 3 * W;      // [3W]
-  const int D = 3 * H;
-  const long row = blockIdx.x;
-  const int j0 = (int)blockIdx.y * W, j1 = min(j0 + W, H);
-  for (int j = threadIdx.x; j < 3 * W; j += blockDim.x) {
-    gw_acc[j] = 0.f;
-    gb_acc[j] = 0.f;
-  }
-  __syncthreads();
-  const T* yr = y + row * (long)D;
-  const T* hr = h + row * hs;
-  const T* ghr = gh + row * (long)H;
-  const T* gh2r = gh2 ? gh2 + row * (long)H : nullptr;
-  const T* gh3r = gh3 ? gh3 + row * gh3s : nullptr;
-  T* gyr = gy + row * (long)D;
-  T* ghp = ghprev + row * (long)H;
-  const float m = mean[row], rs = rstd[row];
-  float s1 = 0.f, s2 = 0.f;
-  for (int j = threadIdx.x; j < H; j += blockDim.x) {
-    float xh_r = (ld(yr, j) - m) * rs;
-    float xh_c = (ld(yr, H + j) - m) * rs;
-    float xh_u = (ld(yr, 2 * H + j) - m) * rs;
-    float zr = xh_r * ld(w, j) + ld(b, j);
-    float zc = xh_c * ld(w, H + j) + ld(b, H + j);
-    float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-    float r = 1.f / (1.f + expf(-zr));
-    float c = tanhf(r * zc);
-    float u = 1.f / (1.f + expf(-(zu - 1.f)));
-    float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
-    float gu = g * (c - ld(hr, j));
-    float gc = g * u;
-    float gzu = gu * u * (1.f - u);
-    float grc = gc * (1.f - c * c);
-    float gzc = grc * r;
-    float gzr = grc * zc * r * (1.f - r);
-    float gxh_r = gzr * ld(w, j);
-    float gxh_c = gzc * ld(w, H + j);
-    float gxh_u = gzu * ld(w, 2 * H + j);
-    s1 += gxh_r + gxh_c + gxh_u;
-    s2 += gxh_r * xh_r + gxh_c * xh_c + gxh_u * xh_u;
-  }
-  block_sum2(s1, s2, lds);
-  const float S1 = s1 / D, S2 = s2 / D;
-  for (int j = j0 + (int)threadIdx.x; j < j1; j += blockDim.x) {
-    float xh_r = (ld(yr, j) - m) * rs;
-    float xh_c = (ld(yr, H + j) - m) * rs;
-    float xh_u = (ld(yr, 2 * H + j) - m) * rs;
-    float zr = xh_r * ld(w, j) + ld(b, j);
-    float zc = xh_c * ld(w, H + j) + ld(b, H + j);
-    float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-    float r = 1.f / (1.f + expf(-zr));
-    float c = tanhf(r * zc);
-    float u = 1.f / (1.f + expf(-(zu - 1.f)));
-    float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
-    float gu = g * (c - ld(hr, j));
-    float gc = g * u;
-    float gzu = gu * u * (1.f - u);
-    float grc = gc * (1.f - c * c);
-    float gzc = grc * r;
-    float gzr = grc * zc * r * (1.f - r);
-    st(ghp, j, g * (1.f - u));
-    const int jc = j - j0;
-    gw_acc[jc] += gzr * xh_r;
-    gb_acc[jc] += gzr;
-    gw_acc[W + jc] += gzc * xh_c;
-    gb_acc[W + jc] += gzc;
-    gw_acc[2 * W + jc] += gzu * xh_u;
-    gb_acc[2 * W + jc] += gzu;
-    st(gyr, j, (gzr * ld(w, j) - S1 - xh_r * S2) * rs);
-    st(gyr, H + j, (gzc * ld(w, H + j) - S1 - xh_c * S2) * rs);
-    st(gyr, 2 * H + j, (gzu * ld(w, 2 * H + j) - S1 - xh_u * S2) * rs);
-  }
-  __syncthreads();
-  for (int jc = threadIdx.x; jc < j1 - j0; jc += blockDim.x) {
-    atomicAdd(&gw[j0 + jc], gw_acc[jc]);
-    atomicAdd(&gb[j0 + jc], gb_acc[jc]);
-    atomicAdd(&gw[H + j0 + jc], gw_acc[W + jc]);
-    atomicAdd(&gb[H + j0 + jc], gb_acc[W + jc]);
-    atomicAdd(&gw[2 * H + j0 + jc], gw_acc[2 * W + jc]);
-    atomicAdd(&gb[2 * H + j0 + jc], gb_acc[2 * W + jc]);
-  }
-}
-
 // Two-stage wide backward: the single wide kernel redundantly recomputed the
 // full-row gate math in every one of its C chunk blocks to form S1/S2
 // (measured 31.7 us at H=4096).  Stage 1 computes each chunk's partial
