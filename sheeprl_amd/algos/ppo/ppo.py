@@ -16,7 +16,7 @@ import numpy as np
 import torch
 
 from sheeprl_amd.algos.ppo.agent import build_agent
-from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, ppo_losses, value_loss
+from sheeprl_amd.algos.ppo.loss import ppo_losses
 from sheeprl_amd.algos.ppo.utils import AGGREGATOR_KEYS, prepare_obs, test
 from sheeprl_amd.config import save_config
 from sheeprl_amd.data import ReplayBuffer

@@ -10,7 +10,7 @@ from typing import Any, Dict
 import numpy as np
 import torch
 
-from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, ppo_losses, value_loss
+from sheeprl_amd.algos.ppo.loss import ppo_losses
 from sheeprl_amd.algos.ppo.utils import AGGREGATOR_KEYS, prepare_obs
 from sheeprl_amd.algos.ppo_recurrent.agent import build_agent
 from sheeprl_amd.config import save_config
