@@ -1173,7 +1173,7 @@ def test_ln_act_dispatch_boundaries(N, D):
 
 
 @requires_gpu
-@pytest.mark.parametrize("B,H", [(65, 2048), (64, 2112), (16, 2048)])
+@pytest.mark.parametrize("B,H", [(65, 2048), (64, 2112), (16, 2048), (1024, 512), (300, 1024)])
 def test_gru_gates_wide_boundaries(B, H):
     """Wide-dispatch gate boundaries: just-over-64 rows (generic), H not a
     multiple of the 128-column chunk, and the standard wide shape."""
