@@ -27,6 +27,25 @@ namespace {
 
 constexpr int kBlock = 256;
 
+// fast transcendental variants for the bf16/half kernel instantiations:
+// v_exp_f32 (~2 instructions) vs libm expf (~16); relative error ~1e-6 is
+// far below bf16 resolution.  fp32 instantiations keep libm for the exact
+// numerics tests.
+template <typename T>
+__device__ __forceinline__ float texp(float v) {
+  if constexpr (sizeof(T) == 2)
+    return __expf(v);
+  else
+    return expf(v);
+}
+template <typename T>
+__device__ __forceinline__ float ttanh(float v) {
+  if constexpr (sizeof(T) == 2)
+    return 1.f - 2.f / (__expf(2.f * v) + 1.f);
+  else
+    return tanhf(v);
+}
+
 __device__ __forceinline__ float wave_sum(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
@@ -255,7 +274,7 @@ union LnVec {
 };
 
 template <typename T, typename TW, bool SILU, int L>
-__global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restrict__ w,
+__global__ void __launch_bounds__(kBlock) ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restrict__ w,
                                      const TW* __restrict__ b, T* __restrict__ y,
                                      float* __restrict__ mean_out, float* __restrict__ rstd_out, long R, int D,
                                      float eps) {
@@ -338,7 +357,7 @@ __global__ void ln_act_fwd_cl_kernel(const T* __restrict__ x, const TW* __restri
 }
 
 template <typename T, typename TW, bool SILU, int L>
-__global__ void ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restrict__ x,
+__global__ void __launch_bounds__(kBlock) ln_act_bwd_cl_kernel(const T* __restrict__ gy, const T* __restrict__ x,
                                      const TW* __restrict__ w, const TW* __restrict__ b,
                                      const float* __restrict__ mean, const float* __restrict__ rstd,
                                      T* __restrict__ gx, float* __restrict__ gw, float* __restrict__ gb, long R,
@@ -544,7 +563,7 @@ __global__ void __launch_bounds__(kBlock) ln_act_fwd_v_kernel(const T* __restric
 #pragma unroll
         for (int e = 0; e < V; ++e) {
           float z = (ld(xv[k].e, e) - mean) * rstd * wv[k * V + e] + bv[k * V + e];
-          if (SILU) z = z / (1.f + expf(-z));
+          if (SILU) z = z / (1.f + texp<T>(-z));
           st(yv.e, e, z);
         }
         *reinterpret_cast<uint4*>(yr + (k * nlanes + lane) * V) = yv.u;
@@ -606,10 +625,12 @@ __global__ void __launch_bounds__(kBlock) ln_act_bwd_v_kernel(const T* __restric
       }
     }
     const float m = mean[row], rs = rstd[row];
-    // xhat/gz stay packed in the xv/gv uint4 registers and are recomputed
-    // in the write pass: storing them as K*V floats spilled to scratch
-    // (68-300 scratch ops in the bf16 instantiations, measured 148 us on
-    // [16384, 1024]); the recompute is pure VALU
+    // gz (post-SILU) is cached in registers across the two passes — the PMC
+    // counters showed this kernel 50% active-issue (VALU-bound), so the
+    // SILU-derivative recompute was the cost, not the loads; with
+    // __launch_bounds__(256) the 16 extra VGPRs no longer spill.  xh is
+    // recomputed (2 ops).
+    float cgz[K * V];
     float s1 = 0.f, s2 = 0.f;
     if (own) {
 #pragma unroll
@@ -621,9 +642,10 @@ __global__ void __launch_bounds__(kBlock) ln_act_bwd_v_kernel(const T* __restric
           float g = ld(gv[k].e, e);
           if (SILU) {
             float z = xh * wv[c] + bv[c];
-            float sig = 1.f / (1.f + expf(-z));
+            float sig = 1.f / (1.f + texp<T>(-z));
             g *= sig * (1.f + z * (1.f - sig));
           }
+          cgz[c] = g;
           float gxhat = g * wv[c];
           s1 += gxhat;
           s2 += gxhat * xh;
@@ -644,12 +666,7 @@ __global__ void __launch_bounds__(kBlock) ln_act_bwd_v_kernel(const T* __restric
         for (int e = 0; e < V; ++e) {
           const int c = k * V + e;
           const float xh = (ld(xv[k].e, e) - m) * rs;
-          float g = ld(gv[k].e, e);
-          if (SILU) {
-            float z = xh * wv[c] + bv[c];
-            float sig = 1.f / (1.f + expf(-z));
-            g *= sig * (1.f + z * (1.f - sig));
-          }
+          const float g = cgz[c];
           gwa[c] += g * xh;
           gba[c] += g;
           st(ov.e, e, (g * wv[c] - S1 - xh * S2) * rs);
@@ -1115,7 +1132,7 @@ void ln_act_bwd_core(const torch::Tensor& gy, const torch::Tensor& x, const torc
         if (K) {
           // vectorized wave-per-row; bounded blocks keep the per-block
           // gw/gb atomic flush (blocks x D adds) off the critical path
-          int vblocks = (int)std::min((N + 3) / 4, (long)384);
+          int vblocks = (int)std::min((N + 3) / 4, (long)768);
           if (silu) {
             if (K == 2)
               hipLaunchKernelGGL((ln_act_bwd_v_kernel<T, TW, true, 2>), dim3(vblocks), dim3(kBlock),
@@ -1316,9 +1333,9 @@ __global__ void gru_gates_fwd_wide_kernel(const T* __restrict__ y, const T* __re
     float zr = ((ld(yr, j) - mean) * rstd) * ld(w, j) + ld(b, j);
     float zc = ((ld(yr, H + j) - mean) * rstd) * ld(w, H + j) + ld(b, H + j);
     float zu = ((ld(yr, 2 * H + j) - mean) * rstd) * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-    float r = 1.f / (1.f + expf(-zr));
-    float c = tanhf(r * zc);
-    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    float r = 1.f / (1.f + texp<T>(-zr));
+    float c = ttanh<T>(r * zc);
+    float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
     float hv = u * c + (1.f - u) * ld(hr, j);
     st(outr, j, hv);
     if (outr2) st(outr2, j, hv);
@@ -1376,9 +1393,9 @@ __global__ void __launch_bounds__(kBlock) gru_gates_fwd_vec_kernel(
         float zr = ((ld(yr_.e, e) - mean) * rstd) * ld(w, j) + ld(b, j);
         float zc = ((ld(yc_.e, e) - mean) * rstd) * ld(w, H + j) + ld(b, H + j);
         float zu = ((ld(yu_.e, e) - mean) * rstd) * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-        float r = 1.f / (1.f + expf(-zr));
-        float c = tanhf(r * zc);
-        float u = 1.f / (1.f + expf(-(zu - 1.f)));
+        float r = 1.f / (1.f + texp<T>(-zr));
+        float c = ttanh<T>(r * zc);
+        float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
         float hv = u * c + (1.f - u) * ld(hr, j);
         st(ov.e, e, hv);
         if (outr2) st(outr2, j, hv);
@@ -1632,9 +1649,9 @@ __device__ __forceinline__ void gru_col_bwd(const T* yr, const T* hr, const T* g
   float zr = xh_r * ld(w, j) + ld(b, j);
   float zc = xh_c * ld(w, H + j) + ld(b, H + j);
   float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-  float r = 1.f / (1.f + expf(-zr));
-  float c = tanhf(r * zc);
-  float u = 1.f / (1.f + expf(-(zu - 1.f)));
+  float r = 1.f / (1.f + texp<T>(-zr));
+  float c = ttanh<T>(r * zc);
+  float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
   float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
   float gu = g * (c - ld(hr, j));
   float gc = g * u;
