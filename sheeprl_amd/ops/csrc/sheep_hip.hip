@@ -5081,7 +5081,7 @@ __global__ void __launch_bounds__(256) scan3_lnsilu_kernel(
     if (m < B) {
       g_out[(long)m * gs + j] = __float2bfloat16(v[r]);
       float z = (v[r] - mr_[m]) * mr_[16 + m] * __bfloat162float(lnw[j]) + __bfloat162float(lnb[j]);
-      z = z / (1.f + expf(-z));
+      z = z / (1.f + __expf(-z));
       st(hu_out, (long)m * hus + hu_off + j, z);
     }
     scratch[(long)(kgrp * 4 + r) * N + col0 + arow] = 0.f;
