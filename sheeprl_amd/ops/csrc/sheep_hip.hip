@@ -39,6 +39,13 @@ __device__ __forceinline__ float texp(float v) {
     return expf(v);
 }
 template <typename T>
+__device__ __forceinline__ float tlog(float v) {
+  if constexpr (sizeof(T) == 2)
+    return __logf(v);
+  else
+    return logf(v);
+}
+template <typename T>
 __device__ __forceinline__ float ttanh(float v) {
   if constexpr (sizeof(T) == 2)
     return 1.f - 2.f / (__expf(2.f * v) + 1.f);
@@ -194,7 +201,7 @@ __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict_
     for (int j = threadIdx.x; j < D; j += blockDim.x) {
       float xhat = (cache[cnt++] - mean) * rstd;
       float z = xhat * ld(w, j) + ld(b, j);
-      if (SILU) z = z / (1.f + expf(-z));
+      if (SILU) z = z / (1.f + texp<T>(-z));
       st(yr, j, z);
     }
     return;
@@ -217,7 +224,7 @@ __global__ void ln_act_fwd_kernel(const T* __restrict__ x, const TW* __restrict_
   for (int j = threadIdx.x; j < D; j += blockDim.x) {
     float xhat = (ld(xr, j) - mean) * rstd;
     float z = xhat * ld(w, j) + ld(b, j);
-    if (SILU) z = z / (1.f + expf(-z));
+    if (SILU) z = z / (1.f + texp<T>(-z));
     st(yr, j, z);
   }
 }
@@ -256,7 +263,7 @@ __global__ void ln_act_fwd_small_kernel(const T* __restrict__ x, const TW* __res
     for (int j = lane; j < D; j += 64) {
       float xhat = (ld(xr, j) - mean) * rstd;
       float z = xhat * ld(w, j) + ld(b, j);
-      if (SILU) z = z / (1.f + expf(-z));
+      if (SILU) z = z / (1.f + texp<T>(-z));
       st(yr, j, z);
     }
   }
@@ -910,7 +917,7 @@ __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict_
         float gz = ld(gr, j);
         if (SILU) {
           float z = xhat * ld(w, j) + ld(b, j);
-          float sig = 1.f / (1.f + expf(-z));
+          float sig = 1.f / (1.f + texp<T>(-z));
           gz *= sig * (1.f + z * (1.f - sig));
         }
         cxh[cnt] = xhat;
@@ -950,7 +957,7 @@ __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict_
       float gz = ld(gr, j);
       if (SILU) {
         float z = xhat * ld(w, j) + ld(b, j);
-        float sig = 1.f / (1.f + expf(-z));
+        float sig = 1.f / (1.f + texp<T>(-z));
         gz *= sig * (1.f + z * (1.f - sig));
       }
       float gxhat = gz * ld(w, j);
@@ -965,7 +972,7 @@ __global__ void ln_act_bwd_kernel(const T* __restrict__ gy, const T* __restrict_
       float gz = ld(gr, j);
       if (SILU) {
         float z = xhat * ld(w, j) + ld(b, j);
-        float sig = 1.f / (1.f + expf(-z));
+        float sig = 1.f / (1.f + texp<T>(-z));
         gz *= sig * (1.f + z * (1.f - sig));
       }
       gw_acc[j] += gz * xhat;
@@ -1013,7 +1020,7 @@ __global__ void ln_act_bwd_small_kernel(const T* __restrict__ gy, const T* __res
       float gz = ld(gr, j);
       if (SILU) {
         float z = xhat * ld(w, j) + ld(b, j);
-        float sig = 1.f / (1.f + expf(-z));
+        float sig = 1.f / (1.f + texp<T>(-z));
         gz *= sig * (1.f + z * (1.f - sig));
       }
       float gxhat = gz * ld(w, j);
@@ -1031,7 +1038,7 @@ __global__ void ln_act_bwd_small_kernel(const T* __restrict__ gy, const T* __res
       float gz = ld(gr, j);
       if (SILU) {
         float z = xhat * ld(w, j) + ld(b, j);
-        float sig = 1.f / (1.f + expf(-z));
+        float sig = 1.f / (1.f + texp<T>(-z));
         gz *= sig * (1.f + z * (1.f - sig));
       }
       gw_acc[j] += gz * xhat;
@@ -1254,9 +1261,9 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
       float zr = ((cache[i] - mean) * rstd) * ld(w, j) + ld(b, j);
       float zc = ((cache[hstep + i] - mean) * rstd) * ld(w, H + j) + ld(b, H + j);
       float zu = ((cache[2 * hstep + i] - mean) * rstd) * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-      float r = 1.f / (1.f + expf(-zr));
-      float c = tanhf(r * zc);
-      float u = 1.f / (1.f + expf(-(zu - 1.f)));
+      float r = 1.f / (1.f + texp<T>(-zr));
+      float c = ttanh<T>(r * zc);
+      float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
       float hv = u * c + (1.f - u) * ld(hr, j);
       st(outr, j, hv);
       if (outr2) st(outr2, j, hv);
@@ -1283,9 +1290,9 @@ __global__ void gru_gates_fwd_kernel(const T* __restrict__ y, const T* __restric
     float zr = ((ld(yr, j) - mean) * rstd) * ld(w, j) + ld(b, j);
     float zc = ((ld(yr, H + j) - mean) * rstd) * ld(w, H + j) + ld(b, H + j);
     float zu = ((ld(yr, 2 * H + j) - mean) * rstd) * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-    float r = 1.f / (1.f + expf(-zr));
-    float c = tanhf(r * zc);
-    float u = 1.f / (1.f + expf(-(zu - 1.f)));
+    float r = 1.f / (1.f + texp<T>(-zr));
+    float c = ttanh<T>(r * zc);
+    float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
     float hv = u * c + (1.f - u) * ld(hr, j);
     st(outr, j, hv);
     if (outr2) st(outr2, j, hv);
@@ -1521,10 +1528,10 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
         float zr = xh_r * ld(w, j) + ld(b, j);
         float zc = xh_c * ld(w, H + j) + ld(b, H + j);
         float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-        float r = 1.f / (1.f + expf(-zr));
+        float r = 1.f / (1.f + texp<T>(-zr));
         float rc = r * zc;
-        float c = tanhf(rc);
-        float u = 1.f / (1.f + expf(-(zu - 1.f)));
+        float c = ttanh<T>(rc);
+        float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
         float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
         float gu = g * (c - ld(hr, j));
         float gc = g * u;
@@ -1585,10 +1592,10 @@ __global__ void gru_gates_bwd_kernel(const T* __restrict__ gh, const T* __restri
       float zr = xh_r * ld(w, j) + ld(b, j);
       float zc = xh_c * ld(w, H + j) + ld(b, H + j);
       float zu = xh_u * ld(w, 2 * H + j) + ld(b, 2 * H + j);
-      float r = 1.f / (1.f + expf(-zr));
+      float r = 1.f / (1.f + texp<T>(-zr));
       float rc = r * zc;
-      float c = tanhf(rc);
-      float u = 1.f / (1.f + expf(-(zu - 1.f)));
+      float c = ttanh<T>(rc);
+      float u = 1.f / (1.f + texp<T>(-(zu - 1.f)));
       float g = ld(ghr, j) + (gh2r ? ld(gh2r, j) : 0.f) + (gh3r ? ld(gh3r, j) : 0.f);
       float gu = g * (c - ld(hr, j));
       float gc = g * u;
@@ -1992,7 +1999,7 @@ __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __rest
   for (int off = 32; off > 0; off >>= 1) lmax = fmaxf(lmax, __shfl_xor(lmax, off, 64));
   // exp + sum
   float lsum = 0.f;
-  for (int j = lane; j < K; j += 64) lsum += expf(ld(L, j) - lmax);
+  for (int j = lane; j < K; j += 64) lsum += texp<T>(ld(L, j) - lmax);
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, 64);
   const float inv = 1.f / lsum;
@@ -2000,16 +2007,16 @@ __global__ void cat_st_fwd_kernel(const T* __restrict__ raw, const float* __rest
   float best = -1e30f;
   int best_j = 0;
   for (int j = lane; j < K; j += 64) {
-    float s = expf(ld(L, j) - lmax) * inv;
+    float s = texp<T>(ld(L, j) - lmax) * inv;
     float p = (1.f - unimix) * s + unimix / K;
-    float m = logf(p);
+    float m = tlog<T>(p);
     sr[j] = s;
     mr[j] = m;
     float score = m;
     if (SAMPLE) {
       float u = urand[row * (long)K + j];
-      float t = fmaxf(-logf(fmaxf(u, 1e-20f)), 1e-20f);
-      score += -logf(t);  // Gumbel(0,1) noise
+      float t = fmaxf(-tlog<T>(fmaxf(u, 1e-20f)), 1e-20f);
+      score += -tlog<T>(t);  // Gumbel(0,1) noise
     }
     if (score > best) {
       best = score;
@@ -2061,21 +2068,21 @@ __global__ void cat_st_resets_fwd_kernel(
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) lmax = fmaxf(lmax, __shfl_xor(lmax, off, 64));
   float lsum = 0.f;
-  for (int j = lane; j < K; j += 64) lsum += expf(ld(L, j) - lmax);
+  for (int j = lane; j < K; j += 64) lsum += texp<T>(ld(L, j) - lmax);
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) lsum += __shfl_xor(lsum, off, 64);
   const float inv = 1.f / lsum;
   float best = -1e30f;
   int best_j = 0;
   for (int j = lane; j < K; j += 64) {
-    float sv = expf(ld(L, j) - lmax) * inv;
+    float sv = texp<T>(ld(L, j) - lmax) * inv;
     float p = (1.f - unimix) * sv + unimix / K;
-    float m = logf(p);
+    float m = tlog<T>(p);
     sr[j] = sv;
     mr[j] = m;
     float u = urand[row * (long)K + j];
-    float t = fmaxf(-logf(fmaxf(u, 1e-20f)), 1e-20f);
-    float score = m - logf(t);
+    float t = fmaxf(-tlog<T>(fmaxf(u, 1e-20f)), 1e-20f);
+    float score = m - tlog<T>(t);
     if (score > best) {
       best = score;
       best_j = j;
