@@ -146,7 +146,7 @@ def run(steps: int = 400, out_path: str = "profiles/loss_parity.json") -> dict:
             batch = {k: v[0] for k, v in s.items()}
             _o.ema_update_(list(target_critic.parameters()), list(critic.parameters()), cfg.algo.critic.tau)
             if graphs and graphed is None and it == 3:
-                graphed = _capture_train_step(
+                graphed, _graph_metrics = _capture_train_step(
                     runtime, wm, actor, critic, target_critic, wo, ao, co,
                     batch, cfg, False, actions_dim, moments,
                 )
