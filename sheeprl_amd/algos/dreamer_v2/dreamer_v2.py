@@ -487,7 +487,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
                             device=device,
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
-                        use_eager = graphed_step is None or (cumulative_steps % 16 == 0)
+                        use_eager = graphed_step is None or (cumulative_steps % 64 == 0)  # every 64th step eager for metrics (DV3 reads the capture's static buffers instead)
                         if use_eager:
                             train(
                                 runtime, world_model, actor, critic, target_critic,

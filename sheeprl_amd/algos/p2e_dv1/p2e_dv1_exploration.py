@@ -399,7 +399,7 @@ def main(runtime: Runtime, cfg: Any) -> None:
                         )
                         batch = {k: v[0].to(device) for k, v in sample.items()}
                         _n_train_calls += 1
-                        if graphed_step is not None and _n_train_calls % 16 != 0:
+                        if graphed_step is not None and _n_train_calls % 64 != 0:  # every 64th step runs eager to feed metrics (~14x a replay; DV3 feeds metrics from the capture's static buffers instead)
                             graphed_step(batch)
                         else:
                             train(
