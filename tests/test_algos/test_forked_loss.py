@@ -125,3 +125,25 @@ def test_forked_loss_viewed_logits_returned():
     )
     assert out[6].shape == (T, B, STOCH, DISC)  # priors viewed
     assert out[7].shape == (T, B, STOCH, DISC)  # posteriors viewed
+
+
+def test_ppo_losses_eager_fallback_matches_components():
+    """On CPU ppo_losses returns exactly the separate policy/value/entropy
+    losses (the fused kernel is GPU-only)."""
+    from sheeprl_amd.algos.ppo.loss import entropy_loss, policy_loss, ppo_losses, value_loss
+
+    torch.manual_seed(5)
+    N = 64
+    lp_new = torch.randn(N, requires_grad=True)
+    lp_old = torch.randn(N)
+    adv = torch.randn(N)
+    v_new = torch.randn(N, 1, requires_grad=True)
+    v_old = torch.randn(N, 1)
+    ret = torch.randn(N, 1)
+    ent = torch.randn(N, requires_grad=True)
+    for clip_vloss in (False, True):
+        for reduction in ("mean", "sum"):
+            pg, vl, el = ppo_losses(lp_new, lp_old, adv, v_new, v_old, ret, ent, 0.2, clip_vloss, reduction)
+            torch.testing.assert_close(pg, policy_loss(lp_new, lp_old, adv, 0.2, reduction))
+            torch.testing.assert_close(vl, value_loss(v_new, v_old, ret, 0.2, clip_vloss, reduction))
+            torch.testing.assert_close(el, entropy_loss(ent, reduction))
